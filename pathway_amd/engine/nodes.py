@@ -1,0 +1,671 @@
+"""Engine dataflow nodes — the MI355X-native operator set.
+
+Each node mirrors one Graph-trait operator family of the reference
+(src/engine/graph.rs:651-1055 — see SURVEY.md §2.3) but is implemented as a
+GPU-columnar micro-batch transform over DeltaBatches:
+
+  InputNode            connector_table / static table
+  ExprMapNode          expression_table (select / with_columns)
+  FilterNode           filter_table
+  ReindexNode          reindex_table / with_id_from
+  GroupReduceNode      group_by_table + reducers
+  JoinNode             join_tables (inner/left/right/outer) + ix
+  ConcatNode           concat_tables
+  KeyedMergeNode       update_rows_table / update_cells_table
+  SemiJoinNode         intersect_tables / subtract_table / restrict
+  FlattenNode          flatten_table
+  CaptureNode          capture for debug / tests
+  OutputNode           output_table (sinks)
+
+State lives in sorted GPU arrangements (engine/state.py).  All per-batch
+work is torch ops (CPU & ROCm); hot paths are overridden by HIP kernels via
+pathway_amd.ops when running on gfx950.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Callable, Sequence
+
+import numpy as np
+import torch
+
+from pathway_amd.internals import dtype as dt
+from pathway_amd.internals.api import MASK64
+from pathway_amd.engine import hashing
+from pathway_amd.engine.batch import DeltaBatch, segment_starts
+from pathway_amd.engine.column import (
+    Column,
+    ObjectColumn,
+    PointerColumn,
+    StringColumn,
+    TensorColumn,
+    column_from_pylist,
+)
+from pathway_amd.engine.expression_eval import EvalContext, evaluate
+from pathway_amd.engine.reducers import REDUCERS, ReducerSpec
+from pathway_amd.engine.state import (
+    Arrangement,
+    consolidate_sorted,
+    lex_sort_words,
+    searchsorted_words,
+)
+
+_node_counter = [0]
+
+
+def _salt(tag: str, node_id: int) -> int:
+    from pathway_amd.internals.api import xxh64
+
+    return xxh64(f"{tag}:{node_id}".encode(), 7)
+
+
+class Node:
+    n_outputs = 1
+
+    def __init__(self, inputs: Sequence["Node"], device):
+        self.inputs = list(inputs)
+        self.device = device
+        self.node_id = _node_counter[0]
+        _node_counter[0] += 1
+        self.name: str | None = None
+
+    def step(self, time: int, inputs: list[DeltaBatch | None]) -> DeltaBatch | None:
+        raise NotImplementedError
+
+    def on_frontier(self, time: int) -> DeltaBatch | None:
+        """Called when the frontier passes `time` (temporal behaviors)."""
+        return None
+
+    def wants_frontier(self) -> bool:
+        return False
+
+    def reset(self) -> None:
+        """Clear mutable state before a fresh run (re-running a graph)."""
+
+
+# ------------------------------------------------------------------ utils --
+
+def batch_vhash(batch: DeltaBatch) -> tuple[torch.Tensor, torch.Tensor]:
+    """128-bit row-value hash of a batch (the arrangement tiebreaker)."""
+    n = len(batch)
+    parts = []
+    for name in batch.columns:
+        lo, hi = batch.columns[name].value_hash()
+        parts.append((lo.to(batch.device), hi.to(batch.device)))
+    if not parts:
+        z = torch.zeros(n, dtype=torch.int64, device=batch.device)
+        return z, z.clone()
+    return hashing.combine_value_hashes(parts)
+
+
+def consolidate_batch(batch: DeltaBatch) -> DeltaBatch | None:
+    """Sort by (key, vhash), sum diffs, drop zeros (consolidation.rs)."""
+    if batch is None or len(batch) == 0:
+        return None
+    v0, v1 = batch_vhash(batch)
+    words = [batch.keys[:, 0].contiguous(), batch.keys[:, 1].contiguous(), v0, v1]
+    perm = lex_sort_words(words)
+    words = [w.index_select(0, perm) for w in words]
+    weights = batch.diffs.index_select(0, perm)
+    cols = {n: c.take(perm) for n, c in batch.columns.items()}
+    out_words, out_w, out_cols = consolidate_sorted(words, weights, cols)
+    if out_w.shape[0] == 0:
+        return None
+    keys = torch.stack(out_words[:2], dim=1)
+    return DeltaBatch(keys, out_cols, out_w, batch.time)
+
+
+def unique_sorted_keys(keys: torch.Tensor) -> torch.Tensor:
+    words = [keys[:, 0].contiguous(), keys[:, 1].contiguous()]
+    perm = lex_sort_words(words)
+    sk = keys.index_select(0, perm)
+    starts = segment_starts(sk)
+    return sk.index_select(0, starts.nonzero(as_tuple=True)[0])
+
+
+def null_column(proto: Column, n: int, device, dtype: dt.DType | None = None) -> Column:
+    d = dtype or proto.dtype
+    if isinstance(proto, TensorColumn):
+        t = torch.zeros((n,), dtype=proto.tensor.dtype, device=device)
+        mask = torch.zeros((n,), dtype=torch.bool, device=device)
+        return TensorColumn(t, dt.Optional(dt.unoptionalize(d)), mask)
+    if isinstance(proto, StringColumn):
+        codes = torch.full((n,), -1, dtype=torch.int64, device=device)
+        return StringColumn(codes, proto.pool, dt.Optional(dt.STR))
+    arr = np.empty(n, dtype=object)
+    return ObjectColumn(arr, dt.Optional(dt.unoptionalize(d)))
+
+
+# ------------------------------------------------------------------ input --
+
+class InputNode(Node):
+    """Source of delta batches; wraps a Source with pull(time)."""
+
+    def __init__(self, source: Any, device):
+        super().__init__([], device)
+        self.source = source
+
+    def step(self, time: int, inputs: list[DeltaBatch | None]) -> DeltaBatch | None:
+        return self.source.pull(time, self.device)
+
+    def reset(self) -> None:
+        r = getattr(self.source, "reset", None)
+        if r is not None:
+            r()
+
+
+class ExprMapNode(Node):
+    """select / with_columns: evaluate expressions, keep keys."""
+
+    def __init__(self, input_node: Node, exprs: dict[str, Any], device):
+        super().__init__([input_node], device)
+        self.exprs = exprs
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        ctx = EvalContext(b.columns, b.keys, self.device)
+        out_cols = {name: evaluate(e, ctx) for name, e in self.exprs.items()}
+        return DeltaBatch(b.keys, out_cols, b.diffs, time)
+
+
+class FilterNode(Node):
+    def __init__(self, input_node: Node, pred_expr: Any, device):
+        super().__init__([input_node], device)
+        self.pred = pred_expr
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        ctx = EvalContext(b.columns, b.keys, self.device)
+        col = evaluate(self.pred, ctx)
+        if isinstance(col, TensorColumn):
+            mask = col.tensor.to(torch.bool)
+            if col.mask is not None:
+                mask = mask & col.mask
+        else:
+            mask = torch.tensor(
+                [bool(v) if v is not None else False for v in col.to_pylist()],
+                dtype=torch.bool,
+                device=b.device,
+            )
+        out = b.filter(mask)
+        return out if len(out) else None
+
+
+class ReindexNode(Node):
+    """Assign new row keys from a pointer expression (reindex/with_id_from)."""
+
+    def __init__(self, input_node: Node, key_expr: Any, device):
+        super().__init__([input_node], device)
+        self.key_expr = key_expr
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        ctx = EvalContext(b.columns, b.keys, self.device)
+        col = evaluate(self.key_expr, ctx)
+        assert isinstance(col, PointerColumn), "reindex needs a pointer expression"
+        return DeltaBatch(col.pairs, b.columns, b.diffs, time)
+
+
+class DeriveKeyNode(Node):
+    """Re-key with a salted hash of the existing key (concat_reindex etc.)."""
+
+    def __init__(self, input_node: Node, tag: str, device):
+        super().__init__([input_node], device)
+        self.salt = _salt(tag, self.node_id)
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        lo, hi = hashing.derive_key_words(
+            self.salt, [(b.keys[:, 0].contiguous(), b.keys[:, 1].contiguous())]
+        )
+        return DeltaBatch(torch.stack([lo, hi], dim=1), b.columns, b.diffs, time)
+
+
+class ConcatNode(Node):
+    def __init__(self, input_nodes: Sequence[Node], device):
+        super().__init__(input_nodes, device)
+
+    def step(self, time, inputs):
+        batches = [b for b in inputs if b is not None and len(b)]
+        if not batches:
+            return None
+        names = list(batches[0].columns.keys())
+        batches = [b.select_columns(names) for b in batches]
+        return DeltaBatch.concat(batches)
+
+
+# ----------------------------------------------------------------- reduce --
+
+class GroupReduceNode(Node):
+    """groupby().reduce() — incremental segmented reduce with retractions.
+
+    Mirrors DataflowGraphInner::group_by_table (dataflow.rs:3761-3868):
+    group key = hash of grouping values; per-reducer state merged per batch;
+    on change emits (old row, -1) and (new row, +1) at batch time.
+    """
+
+    def __init__(
+        self,
+        input_node: Node,
+        group_exprs: dict[str, Any],
+        reducer_calls: dict[str, tuple[str, list[Any], dict]],
+        device,
+        sort_by: Any | None = None,
+    ):
+        super().__init__([input_node], device)
+        self.group_exprs = group_exprs
+        self.reducer_calls = reducer_calls
+        self.sort_by = sort_by
+        self.key_expr = None  # groupby(id=...): group key IS this pointer expr
+        self.seq = 0  # arrival sequence for earliest/latest
+
+        self.group_store: Arrangement | None = None
+        # additive accumulators: sorted keys + acc tensors
+        self.add_keys: list[torch.Tensor] | None = None
+        self.add_accs: dict[str, torch.Tensor] = {}
+        self.multiset_store: Arrangement | None = None
+        self.multiset_colnames: list[str] = []
+        self._spec_cache: dict[str, ReducerSpec] = {}
+
+    def reset(self) -> None:
+        self.seq = 0
+        self.group_store = None
+        self.add_keys = None
+        self.add_accs = {}
+        self.multiset_store = None
+        self.multiset_colnames = []
+
+    # -- helpers --
+
+    def _specs(self) -> dict[str, tuple[ReducerSpec, list[Any], dict]]:
+        out = {}
+        for out_name, (rname, args, kwargs) in self.reducer_calls.items():
+            spec = REDUCERS.get(rname)
+            if spec is None:
+                raise NotImplementedError(f"reducer {rname}")
+            out[out_name] = (spec, args, kwargs)
+        return out
+
+    def _additive_names(self):
+        names = []
+        for out_name, (spec, args, kwargs) in self._specs().items():
+            if spec.family == "additive":
+                if spec.name == "avg":
+                    names += [f"{out_name}__sum", f"{out_name}__cnt"]
+                else:
+                    names.append(out_name)
+        return names
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        device = self.device
+        seq0 = self.seq
+        self.seq += len(b)
+        seq_col = TensorColumn(
+            torch.arange(seq0, seq0 + len(b), dtype=torch.int64, device=device), dt.INT
+        )
+        ctx = EvalContext(b.columns, b.keys, device, extra={"__seq__": seq_col})
+        # 1. evaluate grouping columns and compute group keys
+        gcols = {n: evaluate(e, ctx) for n, e in self.group_exprs.items()}
+        if self.key_expr is not None:
+            kc = evaluate(self.key_expr, ctx)
+            assert isinstance(kc, PointerColumn), "groupby(id=...) needs a pointer"
+            gkeys = kc.pairs
+        else:
+            parts = [
+                (lo.to(device), hi.to(device))
+                for lo, hi in (c.value_hash() for c in gcols.values())
+            ]
+            if parts:
+                glo, ghi = hashing.combine_value_hashes(parts)
+            else:
+                glo = torch.zeros(len(b), dtype=torch.int64, device=device)
+                ghi = glo.clone()
+            gkeys = torch.stack([glo, ghi], dim=1)
+
+        specs = self._specs()
+
+        # 2. evaluate reducer args
+        arg_cols: dict[str, list[Column]] = {}
+        for out_name, (spec, args, kwargs) in specs.items():
+            arg_cols[out_name] = [evaluate(a, ctx) for a in args]
+
+        # 3. affected keys
+        changed = unique_sorted_keys(gkeys)
+        cw = [changed[:, 0].contiguous(), changed[:, 1].contiguous()]
+
+        # 4. old output rows (pre-merge)
+        old_presence, old_cols = self._current_rows(changed, cw, specs)
+
+        # 5. merge states
+        self._merge_group_store(gkeys, gcols, b.diffs)
+        self._merge_additive(gkeys, arg_cols, b.diffs, specs)
+        self._merge_multiset(gkeys, arg_cols, b, specs)
+
+        # 6. new output rows (post-merge)
+        new_presence, new_cols = self._current_rows(changed, cw, specs)
+
+        # 7. emit
+        out_batches = []
+        if bool(old_presence.any()):
+            idx = old_presence.nonzero(as_tuple=True)[0]
+            out_batches.append(
+                DeltaBatch(
+                    changed.index_select(0, idx),
+                    {n: c.take(idx) for n, c in old_cols.items()},
+                    torch.full((idx.shape[0],), -1, dtype=torch.int64, device=device),
+                    time,
+                )
+            )
+        if bool(new_presence.any()):
+            idx = new_presence.nonzero(as_tuple=True)[0]
+            out_batches.append(
+                DeltaBatch(
+                    changed.index_select(0, idx),
+                    {n: c.take(idx) for n, c in new_cols.items()},
+                    torch.ones((idx.shape[0],), dtype=torch.int64, device=device),
+                    time,
+                )
+            )
+        if not out_batches:
+            return None
+        return consolidate_batch(DeltaBatch.concat(out_batches))
+
+    # -- state init --
+
+    def _ensure_states(self, gcols: dict[str, Column], specs):
+        if self.group_store is None:
+            self.group_store = Arrangement(self.device, gcols)
+        if self.add_keys is None:
+            z = torch.zeros((0,), dtype=torch.int64, device=self.device)
+            self.add_keys = [z, z.clone()]
+            self.add_accs = {"__w__": z.clone()}
+            for n in self._additive_names():
+                self.add_accs[n] = z.clone()
+
+    # -- additive state --
+
+    def _merge_additive(self, gkeys, arg_cols, diffs, specs):
+        device = self.device
+        n = gkeys.shape[0]
+        # build per-row acc contributions
+        contribs: dict[str, torch.Tensor] = {"__w__": diffs}
+        for out_name, (spec, args, kwargs) in specs.items():
+            if spec.family != "additive":
+                continue
+            if spec.name == "count":
+                contribs[out_name] = diffs
+            elif spec.name == "sum":
+                t = arg_cols[out_name][0]
+                assert isinstance(t, TensorColumn), "sum needs a numeric column"
+                v = t.tensor
+                if v.dtype == torch.bool:
+                    v = v.to(torch.int64)
+                if t.mask is not None:
+                    v = torch.where(t.mask, v, torch.zeros_like(v))
+                contribs[out_name] = v * diffs.to(v.dtype)
+            elif spec.name == "avg":
+                t = arg_cols[out_name][0]
+                assert isinstance(t, TensorColumn)
+                v = t.tensor.to(torch.float64)
+                if t.mask is not None:
+                    v = torch.where(t.mask, v, torch.zeros_like(v))
+                contribs[f"{out_name}__sum"] = v * diffs.to(torch.float64)
+                contribs[f"{out_name}__cnt"] = diffs
+        # segment-sum contributions by group key
+        words = [gkeys[:, 0].contiguous(), gkeys[:, 1].contiguous()]
+        perm = lex_sort_words(words)
+        swords = [w.index_select(0, perm) for w in words]
+        sk = torch.stack(swords, dim=1)
+        starts = segment_starts(sk)
+        seg = torch.cumsum(starts.to(torch.int64), 0) - 1
+        nseg = int(seg[-1]) + 1 if n else 0
+        ukeys = [w.index_select(0, starts.nonzero(as_tuple=True)[0]) for w in swords]
+        deltas: dict[str, torch.Tensor] = {}
+        for name, c in contribs.items():
+            sc = c.index_select(0, perm)
+            acc = torch.zeros(nseg, dtype=sc.dtype, device=device)
+            acc.index_add_(0, seg, sc)
+            deltas[name] = acc
+        # merge into state: concat + sort + consolidate-sum
+        if self.add_keys is None:
+            return
+        all_words = [torch.cat([s, d]) for s, d in zip(self.add_keys, ukeys)]
+        all_accs = {
+            name: torch.cat(
+                [self.add_accs[name].to(deltas.get(name, self.add_accs[name]).dtype)
+                 if name in deltas else self.add_accs[name],
+                 deltas[name]]
+            )
+            if name in deltas
+            else torch.cat(
+                [self.add_accs[name], torch.zeros(nseg, dtype=self.add_accs[name].dtype, device=device)]
+            )
+            for name in self.add_accs
+        }
+        perm2 = lex_sort_words(all_words)
+        all_words = [w.index_select(0, perm2) for w in all_words]
+        sk2 = torch.stack(all_words, dim=1)
+        starts2 = segment_starts(sk2)
+        seg2 = torch.cumsum(starts2.to(torch.int64), 0) - 1
+        nseg2 = int(seg2[-1]) + 1 if seg2.shape[0] else 0
+        first_idx = starts2.nonzero(as_tuple=True)[0]
+        merged: dict[str, torch.Tensor] = {}
+        for name, acc in all_accs.items():
+            sa = acc.index_select(0, perm2)
+            out = torch.zeros(nseg2, dtype=sa.dtype, device=device)
+            out.index_add_(0, seg2, sa)
+            merged[name] = out
+        keep = merged["__w__"] != 0
+        kidx = keep.nonzero(as_tuple=True)[0]
+        self.add_keys = [
+            w.index_select(0, first_idx).index_select(0, kidx) for w in all_words
+        ]
+        self.add_accs = {name: acc.index_select(0, kidx) for name, acc in merged.items()}
+
+    def _merge_group_store(self, gkeys, gcols, diffs):
+        self._ensure_states(gcols, None)
+        v0, v1 = self._gcols_vhash(gcols, gkeys.shape[0])
+        self.group_store.merge(gkeys, (v0, v1), diffs, gcols)
+
+    def _gcols_vhash(self, gcols: dict[str, Column], n: int):
+        parts = [
+            (lo.to(self.device), hi.to(self.device))
+            for lo, hi in (c.value_hash() for c in gcols.values())
+        ]
+        if not parts:
+            z = torch.zeros(n, dtype=torch.int64, device=self.device)
+            return z, z.clone()
+        return hashing.combine_value_hashes(parts)
+
+    def _merge_multiset(self, gkeys, arg_cols, b: DeltaBatch, specs):
+        ms_cols: dict[str, Column] = {}
+        for out_name, (spec, args, kwargs) in specs.items():
+            if spec.family != "multiset":
+                continue
+            for i, c in enumerate(arg_cols[out_name]):
+                ms_cols[f"{out_name}__{i}"] = c
+        if not ms_cols:
+            return
+        # include row identity so distinct input rows stay distinct
+        ms_cols["__rowkey__"] = PointerColumn(b.keys)
+        if self.multiset_store is None:
+            self.multiset_store = Arrangement(self.device, ms_cols)
+            self.multiset_colnames = list(ms_cols.keys())
+        parts = [
+            (lo.to(self.device), hi.to(self.device))
+            for lo, hi in (c.value_hash() for c in ms_cols.values())
+        ]
+        v0, v1 = hashing.combine_value_hashes(parts)
+        self.multiset_store.merge(gkeys, (v0, v1), b.diffs, ms_cols)
+
+    # -- reading current rows for a set of keys --
+
+    def _current_rows(self, changed: torch.Tensor, cw, specs):
+        device = self.device
+        nq = changed.shape[0]
+        presence = torch.zeros(nq, dtype=torch.bool, device=device)
+        cols: dict[str, Column] = {}
+
+        # group columns from group_store
+        if self.group_store is not None and len(self.group_store):
+            lo, hi = self.group_store.key_range(changed)
+            cnt = hi - lo
+            found = cnt > 0
+            pos = torch.where(found, lo, torch.zeros_like(lo))
+            presence = presence | (
+                found
+                & (
+                    self._gather_weights(self.group_store, pos, found) > 0
+                )
+            )
+            for name, c in self.group_store.columns.items():
+                proto = c
+                taken = c.take(pos.clamp(0, max(len(self.group_store) - 1, 0)))
+                cols[name] = _mask_missing(taken, found, device)
+        else:
+            for name in self.group_exprs:
+                cols[name] = None  # filled below if group_store empty
+        # additive reducer outputs
+        add_pos, add_found = self._additive_lookup(cw)
+        for out_name, (spec, args, kwargs) in specs.items():
+            if spec.family == "additive":
+                if spec.name == "avg":
+                    s = self._gather_acc(f"{out_name}__sum", add_pos, add_found)
+                    c = self._gather_acc(f"{out_name}__cnt", add_pos, add_found)
+                    vals = s.to(torch.float64) / c.clamp(min=1).to(torch.float64)
+                    cols[out_name] = TensorColumn(vals, dt.FLOAT)
+                else:
+                    acc = self._gather_acc(out_name, add_pos, add_found)
+                    if spec.name == "count":
+                        cols[out_name] = TensorColumn(acc.to(torch.int64), dt.INT)
+                    else:
+                        odt = dt.FLOAT if acc.dtype == torch.float64 else dt.INT
+                        cols[out_name] = TensorColumn(acc, odt)
+            elif spec.family == "multiset":
+                cols[out_name] = self._multiset_agg(out_name, spec, changed, nq)
+        # presence from additive weight if group store empty (no group cols)
+        if self.add_keys is not None and "__w__" in self.add_accs:
+            w = self._gather_acc("__w__", add_pos, add_found)
+            presence = presence | (w > 0)
+        # fill any missing group columns (empty store on first batch)
+        for name in list(cols.keys()):
+            if cols[name] is None:
+                cols[name] = column_from_pylist([None] * nq, dt.ANY, device)
+        return presence, cols
+
+    def _gather_weights(self, store: Arrangement, pos, found):
+        if len(store) == 0:
+            return torch.zeros_like(pos)
+        w = store.weights.index_select(0, pos.clamp(0, len(store) - 1))
+        return torch.where(found, w, torch.zeros_like(w))
+
+    def _additive_lookup(self, cw):
+        if self.add_keys is None or self.add_keys[0].shape[0] == 0:
+            z = torch.zeros(cw[0].shape[0], dtype=torch.int64, device=self.device)
+            return z, torch.zeros_like(z, dtype=torch.bool)
+        pos = searchsorted_words(self.add_keys, cw, side="left")
+        m = self.add_keys[0].shape[0]
+        safe = pos.clamp(0, m - 1)
+        found = (
+            (pos < m)
+            & (self.add_keys[0].index_select(0, safe) == cw[0])
+            & (self.add_keys[1].index_select(0, safe) == cw[1])
+        )
+        return safe, found
+
+    def _gather_acc(self, name, pos, found):
+        acc = self.add_accs.get(name)
+        if acc is None or acc.shape[0] == 0:
+            return torch.zeros(pos.shape[0], dtype=torch.int64, device=self.device)
+        v = acc.index_select(0, pos)
+        return torch.where(found, v, torch.zeros_like(v))
+
+    def _multiset_agg(self, out_name, spec, changed, nq) -> Column:
+        device = self.device
+        store = self.multiset_store
+        arg0 = f"{out_name}__0"
+        if store is None or len(store) == 0:
+            proto_dt = dt.ANY
+            return column_from_pylist([None] * nq, proto_dt, device)
+        lo, hi = store.key_range(changed)
+        rows, qidx = store.gather_ranges(lo, hi)
+        col0 = store.columns[arg0]
+        # GPU fast path for min/max over numeric columns
+        if (
+            spec.segment_agg is not None
+            and isinstance(col0, TensorColumn)
+            and col0.mask is None
+        ):
+            vals = col0.tensor.index_select(0, rows)
+            agg = spec.segment_agg(vals, qidx, nq)
+            out = agg
+            if col0.tensor.dtype == torch.int64:
+                out = torch.where(
+                    torch.isfinite(agg), agg, torch.zeros_like(agg)
+                ).to(torch.int64)
+                return TensorColumn(out, dt.INT)
+            return TensorColumn(torch.where(torch.isfinite(agg), agg, torch.zeros_like(agg)), dt.FLOAT)
+        # host path
+        narg = spec.n_args if spec.n_args > 0 else 1
+        arg_lists = []
+        for i in range(narg):
+            cname = f"{out_name}__{i}"
+            if cname in store.columns:
+                arg_lists.append(store.columns[cname].take(rows).to_pylist())
+        weights = store.weights.index_select(0, rows).cpu().tolist()
+        qidx_l = qidx.cpu().tolist()
+        per_group: list[list] = [[] for _ in range(nq)]
+        for j, q in enumerate(qidx_l):
+            tup = tuple(al[j] for al in arg_lists)
+            if len(tup) == 1:
+                tup = (tup[0], None)
+            per_group[q].append((tup, weights[j]))
+        out_vals = []
+        for rows_g in per_group:
+            rows_g = [(t, w) for t, w in rows_g if w > 0]
+            if not rows_g:
+                out_vals.append(None)
+                continue
+            try:
+                out_vals.append(spec.host_agg(rows_g))
+            except ValueError:
+                from pathway_amd.internals.api import ERROR
+
+                out_vals.append(ERROR)
+        col, _ = _build_from_values(out_vals, device)
+        return col
+
+
+def _build_from_values(vals, device):
+    from pathway_amd.engine.column import infer_and_build_column
+
+    return infer_and_build_column(vals, device)
+
+
+def _mask_missing(col: Column, found: torch.Tensor, device) -> Column:
+    """Null out positions where found == False."""
+    if bool(found.all()):
+        return col
+    if isinstance(col, TensorColumn):
+        mask = found & (col.mask if col.mask is not None else torch.ones_like(found))
+        return TensorColumn(col.tensor, col.dtype, mask)
+    if isinstance(col, StringColumn):
+        codes = torch.where(found, col.codes, torch.full_like(col.codes, -1))
+        return StringColumn(codes, col.pool, col.dtype)
+    if isinstance(col, PointerColumn):
+        return col  # pointer columns keep garbage at missing rows; filtered out
+    vals = col.to_pylist()
+    f = found.cpu().tolist()
+    out = [v if ok else None for v, ok in zip(vals, f)]
+    return ObjectColumn(np.array(out, dtype=object), col.dtype)

@@ -1,0 +1,278 @@
+"""Engine runtime: per-timestamp synchronous micro-batch execution.
+
+The MI355X-native replacement of the timely worker loop
+(reference dataflow.rs:7306-7500 run_with_new_dataflow_graph): instead of
+cooperative operator scheduling with progress gossip, the runtime advances
+one global timestamp at a time and pushes that timestamp's delta batches
+through the node DAG in topological order — on a GPU the entire wave is a
+sequence of device-wide kernels, so operator-level interleaving buys
+nothing; cross-worker progress is a tiny min-allreduce per step
+(progress is control-plane, O(operators) sized).
+"""
+
+from __future__ import annotations
+
+import itertools
+from typing import Any, Callable, Iterable
+
+import torch
+
+from pathway_amd.engine.batch import DeltaBatch
+from pathway_amd.engine.column import Column, column_from_pylist
+from pathway_amd.engine.nodes import InputNode, Node, consolidate_batch
+from pathway_amd.internals import dtype as dt
+from pathway_amd.internals.api import BasePointer, DataRow
+
+
+class Source:
+    """Produces delta batches at nondecreasing times."""
+
+    def next_time(self) -> int | None:
+        """Smallest time with pending data; None when exhausted."""
+        raise NotImplementedError
+
+    def pull(self, time: int, device) -> DeltaBatch | None:
+        raise NotImplementedError
+
+
+class StaticSource(Source):
+    """Fixed set of rows with explicit times (debug tables, static io)."""
+
+    def __init__(
+        self,
+        rows: list[tuple[BasePointer, list[Any], int, int]],
+        column_names: list[str],
+        dtypes: list[dt.DType],
+    ):
+        self.column_names = column_names
+        self.dtypes = dtypes
+        by_time: dict[int, list] = {}
+        for key, values, time, diff in rows:
+            by_time.setdefault(time, []).append((key, values, diff))
+        self.by_time = dict(sorted(by_time.items()))
+        self._pending = sorted(self.by_time.keys())
+
+    def reset(self) -> None:
+        self._pending = sorted(self.by_time.keys())
+
+    def next_time(self) -> int | None:
+        return self._pending[0] if self._pending else None
+
+    def pull(self, time: int, device) -> DeltaBatch | None:
+        if not self._pending or self._pending[0] != time:
+            return None
+        self._pending.pop(0)
+        rows = self.by_time[time]
+        keys = torch.tensor(
+            [list(k.as_signed_pair()) for k, _, _ in rows],
+            dtype=torch.int64,
+            device=device,
+        ).reshape(len(rows), 2)
+        diffs = torch.tensor([d for _, _, d in rows], dtype=torch.int64, device=device)
+        cols: dict[str, Column] = {}
+        for j, name in enumerate(self.column_names):
+            vals = [v[j] for _, v, _ in rows]
+            cols[name] = column_from_pylist(vals, self.dtypes[j], device)
+        return DeltaBatch(keys, cols, diffs, time)
+
+
+class CallbackSource(Source):
+    """Streaming source driven by a generator of (time, rows) batches."""
+
+    def __init__(self, gen: Iterable, column_names: list[str], dtypes: list[dt.DType]):
+        self.gen = iter(gen)
+        self.column_names = column_names
+        self.dtypes = dtypes
+        self._buffered: tuple[int, Any] | None = None
+        self._done = False
+        self._advance()
+
+    def _advance(self):
+        if self._done:
+            return
+        try:
+            self._buffered = next(self.gen)
+        except StopIteration:
+            self._buffered = None
+            self._done = True
+
+    def next_time(self) -> int | None:
+        return None if self._buffered is None else self._buffered[0]
+
+    def pull(self, time: int, device) -> DeltaBatch | None:
+        if self._buffered is None or self._buffered[0] != time:
+            return None
+        _, payload = self._buffered
+        self._advance()
+        if isinstance(payload, DeltaBatch):
+            return payload.to_device(device)
+        rows = payload
+        keys = torch.tensor(
+            [list(k.as_signed_pair()) for k, _, _ in rows],
+            dtype=torch.int64,
+            device=device,
+        ).reshape(len(rows), 2)
+        diffs = torch.tensor([d for _, _, d in rows], dtype=torch.int64, device=device)
+        cols: dict[str, Column] = {}
+        for j, name in enumerate(self.column_names):
+            vals = [v[j] for _, v, _ in rows]
+            cols[name] = column_from_pylist(vals, self.dtypes[j], device)
+        return DeltaBatch(keys, cols, diffs, time)
+
+
+class CaptureNode(Node):
+    """Collects the full update stream of a table (debug / tests / sinks)."""
+
+    def __init__(self, input_node: Node, device, column_names: list[str] | None = None):
+        super().__init__([input_node], device)
+        self.rows: list[DataRow] = []
+        self.column_names = column_names
+
+    def reset(self) -> None:
+        self.rows = []
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        b = consolidate_batch(b)
+        if b is None:
+            return None
+        for key, values, t, diff in b.rows():
+            step = 1 if diff > 0 else -1
+            for _ in range(abs(diff)):
+                self.rows.append(DataRow(key=key, values=values, time=t, diff=step))
+        return None
+
+
+class OutputNode(Node):
+    """Delivers consolidated batches to a writer callback (sinks)."""
+
+    def __init__(self, input_node: Node, writer: Callable[[DeltaBatch], None], device):
+        super().__init__([input_node], device)
+        self.writer = writer
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        b = consolidate_batch(b)
+        if b is not None:
+            self.writer(b)
+        return None
+
+    def flush(self, time: int):
+        fl = getattr(self.writer, "flush", None)
+        if fl is not None:
+            fl(time)
+
+
+class SubscribeNode(Node):
+    """io.subscribe: per-row on_change callback + on_time_end/on_end."""
+
+    def __init__(
+        self,
+        input_node: Node,
+        device,
+        on_change: Callable,
+        on_time_end: Callable | None = None,
+        on_end: Callable | None = None,
+    ):
+        super().__init__([input_node], device)
+        self.on_change = on_change
+        self.on_time_end = on_time_end
+        self.on_end = on_end
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        b = consolidate_batch(b)
+        if b is None:
+            return None
+        names = list(b.columns.keys())
+        for key, values, t, diff in b.rows():
+            row = dict(zip(names, values))
+            self.on_change(key=key, row=row, time=t, is_addition=diff > 0)
+        return None
+
+    def finish_time(self, time: int):
+        if self.on_time_end is not None:
+            self.on_time_end(time)
+
+    def finish(self):
+        if self.on_end is not None:
+            self.on_end()
+
+
+def topo_order(sinks: list[Node]) -> list[Node]:
+    seen: dict[int, Node] = {}
+    order: list[Node] = []
+
+    def visit(n: Node, stack: set[int]):
+        if id(n) in seen:
+            return
+        if id(n) in stack:
+            raise RuntimeError("cycle in dataflow graph")
+        stack.add(id(n))
+        for i in n.inputs:
+            visit(i, stack)
+        stack.discard(id(n))
+        seen[id(n)] = n
+        order.append(n)
+
+    for s in sinks:
+        visit(s, set())
+    return order
+
+
+class Runtime:
+    """Single-worker synchronous runtime; multi-GPU coordination lives in
+    parallel/exchange.py (each rank runs its own Runtime in lockstep)."""
+
+    def __init__(self, sinks: list[Node], device="cpu", comm=None):
+        self.device = torch.device(device)
+        self.nodes = topo_order(sinks)
+        self.sources = [n for n in self.nodes if isinstance(n, InputNode)]
+        self.sinks = sinks
+        self.comm = comm  # parallel context or None
+
+    def _next_time(self) -> int | None:
+        times = [s.source.next_time() for s in self.sources]
+        times = [t for t in times if t is not None]
+        local = min(times) if times else None
+        if self.comm is not None:
+            local = self.comm.allreduce_min_time(local)
+        return local
+
+    def step_once(self, time: int) -> None:
+        outputs: dict[int, DeltaBatch | None] = {}
+        for node in self.nodes:
+            if isinstance(node, InputNode):
+                out = node.step(time, [])
+            else:
+                ins = [outputs.get(id(i)) for i in node.inputs]
+                if all(b is None for b in ins) and not node.wants_frontier():
+                    out = None
+                else:
+                    out = node.step(time, ins)
+            outputs[id(node)] = out
+        for node in self.nodes:
+            if isinstance(node, SubscribeNode):
+                node.finish_time(time)
+            if isinstance(node, OutputNode):
+                node.flush(time)
+
+    def run(self, max_steps: int | None = None) -> None:
+        steps = 0
+        while True:
+            t = self._next_time()
+            if t is None:
+                break
+            self.step_once(t)
+            steps += 1
+            if max_steps is not None and steps >= max_steps:
+                break
+        for node in self.nodes:
+            if isinstance(node, SubscribeNode):
+                node.finish()

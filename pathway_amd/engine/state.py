@@ -1,0 +1,229 @@
+"""GPU-resident sorted arrangements — the engine's stateful data structure.
+
+The MI355X-native replacement for differential-dataflow's Spine trace
+(reference: spine_fueled.rs, merge_batcher.rs, ord.rs): instead of an LSM of
+immutable batches with fueled merging, state lives as ONE consolidated
+sorted columnar arrangement in HBM3E and every micro-batch merge is a full
+sorted merge — at ~8 TB/s a full rewrite of even a 10^8-row arrangement is
+sub-millisecond-scale, so the LSM's amortization trick buys nothing on this
+hardware and costs random access.
+
+Rows are ordered lexicographically by (key0, key1, vhash0, vhash1) where
+vhash is the 128-bit hash of the row's values — consolidation (diff
+summing, reference consolidation.rs) merges rows identical in key AND
+value, exactly differential's (data, diff) semantics.
+"""
+
+from __future__ import annotations
+
+from typing import Sequence
+
+import torch
+
+from pathway_amd.engine.batch import segmented_arange
+from pathway_amd.engine.column import Column, concat_columns
+
+
+def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
+    """Stable lexicographic argsort over parallel int64 word tensors."""
+    n = words[0].shape[0]
+    perm = torch.arange(n, dtype=torch.int64, device=words[0].device)
+    if n <= 1:
+        return perm
+    for w in reversed(words):
+        keys = w.index_select(0, perm)
+        p = torch.argsort(keys, stable=True)
+        perm = perm.index_select(0, p)
+    return perm
+
+
+def rows_ne(words: Sequence[torch.Tensor], i_prev_mask: bool = True) -> torch.Tensor:
+    """Mask marking rows that differ from their predecessor (run starts)."""
+    n = words[0].shape[0]
+    starts = torch.ones(n, dtype=torch.bool, device=words[0].device)
+    if n > 1:
+        ne = torch.zeros(n - 1, dtype=torch.bool, device=words[0].device)
+        for w in words:
+            ne |= w[1:] != w[:-1]
+        starts[1:] = ne
+    return starts
+
+
+def consolidate_sorted(
+    words: list[torch.Tensor],
+    weights: torch.Tensor,
+    columns: dict[str, Column],
+) -> tuple[list[torch.Tensor], torch.Tensor, dict[str, Column]]:
+    """Sum weights over equal-row runs, drop zero-weight rows.
+
+    Inputs must already be sorted by `words`.  Column values within a run are
+    identical by construction (vhash is part of the sort key), so the first
+    row of each run is kept.
+    """
+    n = weights.shape[0]
+    if n == 0:
+        return words, weights, columns
+    starts = rows_ne(words)
+    seg = torch.cumsum(starts.to(torch.int64), 0) - 1
+    nseg = int(seg[-1]) + 1
+    wsum = torch.zeros(nseg, dtype=torch.int64, device=weights.device)
+    wsum.index_add_(0, seg, weights)
+    first_idx = starts.nonzero(as_tuple=True)[0]
+    keep = wsum != 0
+    kept_first = first_idx.index_select(0, keep.nonzero(as_tuple=True)[0])
+    kept_w = wsum.index_select(0, keep.nonzero(as_tuple=True)[0])
+    out_words = [w.index_select(0, kept_first) for w in words]
+    out_cols = {name: c.take(kept_first) for name, c in columns.items()}
+    return out_words, kept_w, out_cols
+
+
+def searchsorted_words(
+    sorted_words: Sequence[torch.Tensor],
+    query_words: Sequence[torch.Tensor],
+    side: str = "left",
+) -> torch.Tensor:
+    """Vectorized multiword searchsorted (lexicographic) via binary search."""
+    m = sorted_words[0].shape[0]
+    nq = query_words[0].shape[0]
+    device = query_words[0].device
+    lo = torch.zeros(nq, dtype=torch.int64, device=device)
+    hi = torch.full((nq,), m, dtype=torch.int64, device=device)
+    if m == 0 or nq == 0:
+        return lo
+    it = max(1, m.bit_length() + 1)
+    for _ in range(it):
+        active = lo < hi
+        if not bool(active.any()):
+            break
+        mid = (lo + hi) >> 1
+        mid_safe = mid.clamp(0, m - 1)
+        # lexicographic compare sorted[mid] ? query
+        lt = torch.zeros(nq, dtype=torch.bool, device=device)
+        eq = torch.ones(nq, dtype=torch.bool, device=device)
+        for sw, qw in zip(sorted_words, query_words):
+            sv = sw.index_select(0, mid_safe)
+            lt = lt | (eq & (sv < qw))
+            eq = eq & (sv == qw)
+        if side == "left":
+            go_right = lt
+        else:
+            go_right = lt | eq
+        lo = torch.where(active & go_right, mid + 1, lo)
+        hi = torch.where(active & ~go_right, mid, hi)
+    return lo
+
+
+class Arrangement:
+    """Consolidated sorted multiset of weighted rows, keyed by a 128-bit key.
+
+    key_words: [k0, k1] (m,) int64 each; vhash_words: [v0, v1]; weights (m,)
+    int64 nonzero; columns: the row payload.
+    """
+
+    def __init__(self, device, column_protos: dict[str, Column]):
+        self.device = device
+        z = torch.zeros((0,), dtype=torch.int64, device=device)
+        self.key_words = [z, z.clone()]
+        self.vhash_words = [z.clone(), z.clone()]
+        self.weights = z.clone()
+        idx0 = torch.zeros((0,), dtype=torch.int64)
+        self.columns: dict[str, Column] = {
+            n: c.take(idx0.to(c._device())) for n, c in column_protos.items()
+        }
+
+    def __len__(self) -> int:
+        return int(self.weights.shape[0])
+
+    @property
+    def words(self) -> list[torch.Tensor]:
+        return self.key_words + self.vhash_words
+
+    def key_range(self, query_keys: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+        """[lo, hi) row ranges for (nq,2) query keys (by key prefix)."""
+        q = [query_keys[:, 0].contiguous(), query_keys[:, 1].contiguous()]
+        lo = searchsorted_words(self.key_words, q, side="left")
+        hi = searchsorted_words(self.key_words, q, side="right")
+        return lo, hi
+
+    def gather_ranges(
+        self, lo: torch.Tensor, hi: torch.Tensor
+    ) -> tuple[torch.Tensor, torch.Tensor]:
+        """Expand [lo,hi) ranges → (row_indices, query_index per row)."""
+        counts = hi - lo
+        total = int(counts.sum())
+        qidx = torch.repeat_interleave(
+            torch.arange(counts.shape[0], dtype=torch.int64, device=counts.device),
+            counts,
+        )
+        if total == 0:
+            return torch.zeros((0,), dtype=torch.int64, device=counts.device), qidx
+        offs = segmented_arange(counts)
+        rows = torch.repeat_interleave(lo, counts) + offs
+        return rows, qidx
+
+    def merge(
+        self,
+        keys: torch.Tensor,
+        vhash: tuple[torch.Tensor, torch.Tensor],
+        weights: torch.Tensor,
+        columns: dict[str, Column],
+    ) -> None:
+        """Merge a delta (unsorted ok) into the arrangement, consolidating."""
+        if keys.shape[0] == 0:
+            return
+        dwords = [keys[:, 0].contiguous(), keys[:, 1].contiguous(), vhash[0], vhash[1]]
+        all_words = [torch.cat([s, d]) for s, d in zip(self.words, dwords)]
+        all_w = torch.cat([self.weights, weights])
+        all_cols = {
+            n: concat_columns([self.columns[n], columns[n]]) for n in self.columns
+        }
+        perm = lex_sort_words(all_words)
+        all_words = [w.index_select(0, perm) for w in all_words]
+        all_w = all_w.index_select(0, perm)
+        all_cols = {n: c.take(perm) for n, c in all_cols.items()}
+        out_words, out_w, out_cols = consolidate_sorted(all_words, all_w, all_cols)
+        self.key_words = out_words[:2]
+        self.vhash_words = out_words[2:]
+        self.weights = out_w
+        self.columns = out_cols
+
+    def keys_tensor(self) -> torch.Tensor:
+        return torch.stack(self.key_words, dim=1) if len(self) else torch.zeros(
+            (0, 2), dtype=torch.int64, device=self.device
+        )
+
+
+class AdditiveState:
+    """key -> additive accumulator columns (semigroup reduce fast path).
+
+    Used for count/sum-style reducers (reference reduce.rs:163-560 semigroup
+    states): merging a delta ADDS accumulators; rows with all-zero presence
+    weight are dropped.  Sorted by key; single consolidated array.
+    """
+
+    def __init__(self, device, acc_names: list[str]):
+        z = torch.zeros((0,), dtype=torch.int64, device=device)
+        self.device = device
+        self.key_words = [z, z.clone()]
+        self.weights = z.clone()  # presence weight (count of underlying rows)
+        self.accs: dict[str, torch.Tensor] = {}
+        self.acc_names = acc_names
+
+    def __len__(self) -> int:
+        return int(self.weights.shape[0])
+
+    def lookup(self, keys: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+        """(positions, found_mask) of (nq,2) keys in the state."""
+        q = [keys[:, 0].contiguous(), keys[:, 1].contiguous()]
+        pos = searchsorted_words(self.key_words, q, side="left")
+        m = len(self)
+        safe = pos.clamp(0, max(m - 1, 0))
+        if m == 0:
+            found = torch.zeros(pos.shape[0], dtype=torch.bool, device=pos.device)
+        else:
+            found = (
+                (pos < m)
+                & (self.key_words[0].index_select(0, safe) == q[0])
+                & (self.key_words[1].index_select(0, safe) == q[1])
+            )
+        return pos, found

@@ -1,0 +1,67 @@
+"""Global run graph: registry of sinks + run()/reset machinery.
+
+The analog of the reference's ParseGraph G (internals/parse_graph.py:103)
+plus GraphRunner (graph_runner/__init__.py:40): tables carry their engine
+node eagerly; running = topo walk + Runtime over the registered sinks.
+"""
+
+from __future__ import annotations
+
+from typing import Any
+
+from pathway_amd.internals.config import get_device
+
+
+class RunGraph:
+    def __init__(self) -> None:
+        self.sinks: list[Any] = []  # engine sink nodes (Output/Subscribe/Capture)
+        self.error_log_tables: list[Any] = []
+        self.comm = None  # parallel context (set by parallel.init)
+
+    def add_sink(self, node: Any) -> None:
+        self.sinks.append(node)
+
+    def clear(self) -> None:
+        self.sinks.clear()
+        self.error_log_tables.clear()
+
+    def run_nodes(self, extra_sinks: list[Any] | None = None, monitoring=None, **kwargs):
+        from pathway_amd.engine.runtime import Runtime
+
+        sinks = list(self.sinks) + list(extra_sinks or [])
+        if not sinks:
+            return None
+        rt = Runtime(sinks, device=get_device(), comm=self.comm)
+        reset_all(rt.nodes)
+        rt.run()
+        return rt
+
+
+G = RunGraph()
+
+
+def reset_all(nodes) -> None:
+    for n in nodes:
+        reset = getattr(n, "reset", None)
+        if reset is not None:
+            reset()
+
+
+def run(
+    *,
+    debug: bool = False,
+    monitoring_level: Any = None,
+    with_http_server: bool = False,
+    default_logging: bool = True,
+    persistence_config: Any = None,
+    runtime_typechecking: bool | None = None,
+    license_key: str | None = None,
+    terminate_on_error: bool | None = None,
+    **kwargs: Any,
+):
+    """pw.run(): execute every registered output (reference internals/run.py:13)."""
+    return G.run_nodes(monitoring=monitoring_level)
+
+
+def run_all(**kwargs: Any):
+    return run(**kwargs)

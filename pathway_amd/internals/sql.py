@@ -1,0 +1,204 @@
+"""pw.sql — SQL to dataflow translation (reference internals/sql/, ~1500 LoC).
+
+Round-1: a compact translator covering SELECT / WHERE / GROUP BY / HAVING /
+simple JOIN ... ON / UNION ALL over registered tables.
+"""
+
+from __future__ import annotations
+
+import re
+from typing import Any
+
+from pathway_amd.internals import expression as ex
+from pathway_amd.internals.table import Table
+
+
+class _SqlError(ValueError):
+    pass
+
+
+_AGGS = {"count": "count", "sum": "sum", "avg": "avg", "min": "min", "max": "max"}
+
+
+def _tokenize_expr(s: str):
+    return re.findall(r"[A-Za-z_][A-Za-z0-9_.]*|\d+\.\d+|\d+|<>|<=|>=|!=|=|<|>|[(),*+\-/%]|'[^']*'", s)
+
+
+def _parse_scalar(sql_expr: str, table: Table) -> ex.ColumnExpression:
+    """Tiny recursive-descent expression parser over a single table scope."""
+    toks = _tokenize_expr(sql_expr)
+    pos = [0]
+
+    def peek():
+        return toks[pos[0]] if pos[0] < len(toks) else None
+
+    def eat(t=None):
+        tok = peek()
+        if t is not None and (tok is None or tok.upper() != t.upper()):
+            raise _SqlError(f"expected {t}, got {tok}")
+        pos[0] += 1
+        return tok
+
+    def atom():
+        tok = peek()
+        if tok is None:
+            raise _SqlError("unexpected end of expression")
+        if tok == "(":
+            eat()
+            e = or_expr()
+            eat(")")
+            return e
+        if re.fullmatch(r"\d+", tok):
+            eat()
+            return ex.ColumnConstExpression(int(tok))
+        if re.fullmatch(r"\d+\.\d+", tok):
+            eat()
+            return ex.ColumnConstExpression(float(tok))
+        if tok.startswith("'"):
+            eat()
+            return ex.ColumnConstExpression(tok[1:-1])
+        name = eat()
+        lname = name.lower()
+        if lname in _AGGS and peek() == "(":
+            eat("(")
+            if peek() == "*":
+                eat()
+                arg = None
+            else:
+                arg = or_expr()
+            eat(")")
+            from pathway_amd.internals.expression import ReducerExpression
+
+            if arg is None:
+                return ReducerExpression("count")
+            return ReducerExpression(_AGGS[lname], arg)
+        if lname in ("true", "false"):
+            return ex.ColumnConstExpression(lname == "true")
+        if lname == "null":
+            return ex.ColumnConstExpression(None)
+        col = name.split(".")[-1]
+        return ex.ColumnReference(table, col)
+
+    def mul_expr():
+        e = atom()
+        while peek() in ("*", "/", "%"):
+            op = eat()
+            r = atom()
+            e = {"*": e.__mul__, "/": e.__truediv__, "%": e.__mod__}[op](r)
+        return e
+
+    def add_expr():
+        e = mul_expr()
+        while peek() in ("+", "-"):
+            op = eat()
+            r = mul_expr()
+            e = (e + r) if op == "+" else (e - r)
+        return e
+
+    def cmp_expr():
+        e = add_expr()
+        if peek() in ("=", "!=", "<>", "<", "<=", ">", ">="):
+            op = eat()
+            r = add_expr()
+            return {
+                "=": e.__eq__,
+                "!=": e.__ne__,
+                "<>": e.__ne__,
+                "<": e.__lt__,
+                "<=": e.__le__,
+                ">": e.__gt__,
+                ">=": e.__ge__,
+            }[op](r)
+        return e
+
+    def not_expr():
+        if peek() is not None and peek().upper() == "NOT":
+            eat()
+            return ~not_expr()
+        return cmp_expr()
+
+    def and_expr():
+        e = not_expr()
+        while peek() is not None and peek().upper() == "AND":
+            eat()
+            e = e & not_expr()
+        return e
+
+    def or_expr():
+        e = and_expr()
+        while peek() is not None and peek().upper() == "OR":
+            eat()
+            e = e | and_expr()
+        return e
+
+    return or_expr()
+
+
+def sql(query: str, **tables: Table) -> Table:
+    q = query.strip().rstrip(";")
+    if re.search(r"\bUNION\s+ALL\b", q, re.I):
+        parts = re.split(r"\bUNION\s+ALL\b", q, flags=re.I)
+        result = sql(parts[0], **tables)
+        for p in parts[1:]:
+            result = result.concat_reindex(sql(p, **tables))
+        return result
+    m = re.match(
+        r"SELECT\s+(?P<sel>.*?)\s+FROM\s+(?P<from>[A-Za-z_][A-Za-z0-9_]*)"
+        r"(?:\s+WHERE\s+(?P<where>.*?))?"
+        r"(?:\s+GROUP\s+BY\s+(?P<gb>.*?))?"
+        r"(?:\s+HAVING\s+(?P<having>.*?))?$",
+        q,
+        re.I | re.S,
+    )
+    if not m:
+        raise _SqlError(f"unsupported SQL: {query!r}")
+    tname = m.group("from")
+    if tname not in tables:
+        raise _SqlError(f"unknown table {tname!r}")
+    t = tables[tname]
+    if m.group("where"):
+        t = t.filter(_parse_scalar(m.group("where"), t))
+    sel = m.group("sel").strip()
+    gb = m.group("gb")
+
+    def split_select(s: str) -> list[str]:
+        out, depth, cur = [], 0, []
+        for ch in s:
+            if ch == "(":
+                depth += 1
+            elif ch == ")":
+                depth -= 1
+            if ch == "," and depth == 0:
+                out.append("".join(cur).strip())
+                cur = []
+            else:
+                cur.append(ch)
+        if cur:
+            out.append("".join(cur).strip())
+        return out
+
+    items = split_select(sel)
+    named: dict[str, Any] = {}
+    for it in items:
+        if it == "*":
+            for c in t.column_names():
+                named[c] = ex.ColumnReference(t, c)
+            continue
+        mas = re.match(r"(.*?)\s+AS\s+([A-Za-z_][A-Za-z0-9_]*)$", it, re.I)
+        if mas:
+            expr_s, name = mas.group(1), mas.group(2)
+        else:
+            expr_s = it
+            name = re.sub(r"\W+", "_", it.split(".")[-1]).strip("_") or "col"
+        named[name] = _parse_scalar(expr_s, t)
+    if gb:
+        gcols = [c.strip().split(".")[-1] for c in gb.split(",")]
+        grouped = t.groupby(*[ex.ColumnReference(t, c) for c in gcols])
+        result = grouped.reduce(**named)
+        if m.group("having"):
+            result = result.filter(_parse_scalar(m.group("having"), result))
+        return result
+    has_agg = any(isinstance(e, ex.ReducerExpression) for e in named.values())
+    if has_agg:
+        return t.groupby().reduce(**named)
+    return t.select(**named)

@@ -1,0 +1,487 @@
+"""pw.Table — the user-facing API (reference internals/table.py:53-2760).
+
+Tables are handles on eagerly-built engine nodes; every method composes new
+nodes (see pathway_amd/engine/nodes*.py for operator semantics).  Deferred
+semantics are preserved: nothing computes until pw.run()/pw.debug.* drives
+the Runtime.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Iterable, Mapping, Sequence
+
+from pathway_amd.internals import dtype as dt
+from pathway_amd.internals import expression as ex
+from pathway_amd.internals import thisclass
+from pathway_amd.internals.config import get_device
+from pathway_amd.internals.schema import SchemaMetaclass, schema_from_types
+from pathway_amd.internals.type_inference import infer_dtype
+from pathway_amd.internals.universe import Universe
+
+
+class TableLike:
+    _universe: Universe
+
+
+def _substitute(expr: Any, mapping: dict[type, "Table"]) -> Any:
+    return thisclass.substitute_this(expr, mapping)
+
+
+def _check_refs(expr: ex.ColumnExpression, table: "Table") -> None:
+    for ref in expr._col_refs():
+        t = ref.table
+        if isinstance(t, type):
+            raise ValueError(f"unresolved this/left/right reference {ref!r}")
+        if t is None:
+            continue  # engine-internal reference
+        if isinstance(t, Table):
+            if t._node is table._node:
+                continue
+            if t._universe.is_equal(table._universe):
+                # same key-set: resolution by name against this table's batch
+                # is sound when the column exists here too
+                if ref.name == "id" or ref.name in table._dtypes:
+                    continue
+                raise NotImplementedError(
+                    "referencing columns of another (universe-equal) table in this "
+                    "operation is not supported yet; select them into one table first"
+                )
+            raise ValueError(
+                f"column {ref.name!r} of a table with a different universe used here"
+            )
+
+
+class Table(TableLike):
+    def __init__(self, node, dtypes: dict[str, dt.DType], universe: Universe):
+        self._node = node
+        self._dtypes = dict(dtypes)
+        self._universe = universe
+
+    # -- construction ------------------------------------------------------
+
+    @staticmethod
+    def _from_node(node, dtypes: dict[str, dt.DType], universe: Universe) -> "Table":
+        return Table(node, dtypes, universe)
+
+    @staticmethod
+    def empty(**kwargs: Any) -> "Table":
+        from pathway_amd.engine.nodes import InputNode
+        from pathway_amd.engine.runtime import StaticSource
+
+        dtypes = {n: dt.wrap(t) for n, t in kwargs.items()}
+        src = StaticSource([], list(dtypes.keys()), list(dtypes.values()))
+        node = InputNode(src, get_device())
+        return Table(node, dtypes, Universe())
+
+    # -- schema / columns --------------------------------------------------
+
+    @property
+    def schema(self) -> SchemaMetaclass:
+        return schema_from_types(**{n: t for n, t in self._dtypes.items()})
+
+    @property
+    def id(self) -> ex.ColumnReference:
+        return ex.ColumnReference(self, "id")
+
+    def column_names(self) -> list[str]:
+        return list(self._dtypes.keys())
+
+    def keys(self) -> list[str]:
+        return self.column_names()
+
+    def typehints(self) -> dict[str, Any]:
+        return {n: t.typehint for n, t in self._dtypes.items()}
+
+    def __getattr__(self, name: str) -> ex.ColumnReference:
+        if name.startswith("_"):
+            raise AttributeError(name)
+        if name not in self.__dict__.get("_dtypes", {}):
+            raise AttributeError(
+                f"table has no column {name!r}; columns: {list(self._dtypes)}"
+            )
+        return ex.ColumnReference(self, name)
+
+    def __getitem__(self, arg):
+        if isinstance(arg, str):
+            if arg == "id":
+                return self.id
+            if arg not in self._dtypes:
+                raise KeyError(arg)
+            return ex.ColumnReference(self, arg)
+        if isinstance(arg, ex.ColumnReference):
+            return self[arg.name]
+        if isinstance(arg, (list, tuple)):
+            return self.select(*[self[c] for c in arg])
+        raise TypeError(f"cannot index table with {arg!r}")
+
+    def __iter__(self):
+        raise TypeError("Table is not iterable; use pw.debug.compute_and_print")
+
+    def __repr__(self) -> str:
+        cols = ", ".join(f"{n}: {t!r}" for n, t in self._dtypes.items())
+        return f"<pathway_amd.Table {{{cols}}}>"
+
+    # -- core ops ----------------------------------------------------------
+
+    def _resolve(self, expr: Any) -> ex.ColumnExpression:
+        expr = ex.wrap_expr(expr)
+        expr = _substitute(expr, {thisclass.this: self})
+        _check_refs(expr, self)
+        return expr
+
+    def _named_exprs(
+        self, args: Sequence[Any], kwargs: Mapping[str, Any]
+    ) -> dict[str, ex.ColumnExpression]:
+        out: dict[str, ex.ColumnExpression] = {}
+        for a in args:
+            if isinstance(a, thisclass.ThisMetaclass):
+                raise TypeError("pass pw.this.column, not pw.this")
+            a = _substitute(ex.wrap_expr(a), {thisclass.this: self})
+            if isinstance(a, ex.ColumnReference):
+                if isinstance(a.table, Table) and a.table is not self and a.table._node is self._node:
+                    pass
+                out[a.name] = a
+            else:
+                raise ValueError(
+                    "positional select arguments must be column references; "
+                    "use keyword arguments for expressions"
+                )
+        for name, e in kwargs.items():
+            out[name] = _substitute(ex.wrap_expr(e), {thisclass.this: self})
+        for e in out.values():
+            _check_refs(e, self)
+        return out
+
+    def select(self, *args: Any, **kwargs: Any) -> "Table":
+        from pathway_amd.engine.nodes import ExprMapNode
+
+        exprs = self._named_exprs(args, kwargs)
+        node = ExprMapNode(self._node, exprs, get_device())
+        dtypes = {n: infer_dtype(e, self._dtypes) for n, e in exprs.items()}
+        return Table(node, dtypes, self._universe)
+
+    def with_columns(self, *args: Any, **kwargs: Any) -> "Table":
+        new = self._named_exprs(args, kwargs)
+        exprs: dict[str, ex.ColumnExpression] = {
+            n: ex.ColumnReference(self, n) for n in self._dtypes
+        }
+        exprs.update(new)
+        from pathway_amd.engine.nodes import ExprMapNode
+
+        node = ExprMapNode(self._node, exprs, get_device())
+        dtypes = {n: infer_dtype(e, self._dtypes) for n, e in exprs.items()}
+        return Table(node, dtypes, self._universe)
+
+    def filter(self, filter_expression: Any) -> "Table":
+        from pathway_amd.engine.nodes import FilterNode
+
+        pred = self._resolve(filter_expression)
+        node = FilterNode(self._node, pred, get_device())
+        return Table(node, self._dtypes, self._universe.subuniverse())
+
+    def split(self, split_expression: Any) -> tuple["Table", "Table"]:
+        pos = self.filter(split_expression)
+        neg = self.filter(~ex.wrap_expr(self._resolve(split_expression)))
+        return pos, neg
+
+    def copy(self) -> "Table":
+        return Table(self._node, self._dtypes, self._universe)
+
+    # -- renames / drops ---------------------------------------------------
+
+    def rename_columns(self, **kwargs: str) -> "Table":
+        mapping = {}
+        for new, old in kwargs.items():
+            old_name = old.name if isinstance(old, ex.ColumnReference) else old
+            mapping[old_name] = new
+        exprs = {}
+        for n in self._dtypes:
+            exprs[mapping.get(n, n)] = ex.ColumnReference(self, n)
+        return self.select(**exprs)
+
+    def rename_by_dict(self, names_mapping: Mapping[Any, str]) -> "Table":
+        mapping = {
+            (k.name if isinstance(k, ex.ColumnReference) else k): v
+            for k, v in names_mapping.items()
+        }
+        exprs = {}
+        for n in self._dtypes:
+            exprs[mapping.get(n, n)] = ex.ColumnReference(self, n)
+        return self.select(**exprs)
+
+    def rename(self, names_mapping: Mapping[Any, str] | None = None, **kwargs: str) -> "Table":
+        if names_mapping is not None:
+            return self.rename_by_dict(names_mapping)
+        return self.rename_columns(**kwargs)
+
+    def without(self, *columns: Any) -> "Table":
+        drop = {c.name if isinstance(c, ex.ColumnReference) else c for c in columns}
+        exprs = {
+            n: ex.ColumnReference(self, n) for n in self._dtypes if n not in drop
+        }
+        return self.select(**exprs)
+
+    # -- typing ------------------------------------------------------------
+
+    def update_types(self, **kwargs: Any) -> "Table":
+        exprs = {}
+        for n in self._dtypes:
+            if n in kwargs:
+                exprs[n] = ex.DeclareTypeExpression(ex.ColumnReference(self, n), kwargs[n])
+            else:
+                exprs[n] = ex.ColumnReference(self, n)
+        return self.select(**exprs)
+
+    def cast_to_types(self, **kwargs: Any) -> "Table":
+        exprs = {}
+        for n in self._dtypes:
+            if n in kwargs:
+                exprs[n] = ex.CastExpression(ex.ColumnReference(self, n), kwargs[n])
+            else:
+                exprs[n] = ex.ColumnReference(self, n)
+        return self.select(**exprs)
+
+    # -- groupby / reduce ---------------------------------------------------
+
+    def groupby(
+        self,
+        *args: Any,
+        id: Any = None,
+        sort_by: Any = None,
+        _filter_out_results_of_forgetting: bool = False,
+        instance: Any = None,
+        **kwargs: Any,
+    ):
+        from pathway_amd.internals.groupbys import GroupedTable
+
+        gb: list[ex.ColumnReference] = []
+        for a in args:
+            a = _substitute(ex.wrap_expr(a), {thisclass.this: self})
+            if not isinstance(a, ex.ColumnReference):
+                raise ValueError("groupby arguments must be column references")
+            gb.append(a)
+        if id is not None:
+            idref = _substitute(ex.wrap_expr(id), {thisclass.this: self})
+            return GroupedTable(self, gb, instance=instance, sort_by=sort_by, by_id=idref)
+        return GroupedTable(self, gb, instance=instance, sort_by=sort_by)
+
+    def reduce(self, *args: Any, **kwargs: Any) -> "Table":
+        return self.groupby().reduce(*args, **kwargs)
+
+    def deduplicate(
+        self,
+        *,
+        value: Any,
+        instance: Any = None,
+        acceptor: Any = None,
+        persistent_id: str | None = None,
+        name: str | None = None,
+    ) -> "Table":
+        from pathway_amd.engine.nodes_dedup import DeduplicateNode
+
+        vexpr = self._resolve(value)
+        iexpr = self._resolve(instance) if instance is not None else None
+        node = DeduplicateNode(self._node, vexpr, iexpr, acceptor, list(self._dtypes), get_device())
+        return Table(node, self._dtypes, Universe())
+
+    # -- joins --------------------------------------------------------------
+
+    def join(self, other: "Table", *on: Any, id: Any = None, how: Any = None, **kwargs: Any):
+        from pathway_amd.internals.joins import JoinMode, JoinResult
+
+        mode = how if how is not None else JoinMode.INNER
+        return JoinResult(self, other, list(on), mode, assign_id=id, **kwargs)
+
+    def join_inner(self, other: "Table", *on: Any, id: Any = None, **kwargs: Any):
+        from pathway_amd.internals.joins import JoinMode, JoinResult
+
+        return JoinResult(self, other, list(on), JoinMode.INNER, assign_id=id, **kwargs)
+
+    def join_left(self, other: "Table", *on: Any, id: Any = None, **kwargs: Any):
+        from pathway_amd.internals.joins import JoinMode, JoinResult
+
+        return JoinResult(self, other, list(on), JoinMode.LEFT, assign_id=id, **kwargs)
+
+    def join_right(self, other: "Table", *on: Any, id: Any = None, **kwargs: Any):
+        from pathway_amd.internals.joins import JoinMode, JoinResult
+
+        return JoinResult(self, other, list(on), JoinMode.RIGHT, assign_id=id, **kwargs)
+
+    def join_outer(self, other: "Table", *on: Any, id: Any = None, **kwargs: Any):
+        from pathway_amd.internals.joins import JoinMode, JoinResult
+
+        return JoinResult(self, other, list(on), JoinMode.OUTER, assign_id=id, **kwargs)
+
+    # -- set ops -------------------------------------------------------------
+
+    def concat(self, *others: "Table") -> "Table":
+        from pathway_amd.engine.nodes import ConcatNode
+
+        tables = [self, *others]
+        names = list(self._dtypes.keys())
+        for t in others:
+            if list(t._dtypes.keys()) != names:
+                t_cols = set(t._dtypes)
+                if t_cols != set(names):
+                    raise ValueError("concat requires identical column sets")
+        node = ConcatNode([t._node for t in tables], get_device())
+        dtypes = {
+            n: dt.types_lca(
+                self._dtypes[n],
+                others[0]._dtypes[n] if others else self._dtypes[n],
+            )
+            if others
+            else self._dtypes[n]
+            for n in names
+        }
+        return Table(node, dtypes, Universe())
+
+    def concat_reindex(self, *others: "Table") -> "Table":
+        from pathway_amd.engine.nodes import ConcatNode, DeriveKeyNode
+
+        tables = [self, *others]
+        renamed = [DeriveKeyNode(t._node, f"concat_part{i}", get_device()) for i, t in enumerate(tables)]
+        node = ConcatNode(renamed, get_device())
+        return Table(node, self._dtypes, Universe())
+
+    def update_rows(self, other: "Table") -> "Table":
+        from pathway_amd.engine.nodes_join import KeyedMergeNode
+
+        node = KeyedMergeNode(self._node, other._node, "rows", list(self._dtypes), get_device())
+        dtypes = {
+            n: dt.types_lca(self._dtypes[n], other._dtypes.get(n, self._dtypes[n]))
+            for n in self._dtypes
+        }
+        return Table(node, dtypes, Universe())
+
+    def update_cells(self, other: "Table") -> "Table":
+        from pathway_amd.engine.nodes_join import KeyedMergeNode
+
+        override = [n for n in other._dtypes if n in self._dtypes]
+        node = KeyedMergeNode(self._node, other._node, "cells", override, get_device())
+        return Table(node, self._dtypes, self._universe)
+
+    def __lshift__(self, other: "Table") -> "Table":
+        return self.update_cells(other)
+
+    def intersect(self, *tables: "Table") -> "Table":
+        from pathway_amd.engine.nodes_join import SemiJoinNode
+
+        node = self._node
+        out = self
+        for t in tables:
+            n = SemiJoinNode(out._node, t._node, "intersect", get_device())
+            out = Table(n, out._dtypes, out._universe.subuniverse())
+        return out
+
+    def difference(self, other: "Table") -> "Table":
+        from pathway_amd.engine.nodes_join import SemiJoinNode
+
+        node = SemiJoinNode(self._node, other._node, "difference", get_device())
+        return Table(node, self._dtypes, self._universe.subuniverse())
+
+    def restrict(self, other: TableLike) -> "Table":
+        from pathway_amd.engine.nodes_join import SemiJoinNode
+
+        node = SemiJoinNode(self._node, other._node, "intersect", get_device())  # type: ignore[attr-defined]
+        return Table(node, self._dtypes, other._universe)
+
+    def having(self, *indexers: Any) -> "Table":
+        out = self
+        for ixr in indexers:
+            ixr = self._resolve(ixr) if not isinstance(ixr, Table) else ixr
+            if isinstance(ixr, ex.ColumnReference) and isinstance(ixr.table, Table):
+                keyed = ixr.table.select(_pw_key=ixr).with_id_from_expr(
+                    ex.ColumnReference(ixr.table, ixr.name)
+                )
+                from pathway_amd.engine.nodes_join import SemiJoinNode
+
+                node = SemiJoinNode(out._node, keyed._node, "intersect", get_device())
+                out = Table(node, out._dtypes, out._universe.subuniverse())
+            else:
+                raise TypeError("having() expects column references")
+        return out
+
+    def with_universe_of(self, other: TableLike) -> "Table":
+        self._universe.promise_equal(other._universe)
+        return Table(self._node, self._dtypes, other._universe)
+
+    # -- keys ---------------------------------------------------------------
+
+    def pointer_from(self, *args: Any, optional: bool = False, instance: Any = None):
+        rargs = [self._resolve(a) for a in args]
+        ri = self._resolve(instance) if instance is not None else None
+        return ex.PointerExpression(self, *rargs, optional=optional, instance=ri)
+
+    def with_id_from(self, *args: Any, instance: Any = None) -> "Table":
+        key_expr = self.pointer_from(*args, instance=instance)
+        return self.with_id_from_expr(key_expr)
+
+    def with_id_from_expr(self, key_expr: ex.ColumnExpression) -> "Table":
+        from pathway_amd.engine.nodes import ReindexNode
+
+        node = ReindexNode(self._node, key_expr, get_device())
+        return Table(node, self._dtypes, Universe())
+
+    def with_id(self, new_index: ex.ColumnExpression) -> "Table":
+        return self.with_id_from_expr(self._resolve(new_index))
+
+    # -- flatten / ix -------------------------------------------------------
+
+    def flatten(self, to_flatten: Any, *, origin_id: str | None = None) -> "Table":
+        from pathway_amd.engine.nodes_join import FlattenNode
+
+        ref = self._resolve(to_flatten)
+        if not isinstance(ref, ex.ColumnReference):
+            raise ValueError("flatten expects a column reference")
+        node = FlattenNode(self._node, ref.name, get_device())
+        dtypes = dict(self._dtypes)
+        inner = self._dtypes.get(ref.name, dt.ANY)
+        if isinstance(inner, dt.List):
+            dtypes[ref.name] = inner.wrapped
+        elif dt.unoptionalize(inner) == dt.STR:
+            dtypes[ref.name] = dt.STR
+        else:
+            dtypes[ref.name] = dt.ANY
+        return Table(node, dtypes, Universe())
+
+    def ix(self, expression: Any, *, optional: bool = False, context=None) -> "Table":
+        """Row lookup by pointer (reference table.py ix/ix_ref)."""
+        from pathway_amd.internals.joins import make_ix_table
+
+        if isinstance(expression, ex.ColumnReference) and isinstance(
+            expression.table, Table
+        ):
+            query_table = expression.table
+            pexpr = expression
+        else:
+            query_table = context if isinstance(context, Table) else self
+            pexpr = expression
+        return make_ix_table(query_table, self, pexpr, optional=optional)
+
+    def ix_ref(self, *args: Any, optional: bool = False, context=None, instance=None) -> "Table":
+        query = context if isinstance(context, Table) else self
+        pexpr = query.pointer_from(*args, optional=optional, instance=instance)
+        from pathway_amd.internals.joins import make_ix_table
+
+        return make_ix_table(query, self, pexpr, optional=optional)
+
+    # -- misc ---------------------------------------------------------------
+
+    def remove_errors(self) -> "Table":
+        return self  # Error rows are filtered at output in this implementation
+
+    def await_futures(self) -> "Table":
+        return self
+
+    def debug(self, name: str) -> "Table":
+        return self
+
+    def to(self, sink) -> None:
+        sink.write(self)
+
+    def _capture(self):
+        """Attach a CaptureNode (tests / debug)."""
+        from pathway_amd.engine.runtime import CaptureNode
+
+        return CaptureNode(self._node, get_device(), list(self._dtypes.keys()))

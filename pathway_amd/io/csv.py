@@ -1,0 +1,83 @@
+"""pw.io.csv (reference io/csv)."""
+from __future__ import annotations
+
+import csv as _csv
+import os
+from typing import Any
+
+from pathway_amd.internals import dtype as dt
+
+
+def read(
+    path: str,
+    *,
+    schema=None,
+    mode: str = "streaming",
+    csv_settings: Any = None,
+    autocommit_duration_ms: int | None = 1500,
+    name: str | None = None,
+    **kwargs: Any,
+):
+    from pathway_amd.debug import table_from_rows
+    from pathway_amd.internals.schema import schema_from_csv
+
+    files = []
+    if os.path.isdir(path):
+        for root, _, fnames in os.walk(path):
+            for f in sorted(fnames):
+                files.append(os.path.join(root, f))
+    else:
+        files = [path]
+    if schema is None:
+        schema = schema_from_csv(files[0])
+    names = schema.column_names()
+    rows = []
+    for f in files:
+        with open(f, newline="") as fh:
+            reader = _csv.DictReader(fh)
+            for rec in reader:
+                row = []
+                for n in names:
+                    v = rec.get(n)
+                    d = dt.unoptionalize(schema.__columns__[n].dtype)
+                    if v is None or v == "":
+                        row.append(None)
+                    elif d == dt.INT:
+                        row.append(int(v))
+                    elif d == dt.FLOAT:
+                        row.append(float(v))
+                    elif d == dt.BOOL:
+                        row.append(v.lower() in ("true", "1"))
+                    else:
+                        row.append(v)
+                rows.append(tuple(row))
+    return table_from_rows(schema, rows)
+
+
+class CsvWriter:
+    def __init__(self, filename: str, column_names: list[str]):
+        self.filename = filename
+        self.column_names = column_names
+        self._fh = open(filename, "w", newline="")
+        self._writer = _csv.writer(self._fh)
+        self._writer.writerow(column_names + ["time", "diff"])
+
+    def __call__(self, batch):
+        names = list(batch.columns.keys())
+        for key, values, time, diff in batch.rows():
+            self._writer.writerow(list(values) + [time, diff])
+        self._fh.flush()
+
+    def flush(self, time):
+        self._fh.flush()
+
+
+def write(table, filename: str, *, name: str | None = None, **kwargs):
+    from pathway_amd.engine.runtime import OutputNode
+    from pathway_amd.internals.config import get_device
+    from pathway_amd.internals.rungraph import G
+
+    writer = CsvWriter(filename, table.column_names())
+    node = OutputNode(table._node, writer, get_device())
+    G.add_sink(node)
+    return node

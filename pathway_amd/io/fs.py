@@ -1,0 +1,74 @@
+"""pw.io.fs (reference io/fs) — static mode reads; streaming mode lands with
+the connector-runtime phase."""
+from __future__ import annotations
+
+import json
+import os
+from typing import Any
+
+from pathway_amd.internals import dtype as dt
+
+
+def read(
+    path: str,
+    *,
+    format: str = "plaintext",
+    schema=None,
+    mode: str = "streaming",
+    with_metadata: bool = False,
+    autocommit_duration_ms: int | None = 1500,
+    name: str | None = None,
+    **kwargs: Any,
+):
+    from pathway_amd.io import csv as io_csv, jsonlines as io_jsonlines, plaintext as io_plaintext
+
+    if format in ("csv",):
+        return io_csv.read(path, schema=schema, mode=mode, name=name, **kwargs)
+    if format in ("json", "jsonlines"):
+        return io_jsonlines.read(path, schema=schema, mode=mode, name=name, **kwargs)
+    if format in ("plaintext", "plaintext_by_file"):
+        return io_plaintext.read(path, mode=mode, name=name, **kwargs)
+    if format == "binary":
+        return _read_binary(path, with_metadata=with_metadata)
+    raise ValueError(f"unknown format {format!r}")
+
+
+def _read_binary(path: str, with_metadata: bool = False):
+    from pathway_amd.debug import table_from_rows
+    from pathway_amd.internals.schema import schema_from_types
+
+    files = []
+    if os.path.isdir(path):
+        for root, _, fnames in os.walk(path):
+            for f in sorted(fnames):
+                files.append(os.path.join(root, f))
+    else:
+        files = [path]
+    rows = []
+    for f in files:
+        with open(f, "rb") as fh:
+            data = fh.read()
+        if with_metadata:
+            from pathway_amd.internals.json import Json
+
+            meta = Json({"path": f, "size": len(data), "seen_at": 0,
+                         "modified_at": int(os.path.getmtime(f)),
+                         "owner": "unknown"})
+            rows.append((data, meta))
+        else:
+            rows.append((data,))
+    if with_metadata:
+        schema = schema_from_types(data=bytes, _metadata=dt.JSON)
+    else:
+        schema = schema_from_types(data=bytes)
+    return table_from_rows(schema, rows)
+
+
+def write(table, filename: str, *, format: str = "csv", name: str | None = None, **kwargs):
+    from pathway_amd.io import csv as io_csv, jsonlines as io_jsonlines
+
+    if format == "csv":
+        return io_csv.write(table, filename, name=name)
+    if format in ("json", "jsonlines"):
+        return io_jsonlines.write(table, filename, name=name)
+    raise ValueError(f"unknown format {format!r}")

@@ -1,0 +1,216 @@
+"""Windows + windowby (reference stdlib/temporal/_window.py:39-880;
+engine window assignment: src/engine/dataflow/windows.rs:166).
+
+Window assignment is expression-level (tensor ops on device for numeric
+time columns; host apply for datetimes), followed by the standard
+GroupReduce — so windowed aggregation runs on the same GPU segmented-reduce
+path as plain groupby.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Any
+
+from pathway_amd.internals import dtype as dt
+from pathway_amd.internals import expression as ex
+from pathway_amd.internals import thisclass
+from pathway_amd.internals.table import Table
+
+
+class Window:
+    pass
+
+
+@dataclass
+class TumblingWindow(Window):
+    duration: Any
+    origin: Any = None
+    shift: Any = None
+
+
+@dataclass
+class SlidingWindow(Window):
+    hop: Any
+    duration: Any = None
+    ratio: int | None = None
+    origin: Any = None
+
+
+@dataclass
+class SessionWindow(Window):
+    predicate: Any = None
+    max_gap: Any = None
+
+
+@dataclass
+class IntervalsOverWindow(Window):
+    at: Any = None
+    lower_bound: Any = None
+    upper_bound: Any = None
+    is_outer: bool = True
+
+
+def tumbling(duration: Any = None, origin: Any = None, shift: Any = None) -> TumblingWindow:
+    if duration is None and shift is not None:
+        duration = shift
+    return TumblingWindow(duration, origin, shift)
+
+
+def sliding(hop: Any, duration: Any = None, ratio: int | None = None, origin: Any = None) -> SlidingWindow:
+    return SlidingWindow(hop, duration, ratio, origin)
+
+
+def session(predicate: Any = None, max_gap: Any = None) -> SessionWindow:
+    return SessionWindow(predicate, max_gap)
+
+
+def intervals_over(
+    *, at: Any = None, lower_bound: Any = None, upper_bound: Any = None, is_outer: bool = True
+) -> IntervalsOverWindow:
+    return IntervalsOverWindow(at, lower_bound, upper_bound, is_outer)
+
+
+def _is_numeric(v: Any) -> bool:
+    return isinstance(v, (int, float)) and not isinstance(v, bool)
+
+
+def _as_number(v: Any):
+    import pandas as pd
+
+    if isinstance(v, pd.Timedelta):
+        return int(v.value)
+    if isinstance(v, pd.Timestamp):
+        return int(v.value)
+    return v
+
+
+class WindowedTable:
+    """Result of windowby; reduce() completes the windowed aggregation."""
+
+    def __init__(self, table: Table, time_expr, window: Window, behavior=None, instance=None):
+        self._table = table
+        self._time_expr = table._resolve(time_expr)
+        self._window = window
+        self._behavior = behavior
+        self._instance = table._resolve(instance) if instance is not None else None
+
+    def _prepare(self) -> tuple[Table, list[str]]:
+        """Assign windows: return (table with _pw_window_{start,end}, gb names)."""
+        t = self._table
+        w = self._window
+        texpr = self._time_expr
+        if isinstance(w, TumblingWindow):
+            d = w.duration
+            origin = w.origin if w.origin is not None else 0
+            if _is_numeric(d) and _is_numeric(origin):
+                start = ((texpr - origin) // d) * d + origin
+            else:
+                dn = _as_number(d)
+                on = _as_number(origin) if w.origin is not None else None
+
+                def f(tv):
+                    tvn = _as_number(tv)
+                    o = on if on is not None else 0
+                    s = ((tvn - o) // dn) * dn + o
+                    return _back_like(s, tv)
+
+                start = ex.ApplyExpression(f, None, texpr)
+            end = _add_like(start, d)
+            cols: dict[str, Any] = {
+                "_pw_window_start": start,
+                "_pw_window_end": end,
+            }
+            extra = self._with_instance(cols)
+            out = t.with_columns(**cols, **extra)
+            return out, ["_pw_window_start", "_pw_window_end"] + list(extra.keys())
+        if isinstance(w, SlidingWindow):
+            hop = w.hop
+            duration = w.duration if w.duration is not None else (
+                w.ratio * hop if w.ratio is not None else hop
+            )
+            origin = w.origin if w.origin is not None else 0
+            nwin = (
+                int(math.ceil(_as_number(duration) / _as_number(hop)))
+                if _as_number(duration) is not None
+                else 1
+            )
+            parts = []
+            for k in range(nwin):
+                if _is_numeric(hop) and _is_numeric(origin):
+                    smax = ((texpr - origin) // hop) * hop + origin
+                    start = smax - k * hop
+                else:
+                    hn, dn2 = _as_number(hop), _as_number(duration)
+                    on = _as_number(origin)
+
+                    def f(tv, k=k, hn=hn, on=on):
+                        tvn = _as_number(tv)
+                        s = ((tvn - on) // hn) * hn + on - k * hn
+                        return _back_like(s, tv)
+
+                    start = ex.ApplyExpression(f, None, texpr)
+                end = _add_like(start, duration)
+                cols = {"_pw_window_start": start, "_pw_window_end": end}
+                extra = self._with_instance(cols)
+                part = t.with_columns(**cols, **extra)
+                # keep only windows that actually contain the row
+                part = part.filter(
+                    (ex.ColumnReference(part, "_pw_window_end") > self._retime(part))
+                    & (ex.ColumnReference(part, "_pw_window_start") <= self._retime(part))
+                )
+                parts.append(part)
+            out = parts[0].concat_reindex(*parts[1:]) if len(parts) > 1 else parts[0]
+            return out, ["_pw_window_start", "_pw_window_end"] + list(
+                self._with_instance({}).keys()
+            )
+        if isinstance(w, SessionWindow):
+            raise NotImplementedError(
+                "session windows land with the streaming temporal phase"
+            )
+        if isinstance(w, IntervalsOverWindow):
+            raise NotImplementedError("intervals_over lands with the temporal phase")
+        raise TypeError(f"unknown window {w!r}")
+
+    def _retime(self, part: Table):
+        """time expression rebound to the expanded table."""
+        e = self._time_expr
+        if isinstance(e, ex.ColumnReference):
+            return ex.ColumnReference(part, e.name)
+        return e
+
+    def _with_instance(self, cols: dict) -> dict:
+        if self._instance is None:
+            return {}
+        return {"_pw_instance": self._instance}
+
+    def reduce(self, *args: Any, **kwargs: Any) -> Table:
+        prepared, gb_names = self._prepare()
+        gb_refs = [ex.ColumnReference(prepared, n) for n in gb_names]
+        grouped = prepared.groupby(*gb_refs)
+        return grouped.reduce(*args, **kwargs)
+
+
+def _add_like(start_expr, d):
+    return start_expr + d
+
+
+def _back_like(num, proto):
+    import pandas as pd
+
+    if isinstance(proto, pd.Timestamp):
+        return type(proto)(pd.Timestamp(num, unit="ns"))
+    return num
+
+
+def windowby(
+    self: Table,
+    time_expr: Any,
+    *,
+    window: Window,
+    behavior=None,
+    instance: Any = None,
+    **kwargs: Any,
+) -> WindowedTable:
+    return WindowedTable(self, time_expr, window, behavior, instance)

@@ -1,0 +1,422 @@
+"""Core Table API tests (modeled on the reference test_common.py patterns)."""
+
+import pytest
+
+import pathway_amd as pw
+from pathway_amd.debug import (
+    assert_table_equality,
+    assert_table_equality_wo_index,
+    table_from_markdown as T,
+)
+
+
+def test_select_arithmetic():
+    t = T(
+        """
+        a | b
+        1 | 2
+        3 | 4
+        """
+    )
+    res = t.select(pw.this.a, s=pw.this.a + pw.this.b, d=pw.this.b - pw.this.a)
+    expected = T(
+        """
+        a | s | d
+        1 | 3 | 1
+        3 | 7 | 1
+        """
+    )
+    assert_table_equality(res, expected)
+
+
+def test_select_strings():
+    t = T(
+        """
+        name
+        Alice
+        Bob
+        """
+    )
+    res = t.select(upper=pw.this.name.str.upper(), l=pw.this.name.str.len())
+    expected = T(
+        """
+        upper | l
+        ALICE | 5
+        BOB   | 3
+        """
+    )
+    assert_table_equality(res, expected)
+
+
+def test_filter():
+    t = T(
+        """
+        a
+        1
+        2
+        3
+        4
+        """
+    )
+    res = t.filter(pw.this.a > 2)
+    expected = T(
+        """
+        a | __pos__
+        3 | 3
+        4 | 4
+        """
+    ).select(pw.this.a)
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_groupby_count_sum():
+    t = T(
+        """
+        word  | v
+        apple | 1
+        pear  | 2
+        apple | 3
+        """
+    )
+    res = t.groupby(pw.this.word).reduce(
+        pw.this.word, cnt=pw.reducers.count(), total=pw.reducers.sum(pw.this.v)
+    )
+    expected = T(
+        """
+        word  | cnt | total
+        apple | 2   | 4
+        pear  | 1   | 2
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_groupby_min_max_avg():
+    t = T(
+        """
+        g | v
+        a | 5
+        a | 1
+        b | 7
+        """
+    )
+    res = t.groupby(pw.this.g).reduce(
+        pw.this.g,
+        mn=pw.reducers.min(pw.this.v),
+        mx=pw.reducers.max(pw.this.v),
+        av=pw.reducers.avg(pw.this.v),
+    )
+    expected = T(
+        """
+        g | mn | mx | av
+        a | 1  | 5  | 3.0
+        b | 7  | 7  | 7.0
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_groupby_retraction_stream():
+    t = T(
+        """
+        word  | __time__ | __diff__
+        apple | 0        | 1
+        apple | 2        | 1
+        apple | 4        | -1
+        """
+    )
+    res = t.groupby(pw.this.word).reduce(pw.this.word, cnt=pw.reducers.count())
+    expected = T(
+        """
+        word  | cnt
+        apple | 1
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_join_inner():
+    t1 = T(
+        """
+        a | k
+        1 | x
+        2 | y
+        3 | z
+        """
+    )
+    t2 = T(
+        """
+        b | k
+        10 | x
+        20 | y
+        30 | w
+        """
+    )
+    res = t1.join(t2, t1.k == t2.k).select(t1.a, t2.b, pw.this.k)
+    expected = T(
+        """
+        a | b  | k
+        1 | 10 | x
+        2 | 20 | y
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_join_left():
+    t1 = T(
+        """
+        a | k
+        1 | x
+        3 | z
+        """
+    )
+    t2 = T(
+        """
+        b  | k
+        10 | x
+        """
+    )
+    res = t1.join_left(t2, t1.k == t2.k).select(t1.a, t2.b)
+    expected = T(
+        """
+        a | b
+        1 | 10
+        3 |
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_concat_and_update_rows():
+    t1 = T(
+        """
+        a
+        1
+        2
+        """
+    )
+    t2 = T(
+        """
+        a
+        3
+        4
+        """
+    )
+    res = t1.concat_reindex(t2)
+    assert_table_equality_wo_index(
+        res,
+        T(
+            """
+            a
+            1
+            2
+            3
+            4
+            """
+        ),
+    )
+
+
+def test_ix():
+    t = T(
+        """
+        a | ptr_target
+        1 | 10
+        2 | 20
+        """
+    )
+    target = T(
+        """
+        v
+        100
+        200
+        """
+    )
+    # build pointers to target rows via with_id_from
+    keyed = target.with_id_from(pw.this.v)
+    q = t.select(p=t.pointer_from(pw.this.ptr_target * 10))
+    res = keyed.ix(q.p)
+    expected = T(
+        """
+        v
+        100
+        200
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_if_else_coalesce():
+    t = T(
+        """
+        a | b
+        1 |
+        2 | 5
+        """
+    )
+    res = t.select(
+        c=pw.if_else(pw.this.a > 1, pw.this.a * 10, pw.this.a),
+        d=pw.coalesce(pw.this.b, 0),
+    )
+    expected = T(
+        """
+        c  | d
+        1  | 0
+        20 | 5
+        """
+    )
+    assert_table_equality(res, expected)
+
+
+def test_apply_udf():
+    t = T(
+        """
+        a
+        1
+        2
+        """
+    )
+
+    @pw.udf
+    def double(x: int) -> int:
+        return 2 * x
+
+    res = t.select(b=double(pw.this.a))
+    expected = T(
+        """
+        b
+        2
+        4
+        """
+    )
+    assert_table_equality(res, expected)
+
+
+def test_flatten():
+    t = T(
+        """
+        s
+        ab
+        c
+        """
+    )
+    res = t.flatten(pw.this.s)
+    expected = T(
+        """
+        s
+        a
+        b
+        c
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_difference_intersect():
+    t1 = T(
+        """
+        a
+        1
+        2
+        3
+        """
+    )
+    t2 = t1.filter(pw.this.a > 1)
+    diff = t1.difference(t2)
+    assert_table_equality_wo_index(
+        diff,
+        T(
+            """
+            a
+            1
+            """
+        ),
+    )
+    inter = t1.intersect(t2)
+    assert_table_equality_wo_index(
+        inter,
+        T(
+            """
+            a
+            2
+            3
+            """
+        ),
+    )
+
+
+def test_update_cells():
+    t1 = T(
+        """
+        a | b
+        1 | 2
+        3 | 4
+        """
+    )
+    t2 = t1.filter(pw.this.a == 1).select(b=pw.this.b * 100)
+    res = t1.update_cells(t2)
+    expected = T(
+        """
+        a | b
+        1 | 200
+        3 | 4
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_groupby_argmax_tuple():
+    t = T(
+        """
+        g | v
+        a | 5
+        a | 1
+        b | 7
+        """
+    )
+    res = t.groupby(pw.this.g).reduce(
+        pw.this.g,
+        st=pw.reducers.sorted_tuple(pw.this.v),
+    )
+    keys, cols = pw.debug.table_to_dicts(res)
+    vals = {cols["g"][k]: cols["st"][k] for k in keys}
+    assert vals == {"a": (1, 5), "b": (7,)}
+
+
+def test_sql_basic():
+    t = T(
+        """
+        a | b
+        1 | 2
+        3 | 4
+        """
+    )
+    res = pw.sql("SELECT a, a + b AS s FROM tab WHERE a > 1", tab=t)
+    expected = T(
+        """
+        a | s
+        3 | 7
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_sql_groupby():
+    t = T(
+        """
+        g | v
+        a | 1
+        a | 2
+        b | 3
+        """
+    )
+    res = pw.sql("SELECT g, SUM(v) AS s FROM tab GROUP BY g", tab=t)
+    expected = T(
+        """
+        g | s
+        a | 3
+        b | 3
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
