@@ -85,6 +85,9 @@ class IterateNode(Node):
         feeds: dict[str, DeltaBatch | None] = {
             name: inputs[i] for i, name in enumerate(self.names)
         }
+        # deltas injected from OUTSIDE the feedback loop this round — only
+        # these get subtracted in the feedback formula δ_{r+1} = O_r − ext_r
+        external: dict[str, DeltaBatch | None] = dict(feeds)
         accum: dict[str, list[DeltaBatch]] = {name: [] for name in self.result_tables}
         rounds = 0
         while any(b is not None and len(b) for b in feeds.values()):
@@ -106,11 +109,11 @@ class IterateNode(Node):
                 if ob is not None:
                     accum[name].append(ob)
                 if name in self.port_sources:
-                    # feedback delta: X_{r+1} = F(X_r) ⇒ feed O_r minus any
-                    # delta that was fed from OUTSIDE the feedback loop this
-                    # round (the outer input on round 1) — see derivation in
-                    # the module docstring history: δ_{r+1} = O_r − external_r.
-                    ext = feeds.get(name)
+                    # feedback delta: X_{r+1} = F(X_r) ⇒ δ_{r+1} = O_r minus
+                    # any delta injected from OUTSIDE the loop this round
+                    # (the outer input on round 1); feedback deltas themselves
+                    # are already accounted for in the derivation.
+                    ext = external.get(name)
                     parts = []
                     if ob is not None:
                         parts.append(ob)
@@ -121,6 +124,7 @@ class IterateNode(Node):
                     )
             # ports not in results get no further feed
             feeds = new_feeds
+            external = {}
             rounds += 1
             if self.limit is not None and rounds >= self.limit:
                 break
@@ -137,8 +141,13 @@ class IterateOutNode(Node):
         self.iterate_node = iterate_node
         self.out_name = name
 
+    def wants_frontier(self) -> bool:
+        # IterateNode returns None from step(); outputs are read from
+        # last_outputs, so this node must run even with all-None inputs.
+        return True
+
     def step(self, time, inputs):
-        return self.iterate_node.last_outputs.get(self.out_name)
+        return self.iterate_node.last_outputs.pop(self.out_name, None)
 
 
 def run_iterate(func, iteration_limit: int | None = None, **kwargs: Any):

@@ -93,11 +93,10 @@ class Table(TableLike):
         return {n: t.typehint for n, t in self._dtypes.items()}
 
     def __getattr__(self, name: str) -> ex.ColumnReference:
-        if name.startswith("_"):
-            raise AttributeError(name)
-        if name not in self.__dict__.get("_dtypes", {}):
+        dtypes = self.__dict__.get("_dtypes", {})
+        if name not in dtypes:
             raise AttributeError(
-                f"table has no column {name!r}; columns: {list(self._dtypes)}"
+                f"table has no column {name!r}; columns: {list(dtypes)}"
             )
         return ex.ColumnReference(self, name)
 

@@ -1,0 +1,266 @@
+"""Streaming semantics, windows, iterate, temporal joins."""
+
+import pytest
+
+import pathway_amd as pw
+from pathway_amd.debug import (
+    assert_table_equality_wo_index,
+    table_from_markdown as T,
+)
+
+
+def test_update_stream_times():
+    t = T(
+        """
+        a | __time__ | __diff__
+        1 | 0        | 1
+        2 | 2        | 1
+        1 | 4        | -1
+        """
+    )
+    cap = t._capture()
+    from pathway_amd.engine.runtime import Runtime
+    from pathway_amd.internals.rungraph import reset_all
+
+    rt = Runtime([cap])
+    reset_all(rt.nodes)
+    rt.run()
+    stream = [(r.values[0], r.time, r.diff) for r in cap.rows]
+    assert stream == [(1, 0, 1), (2, 2, 1), (1, 4, -1)]
+
+
+def test_tumbling_window():
+    t = T(
+        """
+        t  | v
+        1  | 1
+        3  | 2
+        7  | 3
+        12 | 4
+        """
+    )
+    res = t.windowby(pw.this.t, window=pw.temporal.tumbling(duration=5)).reduce(
+        start=pw.this._pw_window_start,
+        end=pw.this._pw_window_end,
+        s=pw.reducers.sum(pw.this.v),
+    )
+    expected = T(
+        """
+        start | end | s
+        0     | 5   | 3
+        5     | 10  | 3
+        10    | 15  | 4
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_sliding_window():
+    t = T(
+        """
+        t | v
+        2 | 1
+        6 | 2
+        """
+    )
+    res = t.windowby(
+        pw.this.t, window=pw.temporal.sliding(hop=2, duration=4)
+    ).reduce(
+        start=pw.this._pw_window_start,
+        s=pw.reducers.sum(pw.this.v),
+    )
+    # t=2 -> windows starting 0, 2; t=6 -> windows starting 4, 6
+    expected = T(
+        """
+        start | s
+        0     | 1
+        2     | 1
+        4     | 2
+        6     | 2
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_iterate_collatz():
+    t = T(
+        """
+        n
+        3
+        5
+        """
+    )
+
+    def logic(t):
+        return t.select(
+            n=pw.if_else(
+                t.n == 1,
+                1,
+                pw.if_else(t.n % 2 == 0, t.n // 2, 3 * t.n + 1),
+            )
+        )
+
+    res = pw.iterate(logic, t=t)
+    expected = T(
+        """
+        n
+        1
+        1
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_interval_join_inner():
+    a = T(
+        """
+        t | x
+        1 | a1
+        5 | a5
+        """
+    )
+    b = T(
+        """
+        t | y
+        2 | b2
+        9 | b9
+        """
+    )
+    res = pw.temporal.interval_join(
+        a, b, a.t, b.t, pw.temporal.interval(-2, 2)
+    ).select(pw.left.x, pw.right.y)
+    expected = T(
+        """
+        x  | y
+        a1 | b2
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_asof_join():
+    trades = T(
+        """
+        t  | price
+        3  | 100
+        7  | 101
+        """
+    )
+    quotes = T(
+        """
+        t | bid
+        1 | 99
+        5 | 98
+        6 | 97
+        """
+    )
+    res = pw.temporal.asof_join(
+        trades, quotes, trades.t, quotes.t
+    ).select(pw.left.price, pw.right.bid)
+    expected = T(
+        """
+        price | bid
+        100   | 99
+        101   | 97
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_window_join():
+    a = T(
+        """
+        t | x
+        1 | 1
+        6 | 2
+        """
+    )
+    b = T(
+        """
+        t | y
+        2 | 10
+        7 | 20
+        """
+    )
+    res = pw.temporal.window_join(
+        a, b, a.t, b.t, pw.temporal.tumbling(duration=5)
+    ).select(pw.left.x, pw.right.y)
+    expected = T(
+        """
+        x | y
+        1 | 10
+        2 | 20
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_deduplicate():
+    t = T(
+        """
+        v | __time__ | __diff__
+        1 | 0        | 1
+        1 | 2        | 1
+        3 | 4        | 1
+        """
+    )
+    res = t.deduplicate(value=pw.this.v)
+    expected = T(
+        """
+        v
+        3
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_ordered_diff():
+    t = T(
+        """
+        t | v
+        1 | 10
+        2 | 13
+        4 | 20
+        """
+    )
+    res = pw.ordered.diff(t, pw.this.t, pw.this.v)
+    keys, cols = pw.debug.table_to_dicts(res)
+    vals = sorted(v for v in cols["diff_v"].values() if v is not None)
+    assert vals == [3, 7]
+
+
+def test_stream_generator():
+    sg = pw.debug.StreamGenerator()
+
+    class S(pw.Schema):
+        v: int
+
+    t = sg.table_from_list_of_batches([[{"v": 1}], [{"v": 2}]], S)
+    res = t.groupby().reduce(s=pw.reducers.sum(pw.this.v))
+    expected = T(
+        """
+        s
+        3
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_groupby_stream_incremental_updates():
+    t = T(
+        """
+        w | __time__
+        a | 0
+        a | 2
+        """
+    )
+    res = t.groupby(pw.this.w).reduce(pw.this.w, c=pw.reducers.count())
+    cap = res._capture()
+    from pathway_amd.engine.runtime import Runtime
+    from pathway_amd.internals.rungraph import reset_all
+
+    rt = Runtime([cap])
+    reset_all(rt.nodes)
+    rt.run()
+    stream = sorted((r.time, r.diff, r.values[1]) for r in cap.rows)
+    assert stream == [(0, 1, 1), (2, -1, 1), (2, 1, 2)]
