@@ -51,6 +51,9 @@ class StringPool:
         self._hash_lo: list[int] = []
         self._hash_hi: list[int] = []
         self._device_hashes: dict[Any, tuple[torch.Tensor, torch.Tensor, int]] = {}
+        # True when every worker is known to hold an identical pool (codes
+        # agree across ranks) — lets the exchange ship codes instead of bytes
+        self.synchronized = False
 
     def __len__(self) -> int:
         return len(self._strs)

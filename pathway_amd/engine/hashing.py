@@ -111,6 +111,12 @@ def xxh64_words(words: list[torch.Tensor], seed: int) -> torch.Tensor:
 
 
 def hash128_words(words: list[torch.Tensor]) -> tuple[torch.Tensor, torch.Tensor]:
+    if words[0].is_cuda:
+        # GPU path MUST go through the fused HIP kernel (ops raises loudly
+        # if libpwhip.so is missing on a GPU host)
+        from pathway_amd import ops
+
+        return ops.hash128_words_gpu([w.contiguous() for w in words])
     return xxh64_words(words, SEED_LO), xxh64_words(words, SEED_HI)
 
 
@@ -120,6 +126,10 @@ def _float_bits(x: torch.Tensor) -> torch.Tensor:
 
 def value_hash_words(col_words: torch.Tensor, tag: int) -> tuple[torch.Tensor, torch.Tensor]:
     """128-bit per-value hash of a fixed-width column: hash of [tag, payload]."""
+    if col_words.is_cuda:
+        from pathway_amd import ops
+
+        return ops.value_hash_gpu(col_words, tag)
     tags = torch.full_like(col_words, tag)
     return hash128_words([tags, col_words])
 

@@ -340,6 +340,36 @@ class GroupReduceNode(Node):
         for out_name, (spec, args, kwargs) in specs.items():
             arg_cols[out_name] = [evaluate(a, ctx) for a in args]
 
+        rowkeys = b.keys
+        diffs = b.diffs
+
+        # 2b. multi-worker: all-to-all-v by group-key shard (RCCL over xGMI)
+        from pathway_amd.parallel import get_comm
+
+        comm = get_comm()
+        if comm is not None and comm.world > 1:
+            from pathway_amd.parallel.exchange import exchange_bundle, shard_of
+
+            dest = shard_of(gkeys, comm.world)
+            flat_cols: dict[str, Column] = {}
+            for n_, c_ in gcols.items():
+                flat_cols[f"g.{n_}"] = c_
+            for on_, cl_ in arg_cols.items():
+                for i_, c_ in enumerate(cl_):
+                    flat_cols[f"a.{on_}.{i_}"] = c_
+            tensors = {"gkeys": gkeys, "rowkeys": rowkeys, "diffs": diffs}
+            tensors, flat_cols = exchange_bundle(comm, dest, tensors, flat_cols)
+            gkeys = tensors["gkeys"]
+            rowkeys = tensors["rowkeys"]
+            diffs = tensors["diffs"]
+            gcols = {n_: flat_cols[f"g.{n_}"] for n_ in gcols}
+            arg_cols = {
+                on_: [flat_cols[f"a.{on_}.{i_}"] for i_ in range(len(cl_))]
+                for on_, cl_ in arg_cols.items()
+            }
+            if gkeys.shape[0] == 0:
+                return None
+
         # 3. affected keys
         changed = unique_sorted_keys(gkeys)
         cw = [changed[:, 0].contiguous(), changed[:, 1].contiguous()]
@@ -348,9 +378,9 @@ class GroupReduceNode(Node):
         old_presence, old_cols = self._current_rows(changed, cw, specs)
 
         # 5. merge states
-        self._merge_group_store(gkeys, gcols, b.diffs)
-        self._merge_additive(gkeys, arg_cols, b.diffs, specs)
-        self._merge_multiset(gkeys, arg_cols, b, specs)
+        self._merge_group_store(gkeys, gcols, diffs)
+        self._merge_additive(gkeys, arg_cols, diffs, specs)
+        self._merge_multiset(gkeys, arg_cols, rowkeys, diffs, specs)
 
         # 6. new output rows (post-merge)
         new_presence, new_cols = self._current_rows(changed, cw, specs)
@@ -488,7 +518,7 @@ class GroupReduceNode(Node):
             return z, z.clone()
         return hashing.combine_value_hashes(parts)
 
-    def _merge_multiset(self, gkeys, arg_cols, b: DeltaBatch, specs):
+    def _merge_multiset(self, gkeys, arg_cols, rowkeys, diffs, specs):
         ms_cols: dict[str, Column] = {}
         for out_name, (spec, args, kwargs) in specs.items():
             if spec.family != "multiset":
@@ -498,7 +528,7 @@ class GroupReduceNode(Node):
         if not ms_cols:
             return
         # include row identity so distinct input rows stay distinct
-        ms_cols["__rowkey__"] = PointerColumn(b.keys)
+        ms_cols["__rowkey__"] = PointerColumn(rowkeys)
         if self.multiset_store is None:
             self.multiset_store = Arrangement(self.device, ms_cols)
             self.multiset_colnames = list(ms_cols.keys())
@@ -507,7 +537,7 @@ class GroupReduceNode(Node):
             for lo, hi in (c.value_hash() for c in ms_cols.values())
         ]
         v0, v1 = hashing.combine_value_hashes(parts)
-        self.multiset_store.merge(gkeys, (v0, v1), b.diffs, ms_cols)
+        self.multiset_store.merge(gkeys, (v0, v1), diffs, ms_cols)
 
     # -- reading current rows for a set of keys --
 

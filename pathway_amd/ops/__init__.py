@@ -1,0 +1,148 @@
+"""pathway_amd.ops — native HIP/CDNA4 kernel bindings (ctypes, no shims).
+
+On a GPU host the hash hot path MUST run through libpwhip.so; if the
+library is missing there we raise instead of silently falling back to the
+torch reference implementation (which only serves as the CPU path and the
+numerics reference for the kernels).
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+from typing import Sequence
+
+import torch
+
+_THIS = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_THIS, "libpwhip.so")
+
+_lib = None
+_load_error: Exception | None = None
+
+
+def _try_load():
+    global _lib, _load_error
+    if _lib is not None:
+        return _lib
+    try:
+        if not os.path.exists(_LIB_PATH):
+            from pathway_amd.ops import build as _build
+
+            _build.build(verbose=False)
+        _lib = ctypes.CDLL(_LIB_PATH)
+        _lib.pw_hash128_words.restype = ctypes.c_int
+        _lib.pw_value_hash.restype = ctypes.c_int
+        _lib.pw_varlen_hash.restype = ctypes.c_int
+        _lib.pw_run_starts.restype = ctypes.c_int
+    except Exception as e:  # noqa: BLE001
+        _load_error = e
+        _lib = None
+    return _lib
+
+
+def lib_available() -> bool:
+    return _try_load() is not None
+
+
+def require_lib():
+    lib = _try_load()
+    if lib is None:
+        raise RuntimeError(
+            f"pathway_amd native HIP library missing on a GPU host: {_load_error}"
+        )
+    return lib
+
+
+def _stream_ptr() -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def hash128_words_gpu(words: Sequence[torch.Tensor]) -> tuple[torch.Tensor, torch.Tensor]:
+    """Fused device 128-bit hash of rows of int64 words."""
+    lib = require_lib()
+    n = words[0].shape[0]
+    lo = torch.empty(n, dtype=torch.int64, device=words[0].device)
+    hi = torch.empty(n, dtype=torch.int64, device=words[0].device)
+    arr = (ctypes.c_void_p * len(words))(
+        *[ctypes.c_void_p(w.contiguous().data_ptr()) for w in words]
+    )
+    rc = lib.pw_hash128_words(
+        arr,
+        ctypes.c_int(len(words)),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(lo.data_ptr()),
+        ctypes.c_void_p(hi.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_hash128_words failed: hip error {rc}")
+    return lo, hi
+
+
+def value_hash_gpu(payload: torch.Tensor, tag: int) -> tuple[torch.Tensor, torch.Tensor]:
+    lib = require_lib()
+    payload = payload.contiguous()
+    n = payload.shape[0]
+    lo = torch.empty(n, dtype=torch.int64, device=payload.device)
+    hi = torch.empty(n, dtype=torch.int64, device=payload.device)
+    rc = lib.pw_value_hash(
+        ctypes.c_void_p(payload.data_ptr()),
+        ctypes.c_uint64(tag),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(lo.data_ptr()),
+        ctypes.c_void_p(hi.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_value_hash failed: hip error {rc}")
+    return lo, hi
+
+
+def varlen_hash_gpu(
+    bytes_t: torch.Tensor, offsets: torch.Tensor, tag: int
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Device hash of varlen byte rows: offsets (n+1,) int64, bytes uint8."""
+    lib = require_lib()
+    n = offsets.shape[0] - 1
+    lo = torch.empty(n, dtype=torch.int64, device=bytes_t.device)
+    hi = torch.empty(n, dtype=torch.int64, device=bytes_t.device)
+    rc = lib.pw_varlen_hash(
+        ctypes.c_void_p(bytes_t.contiguous().data_ptr()),
+        ctypes.c_void_p(offsets.contiguous().data_ptr()),
+        ctypes.c_uint64(tag),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(lo.data_ptr()),
+        ctypes.c_void_p(hi.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_varlen_hash failed: hip error {rc}")
+    return lo, hi
+
+
+def run_starts_gpu(words: Sequence[torch.Tensor]) -> torch.Tensor:
+    lib = require_lib()
+    n = words[0].shape[0]
+    starts = torch.empty(n, dtype=torch.bool, device=words[0].device)
+    arr = (ctypes.c_void_p * len(words))(
+        *[ctypes.c_void_p(w.contiguous().data_ptr()) for w in words]
+    )
+    rc = lib.pw_run_starts(
+        arr,
+        ctypes.c_int(len(words)),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(starts.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_run_starts failed: hip error {rc}")
+    return starts
+
+
+def host_hash128_bytes(data: bytes) -> tuple[int, int]:
+    lib = require_lib()
+    lo = ctypes.c_uint64()
+    hi = ctypes.c_uint64()
+    lib.pw_host_hash128_bytes(data, ctypes.c_int64(len(data)), ctypes.byref(lo), ctypes.byref(hi))
+    return lo.value, hi.value
