@@ -295,13 +295,13 @@ class StringColumn(Column):
         lo_t, hi_t = self.pool.hash_tensors(self.codes.device)
         codes = self.codes
         valid = codes >= 0
-        safe = torch.where(valid, codes, torch.zeros_like(codes))
+        safe = codes.clamp_min(0)
         lo = lo_t.index_select(0, safe)
         hi = hi_t.index_select(0, safe)
-        if not bool(valid.all()):
-            nlo, nhi = hashing.none_value_hash(len(self), codes.device)
-            lo = torch.where(valid, lo, nlo)
-            hi = torch.where(valid, hi, nhi)
+        # unconditional None-blend: avoids a device→host .all() sync per call
+        nlo, nhi = hashing.none_value_hash_scalar(codes.device)
+        lo = torch.where(valid, lo, nlo)
+        hi = torch.where(valid, hi, nhi)
         return lo, hi
 
     def to_device(self, device) -> "StringColumn":

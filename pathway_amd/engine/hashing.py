@@ -189,14 +189,26 @@ def derive_key_words(
     return hash128_words(words)
 
 
-NONE_HASH_WORDS = None  # computed lazily
+_NONE_HASH_CACHE: dict[str, tuple[torch.Tensor, torch.Tensor]] = {}
+
+
+def none_value_hash_scalar(device) -> tuple[torch.Tensor, torch.Tensor]:
+    """0-dim broadcastable (lo, hi) hash of None, cached per device."""
+    key = str(device)
+    cached = _NONE_HASH_CACHE.get(key)
+    if cached is None:
+        from pathway_amd.internals.api import hash128, serialize_value
+
+        lo, hi = hash128(serialize_value(None))
+        cached = (
+            torch.tensor(_c(lo), dtype=torch.int64, device=device),
+            torch.tensor(_c(hi), dtype=torch.int64, device=device),
+        )
+        _NONE_HASH_CACHE[key] = cached
+    return cached
 
 
 def none_value_hash(n: int, device) -> tuple[torch.Tensor, torch.Tensor]:
-    from pathway_amd.internals.api import hash128, serialize_value
-
-    lo, hi = hash128(serialize_value(None))
-    t = torch.full((n,), _c(lo), dtype=torch.int64, device=device)
-    u = torch.full((n,), _c(hi), dtype=torch.int64, device=device)
+    lo, hi = none_value_hash_scalar(device)
     _ = TAG_NONE
-    return t, u
+    return lo.expand(n), hi.expand(n)

@@ -342,53 +342,103 @@ class GroupReduceNode(Node):
 
         rowkeys = b.keys
         diffs = b.diffs
+        has_multiset = any(s.family == "multiset" for s, _, _ in specs.values())
 
-        # 2b. multi-worker: all-to-all-v by group-key shard (RCCL over xGMI)
+        # 3. local pre-aggregation: ONE lex sort of the batch's group keys,
+        # then segmented sums per additive accumulator (the combiner — also
+        # what the exchange ships for N>1: unique keys + partials, not rows)
+        from pathway_amd.engine.state import rows_ne
+
+        words = [gkeys[:, 0].contiguous(), gkeys[:, 1].contiguous()]
+        perm = lex_sort_words(words)
+        swords = [w.index_select(0, perm) for w in words]
+        starts = rows_ne(swords)
+        seg = torch.cumsum(starts.to(torch.int64), 0) - 1
+        first_idx = starts.nonzero(as_tuple=True)[0]
+        nseg = int(first_idx.numel())
+        ukeys_w = [w.index_select(0, first_idx) for w in swords]
+        contribs = self._contributions(arg_cols, diffs, specs)
+        acc_deltas: dict[str, torch.Tensor] = {}
+        for name, c in contribs.items():
+            sc = c.index_select(0, perm)
+            acc = torch.zeros(nseg, dtype=sc.dtype, device=device)
+            acc.index_add_(0, seg, sc)
+            acc_deltas[name] = acc
+        gfirst_rows = perm.index_select(0, first_idx)
+        gcols_first = {n: c.take(gfirst_rows) for n, c in gcols.items()}
+
+        # 3b. multi-worker: all-to-all-v of pre-aggregated partials by key
+        # shard (RCCL over xGMI; pact.rs:56 analog with combiner)
         from pathway_amd.parallel import get_comm
 
         comm = get_comm()
         if comm is not None and comm.world > 1:
             from pathway_amd.parallel.exchange import exchange_bundle, shard_of
 
-            dest = shard_of(gkeys, comm.world)
-            flat_cols: dict[str, Column] = {}
-            for n_, c_ in gcols.items():
-                flat_cols[f"g.{n_}"] = c_
-            for on_, cl_ in arg_cols.items():
-                for i_, c_ in enumerate(cl_):
-                    flat_cols[f"a.{on_}.{i_}"] = c_
-            tensors = {"gkeys": gkeys, "rowkeys": rowkeys, "diffs": diffs}
-            tensors, flat_cols = exchange_bundle(comm, dest, tensors, flat_cols)
-            gkeys = tensors["gkeys"]
-            rowkeys = tensors["rowkeys"]
-            diffs = tensors["diffs"]
-            gcols = {n_: flat_cols[f"g.{n_}"] for n_ in gcols}
-            arg_cols = {
-                on_: [flat_cols[f"a.{on_}.{i_}"] for i_ in range(len(cl_))]
-                for on_, cl_ in arg_cols.items()
-            }
-            if gkeys.shape[0] == 0:
+            ukeys_t = torch.stack(ukeys_w, dim=1)
+            dest = shard_of(ukeys_t, comm.world)
+            tensors = {"k": ukeys_t}
+            tensors.update({f"acc.{n}": t for n, t in acc_deltas.items()})
+            tensors, gcols_first = exchange_bundle(comm, dest, tensors, gcols_first)
+            ukeys_t = tensors["k"]
+            if ukeys_t.shape[0] == 0 and not has_multiset:
                 return None
+            # re-consolidate: the same key may arrive from several ranks
+            words2 = [ukeys_t[:, 0].contiguous(), ukeys_t[:, 1].contiguous()]
+            perm2 = lex_sort_words(words2)
+            sw2 = [w.index_select(0, perm2) for w in words2]
+            starts2 = rows_ne(sw2)
+            seg2 = torch.cumsum(starts2.to(torch.int64), 0) - 1
+            first2 = starts2.nonzero(as_tuple=True)[0]
+            nseg2 = int(first2.numel())
+            merged_accs = {}
+            for name in acc_deltas:
+                sc = tensors[f"acc.{name}"].index_select(0, perm2)
+                acc = torch.zeros(nseg2, dtype=sc.dtype, device=device)
+                acc.index_add_(0, seg2, sc)
+                merged_accs[name] = acc
+            acc_deltas = merged_accs
+            ukeys_w = [w.index_select(0, first2) for w in sw2]
+            gcols_first = {
+                n: c.take(perm2.index_select(0, first2)) for n, c in gcols_first.items()
+            }
+            if has_multiset:
+                # multiset values cannot be pre-combined: ship the raw rows
+                destm = shard_of(gkeys, comm.world)
+                flat: dict[str, Column] = {}
+                for on_, cl_ in arg_cols.items():
+                    if specs[on_][0].family == "multiset":
+                        for i_, c_ in enumerate(cl_):
+                            flat[f"a.{on_}.{i_}"] = c_
+                tensors_m = {"gkeys": gkeys, "rowkeys": rowkeys, "diffs": diffs}
+                tensors_m, flat = exchange_bundle(comm, destm, tensors_m, flat)
+                gkeys = tensors_m["gkeys"]
+                rowkeys = tensors_m["rowkeys"]
+                diffs = tensors_m["diffs"]
+                for on_, cl_ in list(arg_cols.items()):
+                    if specs[on_][0].family == "multiset":
+                        arg_cols[on_] = [
+                            flat[f"a.{on_}.{i_}"] for i_ in range(len(cl_))
+                        ]
 
-        # 3. affected keys
-        changed = unique_sorted_keys(gkeys)
-        cw = [changed[:, 0].contiguous(), changed[:, 1].contiguous()]
-
-        # 4. old output rows (pre-merge)
+        # 4. affected keys + old output rows (pre-merge)
+        changed = torch.stack(ukeys_w, dim=1)
+        cw = ukeys_w
         old_presence, old_cols = self._current_rows(changed, cw, specs)
 
-        # 5. merge states
-        self._merge_group_store(gkeys, gcols, diffs)
-        self._merge_additive(gkeys, arg_cols, diffs, specs)
-        self._merge_multiset(gkeys, arg_cols, rowkeys, diffs, specs)
+        # 5. merge states (deltas already consolidated per key)
+        self._merge_group_store_pre(ukeys_w, gcols_first, acc_deltas["__w__"])
+        self._merge_additive_pre(ukeys_w, acc_deltas)
+        if has_multiset:
+            self._merge_multiset(gkeys, arg_cols, rowkeys, diffs, specs)
 
         # 6. new output rows (post-merge)
         new_presence, new_cols = self._current_rows(changed, cw, specs)
 
         # 7. emit
         out_batches = []
-        if bool(old_presence.any()):
-            idx = old_presence.nonzero(as_tuple=True)[0]
+        idx = old_presence.nonzero(as_tuple=True)[0]
+        if idx.numel():
             out_batches.append(
                 DeltaBatch(
                     changed.index_select(0, idx),
@@ -397,8 +447,8 @@ class GroupReduceNode(Node):
                     time,
                 )
             )
-        if bool(new_presence.any()):
-            idx = new_presence.nonzero(as_tuple=True)[0]
+        idx = new_presence.nonzero(as_tuple=True)[0]
+        if idx.numel():
             out_batches.append(
                 DeltaBatch(
                     changed.index_select(0, idx),
@@ -425,10 +475,8 @@ class GroupReduceNode(Node):
 
     # -- additive state --
 
-    def _merge_additive(self, gkeys, arg_cols, diffs, specs):
-        device = self.device
-        n = gkeys.shape[0]
-        # build per-row acc contributions
+    def _contributions(self, arg_cols, diffs, specs) -> dict[str, torch.Tensor]:
+        """Per-row additive accumulator contributions (incl. presence __w__)."""
         contribs: dict[str, torch.Tensor] = {"__w__": diffs}
         for out_name, (spec, args, kwargs) in specs.items():
             if spec.family != "additive":
@@ -442,54 +490,47 @@ class GroupReduceNode(Node):
                 if v.dtype == torch.bool:
                     v = v.to(torch.int64)
                 if t.mask is not None:
-                    v = torch.where(t.mask, v, torch.zeros_like(v))
+                    v = v * t.mask
                 contribs[out_name] = v * diffs.to(v.dtype)
             elif spec.name == "avg":
                 t = arg_cols[out_name][0]
                 assert isinstance(t, TensorColumn)
                 v = t.tensor.to(torch.float64)
                 if t.mask is not None:
-                    v = torch.where(t.mask, v, torch.zeros_like(v))
+                    v = v * t.mask
                 contribs[f"{out_name}__sum"] = v * diffs.to(torch.float64)
                 contribs[f"{out_name}__cnt"] = diffs
-        # segment-sum contributions by group key
-        words = [gkeys[:, 0].contiguous(), gkeys[:, 1].contiguous()]
-        perm = lex_sort_words(words)
-        swords = [w.index_select(0, perm) for w in words]
-        sk = torch.stack(swords, dim=1)
-        starts = segment_starts(sk)
-        seg = torch.cumsum(starts.to(torch.int64), 0) - 1
-        nseg = int(seg[-1]) + 1 if n else 0
-        ukeys = [w.index_select(0, starts.nonzero(as_tuple=True)[0]) for w in swords]
-        deltas: dict[str, torch.Tensor] = {}
-        for name, c in contribs.items():
-            sc = c.index_select(0, perm)
-            acc = torch.zeros(nseg, dtype=sc.dtype, device=device)
-            acc.index_add_(0, seg, sc)
-            deltas[name] = acc
-        # merge into state: concat + sort + consolidate-sum
+        return contribs
+
+    def _merge_additive_pre(self, ukeys_w, acc_deltas):
+        """Merge consolidated (unique sorted keys, acc deltas) into state."""
+        device = self.device
+        nseg = ukeys_w[0].shape[0]
         if self.add_keys is None:
-            return
-        all_words = [torch.cat([s, d]) for s, d in zip(self.add_keys, ukeys)]
-        all_accs = {
-            name: torch.cat(
-                [self.add_accs[name].to(deltas.get(name, self.add_accs[name]).dtype)
-                 if name in deltas else self.add_accs[name],
-                 deltas[name]]
-            )
-            if name in deltas
-            else torch.cat(
-                [self.add_accs[name], torch.zeros(nseg, dtype=self.add_accs[name].dtype, device=device)]
-            )
-            for name in self.add_accs
-        }
+            z = torch.zeros((0,), dtype=torch.int64, device=device)
+            self.add_keys = [z, z.clone()]
+            self.add_accs = {
+                name: torch.zeros((0,), dtype=t.dtype, device=device)
+                for name, t in acc_deltas.items()
+            }
+        from pathway_amd.engine.state import rows_ne
+
+        all_words = [torch.cat([s, d]) for s, d in zip(self.add_keys, ukeys_w)]
+        all_accs = {}
+        for name in self.add_accs:
+            d = acc_deltas.get(name)
+            if d is None:
+                d = torch.zeros(nseg, dtype=self.add_accs[name].dtype, device=device)
+            s = self.add_accs[name]
+            if s.dtype != d.dtype:
+                s = s.to(d.dtype)
+            all_accs[name] = torch.cat([s, d])
         perm2 = lex_sort_words(all_words)
         all_words = [w.index_select(0, perm2) for w in all_words]
-        sk2 = torch.stack(all_words, dim=1)
-        starts2 = segment_starts(sk2)
+        starts2 = rows_ne(all_words)
         seg2 = torch.cumsum(starts2.to(torch.int64), 0) - 1
-        nseg2 = int(seg2[-1]) + 1 if seg2.shape[0] else 0
         first_idx = starts2.nonzero(as_tuple=True)[0]
+        nseg2 = int(first_idx.numel())
         merged: dict[str, torch.Tensor] = {}
         for name, acc in all_accs.items():
             sa = acc.index_select(0, perm2)
@@ -503,10 +544,11 @@ class GroupReduceNode(Node):
         ]
         self.add_accs = {name: acc.index_select(0, kidx) for name, acc in merged.items()}
 
-    def _merge_group_store(self, gkeys, gcols, diffs):
-        self._ensure_states(gcols, None)
-        v0, v1 = self._gcols_vhash(gcols, gkeys.shape[0])
-        self.group_store.merge(gkeys, (v0, v1), diffs, gcols)
+    def _merge_group_store_pre(self, ukeys_w, gcols_first, key_weights):
+        self._ensure_states(gcols_first, None)
+        keys = torch.stack(ukeys_w, dim=1)
+        v0, v1 = self._gcols_vhash(gcols_first, keys.shape[0])
+        self.group_store.merge(keys, (v0, v1), key_weights, gcols_first)
 
     def _gcols_vhash(self, gcols: dict[str, Column], n: int):
         parts = [
@@ -598,12 +640,16 @@ class GroupReduceNode(Node):
         if len(store) == 0:
             return torch.zeros_like(pos)
         w = store.weights.index_select(0, pos.clamp(0, len(store) - 1))
-        return torch.where(found, w, torch.zeros_like(w))
+        return w * found
 
     def _additive_lookup(self, cw):
         if self.add_keys is None or self.add_keys[0].shape[0] == 0:
             z = torch.zeros(cw[0].shape[0], dtype=torch.int64, device=self.device)
             return z, torch.zeros_like(z, dtype=torch.bool)
+        if cw[0].is_cuda:
+            from pathway_amd import ops
+
+            return ops.lookup_gpu(self.add_keys, cw)
         pos = searchsorted_words(self.add_keys, cw, side="left")
         m = self.add_keys[0].shape[0]
         safe = pos.clamp(0, m - 1)
@@ -619,7 +665,7 @@ class GroupReduceNode(Node):
         if acc is None or acc.shape[0] == 0:
             return torch.zeros(pos.shape[0], dtype=torch.int64, device=self.device)
         v = acc.index_select(0, pos)
-        return torch.where(found, v, torch.zeros_like(v))
+        return v * found
 
     def _multiset_agg(self, out_name, spec, changed, nq) -> Column:
         device = self.device
@@ -684,8 +730,8 @@ def _build_from_values(vals, device):
 
 
 def _mask_missing(col: Column, found: torch.Tensor, device) -> Column:
-    """Null out positions where found == False."""
-    if bool(found.all()):
+    """Null out positions where found == False (no host sync on GPU)."""
+    if not found.is_cuda and bool(found.all()):
         return col
     if isinstance(col, TensorColumn):
         mask = found & (col.mask if col.mask is not None else torch.ones_like(found))

@@ -40,6 +40,10 @@ def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
 def rows_ne(words: Sequence[torch.Tensor], i_prev_mask: bool = True) -> torch.Tensor:
     """Mask marking rows that differ from their predecessor (run starts)."""
     n = words[0].shape[0]
+    if words[0].is_cuda and n > 0 and len(words) <= 4:
+        from pathway_amd import ops
+
+        return ops.run_starts_gpu([w.contiguous() for w in words])
     starts = torch.ones(n, dtype=torch.bool, device=words[0].device)
     if n > 1:
         ne = torch.zeros(n - 1, dtype=torch.bool, device=words[0].device)
@@ -82,10 +86,18 @@ def searchsorted_words(
     query_words: Sequence[torch.Tensor],
     side: str = "left",
 ) -> torch.Tensor:
-    """Vectorized multiword searchsorted (lexicographic) via binary search."""
+    """Vectorized multiword searchsorted (lexicographic) via binary search.
+
+    GPU path: one HIP kernel (pw_searchsorted); the torch loop below is the
+    CPU path and the kernel's numerics reference.
+    """
     m = sorted_words[0].shape[0]
     nq = query_words[0].shape[0]
     device = query_words[0].device
+    if query_words[0].is_cuda and len(sorted_words) <= 4 and m > 0 and nq > 0:
+        from pathway_amd import ops
+
+        return ops.searchsorted_gpu(sorted_words, query_words, side)
     lo = torch.zeros(nq, dtype=torch.int64, device=device)
     hi = torch.full((nq,), m, dtype=torch.int64, device=device)
     if m == 0 or nq == 0:
@@ -141,6 +153,10 @@ class Arrangement:
     def key_range(self, query_keys: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
         """[lo, hi) row ranges for (nq,2) query keys (by key prefix)."""
         q = [query_keys[:, 0].contiguous(), query_keys[:, 1].contiguous()]
+        if query_keys.is_cuda and len(self) > 0 and query_keys.shape[0] > 0:
+            from pathway_amd import ops
+
+            return ops.key_range_gpu(self.key_words, q)
         lo = searchsorted_words(self.key_words, q, side="left")
         hi = searchsorted_words(self.key_words, q, side="right")
         return lo, hi
@@ -215,6 +231,10 @@ class AdditiveState:
     def lookup(self, keys: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
         """(positions, found_mask) of (nq,2) keys in the state."""
         q = [keys[:, 0].contiguous(), keys[:, 1].contiguous()]
+        if keys.is_cuda and len(self) > 0 and keys.shape[0] > 0:
+            from pathway_amd import ops
+
+            return ops.lookup_gpu(self.key_words, q)
         pos = searchsorted_words(self.key_words, q, side="left")
         m = len(self)
         safe = pos.clamp(0, max(m - 1, 0))

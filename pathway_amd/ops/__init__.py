@@ -146,3 +146,85 @@ def host_hash128_bytes(data: bytes) -> tuple[int, int]:
     hi = ctypes.c_uint64()
     lib.pw_host_hash128_bytes(data, ctypes.c_int64(len(data)), ctypes.byref(lo), ctypes.byref(hi))
     return lo.value, hi.value
+
+
+def _ptr_arr(tensors):
+    return (ctypes.c_void_p * len(tensors))(
+        *[ctypes.c_void_p(t.data_ptr()) for t in tensors]
+    )
+
+
+def searchsorted_gpu(
+    sorted_words: Sequence[torch.Tensor],
+    query_words: Sequence[torch.Tensor],
+    side: str = "left",
+) -> torch.Tensor:
+    lib = require_lib()
+    m = sorted_words[0].shape[0]
+    nq = query_words[0].shape[0]
+    out = torch.empty(nq, dtype=torch.int64, device=query_words[0].device)
+    s = [t.contiguous() for t in sorted_words]
+    q = [t.contiguous() for t in query_words]
+    rc = lib.pw_searchsorted(
+        _ptr_arr(s),
+        _ptr_arr(q),
+        ctypes.c_int(len(s)),
+        ctypes.c_int64(m),
+        ctypes.c_int64(nq),
+        ctypes.c_int(1 if side == "right" else 0),
+        ctypes.c_void_p(out.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_searchsorted failed: hip error {rc}")
+    return out
+
+
+def lookup_gpu(
+    sorted_words: Sequence[torch.Tensor], query_words: Sequence[torch.Tensor]
+) -> tuple[torch.Tensor, torch.Tensor]:
+    lib = require_lib()
+    m = sorted_words[0].shape[0]
+    nq = query_words[0].shape[0]
+    pos = torch.empty(nq, dtype=torch.int64, device=query_words[0].device)
+    found = torch.empty(nq, dtype=torch.bool, device=query_words[0].device)
+    s = [t.contiguous() for t in sorted_words]
+    q = [t.contiguous() for t in query_words]
+    rc = lib.pw_lookup(
+        _ptr_arr(s),
+        _ptr_arr(q),
+        ctypes.c_int(len(s)),
+        ctypes.c_int64(m),
+        ctypes.c_int64(nq),
+        ctypes.c_void_p(pos.data_ptr()),
+        ctypes.c_void_p(found.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_lookup failed: hip error {rc}")
+    return pos, found
+
+
+def key_range_gpu(
+    sorted_words: Sequence[torch.Tensor], query_words: Sequence[torch.Tensor]
+) -> tuple[torch.Tensor, torch.Tensor]:
+    lib = require_lib()
+    m = sorted_words[0].shape[0]
+    nq = query_words[0].shape[0]
+    lo = torch.empty(nq, dtype=torch.int64, device=query_words[0].device)
+    hi = torch.empty(nq, dtype=torch.int64, device=query_words[0].device)
+    s = [t.contiguous() for t in sorted_words]
+    q = [t.contiguous() for t in query_words]
+    rc = lib.pw_key_range(
+        _ptr_arr(s),
+        _ptr_arr(q),
+        ctypes.c_int(len(s)),
+        ctypes.c_int64(m),
+        ctypes.c_int64(nq),
+        ctypes.c_void_p(lo.data_ptr()),
+        ctypes.c_void_p(hi.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_key_range failed: hip error {rc}")
+    return lo, hi

@@ -248,3 +248,141 @@ extern "C" void pw_host_varlen_hash(const void* bytes, const int64_t* offsets,
     if (buf != stackbuf) delete[] buf;
   }
 }
+
+// ------------------------------------------------------- sorted-key search --
+
+// Lexicographic binary search over up-to-4 parallel sorted word columns.
+// One thread per query — replaces the ~10-kernel-per-iteration torch
+// binary-search loop (the dominant launch-overhead cost in the profile,
+// profiles/wordcount_r01.md).
+struct SWords {
+  const uint64_t* s[4];
+  const uint64_t* q[4];
+};
+
+__device__ __forceinline__ int pw_cmp_row(const SWords& w, int nw, int64_t si,
+                                          int64_t qi) {
+  // compare sorted[si] ? query[qi] as SIGNED int64 lexicographic
+  for (int j = 0; j < nw; ++j) {
+    int64_t a = (int64_t)w.s[j][si];
+    int64_t b = (int64_t)w.q[j][qi];
+    if (a < b) return -1;
+    if (a > b) return 1;
+  }
+  return 0;
+}
+
+template <bool RIGHT>
+__global__ void k_searchsorted(SWords w, int nw, int64_t m, int64_t nq,
+                               int64_t* out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nq;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t lo = 0, hi = m;
+    while (lo < hi) {
+      int64_t mid = (lo + hi) >> 1;
+      int c = pw_cmp_row(w, nw, mid, i);
+      bool go_right = RIGHT ? (c <= 0) : (c < 0);
+      if (go_right)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    out[i] = lo;
+  }
+}
+
+extern "C" int pw_searchsorted(const void** sorted_ptrs, const void** query_ptrs,
+                               int nwords, int64_t m, int64_t nq, int right,
+                               void* out, void* stream) {
+  SWords w;
+  for (int j = 0; j < nwords && j < 4; ++j) {
+    w.s[j] = (const uint64_t*)sorted_ptrs[j];
+    w.q[j] = (const uint64_t*)query_ptrs[j];
+  }
+  dim3 grid(pw_grid(nq)), block(PW_BLOCK);
+  if (right)
+    hipLaunchKernelGGL((k_searchsorted<true>), grid, block, 0,
+                       (hipStream_t)stream, w, nwords, m, nq, (int64_t*)out);
+  else
+    hipLaunchKernelGGL((k_searchsorted<false>), grid, block, 0,
+                       (hipStream_t)stream, w, nwords, m, nq, (int64_t*)out);
+  return (int)hipGetLastError();
+}
+
+// lookup: left-searchsorted + equality check, fused (pos clamped, found flag)
+__global__ void k_lookup(SWords w, int nw, int64_t m, int64_t nq, int64_t* pos,
+                         bool* found) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nq;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t lo = 0, hi = m;
+    while (lo < hi) {
+      int64_t mid = (lo + hi) >> 1;
+      if (pw_cmp_row(w, nw, mid, i) < 0)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    bool f = false;
+    int64_t p = lo;
+    if (lo < m && pw_cmp_row(w, nw, lo, i) == 0) f = true;
+    if (p >= m) p = m > 0 ? m - 1 : 0;
+    pos[i] = p;
+    found[i] = f;
+  }
+}
+
+extern "C" int pw_lookup(const void** sorted_ptrs, const void** query_ptrs,
+                         int nwords, int64_t m, int64_t nq, void* pos,
+                         void* found, void* stream) {
+  SWords w;
+  for (int j = 0; j < nwords && j < 4; ++j) {
+    w.s[j] = (const uint64_t*)sorted_ptrs[j];
+    w.q[j] = (const uint64_t*)query_ptrs[j];
+  }
+  hipLaunchKernelGGL(k_lookup, dim3(pw_grid(nq)), dim3(PW_BLOCK), 0,
+                     (hipStream_t)stream, w, nwords, m, nq, (int64_t*)pos,
+                     (bool*)found);
+  return (int)hipGetLastError();
+}
+
+// key_range: lo (left bound) and hi (right bound) in one kernel
+__global__ void k_key_range(SWords w, int nw, int64_t m, int64_t nq,
+                            int64_t* lo_out, int64_t* hi_out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nq;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t lo = 0, hi = m;
+    while (lo < hi) {
+      int64_t mid = (lo + hi) >> 1;
+      if (pw_cmp_row(w, nw, mid, i) < 0)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    int64_t left = lo;
+    lo = left;
+    hi = m;
+    while (lo < hi) {
+      int64_t mid = (lo + hi) >> 1;
+      if (pw_cmp_row(w, nw, mid, i) <= 0)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    lo_out[i] = left;
+    hi_out[i] = lo;
+  }
+}
+
+extern "C" int pw_key_range(const void** sorted_ptrs, const void** query_ptrs,
+                            int nwords, int64_t m, int64_t nq, void* lo,
+                            void* hi, void* stream) {
+  SWords w;
+  for (int j = 0; j < nwords && j < 4; ++j) {
+    w.s[j] = (const uint64_t*)sorted_ptrs[j];
+    w.q[j] = (const uint64_t*)query_ptrs[j];
+  }
+  hipLaunchKernelGGL(k_key_range, dim3(pw_grid(nq)), dim3(PW_BLOCK), 0,
+                     (hipStream_t)stream, w, nwords, m, nq, (int64_t*)lo,
+                     (int64_t*)hi);
+  return (int)hipGetLastError();
+}
