@@ -304,85 +304,118 @@ class GroupReduceNode(Node):
                     names.append(out_name)
         return names
 
+    def wants_frontier(self) -> bool:
+        # in multi-worker mode every rank must join the exchange collectives
+        # each step, even with no local delta
+        from pathway_amd.parallel import get_comm
+
+        c = get_comm()
+        return c is not None and c.world > 1
+
     def step(self, time, inputs):
         b = inputs[0]
-        if b is None or len(b) == 0:
-            return None
-        device = self.device
-        seq0 = self.seq
-        self.seq += len(b)
-        seq_col = TensorColumn(
-            torch.arange(seq0, seq0 + len(b), dtype=torch.int64, device=device), dt.INT
-        )
-        ctx = EvalContext(b.columns, b.keys, device, extra={"__seq__": seq_col})
-        # 1. evaluate grouping columns and compute group keys
-        gcols = {n: evaluate(e, ctx) for n, e in self.group_exprs.items()}
-        if self.key_expr is not None:
-            kc = evaluate(self.key_expr, ctx)
-            assert isinstance(kc, PointerColumn), "groupby(id=...) needs a pointer"
-            gkeys = kc.pairs
-        else:
-            parts = [
-                (lo.to(device), hi.to(device))
-                for lo, hi in (c.value_hash() for c in gcols.values())
-            ]
-            if parts:
-                glo, ghi = hashing.combine_value_hashes(parts)
-            else:
-                glo = torch.zeros(len(b), dtype=torch.int64, device=device)
-                ghi = glo.clone()
-            gkeys = torch.stack([glo, ghi], dim=1)
-
-        specs = self._specs()
-
-        # 2. evaluate reducer args
-        arg_cols: dict[str, list[Column]] = {}
-        for out_name, (spec, args, kwargs) in specs.items():
-            arg_cols[out_name] = [evaluate(a, ctx) for a in args]
-
-        rowkeys = b.keys
-        diffs = b.diffs
-        has_multiset = any(s.family == "multiset" for s, _, _ in specs.values())
-
-        # 3. local pre-aggregation: ONE lex sort of the batch's group keys,
-        # then segmented sums per additive accumulator (the combiner — also
-        # what the exchange ships for N>1: unique keys + partials, not rows)
-        from pathway_amd.engine.state import rows_ne
-
-        words = [gkeys[:, 0].contiguous(), gkeys[:, 1].contiguous()]
-        perm = lex_sort_words(words)
-        swords = [w.index_select(0, perm) for w in words]
-        starts = rows_ne(swords)
-        seg = torch.cumsum(starts.to(torch.int64), 0) - 1
-        first_idx = starts.nonzero(as_tuple=True)[0]
-        nseg = int(first_idx.numel())
-        ukeys_w = [w.index_select(0, first_idx) for w in swords]
-        contribs = self._contributions(arg_cols, diffs, specs)
-        acc_deltas: dict[str, torch.Tensor] = {}
-        for name, c in contribs.items():
-            sc = c.index_select(0, perm)
-            acc = torch.zeros(nseg, dtype=sc.dtype, device=device)
-            acc.index_add_(0, seg, sc)
-            acc_deltas[name] = acc
-        gfirst_rows = perm.index_select(0, first_idx)
-        gcols_first = {n: c.take(gfirst_rows) for n, c in gcols.items()}
-
-        # 3b. multi-worker: all-to-all-v of pre-aggregated partials by key
-        # shard (RCCL over xGMI; pact.rs:56 analog with combiner)
         from pathway_amd.parallel import get_comm
 
         comm = get_comm()
-        if comm is not None and comm.world > 1:
+        distributed = comm is not None and comm.world > 1
+        if b is None or len(b) == 0:
+            if not distributed:
+                return None
+            b = None
+        device = self.device
+        specs = self._specs()
+        has_multiset = any(s.family == "multiset" for s, _, _ in specs.values())
+        from pathway_amd.engine.state import rows_ne
+
+        gkeys = rowkeys = diffs = None
+        gcols: dict[str, Column] = {}
+        arg_cols: dict[str, list[Column]] = {}
+        ukeys_w = acc_deltas = gcols_first = None
+        if b is not None:
+            seq0 = self.seq
+            self.seq += len(b)
+            seq_col = TensorColumn(
+                torch.arange(seq0, seq0 + len(b), dtype=torch.int64, device=device),
+                dt.INT,
+            )
+            ctx = EvalContext(b.columns, b.keys, device, extra={"__seq__": seq_col})
+            # 1. evaluate grouping columns and compute group keys
+            gcols = {n: evaluate(e, ctx) for n, e in self.group_exprs.items()}
+            if self.key_expr is not None:
+                kc = evaluate(self.key_expr, ctx)
+                assert isinstance(kc, PointerColumn), "groupby(id=...) needs a pointer"
+                gkeys = kc.pairs
+            else:
+                parts = [
+                    (lo.to(device), hi.to(device))
+                    for lo, hi in (c.value_hash() for c in gcols.values())
+                ]
+                if parts:
+                    glo, ghi = hashing.combine_value_hashes(parts)
+                else:
+                    glo = torch.zeros(len(b), dtype=torch.int64, device=device)
+                    ghi = glo.clone()
+                gkeys = torch.stack([glo, ghi], dim=1)
+
+            # 2. evaluate reducer args
+            for out_name, (spec, args, kwargs) in specs.items():
+                arg_cols[out_name] = [evaluate(a, ctx) for a in args]
+
+            rowkeys = b.keys
+            diffs = b.diffs
+
+            # 3. local pre-aggregation: ONE lex sort of the batch's group
+            # keys, then segmented sums per additive accumulator (the
+            # combiner — also what the exchange ships for N>1)
+            words = [gkeys[:, 0].contiguous(), gkeys[:, 1].contiguous()]
+            perm = lex_sort_words(words)
+            swords = [w.index_select(0, perm) for w in words]
+            starts = rows_ne(swords)
+            seg = torch.cumsum(starts.to(torch.int64), 0) - 1
+            first_idx = starts.nonzero(as_tuple=True)[0]
+            nseg = int(first_idx.numel())
+            ukeys_w = [w.index_select(0, first_idx) for w in swords]
+            contribs = self._contributions(arg_cols, diffs, specs)
+            acc_deltas = {}
+            for name, c in contribs.items():
+                sc = c.index_select(0, perm)
+                acc = torch.zeros(nseg, dtype=sc.dtype, device=device)
+                acc.index_add_(0, seg, sc)
+                acc_deltas[name] = acc
+            gfirst_rows = perm.index_select(0, first_idx)
+            gcols_first = {n: c.take(gfirst_rows) for n, c in gcols.items()}
+
+        # 3b. multi-worker: all-to-all-v of pre-aggregated partials by key
+        # shard (RCCL over xGMI; pact.rs:56 analog with combiner)
+        if distributed:
             from pathway_amd.parallel.exchange import exchange_bundle, shard_of
 
-            ukeys_t = torch.stack(ukeys_w, dim=1)
-            dest = shard_of(ukeys_t, comm.world)
-            tensors = {"k": ukeys_t}
-            tensors.update({f"acc.{n}": t for n, t in acc_deltas.items()})
+            if ukeys_w is not None:
+                ukeys_t = torch.stack(ukeys_w, dim=1)
+                dest = shard_of(ukeys_t, comm.world)
+                tensors = {"k": ukeys_t}
+                tensors.update({f"acc.{n}": t for n, t in acc_deltas.items()})
+            else:
+                dest = tensors = None
             tensors, gcols_first = exchange_bundle(comm, dest, tensors, gcols_first)
+            ms_exchanged = None
+            if has_multiset:
+                # multiset values cannot be pre-combined: ship the raw rows
+                if gkeys is not None:
+                    destm = shard_of(gkeys, comm.world)
+                    flat: dict[str, Column] = {}
+                    for on_, cl_ in arg_cols.items():
+                        if specs[on_][0].family == "multiset":
+                            for i_, c_ in enumerate(cl_):
+                                flat[f"a.{on_}.{i_}"] = c_
+                    tensors_m = {"gkeys": gkeys, "rowkeys": rowkeys, "diffs": diffs}
+                else:
+                    destm = tensors_m = flat = None
+                tensors_m, flat = exchange_bundle(comm, destm, tensors_m, flat)
+                ms_exchanged = (tensors_m, flat)
+            if tensors is None:
+                return None  # no rank had data this step
             ukeys_t = tensors["k"]
-            if ukeys_t.shape[0] == 0 and not has_multiset:
-                return None
             # re-consolidate: the same key may arrive from several ranks
             words2 = [ukeys_t[:, 0].contiguous(), ukeys_t[:, 1].contiguous()]
             perm2 = lex_sort_words(words2)
@@ -391,8 +424,9 @@ class GroupReduceNode(Node):
             seg2 = torch.cumsum(starts2.to(torch.int64), 0) - 1
             first2 = starts2.nonzero(as_tuple=True)[0]
             nseg2 = int(first2.numel())
+            acc_names = [k[4:] for k in tensors if k.startswith("acc.")]
             merged_accs = {}
-            for name in acc_deltas:
+            for name in acc_names:
                 sc = tensors[f"acc.{name}"].index_select(0, perm2)
                 acc = torch.zeros(nseg2, dtype=sc.dtype, device=device)
                 acc.index_add_(0, seg2, sc)
@@ -402,24 +436,17 @@ class GroupReduceNode(Node):
             gcols_first = {
                 n: c.take(perm2.index_select(0, first2)) for n, c in gcols_first.items()
             }
-            if has_multiset:
-                # multiset values cannot be pre-combined: ship the raw rows
-                destm = shard_of(gkeys, comm.world)
-                flat: dict[str, Column] = {}
-                for on_, cl_ in arg_cols.items():
-                    if specs[on_][0].family == "multiset":
-                        for i_, c_ in enumerate(cl_):
-                            flat[f"a.{on_}.{i_}"] = c_
-                tensors_m = {"gkeys": gkeys, "rowkeys": rowkeys, "diffs": diffs}
-                tensors_m, flat = exchange_bundle(comm, destm, tensors_m, flat)
+            if has_multiset and ms_exchanged is not None and ms_exchanged[0] is not None:
+                tensors_m, flat = ms_exchanged
                 gkeys = tensors_m["gkeys"]
                 rowkeys = tensors_m["rowkeys"]
                 diffs = tensors_m["diffs"]
                 for on_, cl_ in list(arg_cols.items()):
                     if specs[on_][0].family == "multiset":
-                        arg_cols[on_] = [
-                            flat[f"a.{on_}.{i_}"] for i_ in range(len(cl_))
-                        ]
+                        ncols = sum(1 for k in flat if k.startswith(f"a.{on_}."))
+                        arg_cols[on_] = [flat[f"a.{on_}.{i_}"] for i_ in range(ncols)]
+        elif ukeys_w is None:
+            return None
 
         # 4. affected keys + old output rows (pre-merge)
         changed = torch.stack(ukeys_w, dim=1)
@@ -429,7 +456,7 @@ class GroupReduceNode(Node):
         # 5. merge states (deltas already consolidated per key)
         self._merge_group_store_pre(ukeys_w, gcols_first, acc_deltas["__w__"])
         self._merge_additive_pre(ukeys_w, acc_deltas)
-        if has_multiset:
+        if has_multiset and gkeys is not None:
             self._merge_multiset(gkeys, arg_cols, rowkeys, diffs, specs)
 
         # 6. new output rows (post-merge)

@@ -50,6 +50,28 @@ def _join_keys(exprs: list[Any], b: DeltaBatch, device) -> torch.Tensor:
     return torch.stack([lo, hi], dim=1)
 
 
+def _exchange_side(comm, b: DeltaBatch | None, jk: torch.Tensor | None, time: int):
+    """All-to-all one join side's delta rows by join-key shard.
+
+    Every rank must participate in the collective even with an empty local
+    batch (peers may still route rows here)."""
+    from pathway_amd.parallel.exchange import exchange_bundle, shard_of
+
+    if b is None:
+        tensors, cols = exchange_bundle(comm, None, None, None)
+    else:
+        dest = shard_of(jk, comm.world)
+        tensors, cols = exchange_bundle(
+            comm, dest, {"jk": jk, "keys": b.keys, "diffs": b.diffs}, dict(b.columns)
+        )
+    if tensors is None:
+        return None, None
+    nb = DeltaBatch(tensors["keys"], cols, tensors["diffs"], time)
+    if len(nb) == 0:
+        return None, None
+    return nb, tensors["jk"]
+
+
 class _SideStore:
     """Arrangement of one join side keyed by join key, payload = row."""
 
@@ -144,18 +166,32 @@ class JoinNode(Node):
         self._left_proto = None
         self._right_proto = None
 
+    def wants_frontier(self) -> bool:
+        from pathway_amd.parallel import get_comm
+
+        c = get_comm()
+        return c is not None and c.world > 1
+
     def step(self, time, inputs):
         bl, br = inputs
         device = self.device
         out_parts: list[DeltaBatch] = []
 
+        jl = _join_keys(self.left_on, bl, device) if bl is not None and len(bl) else None
+        jr = _join_keys(self.right_on, br, device) if br is not None and len(br) else None
+
+        # multi-worker: co-shard both sides by join key (RCCL all-to-all)
+        from pathway_amd.parallel import get_comm
+
+        comm = get_comm()
+        if comm is not None and comm.world > 1:
+            bl, jl = _exchange_side(comm, bl, jl, time)
+            br, jr = _exchange_side(comm, br, jr, time)
+
         if bl is not None and len(bl):
             self._left_proto = self._left_proto or dict(bl.columns)
         if br is not None and len(br):
             self._right_proto = self._right_proto or dict(br.columns)
-
-        jl = _join_keys(self.left_on, bl, device) if bl is not None and len(bl) else None
-        jr = _join_keys(self.right_on, br, device) if br is not None and len(br) else None
 
         affected_list = []
         if jl is not None:

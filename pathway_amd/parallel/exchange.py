@@ -26,15 +26,86 @@ def shard_of(keys: torch.Tensor, world: int) -> torch.Tensor:
     return (keys[:, 0] & SHARD_MASK) % world
 
 
+def _bundle_meta(tensors, columns):
+    meta = {"t": {}, "c": {}}
+    for name, t in tensors.items():
+        meta["t"][name] = (str(t.dtype), tuple(t.shape[1:]))
+    for name, c in columns.items():
+        if isinstance(c, TensorColumn):
+            meta["c"][name] = ("tensor", str(c.tensor.dtype), c.dtype, c.mask is not None)
+        elif isinstance(c, PointerColumn):
+            meta["c"][name] = ("pointer", None, c.dtype, False)
+        elif isinstance(c, StringColumn):
+            meta["c"][name] = ("string", None, c.dtype, False)
+        else:
+            meta["c"][name] = ("object", None, c.dtype, False)
+    return meta
+
+
+_TORCH_DTYPES = {
+    "torch.int64": torch.int64,
+    "torch.int32": torch.int32,
+    "torch.float64": torch.float64,
+    "torch.float32": torch.float32,
+    "torch.bool": torch.bool,
+    "torch.uint8": torch.uint8,
+    "torch.bfloat16": torch.bfloat16,
+    "torch.float16": torch.float16,
+}
+
+
+def _empty_bundle_from_meta(meta, device):
+    import numpy as np
+
+    from pathway_amd.engine.column import GLOBAL_STRING_POOL
+
+    tensors = {}
+    for name, (dts, trail) in meta["t"].items():
+        tensors[name] = torch.zeros((0, *trail), dtype=_TORCH_DTYPES[dts], device=device)
+    columns = {}
+    for name, (kind, tds, ddt, has_mask) in meta["c"].items():
+        if kind == "tensor":
+            t = torch.zeros((0,), dtype=_TORCH_DTYPES[tds], device=device)
+            mask = torch.zeros((0,), dtype=torch.bool, device=device) if has_mask else None
+            columns[name] = TensorColumn(t, ddt, mask)
+        elif kind == "pointer":
+            columns[name] = PointerColumn(
+                torch.zeros((0, 2), dtype=torch.int64, device=device), ddt
+            )
+        elif kind == "string":
+            columns[name] = StringColumn(
+                torch.zeros((0,), dtype=torch.int64, device=device),
+                GLOBAL_STRING_POOL,
+                ddt,
+            )
+        else:
+            columns[name] = ObjectColumn(np.empty(0, dtype=object), ddt)
+    return tensors, columns
+
+
 def exchange_bundle(
     comm,
-    dest: torch.Tensor,
-    tensors: dict[str, torch.Tensor],
-    columns: dict[str, Column],
-) -> tuple[dict[str, torch.Tensor], dict[str, Column]]:
+    dest: torch.Tensor | None,
+    tensors: dict[str, torch.Tensor] | None,
+    columns: dict[str, Column] | None,
+) -> tuple[dict[str, torch.Tensor] | None, dict[str, Column] | None]:
     """Exchange rows by dest; tensors dict = named per-row tensors (first dim
-    n); columns = engine Columns.  Returns received (tensors, columns)."""
+    n); columns = engine Columns.  Returns received (tensors, columns).
+
+    Pass None tensors/columns for "no local data": the rank still joins the
+    collective; the bundle schema is learned from peers via a small
+    metadata all-gather.  Returns (None, None) when NO rank had data.
+    """
     world = comm.world
+    local_meta = None if tensors is None else _bundle_meta(tensors, columns or {})
+    metas = comm.all_to_all_objects([local_meta] * world)
+    merged = next((m for m in metas if m is not None), None)
+    if local_meta is None:
+        if merged is None:
+            return None, None
+        tensors, columns = _empty_bundle_from_meta(merged, comm.device)
+        dest = torch.zeros((0,), dtype=torch.int64, device=comm.device)
+    columns = columns or {}
     perm = torch.argsort(dest)
     counts = torch.bincount(dest, minlength=world)
     out_tensors: dict[str, torch.Tensor] = {}

@@ -75,3 +75,79 @@ def test_distributed_groupby_gloo():
     assert all_rows == [("a", 2, 11), ("b", 1, 2), ("c", 1, 5)]
     # shards must be disjoint
     assert not (set(results[0]) & set(results[1]))
+
+
+def _worker_join(rank: int, world: int, port: int, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["PW_DEVICE"] = "cpu"
+    import pathway_amd as pw
+    import pathway_amd.parallel as par
+    from pathway_amd.debug import table_from_markdown as T
+
+    par.init(backend="gloo")
+    if rank == 0:
+        t1 = T(
+            """
+            a | k
+            1 | x
+            2 | y
+            """
+        )
+        t2 = T(
+            """
+            b  | k
+            20 | y
+            """
+        )
+    else:
+        t1 = T(
+            """
+            a | k
+            3 | z
+            """
+        )
+        t2 = T(
+            """
+            b  | k
+            10 | x
+            30 | z
+            """
+        )
+    res = t1.join(t2, t1.k == t2.k).select(t1.a, t2.b, pw.this.k)
+    cap = res._capture()
+    from pathway_amd.engine.runtime import Runtime
+    from pathway_amd.internals.rungraph import reset_all
+
+    rt = Runtime([cap], comm=par.get_comm())
+    reset_all(rt.nodes)
+    rt.run()
+    from pathway_amd.internals.api import squash_updates
+
+    state = squash_updates(cap.rows)
+    rows = sorted(tuple(v) for v in state.values())
+    q.put((rank, rows))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_distributed_join_gloo():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = 29720
+    procs = [ctx.Process(target=_worker_join, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, rows = q.get()
+        results[rank] = rows
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0
+    all_rows = sorted(results[0] + results[1])
+    assert all_rows == [(1, 10, "x"), (2, 20, "y"), (3, 30, "z")]
