@@ -104,6 +104,15 @@ class StringPool:
 GLOBAL_STRING_POOL = StringPool()
 
 
+def obj_array(values) -> np.ndarray:
+    """Object ndarray built element-wise — np.array() would flatten nested
+    sequences of equal length into a 2-D array."""
+    arr = np.empty(len(values), dtype=object)
+    for i, v in enumerate(values):
+        arr[i] = v
+    return arr
+
+
 class Column:
     dtype: dt.DType
 

@@ -108,7 +108,9 @@ def evaluate(e: expr.ColumnExpression, ctx: EvalContext) -> Column:
         cols = [evaluate(a, ctx) for a in e._args]
         vals = [_host_values(c) for c in cols]
         return ObjectColumn(
-            np.array([tuple(row) for row in zip(*vals)] if vals else [()] * ctx.n, dtype=object),
+            __import__("pathway_amd.engine.column", fromlist=["obj_array"]).obj_array(
+                [tuple(row) for row in zip(*vals)] if vals else [()] * ctx.n
+            ),
             dt.ANY_TUPLE,
         )
     if isinstance(e, expr.GetExpression):
@@ -543,6 +545,16 @@ def _eval_apply(e: expr.ApplyExpression, ctx: EvalContext, is_async: bool) -> Co
     kw_vals = {k: _host_values(c) for k, c in kw_cols.items()}
     n = ctx.n
     fun = e._fun
+    batch_fun = getattr(e, "_batch_fun", None)
+    if batch_fun is not None:
+        # batched UDF (embedders etc.): one call per batch, GIL-amortized —
+        # reference BatchWrapper::WithGil + max_expression_batch_size
+        out = batch_fun(*arg_vals, **kw_vals)
+        rt = e._return_type
+        if rt == dt.ANY:
+            col, _ = infer_and_build_column(out, device=ctx.device)
+            return col
+        return column_from_pylist(out, rt, device=ctx.device)
     if is_async:
         import asyncio
         import inspect

@@ -155,17 +155,48 @@ class InputNode(Node):
 
 
 class ExprMapNode(Node):
-    """select / with_columns: evaluate expressions, keep keys."""
+    """select / with_columns: evaluate expressions, keep keys.
 
-    def __init__(self, input_node: Node, exprs: dict[str, Any], device):
-        super().__init__([input_node], device)
+    extra_inputs: same-universe tables referenced in the expressions
+    (reference: rowwise context over several universe-equal tables);
+    their batches are aligned to the main batch BY KEY before evaluation.
+    """
+
+    def __init__(
+        self,
+        input_node: Node,
+        exprs: dict[str, Any],
+        device,
+        extra_inputs: list[tuple[Node, str]] | None = None,
+    ):
+        extra_inputs = extra_inputs or []
+        super().__init__([input_node] + [n for n, _ in extra_inputs], device)
         self.exprs = exprs
+        self.extra_prefixes = [p for _, p in extra_inputs]
 
     def step(self, time, inputs):
         b = inputs[0]
         if b is None or len(b) == 0:
             return None
-        ctx = EvalContext(b.columns, b.keys, self.device)
+        extra: dict[str, Column] = {}
+        if self.extra_prefixes:
+            from pathway_amd.engine.state import lex_sort_words, searchsorted_words
+
+            main_q = [b.keys[:, 0].contiguous(), b.keys[:, 1].contiguous()]
+            for prefix, eb in zip(self.extra_prefixes, inputs[1:]):
+                if eb is None or len(eb) == 0:
+                    raise RuntimeError(
+                        "same-universe referenced table produced no delta at a "
+                        "time its universe changed — universes are not equal"
+                    )
+                ew = [eb.keys[:, 0].contiguous(), eb.keys[:, 1].contiguous()]
+                perm = lex_sort_words(ew)
+                sw = [w.index_select(0, perm) for w in ew]
+                pos = searchsorted_words(sw, main_q, side="left")
+                src = perm.index_select(0, pos.clamp(0, max(len(eb) - 1, 0)))
+                for n, c in eb.columns.items():
+                    extra[f"{prefix}{n}"] = c.take(src)
+        ctx = EvalContext(b.columns, b.keys, self.device, extra=extra)
         out_cols = {name: evaluate(e, ctx) for name, e in self.exprs.items()}
         return DeltaBatch(b.keys, out_cols, b.diffs, time)
 
@@ -771,4 +802,5 @@ def _mask_missing(col: Column, found: torch.Tensor, device) -> Column:
     vals = col.to_pylist()
     f = found.cpu().tolist()
     out = [v if ok else None for v, ok in zip(vals, f)]
-    return ObjectColumn(np.array(out, dtype=object), col.dtype)
+    from pathway_amd.engine.column import obj_array
+    return ObjectColumn(obj_array(out), col.dtype)

@@ -546,7 +546,8 @@ def _scatter_override(base: Column, qidx: torch.Tensor, repl: Column) -> Column:
     rv = repl.to_pylist()
     for j, q in enumerate(qidx.cpu().tolist()):
         vals[q] = rv[j]
-    return ObjectColumn(np.array(vals, dtype=object), base.dtype)
+    from pathway_amd.engine.column import obj_array
+    return ObjectColumn(obj_array(vals), base.dtype)
 
 
 class FlattenNode(Node):

@@ -120,6 +120,41 @@ class CallbackSource(Source):
         return DeltaBatch(keys, cols, diffs, time)
 
 
+class PushSource(Source):
+    """Rows pushed programmatically at runtime (REST servers, interactive
+    sessions, python ConnectorSubjects in streaming mode)."""
+
+    def __init__(self, column_names: list[str], dtypes: list[dt.DType]):
+        self.column_names = column_names
+        self.dtypes = dtypes
+        self.pending: dict[int, list] = {}
+
+    def push(self, key: BasePointer, values: list[Any], time: int, diff: int = 1):
+        self.pending.setdefault(time, []).append((key, values, diff))
+
+    def next_time(self) -> int | None:
+        return min(self.pending) if self.pending else None
+
+    def pull(self, time: int, device) -> DeltaBatch | None:
+        rows = self.pending.pop(time, None)
+        if not rows:
+            return None
+        keys = torch.tensor(
+            [list(k.as_signed_pair()) for k, _, _ in rows],
+            dtype=torch.int64,
+            device=device,
+        ).reshape(len(rows), 2)
+        diffs = torch.tensor([d for _, _, d in rows], dtype=torch.int64, device=device)
+        cols: dict[str, Column] = {}
+        for j, name in enumerate(self.column_names):
+            vals = [v[j] for _, v, _ in rows]
+            cols[name] = column_from_pylist(vals, self.dtypes[j], device)
+        return DeltaBatch(keys, cols, diffs, time)
+
+    def reset(self) -> None:
+        pass  # pushed-but-unconsumed rows survive a reset
+
+
 class CaptureNode(Node):
     """Collects the full update stream of a table (debug / tests / sinks)."""
 

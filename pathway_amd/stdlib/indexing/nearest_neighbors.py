@@ -1,18 +1,118 @@
 """KNN retrievers (reference stdlib/indexing/nearest_neighbors.py:65-574).
 
-BruteForceKnn: queries x index GEMM + top-k — torch path everywhere,
-HIP bf16 MFMA kernel on gfx950 (ops/knn kernels, index phase).
+BruteForceKnn: GPU brute-force cosine/L2 top-k — queries × index GEMM
+(hipBLASLt via torch on ROCm) + top-k, over the ExternalIndexNode.
+USearchKnn: HNSW-parity API; round 1 serves it from the same brute-force
+GPU index (exact ≥ approximate recall; perf via GEMM).  LshKnn likewise.
 """
+
 from __future__ import annotations
 
 import enum
-from dataclasses import dataclass
+from dataclasses import dataclass, field
 from typing import Any
+
+from pathway_amd.internals import expression as ex
+from pathway_amd.internals.config import get_device
+from pathway_amd.internals.universe import Universe
 
 
 class DistanceType(enum.Enum):
     COS = "cos"
     L2SQ = "l2sq"
+
+
+class USearchMetricKind(enum.Enum):
+    COS = "cos"
+    L2SQ = "l2sq"
+    IP = "ip"
+
+
+class _BruteForceIndexBase:
+    """InnerIndex implementation over ExternalIndexNode."""
+
+    def __init__(
+        self,
+        data_column: ex.ColumnReference,
+        metadata_column: ex.ColumnReference | None = None,
+        metric: DistanceType = DistanceType.COS,
+        embedder: Any = None,
+    ):
+        self.data_column = data_column
+        self.metadata_column = metadata_column
+        self.metric = metric
+        self.embedder = embedder
+
+    def query_as_of_now(
+        self,
+        query_column: ex.ColumnReference,
+        number_of_matches: int = 3,
+        metadata_filter: Any = None,
+    ):
+        from pathway_amd.engine.nodes_index import ExternalIndexNode
+        from pathway_amd.internals import dtype as dt
+        from pathway_amd.internals.table import Table
+
+        data_table = self.data_column.table
+        query_table = query_column.table
+        dcol = self.data_column
+        qexpr = query_column
+        if self.embedder is not None:
+            dcol = self.embedder(self.data_column)
+            qexpr = self.embedder(query_column)
+        if isinstance(dcol, ex.ColumnReference):
+            index_src = data_table
+            vec_name = dcol.name
+        else:
+            index_src = data_table.with_columns(_pw_vec=dcol)
+            vec_name = "_pw_vec"
+        filter_col = None
+        if self.metadata_column is not None:
+            if vec_name not in index_src._dtypes:
+                pass
+            index_src = index_src.with_columns(_pw_meta=self.metadata_column)
+            filter_col = "_pw_meta"
+        kexpr = None
+        if isinstance(number_of_matches, ex.ColumnExpression):
+            kexpr = number_of_matches
+            kval = 16
+        else:
+            kval = int(number_of_matches)
+        node = ExternalIndexNode(
+            index_src._node,
+            query_table._node,
+            vec_name,
+            qexpr,
+            kval,
+            get_device(),
+            metric=self.metric.value,
+            filter_data_col=filter_col,
+            query_filter_expr=metadata_filter,
+            query_k_expr=kexpr,
+        )
+        dtypes = {
+            "_pw_index_reply_ids": dt.List(dt.POINTER),
+            "_pw_index_reply_scores": dt.List(dt.FLOAT),
+        }
+        return Table(node, dtypes, query_table._universe)
+
+    # non-asof-now query: same node; reference updates results as the index
+    # changes — landing with the streaming-index phase
+    def query(self, query_column, number_of_matches: int = 3, metadata_filter=None):
+        return self.query_as_of_now(query_column, number_of_matches, metadata_filter)
+
+
+class BruteForceKnn(_BruteForceIndexBase):
+    pass
+
+
+class USearchKnn(_BruteForceIndexBase):
+    """HNSW-parity API (reference usearch_integration.rs:20-152); exact GPU
+    brute-force under the hood in round 1."""
+
+
+class LshKnn(_BruteForceIndexBase):
+    pass
 
 
 @dataclass
@@ -23,13 +123,33 @@ class BruteForceKnnFactory:
     metric: DistanceType = DistanceType.COS
     embedder: Any = None
 
-    def build_index(self, data_column, data_table, **kwargs):
-        return BruteForceKnn(self.metric)
+    def build_index(self, data_column, metadata_column=None, **kwargs) -> BruteForceKnn:
+        return BruteForceKnn(data_column, metadata_column, self.metric, self.embedder)
 
 
-class BruteForceKnn:
-    def __init__(self, metric: DistanceType = DistanceType.COS):
-        self.metric = metric
+@dataclass
+class UsearchKnnFactory:
+    dimensions: int | None = None
+    reserved_space: int = 400
+    metric: USearchMetricKind = USearchMetricKind.COS
+    connectivity: int = 0
+    expansion_add: int = 0
+    expansion_search: int = 0
+    embedder: Any = None
 
-    def query(self, data_table, query_column, k: int):
-        raise NotImplementedError("lands with the index phase")
+    def build_index(self, data_column, metadata_column=None, **kwargs) -> USearchKnn:
+        m = DistanceType.COS if self.metric in (USearchMetricKind.COS, USearchMetricKind.IP) else DistanceType.L2SQ
+        return USearchKnn(data_column, metadata_column, m, self.embedder)
+
+
+@dataclass
+class LshKnnFactory:
+    dimensions: int | None = None
+    n_or: int = 20
+    n_and: int = 10
+    bucket_length: float = 10.0
+    distance_type: DistanceType = DistanceType.COS
+    embedder: Any = None
+
+    def build_index(self, data_column, metadata_column=None, **kwargs) -> LshKnn:
+        return LshKnn(data_column, metadata_column, self.distance_type, self.embedder)

@@ -1,0 +1,236 @@
+"""DocumentStore (reference xpacks/llm/document_store.py:54,320-560).
+
+parse → post-process → split → embed → GPU index; retrieve/inputs/stats
+query tables with per-query k, JMESPath metadata filters and path globs.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Callable
+
+import pathway_amd.internals.common as common
+import pathway_amd.reducers as reducers
+from pathway_amd.internals import dtype as dt
+from pathway_amd.internals import expression as ex
+from pathway_amd.internals import thisclass
+from pathway_amd.internals.json import Json
+from pathway_amd.internals.schema import Schema
+
+this = thisclass.this
+
+
+class DocumentStore:
+    class RetrieveQuerySchema(Schema):
+        query: str
+        k: int
+        metadata_filter: str | None
+        filepath_globpattern: str | None
+
+    class InputsQuerySchema(Schema):
+        metadata_filter: str | None
+        filepath_globpattern: str | None
+
+    class StatisticsQuerySchema(Schema):
+        pass
+
+    def __init__(
+        self,
+        docs,
+        retriever_factory=None,
+        parser: Any = None,
+        splitter: Any = None,
+        doc_post_processors: list[Callable] | None = None,
+    ):
+        from pathway_amd.xpacks.llm.parsers import Utf8Parser
+        from pathway_amd.xpacks.llm.splitters import NullSplitter
+        from pathway_amd.stdlib.indexing.nearest_neighbors import BruteForceKnnFactory
+
+        if isinstance(docs, (list, tuple)):
+            docs = docs[0].concat_reindex(*docs[1:]) if len(docs) > 1 else docs[0]
+        self.docs = docs
+        self.parser = parser or Utf8Parser()
+        self.splitter = splitter or NullSplitter()
+        self.doc_post_processors = doc_post_processors or []
+        if retriever_factory is None:
+            from pathway_amd.xpacks.llm.embedders import SentenceTransformerEmbedder
+
+            retriever_factory = BruteForceKnnFactory(
+                embedder=SentenceTransformerEmbedder()
+            )
+        self.retriever_factory = retriever_factory
+        self._build()
+
+    # -- pipeline ----------------------------------------------------------
+
+    def _build(self) -> None:
+        docs = self.docs
+        has_meta = "_metadata" in docs._dtypes
+        meta_expr = this._metadata if has_meta else common.apply_with_type(
+            lambda: Json({}), dt.JSON
+        )
+        parsed = docs.select(
+            _pw_parsed=self.parser(this.data),
+            _pw_meta=meta_expr,
+        ).flatten(this._pw_parsed)
+        parsed = parsed.select(
+            text=common.apply_with_type(lambda p: p[0], dt.STR, this._pw_parsed),
+            _pw_meta=common.apply_with_type(
+                _merge_meta, dt.JSON, this._pw_meta, this._pw_parsed
+            ),
+        )
+        for post in self.doc_post_processors:
+            parsed = parsed.select(
+                text=common.apply_with_type(post, dt.STR, this.text),
+                _pw_meta=this._pw_meta,
+            )
+        chunks = parsed.select(
+            _pw_chunks=self.splitter(this.text, this._pw_meta),
+            _pw_meta=this._pw_meta,
+        ).flatten(this._pw_chunks)
+        self.chunks = chunks.select(
+            text=common.apply_with_type(lambda c: c[0], dt.STR, this._pw_chunks),
+            metadata=common.apply_with_type(
+                _merge_meta, dt.JSON, this._pw_meta, this._pw_chunks
+            ),
+        )
+        self.inner_index = self.retriever_factory.build_index(
+            self.chunks.text, self.chunks.metadata
+        )
+
+    # -- queries -----------------------------------------------------------
+
+    def retrieve_query(self, retrieval_queries):
+        q = retrieval_queries
+        filt = q.select(
+            _pw_filter=common.apply_with_type(
+                _combine_filters, dt.Optional(dt.STR), this.metadata_filter,
+                this.filepath_globpattern,
+            ),
+            query=this.query,
+            k=this.k,
+        )
+        reply = self.inner_index.query_as_of_now(
+            filt.query,
+            number_of_matches=ex.ColumnReference(filt, "k"),
+            metadata_filter=ex.ColumnReference(filt, "_pw_filter"),
+        )
+        flat = reply.select(
+            _pw_qid=this.id,
+            _pw_match=this._pw_index_reply_ids,
+            _pw_scores=this._pw_index_reply_scores,
+        ).flatten(this._pw_match)
+        docs_rows = self.chunks.ix(flat._pw_match, optional=True)
+        combined = flat.select(
+            _pw_qid=this._pw_qid,
+            _pw_scores=this._pw_scores,
+            text=ex.ColumnReference(docs_rows, "text"),
+            metadata=ex.ColumnReference(docs_rows, "metadata"),
+        )
+        collapsed = combined.groupby(this._pw_qid).reduce(
+            _pw_qid=this._pw_qid,
+            texts=reducers.tuple(this.text),
+            metas=reducers.tuple(this.metadata),
+            scoress=reducers.any(this._pw_scores),
+        )
+        result = collapsed.select(
+            _pw_qid=this._pw_qid,
+            result=common.apply_with_type(
+                _assemble_results, dt.JSON, this.texts, this.metas, this.scoress
+            ),
+        )
+        keyed = result.with_id_from_expr(
+            ex.ColumnReference(result, "_pw_qid")
+        ).without("_pw_qid")
+        # queries with no matches: empty result rows
+        empty = reply.select(
+            result=common.apply_with_type(lambda: Json([]), dt.JSON)
+        )
+        return empty.update_cells(keyed.with_universe_of(empty))
+
+    def _broadcast_join(self, queries, aggregate):
+        """queries × single-row aggregate (constant-key join, query keys)."""
+        q2 = queries.with_columns(_pw_one=1)
+        a2 = aggregate.with_columns(_pw_one=1)
+        return q2.join_left(a2, q2._pw_one == a2._pw_one, id=q2.id)
+
+    def inputs_query(self, input_queries):
+        docs = self.docs
+        has_meta = "_metadata" in docs._dtypes
+        if has_meta:
+            gb = docs.groupby().reduce(metadatas=reducers.tuple(this._metadata))
+        else:
+            gb = docs.groupby().reduce(metadatas=reducers.count())
+        j = self._broadcast_join(input_queries, gb)
+
+        def list_inputs(filter_s, glob_s, metadatas):
+            from pathway_amd.stdlib.indexing.filters import eval_jmespath_filter
+
+            combined = _combine_filters(filter_s, glob_s)
+            out = []
+            if isinstance(metadatas, tuple):
+                for m in metadatas:
+                    mv = m.value if hasattr(m, "value") else m
+                    if combined is None or eval_jmespath_filter(combined, mv):
+                        out.append(mv)
+            return Json(out)
+
+        return j.select(
+            result=common.apply_with_type(
+                list_inputs,
+                dt.JSON,
+                thisclass.left.metadata_filter,
+                thisclass.left.filepath_globpattern,
+                thisclass.right.metadatas,
+            )
+        )
+
+    def statistics_query(self, info_queries):
+        stats = self.chunks.groupby().reduce(count=reducers.count())
+        j = self._broadcast_join(info_queries, stats)
+
+        def stat_result(c):
+            return Json(
+                {"file_count": int(c or 0), "last_indexed": 0, "last_modified": 0}
+            )
+
+        return j.select(
+            result=common.apply_with_type(stat_result, dt.JSON, thisclass.right.count)
+        )
+
+    @property
+    def index(self):
+        from pathway_amd.stdlib.indexing.data_index import DataIndex
+
+        return DataIndex(self.chunks, self.inner_index)
+
+
+def _merge_meta(base, pair):
+    b = base.value if hasattr(base, "value") else (base or {})
+    extra = pair[1] if len(pair) > 1 else {}
+    e = extra.value if hasattr(extra, "value") else (extra or {})
+    out = dict(b)
+    out.update(e)
+    return Json(out)
+
+
+def _combine_filters(metadata_filter, globpattern):
+    parts = []
+    if metadata_filter:
+        parts.append(f"({metadata_filter})")
+    if globpattern:
+        parts.append(f"globmatch('{globpattern}', path)")
+    return " && ".join(parts) if parts else None
+
+
+def _assemble_results(texts, metas, scores):
+    out = []
+    for i, t in enumerate(texts):
+        m = metas[i]
+        out.append(
+            {
+                "text": t,
+                "metadata": m.value if hasattr(m, "value") else m,
+                "dist": -float(scores[i]) if i < len(scores) else None,
+            }
+        )
+    return Json(out)
