@@ -147,3 +147,63 @@ def test_native_lib_is_used_on_gpu():
     lo, hi = hashing.hash128_words([w, w])
     ref_lo, ref_hi = hashing.xxh64_words([w.cpu(), w.cpu()], 0), None
     assert torch.equal(lo.cpu(), ref_lo)
+
+
+@gpu
+@requires_cuda
+def test_embedder_gpu_bf16():
+    from pathway_amd.xpacks.llm._encoder import get_encoder
+
+    enc = get_encoder(device="cuda:0")
+    assert enc.dtype == torch.bfloat16
+    out = enc.encode(["hello world", "hello world", "different text"])
+    import numpy as np
+
+    assert np.allclose(out[0], out[1])
+    assert not np.allclose(out[0], out[2])
+    assert abs(float(np.linalg.norm(out[0])) - 1.0) < 1e-2
+
+
+@gpu
+@requires_cuda
+def test_knn_index_gpu():
+    from pathway_amd.engine.nodes_index import VectorIndexState
+
+    st = VectorIndexState(torch.device("cuda:0"), metric="cos")
+    vecs = torch.eye(4, dtype=torch.float32).cuda()
+    keys = torch.arange(8, dtype=torch.int64).reshape(4, 2).cuda()
+    st.update(keys, vecs, torch.ones(4, dtype=torch.int64).cuda())
+    q = torch.tensor([[0.0, 1.0, 0.05, 0.0]]).cuda()
+    ids, scores, _ = st.search(q, 2)
+    assert ids[0, 0].cpu().tolist() == [2, 3]  # second basis vector's key
+
+
+@gpu
+@requires_cuda
+def test_document_store_gpu():
+    from pathway_amd.internals.config import pathway_config
+
+    pathway_config.device = "cuda:0"
+    try:
+        from pathway_amd.debug import table_from_rows, table_to_dicts
+        from pathway_amd.internals.schema import schema_from_types
+        from pathway_amd.xpacks.llm.document_store import DocumentStore
+
+        schema = schema_from_types(data=bytes, _metadata=dict)
+        docs = table_from_rows(
+            schema,
+            [
+                (b"gpu streaming dataflow engine", {"path": "a.txt"}),
+                (b"other thing entirely", {"path": "b.txt"}),
+            ],
+        )
+        store = DocumentStore(docs)
+        queries = table_from_rows(
+            DocumentStore.RetrieveQuerySchema, [("streaming dataflow", 1, None, None)]
+        )
+        keys, cols = table_to_dicts(store.retrieve_query(queries))
+        rv = cols["result"][keys[0]]
+        rv = rv.value if hasattr(rv, "value") else rv
+        assert len(rv) == 1 and "streaming" in rv[0]["text"]
+    finally:
+        pathway_config.device = None

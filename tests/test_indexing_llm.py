@@ -154,3 +154,39 @@ def test_rag_answerer():
     out = cols["result"][keys[0]]
     assert out.startswith("ECHO: ")
     assert "answerable content" in out
+
+
+def test_document_store_rest_server():
+    import json
+    import urllib.request
+
+    from pathway_amd.xpacks.llm.document_store import DocumentStore
+    from pathway_amd.xpacks.llm.servers import DocumentStoreServer
+    from pathway_amd.xpacks.llm.vector_store import VectorStoreClient
+
+    schema = schema_from_types(data=bytes, _metadata=dict)
+    docs = table_from_rows(
+        schema,
+        [
+            (b"alpha document about streaming", {"path": "a.txt"}),
+            (b"beta document about graphs", {"path": "b.txt"}),
+        ],
+    )
+    store = DocumentStore(docs)
+    port = 18231
+    srv = DocumentStoreServer("127.0.0.1", port, store)
+    th = srv.run(threaded=True)
+    try:
+        client = VectorStoreClient(port=port)
+        out = client.query("streaming", k=1)
+        assert isinstance(out, list) and len(out) == 1
+        assert "streaming" in out[0]["text"]
+        stats = client.get_vectorstore_statistics()
+        assert stats["file_count"] == 2
+        files = client.get_input_files()
+        assert len(files) == 2
+        # second query exercises incremental stepping
+        out2 = client.query("graphs", k=1)
+        assert "graphs" in out2[0]["text"]
+    finally:
+        srv.shutdown()
