@@ -55,6 +55,10 @@ class StaticSource(Source):
     def reset(self) -> None:
         self._pending = sorted(self.by_time.keys())
 
+    def seek(self, threshold_time: int) -> None:
+        """Skip data at times ≤ threshold (recovery rewind, mod.rs:567)."""
+        self._pending = [t for t in self._pending if t > threshold_time]
+
     def next_time(self) -> int | None:
         return self._pending[0] if self._pending else None
 
@@ -154,6 +158,14 @@ class PushSource(Source):
     def reset(self) -> None:
         pass  # pushed-but-unconsumed rows survive a reset
 
+    def seek(self, threshold_time: int) -> None:
+        self.pending = {t: r for t, r in self.pending.items() if t > threshold_time}
+
+
+#: process-wide "replay in progress" flag — sinks suppress re-emission of
+#: already-delivered outputs during recovery replay
+REPLAY_ACTIVE = [False]
+
 
 class CaptureNode(Node):
     """Collects the full update stream of a table (debug / tests / sinks)."""
@@ -168,7 +180,7 @@ class CaptureNode(Node):
 
     def step(self, time, inputs):
         b = inputs[0]
-        if b is None or len(b) == 0:
+        if b is None or len(b) == 0 or REPLAY_ACTIVE[0]:
             return None
         b = consolidate_batch(b)
         if b is None:
@@ -189,7 +201,7 @@ class OutputNode(Node):
 
     def step(self, time, inputs):
         b = inputs[0]
-        if b is None or len(b) == 0:
+        if b is None or len(b) == 0 or REPLAY_ACTIVE[0]:
             return None
         b = consolidate_batch(b)
         if b is not None:
@@ -220,7 +232,7 @@ class SubscribeNode(Node):
 
     def step(self, time, inputs):
         b = inputs[0]
-        if b is None or len(b) == 0:
+        if b is None or len(b) == 0 or REPLAY_ACTIVE[0]:
             return None
         b = consolidate_batch(b)
         if b is None:
@@ -265,12 +277,16 @@ class Runtime:
     """Single-worker synchronous runtime; multi-GPU coordination lives in
     parallel/exchange.py (each rank runs its own Runtime in lockstep)."""
 
-    def __init__(self, sinks: list[Node], device="cpu", comm=None):
+    def __init__(self, sinks: list[Node], device="cpu", comm=None, persistence=None):
         self.device = torch.device(device)
         self.nodes = topo_order(sinks)
         self.sources = [n for n in self.nodes if isinstance(n, InputNode)]
         self.sinks = sinks
         self.comm = comm  # parallel context or None
+        self.persistence = persistence
+        for i, src in enumerate(self.sources):
+            if not getattr(src, "persistent_id", None):
+                src.persistent_id = getattr(src.source, "name", None) or f"src{i}"
 
     def _next_time(self) -> int | None:
         times = [s.source.next_time() for s in self.sources]
@@ -280,11 +296,17 @@ class Runtime:
             local = self.comm.allreduce_min_time(local)
         return local
 
-    def step_once(self, time: int) -> None:
+    def step_once(self, time: int, injected: dict[int, DeltaBatch | None] | None = None) -> None:
         outputs: dict[int, DeltaBatch | None] = {}
+        pm = self.persistence
         for node in self.nodes:
             if isinstance(node, InputNode):
-                out = node.step(time, [])
+                if injected is not None:
+                    out = injected.get(id(node))
+                else:
+                    out = node.step(time, [])
+                    if pm is not None and out is not None:
+                        pm.record(node.persistent_id, time, out)
             else:
                 ins = [outputs.get(id(i)) for i in node.inputs]
                 if all(b is None for b in ins) and not node.wants_frontier():
@@ -298,13 +320,61 @@ class Runtime:
             if isinstance(node, OutputNode):
                 node.flush(time)
 
+    def replay_persisted(self) -> None:
+        """Recovery: push snapshotted input batches through the graph
+        (sinks suppressed), then seek live sources past the threshold."""
+        pm = self.persistence
+        if pm is None or pm.threshold_time < 0:
+            return
+        from pathway_amd.engine.column import infer_and_build_column
+        from pathway_amd.internals.api import BasePointer
+
+        by_time: dict[int, dict[int, DeltaBatch]] = {}
+        for src in self.sources:
+            for time, block in pm.replay_blocks(src.persistent_id):
+                names = block["names"]
+                rows = block["rows"]
+                keys = torch.tensor(
+                    [
+                        list(BasePointer(lo, hi).as_signed_pair())
+                        for (lo, hi), _, _ in rows
+                    ],
+                    dtype=torch.int64,
+                    device=self.device,
+                ).reshape(len(rows), 2)
+                diffs = torch.tensor(
+                    [d for _, _, d in rows], dtype=torch.int64, device=self.device
+                )
+                cols = {}
+                for j, n in enumerate(names):
+                    vals = [v[j] for _, v, _ in rows]
+                    cols[n], _ = infer_and_build_column(vals, self.device)
+                by_time.setdefault(time, {})[id(src)] = DeltaBatch(
+                    keys, cols, diffs, time
+                )
+        REPLAY_ACTIVE[0] = True
+        try:
+            for t in sorted(by_time):
+                self.step_once(t, injected=by_time[t])
+        finally:
+            REPLAY_ACTIVE[0] = False
+        for src in self.sources:
+            seek = getattr(src.source, "seek", None)
+            if seek is not None:
+                seek(pm.threshold_time)
+
     def run(self, max_steps: int | None = None) -> None:
         steps = 0
+        if self.persistence is not None and not getattr(self, "_replayed", False):
+            self.replay_persisted()
+            self._replayed = True
         while True:
             t = self._next_time()
             if t is None:
                 break
             self.step_once(t)
+            if self.persistence is not None:
+                self.persistence.commit(t)
             steps += 1
             if max_steps is not None and steps >= max_steps:
                 break

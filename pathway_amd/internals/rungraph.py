@@ -25,15 +25,29 @@ class RunGraph:
         self.sinks.clear()
         self.error_log_tables.clear()
 
-    def run_nodes(self, extra_sinks: list[Any] | None = None, monitoring=None, **kwargs):
+    def run_nodes(
+        self,
+        extra_sinks: list[Any] | None = None,
+        monitoring=None,
+        persistence_config=None,
+        **kwargs,
+    ):
         from pathway_amd.engine.runtime import Runtime
 
         sinks = list(self.sinks) + list(extra_sinks or [])
         if not sinks:
             return None
-        rt = Runtime(sinks, device=get_device(), comm=self.comm)
+        pm = None
+        if persistence_config is not None:
+            from pathway_amd.persistence.engine import PersistenceManager
+            from pathway_amd.internals.config import pathway_config
+
+            pm = PersistenceManager(persistence_config, worker=pathway_config.process_id)
+        rt = Runtime(sinks, device=get_device(), comm=self.comm, persistence=pm)
         reset_all(rt.nodes)
         rt.run()
+        if pm is not None:
+            pm.close()
         return rt
 
 
@@ -60,7 +74,9 @@ def run(
     **kwargs: Any,
 ):
     """pw.run(): execute every registered output (reference internals/run.py:13)."""
-    return G.run_nodes(monitoring=monitoring_level)
+    return G.run_nodes(
+        monitoring=monitoring_level, persistence_config=persistence_config
+    )
 
 
 def run_all(**kwargs: Any):
