@@ -1,0 +1,176 @@
+"""Temporal-behavior nodes: buffer / forget / freeze
+(reference src/engine/dataflow/operators/time_column.rs:163-660;
+Graph ops forget/buffer/freeze, graph.rs:753-793).
+
+Watermark model: the watermark of a "time column" is the max value seen so
+far (per worker; allreduce-max across workers when a comm is active).
+  BufferNode: hold rows until watermark ≥ threshold(row)   (delay)
+  ForgetNode: retract rows once watermark > threshold(row) (cutoff), and
+              drop late arrivals whose threshold ≤ current watermark
+  FreezeNode: drop BOTH late arrivals and any updates to frozen rows
+"""
+
+from __future__ import annotations
+
+from typing import Any
+
+import torch
+
+from pathway_amd.engine.batch import DeltaBatch
+from pathway_amd.engine.column import Column, TensorColumn, concat_columns
+from pathway_amd.engine.expression_eval import EvalContext, evaluate
+from pathway_amd.engine.nodes import Node, consolidate_batch
+
+
+def _threshold_tensor(expr, b: DeltaBatch, device) -> torch.Tensor:
+    ctx = EvalContext(b.columns, b.keys, device)
+    col = evaluate(expr, ctx)
+    assert isinstance(col, TensorColumn), "time/threshold column must be numeric"
+    t = col.tensor
+    if t.dtype != torch.int64:
+        t = t.to(torch.float64)
+    return t
+
+
+class _WatermarkMixin:
+    def _init_wm(self):
+        self.watermark: float | None = None
+
+    def _advance_watermark(self, values: torch.Tensor) -> None:
+        if values.numel():
+            m = float(values.max())
+            self.watermark = m if self.watermark is None else max(self.watermark, m)
+
+
+class BufferNode(Node, _WatermarkMixin):
+    """Delay rows until the watermark passes their threshold (buffer)."""
+
+    def __init__(self, input_node: Node, threshold_expr: Any, time_expr: Any, device):
+        super().__init__([input_node], device)
+        self.threshold_expr = threshold_expr
+        self.time_expr = time_expr
+        self._init_wm()
+        self.held: list[tuple[DeltaBatch, torch.Tensor]] = []
+
+    def reset(self):
+        self._init_wm()
+        self.held = []
+
+    def wants_frontier(self) -> bool:
+        return True  # may release held rows even without new input
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        device = self.device
+        if b is not None and len(b):
+            thr = _threshold_tensor(self.threshold_expr, b, device)
+            now = _threshold_tensor(self.time_expr, b, device)
+            self._advance_watermark(now)
+            self.held.append((b, thr))
+        if self.watermark is None or not self.held:
+            return None
+        out = []
+        still = []
+        for hb, thr in self.held:
+            ready = thr <= self.watermark
+            ridx = ready.nonzero(as_tuple=True)[0]
+            hidx = (~ready).nonzero(as_tuple=True)[0]
+            if ridx.numel():
+                rb = hb.take(ridx)
+                rb.time = time
+                out.append(rb)
+            if hidx.numel():
+                still.append((hb.take(hidx), thr.index_select(0, hidx)))
+        self.held = still
+        if not out:
+            return None
+        return consolidate_batch(DeltaBatch.concat(out))
+
+
+class ForgetNode(Node, _WatermarkMixin):
+    """Retract rows whose threshold falls behind the watermark; drop late
+    arrivals (cutoff behavior / `forget`)."""
+
+    def __init__(
+        self,
+        input_node: Node,
+        threshold_expr: Any,
+        time_expr: Any,
+        device,
+        mark_forgetting_records: bool = False,
+    ):
+        super().__init__([input_node], device)
+        self.threshold_expr = threshold_expr
+        self.time_expr = time_expr
+        self._init_wm()
+        # live rows, with their thresholds (kept until forgotten)
+        self.live: list[tuple[DeltaBatch, torch.Tensor]] = []
+
+    def reset(self):
+        self._init_wm()
+        self.live = []
+
+    def wants_frontier(self) -> bool:
+        return True
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        device = self.device
+        out = []
+        if b is not None and len(b):
+            thr = _threshold_tensor(self.threshold_expr, b, device)
+            now = _threshold_tensor(self.time_expr, b, device)
+            self._advance_watermark(now)
+            # drop LATE rows (threshold already passed)
+            fresh = thr > (self.watermark if self.watermark is not None else float("-inf"))
+            fidx = fresh.nonzero(as_tuple=True)[0]
+            if fidx.numel():
+                fb = b.take(fidx)
+                out.append(fb)
+                self.live.append((fb, thr.index_select(0, fidx)))
+        # forget rows that just fell behind the watermark
+        if self.watermark is not None and self.live:
+            still = []
+            for lb, thr in self.live:
+                gone = thr <= self.watermark
+                gidx = gone.nonzero(as_tuple=True)[0]
+                kidx = (~gone).nonzero(as_tuple=True)[0]
+                if gidx.numel():
+                    gb = lb.take(gidx)
+                    out.append(DeltaBatch(gb.keys, gb.columns, -gb.diffs, time))
+                if kidx.numel():
+                    still.append((lb.take(kidx), thr.index_select(0, kidx)))
+            self.live = still
+        if not out:
+            return None
+        for o in out:
+            o.time = time
+        return consolidate_batch(DeltaBatch.concat(out))
+
+
+class FreezeNode(Node, _WatermarkMixin):
+    """Drop updates (insertions AND retractions) for frozen rows."""
+
+    def __init__(self, input_node: Node, threshold_expr: Any, time_expr: Any, device):
+        super().__init__([input_node], device)
+        self.threshold_expr = threshold_expr
+        self.time_expr = time_expr
+        self._init_wm()
+
+    def reset(self):
+        self._init_wm()
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        device = self.device
+        thr = _threshold_tensor(self.threshold_expr, b, device)
+        now = _threshold_tensor(self.time_expr, b, device)
+        wm = self.watermark if self.watermark is not None else float("-inf")
+        keep = thr > wm
+        self._advance_watermark(now)
+        kidx = keep.nonzero(as_tuple=True)[0]
+        if not kidx.numel():
+            return None
+        return b.take(kidx)

@@ -264,3 +264,77 @@ def test_groupby_stream_incremental_updates():
     rt.run()
     stream = sorted((r.time, r.diff, r.values[1]) for r in cap.rows)
     assert stream == [(0, 1, 1), (2, -1, 1), (2, 1, 2)]
+
+
+def test_window_behavior_cutoff():
+    # late row (t=1 arriving after watermark passed its window end + cutoff)
+    t = T(
+        """
+        t  | v | __time__
+        1  | 1 | 0
+        12 | 5 | 2
+        1  | 100 | 4
+        """
+    )
+    res = t.windowby(
+        pw.this.t,
+        window=pw.temporal.tumbling(duration=5),
+        behavior=pw.temporal.common_behavior(cutoff=2, keep_results=True),
+    ).reduce(start=pw.this._pw_window_start, s=pw.reducers.sum(pw.this.v))
+    # the t=1/v=100 row is late (watermark 12 > window_end 5 + cutoff 2)
+    expected = T(
+        """
+        start | s
+        0     | 1
+        10    | 5
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_window_behavior_forget():
+    t = T(
+        """
+        t  | v | __time__
+        1  | 1 | 0
+        12 | 5 | 2
+        """
+    )
+    res = t.windowby(
+        pw.this.t,
+        window=pw.temporal.tumbling(duration=5),
+        behavior=pw.temporal.common_behavior(cutoff=2, keep_results=False),
+    ).reduce(start=pw.this._pw_window_start, s=pw.reducers.sum(pw.this.v))
+    # first window's results are dropped once the watermark passes end+cutoff
+    expected = T(
+        """
+        start | s
+        10    | 5
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_window_behavior_delay_buffer():
+    t = T(
+        """
+        t | v | __time__
+        1 | 1 | 0
+        2 | 2 | 2
+        9 | 3 | 4
+        """
+    )
+    res = t.windowby(
+        pw.this.t,
+        window=pw.temporal.tumbling(duration=5),
+        behavior=pw.temporal.common_behavior(delay=3),
+    ).reduce(start=pw.this._pw_window_start, s=pw.reducers.sum(pw.this.v))
+    # rows buffered until watermark >= window_start+3; all eventually released
+    expected = T(
+        """
+        start | s
+        0     | 3
+        5     | 3
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
