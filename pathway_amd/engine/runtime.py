@@ -166,6 +166,10 @@ class PushSource(Source):
 #: already-delivered outputs during recovery replay
 REPLAY_ACTIVE = [False]
 
+#: streaming-source next_time() sentinels
+STREAM_READY = object()   # has data now, any commit time works
+STREAM_WAITING = object()  # alive but idle — the runtime should wait
+
 
 class CaptureNode(Node):
     """Collects the full update stream of a table (debug / tests / sinks)."""
@@ -284,17 +288,34 @@ class Runtime:
         self.sinks = sinks
         self.comm = comm  # parallel context or None
         self.persistence = persistence
+        self._clock = 0
         for i, src in enumerate(self.sources):
             if not getattr(src, "persistent_id", None):
                 src.persistent_id = getattr(src.source, "name", None) or f"src{i}"
 
-    def _next_time(self) -> int | None:
-        times = [s.source.next_time() for s in self.sources]
-        times = [t for t in times if t is not None]
-        local = min(times) if times else None
+    def _next_time(self):
+        """Returns (time, waiting): time=None & waiting=False means done."""
+        BIG = 2**62
+        raw = [s.source.next_time() for s in self.sources]
+        numeric = [t for t in raw if isinstance(t, int)]
+        ready = any(t is STREAM_READY for t in raw)
+        waiting = any(t is STREAM_WAITING for t in raw)
+        if numeric:
+            local = min(numeric)
+        elif ready:
+            local = self._clock
+        elif waiting:
+            local = BIG - 1
+        else:
+            local = BIG
         if self.comm is not None:
-            local = self.comm.allreduce_min_time(local)
-        return local
+            enc = self.comm.allreduce_min_time(local)
+            local = enc if enc is not None else BIG
+        if local >= BIG:
+            return None, False
+        if local == BIG - 1:
+            return None, True
+        return local, False
 
     def step_once(self, time: int, injected: dict[int, DeltaBatch | None] | None = None) -> None:
         outputs: dict[int, DeltaBatch | None] = {}
@@ -368,11 +389,17 @@ class Runtime:
         if self.persistence is not None and not getattr(self, "_replayed", False):
             self.replay_persisted()
             self._replayed = True
+        import time as _time
+
         while True:
-            t = self._next_time()
+            t, waiting = self._next_time()
             if t is None:
-                break
+                if not waiting:
+                    break
+                _time.sleep(0.005)
+                continue
             self.step_once(t)
+            self._clock = max(self._clock, t + 2)
             if self.persistence is not None:
                 self.persistence.commit(t)
             steps += 1

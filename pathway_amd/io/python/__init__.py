@@ -24,7 +24,11 @@ class ConnectorSubject:
         raise NotImplementedError
 
     def next(self, **kwargs: Any) -> None:
-        self._buffer.append((self._time, kwargs))
+        emit = getattr(self, "_emit", None)
+        if emit is not None:
+            emit(kwargs)
+        else:
+            self._buffer.append((self._time, kwargs))
 
     def next_json(self, message: dict) -> None:
         self.next(**message)
@@ -51,19 +55,33 @@ def read(
     schema=None,
     autocommit_duration_ms: int | None = 1500,
     name: str | None = None,
+    mode: str = "streaming",
     **kwargs: Any,
 ):
-    from pathway_amd.debug import table_from_rows
+    """Streaming mode: subject.run() on a reader thread feeding the engine
+    live (reference Connector::run input thread, mod.rs:660)."""
+    from pathway_amd.engine.nodes import InputNode
+    from pathway_amd.engine.streaming import StreamingSource, spawn_reader
+    from pathway_amd.internals.config import get_device
+    from pathway_amd.internals.table import Table
+    from pathway_amd.internals.universe import Universe
 
-    subject.run()
-    subject.on_stop()
     names = schema.column_names()
-    rows = []
-    for t, rec in subject._buffer:
-        if rec is None:
-            continue
-        rows.append(tuple(rec.get(n) for n in names) + (t, 1))
-    return table_from_rows(schema, rows, is_stream=True)
+    dtypes = [schema.__columns__[n].dtype for n in names]
+    src = StreamingSource(names, dtypes, name=name)
+
+    subject._emit = lambda rec: src.emit([rec.get(n) for n in names])
+
+    def run_subject():
+        try:
+            subject.run()
+            subject.on_stop()
+        finally:
+            src.finish()
+
+    spawn_reader(run_subject)
+    node = InputNode(src, get_device())
+    return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 
 
 ConnectorObserver = ConnectorSubject
