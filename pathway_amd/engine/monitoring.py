@@ -1,0 +1,137 @@
+"""Progress monitoring + HTTP metrics
+(reference src/engine/dataflow/monitoring.rs ProberStats,
+src/engine/http_server.rs /metrics OpenMetrics + JSON status,
+python internals/monitoring.py rich dashboard).
+"""
+
+from __future__ import annotations
+
+import json
+import threading
+import time
+from collections import defaultdict
+from dataclasses import dataclass, field
+from typing import Any
+
+
+@dataclass
+class OperatorStats:
+    rows_in: int = 0
+    rows_out: int = 0
+    steps: int = 0
+    total_time_s: float = 0.0
+
+
+@dataclass
+class RunStats:
+    """Engine-wide counters, updated per step by the Runtime."""
+
+    started_at: float = field(default_factory=time.time)
+    current_time: int | None = None
+    steps: int = 0
+    rows_ingested: int = 0
+    rows_output: int = 0
+    step_latencies_ms: list = field(default_factory=list)
+    operators: dict = field(default_factory=lambda: defaultdict(OperatorStats))
+
+    def record_step(self, t: int, latency_s: float, ingested: int, output: int):
+        self.current_time = t
+        self.steps += 1
+        self.rows_ingested += ingested
+        self.rows_output += output
+        self.step_latencies_ms.append(latency_s * 1000)
+        if len(self.step_latencies_ms) > 1000:
+            self.step_latencies_ms = self.step_latencies_ms[-1000:]
+
+    def latency_quantile(self, q: float) -> float | None:
+        if not self.step_latencies_ms:
+            return None
+        xs = sorted(self.step_latencies_ms)
+        return xs[min(int(len(xs) * q), len(xs) - 1)]
+
+    def snapshot(self) -> dict:
+        return {
+            "uptime_s": time.time() - self.started_at,
+            "current_time": self.current_time,
+            "steps": self.steps,
+            "rows_ingested": self.rows_ingested,
+            "rows_output": self.rows_output,
+            "p50_step_ms": self.latency_quantile(0.5),
+            "p95_step_ms": self.latency_quantile(0.95),
+            "operators": {
+                name: vars(st) for name, st in self.operators.items()
+            },
+        }
+
+    def openmetrics(self) -> str:
+        s = self.snapshot()
+        lines = [
+            "# TYPE pathway_steps counter",
+            f"pathway_steps_total {s['steps']}",
+            "# TYPE pathway_rows_ingested counter",
+            f"pathway_rows_ingested_total {s['rows_ingested']}",
+            "# TYPE pathway_rows_output counter",
+            f"pathway_rows_output_total {s['rows_output']}",
+            "# TYPE pathway_step_latency_ms gauge",
+            f"pathway_step_latency_ms{{quantile=\"0.5\"}} {s['p50_step_ms'] or 0}",
+            f"pathway_step_latency_ms{{quantile=\"0.95\"}} {s['p95_step_ms'] or 0}",
+            "# EOF",
+        ]
+        return "\n".join(lines) + "\n"
+
+
+GLOBAL_STATS = RunStats()
+
+
+def start_http_server(stats: RunStats, port: int | None = None) -> Any:
+    """/metrics (OpenMetrics) + /status (JSON) — reference http_server.rs:
+    port 20000 + process id."""
+    import os
+    from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+    if port is None:
+        port = 20000 + int(os.environ.get("PATHWAY_PROCESS_ID", "0"))
+
+    class Handler(BaseHTTPRequestHandler):
+        def do_GET(self):
+            if self.path.startswith("/metrics"):
+                body = stats.openmetrics().encode()
+                ctype = "application/openmetrics-text; version=1.0.0"
+            else:
+                body = json.dumps(stats.snapshot(), default=str).encode()
+                ctype = "application/json"
+            self.send_response(200)
+            self.send_header("Content-Type", ctype)
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def log_message(self, *a):
+            pass
+
+    httpd = ThreadingHTTPServer(("0.0.0.0", port), Handler)
+    th = threading.Thread(target=httpd.serve_forever, daemon=True)
+    th.start()
+    return httpd
+
+
+class ConsoleMonitor:
+    """Lightweight terminal dashboard (reference internals/monitoring.py
+    rich table) — prints a periodic one-line progress summary."""
+
+    def __init__(self, stats: RunStats, interval_s: float = 5.0):
+        self.stats = stats
+        self.interval_s = interval_s
+        self._last = 0.0
+
+    def maybe_report(self):
+        now = time.time()
+        if now - self._last >= self.interval_s:
+            self._last = now
+            s = self.stats.snapshot()
+            print(
+                f"[pathway_amd] t={s['current_time']} steps={s['steps']} "
+                f"in={s['rows_ingested']} out={s['rows_output']} "
+                f"p95={s['p95_step_ms'] and round(s['p95_step_ms'], 2)}ms",
+                flush=True,
+            )

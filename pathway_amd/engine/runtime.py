@@ -289,6 +289,10 @@ class Runtime:
         self.comm = comm  # parallel context or None
         self.persistence = persistence
         self._clock = 0
+        from pathway_amd.engine.monitoring import RunStats
+
+        self.stats = RunStats()
+        self.monitor = None
         for i, src in enumerate(self.sources):
             if not getattr(src, "persistent_id", None):
                 src.persistent_id = getattr(src.source, "name", None) or f"src{i}"
@@ -317,8 +321,10 @@ class Runtime:
             return None, True
         return local, False
 
-    def step_once(self, time: int, injected: dict[int, DeltaBatch | None] | None = None) -> None:
+    def step_once(self, time: int, injected: dict[int, DeltaBatch | None] | None = None):
         outputs: dict[int, DeltaBatch | None] = {}
+        n_ingested = 0
+        n_output = 0
         pm = self.persistence
         for node in self.nodes:
             if isinstance(node, InputNode):
@@ -326,6 +332,8 @@ class Runtime:
                     out = injected.get(id(node))
                 else:
                     out = node.step(time, [])
+                    if out is not None:
+                        n_ingested += len(out)
                     if pm is not None and out is not None:
                         pm.record(node.persistent_id, time, out)
             else:
@@ -336,10 +344,15 @@ class Runtime:
                     out = node.step(time, ins)
             outputs[id(node)] = out
         for node in self.nodes:
+            if isinstance(node, (SubscribeNode, OutputNode, CaptureNode)):
+                b = outputs.get(id(node.inputs[0]))
+                if b is not None:
+                    n_output += len(b)
             if isinstance(node, SubscribeNode):
                 node.finish_time(time)
             if isinstance(node, OutputNode):
                 node.flush(time)
+        return n_ingested, n_output
 
     def replay_persisted(self) -> None:
         """Recovery: push snapshotted input batches through the graph
@@ -398,7 +411,11 @@ class Runtime:
                     break
                 _time.sleep(0.005)
                 continue
-            self.step_once(t)
+            _s0 = _time.perf_counter()
+            ingested, output = self.step_once(t)
+            self.stats.record_step(t, _time.perf_counter() - _s0, ingested, output)
+            if self.monitor is not None:
+                self.monitor.maybe_report()
             self._clock = max(self._clock, t + 2)
             if self.persistence is not None:
                 self.persistence.commit(t)

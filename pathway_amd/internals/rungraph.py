@@ -44,10 +44,21 @@ class RunGraph:
 
             pm = PersistenceManager(persistence_config, worker=pathway_config.process_id)
         rt = Runtime(sinks, device=get_device(), comm=self.comm, persistence=pm)
+        http_server = None
+        if kwargs.get("with_http_server"):
+            from pathway_amd.engine.monitoring import start_http_server
+
+            http_server = start_http_server(rt.stats)
+        if monitoring is not None and str(monitoring) not in ("none", "MonitoringLevel.NONE"):
+            from pathway_amd.engine.monitoring import ConsoleMonitor
+
+            rt.monitor = ConsoleMonitor(rt.stats)
         reset_all(rt.nodes)
         rt.run()
         if pm is not None:
             pm.close()
+        if http_server is not None:
+            http_server.shutdown()
         return rt
 
 
@@ -75,7 +86,9 @@ def run(
 ):
     """pw.run(): execute every registered output (reference internals/run.py:13)."""
     return G.run_nodes(
-        monitoring=monitoring_level, persistence_config=persistence_config
+        monitoring=monitoring_level,
+        persistence_config=persistence_config,
+        with_http_server=with_http_server,
     )
 
 
