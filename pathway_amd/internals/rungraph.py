@@ -16,6 +16,7 @@ class RunGraph:
     def __init__(self) -> None:
         self.sinks: list[Any] = []  # engine sink nodes (Output/Subscribe/Capture)
         self.error_log_tables: list[Any] = []
+        self.services: list[Any] = []  # rest_connector webservers
         self.comm = None  # parallel context (set by parallel.init)
 
     def add_sink(self, node: Any) -> None:
@@ -24,6 +25,7 @@ class RunGraph:
     def clear(self) -> None:
         self.sinks.clear()
         self.error_log_tables.clear()
+        self.services.clear()
 
     def run_nodes(
         self,
@@ -55,6 +57,23 @@ class RunGraph:
             rt.monitor = ConsoleMonitor(rt.stats)
         reset_all(rt.nodes)
         rt.run()
+        if self.services:
+            # serving mode: keep the engine live for rest_connector traffic
+            from pathway_amd.io.http import serve_forever
+
+            servers = serve_forever(self.services, rt)
+            if not kwargs.get("_serve_in_background"):
+                import time as _time
+
+                try:
+                    while True:
+                        _time.sleep(0.2)
+                        rt.run()  # drain any streaming sources
+                except KeyboardInterrupt:
+                    pass
+                finally:
+                    for s_ in servers:
+                        s_.shutdown()
         if pm is not None:
             pm.close()
         if http_server is not None:
@@ -89,6 +108,7 @@ def run(
         monitoring=monitoring_level,
         persistence_config=persistence_config,
         with_http_server=with_http_server,
+        **kwargs,
     )
 
 

@@ -1,11 +1,53 @@
 """pw.io — connectors (reference python/pathway/io, 46 modules).
 
-Round-1 set: csv, jsonlines, fs, plaintext, python (ConnectorSubject),
-subscribe, null.  Streaming connector runtime lands in the streaming phase;
-message-queue and DB connectors are stubbed with the reference API surface.
+Working offline: fs, csv, jsonlines, plaintext, python (ConnectorSubject),
+subscribe, sqlite, http (+rest_connector), null, debezium-envelope parsing.
+Service-backed connectors (kafka/nats/postgres/...) expose the reference
+API and activate when their client library is installed.
 """
-from pathway_amd.io import csv, fs, jsonlines, plaintext, python
+from pathway_amd.io import (
+    airbyte,
+    azure,
+    bigquery,
+    clickhouse,
+    csv,
+    debezium,
+    deltalake,
+    dynamodb,
+    elasticsearch,
+    fs,
+    gdrive,
+    http,
+    iceberg,
+    jsonlines,
+    kafka,
+    kinesis,
+    logstash,
+    minio,
+    mongodb,
+    mqtt,
+    mssql,
+    mysql,
+    nats,
+    null,
+    plaintext,
+    postgres,
+    pubsub,
+    pulsar,
+    python,
+    questdb,
+    rabbitmq,
+    redpanda,
+    s3,
+    sqlite,
+)
 from pathway_amd.io._subscribe import subscribe
-from pathway_amd.io import null
 
-__all__ = ["csv", "fs", "jsonlines", "plaintext", "python", "subscribe", "null"]
+__all__ = [
+    "airbyte", "azure", "bigquery", "clickhouse", "csv", "debezium",
+    "deltalake", "dynamodb", "elasticsearch", "fs", "gdrive", "http",
+    "iceberg", "jsonlines", "kafka", "kinesis", "logstash", "minio",
+    "mongodb", "mqtt", "mssql", "mysql", "nats", "null", "plaintext",
+    "postgres", "pubsub", "pulsar", "python", "questdb", "rabbitmq",
+    "redpanda", "s3", "sqlite", "subscribe",
+]

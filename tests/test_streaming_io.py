@@ -46,3 +46,53 @@ def test_python_connector_streaming():
     keys, cols = pw.debug.table_to_dicts(res)
     assert list(cols["s"].values()) == [10]
     assert list(cols["c"].values()) == [5]
+
+
+@pytest.mark.timeout(120)
+def test_rest_connector_serving():
+    import json
+    import urllib.request
+
+    from pathway_amd.internals.rungraph import G
+
+    webserver = pw.io.http.PathwayWebserver("127.0.0.1", 18341)
+    schema = schema_from_types(query=str)
+    queries, response_writer = pw.io.http.rest_connector(
+        webserver=webserver, schema=schema, route="/ask", delete_completed_queries=True
+    )
+    result = queries.select(result=pw.this.query.str.upper())
+    response_writer(result)
+    rt = pw.run(_serve_in_background=True)
+    try:
+        req = urllib.request.Request(
+            "http://127.0.0.1:18341/ask",
+            data=json.dumps({"query": "hello"}).encode(),
+            headers={"Content-Type": "application/json"},
+        )
+        with urllib.request.urlopen(req, timeout=10) as resp:
+            out = json.loads(resp.read())
+        assert out == "HELLO"
+    finally:
+        for ws in G.services:
+            if hasattr(ws, "_httpd"):
+                ws._httpd.shutdown()
+
+
+@pytest.mark.timeout(60)
+def test_sqlite_roundtrip(tmp_path):
+    import sqlite3
+
+    db = str(tmp_path / "t.db")
+    con = sqlite3.connect(db)
+    con.execute("CREATE TABLE src (a INTEGER, b TEXT)")
+    con.execute("INSERT INTO src VALUES (1, 'x'), (2, 'y')")
+    con.commit()
+    con.close()
+    schema = schema_from_types(a=int, b=str)
+    t = pw.io.sqlite.read(db, "src", schema)
+    res = t.select(a2=pw.this.a * 10, b=pw.this.b)
+    pw.io.sqlite.write(res, db, "dst")
+    pw.run()
+    con = sqlite3.connect(db)
+    rows = sorted(con.execute("SELECT a2, b FROM dst").fetchall())
+    assert rows == [(10, "x"), (20, "y")]
