@@ -308,6 +308,8 @@ class GroupReduceNode(Node):
 
     def reset(self) -> None:
         self.seq = 0
+        self._xmeta_add = {}
+        self._xmeta_ms = {}
         self.group_store = None
         self.add_keys = None
         self.add_accs = {}
@@ -428,7 +430,12 @@ class GroupReduceNode(Node):
                 tensors.update({f"acc.{n}": t for n, t in acc_deltas.items()})
             else:
                 dest = tensors = None
-            tensors, gcols_first = exchange_bundle(comm, dest, tensors, gcols_first)
+            if not hasattr(self, "_xmeta_add"):
+                self._xmeta_add = {}
+                self._xmeta_ms = {}
+            tensors, gcols_first = exchange_bundle(
+                comm, dest, tensors, gcols_first, meta_state=self._xmeta_add
+            )
             ms_exchanged = None
             if has_multiset:
                 # multiset values cannot be pre-combined: ship the raw rows
@@ -442,7 +449,9 @@ class GroupReduceNode(Node):
                     tensors_m = {"gkeys": gkeys, "rowkeys": rowkeys, "diffs": diffs}
                 else:
                     destm = tensors_m = flat = None
-                tensors_m, flat = exchange_bundle(comm, destm, tensors_m, flat)
+                tensors_m, flat = exchange_bundle(
+                    comm, destm, tensors_m, flat, meta_state=self._xmeta_ms
+                )
                 ms_exchanged = (tensors_m, flat)
             if tensors is None:
                 return None  # no rank had data this step

@@ -50,7 +50,7 @@ def _join_keys(exprs: list[Any], b: DeltaBatch, device) -> torch.Tensor:
     return torch.stack([lo, hi], dim=1)
 
 
-def _exchange_side(comm, b: DeltaBatch | None, jk: torch.Tensor | None, time: int):
+def _exchange_side(comm, b: DeltaBatch | None, jk: torch.Tensor | None, time: int, meta_state: dict | None = None):
     """All-to-all one join side's delta rows by join-key shard.
 
     Every rank must participate in the collective even with an empty local
@@ -58,11 +58,15 @@ def _exchange_side(comm, b: DeltaBatch | None, jk: torch.Tensor | None, time: in
     from pathway_amd.parallel.exchange import exchange_bundle, shard_of
 
     if b is None:
-        tensors, cols = exchange_bundle(comm, None, None, None)
+        tensors, cols = exchange_bundle(comm, None, None, None, meta_state=meta_state)
     else:
         dest = shard_of(jk, comm.world)
         tensors, cols = exchange_bundle(
-            comm, dest, {"jk": jk, "keys": b.keys, "diffs": b.diffs}, dict(b.columns)
+            comm,
+            dest,
+            {"jk": jk, "keys": b.keys, "diffs": b.diffs},
+            dict(b.columns),
+            meta_state=meta_state,
         )
     if tensors is None:
         return None, None
@@ -165,6 +169,8 @@ class JoinNode(Node):
         self.rstore = _SideStore(self.device)
         self._left_proto = None
         self._right_proto = None
+        self._xmeta_l = {}
+        self._xmeta_r = {}
 
     def wants_frontier(self) -> bool:
         from pathway_amd.parallel import get_comm
@@ -185,8 +191,11 @@ class JoinNode(Node):
 
         comm = get_comm()
         if comm is not None and comm.world > 1:
-            bl, jl = _exchange_side(comm, bl, jl, time)
-            br, jr = _exchange_side(comm, br, jr, time)
+            if not hasattr(self, "_xmeta_l"):
+                self._xmeta_l = {}
+                self._xmeta_r = {}
+            bl, jl = _exchange_side(comm, bl, jl, time, self._xmeta_l)
+            br, jr = _exchange_side(comm, br, jr, time, self._xmeta_r)
 
         if bl is not None and len(bl):
             self._left_proto = self._left_proto or dict(bl.columns)

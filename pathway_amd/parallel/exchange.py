@@ -88,6 +88,7 @@ def exchange_bundle(
     dest: torch.Tensor | None,
     tensors: dict[str, torch.Tensor] | None,
     columns: dict[str, Column] | None,
+    meta_state: dict | None = None,
 ) -> tuple[dict[str, torch.Tensor] | None, dict[str, Column] | None]:
     """Exchange rows by dest; tensors dict = named per-row tensors (first dim
     n); columns = engine Columns.  Returns received (tensors, columns).
@@ -95,11 +96,27 @@ def exchange_bundle(
     Pass None tensors/columns for "no local data": the rank still joins the
     collective; the bundle schema is learned from peers via a small
     metadata all-gather.  Returns (None, None) when NO rank had data.
+
+    meta_state: optional per-call-site dict caching the negotiated schema —
+    once every rank holds it, the per-step metadata all-gather is replaced
+    by a 1-int allreduce (hot-path cost on RCCL).
     """
     world = comm.world
     local_meta = None if tensors is None else _bundle_meta(tensors, columns or {})
-    metas = comm.all_to_all_objects([local_meta] * world)
-    merged = next((m for m in metas if m is not None), None)
+    cached = meta_state.get("meta") if meta_state is not None else None
+    if meta_state is not None:
+        # cheap agreement: does any rank still lack the schema?
+        have = 1 if (local_meta or cached) else 0
+        need_handshake = comm.allreduce_min_time(have) == 0
+    else:
+        need_handshake = True
+    if need_handshake:
+        metas = comm.all_to_all_objects([local_meta or cached] * world)
+        merged = next((m for m in metas if m is not None), None)
+    else:
+        merged = local_meta or cached
+    if meta_state is not None and merged is not None:
+        meta_state["meta"] = merged
     if local_meta is None:
         if merged is None:
             return None, None

@@ -1,54 +1,48 @@
 """Bellman-Ford shortest paths via pw.iterate (reference stdlib/graphs)."""
+
 from __future__ import annotations
 
 import math
 
+import pathway_amd.internals.common as common
 import pathway_amd.reducers as reducers
-from pathway_amd.internals import iterate as it
 from pathway_amd.internals import thisclass
+from pathway_amd.internals.iterate import run_iterate
 
 this = thisclass.this
+left = thisclass.left
+right = thisclass.right
 
 
 def bellman_ford(vertices, edges, iteration_limit: int = 50):
-    """vertices: table with is_source bool; edges: u, v, dist float.
-    Returns dist_from_source per vertex (float, inf if unreachable)."""
-    import pathway_amd.internals.common as common
-
+    """vertices: table with `vtx` id + `is_source` bool; edges: table with
+    u, v, dist (float).  Returns per-vertex `vtx`, `dist_from_source`."""
     d0 = vertices.select(
+        vtx=this.vtx,
         dist=common.if_else(this.is_source, 0.0, math.inf),
     )
 
-    def step(state):
-        e = edges.with_columns(_pw_uk=this.u)
-        du = state.ix(e._pw_uk, context=e)
-        cand = e.with_columns(reach=du.with_universe_of(e).dist + this.dist)
-        best = cand.groupby(cand.v).reduce(
-            _pw_v=this.v, best=reducers.min(this.reach)
+    def step(state, edges):
+        cand = edges.join(state, edges.u == state.vtx).select(
+            v=left.v, reach=right.dist + left.dist
         )
-        best_keyed = best.with_id_from_expr(
-            best._pw_v.to_column_expression()
-            if hasattr(best._pw_v, "to_column_expression")
-            else best._pw_v
+        best = cand.groupby(this.v).reduce(
+            vtx=this.v, best=reducers.min(this.reach)
         )
-        # actually key by the vertex pointer itself
-        improved = state.copy()
-        from pathway_amd.engine.nodes_join import KeyedMergeNode
-        # relax: new dist = min(old, best inbound)
-        joined = state.join_left(best, state.id == best._pw_v).select(
-            dist=common.coalesce(
-                common.if_else(
-                    thisclass.right.best.is_not_none()
-                    & (thisclass.right.best < thisclass.left.dist),
-                    thisclass.right.best,
-                    thisclass.left.dist,
-                ),
-                thisclass.left.dist,
+        relaxed = state.join_left(best, state.vtx == best.vtx).select(
+            vtx=left.vtx,
+            dist=common.if_else(
+                right.best.is_not_none() & (common.coalesce(right.best, math.inf) < left.dist),
+                common.coalesce(right.best, math.inf),
+                left.dist,
             ),
-            _pw_vid=thisclass.left.id,
         )
-        out = joined.with_id_from_expr(joined._pw_vid).without("_pw_vid")
-        return out
+        return relaxed
 
-    res = it.run_iterate(lambda state: step(state), iteration_limit=iteration_limit, state=d0)
-    return res.select(dist_from_source=this.dist)
+    res = run_iterate(
+        lambda state, edges: step(state, edges),
+        iteration_limit=iteration_limit,
+        state=d0,
+        edges=edges,
+    )
+    return res.select(vtx=this.vtx, dist_from_source=this.dist)
