@@ -431,11 +431,13 @@ def column_from_pylist(
     if base == dt.STR:
         return StringColumn.from_strings(list(values), device=device)
     if isinstance(base, dt.Pointer):
+        if has_none:
+            arr = np.empty(n, dtype=object)
+            for i, v in enumerate(values):
+                arr[i] = v
+            return ObjectColumn(arr, dtype)
         pairs = torch.tensor(
-            [
-                list(v.as_signed_pair()) if v is not None else [0, 0]
-                for v in values
-            ],
+            [list(v.as_signed_pair()) for v in values],
             dtype=torch.int64,
             device=device,
         ).reshape(n, 2)

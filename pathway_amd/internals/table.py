@@ -517,6 +517,54 @@ class Table(TableLike):
 
         return make_ix_table(query, self, pexpr, optional=optional)
 
+    def sort(self, key: Any, instance: Any = None) -> "Table":
+        """prev/next pointer columns by sort order (reference sort_table +
+        add_prev_next_pointers, operators/prev_next.rs:775)."""
+        from pathway_amd.engine.nodes_recompute import RecomputeNode
+
+        kref = self._resolve(key)
+        iref = self._resolve(instance) if instance is not None else None
+        if not isinstance(kref, ex.ColumnReference):
+            raise NotImplementedError("sort key must be a column")
+        kname = kref.name
+        iname = iref.name if isinstance(iref, ex.ColumnReference) else None
+        src = self
+
+        def fn(in_rows, in_keys):
+            rows, keys = in_rows[0], in_keys[0]
+            groups: dict = {}
+            for row, k in zip(rows, keys):
+                g = row.get(iname) if iname else None
+                groups.setdefault(g, []).append((row[kname], k))
+            out = []
+            for g, rl in groups.items():
+                rl.sort(key=lambda x: (x[0], repr(x[1])))
+                for i, (_, k) in enumerate(rl):
+                    out.append(
+                        (
+                            k,
+                            {
+                                "prev": rl[i - 1][1] if i > 0 else None,
+                                "next": rl[i + 1][1] if i + 1 < len(rl) else None,
+                            },
+                        )
+                    )
+            return out
+
+        out_dtypes = {
+            "prev": dt.Optional(dt.POINTER),
+            "next": dt.Optional(dt.POINTER),
+        }
+        node = RecomputeNode(
+            [src._node], fn, ["prev", "next"], out_dtypes, get_device()
+        )
+        return Table(node, out_dtypes, self._universe)
+
+    def diff(self, timestamp: Any, *values: Any, instance: Any = None) -> "Table":
+        from pathway_amd.stdlib.ordered import diff as _diff
+
+        return _diff(self, timestamp, *values, instance=instance)
+
     # -- misc ---------------------------------------------------------------
 
     def remove_errors(self) -> "Table":

@@ -166,12 +166,113 @@ class WindowedTable:
                 self._with_instance({}).keys()
             )
         if isinstance(w, SessionWindow):
-            raise NotImplementedError(
-                "session windows land with the streaming temporal phase"
-            )
+            return self._prepare_session(w)
         if isinstance(w, IntervalsOverWindow):
-            raise NotImplementedError("intervals_over lands with the temporal phase")
+            return self._prepare_intervals_over(w)
         raise TypeError(f"unknown window {w!r}")
+
+    def _prepare_session(self, w: "SessionWindow"):
+        """Session windows: merge rows whose gap ≤ max_gap (or predicate).
+
+        Order-dependent merge → RecomputeNode host path (reference
+        windows.rs session merge); downstream groupby stays incremental.
+        """
+        from pathway_amd.engine.nodes_recompute import RecomputeNode
+        from pathway_amd.internals import dtype as dt
+        from pathway_amd.internals.config import get_device
+        from pathway_amd.internals.universe import Universe
+
+        t = self._table
+        texpr = self._time_expr
+        if not isinstance(texpr, ex.ColumnExpression):
+            raise TypeError("session windowby needs a time expression")
+        # materialize the needed columns (time + instance + all original)
+        cols = {n: ex.ColumnReference(t, n) for n in t._dtypes}
+        cols["_pw_t"] = texpr
+        if self._instance is not None:
+            cols["_pw_instance"] = self._instance
+        src = t.select(**cols)
+        max_gap = _as_number(w.max_gap) if w.max_gap is not None else None
+        predicate = w.predicate
+        out_columns = list(src._dtypes.keys()) + [
+            "_pw_window_start",
+            "_pw_window_end",
+        ]
+
+        def fn(in_rows, in_keys):
+            rows, keys = in_rows[0], in_keys[0]
+            groups: dict = {}
+            for row, key in zip(rows, keys):
+                g = row.get("_pw_instance")
+                groups.setdefault(g, []).append((row["_pw_t"], row, key))
+            out = []
+            for g, rl in groups.items():
+                rl.sort(key=lambda x: x[0])
+                sess: list = []
+                for tv, row, key in rl:
+                    if sess:
+                        prev_t = sess[-1][0]
+                        merge = (
+                            predicate(prev_t, tv)
+                            if predicate is not None
+                            else (max_gap is not None and tv - prev_t <= max_gap)
+                        )
+                    else:
+                        merge = False
+                    if sess and not merge:
+                        out.extend(_emit_session(sess))
+                        sess = []
+                    sess.append((tv, row, key))
+                if sess:
+                    out.extend(_emit_session(sess))
+            return out
+
+        def _emit_session(sess):
+            start = sess[0][0]
+            end = sess[-1][0]
+            for tv, row, key in sess:
+                vals = dict(row)
+                vals["_pw_window_start"] = start
+                vals["_pw_window_end"] = end
+                yield (key, vals)
+
+        out_dtypes = dict(src._dtypes)
+        out_dtypes["_pw_window_start"] = src._dtypes.get("_pw_t", dt.ANY)
+        out_dtypes["_pw_window_end"] = src._dtypes.get("_pw_t", dt.ANY)
+        node = RecomputeNode([src._node], fn, out_columns, out_dtypes, get_device())
+        out = Table(node, out_dtypes, Universe())
+        gb = ["_pw_window_start", "_pw_window_end"]
+        if self._instance is not None:
+            gb.append("_pw_instance")
+        return out, gb
+
+    def _prepare_intervals_over(self, w: "IntervalsOverWindow"):
+        """intervals_over: windows [at+lb, at+ub] centered on the `at`
+        column's values — an interval join + groupby on the at-point."""
+        from pathway_amd.stdlib.temporal._interval_join import (
+            IntervalJoinResult,
+            interval,
+        )
+        from pathway_amd.internals import thisclass
+
+        at_ref = w.at
+        if not isinstance(at_ref, ex.ColumnReference):
+            raise TypeError("intervals_over(at=...) must be a column reference")
+        at_table = at_ref.table
+        t = self._table
+        lb = _as_number(w.lower_bound)
+        ub = _as_number(w.upper_bound)
+        ij = IntervalJoinResult(
+            at_table, t, at_ref, self._time_expr, interval(lb, ub), [], "inner"
+        )
+        sel = {n: ex.ColumnReference(thisclass.right, n) for n in t._dtypes}
+        sel["_pw_window_location"] = ex.ColumnReference(thisclass.left, at_ref.name)
+        joined = ij.select(**sel)
+        joined = joined.with_columns(
+            _pw_window_start=ex.ColumnReference(joined, "_pw_window_location") + lb,
+            _pw_window_end=ex.ColumnReference(joined, "_pw_window_location") + ub,
+        )
+        return joined, ["_pw_window_location", "_pw_window_start", "_pw_window_end"]
 
     def _retime(self, part: Table):
         """time expression rebound to the expanded table."""

@@ -338,3 +338,92 @@ def test_window_behavior_delay_buffer():
         """
     )
     assert_table_equality_wo_index(res, expected)
+
+
+def test_session_window():
+    t = T(
+        """
+        t  | v
+        1  | 1
+        2  | 2
+        3  | 3
+        10 | 4
+        11 | 5
+        """
+    )
+    res = t.windowby(
+        pw.this.t, window=pw.temporal.session(max_gap=2)
+    ).reduce(
+        start=pw.this._pw_window_start,
+        end=pw.this._pw_window_end,
+        s=pw.reducers.sum(pw.this.v),
+    )
+    expected = T(
+        """
+        start | end | s
+        1     | 3   | 6
+        10    | 11  | 9
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_intervals_over():
+    data = T(
+        """
+        t | v
+        1 | 1
+        3 | 2
+        5 | 4
+        """
+    )
+    probes = T(
+        """
+        pt
+        2
+        6
+        """
+    )
+    res = data.windowby(
+        pw.this.t,
+        window=pw.temporal.intervals_over(
+            at=probes.pt, lower_bound=-2, upper_bound=1
+        ),
+    ).reduce(
+        loc=pw.this._pw_window_location,
+        s=pw.reducers.sum(pw.this.v),
+    )
+    # at=2: t in [0,3] -> 1+2=3 ; at=6: t in [4,7] -> 4
+    expected = T(
+        """
+        loc | s
+        2   | 3
+        6   | 4
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
+
+
+def test_table_sort_prev_next():
+    t = T(
+        """
+        v
+        30
+        10
+        20
+        """
+    )
+    s = t.sort(pw.this.v)
+    keys, cols = pw.debug.table_to_dicts(s + t if False else s)
+    # row with v=10 has no prev; v=30 has no next
+    tkeys, tcols = pw.debug.table_to_dicts(t)
+    v_by_key = {k: tcols["v"][k] for k in tkeys}
+    for k in keys:
+        prev, nxt = cols["prev"][k], cols["next"][k]
+        v = v_by_key[k]
+        if v == 10:
+            assert prev is None and v_by_key[nxt] == 20
+        elif v == 20:
+            assert v_by_key[prev] == 10 and v_by_key[nxt] == 30
+        else:
+            assert v_by_key[prev] == 20 and nxt is None
