@@ -1,0 +1,75 @@
+"""HybridIndex: reciprocal-rank fusion of several retrievers
+(reference stdlib/indexing/hybrid_index.py:14)."""
+from __future__ import annotations
+
+from typing import Any
+
+import pathway_amd.internals.common as common
+from pathway_amd.internals import dtype as dt
+from pathway_amd.internals import thisclass
+
+this = thisclass.this
+
+
+class HybridIndex:
+    def __init__(self, retrievers: list, k: int = 60):
+        self.retrievers = retrievers
+        self.k = k  # RRF constant
+        self.embedder = None
+
+    def query_as_of_now(self, query_column, number_of_matches: int = 3, metadata_filter=None):
+        replies = [
+            r.query_as_of_now(
+                query_column,
+                number_of_matches=max(
+                    number_of_matches if isinstance(number_of_matches, int) else 16,
+                    10,
+                ),
+                metadata_filter=metadata_filter,
+            )
+            for r in self.retrievers
+        ]
+        base = replies[0]
+        sel = {
+            "ids0": this._pw_index_reply_ids,
+            "scores0": this._pw_index_reply_scores,
+        }
+        combined = base.select(**sel)
+        for i, rep in enumerate(replies[1:], start=1):
+            from pathway_amd.internals import expression as ex
+
+            combined = combined.with_columns(
+                **{
+                    f"ids{i}": ex.ColumnReference(
+                        rep.with_universe_of(combined), "_pw_index_reply_ids"
+                    ),
+                    f"scores{i}": ex.ColumnReference(
+                        rep.with_universe_of(combined), "_pw_index_reply_scores"
+                    ),
+                }
+            )
+        kk = self.k
+        nm = number_of_matches if isinstance(number_of_matches, int) else 3
+
+        def fuse(*id_lists):
+            scores: dict = {}
+            keep: dict = {}
+            for ids in id_lists:
+                for rank, p in enumerate(ids or ()):
+                    keep[repr(p)] = p
+                    scores[repr(p)] = scores.get(repr(p), 0.0) + 1.0 / (kk + rank + 1)
+            ranked = sorted(scores.items(), key=lambda kv: -kv[1])[:nm]
+            return tuple(keep[r] for r, _ in ranked), tuple(s for _, s in ranked)
+
+        id_args = [
+            thisclass.this[f"ids{i}"] for i in range(len(replies))
+        ]
+        fused = combined.select(
+            _pw_fused=common.apply_with_type(fuse, dt.ANY_TUPLE, *id_args)
+        )
+        return fused.select(
+            _pw_index_reply_ids=fused._pw_fused[0],
+            _pw_index_reply_scores=fused._pw_fused[1],
+        )
+
+    query = query_as_of_now

@@ -190,3 +190,49 @@ def test_document_store_rest_server():
         assert "graphs" in out2[0]["text"]
     finally:
         srv.shutdown()
+
+
+def test_bm25_index():
+    from pathway_amd.stdlib.indexing import TantivyBM25
+
+    schema = schema_from_types(text=str)
+    docs = table_from_rows(
+        schema,
+        [
+            ("the quick brown fox",),
+            ("lazy dogs sleep all day",),
+            ("quick reactions win races",),
+        ],
+    )
+    queries = table_from_rows(schema_from_types(q=str), [("quick fox",)])
+    idx = TantivyBM25(docs.text)
+    reply = idx.query_as_of_now(queries.q, number_of_matches=2)
+    keys, cols = table_to_dicts(reply)
+    ids = cols["_pw_index_reply_ids"][keys[0]]
+    dkeys, dcols = table_to_dicts(docs)
+    names = {k: dcols["text"][k] for k in dkeys}
+    got = [names[i] for i in ids]
+    assert got[0] == "the quick brown fox"
+    assert len(got) == 2
+
+
+def test_hybrid_index_rrf():
+    from pathway_amd.stdlib.indexing import HybridIndex, TantivyBM25, BruteForceKnn
+
+    schema = schema_from_types(text=str, vec=tuple)
+    docs = table_from_rows(
+        schema,
+        [
+            ("alpha beta", (1.0, 0.0)),
+            ("gamma delta", (0.0, 1.0)),
+        ],
+    )
+    queries = table_from_rows(
+        schema_from_types(q=str, qv=tuple), [("alpha", (0.9, 0.1))]
+    )
+    hybrid = HybridIndex([TantivyBM25(docs.text)])
+    reply = hybrid.query_as_of_now(queries.q, number_of_matches=1)
+    keys, cols = table_to_dicts(reply)
+    ids = cols["_pw_index_reply_ids"][keys[0]]
+    dkeys, dcols = table_to_dicts(docs)
+    assert dcols["text"][ids[0]] == "alpha beta"
