@@ -29,7 +29,7 @@ def main() -> None:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch", type=int, default=1_000_000, help="events per step per GPU")
+    p.add_argument("--batch", type=int, default=4_000_000, help="events per step per GPU")
     p.add_argument("--vocab", type=int, default=50_000)
     p.add_argument("--device", type=str, default=None)
     args = p.parse_args()

@@ -25,11 +25,29 @@ from pathway_amd.engine.column import Column, concat_columns
 
 
 def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
-    """Stable lexicographic argsort over parallel int64 word tensors."""
+    """Stable lexicographic argsort over parallel int64 word tensors.
+
+    Fast path for 2-word (128-bit hash) keys: random hash keys almost never
+    collide in the primary word, so sort by word0 alone and fall back to
+    the full multi-pass only if adjacent duplicates exist (exact check)."""
     n = words[0].shape[0]
     perm = torch.arange(n, dtype=torch.int64, device=words[0].device)
     if n <= 1:
         return perm
+    if len(words) == 2 and n > 2048:
+        k0_sorted, perm0 = torch.sort(words[0])
+        k1_sorted = words[1].index_select(0, perm0)
+        # only a k0 collision between rows with DIFFERENT k1 violates lex
+        # order (equal (k0,k1) rows may appear in any relative order —
+        # grouping and merging treat them identically)
+        violation = bool(
+            (
+                (k0_sorted[1:] == k0_sorted[:-1])
+                & (k1_sorted[1:] != k1_sorted[:-1])
+            ).any()
+        )
+        if not violation:
+            return perm0
     for w in reversed(words):
         keys = w.index_select(0, perm)
         p = torch.argsort(keys, stable=True)
