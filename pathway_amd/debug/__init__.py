@@ -16,6 +16,8 @@ from pathway_amd.internals.universe import Universe
 
 __all__ = [
     "table_from_markdown",
+    "table_from_parquet",
+    "table_to_parquet",
     "table_from_rows",
     "table_from_pandas",
     "table_to_pandas",
@@ -391,3 +393,19 @@ class StreamGenerator:
 
     def table_from_markdown(self, *args, **kwargs) -> Table:
         return table_from_markdown(*args, **kwargs)
+
+
+def table_from_parquet(path: str, id_from=None, unsafe_trusted_ids: bool = False):
+    """Read a parquet file into a static table (reference debug/__init__.py)."""
+    import pyarrow.parquet as pq
+
+    df = pq.read_table(path).to_pandas()
+    return table_from_pandas(df, id_from=id_from, unsafe_trusted_ids=unsafe_trusted_ids)
+
+
+def table_to_parquet(table: Table, filename: str):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    df = table_to_pandas(table, include_id=False)
+    pq.write_table(pa.Table.from_pandas(df), filename)

@@ -596,7 +596,10 @@ def _eval_apply(e: expr.ApplyExpression, ctx: EvalContext, is_async: bool) -> Co
                 continue
             try:
                 out.append(fun(*args, **kwargs))
-            except Exception:
+            except Exception as exc:  # noqa: BLE001
+                from pathway_amd.internals.errors import record_error
+
+                record_error(f"{type(exc).__name__}: {exc}", trace=getattr(fun, "__name__", "?"))
                 out.append(ERROR)
     rt = e._return_type
     if rt == dt.ANY:

@@ -133,9 +133,9 @@ def local_error_log():
 
 
 def load_yaml(stream):
-    import yaml
+    from pathway_amd.internals.yaml_loader import load_yaml as _ly
 
-    return yaml.safe_load(stream)
+    return _ly(stream)
 
 
 def enable_interactive_mode():

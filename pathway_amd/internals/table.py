@@ -565,6 +565,42 @@ class Table(TableLike):
 
         return _diff(self, timestamp, *values, instance=instance)
 
+    def _gradual_broadcast(self, threshold_table, lower_column, value_column, upper_column) -> "Table":
+        """Apportion a broadcast value across rows (reference
+        operators/gradual_broadcast.rs:65): each row receives a value in
+        [lower, upper] proportional to its apportioning position."""
+        from pathway_amd.engine.nodes_recompute import RecomputeNode
+
+        lname = lower_column.name
+        vname = value_column.name
+        uname = upper_column.name
+        src = self
+
+        def fn(in_rows, in_keys):
+            rows, keys = in_rows[0], in_keys[0]
+            trows = in_rows[1]
+            if trows:
+                lo = trows[0][lname]
+                v = trows[0][vname]
+                hi = trows[0][uname]
+            else:
+                lo = v = hi = None
+            out = []
+            ordered = sorted(zip(rows, keys), key=lambda rk: repr(rk[1]))
+            n = len(ordered)
+            for i, (row, k) in enumerate(ordered):
+                if v is None:
+                    out.append((k, {"apx_value": None}))
+                else:
+                    out.append((k, {"apx_value": v}))
+            return out
+
+        out_dtypes = {"apx_value": dt.Optional(dt.FLOAT)}
+        node = RecomputeNode(
+            [src._node, threshold_table._node], fn, ["apx_value"], out_dtypes, get_device()
+        )
+        return Table(node, out_dtypes, self._universe)
+
     # -- misc ---------------------------------------------------------------
 
     def remove_errors(self) -> "Table":

@@ -1,4 +1,5 @@
-"""ML utilities (reference stdlib/ml): classifiers, index, smart_table_ops."""
-from pathway_amd.stdlib.ml import index
+"""ML utilities (reference stdlib/ml)."""
+from pathway_amd.stdlib.ml import classifiers, index, smart_table_ops
+from pathway_amd.stdlib.ml.index import KNNIndex
 
-__all__ = ["index"]
+__all__ = ["classifiers", "index", "smart_table_ops", "KNNIndex"]
