@@ -328,12 +328,14 @@ class GroupReduceNode(Node):
         return out
 
     def _additive_names(self):
+        # count-style accumulators alias the presence weight __w__ (same
+        # per-row contribution: diffs) — no separate state arrays for them
         names = []
         for out_name, (spec, args, kwargs) in self._specs().items():
             if spec.family == "additive":
                 if spec.name == "avg":
-                    names += [f"{out_name}__sum", f"{out_name}__cnt"]
-                else:
+                    names.append(f"{out_name}__sum")
+                elif spec.name == "sum":
                     names.append(out_name)
         return names
 
@@ -502,9 +504,16 @@ class GroupReduceNode(Node):
         # 6. new output rows (post-merge)
         new_presence, new_cols = self._current_rows(changed, cw, specs)
 
-        # 7. emit
+        # 7. emit — sort-free consolidation: old/new rows are aligned on the
+        # same (sorted, unique) changed keys, so cancellation is a direct
+        # per-column equality compare instead of sort+vhash+segsum
+        unchanged = old_presence & new_presence
+        for name in old_cols:
+            unchanged = unchanged & columns_equal_mask(
+                old_cols[name], new_cols[name], device
+            )
         out_batches = []
-        idx = old_presence.nonzero(as_tuple=True)[0]
+        idx = (old_presence & ~unchanged).nonzero(as_tuple=True)[0]
         if idx.numel():
             out_batches.append(
                 DeltaBatch(
@@ -514,7 +523,7 @@ class GroupReduceNode(Node):
                     time,
                 )
             )
-        idx = new_presence.nonzero(as_tuple=True)[0]
+        idx = (new_presence & ~unchanged).nonzero(as_tuple=True)[0]
         if idx.numel():
             out_batches.append(
                 DeltaBatch(
@@ -526,7 +535,7 @@ class GroupReduceNode(Node):
             )
         if not out_batches:
             return None
-        return consolidate_batch(DeltaBatch.concat(out_batches))
+        return DeltaBatch.concat(out_batches)
 
     # -- state init --
 
@@ -549,7 +558,7 @@ class GroupReduceNode(Node):
             if spec.family != "additive":
                 continue
             if spec.name == "count":
-                contribs[out_name] = diffs
+                pass  # aliases __w__
             elif spec.name == "sum":
                 t = arg_cols[out_name][0]
                 assert isinstance(t, TensorColumn), "sum needs a numeric column"
@@ -566,7 +575,6 @@ class GroupReduceNode(Node):
                 if t.mask is not None:
                     v = v * t.mask
                 contribs[f"{out_name}__sum"] = v * diffs.to(torch.float64)
-                contribs[f"{out_name}__cnt"] = diffs
         return contribs
 
     def _merge_additive_pre(self, ukeys_w, acc_deltas):
@@ -615,7 +623,10 @@ class GroupReduceNode(Node):
         self._ensure_states(gcols_first, None)
         keys = torch.stack(ukeys_w, dim=1)
         v0, v1 = self._gcols_vhash(gcols_first, keys.shape[0])
-        self.group_store.merge(keys, (v0, v1), key_weights, gcols_first)
+        # group values define the group key ⇒ vhash is key-determined
+        self.group_store.merge(
+            keys, (v0, v1), key_weights, gcols_first, key_determined_vhash=True
+        )
 
     def _gcols_vhash(self, gcols: dict[str, Column], n: int):
         parts = [
@@ -677,20 +688,27 @@ class GroupReduceNode(Node):
                 cols[name] = None  # filled below if group_store empty
         # additive reducer outputs
         add_pos, add_found = self._additive_lookup(cw)
+        w_cache: torch.Tensor | None = None
+
+        def wacc():
+            nonlocal w_cache
+            if w_cache is None:
+                w_cache = self._gather_acc("__w__", add_pos, add_found)
+            return w_cache
+
         for out_name, (spec, args, kwargs) in specs.items():
             if spec.family == "additive":
                 if spec.name == "avg":
                     s = self._gather_acc(f"{out_name}__sum", add_pos, add_found)
-                    c = self._gather_acc(f"{out_name}__cnt", add_pos, add_found)
+                    c = wacc()
                     vals = s.to(torch.float64) / c.clamp(min=1).to(torch.float64)
                     cols[out_name] = TensorColumn(vals, dt.FLOAT)
+                elif spec.name == "count":
+                    cols[out_name] = TensorColumn(wacc().to(torch.int64), dt.INT)
                 else:
                     acc = self._gather_acc(out_name, add_pos, add_found)
-                    if spec.name == "count":
-                        cols[out_name] = TensorColumn(acc.to(torch.int64), dt.INT)
-                    else:
-                        odt = dt.FLOAT if acc.dtype == torch.float64 else dt.INT
-                        cols[out_name] = TensorColumn(acc, odt)
+                    odt = dt.FLOAT if acc.dtype == torch.float64 else dt.INT
+                    cols[out_name] = TensorColumn(acc, odt)
             elif spec.family == "multiset":
                 cols[out_name] = self._multiset_agg(out_name, spec, changed, nq)
         # presence from additive weight if group store empty (no group cols)
@@ -794,6 +812,34 @@ def _build_from_values(vals, device):
     from pathway_amd.engine.column import infer_and_build_column
 
     return infer_and_build_column(vals, device)
+
+
+def columns_equal_mask(a: Column, b: Column, device) -> torch.Tensor:
+    """Per-row equality between two aligned columns (None == None)."""
+    n = len(a)
+    if isinstance(a, TensorColumn) and isinstance(b, TensorColumn):
+        if a.tensor.dtype != b.tensor.dtype:
+            eq = a.tensor.to(torch.float64) == b.tensor.to(torch.float64)
+        else:
+            eq = a.tensor == b.tensor
+        ma = a.mask
+        mb = b.mask
+        if ma is None and mb is None:
+            return eq
+        ones = torch.ones(n, dtype=torch.bool, device=a.tensor.device)
+        ma = ma if ma is not None else ones
+        mb = mb if mb is not None else ones
+        return (ma & mb & eq) | (~ma & ~mb)
+    if isinstance(a, StringColumn) and isinstance(b, StringColumn) and a.pool is b.pool:
+        return a.codes == b.codes
+    if isinstance(a, PointerColumn) and isinstance(b, PointerColumn):
+        return (a.pairs[:, 0] == b.pairs[:, 0]) & (a.pairs[:, 1] == b.pairs[:, 1])
+    va, vb = a.to_pylist(), b.to_pylist()
+    return torch.tensor(
+        [x == y or (x is None and y is None) for x, y in zip(va, vb)],
+        dtype=torch.bool,
+        device=device,
+    )
 
 
 def _mask_missing(col: Column, found: torch.Tensor, device) -> Column:

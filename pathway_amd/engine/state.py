@@ -201,8 +201,13 @@ class Arrangement:
         vhash: tuple[torch.Tensor, torch.Tensor],
         weights: torch.Tensor,
         columns: dict[str, Column],
+        key_determined_vhash: bool = False,
     ) -> None:
-        """Merge a delta (unsorted ok) into the arrangement, consolidating."""
+        """Merge a delta (unsorted ok) into the arrangement, consolidating.
+
+        key_determined_vhash: caller guarantees vhash is a function of the
+        key (e.g. group stores: values define the key) — sorting by the key
+        words alone is then lexicographic for the full 4-word rows."""
         if keys.shape[0] == 0:
             return
         dwords = [keys[:, 0].contiguous(), keys[:, 1].contiguous(), vhash[0], vhash[1]]
@@ -211,7 +216,10 @@ class Arrangement:
         all_cols = {
             n: concat_columns([self.columns[n], columns[n]]) for n in self.columns
         }
-        perm = lex_sort_words(all_words)
+        if key_determined_vhash:
+            perm = lex_sort_words(all_words[:2])
+        else:
+            perm = lex_sort_words(all_words)
         all_words = [w.index_select(0, perm) for w in all_words]
         all_w = all_w.index_select(0, perm)
         all_cols = {n: c.take(perm) for n, c in all_cols.items()}
