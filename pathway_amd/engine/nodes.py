@@ -411,11 +411,30 @@ class GroupReduceNode(Node):
             nseg = int(first_idx.numel())
             ukeys_w = [w.index_select(0, first_idx) for w in swords]
             contribs = self._contributions(arg_cols, diffs, specs)
+            # segmented sums over the sorted batch via prefix-sum boundary
+            # differences — exact for int64, atomic-free (index_add_ was the
+            # contention hot spot at high duplicates-per-group)
+            nrows = perm.shape[0]
+            seg_ends = (
+                torch.cat(
+                    [
+                        first_idx[1:],
+                        torch.tensor([nrows], dtype=torch.int64, device=device),
+                    ]
+                )
+                - 1
+            )
             acc_deltas = {}
             for name, c in contribs.items():
                 sc = c.index_select(0, perm)
-                acc = torch.zeros(nseg, dtype=sc.dtype, device=device)
-                acc.index_add_(0, seg, sc)
+                if sc.dtype == torch.float64:
+                    acc = torch.zeros(nseg, dtype=sc.dtype, device=device)
+                    acc.index_add_(0, seg, sc)
+                else:
+                    cs = torch.cumsum(sc, 0)
+                    seg_tot = cs.index_select(0, seg_ends)
+                    acc = seg_tot.clone()
+                    acc[1:] -= seg_tot[:-1]
                 acc_deltas[name] = acc
             gfirst_rows = perm.index_select(0, first_idx)
             gcols_first = {n: c.take(gfirst_rows) for n, c in gcols.items()}
