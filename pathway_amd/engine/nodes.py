@@ -321,7 +321,47 @@ class GroupReduceNode(Node):
     def _specs(self) -> dict[str, tuple[ReducerSpec, list[Any], dict]]:
         out = {}
         for out_name, (rname, args, kwargs) in self.reducer_calls.items():
-            spec = REDUCERS.get(rname)
+            if rname == "stateful_many":
+                combine = kwargs["_combine_many"]
+
+                _n = max(len(args), 1)
+
+                def host_agg(rows, _c=combine, _n=_n):
+                    return _c(None, [(list(t)[:_n], w) for t, w in rows])
+
+                spec = ReducerSpec(
+                    name="stateful_many",
+                    family="multiset",
+                    n_args=len(args),
+                    host_agg=host_agg,
+                    out_dtype=lambda args_dt: dt.ANY,
+                )
+            elif rname == "udf_reducer":
+                acc_cls = kwargs["_accumulator_cls"]
+
+                _n = max(len(args), 1)
+
+                def host_agg(rows, _cls=acc_cls, _n=_n):
+                    acc = None
+                    for t, w in rows:
+                        row = list(t)[:_n]
+                        for _ in range(w):
+                            a = _cls.from_row(row)
+                            if acc is None:
+                                acc = a
+                            else:
+                                acc.update(a)
+                    return acc.compute_result() if acc is not None else None
+
+                spec = ReducerSpec(
+                    name="udf_reducer",
+                    family="multiset",
+                    n_args=len(args),
+                    host_agg=host_agg,
+                    out_dtype=lambda args_dt: dt.ANY,
+                )
+            else:
+                spec = REDUCERS.get(rname)
             if spec is None:
                 raise NotImplementedError(f"reducer {rname}")
             out[out_name] = (spec, args, kwargs)

@@ -75,7 +75,12 @@ class GroupedTable:
                     rargs = [rargs[0], _raw_ref("__seq__")]
                 out_name = f"_pw_r{counter[0]}"
                 counter[0] += 1
-                reducer_calls[out_name] = (rname, rargs, dict(e._kwargs))
+                rkw = dict(e._kwargs)
+                if hasattr(e, "_combine_many"):
+                    rkw["_combine_many"] = e._combine_many
+                if hasattr(e, "_accumulator_cls"):
+                    rkw["_accumulator_cls"] = e._accumulator_cls
+                reducer_calls[out_name] = (rname, rargs, rkw)
                 return _raw_ref(out_name)
             if isinstance(e, ex.ColumnReference):
                 if e.table is table or (
