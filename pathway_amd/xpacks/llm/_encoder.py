@@ -111,10 +111,21 @@ class NativeEncoder(torch.nn.Module):
     def tokenize(self, texts: list[str], max_len: int | None = None):
         c = self.cfg
         max_len = min(max_len or c.max_len, c.max_len)
+        cache = getattr(self, "_tok_cache", None)
+        if cache is None:
+            cache = self._tok_cache = {}
+        vocab_span = c.vocab - 3
         tok_rows = []
         for t in texts:
-            words = (t or "").lower().split()[: max_len]
-            ids = [3 + (xxh64(w.encode(), 77) % (c.vocab - 3)) for w in words]
+            words = (t or "").lower().split()[:max_len]
+            ids = []
+            for w in words:
+                tid = cache.get(w)
+                if tid is None:
+                    tid = 3 + (xxh64(w.encode(), 77) % vocab_span)
+                    if len(cache) < 1_000_000:
+                        cache[w] = tid
+                ids.append(tid)
             if not ids:
                 ids = [1]
             tok_rows.append(ids)
