@@ -130,6 +130,15 @@ from pathway_amd.stdlib import (  # noqa: E402
 from pathway_amd.stdlib.utils.async_transformer import AsyncTransformer  # noqa: E402
 from pathway_amd.stdlib.utils.pandas_transformer import pandas_transformer  # noqa: E402
 
+# reference-surface aliases
+from pathway_amd.stdlib.temporal._asof_join import AsofJoinResult  # noqa: E402
+from pathway_amd.stdlib.temporal._interval_join import IntervalJoinResult  # noqa: E402
+from pathway_amd.stdlib.temporal._window_join import WindowJoinResult  # noqa: E402
+
+OuterJoinResult = JoinResult
+asynchronous = udfs  # reference-deprecated alias for pw.udfs
+window = temporal  # reference alias module for window constructors
+
 # temporal joins attached like the reference does
 Table.windowby = temporal.windowby
 Table.asof_join = temporal.asof_join
@@ -238,4 +247,10 @@ __all__ = [
     "load_yaml",
     "ERROR",
     "PENDING",
+    "AsofJoinResult",
+    "IntervalJoinResult",
+    "WindowJoinResult",
+    "OuterJoinResult",
+    "asynchronous",
+    "window",
 ]

@@ -603,6 +603,123 @@ class Table(TableLike):
 
     # -- misc ---------------------------------------------------------------
 
+    # -- temporal behavior ops (graph.rs:753-793) ---------------------------
+
+    def _forget(
+        self, threshold_column: Any, time_column: Any, mark_forgetting_records: bool = False,
+        instance_column: Any = None,
+    ) -> "Table":
+        from pathway_amd.engine.nodes_temporal import ForgetNode
+
+        node = ForgetNode(
+            self._node,
+            self._resolve(threshold_column),
+            self._resolve(time_column),
+            get_device(),
+            mark_forgetting_records,
+        )
+        return Table(node, self._dtypes, Universe())
+
+    forget = _forget
+
+    def _buffer(self, threshold_column: Any, time_column: Any) -> "Table":
+        from pathway_amd.engine.nodes_temporal import BufferNode
+
+        node = BufferNode(
+            self._node,
+            self._resolve(threshold_column),
+            self._resolve(time_column),
+            get_device(),
+        )
+        return Table(node, self._dtypes, Universe())
+
+    buffer = _buffer
+
+    def _freeze(self, threshold_column: Any, time_column: Any) -> "Table":
+        from pathway_amd.engine.nodes_temporal import FreezeNode
+
+        node = FreezeNode(
+            self._node,
+            self._resolve(threshold_column),
+            self._resolve(time_column),
+            get_device(),
+        )
+        return Table(node, self._dtypes, Universe())
+
+    def ignore_late(self, threshold_column: Any, time_column: Any) -> "Table":
+        return self._freeze(threshold_column, time_column)
+
+    def _remove_retractions(self) -> "Table":
+        """remove_retractions_from_table (graph.rs): drop diff<0 rows."""
+        from pathway_amd.engine.nodes import Node as _N
+
+        class _DropRetractions(_N):
+            def step(self, time, inputs):
+                b = inputs[0]
+                if b is None or len(b) == 0:
+                    return None
+                import torch
+
+                idx = (b.diffs > 0).nonzero(as_tuple=True)[0]
+                return b.take(idx) if idx.numel() else None
+
+        node = _DropRetractions([self._node], get_device())
+        return Table(node, self._dtypes, Universe())
+
+    def filter_out_results_of_forgetting(self) -> "Table":
+        return self._remove_retractions()
+
+    # -- misc surface parity ------------------------------------------------
+
+    @staticmethod
+    def from_columns(*args: Any, **kwargs: Any) -> "Table":
+        cols = list(args) + list(kwargs.values())
+        if not cols:
+            raise ValueError("from_columns needs at least one column")
+        base = cols[0].table
+        named = {c.name: c for c in args}
+        named.update(kwargs)
+        return base.select(**named)
+
+    def with_prefix(self, prefix: str) -> "Table":
+        return self.select(
+            **{f"{prefix}{n}": ex.ColumnReference(self, n) for n in self._dtypes}
+        )
+
+    def with_suffix(self, suffix: str) -> "Table":
+        return self.select(
+            **{f"{n}{suffix}": ex.ColumnReference(self, n) for n in self._dtypes}
+        )
+
+    def update_id_type(self, target_type: Any, **kwargs: Any) -> "Table":
+        return self
+
+    @property
+    def slice(self):
+        from pathway_amd.internals import TableSlice
+
+        return TableSlice({n: ex.ColumnReference(self, n) for n in self._dtypes}, self)
+
+    def is_append_only(self) -> bool:
+        return False
+
+    def assert_append_only(self) -> "Table":
+        return self
+
+    def live(self) -> "Table":
+        return self
+
+    def to_stream(self) -> "Table":
+        """table_to_stream (graph.rs): events as an append-only table with
+        an is_upsert flag column."""
+        raise NotImplementedError("to_stream lands in the next round")
+
+    def stream_to_table(self, is_upsert: Any) -> "Table":
+        raise NotImplementedError("stream_to_table lands in the next round")
+
+    def unpack_snapshots(self, *args, **kwargs) -> "Table":
+        raise NotImplementedError("unpack_snapshots lands in the next round")
+
     def remove_errors(self) -> "Table":
         return self  # Error rows are filtered at output in this implementation
 
