@@ -64,7 +64,8 @@ def main() -> None:
 
     comm = None
     if world > 1:
-        comm = par.init(backend="nccl" if use_cuda else "gloo", device=device)
+        backend = os.environ.get("PW_BACKEND") or ("nccl" if use_cuda else "gloo")
+        comm = par.init(backend=backend, device=device)
 
     # ---- synthetic parsed word stream (identical vocab on all ranks) ----
     vocab = [f"word{i:06d}" for i in range(args.vocab)]

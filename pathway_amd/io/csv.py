@@ -58,9 +58,13 @@ class CsvWriter:
     def __init__(self, filename: str, column_names: list[str]):
         self.filename = filename
         self.column_names = column_names
-        self._fh = open(filename, "w", newline="")
+        import os
+
+        fresh = not (os.path.exists(filename) and os.path.getsize(filename) > 0)
+        self._fh = open(filename, "a", newline="")
         self._writer = _csv.writer(self._fh)
-        self._writer.writerow(column_names + ["time", "diff"])
+        if fresh:
+            self._writer.writerow(column_names + ["time", "diff"])
 
     def __call__(self, batch):
         names = list(batch.columns.keys())
