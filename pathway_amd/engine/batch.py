@@ -18,7 +18,7 @@ from pathway_amd.engine.column import Column, concat_columns
 
 
 class DeltaBatch:
-    __slots__ = ("keys", "columns", "diffs", "time")
+    __slots__ = ("keys", "columns", "diffs", "time", "consolidated")
 
     def __init__(
         self,
@@ -26,12 +26,16 @@ class DeltaBatch:
         columns: dict[str, Column],
         diffs: torch.Tensor,
         time: int,
+        consolidated: bool = False,
     ):
         assert keys.dim() == 2 and keys.shape[1] == 2, keys.shape
         self.keys = keys
         self.columns = columns
         self.diffs = diffs
         self.time = time
+        #: producer guarantees rows are already consolidated (no equal-row
+        #: cancellation possible) — sinks skip the consolidation sort
+        self.consolidated = consolidated
 
     def __len__(self) -> int:
         return int(self.keys.shape[0])

@@ -303,12 +303,16 @@ class StringColumn(Column):
     def value_hash(self) -> tuple[torch.Tensor, torch.Tensor]:
         lo_t, hi_t = self.pool.hash_tensors(self.codes.device)
         codes = self.codes
+        nlo, nhi = hashing.none_value_hash_scalar(codes.device)
+        if codes.is_cuda:
+            from pathway_amd import ops
+
+            return ops.pool_hash_gpu(codes, lo_t, hi_t, int(nlo), int(nhi))
         valid = codes >= 0
         safe = codes.clamp_min(0)
         lo = lo_t.index_select(0, safe)
         hi = hi_t.index_select(0, safe)
         # unconditional None-blend: avoids a device→host .all() sync per call
-        nlo, nhi = hashing.none_value_hash_scalar(codes.device)
         lo = torch.where(valid, lo, nlo)
         hi = torch.where(valid, hi, nhi)
         return lo, hi

@@ -103,6 +103,8 @@ def consolidate_batch(batch: DeltaBatch) -> DeltaBatch | None:
     """Sort by (key, vhash), sum diffs, drop zeros (consolidation.rs)."""
     if batch is None or len(batch) == 0:
         return None
+    if batch.consolidated:
+        return batch
     v0, v1 = batch_vhash(batch)
     words = [batch.keys[:, 0].contiguous(), batch.keys[:, 1].contiguous(), v0, v1]
     perm = lex_sort_words(words)
@@ -595,7 +597,9 @@ class GroupReduceNode(Node):
             )
         if not out_batches:
             return None
-        return DeltaBatch.concat(out_batches)
+        out = DeltaBatch.concat(out_batches)
+        out.consolidated = True
+        return out
 
     # -- state init --
 

@@ -228,3 +228,30 @@ def key_range_gpu(
     if rc != 0:
         raise RuntimeError(f"pw_key_range failed: hip error {rc}")
     return lo, hi
+
+
+def pool_hash_gpu(
+    codes: torch.Tensor,
+    pool_lo: torch.Tensor,
+    pool_hi: torch.Tensor,
+    none_lo: int,
+    none_hi: int,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    lib = require_lib()
+    n = codes.shape[0]
+    lo = torch.empty(n, dtype=torch.int64, device=codes.device)
+    hi = torch.empty(n, dtype=torch.int64, device=codes.device)
+    rc = lib.pw_pool_hash(
+        ctypes.c_void_p(codes.contiguous().data_ptr()),
+        ctypes.c_void_p(pool_lo.contiguous().data_ptr()),
+        ctypes.c_void_p(pool_hi.contiguous().data_ptr()),
+        ctypes.c_uint64(none_lo & ((1 << 64) - 1)),
+        ctypes.c_uint64(none_hi & ((1 << 64) - 1)),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(lo.data_ptr()),
+        ctypes.c_void_p(hi.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_pool_hash failed: hip error {rc}")
+    return lo, hi

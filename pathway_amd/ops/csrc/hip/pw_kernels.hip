@@ -386,3 +386,36 @@ extern "C" int pw_key_range(const void** sorted_ptrs, const void** query_ptrs,
                      (int64_t*)hi);
   return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------- pooled string hash --
+
+// Fused dictionary-column value hash: out = pool_hash[code] blended with
+// the canonical None hash for code < 0.  Replaces a clamp + 2 gathers +
+// 2 wheres + mask compare torch chain with one pass.
+__global__ void k_pool_hash(const int64_t* codes, const uint64_t* pool_lo,
+                            const uint64_t* pool_hi, uint64_t none_lo,
+                            uint64_t none_hi, int64_t n, uint64_t* lo,
+                            uint64_t* hi) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t c = codes[i];
+    if (c >= 0) {
+      lo[i] = pool_lo[c];
+      hi[i] = pool_hi[c];
+    } else {
+      lo[i] = none_lo;
+      hi[i] = none_hi;
+    }
+  }
+}
+
+extern "C" int pw_pool_hash(const void* codes, const void* pool_lo,
+                            const void* pool_hi, uint64_t none_lo,
+                            uint64_t none_hi, int64_t n, void* lo, void* hi,
+                            void* stream) {
+  hipLaunchKernelGGL(k_pool_hash, dim3(pw_grid(n)), dim3(PW_BLOCK), 0,
+                     (hipStream_t)stream, (const int64_t*)codes,
+                     (const uint64_t*)pool_lo, (const uint64_t*)pool_hi,
+                     none_lo, none_hi, n, (uint64_t*)lo, (uint64_t*)hi);
+  return (int)hipGetLastError();
+}
