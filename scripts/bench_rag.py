@@ -148,3 +148,43 @@ if __name__ == "__main__":
         bench_embedder()
     if which in ("all", "rag"):
         bench_rag_serving()
+    if which in ("all", "knn_e2e"):
+        bench_knn_e2e()
+
+
+def bench_knn_e2e(n_index=1_000_000, n_queries=1024, k=10, iters=10):
+    """BASELINE config 3 shape: bge-small-class embedder bf16 + cosine
+    top-k over a 1M-vector index — text in, neighbors out."""
+    from pathway_amd.engine.nodes_index import VectorIndexState
+    from pathway_amd.xpacks.llm._encoder import get_encoder
+
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    enc = get_encoder()
+    dim = enc.cfg.dim
+    st = VectorIndexState(torch.device(device), metric="cos")
+    g = torch.Generator(device="cpu").manual_seed(0)
+    vecs = torch.randn(n_index, dim, generator=g).to(device)
+    keys = torch.randint(-2**62, 2**62, (n_index, 2), dtype=torch.int64, generator=g).to(device)
+    st.update(keys, vecs, torch.ones(n_index, dtype=torch.int64, device=device))
+    texts = [f"query about topic {i} with some more words appended" for i in range(n_queries)]
+    # warmup
+    q = torch.from_numpy(np.stack(enc.encode(texts[:64]))).to(device)
+    st.search(q, k)
+    if device != "cpu":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        emb = torch.from_numpy(np.stack(enc.encode(texts))).to(device)
+        ids, scores, _ = st.search(emb, k)
+    if device != "cpu":
+        torch.cuda.synchronize()
+    dt_s = time.perf_counter() - t0
+    print(json.dumps({
+        "bench": "knn_e2e_embed_plus_topk",
+        "index_size": n_index, "dim": dim, "k": k,
+        "batch_queries": n_queries,
+        "qps": n_queries * iters / dt_s,
+        "ms_per_batch": dt_s / iters * 1000,
+        "dtype": "bf16 embedder + f32 scores",
+        "device": device,
+    }))
