@@ -138,20 +138,6 @@ def bench_rag_serving(n_docs=20000, n_queries=200):
     }))
 
 
-if __name__ == "__main__":
-    import sys
-
-    which = sys.argv[1] if len(sys.argv) > 1 else "all"
-    if which in ("all", "knn"):
-        bench_knn()
-    if which in ("all", "embedder"):
-        bench_embedder()
-    if which in ("all", "rag"):
-        bench_rag_serving()
-    if which in ("all", "knn_e2e"):
-        bench_knn_e2e()
-
-
 def bench_knn_e2e(n_index=1_000_000, n_queries=1024, k=10, iters=10):
     """BASELINE config 3 shape: bge-small-class embedder bf16 + cosine
     top-k over a 1M-vector index — text in, neighbors out."""
@@ -188,3 +174,17 @@ def bench_knn_e2e(n_index=1_000_000, n_queries=1024, k=10, iters=10):
         "dtype": "bf16 embedder + f32 scores",
         "device": device,
     }))
+
+
+if __name__ == "__main__":
+    import sys
+
+    which = sys.argv[1] if len(sys.argv) > 1 else "all"
+    if which in ("all", "knn"):
+        bench_knn()
+    if which in ("all", "embedder"):
+        bench_embedder()
+    if which in ("all", "rag"):
+        bench_rag_serving()
+    if which in ("all", "knn_e2e"):
+        bench_knn_e2e()
