@@ -176,6 +176,83 @@ class JoinResult(Joinable):
         dtypes = {n: infer_dtype(e, mid_dtypes) for n, e in final.items()}
         return Table(proj, dtypes, Universe())
 
+    def _all_columns_table(self):
+        """Materialize the join with every column of both sides (prefixed
+        names resolved) — backing for filter/groupby/reduce on JoinResult."""
+        left, right = self._left, self._right
+        sel = {}
+        for n in left._dtypes:
+            sel[n if n not in right._dtypes or n in self._unified else f"{n}_left"] = (
+                ex.ColumnReference(left, n)
+            )
+        for n in right._dtypes:
+            if n in self._unified:
+                continue
+            sel[n if n not in left._dtypes else f"{n}_right"] = ex.ColumnReference(
+                right, n
+            )
+        return self.select(**sel)
+
+    def filter(self, expression):
+        return self._all_columns_table().filter(
+            _remap_joined(expression, self._left, self._right, self._unified)
+        )
+
+    def groupby(self, *args, **kwargs):
+        t = self._all_columns_table()
+        args = [_remap_joined(a, self._left, self._right, self._unified) for a in args]
+        return t.groupby(*args, **kwargs)
+
+    def reduce(self, *args, **kwargs):
+        t = self._all_columns_table()
+        args = [_remap_joined(a, self._left, self._right, self._unified) for a in args]
+        kwargs = {
+            k: _remap_joined(v, self._left, self._right, self._unified)
+            for k, v in kwargs.items()
+        }
+        return t.reduce(*args, **kwargs)
+
+
+def _remap_joined(e, left, right, unified):
+    """Rebind left/right-table refs to the materialized joined table's
+    column names (suffix disambiguation as in the reference)."""
+    from pathway_amd.internals import expression as exm
+
+    e = thisclass.substitute_this(
+        exm.wrap_expr(e), {thisclass.left: left, thisclass.right: right}
+    )
+
+    def rec(x):
+        if isinstance(x, exm.ColumnReference):
+            t = x.table
+            if t is left or getattr(t, "_node", None) is getattr(left, "_node", None):
+                name = x.name
+                if name in right._dtypes and name not in unified:
+                    name = f"{name}_left"
+                return exm.ColumnReference(None, name)
+            if t is right or getattr(t, "_node", None) is getattr(right, "_node", None):
+                name = x.name
+                if name in left._dtypes and name not in unified:
+                    name = f"{name}_right"
+                return exm.ColumnReference(None, name)
+            return x
+        if not isinstance(x, exm.ColumnExpression):
+            return x
+        new = object.__new__(type(x))
+        new.__dict__.update(x.__dict__)
+        for attr, val in list(x.__dict__.items()):
+            if isinstance(val, exm.ColumnExpression):
+                new.__dict__[attr] = rec(val)
+            elif isinstance(val, tuple) and any(
+                isinstance(v, exm.ColumnExpression) for v in val
+            ):
+                new.__dict__[attr] = tuple(
+                    rec(v) if isinstance(v, exm.ColumnExpression) else v for v in val
+                )
+        return new
+
+    return rec(e)
+
 
 def make_ix_table(query, source, pexpr, optional: bool):
     """t.ix(expr): row of `source` addressed by pointer per `query` row."""

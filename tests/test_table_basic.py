@@ -420,3 +420,46 @@ def test_sql_groupby():
         """
     )
     assert_table_equality_wo_index(res, expected)
+
+
+def test_join_result_filter_and_reduce():
+    t1 = T(
+        """
+        a | k
+        1 | x
+        2 | y
+        5 | x
+        """
+    )
+    t2 = T(
+        """
+        b  | k
+        10 | x
+        20 | y
+        """
+    )
+    j = t1.join(t2, t1.k == t2.k)
+    res = j.filter(pw.this.a > 1).select(pw.this.a, pw.this.b)
+    assert_table_equality_wo_index(
+        res,
+        T(
+            """
+            a | b
+            2 | 20
+            5 | 10
+            """
+        ),
+    )
+    red = t1.join(t2, t1.k == t2.k).groupby(pw.this.k).reduce(
+        pw.this.k, s=pw.reducers.sum(pw.left.a), sb=pw.reducers.sum(pw.right.b)
+    )
+    assert_table_equality_wo_index(
+        red,
+        T(
+            """
+            k | s | sb
+            x | 6 | 20
+            y | 2 | 20
+            """
+        ),
+    )
