@@ -334,7 +334,11 @@ class Runtime:
                     out = node.step(time, [])
                     if out is not None:
                         n_ingested += len(out)
-                    if pm is not None and out is not None:
+                    if (
+                        pm is not None
+                        and out is not None
+                        and not getattr(pm, "operator_persisting", False)
+                    ):
                         pm.record(node.persistent_id, time, out)
             else:
                 ins = [outputs.get(id(i)) for i in node.inputs]
@@ -353,6 +357,40 @@ class Runtime:
             if isinstance(node, OutputNode):
                 node.flush(time)
         return n_ingested, n_output
+
+    def _stateful_node_ids(self):
+        out = []
+        for i, n in enumerate(self.nodes):
+            out.append((f"{i}:{type(n).__name__}", n))
+        return out
+
+    def save_operator_snapshot(self, time: int) -> None:
+        from pathway_amd.persistence.operator_snapshot import node_state_save
+
+        states = {}
+        for nid, node in self._stateful_node_ids():
+            st = node_state_save(node)
+            if st is not None:
+                states[nid] = st
+        self.persistence.op_store.save(time, states)
+
+    def load_operator_snapshot(self) -> bool:
+        from pathway_amd.persistence.operator_snapshot import node_state_load
+
+        snap = self.persistence.op_store.load()
+        if snap is None:
+            return False
+        time, states = snap
+        by_id = dict(self._stateful_node_ids())
+        for nid, st in states.items():
+            node = by_id.get(nid)
+            if node is not None:
+                node_state_load(node, st, self.device)
+        for src in self.sources:
+            seek = getattr(src.source, "seek", None)
+            if seek is not None:
+                seek(time)
+        return True
 
     def replay_persisted(self) -> None:
         """Recovery: push snapshotted input batches through the graph
@@ -400,7 +438,10 @@ class Runtime:
     def run(self, max_steps: int | None = None) -> None:
         steps = 0
         if self.persistence is not None and not getattr(self, "_replayed", False):
-            self.replay_persisted()
+            if getattr(self.persistence, "operator_persisting", False):
+                self.load_operator_snapshot()
+            else:
+                self.replay_persisted()
             self._replayed = True
         import time as _time
 
@@ -418,6 +459,8 @@ class Runtime:
                 self.monitor.maybe_report()
             self._clock = max(self._clock, t + 2)
             if self.persistence is not None:
+                if getattr(self.persistence, "operator_persisting", False):
+                    self.save_operator_snapshot(t)
                 self.persistence.commit(t)
             steps += 1
             if max_steps is not None and steps >= max_steps:

@@ -126,6 +126,11 @@ class PersistenceManager:
         root = getattr(backend, "path", None) or "/tmp/pw_persist"
         self.backend = FilesystemSnapshotBackend(os.path.join(root, f"w{worker}"))
         self.writers: dict[str, SnapshotWriter] = {}
+        mode = getattr(config, "persistence_mode", None)
+        self.operator_persisting = str(mode).lower().endswith("operator_persisting")
+        from pathway_amd.persistence.operator_snapshot import OperatorSnapshotStore
+
+        self.op_store = OperatorSnapshotStore(self.backend.root)
         meta = self.backend.load_metadata()
         self.threshold_time: int = meta["threshold_time"] if meta else -1
 

@@ -1,0 +1,298 @@
+"""Operator snapshots (reference persistence/operator_snapshot.rs:21-380 +
+engine/dataflow/persist.rs maybe_persist).
+
+Stateful nodes expose save_state()/load_state(); the Runtime snapshots
+every node's state at commit boundaries (PersistenceMode.OPERATOR_PERSISTING)
+so recovery restores operator state directly instead of replaying input
+snapshots.  Serialization is host-side (tensors → numpy, dictionary
+strings decoded — pool codes are process-local) with zlib compression.
+"""
+
+from __future__ import annotations
+
+import os
+import pickle
+import zlib
+from typing import Any
+
+import numpy as np
+import torch
+
+
+def column_to_portable(col) -> dict:
+    from pathway_amd.engine.column import (
+        ObjectColumn,
+        PointerColumn,
+        StringColumn,
+        TensorColumn,
+    )
+
+    if isinstance(col, TensorColumn):
+        return {
+            "kind": "tensor",
+            "data": col.tensor.cpu().numpy(),
+            "mask": col.mask.cpu().numpy() if col.mask is not None else None,
+            "dtype": col.dtype,
+        }
+    if isinstance(col, PointerColumn):
+        return {"kind": "pointer", "data": col.pairs.cpu().numpy(), "dtype": col.dtype}
+    if isinstance(col, StringColumn):
+        return {"kind": "string", "data": col.to_pylist(), "dtype": col.dtype}
+    if isinstance(col, ObjectColumn):
+        return {"kind": "object", "data": list(col.values), "dtype": col.dtype}
+    raise TypeError(f"unsnapshotable column {type(col)}")
+
+
+def column_from_portable(d: dict, device):
+    from pathway_amd.engine.column import (
+        ObjectColumn,
+        PointerColumn,
+        StringColumn,
+        TensorColumn,
+        obj_array,
+    )
+
+    kind = d["kind"]
+    if kind == "tensor":
+        t = torch.from_numpy(np.ascontiguousarray(d["data"])).to(device)
+        mask = (
+            torch.from_numpy(np.ascontiguousarray(d["mask"])).to(device)
+            if d["mask"] is not None
+            else None
+        )
+        return TensorColumn(t, d["dtype"], mask)
+    if kind == "pointer":
+        return PointerColumn(
+            torch.from_numpy(np.ascontiguousarray(d["data"])).to(device), d["dtype"]
+        )
+    if kind == "string":
+        return StringColumn.from_strings(d["data"], device=device)
+    if kind == "object":
+        return ObjectColumn(obj_array(d["data"]), d["dtype"])
+    raise TypeError(kind)
+
+
+def tensor_to_portable(t: torch.Tensor):
+    return t.cpu().numpy()
+
+
+def tensor_from_portable(a, device) -> torch.Tensor:
+    return torch.from_numpy(np.ascontiguousarray(a)).to(device)
+
+
+class OperatorSnapshotStore:
+    def __init__(self, root: str):
+        self.root = os.path.join(root, "operator_snapshots")
+        os.makedirs(self.root, exist_ok=True)
+
+    def save(self, snapshot_time: int, states: dict[str, Any]) -> None:
+        payload = zlib.compress(
+            pickle.dumps({"time": snapshot_time, "states": states}, protocol=4), 1
+        )
+        tmp = os.path.join(self.root, "snapshot.bin.tmp")
+        with open(tmp, "wb") as f:
+            f.write(payload)
+            f.flush()
+            os.fsync(f.fileno())
+        os.replace(tmp, os.path.join(self.root, "snapshot.bin"))
+
+    def load(self) -> tuple[int, dict[str, Any]] | None:
+        path = os.path.join(self.root, "snapshot.bin")
+        if not os.path.exists(path):
+            return None
+        with open(path, "rb") as f:
+            d = pickle.loads(zlib.decompress(f.read()))
+        return d["time"], d["states"]
+
+
+# ------------------------------------------------- node state dispatchers --
+
+def _arrangement_to_portable(arr) -> dict | None:
+    if arr is None:
+        return None
+    return {
+        "key_words": [tensor_to_portable(w) for w in arr.key_words],
+        "vhash_words": [tensor_to_portable(w) for w in arr.vhash_words],
+        "weights": tensor_to_portable(arr.weights),
+        "columns": {n: column_to_portable(c) for n, c in arr.columns.items()},
+    }
+
+
+def _arrangement_from_portable(d, device):
+    if d is None:
+        return None
+    from pathway_amd.engine.state import Arrangement
+
+    cols = {n: column_from_portable(c, device) for n, c in d["columns"].items()}
+    arr = Arrangement(device, cols)
+    arr.key_words = [tensor_from_portable(w, device) for w in d["key_words"]]
+    arr.vhash_words = [tensor_from_portable(w, device) for w in d["vhash_words"]]
+    arr.weights = tensor_from_portable(d["weights"], device)
+    arr.columns = cols
+    return arr
+
+
+def _sidestore_to_portable(store) -> dict:
+    return {"arr": _arrangement_to_portable(store.arr)}
+
+
+def _sidestore_from_portable(d, device):
+    from pathway_amd.engine.nodes_join import _SideStore
+
+    st = _SideStore(device)
+    st.arr = _arrangement_from_portable(d["arr"], device)
+    return st
+
+
+def _batch_to_portable(b) -> dict:
+    return {
+        "keys": tensor_to_portable(b.keys),
+        "diffs": tensor_to_portable(b.diffs),
+        "time": b.time,
+        "columns": {n: column_to_portable(c) for n, c in b.columns.items()},
+    }
+
+
+def _batch_from_portable(d, device):
+    from pathway_amd.engine.batch import DeltaBatch
+
+    return DeltaBatch(
+        tensor_from_portable(d["keys"], device),
+        {n: column_from_portable(c, device) for n, c in d["columns"].items()},
+        tensor_from_portable(d["diffs"], device),
+        d["time"],
+    )
+
+
+def node_state_save(node):
+    """Portable state of a stateful node, or None (reference maybe_persist,
+    persist.rs:679)."""
+    from pathway_amd.engine.nodes import GroupReduceNode
+    from pathway_amd.engine.nodes_dedup import DeduplicateNode
+    from pathway_amd.engine.nodes_index import ExternalIndexNode
+    from pathway_amd.engine.nodes_join import JoinNode, KeyedMergeNode, SemiJoinNode
+    from pathway_amd.engine.nodes_recompute import RecomputeNode
+    from pathway_amd.engine.nodes_temporal import BufferNode, ForgetNode, FreezeNode
+
+    if isinstance(node, GroupReduceNode):
+        return {
+            "kind": "reduce",
+            "seq": node.seq,
+            "add_keys": [tensor_to_portable(w) for w in node.add_keys]
+            if node.add_keys is not None
+            else None,
+            "add_accs": {n: tensor_to_portable(t) for n, t in node.add_accs.items()},
+            "add_carried": {
+                n: column_to_portable(c) for n, c in node.add_carried.items()
+            },
+            "multiset": _arrangement_to_portable(node.multiset_store),
+        }
+    if isinstance(node, JoinNode) or isinstance(node, KeyedMergeNode):
+        return {
+            "kind": "two_store",
+            "l": _sidestore_to_portable(node.lstore),
+            "r": _sidestore_to_portable(node.rstore),
+        }
+    if isinstance(node, SemiJoinNode):
+        return {
+            "kind": "semi",
+            "l": _sidestore_to_portable(node.lstore),
+            "r": _sidestore_to_portable(node.rcount),
+        }
+    if isinstance(node, DeduplicateNode):
+        return {"kind": "dedup", "state": node.state}
+    if isinstance(node, RecomputeNode):
+        return {
+            "kind": "recompute",
+            "stores": [_sidestore_to_portable(s) for s in node.stores],
+            "prev": node.prev_output,
+        }
+    if isinstance(node, BufferNode):
+        return {
+            "kind": "buffer",
+            "wm": node.watermark,
+            "held": [
+                (_batch_to_portable(b), tensor_to_portable(t)) for b, t in node.held
+            ],
+        }
+    if isinstance(node, ForgetNode):
+        return {
+            "kind": "forget",
+            "wm": node.watermark,
+            "live": [
+                (_batch_to_portable(b), tensor_to_portable(t)) for b, t in node.live
+            ],
+        }
+    if isinstance(node, FreezeNode):
+        return {"kind": "freeze", "wm": node.watermark}
+    if isinstance(node, ExternalIndexNode):
+        st = node.state
+        return {
+            "kind": "vindex",
+            "keys": tensor_to_portable(st.keys),
+            "vectors": tensor_to_portable(st.vectors) if st.vectors is not None else None,
+            "payload": st.payload,
+            "answers": node.answers,
+        }
+    if type(node).__name__ == "_BM25IndexNode":
+        return {"kind": "bm25", "state": node.state, "answers": node.answers}
+    return None
+
+
+def node_state_load(node, state, device) -> None:
+    from pathway_amd.engine.nodes_index import VectorIndexState
+
+    kind = state["kind"]
+    if kind == "reduce":
+        node.seq = state["seq"]
+        node.add_keys = (
+            [tensor_from_portable(w, device) for w in state["add_keys"]]
+            if state["add_keys"] is not None
+            else None
+        )
+        node.add_accs = {
+            n: tensor_from_portable(t, device) for n, t in state["add_accs"].items()
+        }
+        node.add_carried = {
+            n: column_from_portable(c, device) for n, c in state["add_carried"].items()
+        }
+        node.multiset_store = _arrangement_from_portable(state["multiset"], device)
+    elif kind == "two_store":
+        node.lstore = _sidestore_from_portable(state["l"], device)
+        node.rstore = _sidestore_from_portable(state["r"], device)
+    elif kind == "semi":
+        node.lstore = _sidestore_from_portable(state["l"], device)
+        node.rcount = _sidestore_from_portable(state["r"], device)
+    elif kind == "dedup":
+        node.state = state["state"]
+    elif kind == "recompute":
+        node.stores = [_sidestore_from_portable(s, device) for s in state["stores"]]
+        node.prev_output = state["prev"]
+    elif kind == "buffer":
+        node.watermark = state["wm"]
+        node.held = [
+            (_batch_from_portable(b, device), tensor_from_portable(t, device))
+            for b, t in state["held"]
+        ]
+    elif kind == "forget":
+        node.watermark = state["wm"]
+        node.live = [
+            (_batch_from_portable(b, device), tensor_from_portable(t, device))
+            for b, t in state["live"]
+        ]
+    elif kind == "freeze":
+        node.watermark = state["wm"]
+    elif kind == "vindex":
+        st = VectorIndexState(device, node.state.metric)
+        st.keys = tensor_from_portable(state["keys"], device)
+        st.vectors = (
+            tensor_from_portable(state["vectors"], device)
+            if state["vectors"] is not None
+            else None
+        )
+        st.payload = state["payload"]
+        node.state = st
+        node.answers = state["answers"]
+    elif kind == "bm25":
+        node.state = state["state"]
+        node.answers = state["answers"]

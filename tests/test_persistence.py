@@ -118,3 +118,31 @@ def test_pw_run_with_persistence(tmp_path):
     assert os.path.exists(os.path.join(pdir, "w0", "metadata.json"))
     with open(outfile) as f:
         assert len(f.read().strip().splitlines()) >= 2
+
+
+def test_operator_persisting_recovery(tmp_path):
+    pdir = str(tmp_path / "opsnap")
+    cfg = pw.persistence.Config(
+        backend=pw.persistence.Backend.filesystem(pdir),
+        persistence_mode=pw.PersistenceMode.OPERATOR_PERSISTING,
+    )
+    t, res, cap = _build_pipeline()
+    pm = PersistenceManager(cfg)
+    assert pm.operator_persisting
+    rt = Runtime([cap], persistence=pm)
+    reset_all(rt.nodes)
+    rt.run(max_steps=2)
+    pm.close()
+    rows_before = list(cap.rows)
+
+    t2, res2, cap2 = _build_pipeline()
+    pm2 = PersistenceManager(cfg)
+    rt2 = Runtime([cap2], persistence=pm2)
+    reset_all(rt2.nodes)
+    rt2.run()
+    pm2.close()
+    from pathway_amd.internals.api import squash_updates
+
+    combined = squash_updates(rows_before + list(cap2.rows))
+    got = sorted(tuple(v) for v in combined.values())
+    assert got == [("a", 3), ("c", 1)]
