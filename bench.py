@@ -32,6 +32,13 @@ def main() -> None:
     p.add_argument("--batch", type=int, default=4_000_000, help="events per step per GPU")
     p.add_argument("--vocab", type=int, default=50_000)
     p.add_argument("--device", type=str, default=None)
+    p.add_argument(
+        "--ingest",
+        choices=["codes", "bytes"],
+        default="codes",
+        help="codes: pre-parsed dictionary codes; bytes: raw newline-separated "
+        "byte stream tokenized+hashed on-device (HIP varlen kernel)",
+    )
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -49,6 +56,8 @@ def main() -> None:
         torch.cuda.set_device(device)
 
     os.environ["PW_DEVICE"] = str(device)
+
+    import numpy as np
 
     import pathway_amd as pw
     import pathway_amd.parallel as par
@@ -105,12 +114,94 @@ def main() -> None:
         def reset(self):
             self.counter = 0
 
-    source = WordStream(args.batch, args.vocab)
-    in_node = InputNode(source, device)
-    words = Table(in_node, {"word": dt.STR}, Universe())
-    counts = words.groupby(pw.this.word).reduce(
-        pw.this.word, count=pw.reducers.count()
-    )
+    class ByteStream:
+        """Raw ingest: per pull, a newline-separated byte buffer is
+        tokenized and xxh-128-keyed ON DEVICE (pw_varlen_hash) — the parse
+        path a GPU-native text connector runs (message parsing included in
+        the timed region)."""
+
+        def __init__(self, batch: int, vocab_n: int):
+            self.batch = batch
+            self.vocab_n = vocab_n
+            self.counter = 0
+            # device matrix of the vocabulary's word bytes (fixed width 10)
+            wb = np.frombuffer(
+                "".join(f"word{i:06d}" for i in range(vocab_n)).encode(),
+                dtype=np.uint8,
+            ).reshape(vocab_n, 10)
+            self.word_bytes = torch.from_numpy(wb.copy()).to(device)
+
+        def next_time(self):
+            return None
+
+        def pull(self, t, dev):
+            from pathway_amd import ops
+            from pathway_amd.engine.column import PointerColumn
+            from pathway_amd.internals.api import TAG_STR
+
+            n = self.batch
+            codes = torch.randint(
+                0, self.vocab_n, (n,), dtype=torch.int64, generator=gen
+            ).to(dev, non_blocking=True)
+            # build the wire buffer: "<word>\n" per message (device gather)
+            msg = torch.empty((n, 11), dtype=torch.uint8, device=dev)
+            msg[:, :10] = self.word_bytes.index_select(0, codes)
+            msg[:, 10] = 10  # newline
+            buf = msg.reshape(-1).contiguous()
+            # PARSE on device: newline scan -> token [start, end) spans
+            nl = (buf == 10).nonzero(as_tuple=True)[0]
+            starts = torch.cat(
+                [torch.zeros(1, dtype=torch.int64, device=dev), nl[:-1] + 1]
+            )
+            ends = nl
+            # per-token canonical string hash == the word's group key
+            if dev.type == "cuda":
+                glo, ghi = ops.varlen_hash_se_gpu(buf, starts, ends, TAG_STR)
+            else:
+                from pathway_amd.internals.api import hash128, serialize_value
+
+                bb = buf.cpu().numpy().tobytes()
+                st = starts.cpu().tolist()
+                en = ends.cpu().tolist()
+                glo_l, ghi_l = [], []
+                for a, b in zip(st, en):
+                    l_, h_ = hash128(serialize_value(bb[a:b].decode()))
+                    glo_l.append(l_ - (1 << 64) if l_ >= 1 << 63 else l_)
+                    ghi_l.append(h_ - (1 << 64) if h_ >= 1 << 63 else h_)
+                glo = torch.tensor(glo_l, dtype=torch.int64)
+                ghi = torch.tensor(ghi_l, dtype=torch.int64)
+            seq = torch.arange(
+                self.counter, self.counter + n, dtype=torch.int64, device=dev
+            )
+            self.counter += n
+            klo, khi = hashing.value_hash_words(seq, TAG_INT)
+            keys = torch.stack([klo, khi], dim=1)
+            cols = {
+                "word": StringColumn(codes, GLOBAL_STRING_POOL, dt.STR),
+                "wkey": PointerColumn(torch.stack([glo, ghi], dim=1)),
+            }
+            diffs = torch.ones(n, dtype=torch.int64, device=dev)
+            return DeltaBatch(keys, cols, diffs, t)
+
+        def reset(self):
+            self.counter = 0
+
+    if args.ingest == "bytes":
+        source = ByteStream(args.batch, args.vocab)
+        in_node = InputNode(source, device)
+        words = Table(
+            in_node, {"word": dt.STR, "wkey": dt.POINTER}, Universe()
+        )
+        counts = words.groupby(pw.this.word, id=pw.this.wkey).reduce(
+            pw.this.word, count=pw.reducers.count()
+        )
+    else:
+        source = WordStream(args.batch, args.vocab)
+        in_node = InputNode(source, device)
+        words = Table(in_node, {"word": dt.STR}, Universe())
+        counts = words.groupby(pw.this.word).reduce(
+            pw.this.word, count=pw.reducers.count()
+        )
 
     emitted = [0]
 
@@ -187,6 +278,7 @@ def main() -> None:
                         "seq_len": 1,
                         "parallelism": f"dp{n_gpus} (shard exchange: RCCL all-to-all)",
                         "vocab": args.vocab,
+                        "ingest": args.ingest,
                         "p95_step_latency_ms": p95,
                         "emitted_delta_rows": emitted[0],
                     },

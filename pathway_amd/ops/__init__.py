@@ -255,3 +255,29 @@ def pool_hash_gpu(
     if rc != 0:
         raise RuntimeError(f"pw_pool_hash failed: hip error {rc}")
     return lo, hi
+
+
+def varlen_hash_se_gpu(
+    bytes_t: torch.Tensor,
+    starts: torch.Tensor,
+    ends: torch.Tensor,
+    tag: int,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Device hash of varlen rows with explicit [start, end) spans."""
+    lib = require_lib()
+    n = starts.shape[0]
+    lo = torch.empty(n, dtype=torch.int64, device=bytes_t.device)
+    hi = torch.empty(n, dtype=torch.int64, device=bytes_t.device)
+    rc = lib.pw_varlen_hash_se(
+        ctypes.c_void_p(bytes_t.contiguous().data_ptr()),
+        ctypes.c_void_p(starts.contiguous().data_ptr()),
+        ctypes.c_void_p(ends.contiguous().data_ptr()),
+        ctypes.c_uint64(tag),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(lo.data_ptr()),
+        ctypes.c_void_p(hi.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_varlen_hash_se failed: hip error {rc}")
+    return lo, hi

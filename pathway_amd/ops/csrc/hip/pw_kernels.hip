@@ -419,3 +419,36 @@ extern "C" int pw_pool_hash(const void* codes, const void* pool_lo,
                      none_lo, none_hi, n, (uint64_t*)lo, (uint64_t*)hi);
   return (int)hipGetLastError();
 }
+
+// varlen hash with explicit [start, end) per row (token parse output —
+// tokens exclude separators so rows are not contiguous)
+__global__ void k_varlen_hash_se(const uint8_t* bytes, const int64_t* starts,
+                                 const int64_t* ends, uint64_t tag, int64_t n,
+                                 uint64_t* lo, uint64_t* hi) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t a = starts[i], b = ends[i];
+    uint8_t buf[64];
+    int64_t len = b - a;
+    if (len <= 56) {
+      uint64_t t = tag;
+      __builtin_memcpy(buf, &t, 8);
+      for (int64_t k = 0; k < len; ++k) buf[8 + k] = bytes[a + k];
+      lo[i] = pw_xxh64_bytes(buf, 8 + len, PW_SEED_LO);
+      hi[i] = pw_xxh64_bytes(buf, 8 + len, PW_SEED_HI);
+    } else {
+      lo[i] = pw_xxh64_long(bytes + a, len, tag, PW_SEED_LO);
+      hi[i] = pw_xxh64_long(bytes + a, len, tag, PW_SEED_HI);
+    }
+  }
+}
+
+extern "C" int pw_varlen_hash_se(const void* bytes, const void* starts,
+                                 const void* ends, uint64_t tag, int64_t n,
+                                 void* lo, void* hi, void* stream) {
+  hipLaunchKernelGGL(k_varlen_hash_se, dim3(pw_grid(n)), dim3(PW_BLOCK), 0,
+                     (hipStream_t)stream, (const uint8_t*)bytes,
+                     (const int64_t*)starts, (const int64_t*)ends, tag, n,
+                     (uint64_t*)lo, (uint64_t*)hi);
+  return (int)hipGetLastError();
+}

@@ -207,3 +207,21 @@ def test_document_store_gpu():
         assert len(rv) == 1 and "streaming" in rv[0]["text"]
     finally:
         pathway_config.device = None
+
+
+@gpu
+@requires_cuda
+def test_varlen_hash_se_kernel():
+    from pathway_amd import ops
+    from pathway_amd.internals.api import MASK64, TAG_STR, hash128, serialize_value
+
+    words = ["alpha", "beta", "x" * 80]
+    wire = ("\n".join(words) + "\n").encode()
+    buf = torch.frombuffer(bytearray(wire), dtype=torch.uint8).cuda()
+    nl = (buf == 10).nonzero(as_tuple=True)[0]
+    starts = torch.cat([torch.zeros(1, dtype=torch.int64).cuda(), nl[:-1] + 1])
+    lo, hi = ops.varlen_hash_se_gpu(buf, starts, nl, TAG_STR)
+    for i, w in enumerate(words):
+        elo, ehi = hash128(serialize_value(w))
+        assert int(lo[i]) & MASK64 == elo
+        assert int(hi[i]) & MASK64 == ehi
