@@ -30,7 +30,9 @@ import numpy as np
 import torch
 
 from pathway_amd.internals import dtype as dt
-from pathway_amd.internals.api import MASK64
+from pathway_amd.internals.api import MASK64, SHARD_MASK
+
+MASK_SHARD = SHARD_MASK
 from pathway_amd.engine import hashing
 from pathway_amd.engine.batch import DeltaBatch, segment_starts
 from pathway_amd.engine.column import (
@@ -299,6 +301,9 @@ class GroupReduceNode(Node):
         self.reducer_calls = reducer_calls
         self.sort_by = sort_by
         self.key_expr = None  # groupby(id=...): group key IS this pointer expr
+        #: instance column name — shard bits of the group key come from the
+        #: instance hash alone (ShardPolicy::LastKeyColumn, value.rs:96-118)
+        self.instance_name: str | None = None
         self.seq = 0  # arrival sequence for earliest/latest
 
         # combined reduce state: sorted keys + additive acc tensors +
@@ -434,6 +439,10 @@ class GroupReduceNode(Node):
                 else:
                     glo = torch.zeros(len(b), dtype=torch.int64, device=device)
                     ghi = glo.clone()
+                if self.instance_name and self.instance_name in gcols:
+                    ilo, _ihi = gcols[self.instance_name].value_hash()
+                    ilo = ilo.to(device)
+                    glo = (glo & ~MASK_SHARD) | (ilo & MASK_SHARD)
                 gkeys = torch.stack([glo, ghi], dim=1)
 
             # 2. evaluate reducer args

@@ -130,6 +130,8 @@ class GroupedTable:
             get_device(),
             sort_by=self._sort_by,
         )
+        if self._instance is not None:
+            node.instance_name = "_pw_instance"
         if key_expr is not None:
             node.key_expr = key_expr
 

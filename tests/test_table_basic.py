@@ -463,3 +463,34 @@ def test_join_result_filter_and_reduce():
             """
         ),
     )
+
+
+def test_groupby_instance_colocation():
+    t = T(
+        """
+        g | inst | v
+        a | 1    | 1
+        a | 1    | 2
+        b | 1    | 3
+        a | 2    | 4
+        """
+    )
+    res = t.groupby(pw.this.g, instance=pw.this.inst).reduce(
+        pw.this.g, s=pw.reducers.sum(pw.this.v)
+    )
+    keys, cols = pw.debug.table_to_dicts(res)
+    got = sorted((cols["g"][k], cols["s"][k]) for k in keys)
+    assert got == [("a", 3), ("a", 4), ("b", 3)]
+    # rows with the same instance share the shard bits of their group key
+    from pathway_amd.internals.api import SHARD_MASK
+
+    shards = {}
+    gk, ck = pw.debug.table_to_dicts(
+        t.groupby(pw.this.g, instance=pw.this.inst).reduce(
+            pw.this.g, i=pw.reducers.any(pw.this.inst)
+        )
+    )
+    for k in gk:
+        shards.setdefault(ck["i"][k], set()).add(k.lo & SHARD_MASK)
+    for inst, sh in shards.items():
+        assert len(sh) == 1, f"instance {inst} split across shards {sh}"
