@@ -35,7 +35,7 @@ def main() -> None:
     p.add_argument(
         "--ingest",
         choices=["codes", "bytes"],
-        default="codes",
+        default="bytes",
         help="codes: pre-parsed dictionary codes; bytes: raw newline-separated "
         "byte stream tokenized+hashed on-device (HIP varlen kernel)",
     )
