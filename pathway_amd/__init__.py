@@ -130,6 +130,15 @@ from pathway_amd.stdlib import (  # noqa: E402
 from pathway_amd.stdlib.utils.async_transformer import AsyncTransformer  # noqa: E402
 from pathway_amd.stdlib.utils.pandas_transformer import pandas_transformer  # noqa: E402
 
+from pathway_amd.internals.row_transformer import (  # noqa: E402
+    ClassArg,
+    input_attribute,
+    input_method,
+    method,
+    output_attribute,
+    transformer,
+)
+
 # reference-surface aliases
 from pathway_amd.stdlib.temporal._asof_join import AsofJoinResult  # noqa: E402
 from pathway_amd.stdlib.temporal._interval_join import IntervalJoinResult  # noqa: E402
@@ -253,4 +262,10 @@ __all__ = [
     "OuterJoinResult",
     "asynchronous",
     "window",
+    "transformer",
+    "ClassArg",
+    "input_attribute",
+    "input_method",
+    "output_attribute",
+    "method",
 ]

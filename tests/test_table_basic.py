@@ -494,3 +494,36 @@ def test_groupby_instance_colocation():
         shards.setdefault(ck["i"][k], set()).add(k.lo & SHARD_MASK)
     for inst, sh in shards.items():
         assert len(sh) == 1, f"instance {inst} split across shards {sh}"
+
+
+def test_row_transformer():
+    @pw.transformer
+    class fib_tr:
+        class series(pw.ClassArg):
+            n = pw.input_attribute()
+
+            @pw.output_attribute
+            def fib(self) -> int:
+                if self.n <= 1:
+                    return self.n
+                p1 = self.pointer_from(self.n - 1)
+                p2 = self.pointer_from(self.n - 2)
+                return (
+                    self.transformer.series[p1].fib
+                    + self.transformer.series[p2].fib
+                )
+
+    t = T(
+        """
+        n
+        0
+        1
+        2
+        3
+        4
+        5
+        """
+    ).with_id_from(pw.this.n)
+    res = fib_tr(series=t).series
+    keys, cols = pw.debug.table_to_dicts(res)
+    assert sorted(cols["fib"].values()) == [0, 1, 1, 2, 3, 5]
