@@ -225,3 +225,56 @@ def test_varlen_hash_se_kernel():
         elo, ehi = hash128(serialize_value(w))
         assert int(lo[i]) & MASK64 == elo
         assert int(hi[i]) & MASK64 == ehi
+
+
+@gpu
+@requires_cuda
+def test_gpu_cpu_pipeline_equivalence():
+    """Same pipeline on cpu and cuda must produce identical results
+    (determinism across devices — SURVEY §5.2 analog)."""
+    from pathway_amd.internals.config import pathway_config
+
+    def build_and_run():
+        import pathway_amd as pw
+        from pathway_amd.debug import table_from_markdown as T, table_to_dicts
+
+        t = T(
+            """
+            g | v | t
+            a | 1 | 1
+            a | 4 | 3
+            b | 2 | 2
+            b | 3 | 9
+            a | 7 | 12
+            """
+        )
+        win = t.windowby(pw.this.t, window=pw.temporal.tumbling(duration=5)).reduce(
+            start=pw.this._pw_window_start,
+            s=pw.reducers.sum(pw.this.v),
+            m=pw.reducers.max(pw.this.v),
+        )
+        other = T(
+            """
+            g | w
+            a | 100
+            b | 200
+            """
+        )
+        j = t.join(other, t.g == other.g).select(pw.this.g, pw.this.v, pw.this.w)
+        red = j.groupby(pw.this.g).reduce(
+            pw.this.g, tot=pw.reducers.sum(pw.this.v + pw.this.w)
+        )
+        k1, c1 = table_to_dicts(win)
+        k2, c2 = table_to_dicts(red)
+        rows1 = sorted((c1["start"][k], c1["s"][k], c1["m"][k]) for k in k1)
+        rows2 = sorted((c2["g"][k], c2["tot"][k]) for k in k2)
+        return rows1, rows2
+
+    pathway_config.device = "cpu"
+    cpu_result = build_and_run()
+    pathway_config.device = "cuda:0"
+    try:
+        gpu_result = build_and_run()
+    finally:
+        pathway_config.device = None
+    assert cpu_result == gpu_result
