@@ -144,6 +144,7 @@ def sql(query: str, **tables: Table) -> Table:
         return result
     m = re.match(
         r"SELECT\s+(?P<sel>.*?)\s+FROM\s+(?P<from>[A-Za-z_][A-Za-z0-9_]*)"
+        r"(?:\s+(?:INNER\s+)?JOIN\s+(?P<jt>[A-Za-z_][A-Za-z0-9_]*)\s+ON\s+(?P<on>.*?))?"
         r"(?:\s+WHERE\s+(?P<where>.*?))?"
         r"(?:\s+GROUP\s+BY\s+(?P<gb>.*?))?"
         r"(?:\s+HAVING\s+(?P<having>.*?))?$",
@@ -156,6 +157,24 @@ def sql(query: str, **tables: Table) -> Table:
     if tname not in tables:
         raise _SqlError(f"unknown table {tname!r}")
     t = tables[tname]
+    if m.group("jt"):
+        jt = m.group("jt")
+        if jt not in tables:
+            raise _SqlError(f"unknown table {jt!r}")
+        right = tables[jt]
+        on = m.group("on")
+        mo = re.match(
+            r"\s*([A-Za-z_][\w.]*)\s*=\s*([A-Za-z_][\w.]*)\s*$", on
+        )
+        if not mo:
+            raise _SqlError(f"unsupported JOIN condition {on!r}")
+        lcol = mo.group(1).split(".")[-1]
+        rcol = mo.group(2).split(".")[-1]
+        if lcol not in t._dtypes:
+            lcol, rcol = rcol, lcol
+        j = t.join(right, t[lcol] == right[rcol])
+        # materialize all columns of both sides (suffix disambiguation)
+        t = j._all_columns_table()
     if m.group("where"):
         t = t.filter(_parse_scalar(m.group("where"), t))
     sel = m.group("sel").strip()

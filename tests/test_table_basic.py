@@ -527,3 +527,31 @@ def test_row_transformer():
     res = fib_tr(series=t).series
     keys, cols = pw.debug.table_to_dicts(res)
     assert sorted(cols["fib"].values()) == [0, 1, 1, 2, 3, 5]
+
+
+def test_sql_join():
+    t1 = T(
+        """
+        a | k
+        1 | x
+        2 | y
+        """
+    )
+    t2 = T(
+        """
+        b  | k
+        10 | x
+        20 | y
+        """
+    )
+    res = pw.sql("SELECT a, b, a + b AS s FROM l JOIN r ON l.k = r.k", l=t1, r=t2)
+    assert_table_equality_wo_index(
+        res,
+        T(
+            """
+            a | b  | s
+            1 | 10 | 11
+            2 | 20 | 22
+            """
+        ),
+    )
