@@ -676,7 +676,10 @@ class GroupReduceNode(Node):
             n: concat_columns([self.add_carried[n], gcols_first[n]])
             for n in self.add_carried
         }
-        perm2 = lex_sort_words(all_words)
+        # state and delta are both sorted: O(m+n) merge, no re-sort
+        from pathway_amd.engine.state import merge_sorted_select
+
+        perm2 = merge_sorted_select(self.add_keys, ukeys_w)
         all_words = [w.index_select(0, perm2) for w in all_words]
         starts2 = rows_ne(all_words)
         seg2 = torch.cumsum(starts2.to(torch.int64), 0) - 1

@@ -143,6 +143,36 @@ def searchsorted_words(
     return lo
 
 
+def merge_sorted_select(
+    state_words: Sequence[torch.Tensor], delta_words: Sequence[torch.Tensor]
+) -> torch.Tensor:
+    """Source-selection permutation merging two SORTED row sets in O(m+n).
+
+    Returns src_sel of length m+n indexing into concat([state, delta]):
+    taking concat-columns with src_sel yields the merged sorted order
+    (state rows precede equal delta rows).  Replaces the concat+re-sort of
+    the whole arrangement — the LSM-merge analog, two binary-search kernels
+    plus one gather instead of an O((m+n) log) sort.
+    """
+    m = state_words[0].shape[0]
+    n = delta_words[0].shape[0]
+    device = state_words[0].device
+    if m == 0:
+        return torch.arange(n, dtype=torch.int64, device=device)
+    if n == 0:
+        return torch.arange(m, dtype=torch.int64, device=device)
+    # delta row i lands after all state rows ≤ it
+    pos_d = searchsorted_words(state_words, delta_words, side="right")
+    final_d = pos_d + torch.arange(n, dtype=torch.int64, device=device)
+    # state row j lands after delta rows strictly before it
+    pos_s = searchsorted_words(delta_words, state_words, side="left")
+    final_s = pos_s + torch.arange(m, dtype=torch.int64, device=device)
+    src_sel = torch.empty(m + n, dtype=torch.int64, device=device)
+    src_sel[final_s] = torch.arange(m, dtype=torch.int64, device=device)
+    src_sel[final_d] = m + torch.arange(n, dtype=torch.int64, device=device)
+    return src_sel
+
+
 class Arrangement:
     """Consolidated sorted multiset of weighted rows, keyed by a 128-bit key.
 
@@ -211,18 +241,25 @@ class Arrangement:
         if keys.shape[0] == 0:
             return
         dwords = [keys[:, 0].contiguous(), keys[:, 1].contiguous(), vhash[0], vhash[1]]
-        all_words = [torch.cat([s, d]) for s, d in zip(self.words, dwords)]
-        all_w = torch.cat([self.weights, weights])
+        # 1. sort the (small) delta
+        sort_words = dwords[:2] if key_determined_vhash else dwords
+        dperm = lex_sort_words(sort_words)
+        dwords = [w.index_select(0, dperm) for w in dwords]
+        dweights = weights.index_select(0, dperm)
+        dcols = {n: columns[n].take(dperm) for n in self.columns}
+        # 2. O(m+n) sorted merge with the state (no state re-sort)
+        mwords = self.words[:2] if key_determined_vhash else self.words
+        qwords = dwords[:2] if key_determined_vhash else dwords
+        src_sel = merge_sorted_select(mwords, qwords)
+        all_words = [
+            torch.cat([s, d]).index_select(0, src_sel)
+            for s, d in zip(self.words, dwords)
+        ]
+        all_w = torch.cat([self.weights, dweights]).index_select(0, src_sel)
         all_cols = {
-            n: concat_columns([self.columns[n], columns[n]]) for n in self.columns
+            n: concat_columns([self.columns[n], dcols[n]]).take(src_sel)
+            for n in self.columns
         }
-        if key_determined_vhash:
-            perm = lex_sort_words(all_words[:2])
-        else:
-            perm = lex_sort_words(all_words)
-        all_words = [w.index_select(0, perm) for w in all_words]
-        all_w = all_w.index_select(0, perm)
-        all_cols = {n: c.take(perm) for n, c in all_cols.items()}
         out_words, out_w, out_cols = consolidate_sorted(all_words, all_w, all_cols)
         self.key_words = out_words[:2]
         self.vhash_words = out_words[2:]
