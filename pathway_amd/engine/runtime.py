@@ -171,6 +171,26 @@ STREAM_READY = object()   # has data now, any commit time works
 STREAM_WAITING = object()  # alive but idle — the runtime should wait
 
 
+def _check_errors(batch) -> None:
+    """Raise on Value::Error reaching a sink when terminate_on_error is set
+    (reference parse_graph.py:182-201 / table.py:2753 semantics)."""
+    from pathway_amd.internals.config import pathway_config
+
+    if not pathway_config.terminate_on_error:
+        return
+    from pathway_amd.engine.column import ObjectColumn
+    from pathway_amd.internals.api import ERROR
+
+    for name, col in batch.columns.items():
+        if isinstance(col, ObjectColumn):
+            for v in col.values:
+                if v is ERROR:
+                    raise RuntimeError(
+                        f"Error value in column {name!r} reached an output; "
+                        "use pw.fill_error(...) or set terminate_on_error=False"
+                    )
+
+
 class CaptureNode(Node):
     """Collects the full update stream of a table (debug / tests / sinks)."""
 
@@ -189,6 +209,7 @@ class CaptureNode(Node):
         b = consolidate_batch(b)
         if b is None:
             return None
+        _check_errors(b)
         for key, values, t, diff in b.rows():
             step = 1 if diff > 0 else -1
             for _ in range(abs(diff)):

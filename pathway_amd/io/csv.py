@@ -28,13 +28,18 @@ def read(
                 files.append(os.path.join(root, f))
     else:
         files = [path]
+    delimiter = ","
+    quotechar = '"'
+    if csv_settings is not None:
+        delimiter = getattr(csv_settings, "delimiter", ",") or ","
+        quotechar = getattr(csv_settings, "quote", '"') or '"'
     if schema is None:
-        schema = schema_from_csv(files[0])
+        schema = schema_from_csv(files[0], delimiter=delimiter, quote=quotechar)
     names = schema.column_names()
     rows = []
     for f in files:
         with open(f, newline="") as fh:
-            reader = _csv.DictReader(fh)
+            reader = _csv.DictReader(fh, delimiter=delimiter, quotechar=quotechar)
             for rec in reader:
                 row = []
                 for n in names:

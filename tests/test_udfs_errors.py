@@ -230,3 +230,22 @@ def test_json_column():
     k = keys[0]
     assert cols["b"][k] == 7
     assert cols["l1"][k] == 2
+
+
+def test_terminate_on_error_raises_at_sink():
+    from pathway_amd.internals.config import pathway_config
+
+    @pw.udf
+    def boom(x: int) -> int:
+        raise ValueError("nope")
+
+    t = T(
+        """
+        a
+        1
+        """
+    )
+    res = t.select(b=pw.declare_type(pw.Type.ANY, boom(pw.this.a)))
+    assert pathway_config.terminate_on_error
+    with pytest.raises(RuntimeError, match="Error value"):
+        pw.debug.table_to_dicts(res)
