@@ -34,7 +34,7 @@ from pathway_amd.engine.nodes import (
     null_column,
     unique_sorted_keys,
 )
-from pathway_amd.engine.state import Arrangement
+from pathway_amd.engine.state import Arrangement, SpineArrangement
 
 
 def _join_keys(exprs: list[Any], b: DeltaBatch, device) -> torch.Tensor:
@@ -77,15 +77,20 @@ def _exchange_side(comm, b: DeltaBatch | None, jk: torch.Tensor | None, time: in
 
 
 class _SideStore:
-    """Arrangement of one join side keyed by join key, payload = row."""
+    """Spine of one join side keyed by join key, payload = row
+    (reference join.rs arrangements; GPU spine: state.SpineArrangement)."""
 
     def __init__(self, device):
         self.device = device
-        self.arr: Arrangement | None = None
+        self.spine: SpineArrangement | None = None
+
+    @property
+    def arr(self):  # back-compat for snapshots/tools
+        return self.spine
 
     def ensure(self, cols: dict[str, Column]):
-        if self.arr is None:
-            self.arr = Arrangement(self.device, cols)
+        if self.spine is None:
+            self.spine = SpineArrangement(self.device, cols)
 
     def merge(self, jkeys: torch.Tensor, b: DeltaBatch):
         cols = dict(b.columns)
@@ -96,38 +101,25 @@ class _SideStore:
             for lo, hi in (c.value_hash() for c in cols.values())
         ]
         v0, v1 = hashing.combine_value_hashes(parts)
-        self.arr.merge(jkeys, (v0, v1), b.diffs, cols)
+        self.spine.merge(jkeys, (v0, v1), b.diffs, cols)
+
+    def __len__(self) -> int:
+        return len(self.spine) if self.spine is not None else 0
 
     def probe(self, jkeys: torch.Tensor):
-        """rows, qidx, weights matched for each probe row."""
-        if self.arr is None or len(self.arr) == 0:
+        """(gathered columns incl __id__, qidx, weights) per matched row."""
+        if self.spine is None or len(self.spine) == 0:
             z = torch.zeros((0,), dtype=torch.int64, device=self.device)
-            return z, z.clone(), z.clone()
-        lo, hi = self.arr.key_range(jkeys)
-        rows, qidx = self.arr.gather_ranges(lo, hi)
-        w = self.arr.weights.index_select(0, rows)
-        return rows, qidx, w
+            return None, z, z.clone()
+        return self.spine.probe_rows(jkeys)
 
     def count_for(self, keys2: torch.Tensor) -> torch.Tensor:
-        """Total weight per query key."""
         nq = keys2.shape[0]
-        if self.arr is None or len(self.arr) == 0:
+        if self.spine is None or len(self.spine) == 0:
             return torch.zeros(nq, dtype=torch.int64, device=self.device)
-        lo, hi = self.arr.key_range(keys2)
-        rows, qidx = self.arr.gather_ranges(lo, hi)
-        w = self.arr.weights.index_select(0, rows)
-        out = torch.zeros(nq, dtype=torch.int64, device=self.device)
-        out.index_add_(0, qidx, w)
-        return out
+        return self.spine.count_for(keys2)
 
-    def rows_for(self, keys2: torch.Tensor):
-        if self.arr is None or len(self.arr) == 0:
-            z = torch.zeros((0,), dtype=torch.int64, device=self.device)
-            return z, z.clone(), z.clone()
-        lo, hi = self.arr.key_range(keys2)
-        rows, qidx = self.arr.gather_ranges(lo, hi)
-        w = self.arr.weights.index_select(0, rows)
-        return rows, qidx, w
+    rows_for = probe
 
 
 class JoinNode(Node):
@@ -253,13 +245,12 @@ class JoinNode(Node):
     # -- pair emission --
 
     def _emit_pairs(self, b: DeltaBatch, jk: torch.Tensor, store: _SideStore, probe_is_right: bool):
-        rows, qidx, w = store.probe(jk)
-        if rows.shape[0] == 0:
+        stored_cols, qidx, w = store.probe(jk)
+        if qidx.shape[0] == 0:
             return []
         probe = b.take(qidx)
         diffs = probe.diffs * w
-        arr = store.arr
-        stored_cols = {n: c.take(rows) for n, c in arr.columns.items()}
+        stored_cols = dict(stored_cols)
         stored_ids = stored_cols.pop("__id__")
         if probe_is_right:
             lkeys, rkeys = stored_ids.pairs, probe.keys
@@ -273,12 +264,12 @@ class JoinNode(Node):
         # match dL and dR directly: sort dR by key, range-probe with dL
         tmp = _SideStore(self.device)
         tmp.merge(jr, br)
-        rows, qidx, w = tmp.probe(jl)
-        if rows.shape[0] == 0:
+        stored_cols, qidx, w = tmp.probe(jl)
+        if qidx.shape[0] == 0:
             return []
         probe = bl.take(qidx)
         diffs = probe.diffs * w
-        stored_cols = {n: c.take(rows) for n, c in tmp.arr.columns.items()}
+        stored_cols = dict(stored_cols)
         stored_ids = stored_cols.pop("__id__")
         return [
             self._make_pair_batch(
@@ -319,18 +310,19 @@ class JoinNode(Node):
         device = self.device
         own = self.lstore if side == "left" else self.rstore
         other = self.rstore if side == "left" else self.lstore
-        if own.arr is None or len(own.arr) == 0:
+        if len(own) == 0:
             return None
         other_cnt = other.count_for(affected)
         un_keys_mask = other_cnt == 0
         if not bool(un_keys_mask.any()):
             return None
         ukeys = affected.index_select(0, un_keys_mask.nonzero(as_tuple=True)[0])
-        rows, qidx, w = own.rows_for(ukeys)
-        if rows.shape[0] == 0:
+        cols, qidx, w = own.rows_for(ukeys)
+        if qidx.shape[0] == 0:
             return None
-        cols = {n: c.take(rows) for n, c in own.arr.columns.items()}
+        cols = dict(cols)
         ids = cols.pop("__id__")
+        n = qidx.shape[0]
         if (self.key_mode == "left" and side == "left") or (
             self.key_mode == "right" and side == "right"
         ):
@@ -342,7 +334,7 @@ class JoinNode(Node):
                 [(ids.pairs[:, 0].contiguous(), ids.pairs[:, 1].contiguous())],
             )
             okeys = torch.stack([lo, hi], dim=1)
-        n = rows.shape[0]
+        n = qidx.shape[0]
         out_cols: dict[str, Column] = {}
         own_map = self.left_out if side == "left" else self.right_out
         other_map = self.right_out if side == "left" else self.left_out
@@ -426,17 +418,17 @@ class SemiJoinNode(Node):
 
     def _visible(self, aff: torch.Tensor) -> DeltaBatch | None:
         device = self.device
-        if self.lstore.arr is None or len(self.lstore.arr) == 0:
+        if len(self.lstore) == 0:
             return None
         rc = self.rcount.count_for(aff)
         want = (rc > 0) if self.mode == "intersect" else (rc == 0)
         if not bool(want.any()):
             return None
         keys = aff.index_select(0, want.nonzero(as_tuple=True)[0])
-        rows, qidx, w = self.lstore.rows_for(keys)
-        if rows.shape[0] == 0:
+        cols, qidx, w = self.lstore.rows_for(keys)
+        if qidx.shape[0] == 0:
             return None
-        cols = {n: c.take(rows) for n, c in self.lstore.arr.columns.items()}
+        cols = dict(cols)
         ids = cols.pop("__id__")
         return DeltaBatch(ids.pairs, cols, w, 0)
 
@@ -485,8 +477,8 @@ class KeyedMergeNode(Node):
 
     def _visible(self, aff: torch.Tensor) -> DeltaBatch | None:
         device = self.device
-        lhas = self.lstore.arr is not None and len(self.lstore.arr) > 0
-        rhas = self.rstore.arr is not None and len(self.rstore.arr) > 0
+        lhas = len(self.lstore) > 0
+        rhas = len(self.rstore) > 0
         if not lhas and not rhas:
             return None
         lcnt = self.lstore.count_for(aff) if lhas else torch.zeros(aff.shape[0], dtype=torch.int64, device=device)
@@ -512,12 +504,13 @@ class KeyedMergeNode(Node):
                 return None
             if rhas:
                 # override cells where right row exists for same key
-                rows, qidx, w = self.rstore.rows_for(lb.keys)
-                if rows.shape[0]:
+                rcols, qidx, w = self.rstore.rows_for(lb.keys)
+                if qidx.shape[0]:
                     cols = dict(lb.columns)
                     for cname in self.override_cols:
-                        rcol = self.rstore.arr.columns[cname].take(rows)
-                        cols[cname] = _scatter_override(cols[cname], qidx, rcol)
+                        cols[cname] = _scatter_override(
+                            cols[cname], qidx, rcols[cname]
+                        )
                     lb = lb.with_columns(cols)
             batches.append(lb)
         batches = [b for b in batches if b is not None and len(b)]
@@ -528,10 +521,10 @@ class KeyedMergeNode(Node):
         return DeltaBatch.concat(batches)
 
     def _gather_side(self, store: _SideStore, keys: torch.Tensor) -> DeltaBatch | None:
-        rows, qidx, w = store.rows_for(keys)
-        if rows.shape[0] == 0:
+        cols, qidx, w = store.rows_for(keys)
+        if qidx.shape[0] == 0:
             return None
-        cols = {n: c.take(rows) for n, c in store.arr.columns.items()}
+        cols = dict(cols)
         ids = cols.pop("__id__")
         return DeltaBatch(ids.pairs, cols, w, 0)
 

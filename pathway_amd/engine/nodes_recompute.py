@@ -53,24 +53,36 @@ class RecomputeNode(Node):
                 changed = True
         if not changed:
             return None
-        # materialize inputs on host
+        # materialize inputs on host (summing weights across spine levels)
         in_rows = []
         in_keys = []
         for st in self.stores:
             rows = []
             keys = []
-            if st.arr is not None and len(st.arr):
-                cols = {n: c.to_pylist() for n, c in st.arr.columns.items()}
-                ids = cols.pop("__id__")
-                w = st.arr.weights.cpu().tolist()
-                names = list(cols.keys())
-                m = len(ids)
-                for i in range(m):
-                    if w[i] <= 0:
+            if len(st):
+                from collections import defaultdict
+
+                acc: dict = defaultdict(int)
+                payload: dict = {}
+                for lvl in st.spine.levels:
+                    if len(lvl) == 0:
                         continue
-                    for _ in range(w[i]):
-                        rows.append({n: cols[n][i] for n in names})
-                        keys.append(ids[i])
+                    cols = {n: c.to_pylist() for n, c in lvl.columns.items()}
+                    ids = cols.pop("__id__")
+                    w = lvl.weights.cpu().tolist()
+                    names = list(cols.keys())
+                    for i in range(len(ids)):
+                        row = tuple((n, cols[n][i]) for n in names)
+                        k = (repr(ids[i]), row)
+                        acc[k] += w[i]
+                        payload[k] = (ids[i], {n: cols[n][i] for n in names})
+                for k, wsum in acc.items():
+                    if wsum <= 0:
+                        continue
+                    key_obj, row = payload[k]
+                    for _ in range(wsum):
+                        rows.append(row)
+                        keys.append(key_obj)
             in_rows.append(rows)
             in_keys.append(keys)
         new_output = self.fn(in_rows, in_keys)

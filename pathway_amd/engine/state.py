@@ -272,6 +272,97 @@ class Arrangement:
         )
 
 
+class SpineArrangement:
+    """Leveled arrangement — the GPU spine (reference spine_fueled.rs).
+
+    State lives as ≤log₂(n) sorted runs with geometric compaction: a delta
+    becomes a new run; adjacent runs merge (merge_sorted_select, O(sum))
+    when within 2× of each other.  Probes binary-search every run and sum —
+    amortizes maintenance to O(n log n) total instead of a full O(state)
+    rewrite per micro-batch (which the flat Arrangement pays; that one is
+    kept for small bounded states like reduce groups).
+    Rows may appear in several runs with partial weights; probe results sum
+    weights per (key, vhash) implicitly through downstream consolidation.
+    """
+
+    def __init__(self, device, column_protos: dict[str, Column]):
+        self.device = device
+        self.protos = column_protos
+        self.levels: list[Arrangement] = []
+
+    def __len__(self) -> int:
+        return sum(len(l) for l in self.levels)
+
+    def merge(self, keys, vhash, weights, columns, key_determined_vhash=False):
+        if keys.shape[0] == 0:
+            return
+        run = Arrangement(self.device, columns)
+        run.merge(keys, vhash, weights, columns, key_determined_vhash)
+        self.levels.append(run)
+        # geometric compaction
+        while len(self.levels) >= 2 and (
+            len(self.levels[-1]) * 2 >= len(self.levels[-2])
+        ):
+            b = self.levels.pop()
+            a = self.levels.pop()
+            if len(b) == 0:
+                merged = a
+            elif len(a) == 0:
+                merged = b
+            else:
+                a.merge(
+                    b.keys_tensor(),
+                    (b.vhash_words[0], b.vhash_words[1]),
+                    b.weights,
+                    b.columns,
+                )
+                merged = a
+            if len(merged):
+                self.levels.append(merged)
+        # keep levels sorted by size descending (compaction can disorder)
+        self.levels.sort(key=len, reverse=True)
+
+    def probe_rows(self, query_keys: torch.Tensor):
+        """All matching rows for (nq,2) query keys across levels:
+        returns (columns dict gathered, qidx, weights)."""
+        device = self.device
+        parts_cols: list[dict[str, Column]] = []
+        parts_qidx: list[torch.Tensor] = []
+        parts_w: list[torch.Tensor] = []
+        for lvl in self.levels:
+            if len(lvl) == 0:
+                continue
+            lo, hi = lvl.key_range(query_keys)
+            rows, qidx = lvl.gather_ranges(lo, hi)
+            if rows.shape[0] == 0:
+                continue
+            parts_cols.append({n: c.take(rows) for n, c in lvl.columns.items()})
+            parts_qidx.append(qidx)
+            parts_w.append(lvl.weights.index_select(0, rows))
+        if not parts_qidx:
+            z = torch.zeros((0,), dtype=torch.int64, device=device)
+            idx0 = torch.zeros((0,), dtype=torch.int64)
+            cols = {n: c.take(idx0.to(c._device())) for n, c in self.protos.items()}
+            return cols, z, z.clone()
+        if len(parts_qidx) == 1:
+            return parts_cols[0], parts_qidx[0], parts_w[0]
+        names = list(parts_cols[0].keys())
+        cols = {n: concat_columns([p[n] for p in parts_cols]) for n in names}
+        return cols, torch.cat(parts_qidx), torch.cat(parts_w)
+
+    def count_for(self, query_keys: torch.Tensor) -> torch.Tensor:
+        nq = query_keys.shape[0]
+        out = torch.zeros(nq, dtype=torch.int64, device=self.device)
+        for lvl in self.levels:
+            if len(lvl) == 0:
+                continue
+            lo, hi = lvl.key_range(query_keys)
+            rows, qidx = lvl.gather_ranges(lo, hi)
+            if rows.shape[0]:
+                out.index_add_(0, qidx, lvl.weights.index_select(0, rows))
+        return out
+
+
 class AdditiveState:
     """key -> additive accumulator columns (semigroup reduce fast path).
 

@@ -133,14 +133,33 @@ def _arrangement_from_portable(d, device):
 
 
 def _sidestore_to_portable(store) -> dict:
-    return {"arr": _arrangement_to_portable(store.arr)}
+    spine = getattr(store, "spine", None)
+    if spine is None:
+        return {"levels": None}
+    return {
+        "levels": [_arrangement_to_portable(l) for l in spine.levels],
+        "protos": {n: column_to_portable(c.take(
+            __import__("torch").zeros(0, dtype=__import__("torch").int64)
+        )) for n, c in spine.protos.items()} if spine.protos else {},
+    }
 
 
 def _sidestore_from_portable(d, device):
     from pathway_amd.engine.nodes_join import _SideStore
+    from pathway_amd.engine.state import SpineArrangement
 
     st = _SideStore(device)
-    st.arr = _arrangement_from_portable(d["arr"], device)
+    if d.get("levels") is None:
+        return st
+    levels = [_arrangement_from_portable(l, device) for l in d["levels"]]
+    protos = (
+        levels[0].columns
+        if levels
+        else {n: column_from_portable(c, device) for n, c in d.get("protos", {}).items()}
+    )
+    sp = SpineArrangement(device, protos)
+    sp.levels = [l for l in levels if l is not None]
+    st.spine = sp
     return st
 
 
