@@ -38,9 +38,7 @@ class IntervalJoinResult:
 
     def select(self, *args: Any, **kwargs: Any) -> Table:
         if self._mode != "inner":
-            raise NotImplementedError(
-                f"interval_join mode {self._mode!r} lands with the temporal phase"
-            )
+            return self._select_outer(*args, **kwargs)
         left, right = self._left, self._right
         lo, hi = self._interval.lower_bound, self._interval.upper_bound
         import pandas as pd
@@ -59,15 +57,17 @@ class IntervalJoinResult:
             ex.wrap_expr(self._right_time), {thisclass.this: right, thisclass.right: right}
         )
         if W == 0:
-            lx = left.with_columns(_pw_b=lt + lon)
-            rx = right.with_columns(_pw_b=rt)
+            lx = left.with_columns(_pw_b=lt + lon, _pw_olid=left.id)
+            rx = right.with_columns(_pw_b=rt, _pw_orid=right.id)
         else:
-            l0 = left.with_columns(_pw_b=(lt + lon) // W)
-            l1 = left.with_columns(_pw_b=(lt + hin) // W).filter(
+            l0 = left.with_columns(_pw_b=(lt + lon) // W, _pw_olid=left.id)
+            l1 = left.with_columns(
+                _pw_b=(lt + hin) // W, _pw_olid=left.id
+            ).filter(
                 (thisclass.this._pw_b) != ((self._rebind(lt, "left") + lon) // W)
             )
             lx = l0.concat_reindex(l1)
-            rx = right.with_columns(_pw_b=rt // W)
+            rx = right.with_columns(_pw_b=rt // W, _pw_orid=right.id)
         conds = [lx._pw_b == rx._pw_b]
         for c in self._on:
             c = thisclass.substitute_this(
@@ -90,6 +90,64 @@ class IntervalJoinResult:
                 & (thisclass.this._pw_rt <= thisclass.this._pw_lt + hin)
             )
         return flt.without("_pw_lt", "_pw_rt")
+
+    def _select_outer(self, *args: Any, **kwargs: Any) -> Table:
+        """left/right/outer interval joins: inner pairs + None-padded
+        unmatched rows (plain column references only in select)."""
+        import pathway_amd.internals.common as common
+        import pathway_amd.reducers as reducers
+
+        left, right = self._left, self._right
+        inner = IntervalJoinResult(
+            left, right, self._left_time, self._right_time, self._interval,
+            self._on, "inner",
+        )
+        # resolve requested columns to (side, name, out_name)
+        wanted: list[tuple[str, str, str]] = []
+        for a in args:
+            a2 = thisclass.substitute_this(
+                ex.wrap_expr(a), {thisclass.left: left, thisclass.right: right}
+            )
+            if not isinstance(a2, ex.ColumnReference):
+                raise NotImplementedError("outer interval select needs plain columns")
+            side = "l" if (a2.table is left or getattr(a2.table, "_node", None) is left._node) else "r"
+            wanted.append((side, a2.name, a2.name))
+        for name, e in kwargs.items():
+            e2 = thisclass.substitute_this(
+                ex.wrap_expr(e), {thisclass.left: left, thisclass.right: right}
+            )
+            if not isinstance(e2, ex.ColumnReference):
+                raise NotImplementedError("outer interval select needs plain columns")
+            side = "l" if (e2.table is left or getattr(e2.table, "_node", None) is left._node) else "r"
+            wanted.append((side, e2.name, name))
+        sel_kwargs = {out: (thisclass.left[src] if side == "l" else thisclass.right[src])
+                      for side, src, out in wanted}
+        sel_kwargs["_pw_lid"] = thisclass.left._pw_olid
+        sel_kwargs["_pw_rid"] = thisclass.right._pw_orid
+        inner_t = inner.select(**sel_kwargs)
+        parts = [inner_t.without("_pw_lid", "_pw_rid")]
+        if self._mode in ("left", "outer"):
+            matched = inner_t.groupby(inner_t._pw_lid).reduce(
+                _pw_k=thisclass.this._pw_lid
+            ).with_id_from_expr(ex.ColumnReference(None, "_pw_k"))
+            unmatched = left.difference(matched)
+            pad = {
+                out: (ex.ColumnReference(unmatched, src) if side == "l" else None)
+                for side, src, out in wanted
+            }
+            parts.append(unmatched.select(**pad))
+        if self._mode in ("right", "outer"):
+            matched = inner_t.groupby(inner_t._pw_rid).reduce(
+                _pw_k=thisclass.this._pw_rid
+            ).with_id_from_expr(ex.ColumnReference(None, "_pw_k"))
+            unmatched = right.difference(matched)
+            pad = {
+                out: (ex.ColumnReference(unmatched, src) if side == "r" else None)
+                for side, src, out in wanted
+            }
+            parts.append(unmatched.select(**pad))
+        out = parts[0].concat_reindex(*parts[1:]) if len(parts) > 1 else parts[0]
+        return out
 
     def _rebind(self, e, side):
         return e

@@ -427,3 +427,31 @@ def test_table_sort_prev_next():
             assert v_by_key[prev] == 10 and v_by_key[nxt] == 30
         else:
             assert v_by_key[prev] == 20 and nxt is None
+
+
+def test_interval_join_left_outer():
+    a = T(
+        """
+        t | x
+        1 | a1
+        5 | a5
+        """
+    )
+    b = T(
+        """
+        t | y
+        2 | b2
+        9 | b9
+        """
+    )
+    res = pw.temporal.interval_join_left(
+        a, b, a.t, b.t, pw.temporal.interval(-2, 2)
+    ).select(pw.left.x, pw.right.y)
+    expected = T(
+        """
+        x  | y
+        a1 | b2
+        a5 |
+        """
+    )
+    assert_table_equality_wo_index(res, expected)
