@@ -902,3 +902,116 @@ def _mask_missing(col: Column, found: torch.Tensor, device) -> Column:
     out = [v if ok else None for v, ok in zip(vals, f)]
     from pathway_amd.engine.column import obj_array
     return ObjectColumn(obj_array(out), col.dtype)
+
+
+class ToStreamNode(Node):
+    """table_to_stream (graph.rs): each update becomes an append-only event
+    row keyed uniquely, with an is_upsert flag."""
+
+    def __init__(self, input_node, device):
+        super().__init__([input_node], device)
+        self.salt = _salt("to_stream", self.node_id)
+        self.seq = 0
+
+    def reset(self):
+        self.seq = 0
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        n = len(b)
+        seq = torch.arange(self.seq, self.seq + n, dtype=torch.int64, device=self.device)
+        self.seq += n
+        slo, shi = hashing.value_hash_words(seq, 2)
+        lo, hi = hashing.derive_key_words(
+            self.salt,
+            [
+                (b.keys[:, 0].contiguous(), b.keys[:, 1].contiguous()),
+                (slo, shi),
+            ],
+        )
+        cols = dict(b.columns)
+        cols["is_upsert"] = TensorColumn(b.diffs > 0, dt.BOOL)
+        cols["_pw_source_id"] = PointerColumn(b.keys)
+        keys = torch.stack([lo, hi], dim=1)
+        diffs = torch.ones(n, dtype=torch.int64, device=self.device)
+        return DeltaBatch(keys, cols, diffs, time)
+
+
+class StreamToTableNode(Node):
+    """stream_to_table: interpret an event stream (with is_upsert) as
+    upserts keyed by a source-id column."""
+
+    def __init__(self, input_node, upsert_name: str, device):
+        super().__init__([input_node], device)
+        self.upsert_name = upsert_name
+        from pathway_amd.engine.nodes_join import _SideStore
+
+        self.store = _SideStore(device)
+
+    def reset(self):
+        from pathway_amd.engine.nodes_join import _SideStore
+
+        self.store = _SideStore(self.device)
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        ups = b.columns[self.upsert_name]
+        src = b.columns.get("_pw_source_id")
+        if src is None:
+            raise ValueError("stream_to_table needs a _pw_source_id column")
+        keys = src.pairs
+        up = ups.tensor if isinstance(ups, TensorColumn) else torch.tensor(
+            [bool(v) for v in ups.to_pylist()], device=self.device
+        )
+        diffs = torch.where(
+            up.to(torch.bool),
+            torch.ones(len(b), dtype=torch.int64, device=self.device),
+            -torch.ones(len(b), dtype=torch.int64, device=self.device),
+        )
+        cols = {
+            n: c
+            for n, c in b.columns.items()
+            if n not in (self.upsert_name, "_pw_source_id")
+        }
+        return DeltaBatch(keys, cols, diffs, time)
+
+
+class FreezeAnswersNode(Node):
+    """as-of-now answer freezing: the first emission per key wins; later
+    positive updates for an answered key are dropped (retractions of the
+    answered row pass through once)."""
+
+    def __init__(self, input_node, device):
+        super().__init__([input_node], device)
+        self.answered: dict = {}
+
+    def reset(self):
+        self.answered = {}
+
+    def step(self, time, inputs):
+        b = inputs[0]
+        if b is None or len(b) == 0:
+            return None
+        b = consolidate_batch(b)
+        if b is None:
+            return None
+        keep = []
+        keys = b.keys.cpu().tolist()
+        diffs = b.diffs.cpu().tolist()
+        for i in range(len(b)):
+            k = tuple(keys[i])
+            if diffs[i] > 0:
+                if k not in self.answered:
+                    self.answered[k] = True
+                    keep.append(i)
+            else:
+                if self.answered.pop(k, None):
+                    keep.append(i)
+        if not keep:
+            return None
+        idx = torch.tensor(keep, dtype=torch.int64, device=self.device)
+        return b.take(idx)

@@ -712,10 +712,29 @@ class Table(TableLike):
     def to_stream(self) -> "Table":
         """table_to_stream (graph.rs): events as an append-only table with
         an is_upsert flag column."""
-        raise NotImplementedError("to_stream lands in the next round")
+        from pathway_amd.engine.nodes import ToStreamNode
 
-    def stream_to_table(self, is_upsert: Any) -> "Table":
-        raise NotImplementedError("stream_to_table lands in the next round")
+        node = ToStreamNode(self._node, get_device())
+        dtypes = dict(self._dtypes)
+        dtypes["is_upsert"] = dt.BOOL
+        dtypes["_pw_source_id"] = dt.POINTER
+        return Table(node, dtypes, Universe())
+
+    def stream_to_table(self, is_upsert: Any = None) -> "Table":
+        from pathway_amd.engine.nodes import StreamToTableNode
+
+        name = "is_upsert"
+        if is_upsert is not None:
+            r = self._resolve(is_upsert)
+            if isinstance(r, ex.ColumnReference):
+                name = r.name
+        node = StreamToTableNode(self._node, name, get_device())
+        dtypes = {
+            n: t
+            for n, t in self._dtypes.items()
+            if n not in (name, "_pw_source_id")
+        }
+        return Table(node, dtypes, Universe())
 
     def unpack_snapshots(self, *args, **kwargs) -> "Table":
         raise NotImplementedError("unpack_snapshots lands in the next round")

@@ -179,12 +179,32 @@ def asof_join_outer(self, other, self_time, other_time, *on, defaults=None, dire
     return AsofJoinResult(self, other, self_time, other_time, on, "outer", direction, defaults)
 
 
-def asof_now_join(self, other, *on, how=None, **kw):
-    """Join each left row with the CURRENT right side (query-style join)."""
-    mode = how.value if hasattr(how, "value") else (how or "inner")
-    from pathway_amd.internals.joins import JoinMode, JoinResult
+class AsofNowJoinResult:
+    """asof_now: each left row is answered with the right side AS OF its
+    arrival; later right-side changes do not retro-update the answer
+    (reference asof_now_join / use-as-of-now semantics)."""
 
-    return JoinResult(self, other, list(on), JoinMode(mode))
+    def __init__(self, left, right, on, mode):
+        from pathway_amd.internals.joins import JoinMode, JoinResult
+
+        self._inner = JoinResult(
+            left, right, list(on), JoinMode(mode), assign_id=left.id
+        )
+        self._left = left
+
+    def select(self, *args, **kwargs):
+        from pathway_amd.engine.nodes import FreezeAnswersNode
+        from pathway_amd.internals.config import get_device
+        from pathway_amd.internals.table import Table as _Table
+
+        t = self._inner.select(*args, **kwargs)
+        node = FreezeAnswersNode(t._node, get_device())
+        return _Table(node, t._dtypes, t._universe)
+
+
+def asof_now_join(self, other, *on, how=None, **kw):
+    mode = how.value if hasattr(how, "value") else (how or "inner")
+    return AsofNowJoinResult(self, other, on, mode)
 
 
 def asof_now_join_inner(self, other, *on, **kw):

@@ -455,3 +455,53 @@ def test_interval_join_left_outer():
         """
     )
     assert_table_equality_wo_index(res, expected)
+
+
+def test_to_stream_and_back():
+    t = T(
+        """
+        v | __time__ | __diff__
+        1 | 0        | 1
+        2 | 2        | 1
+        1 | 4        | -1
+        """,
+        id_from=["v"],
+    )
+    ev = t.to_stream()
+    keys, cols = pw.debug.table_to_dicts(ev)
+    flags = sorted(cols["is_upsert"].values())
+    assert flags == [False, True, True]
+    back = ev.stream_to_table()
+    assert_table_equality_wo_index(
+        back,
+        T(
+            """
+            v
+            2
+            """
+        ),
+    )
+
+
+def test_asof_now_join_freezes_answers():
+    queries = T(
+        """
+        q | __time__
+        1 | 2
+        """
+    )
+    state = T(
+        """
+        q | v | __time__ | __diff__
+        1 | 10 | 0       | 1
+        1 | 10 | 4       | -1
+        1 | 99 | 4       | 1
+        """
+    )
+    res = pw.temporal.asof_now_join(queries, state, queries.q == state.q).select(
+        pw.left.q, pw.right.v
+    )
+    keys, cols = pw.debug.table_to_dicts(res)
+    # the query was answered at t=2 with v=10; the t=4 state change must
+    # not retro-update the frozen answer
+    assert list(cols["v"].values()) == [10]
