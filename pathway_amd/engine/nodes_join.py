@@ -138,6 +138,7 @@ class JoinNode(Node):
         key_mode: str = "pair",
         left_id_name: str | None = None,
         right_id_name: str | None = None,
+        probe_only_left: bool = False,
     ):
         super().__init__([left, right], device)
         self.left_on = left_on
@@ -148,6 +149,9 @@ class JoinNode(Node):
         self.key_mode = key_mode
         self.left_id_name = left_id_name
         self.right_id_name = right_id_name
+        #: as-of-now semantics: right deltas update state without emitting;
+        #: only left (query) deltas produce output (answers frozen)
+        self.probe_only_left = probe_only_left
         self.lstore = _SideStore(device)
         self.rstore = _SideStore(device)
         self.pair_salt = _salt("join_pair", self.node_id)
@@ -210,22 +214,29 @@ class JoinNode(Node):
         old_lpad = self._padded_rows(affected, "left") if pad_left else None
         old_rpad = self._padded_rows(affected, "right") if pad_right else None
 
-        # dR >< L_old
-        if jr is not None:
-            out_parts += self._emit_pairs(br, jr, self.lstore, probe_is_right=True)
-        # merge left
-        if jl is not None:
-            self.lstore.merge(jl, bl)
-        # dL >< R_old (right not yet merged) would double count dLxdR when we
-        # also merged dR first; we merge dR AFTER probing with dL:
-        if jl is not None:
-            out_parts += self._emit_pairs(bl, jl, self.rstore, probe_is_right=False)
-        if jr is not None:
-            self.rstore.merge(jr, br)
-        # wait: dL><dR must appear exactly once. Above: dR probed L_old
-        # (excludes dL), dL probed R_old (excludes dR) — dLxdR missing.
-        if jl is not None and jr is not None:
-            out_parts += self._emit_delta_cross(bl, jl, br, jr)
+        if self.probe_only_left:
+            # as-of-now: right state updates first (queries at time t see
+            # the state AS OF t), then left deltas probe; no dR emissions
+            if jr is not None:
+                self.rstore.merge(jr, br)
+            if jl is not None:
+                out_parts += self._emit_pairs(bl, jl, self.rstore, probe_is_right=False)
+                self.lstore.merge(jl, bl)
+        else:
+            # dR >< L_old
+            if jr is not None:
+                out_parts += self._emit_pairs(br, jr, self.lstore, probe_is_right=True)
+            # merge left
+            if jl is not None:
+                self.lstore.merge(jl, bl)
+            # dL >< R_old, then merge dR; the dL><dR term is emitted once
+            # via _emit_delta_cross below
+            if jl is not None:
+                out_parts += self._emit_pairs(bl, jl, self.rstore, probe_is_right=False)
+            if jr is not None:
+                self.rstore.merge(jr, br)
+            if jl is not None and jr is not None:
+                out_parts += self._emit_delta_cross(bl, jl, br, jr)
 
         # new padded rows (post-merge)
         if pad_left:

@@ -39,11 +39,13 @@ class JoinResult(Joinable):
         assign_id: Any = None,
         left_instance: Any = None,
         right_instance: Any = None,
+        probe_only_left: bool = False,
     ):
         self._left = left
         self._right = right
         self._mode = mode if isinstance(mode, JoinMode) else JoinMode(mode)
         self._assign_id = assign_id
+        self._probe_only_left = probe_only_left
         self._left_on: list[ex.ColumnExpression] = []
         self._right_on: list[ex.ColumnExpression] = []
         for cond in on:
@@ -112,6 +114,7 @@ class JoinResult(Joinable):
             key_mode=self._key_mode(),
             left_id_name="l.__id__",
             right_id_name="r.__id__",
+            probe_only_left=self._probe_only_left,
         )
 
         out_exprs: dict[str, ex.ColumnExpression] = {}

@@ -188,18 +188,17 @@ class AsofNowJoinResult:
         from pathway_amd.internals.joins import JoinMode, JoinResult
 
         self._inner = JoinResult(
-            left, right, list(on), JoinMode(mode), assign_id=left.id
+            left,
+            right,
+            list(on),
+            JoinMode(mode),
+            assign_id=left.id,
+            probe_only_left=True,
         )
         self._left = left
 
     def select(self, *args, **kwargs):
-        from pathway_amd.engine.nodes import FreezeAnswersNode
-        from pathway_amd.internals.config import get_device
-        from pathway_amd.internals.table import Table as _Table
-
-        t = self._inner.select(*args, **kwargs)
-        node = FreezeAnswersNode(t._node, get_device())
-        return _Table(node, t._dtypes, t._universe)
+        return self._inner.select(*args, **kwargs)
 
 
 def asof_now_join(self, other, *on, how=None, **kw):
