@@ -347,7 +347,11 @@ class Runtime:
         n_ingested = 0
         n_output = 0
         pm = self.persistence
+        import time as _t
+
+        ops_stats = self.stats.operators
         for node in self.nodes:
+            _s0 = _t.perf_counter()
             if isinstance(node, InputNode):
                 if injected is not None:
                     out = injected.get(id(node))
@@ -367,6 +371,12 @@ class Runtime:
                     out = None
                 else:
                     out = node.step(time, ins)
+                    st = ops_stats[type(node).__name__]
+                    st.steps += 1
+                    st.total_time_s += _t.perf_counter() - _s0
+                    st.rows_in += sum(len(b) for b in ins if b is not None)
+                    if out is not None:
+                        st.rows_out += len(out)
             outputs[id(node)] = out
         for node in self.nodes:
             if isinstance(node, (SubscribeNode, OutputNode, CaptureNode)):
