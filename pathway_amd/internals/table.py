@@ -438,13 +438,24 @@ class Table(TableLike):
         return Table(node, self._dtypes, other._universe)
 
     def having(self, *indexers: Any) -> "Table":
+        """Restrict to rows whose id appears in the indexer pointer columns
+        (reference table.py having)."""
         out = self
         for ixr in indexers:
-            ixr = self._resolve(ixr) if not isinstance(ixr, Table) else ixr
             if isinstance(ixr, ex.ColumnReference) and isinstance(ixr.table, Table):
-                keyed = ixr.table.select(_pw_key=ixr).with_id_from_expr(
-                    ex.ColumnReference(ixr.table, ixr.name)
-                )
+                src = ixr.table
+                if self._dtypes.get(ixr.name) is not None and ixr.name in self._dtypes:
+                    pass
+                # pointer column: key the indexer table by the pointer value
+                if isinstance(src._dtypes.get(ixr.name), dt.Pointer):
+                    keyed = src.select(_pw_key=ixr).with_id_from_expr(
+                        ex.ColumnReference(None, "_pw_key")
+                    )
+                else:
+                    # value column: ids of self are hashes of these values
+                    keyed = src.select(_pw_key=ixr).with_id_from(
+                        ex.ColumnReference(None, "_pw_key")
+                    )
                 from pathway_amd.engine.nodes_join import SemiJoinNode
 
                 node = SemiJoinNode(out._node, keyed._node, "intersect", get_device())
