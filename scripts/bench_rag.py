@@ -76,7 +76,7 @@ def bench_embedder(batch=1024, iters=10, seq_len=64):
     return tps
 
 
-def bench_rag_serving(n_docs=20000, n_queries=200):
+def bench_rag_serving(n_docs=20000, n_queries=200, qps: float | None = None):
     """DocumentStore retrieve latency: docs indexed once, then per-query
     incremental steps (p50/p95 end-to-end in-engine latency)."""
     from pathway_amd.debug import table_from_rows
@@ -115,7 +115,13 @@ def bench_rag_serving(n_docs=20000, n_queries=200):
     ingest_s = time.perf_counter() - t_ing0
     lat = []
     t = 100
+    start = time.perf_counter()
     for i in range(n_queries):
+        if qps:
+            target = start + i / qps
+            now = time.perf_counter()
+            if now < target:
+                time.sleep(target - now)
         qtext = " ".join(rng.choice(words, size=8))
         lo, hi = hash_values([i, "q"])
         s0 = time.perf_counter()
@@ -133,6 +139,7 @@ def bench_rag_serving(n_docs=20000, n_queries=200):
         "n_docs": n_docs, "n_queries": n_queries, "k": 5,
         "ingest_s": ingest_s,
         "p50_ms": p50, "p95_ms": p95,
+        "fixed_qps": qps,
         "qps_serial": 1000.0 / p50,
         "device": str(get_device()),
     }))
@@ -186,5 +193,7 @@ if __name__ == "__main__":
         bench_embedder()
     if which in ("all", "rag"):
         bench_rag_serving()
+    if which in ("all", "rag_qps"):
+        bench_rag_serving(qps=40.0)
     if which in ("all", "knn_e2e"):
         bench_knn_e2e()
