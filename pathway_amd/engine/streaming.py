@@ -37,6 +37,10 @@ class StreamingSource(Source):
         self.q: queue.Queue = queue.Queue()
         self._finished = threading.Event()
         self._seq = 0
+        self._sync_group = None
+        self._sync_member = None
+        self._sync_idx = None
+        self._held: list = []
 
     # -- producer side (reader thread) --
 
@@ -49,11 +53,19 @@ class StreamingSource(Source):
 
     def finish(self):
         self._finished.set()
+        if self._sync_group is not None:
+            # a finished member's frontier is +inf: stop throttling peers
+            self._sync_group.observe(self._sync_member, float("inf"))
+
+    def attach_sync_group(self, group, member_id, col_idx) -> None:
+        self._sync_group = group
+        self._sync_member = member_id
+        self._sync_idx = col_idx
 
     # -- consumer side (engine loop) --
 
     def has_pending(self) -> bool:
-        return not self.q.empty()
+        return bool(self._held) or not self.q.empty()
 
     def is_live(self) -> bool:
         return not self._finished.is_set() or self.has_pending()
@@ -68,12 +80,24 @@ class StreamingSource(Source):
         return None
 
     def pull(self, time: int, device) -> DeltaBatch | None:
-        rows = []
+        rows = list(self._held)
+        self._held = []
         try:
             while len(rows) < MAX_EVENTS_PER_STEP:
                 rows.append(self.q.get_nowait())
         except queue.Empty:
             pass
+        if self._sync_group is not None and rows:
+            # advance our watermark, then hold rows past the group threshold
+            idx = self._sync_idx
+            for _, values, _ in rows:
+                v = values[idx]
+                if v is not None:
+                    self._sync_group.observe(self._sync_member, float(v))
+            thr = self._sync_group.release_threshold()
+            ready = [r for r in rows if r[1][idx] is None or float(r[1][idx]) <= thr]
+            self._held = [r for r in rows if not (r[1][idx] is None or float(r[1][idx]) <= thr)]
+            rows = ready
         if not rows:
             return None
         keys = torch.tensor(

@@ -42,6 +42,7 @@ from pathway_amd.io import (
     sqlite,
 )
 from pathway_amd.io._subscribe import subscribe
+from pathway_amd.io.synchronization import register_input_synchronization_group
 
 __all__ = [
     "airbyte", "azure", "bigquery", "clickhouse", "csv", "debezium",
@@ -49,5 +50,5 @@ __all__ = [
     "iceberg", "jsonlines", "kafka", "kinesis", "logstash", "minio",
     "mongodb", "mqtt", "mssql", "mysql", "nats", "null", "plaintext",
     "postgres", "pubsub", "pulsar", "python", "questdb", "rabbitmq",
-    "redpanda", "s3", "sqlite", "subscribe",
+    "redpanda", "s3", "sqlite", "subscribe", "register_input_synchronization_group",
 ]

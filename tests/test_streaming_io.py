@@ -174,3 +174,53 @@ def test_telemetry_spans(tmp_path):
 
     recs = [json.loads(l) for l in open(path)]
     assert {r["kind"] for r in recs} == {"span", "metric"}
+
+
+@pytest.mark.timeout(120)
+def test_input_synchronization_group():
+    class Fast(pw.io.python.ConnectorSubject):
+        def run(self):
+            for i in range(10):
+                self.next(t=i * 10, v=1)
+
+    class Slow(pw.io.python.ConnectorSubject):
+        def run(self):
+            for i in range(3):
+                self.next(t=i * 10, v=2)
+                time.sleep(0.05)
+
+    schema = schema_from_types(t=int, v=int)
+    fast = pw.io.python.read(Fast(), schema=schema)
+    slow = pw.io.python.read(Slow(), schema=schema)
+    pw.io.register_input_synchronization_group(
+        fast.t, slow.t, max_difference=15
+    )
+    both = fast.concat_reindex(slow)
+    cap = both._capture()
+    from pathway_amd.engine.runtime import Runtime
+    from pathway_amd.internals.rungraph import reset_all
+
+    rt = Runtime([cap])
+    reset_all(rt.nodes)
+    # run a few steps while the slow source lags: released fast rows must
+    # stay within max_difference of the slow watermark
+    import time as _time
+
+    deadline = _time.time() + 5
+    while _time.time() < deadline:
+        t_, waiting = rt._next_time()
+        if t_ is None and not waiting:
+            break
+        if t_ is not None:
+            rt.step_once(t_)
+            rt._clock = max(rt._clock, t_ + 2)
+            fast_rows = [r for r in cap.rows if r.values[1] == 1]
+            slow_max = max(
+                [r.values[0] for r in cap.rows if r.values[1] == 2], default=None
+            )
+            if slow_max is not None and slow_max < 20:
+                assert all(r.values[0] <= slow_max + 15 for r in fast_rows)
+        else:
+            _time.sleep(0.01)
+    # eventually everything is released
+    assert len(cap.rows) == 13
