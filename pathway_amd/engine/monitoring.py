@@ -135,3 +135,34 @@ class ConsoleMonitor:
                 f"p95={s['p95_step_ms'] and round(s['p95_step_ms'], 2)}ms",
                 flush=True,
             )
+
+
+EXIT_CODE_UPSCALE = 77
+EXIT_CODE_DOWNSCALE = 78
+
+
+class WorkloadTracker:
+    """Elastic-scaling advice (reference src/engine/workload_tracker.rs +
+    dataflow.rs:7455-7499): sliding-window busy-fraction → ScaleUp/Down.
+
+    Enabled by PATHWAY_ELASTIC=1 under the `pathway_amd spawn` supervisor
+    (which restarts with ±workers on the exit codes); requires persistence
+    so the restart can recover state."""
+
+    def __init__(self, window: int = 50, high: float = 0.85, low: float = 0.15):
+        self.window = window
+        self.high = high
+        self.low = low
+        self.samples: list[float] = []
+
+    def add_point(self, busy_fraction: float) -> str | None:
+        self.samples.append(busy_fraction)
+        if len(self.samples) < self.window:
+            return None
+        self.samples = self.samples[-self.window :]
+        avg = sum(self.samples) / len(self.samples)
+        if avg > self.high:
+            return "up"
+        if avg < self.low:
+            return "down"
+        return None

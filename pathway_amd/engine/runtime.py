@@ -474,8 +474,19 @@ class Runtime:
             else:
                 self.replay_persisted()
             self._replayed = True
+        import os
         import time as _time
 
+        elastic = None
+        if os.environ.get("PATHWAY_ELASTIC") and self.persistence is not None:
+            from pathway_amd.engine.monitoring import (
+                EXIT_CODE_DOWNSCALE,
+                EXIT_CODE_UPSCALE,
+                WorkloadTracker,
+            )
+
+            elastic = WorkloadTracker()
+        _last_loop = _time.perf_counter()
         while True:
             t, waiting = self._next_time()
             if t is None:
@@ -485,7 +496,24 @@ class Runtime:
                 continue
             _s0 = _time.perf_counter()
             ingested, output = self.step_once(t)
-            self.stats.record_step(t, _time.perf_counter() - _s0, ingested, output)
+            _now = _time.perf_counter()
+            self.stats.record_step(t, _now - _s0, ingested, output)
+            if elastic is not None:
+                busy = (_now - _s0) / max(_now - _last_loop, 1e-9)
+                advice = elastic.add_point(min(busy, 1.0))
+                _last_loop = _now
+                if advice == "up":
+                    import sys
+
+                    from pathway_amd.engine.monitoring import EXIT_CODE_UPSCALE
+
+                    sys.exit(EXIT_CODE_UPSCALE)
+                if advice == "down":
+                    import sys
+
+                    from pathway_amd.engine.monitoring import EXIT_CODE_DOWNSCALE
+
+                    sys.exit(EXIT_CODE_DOWNSCALE)
             if self.monitor is not None:
                 self.monitor.maybe_report()
             self._clock = max(self._clock, t + 2)
