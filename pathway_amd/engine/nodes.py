@@ -760,7 +760,7 @@ class GroupReduceNode(Node):
                     odt = dt.FLOAT if acc.dtype == torch.float64 else dt.INT
                     cols[out_name] = TensorColumn(acc, odt)
             elif spec.family == "multiset":
-                cols[out_name] = self._multiset_agg(out_name, spec, changed, nq)
+                cols[out_name] = self._multiset_agg(out_name, spec, changed, nq, kwargs)
         return presence, cols
 
 
@@ -795,7 +795,7 @@ class GroupReduceNode(Node):
         v = acc.index_select(0, pos)
         return v * found
 
-    def _multiset_agg(self, out_name, spec, changed, nq) -> Column:
+    def _multiset_agg(self, out_name, spec, changed, nq, rkw=None) -> Column:
         device = self.device
         store = self.multiset_store
         arg0 = f"{out_name}__0"
@@ -836,8 +836,11 @@ class GroupReduceNode(Node):
                 tup = (tup[0], None)
             per_group[q].append((tup, weights[j]))
         out_vals = []
+        skip_nones = bool(rkw and rkw.get("skip_nones"))
         for rows_g in per_group:
             rows_g = [(t, w) for t, w in rows_g if w > 0]
+            if skip_nones:
+                rows_g = [(t, w) for t, w in rows_g if t[0] is not None]
             if not rows_g:
                 out_vals.append(None)
                 continue
@@ -878,8 +881,22 @@ def columns_equal_mask(a: Column, b: Column, device) -> torch.Tensor:
     if isinstance(a, PointerColumn) and isinstance(b, PointerColumn):
         return (a.pairs[:, 0] == b.pairs[:, 0]) & (a.pairs[:, 1] == b.pairs[:, 1])
     va, vb = a.to_pylist(), b.to_pylist()
+
+    def _veq(x, y):
+        if x is None or y is None:
+            return x is None and y is None
+        if isinstance(x, np.ndarray) or isinstance(y, np.ndarray):
+            return (
+                isinstance(x, np.ndarray)
+                and isinstance(y, np.ndarray)
+                and x.shape == y.shape
+                and bool(np.array_equal(x, y))
+            )
+        r = x == y
+        return bool(r) if isinstance(r, (bool, np.bool_)) else False
+
     return torch.tensor(
-        [x == y or (x is None and y is None) for x, y in zip(va, vb)],
+        [_veq(x, y) for x, y in zip(va, vb)],
         dtype=torch.bool,
         device=device,
     )

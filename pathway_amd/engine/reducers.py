@@ -166,7 +166,9 @@ _multiset(
     "sorted_tuple",
     None,
     lambda rows: tuple(
-        v for r in sorted(rows, key=lambda r: r[0][0]) for v in [r[0][0]] * r[1]
+        v
+        for r in sorted(rows, key=lambda r: (r[0][0] is not None, r[0][0]))
+        for v in [r[0][0]] * r[1]
     ),
     lambda args_dt: dt.List(dt.unoptionalize(args_dt[0])),
 )
