@@ -1,0 +1,203 @@
+"""Temporal-join variant coverage: asof directions/modes, interval join
+right/outer, window join modes (modeled on the reference's
+test_asof_join.py / test_interval_join.py / test_window_join.py)."""
+
+import pytest
+
+import pathway_amd as pw
+from pathway_amd.debug import table_from_markdown as T, table_to_dicts
+from pathway_amd.stdlib.temporal import Direction
+
+
+def _rows(table, *names):
+    _, cols = table_to_dicts(table)
+    ids = list(cols[names[0]].keys())
+    return sorted(
+        (tuple(cols[n][i] for n in names) for i in ids),
+        key=lambda r: tuple((x is None, x) for x in r),
+    )
+
+
+LEFT = """
+t  | v
+1  | a
+5  | b
+9  | c
+"""
+
+RIGHT = """
+s  | w
+2  | X
+6  | Y
+"""
+
+
+def test_asof_backward():
+    l, r = T(LEFT), T(RIGHT)
+    res = l.asof_join(
+        r, l.t, r.s, how=pw.JoinMode.INNER, direction=Direction.BACKWARD
+    ).select(pw.left.v, pw.right.w)
+    # backward: right row with largest s <= t
+    assert _rows(res, "v", "w") == [("b", "X"), ("c", "Y")]
+
+
+def test_asof_forward():
+    l, r = T(LEFT), T(RIGHT)
+    res = l.asof_join(
+        r, l.t, r.s, how=pw.JoinMode.INNER, direction=Direction.FORWARD
+    ).select(pw.left.v, pw.right.w)
+    # forward: right row with smallest s >= t
+    assert _rows(res, "v", "w") == [("a", "X"), ("b", "Y")]
+
+
+def test_asof_nearest():
+    l, r = T(LEFT), T(RIGHT)
+    res = l.asof_join(
+        r, l.t, r.s, how=pw.JoinMode.INNER, direction=Direction.NEAREST
+    ).select(pw.left.v, pw.right.w)
+    # nearest: 1→2(X), 5→6(Y), 9→6(Y)
+    assert _rows(res, "v", "w") == [("a", "X"), ("b", "Y"), ("c", "Y")]
+
+
+def test_asof_left_with_defaults():
+    l, r = T(LEFT), T(RIGHT)
+    res = l.asof_join_left(
+        r, l.t, r.s, defaults={r.w: "none"}, direction=Direction.BACKWARD
+    ).select(pw.left.v, pw.right.w)
+    assert _rows(res, "v", "w") == [("a", "none"), ("b", "X"), ("c", "Y")]
+
+
+def test_interval_join_right_and_outer():
+    l = T(
+        """
+        t | a
+        0 | p
+        10 | q
+        """
+    )
+    r = T(
+        """
+        s | b
+        1 | u
+        50 | v
+        """
+    )
+    inner_right = l.interval_join_right(
+        r, l.t, r.s, pw.temporal.interval(-2, 2)
+    ).select(pw.left.a, pw.right.b)
+    assert _rows(inner_right, "a", "b") == [("p", "u"), (None, "v")]
+
+    outer = l.interval_join_outer(
+        r, l.t, r.s, pw.temporal.interval(-2, 2)
+    ).select(pw.left.a, pw.right.b)
+    assert _rows(outer, "a", "b") == [("p", "u"), ("q", None), (None, "v")]
+
+
+def test_interval_join_with_exact_on_condition():
+    l = T(
+        """
+        t | k | a
+        0 | 1 | p
+        0 | 2 | q
+        """
+    )
+    r = T(
+        """
+        s | k | b
+        1 | 1 | u
+        1 | 3 | v
+        """
+    )
+    res = l.interval_join(
+        r, l.t, r.s, pw.temporal.interval(-2, 2), l.k == r.k
+    ).select(pw.left.a, pw.right.b)
+    assert _rows(res, "a", "b") == [("p", "u")]
+
+
+def test_window_join_left():
+    l = T(
+        """
+        t | a
+        1 | p
+        7 | q
+        """
+    )
+    r = T(
+        """
+        t | b
+        2 | u
+        """
+    )
+    res = l.window_join_left(
+        r, l.t, r.t, pw.temporal.tumbling(duration=5)
+    ).select(pw.left.a, pw.right.b)
+    assert _rows(res, "a", "b") == [("p", "u"), ("q", None)]
+
+
+def test_window_join_sliding_multiplicity():
+    l = T(
+        """
+        t | a
+        3 | p
+        """
+    )
+    r = T(
+        """
+        t | b
+        4 | u
+        """
+    )
+    # hop 2, duration 4: windows [0,4), [2,6) — t=3 in both; t=4 in [2,6),[4,8)
+    res = l.window_join_inner(
+        r, l.t, r.t, pw.temporal.sliding(hop=2, duration=4)
+    ).select(pw.left.a, pw.right.b)
+    assert _rows(res, "a", "b") == [("p", "u")]
+
+
+def test_windowby_sliding_counts():
+    t = T(
+        """
+        t | v
+        0 | 1
+        1 | 1
+        3 | 1
+        """
+    )
+    res = t.windowby(t.t, window=pw.temporal.sliding(hop=2, duration=4)).reduce(
+        start=pw.this._pw_window_start,
+        n=pw.reducers.sum(pw.this.v),
+    )
+    assert _rows(res, "start", "n") == [(-2, 2), (0, 3), (2, 1)]
+
+
+def test_session_window_max_gap_merges():
+    t = T(
+        """
+        t | v
+        1 | 1
+        2 | 1
+        9 | 1
+        """
+    )
+    res = t.windowby(
+        t.t, window=pw.temporal.session(max_gap=3)
+    ).reduce(n=pw.reducers.sum(pw.this.v))
+    assert sorted(x[0] for x in _rows(res, "n")) == [1, 2]
+
+
+def test_asof_now_join_left_pads():
+    t = T(
+        """
+        a | __time__
+        1 |    2
+        2 |    4
+        """
+    )
+    r = T(
+        """
+        a | b | __time__
+        1 | x |    2
+        """
+    )
+    res = t.asof_now_join_left(r, t.a == r.a).select(t.a, r.b)
+    assert _rows(res, "a", "b") == [(1, "x"), (2, None)]
