@@ -30,6 +30,21 @@ class ConnectorSubject:
         else:
             self._buffer.append((self._time, kwargs))
 
+    def _add(self, key: Any, message: dict) -> None:
+        """Insert a row under an explicit key (reference ConnectorSubject._add)."""
+        emit = getattr(self, "_emit_keyed", None)
+        if emit is not None:
+            emit(message, key, 1)
+        else:
+            self._buffer.append((self._time, message))
+
+    def _remove(self, key: Any, message: dict) -> None:
+        """Retract a previously added row (reference ConnectorSubject._remove):
+        the key and values must match the insertion for the diffs to cancel."""
+        emit = getattr(self, "_emit_keyed", None)
+        if emit is not None:
+            emit(message, key, -1)
+
     def next_json(self, message: dict) -> None:
         self.next(**message)
 
@@ -71,6 +86,20 @@ def read(
     src = StreamingSource(names, dtypes, name=name)
 
     subject._emit = lambda rec: src.emit([rec.get(n) for n in names])
+
+    def _emit_keyed(rec, key, diff):
+        from pathway_amd.internals.api import BasePointer, Pointer, hash_values
+
+        if key is None:
+            p = None
+        elif isinstance(key, BasePointer):
+            p = key
+        else:
+            lo, hi = hash_values(list(key) if isinstance(key, (list, tuple)) else [key])
+            p = Pointer(lo, hi)
+        src.emit([rec.get(n) for n in names], key=p, diff=diff)
+
+    subject._emit_keyed = _emit_keyed
 
     def run_subject():
         try:

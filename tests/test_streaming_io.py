@@ -237,3 +237,20 @@ def test_workload_tracker_advice():
     for _ in range(4):
         wt2.add_point(0.05)
     assert wt2.add_point(0.05) == "down"
+
+
+def test_python_connector_add_remove():
+    # _add/_remove with explicit keys: the retraction cancels the insert
+    class Subject(pw.io.python.ConnectorSubject):
+        def run(self):
+            self._add("k1", {"v": 10})
+            self._add("k2", {"v": 20})
+            time.sleep(0.02)
+            self._remove("k1", {"v": 10})
+
+    schema = schema_from_types(v=int)
+    t = pw.io.python.read(Subject(), schema=schema)
+    res = t.groupby().reduce(s=pw.reducers.sum(pw.this.v), c=pw.reducers.count())
+    keys, cols = pw.debug.table_to_dicts(res)
+    assert list(cols["s"].values()) == [20]
+    assert list(cols["c"].values()) == [1]
