@@ -468,6 +468,10 @@ class Table(TableLike):
         self._universe.promise_equal(other._universe)
         return Table(self._node, self._dtypes, other._universe)
 
+    def unsafe_promise_universes_are_equal(self, other: TableLike) -> "Table":
+        """Deprecated alias of with_universe_of (reference table.py)."""
+        return self.with_universe_of(other)
+
     # -- keys ---------------------------------------------------------------
 
     def pointer_from(self, *args: Any, optional: bool = False, instance: Any = None):

@@ -2,5 +2,18 @@
 from pathway_amd.stdlib.graphs.common import Edge, Vertex
 from pathway_amd.stdlib.graphs.pagerank import pagerank
 from pathway_amd.stdlib.graphs.bellman_ford import bellman_ford
+from pathway_amd.stdlib.graphs.louvain import (
+    exact_modularity,
+    louvain_communities,
+    louvain_level,
+)
 
-__all__ = ["pagerank", "bellman_ford", "Edge", "Vertex"]
+__all__ = [
+    "pagerank",
+    "bellman_ford",
+    "louvain_level",
+    "louvain_communities",
+    "exact_modularity",
+    "Edge",
+    "Vertex",
+]

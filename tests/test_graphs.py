@@ -66,3 +66,38 @@ def test_bellman_ford():
     keys, cols = table_to_dicts(res)
     dists = {cols["vtx"][k]: cols["dist_from_source"][k] for k in keys}
     assert dists == {"a": 0.0, "b": 1.0, "c": 3.0, "d": math.inf}
+
+
+def test_louvain_two_cliques():
+    # two triangles joined by one weak edge → two communities
+    import pathway_amd as pw
+    from pathway_amd.debug import table_from_markdown as T, table_to_dicts
+    from pathway_amd.stdlib.graphs import exact_modularity, louvain_level
+
+    raw = T(
+        """
+        a | b
+        1 | 2
+        2 | 3
+        1 | 3
+        4 | 5
+        5 | 6
+        4 | 6
+        3 | 4
+        """
+    )
+    edges = raw.select(
+        u=raw.pointer_from(pw.this.a), v=raw.pointer_from(pw.this.b)
+    )
+    cl = louvain_level(edges)
+    _, cols = table_to_dicts(cl)
+    groups = {}
+    for i in cols["u"]:
+        groups.setdefault(repr(cols["c"][i]), set()).add(repr(cols["u"][i]))
+    assert len(groups) == 2
+    assert sorted(len(g) for g in groups.values()) == [3, 3]
+
+    q = exact_modularity(edges, cl)
+    _, qc = table_to_dicts(q)
+    (qv,) = qc["modularity"].values()
+    assert qv > 0.3  # two-clique split has high modularity
