@@ -117,6 +117,16 @@ class UDF:
         fun2 = fun
         if self.cache_strategy is not None:
             fun2 = self.cache_strategy.wrap(fun)
+        from pathway_amd.udfs import AsyncExecutor, FullyAsyncExecutor
+
+        if isinstance(self.executor, AsyncExecutor):
+            wrapped = _apply_executor_options(fun2, self.executor)
+            cls = (
+                ex.FullyAsyncApplyExpression
+                if isinstance(self.executor, FullyAsyncExecutor)
+                else ex.AsyncApplyExpression
+            )
+            return cls(wrapped, ret, *args, propagate_none=self.propagate_none, **kwargs)
         if is_async:
             return ex.AsyncApplyExpression(
                 fun2, ret, *args, propagate_none=self.propagate_none, **kwargs
@@ -130,6 +140,42 @@ class UDF:
             max_batch_size=self.max_batch_size,
             **kwargs,
         )
+
+
+def _apply_executor_options(fun: Callable, executor) -> Callable:
+    """Wrap `fun` per the AsyncExecutor knobs (reference udfs/executors.py:
+    capacity semaphore, per-attempt timeout, retry strategy)."""
+    import asyncio
+    import inspect
+
+    sems: dict[int, Any] = {}
+
+    async def awrap(*a: Any, **k: Any):
+        async def call():
+            r = fun(*a, **k)
+            return await r if inspect.isawaitable(r) else r
+
+        async def timed():
+            if executor.timeout is not None:
+                return await asyncio.wait_for(call(), executor.timeout)
+            return await call()
+
+        async def retried():
+            if executor.retry_strategy is not None:
+                return await executor.retry_strategy.invoke(timed)
+            return await timed()
+
+        if executor.capacity:
+            loop = asyncio.get_running_loop()
+            sem = sems.get(id(loop))
+            if sem is None:
+                sem = sems[id(loop)] = asyncio.Semaphore(executor.capacity)
+            async with sem:
+                return await retried()
+        return await retried()
+
+    functools.update_wrapper(awrap, fun, updated=())
+    return awrap
 
 
 def udf(
