@@ -1,0 +1,173 @@
+"""Misc stdlib + io coverage: pw.io.subscribe, statistical.interpolate,
+utils.col/filtering, monitoring HTTP server (reference misc suites)."""
+
+import json
+import time
+import urllib.request
+
+import pytest
+
+import pathway_amd as pw
+from pathway_amd.debug import table_from_markdown as T, table_from_rows, table_to_dicts
+from pathway_amd.internals.schema import schema_from_types
+
+
+@pytest.fixture(autouse=True)
+def _clean_graph():
+    yield
+    pw.internals.rungraph.G.clear()
+
+
+def test_io_subscribe_callbacks():
+    t = T(
+        """
+        a | __time__ | __diff__
+        1 |    2     |    1
+        2 |    4     |    1
+        1 |    6     |   -1
+        """,
+        id_from=["a"],
+    )
+    events = []
+    done = []
+    pw.io.subscribe(
+        t,
+        on_change=lambda key, row, time, is_addition: events.append(
+            (row["a"], time, is_addition)
+        ),
+        on_end=lambda: done.append(True),
+    )
+    pw.run(monitoring_level=pw.MonitoringLevel.NONE)
+    assert (1, 2, True) in events and (2, 4, True) in events
+    assert (1, 6, False) in events
+    assert done == [True]
+
+
+def test_statistical_interpolate():
+    from pathway_amd.stdlib.statistical import interpolate
+
+    t = T(
+        """
+        t  | v
+        0  | 0.0
+        10 | 100.0
+        """
+    )
+    q = T(
+        """
+        t
+        5
+        """
+    )
+    all_t = pw.Table.concat_reindex(
+        t.select(pw.this.t, v=pw.cast(float, pw.this.v)),
+        q.select(pw.this.t, v=pw.declare_type(float, None)),
+    )
+    res = interpolate(all_t, pw.this.t, pw.this.v)
+    _, cols = table_to_dicts(res)
+    vals = sorted(v for v in cols["v"].values() if v is not None)
+    assert 50.0 in vals  # linear midpoint filled in
+
+
+def test_utils_col_unpack_col():
+    from pathway_amd.stdlib.utils.col import unpack_col
+
+    t = T(
+        """
+        a | b
+        1 | 2
+        """
+    )
+    packed = t.select(tup=pw.make_tuple(pw.this.a, pw.this.b))
+    un = unpack_col(packed.tup, "x", "y")
+    _, cols = table_to_dicts(un)
+    assert list(cols["x"].values()) == [1]
+    assert list(cols["y"].values()) == [2]
+
+
+def test_utils_filtering_argmin_rows():
+    from pathway_amd.stdlib.utils.filtering import argmin_rows
+
+    t = T(
+        """
+        g | v | extra
+        a | 3 | p
+        a | 1 | q
+        b | 7 | r
+        """
+    )
+    res = argmin_rows(t, pw.this.g, what=pw.this.v)
+    _, cols = table_to_dicts(res)
+    got = sorted(zip(cols["g"].values(), cols["v"].values(), cols["extra"].values()))
+    assert got == [("a", 1, "q"), ("b", 7, "r")]
+
+
+def test_bucketing_helpers():
+    import datetime
+
+    from pathway_amd.stdlib.utils.bucketing import (
+        truncate_to_hours,
+        truncate_to_minutes,
+    )
+
+    d = datetime.datetime(2023, 3, 25, 12, 34, 56, 789)
+    assert truncate_to_minutes(d) == datetime.datetime(2023, 3, 25, 12, 34)
+    assert truncate_to_hours(d) == datetime.datetime(2023, 3, 25, 12)
+
+
+def test_monitoring_status_endpoint():
+    from pathway_amd.engine.monitoring import RunStats, start_http_server
+
+    stats = RunStats()
+    stats.record_step(2, 0.001, 10, 5)
+    httpd = start_http_server(stats, port=0)
+    try:
+        port = httpd.server_address[1]
+        with urllib.request.urlopen(f"http://127.0.0.1:{port}/status") as r:
+            st = json.loads(r.read())
+        assert st["rows_ingested"] == 10 and st["rows_output"] == 5
+        with urllib.request.urlopen(f"http://127.0.0.1:{port}/metrics") as r:
+            body = r.read().decode()
+        assert "pathway_rows_ingested_total 10" in body
+    finally:
+        httpd.shutdown()
+
+
+def test_deduplicate_with_acceptor():
+    t = T(
+        """
+        g | v | __time__
+        a | 1 |    2
+        a | 3 |    4
+        a | 2 |    6
+        a | 7 |    8
+        """
+    )
+    # accept a new value only if it is at least 2 bigger than the last kept
+    res = t.deduplicate(
+        value=pw.this.v,
+        instance=pw.this.g,
+        acceptor=lambda new, old: new >= old + 2,
+    )
+    _, cols = table_to_dicts(res)
+    assert list(cols["v"].values()) == [7]  # 1 -> 3 -> (2 rejected) -> 7; last state
+
+
+def test_ordered_diff():
+    from pathway_amd.stdlib.ordered import diff
+
+    t = T(
+        """
+        t | v
+        1 | 10
+        2 | 13
+        3 | 11
+        """
+    )
+    res = diff(t, t.t, t.v)
+    _, cols = table_to_dicts(res)
+    by_t = {}
+    keys, _ = table_to_dicts(t)
+    # diff returns per-row deltas vs the previous row in t-order
+    vals = sorted(v for v in cols["diff_v"].values() if v is not None)
+    assert vals == [-2, 3]
