@@ -249,3 +249,46 @@ def test_terminate_on_error_raises_at_sink():
     assert pathway_config.terminate_on_error
     with pytest.raises(RuntimeError, match="Error value"):
         pw.debug.table_to_dicts(res)
+
+
+def test_global_error_log_records_udf_failures():
+    from pathway_amd.internals.errors import _global_error_rows
+
+    before = len(_global_error_rows)
+
+    @pw.udf
+    def boom(x: int) -> int:
+        if x == 2:
+            raise ValueError("bad row 2")
+        return x
+
+    t = T(
+        """
+        a
+        1
+        2
+        """
+    )
+    res = t.select(b=pw.fill_error(boom(pw.this.a), -1))
+    _, cols = pw.debug.table_to_dicts(res)
+    assert sorted(cols["b"].values()) == [-1, 1]
+    log = pw.global_error_log()
+    _, lcols = pw.debug.table_to_dicts(log)
+    msgs = list(lcols["message"].values())
+    assert any("bad row 2" in m for m in msgs)
+    assert len(msgs) > before
+
+
+def test_iterate_iteration_limit():
+    def step(t):
+        return t.select(v=pw.if_else(pw.this.v < 100, pw.this.v * 2, pw.this.v))
+
+    t = T(
+        """
+        v
+        1
+        """
+    )
+    limited = pw.iterate(step, iteration_limit=3, t=t)
+    _, cols = pw.debug.table_to_dicts(limited)
+    assert list(cols["v"].values()) == [8]  # 3 doublings only
