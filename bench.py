@@ -77,7 +77,10 @@ def main() -> None:
         comm = par.init(backend=backend, device=device)
 
     # ---- synthetic parsed word stream (identical vocab on all ranks) ----
-    vocab = [f"word{i:06d}" for i in range(args.vocab)]
+    # fixed word width: zero-pad to however many digits the vocab needs
+    DIG = max(6, len(str(args.vocab - 1)))
+    WW = 4 + DIG  # b"word" + digits
+    vocab = [f"word{i:0{DIG}d}" for i in range(args.vocab)]
     GLOBAL_STRING_POOL.codes(vocab)
     GLOBAL_STRING_POOL.synchronized = True
     GLOBAL_STRING_POOL.hash_tensors(device)  # pre-stage pool hashes in HBM
@@ -124,11 +127,11 @@ def main() -> None:
             self.batch = batch
             self.vocab_n = vocab_n
             self.counter = 0
-            # device matrix of the vocabulary's word bytes (fixed width 10)
+            # device matrix of the vocabulary's word bytes (fixed width WW)
             wb = np.frombuffer(
-                "".join(f"word{i:06d}" for i in range(vocab_n)).encode(),
+                "".join(f"word{i:0{DIG}d}" for i in range(vocab_n)).encode(),
                 dtype=np.uint8,
-            ).reshape(vocab_n, 10)
+            ).reshape(vocab_n, WW)
             self.word_bytes = torch.from_numpy(wb.copy()).to(device)
 
         def next_time(self):
@@ -144,9 +147,9 @@ def main() -> None:
                 0, self.vocab_n, (n,), dtype=torch.int64, generator=gen
             ).to(dev, non_blocking=True)
             # build the wire buffer: "<word>\n" per message (device gather)
-            msg = torch.empty((n, 11), dtype=torch.uint8, device=dev)
-            msg[:, :10] = self.word_bytes.index_select(0, codes)
-            msg[:, 10] = 10  # newline
+            msg = torch.empty((n, WW + 1), dtype=torch.uint8, device=dev)
+            msg[:, :WW] = self.word_bytes.index_select(0, codes)
+            msg[:, WW] = 10  # newline
             buf = msg.reshape(-1).contiguous()
             # PARSE on device: newline scan -> token [start, end) spans
             nl = (buf == 10).nonzero(as_tuple=True)[0]
