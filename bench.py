@@ -231,6 +231,23 @@ def main() -> None:
     barrier()
     sync()
 
+    if os.environ.get("PW_TORCH_PROF"):
+        # host/device breakdown of the steady-state step (diagnostics only)
+        from torch.profiler import ProfilerActivity, profile
+
+        acts = [ProfilerActivity.CPU]
+        if use_cuda:
+            acts.append(ProfilerActivity.CUDA)
+        with profile(activities=acts) as prof:
+            for _ in range(3):
+                rt.step_once(t)
+                t += 2
+            sync()
+        ka = prof.key_averages()
+        print(ka.table(sort_by="self_cpu_time_total", row_limit=30))
+        if use_cuda:
+            print(ka.table(sort_by="self_cuda_time_total", row_limit=20))
+
     # ---- timed region: exactly K steps ----
     step_times = []
     t0 = time.perf_counter()
