@@ -85,7 +85,10 @@ def main() -> None:
     GLOBAL_STRING_POOL.synchronized = True
     GLOBAL_STRING_POOL.hash_tensors(device)  # pre-stage pool hashes in HBM
 
-    gen = torch.Generator(device="cpu")
+    # synthetic stream RNG lives ON DEVICE: a real connector DMAs messages
+    # into HBM; generating on host and copying would bill a host artifact
+    # to the engine. Per-rank independent streams via distinct seeds.
+    gen = torch.Generator(device=device if use_cuda else "cpu")
     gen.manual_seed(1234 + rank)
 
     class WordStream:
@@ -102,7 +105,8 @@ def main() -> None:
         def pull(self, t, dev):
             n = self.batch
             codes = torch.randint(
-                0, self.vocab_n, (n,), dtype=torch.int64, generator=gen
+                0, self.vocab_n, (n,), dtype=torch.int64, generator=gen,
+                device=gen.device,
             ).to(dev, non_blocking=True)
             seq = torch.arange(
                 self.counter, self.counter + n, dtype=torch.int64, device=dev
@@ -144,7 +148,8 @@ def main() -> None:
 
             n = self.batch
             codes = torch.randint(
-                0, self.vocab_n, (n,), dtype=torch.int64, generator=gen
+                0, self.vocab_n, (n,), dtype=torch.int64, generator=gen,
+                device=gen.device,
             ).to(dev, non_blocking=True)
             # build the wire buffer: "<word>\n" per message (device gather)
             msg = torch.empty((n, WW + 1), dtype=torch.uint8, device=dev)

@@ -57,9 +57,9 @@ def main():
         def pull(self, t, dev):
             base = self.counter
             self.counter += 1
-            g = torch.Generator(device="cpu").manual_seed(base)
+            g = torch.Generator(device=dev).manual_seed(base)
             ts = (
-                torch.rand(B, generator=g).to(dev) * args.duration + base
+                torch.rand(B, generator=g, device=dev) * args.duration + base
             ).to(torch.int64)
             vals = torch.arange(B, dtype=torch.int64, device=dev) % 1000
             seq = torch.arange(base * B, base * B + B, dtype=torch.int64, device=dev)
