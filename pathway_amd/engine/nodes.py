@@ -53,6 +53,11 @@ from pathway_amd.engine.state import (
     searchsorted_words,
 )
 
+import os as _os
+
+#: disable the HIP hash-aggregation pre-agg path (A/B testing)
+_PW_NO_HASHAGG = bool(_os.environ.get("PW_NO_HASHAGG"))
+
 _node_counter = [0]
 
 
@@ -452,45 +457,99 @@ class GroupReduceNode(Node):
             rowkeys = b.keys
             diffs = b.diffs
 
-            # 3. local pre-aggregation: ONE lex sort of the batch's group
-            # keys, then segmented sums per additive accumulator (the
-            # combiner — also what the exchange ships for N>1)
-            words = [gkeys[:, 0].contiguous(), gkeys[:, 1].contiguous()]
-            perm = lex_sort_words(words)
-            swords = [w.index_select(0, perm) for w in words]
-            starts = rows_ne(swords)
-            seg = torch.cumsum(starts.to(torch.int64), 0) - 1
-            first_idx = starts.nonzero(as_tuple=True)[0]
-            nseg = int(first_idx.numel())
-            ukeys_w = [w.index_select(0, first_idx) for w in swords]
+            # 3. local pre-aggregation (the combiner — also what the
+            # exchange ships for N>1)
             contribs = self._contributions(arg_cols, diffs, specs)
-            # segmented sums over the sorted batch via prefix-sum boundary
-            # differences — exact for int64, atomic-free (index_add_ was the
-            # contention hot spot at high duplicates-per-group)
-            nrows = perm.shape[0]
-            seg_ends = (
-                torch.cat(
-                    [
-                        first_idx[1:],
-                        torch.tensor([nrows], dtype=torch.int64, device=device),
-                    ]
+            hashagg = None
+            if (
+                torch.device(device).type == "cuda"
+                and not _PW_NO_HASHAGG
+                and all(c.dtype == torch.int64 for c in contribs.values())
+            ):
+                # sort-free path: one HIP hash-aggregation pass over the
+                # batch (pw_hash_agg) — replaces the 4M-row radix sort +
+                # segment-boundary chain with atomics into an open-
+                # addressing table, then sort/consolidate only the (small)
+                # distinct-key set
+                from pathway_amd import ops
+
+                names = list(contribs)
+                uk0, uk1, accs, rep = ops.hash_agg_gpu(
+                    gkeys[:, 0].contiguous(),
+                    gkeys[:, 1].contiguous(),
+                    [contribs[nm] for nm in names],
                 )
-                - 1
-            )
-            acc_deltas = {}
-            for name, c in contribs.items():
-                sc = c.index_select(0, perm)
-                if sc.dtype == torch.float64:
-                    acc = torch.zeros(nseg, dtype=sc.dtype, device=device)
-                    acc.index_add_(0, seg, sc)
+                perm = lex_sort_words([uk0, uk1])
+                uk0s = uk0.index_select(0, perm)
+                uk1s = uk1.index_select(0, perm)
+                starts = rows_ne([uk0s, uk1s])
+                first_idx = starts.nonzero(as_tuple=True)[0]
+                nseg = int(first_idx.numel())
+                if nseg == uk0s.shape[0]:
+                    # no duplicate slots (the common case)
+                    ukeys_w = [uk0s, uk1s]
+                    acc_deltas = {
+                        nm: a.index_select(0, perm) for nm, a in zip(names, accs)
+                    }
+                    rep_first = rep.index_select(0, perm)
                 else:
-                    cs = torch.cumsum(sc, 0)
-                    seg_tot = cs.index_select(0, seg_ends)
-                    acc = seg_tot.clone()
-                    acc[1:] -= seg_tot[:-1]
-                acc_deltas[name] = acc
-            gfirst_rows = perm.index_select(0, first_idx)
-            gcols_first = {n: c.take(gfirst_rows) for n, c in gcols.items()}
+                    # rare publication race: merge duplicate-key slots
+                    seg = torch.cumsum(starts.to(torch.int64), 0) - 1
+                    ukeys_w = [
+                        w.index_select(0, first_idx) for w in (uk0s, uk1s)
+                    ]
+                    acc_deltas = {}
+                    for nm, a in zip(names, accs):
+                        sa = a.index_select(0, perm)
+                        out = torch.zeros(nseg, dtype=sa.dtype, device=device)
+                        out.index_add_(0, seg, sa)
+                        acc_deltas[nm] = out
+                    rep_first = rep.index_select(0, perm).index_select(
+                        0, first_idx
+                    )
+                gcols_first = {n: c.take(rep_first) for n, c in gcols.items()}
+                hashagg = True
+            if hashagg is None:
+                # sort path: ONE lex sort of the batch's group keys, then
+                # segmented sums per additive accumulator
+                words = [gkeys[:, 0].contiguous(), gkeys[:, 1].contiguous()]
+                perm = lex_sort_words(words)
+                swords = [w.index_select(0, perm) for w in words]
+                starts = rows_ne(swords)
+                seg = torch.cumsum(starts.to(torch.int64), 0) - 1
+                first_idx = starts.nonzero(as_tuple=True)[0]
+                nseg = int(first_idx.numel())
+                ukeys_w = [w.index_select(0, first_idx) for w in swords]
+                # segmented sums over the sorted batch via prefix-sum
+                # boundary differences — exact for int64, atomic-free
+                # (index_add_ was the contention hot spot at high
+                # duplicates-per-group)
+                nrows = perm.shape[0]
+                seg_ends = (
+                    torch.cat(
+                        [
+                            first_idx[1:],
+                            torch.tensor(
+                                [nrows], dtype=torch.int64, device=device
+                            ),
+                        ]
+                    )
+                    - 1
+                )
+                acc_deltas = {}
+                for name, c in contribs.items():
+                    sc = c.index_select(0, perm)
+                    if sc.dtype == torch.float64:
+                        acc = torch.zeros(nseg, dtype=sc.dtype, device=device)
+                        acc.index_add_(0, seg, sc)
+                    else:
+                        cs = torch.cumsum(sc, 0)
+                        seg_tot = cs.index_select(0, seg_ends)
+                        acc = seg_tot.clone()
+                        acc[1:] -= seg_tot[:-1]
+                    acc_deltas[name] = acc
+                gfirst_rows = perm.index_select(0, first_idx)
+                gcols_first = {n: c.take(gfirst_rows) for n, c in gcols.items()}
 
         # 3b. multi-worker: all-to-all-v of pre-aggregated partials by key
         # shard (RCCL over xGMI; pact.rs:56 analog with combiner)

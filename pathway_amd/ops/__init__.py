@@ -281,3 +281,63 @@ def varlen_hash_se_gpu(
     if rc != 0:
         raise RuntimeError(f"pw_varlen_hash_se failed: hip error {rc}")
     return lo, hi
+
+
+def hash_agg_gpu(
+    k0: torch.Tensor, k1: torch.Tensor, contribs: Sequence[torch.Tensor]
+) -> tuple[torch.Tensor, torch.Tensor, list, torch.Tensor]:
+    """Sort-free additive pre-aggregation (pw_hash_agg): returns
+    (uk0, uk1, [accs...], rep_row_idx) for the distinct group keys of the
+    batch — UNSORTED and possibly with rare duplicate keys (publication
+    race); callers sort + consolidate the (small) result."""
+    lib = require_lib()
+    n = k0.shape[0]
+    dev = k0.device
+    nacc = len(contribs)
+    cap = 1 << max(10, (2 * n - 1).bit_length())
+    tk0 = torch.empty(cap, dtype=torch.int64, device=dev)
+    tk1 = torch.zeros(cap, dtype=torch.int64, device=dev)
+    rep = torch.zeros(cap, dtype=torch.int64, device=dev)
+    taccs = [torch.zeros(cap, dtype=torch.int64, device=dev) for _ in range(nacc)]
+    counter = torch.zeros(1, dtype=torch.int32, device=dev)
+    out_k0 = torch.empty(n, dtype=torch.int64, device=dev)
+    out_k1 = torch.empty(n, dtype=torch.int64, device=dev)
+    out_rep = torch.empty(n, dtype=torch.int64, device=dev)
+    out_accs = [torch.empty(n, dtype=torch.int64, device=dev) for _ in range(nacc)]
+    carr = (ctypes.c_void_p * max(nacc, 1))(
+        *[ctypes.c_void_p(c.contiguous().data_ptr()) for c in contribs]
+    )
+    tarr = (ctypes.c_void_p * max(nacc, 1))(
+        *[ctypes.c_void_p(c.data_ptr()) for c in taccs]
+    )
+    oarr = (ctypes.c_void_p * max(nacc, 1))(
+        *[ctypes.c_void_p(c.data_ptr()) for c in out_accs]
+    )
+    rc = lib.pw_hash_agg(
+        ctypes.c_void_p(k0.contiguous().data_ptr()),
+        ctypes.c_void_p(k1.contiguous().data_ptr()),
+        carr,
+        ctypes.c_int(nacc),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(tk0.data_ptr()),
+        ctypes.c_void_p(tk1.data_ptr()),
+        tarr,
+        ctypes.c_void_p(rep.data_ptr()),
+        ctypes.c_int64(cap),
+        ctypes.c_void_p(counter.data_ptr()),
+        ctypes.c_void_p(out_k0.data_ptr()),
+        ctypes.c_void_p(out_k1.data_ptr()),
+        oarr,
+        ctypes.c_void_p(out_rep.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_hash_agg failed: hip error {rc}")
+    # counter is the only host-visible size: one small D2H sync
+    m = int(counter.item())
+    return (
+        out_k0.narrow(0, 0, m),
+        out_k1.narrow(0, 0, m),
+        [a.narrow(0, 0, m) for a in out_accs],
+        out_rep.narrow(0, 0, m),
+    )

@@ -452,3 +452,120 @@ extern "C" int pw_varlen_hash_se(const void* bytes, const void* starts,
                      (uint64_t*)lo, (uint64_t*)hi);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------------- hash aggregation --
+//
+// Sort-free per-batch pre-aggregation for additive group-reduce
+// (count/int-sum): one pass over the batch accumulating into an open-
+// addressing hash table keyed by the 128-bit group key, instead of
+// radix-sorting the whole batch and running the segment-boundary chain
+// (rocprof: sort family + partition/nonzero were ~0.9 ms of a 2.9 ms
+// wordcount step at 4M rows).
+//
+// Claim protocol (no intra-wave spinning): CAS the first key word against
+// EMPTY; the winner then publishes word1 and the representative row.  A
+// reader that observes word0 == its key but an unpublished word1 simply
+// mismatches and claims the next slot — the same key may transiently own
+// two slots, which the (tiny) sorted consolidation of the collected
+// uniques merges afterwards.  EMPTY is a reserved 64-bit value; group keys
+// are xxh64 outputs, so a real key equals it with probability 2^-64.
+
+#define PW_HA_EMPTY 0x8000000000000000ULL
+
+struct AccPtrs {
+  const long long* p[8];
+};
+struct AccPtrsMut {
+  long long* p[8];
+};
+
+__global__ void k_hash_agg(const long long* k0, const long long* k1,
+                           AccPtrs contribs, int nacc, int64_t n,
+                           unsigned long long* tk0, long long* tk1,
+                           AccPtrsMut taccs, long long* rep, uint64_t mask) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    unsigned long long key0 = (unsigned long long)k0[i];
+    long long key1 = k1[i];
+    uint64_t slot = ((uint64_t)key0) & mask;
+    for (;;) {
+      unsigned long long prev =
+          atomicCAS(&tk0[slot], PW_HA_EMPTY, key0);
+      if (prev == PW_HA_EMPTY) {
+        // claimed: publish the rest of the key + a representative row
+        tk1[slot] = key1;
+        rep[slot] = (long long)i;
+        __threadfence();
+      } else if (prev != key0) {
+        slot = (slot + 1) & mask;
+        continue;
+      } else if (tk1[slot] != key1) {
+        // word0 matches but word1 is different OR not yet published:
+        // treat as mismatch and probe on (duplicate slots merge later)
+        slot = (slot + 1) & mask;
+        continue;
+      }
+      for (int a = 0; a < nacc; ++a)
+        atomicAdd((unsigned long long*)&taccs.p[a][slot],
+                  (unsigned long long)contribs.p[a][i]);
+      break;
+    }
+  }
+}
+
+__global__ void k_hash_agg_init(unsigned long long* tk0, uint64_t cap) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < cap;
+       i += (uint64_t)gridDim.x * blockDim.x)
+    tk0[i] = PW_HA_EMPTY;
+}
+
+__global__ void k_hash_agg_collect(const unsigned long long* tk0,
+                                   const long long* tk1, AccPtrs taccs,
+                                   const long long* rep, int nacc,
+                                   uint64_t cap, unsigned int* counter,
+                                   long long* out_k0, long long* out_k1,
+                                   AccPtrsMut out_accs, long long* out_rep) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < cap;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    unsigned long long k = tk0[i];
+    if (k == PW_HA_EMPTY) continue;
+    unsigned int idx = atomicAdd(counter, 1u);
+    out_k0[idx] = (long long)k;
+    out_k1[idx] = tk1[i];
+    out_rep[idx] = rep[i];
+    for (int a = 0; a < nacc; ++a) out_accs.p[a][idx] = taccs.p[a][i];
+  }
+}
+
+extern "C" int pw_hash_agg(const void* k0, const void* k1,
+                           const void** contrib_ptrs, int nacc, int64_t n,
+                           void* tk0, void* tk1, void** tacc_ptrs, void* rep,
+                           int64_t capacity, void* counter, void* out_k0,
+                           void* out_k1, void** out_acc_ptrs, void* out_rep,
+                           void* stream) {
+  if (nacc > 8) return 1;
+  hipStream_t s = (hipStream_t)stream;
+  AccPtrs cp;
+  AccPtrsMut tp, op;
+  for (int a = 0; a < nacc; ++a) {
+    cp.p[a] = (const long long*)contrib_ptrs[a];
+    tp.p[a] = (long long*)tacc_ptrs[a];
+    op.p[a] = (long long*)out_acc_ptrs[a];
+  }
+  AccPtrs tpc;
+  for (int a = 0; a < nacc; ++a) tpc.p[a] = tp.p[a];
+  uint64_t mask = (uint64_t)capacity - 1;
+  dim3 block(PW_BLOCK);
+  hipLaunchKernelGGL(k_hash_agg_init, dim3(pw_grid(capacity)), block, 0, s,
+                     (unsigned long long*)tk0, (uint64_t)capacity);
+  hipLaunchKernelGGL(k_hash_agg, dim3(pw_grid(n)), block, 0, s,
+                     (const long long*)k0, (const long long*)k1, cp, nacc, n,
+                     (unsigned long long*)tk0, (long long*)tk1, tp,
+                     (long long*)rep, mask);
+  hipLaunchKernelGGL(k_hash_agg_collect, dim3(pw_grid(capacity)), block, 0, s,
+                     (const unsigned long long*)tk0, (const long long*)tk1,
+                     tpc, (long long*)rep, nacc, (uint64_t)capacity,
+                     (unsigned int*)counter, (long long*)out_k0,
+                     (long long*)out_k1, op, (long long*)out_rep);
+  return (int)hipGetLastError();
+}
