@@ -365,7 +365,7 @@ out = sorted(zip(cols["w"].values(), cols["c"].values()))
 print(json.dumps(out))
 """
     outs = []
-    for env_extra in ({}, {"PW_NO_HASHAGG": "1"}):
+    for env_extra in ({"PW_HASHAGG": "1"}, {}):
         env = dict(_o.environ, **env_extra)
         res = subprocess.run(
             [sys.executable, "-c", code], capture_output=True, text=True, env=env
