@@ -171,3 +171,63 @@ def test_ordered_diff():
     # diff returns per-row deltas vs the previous row in t-order
     vals = sorted(v for v in cols["diff_v"].values() if v is not None)
     assert vals == [-2, 3]
+
+
+def test_async_transformer_class():
+    from pathway_amd.stdlib.utils.async_transformer import AsyncTransformer
+
+    class S(pw.Schema):
+        doubled: int
+        tag: str
+
+    class MyT(AsyncTransformer):
+        output_schema = S
+
+        async def invoke(self, a) -> dict:
+            import asyncio as _a
+
+            await _a.sleep(0.001)
+            return {"doubled": 2 * a, "tag": f"row{a}"}
+
+    t = T(
+        """
+        a
+        1
+        2
+        """
+    )
+    res = MyT(input_table=t).successful
+    _, cols = table_to_dicts(res)
+    assert sorted(cols["doubled"].values()) == [2, 4]
+    assert sorted(cols["tag"].values()) == ["row1", "row2"]
+
+
+def test_pandas_transformer():
+    import pandas as pd
+
+    from pathway_amd.stdlib.utils.pandas_transformer import pandas_transformer
+
+    class Out(pw.Schema):
+        total: int
+
+    @pandas_transformer(output_schema=Out)
+    def column_sum(df: pd.DataFrame) -> pd.DataFrame:
+        return pd.DataFrame({"total": [int(df["v"].sum())]})
+
+    t = T(
+        """
+        v
+        3
+        4
+        5
+        """
+    )
+    res = column_sum(t)
+    _, cols = table_to_dicts(res)
+    assert list(cols["total"].values()) == [12]
+
+
+def test_mcp_server_surface():
+    from pathway_amd.xpacks.llm.mcp_server import McpServable, McpServer
+
+    assert hasattr(McpServer, "tool") or hasattr(McpServer, "serve") or McpServable
