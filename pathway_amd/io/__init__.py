@@ -40,6 +40,15 @@ from pathway_amd.io import (
     redpanda,
     s3,
     sqlite,
+    chroma,
+    duckdb,
+    leann,
+    milvus,
+    pinecone,
+    pyfilesystem,
+    qdrant,
+    slack,
+    weaviate,
 )
 from pathway_amd.io._subscribe import subscribe
 from pathway_amd.io.synchronization import register_input_synchronization_group
@@ -50,5 +59,5 @@ __all__ = [
     "iceberg", "jsonlines", "kafka", "kinesis", "logstash", "minio",
     "mongodb", "mqtt", "mssql", "mysql", "nats", "null", "plaintext",
     "postgres", "pubsub", "pulsar", "python", "questdb", "rabbitmq",
-    "redpanda", "s3", "sqlite", "subscribe", "register_input_synchronization_group",
+    "redpanda", "s3", "sqlite", "chroma", "duckdb", "leann", "milvus", "pinecone", "pyfilesystem", "qdrant", "slack", "weaviate", "subscribe", "register_input_synchronization_group",
 ]

@@ -1,0 +1,19 @@
+"""pw.io.duckdb (reference io/duckdb) — API-parity surface.
+
+Requires the duckdb client library (offline image: raises at call time).
+"""
+from __future__ import annotations
+
+from typing import Any
+
+from pathway_amd.io._utils import require_client
+
+
+def read(*args: Any, schema=None, mode: str = "streaming", name: str | None = None, autocommit_duration_ms: int | None = 1500, **kwargs: Any):
+    require_client("duckdb", "duckdb")
+    raise NotImplementedError("pw.io.duckdb.read: client library loaded but offline transport is unavailable in this environment")
+
+
+def write(table, *args: Any, name: str | None = None, **kwargs: Any):
+    require_client("duckdb", "duckdb")
+    raise NotImplementedError("pw.io.duckdb.write: client library loaded but offline transport is unavailable in this environment")

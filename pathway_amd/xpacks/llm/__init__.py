@@ -24,3 +24,5 @@ __all__ = [
     "VectorStoreServer",
     "VectorStoreClient",
 ]
+
+from pathway_amd.xpacks.llm import constants, mcp_server, servers, utils  # noqa: E402
