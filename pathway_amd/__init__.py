@@ -151,6 +151,8 @@ window = temporal  # reference alias module for window constructors
 # temporal joins attached like the reference does
 Table.windowby = temporal.windowby
 Table.asof_join = temporal.asof_join
+Table.inactivity_detection = temporal.inactivity_detection
+Table.add_update_timestamp_utc = temporal.add_update_timestamp_utc
 Table.asof_join_left = temporal.asof_join_left
 Table.asof_join_right = temporal.asof_join_right
 Table.asof_join_outer = temporal.asof_join_outer

@@ -254,3 +254,72 @@ def test_python_connector_add_remove():
     keys, cols = pw.debug.table_to_dicts(res)
     assert list(cols["s"].values()) == [20]
     assert list(cols["c"].values()) == [1]
+
+
+def test_utc_now_and_update_timestamp():
+    import datetime
+
+    import pathway_amd.stdlib.temporal.time_utils as tu
+
+    tu.utc_now.cache_clear()
+    t0 = datetime.datetime.now(tz=datetime.timezone.utc)
+    ticks = tu.utc_now(
+        refresh_rate=datetime.timedelta(milliseconds=20), max_ticks=3
+    )
+    latest = ticks.reduce(ts=pw.reducers.latest(pw.this.timestamp_utc), n=pw.reducers.count())
+    _, cols = pw.debug.table_to_dicts(latest)
+    (n,) = cols["n"].values()
+    (ts,) = cols["ts"].values()
+    assert n == 3
+    assert abs((ts - t0).total_seconds()) < 30
+    tu.utc_now.cache_clear()
+
+
+def test_add_update_timestamp_utc():
+    import datetime
+
+    import pathway_amd.stdlib.temporal.time_utils as tu
+
+    tu.utc_now.cache_clear()
+    t = pw.debug.table_from_markdown(
+        """
+        a
+        1
+        2
+        """
+    )
+    res = t.add_update_timestamp_utc(
+        refresh_rate=datetime.timedelta(milliseconds=20), _max_ticks=2
+    )
+    _, cols = pw.debug.table_to_dicts(res)
+    assert sorted(cols["a"].values()) == [1, 2]
+    for ts in cols["updated_timestamp_utc"].values():
+        assert ts is not None
+    tu.utc_now.cache_clear()
+
+
+@pytest.mark.timeout(60)
+def test_inactivity_detection():
+    import datetime
+
+    import pathway_amd.stdlib.temporal.time_utils as tu
+
+    tu.utc_now.cache_clear()
+
+    class Activity(pw.io.python.ConnectorSubject):
+        def run(self):
+            self.next(v=1)
+            self.commit()
+            # then go silent; the utc_now ticks keep arriving
+            time.sleep(0.8)
+
+    t = pw.io.python.read(Activity(), schema=schema_from_types(v=int))
+    inactive = t.inactivity_detection(
+        allowed_inactivity_period=datetime.timedelta(milliseconds=150),
+        refresh_rate=datetime.timedelta(milliseconds=50),
+        _max_ticks=14,
+    )
+    _, cols = pw.debug.table_to_dicts(inactive)
+    stamps = list(cols["inactivity_timestamp_utc"].values())
+    assert len(stamps) >= 1  # the silence after the first row was flagged
+    tu.utc_now.cache_clear()
