@@ -73,3 +73,22 @@ class HybridIndex:
         )
 
     query = query_as_of_now
+
+
+class HybridIndexFactory:
+    """Builds a HybridIndex from several retriever factories (reference
+    hybrid_index.py:161)."""
+
+    def __init__(self, retriever_factories, k: int = 60):
+        self.retriever_factories = list(retriever_factories)
+        self.k = k
+
+    def build_index(self, data_column, data_table, metadata_column=None):
+        from pathway_amd.stdlib.indexing.data_index import DataIndex
+
+        inners = []
+        for f in self.retriever_factories:
+            # inner-index factories take (data_column, metadata_column)
+            built = f.build_index(data_column, metadata_column)
+            inners.append(built.inner if isinstance(built, DataIndex) else built)
+        return DataIndex(data_table, HybridIndex(inners, k=self.k))

@@ -153,3 +153,7 @@ class LshKnnFactory:
 
     def build_index(self, data_column, metadata_column=None, **kwargs) -> LshKnn:
         return LshKnn(data_column, metadata_column, self.distance_type, self.embedder)
+
+
+class DefaultKnnFactory(BruteForceKnnFactory):
+    """Good-default KNN factory (reference nearest_neighbors.py:574)."""

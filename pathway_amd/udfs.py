@@ -171,3 +171,23 @@ def async_options(**kwargs):
         return fun
 
     return decorator
+
+
+from pathway_amd.internals.common import UDF, udf  # noqa: E402
+
+
+def with_cache_strategy(fun, cache_strategy: CacheStrategy):
+    """Wrap a callable with a cache strategy (reference udfs surface)."""
+    return cache_strategy.wrap(fun)
+
+
+def with_retry_strategy(fun, retry_strategy: AsyncRetryStrategy):
+    """Wrap an async callable with a retry strategy."""
+
+    async def wrapped(*args, **kwargs):
+        async def call():
+            return await fun(*args, **kwargs)
+
+        return await retry_strategy.invoke(call)
+
+    return wrapped
