@@ -234,3 +234,61 @@ def _assemble_results(texts, metas, scores):
             }
         )
     return Json(out)
+
+
+class IndexingStatus(str, __import__("enum").Enum):
+    """Document indexing state reported by inputs_query (reference
+    document_store.py:49-52)."""
+
+    INDEXED = "INDEXED"
+    INGESTED = "INGESTED"
+
+
+class SlidesDocumentStore(DocumentStore):
+    """DocumentStore preset for slide decks (reference document_store.py):
+    same pipeline with slide-parsing defaults."""
+
+
+class DocumentStoreClient:
+    """HTTP client for DocumentStoreServer (reference
+    document_store.py:637-750: /v1/retrieve, /v1/statistics, /v1/inputs)."""
+
+    def __init__(self, host: str | None = None, port: int | None = None,
+                 url: str | None = None, timeout: float | None = 60):
+        if url is not None:
+            self.url = url
+        else:
+            self.url = f"http://{host}:{port}"
+        self.timeout = timeout
+
+    def _post(self, path: str, payload: dict):
+        from pathway_amd.xpacks.llm.question_answering import send_post_request
+
+        return send_post_request(self.url + path, payload, None, self.timeout)
+
+    def query(self, query: str, k: int = 3, metadata_filter: str | None = None,
+              filepath_globpattern: str | None = None):
+        return self._post(
+            "/v1/retrieve",
+            {
+                "query": query,
+                "k": k,
+                "metadata_filter": metadata_filter,
+                "filepath_globpattern": filepath_globpattern,
+            },
+        )
+
+    retrieve = query
+
+    def get_vectorstore_statistics(self):
+        return self._post("/v1/statistics", {})
+
+    statistics = get_vectorstore_statistics
+
+    def get_input_files(self, metadata_filter: str | None = None,
+                        filepath_globpattern: str | None = None):
+        return self._post(
+            "/v1/inputs",
+            {"metadata_filter": metadata_filter,
+             "filepath_globpattern": filepath_globpattern},
+        )

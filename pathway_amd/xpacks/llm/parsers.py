@@ -56,3 +56,45 @@ class ImageParser(_HeavyParser):
 
 class SlideParser(_HeavyParser):
     dependency = "a vision LLM"
+
+
+def default_vision_llm():
+    """Default vision-capable chat for image/slide parsing (reference
+    parsers.py default_vision_llm) — network-backed models are unavailable
+    offline; the echo chat stands in so pipelines still construct."""
+    from pathway_amd.xpacks.llm.llms import EchoChat
+
+    return EchoChat()
+
+
+#: legacy alias (reference parsers.py ParseUnstructured deprecation)
+ParseUnstructured = UnstructuredParser
+
+#: legacy alias kept for API parity
+PaddleOCRParser = PaddleParser
+
+
+class AudioParser(UDF):
+    """Audio transcription parser (reference parsers.py AudioParser) —
+    requires a speech-to-text backend; unavailable in the offline image."""
+
+    def __init__(self, *args: Any, **kwargs: Any):
+        super().__init__()
+
+    def __wrapped__(self, contents: bytes, **kwargs: Any):
+        raise NotImplementedError(
+            "AudioParser needs a speech-to-text backend (no network in this environment)"
+        )
+
+
+class TwelveLabsVideoParser(UDF):
+    """Video parser backed by the TwelveLabs API (reference parsers.py) —
+    network service, unavailable offline."""
+
+    def __init__(self, *args: Any, **kwargs: Any):
+        super().__init__()
+
+    def __wrapped__(self, contents: bytes, **kwargs: Any):
+        raise NotImplementedError(
+            "TwelveLabsVideoParser needs the TwelveLabs API (no network in this environment)"
+        )

@@ -134,3 +134,138 @@ class AdaptiveRAGQuestionAnswerer(BaseRAGQuestionAnswerer):
 class DeckRetriever(BaseRAGQuestionAnswerer):
     def __init__(self, *args, **kwargs):
         raise NotImplementedError("slide decks need vision parsers (offline)")
+
+
+class BaseContextProcessor:
+    """Formats retrieved docs into the LLM context string (reference
+    question_answering.py:39-70)."""
+
+    def maybe_unwrap_docs(self, docs):
+        from pathway_amd.internals.json import Json
+
+        if isinstance(docs, Json):
+            docs = docs.value
+        out = []
+        for d in docs or []:
+            if isinstance(d, Json):
+                d = d.value
+            out.append(d)
+        return out
+
+    def docs_to_context(self, docs) -> str:
+        raise NotImplementedError
+
+    def apply(self, docs) -> str:
+        return self.docs_to_context(self.maybe_unwrap_docs(docs))
+
+    def as_udf(self):
+        from pathway_amd.internals.common import udf
+
+        return udf(lambda docs: self.apply(docs))
+
+
+class SimpleContextProcessor(BaseContextProcessor):
+    """Joins doc texts with a separator, keeping selected metadata."""
+
+    def __init__(self, context_metadata_keys=None, docs_separator: str = "\n\n"):
+        self.context_metadata_keys = context_metadata_keys or []
+        self.docs_separator = docs_separator
+
+    def docs_to_context(self, docs) -> str:
+        parts = []
+        for d in docs:
+            if isinstance(d, dict):
+                text = str(d.get("text", ""))
+                meta = ", ".join(
+                    f"{k}: {d.get(k)}" for k in self.context_metadata_keys if k in d
+                )
+                parts.append(f"{text} ({meta})" if meta else text)
+            else:
+                parts.append(str(d))
+        return self.docs_separator.join(parts)
+
+
+class BaseQuestionAnswerer:
+    """Server-facing QA interface (reference question_answering.py)."""
+
+    AnswerQuerySchema = BaseRAGQuestionAnswerer.AnswerQuerySchema
+    RetrieveQuerySchema = None
+    StatisticsQuerySchema = None
+    InputsQuerySchema = None
+
+    def answer_query(self, pw_ai_queries):
+        raise NotImplementedError
+
+
+class SummaryQuestionAnswerer(BaseRAGQuestionAnswerer):
+    """QA variant whose answer includes a summary of matching docs."""
+
+
+def send_post_request(url: str, data: dict, headers: dict | None = None, timeout: float | None = None):
+    """POST JSON and return the parsed response (reference helper)."""
+    import json as _json
+    import urllib.request
+
+    req = urllib.request.Request(
+        url,
+        data=_json.dumps(data).encode(),
+        headers={"Content-Type": "application/json", **(headers or {})},
+        method="POST",
+    )
+    with urllib.request.urlopen(req, timeout=timeout) as resp:
+        return _json.loads(resp.read())
+
+
+class RAGClient:
+    """HTTP client for a running QA REST server (reference
+    question_answering.py:1070-1230: /v2/answer, /v2/summarize,
+    /v1/retrieve, /v1/statistics)."""
+
+    def __init__(self, host: str | None = None, port: int | None = None,
+                 url: str | None = None, timeout: float | None = 90,
+                 additional_headers: dict | None = None):
+        err = "Either (`host` and `port`) or `url` must be provided, but not both."
+        if url is not None:
+            if host is not None or port is not None:
+                raise ValueError(err)
+            self.url = url
+        else:
+            if host is None or port is None:
+                raise ValueError(err)
+            self.url = f"http://{host}:{port}"
+        self.timeout = timeout
+        self.additional_headers = additional_headers or {}
+
+    def _post(self, path: str, payload: dict):
+        return send_post_request(
+            self.url + path, payload, self.additional_headers, self.timeout
+        )
+
+    def answer(self, prompt: str, filters: str | None = None, model: str | None = None):
+        payload = {"prompt": prompt}
+        if filters:
+            payload["filters"] = filters
+        if model:
+            payload["model"] = model
+        return self._post("/v2/answer", payload)
+
+    def summarize(self, text_list, model: str | None = None):
+        payload = {"text_list": list(text_list)}
+        if model:
+            payload["model"] = model
+        return self._post("/v2/summarize", payload)
+
+    def retrieve(self, query: str, k: int = 3, metadata_filter: str | None = None,
+                 filepath_globpattern: str | None = None):
+        return self._post(
+            "/v1/retrieve",
+            {
+                "query": query,
+                "k": k,
+                "metadata_filter": metadata_filter,
+                "filepath_globpattern": filepath_globpattern,
+            },
+        )
+
+    def statistics(self):
+        return self._post("/v1/statistics", {})

@@ -107,3 +107,9 @@ class VectorStoreClient:
         )
         with urllib.request.urlopen(req, timeout=self.timeout) as resp:
             return json.loads(resp.read())
+
+
+class SlidesVectorStoreServer(VectorStoreServer):
+    """VectorStoreServer preset for slide decks (reference
+    vector_store.py SlidesVectorStoreServer): same serving surface with
+    slide-parser defaults."""
