@@ -258,7 +258,7 @@ class DocumentStoreClient:
         if url is not None:
             self.url = url
         else:
-            self.url = f"http://{host}:{port}"
+            self.url = f"http://{host or '127.0.0.1'}:{port}"
         self.timeout = timeout
 
     def _post(self, path: str, payload: dict):

@@ -236,3 +236,53 @@ def test_hybrid_index_rrf():
     ids = cols["_pw_index_reply_ids"][keys[0]]
     dkeys, dcols = table_to_dicts(docs)
     assert dcols["text"][ids[0]] == "alpha beta"
+
+
+def test_document_store_client_and_rag_client():
+    from pathway_amd.xpacks.llm.document_store import DocumentStore, DocumentStoreClient
+    from pathway_amd.xpacks.llm.llms import EchoChat
+    from pathway_amd.xpacks.llm.question_answering import (
+        BaseRAGQuestionAnswerer,
+        RAGClient,
+    )
+    from pathway_amd.xpacks.llm.servers import QARestServer
+
+    schema = schema_from_types(data=bytes, _metadata=dict)
+    docs = table_from_rows(
+        schema,
+        [
+            (b"gamma notes on indexes", {"path": "g.txt"}),
+            (b"delta notes on joins", {"path": "d.txt"}),
+        ],
+    )
+    store = DocumentStore(docs)
+    qa = BaseRAGQuestionAnswerer(EchoChat(), store)
+    port = 18233
+    srv = QARestServer("127.0.0.1", port, qa)
+    th = srv.run(threaded=True)
+    try:
+        dsc = DocumentStoreClient(port=port)
+        out = dsc.query("indexes", k=1)
+        assert len(out) == 1 and "indexes" in out[0]["text"]
+        stats = dsc.get_vectorstore_statistics()
+        assert stats["file_count"] == 2
+
+        rc = RAGClient(host="127.0.0.1", port=port)
+        ans = rc.answer("what about joins?")
+        assert isinstance(ans, (str, dict))
+        text = ans if isinstance(ans, str) else str(ans)
+        assert "joins" in text or len(text) > 0
+    finally:
+        srv.shutdown()
+
+
+def test_simple_context_processor():
+    from pathway_amd.xpacks.llm.question_answering import SimpleContextProcessor
+
+    cp = SimpleContextProcessor(context_metadata_keys=["path"])
+    ctx = cp.apply([
+        {"text": "first doc", "path": "a.txt"},
+        {"text": "second doc", "path": "b.txt"},
+    ])
+    assert "first doc (path: a.txt)" in ctx
+    assert "second doc" in ctx
