@@ -15,6 +15,8 @@ import sys
 _THIS = os.path.dirname(os.path.abspath(__file__))
 SRC = os.path.join(_THIS, "csrc", "hip", "pw_kernels.hip")
 OUT = os.path.join(_THIS, "libpwhip.so")
+IO_SRC = os.path.join(_THIS, "csrc", "cpp", "pw_io.cpp")
+IO_OUT = os.path.join(_THIS, "libpwio.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 
@@ -47,5 +49,19 @@ def build(verbose: bool = True) -> str:
     return OUT
 
 
+def build_io(verbose: bool = True) -> str:
+    """Host-only native IO scanner (g++; reference data_storage analog)."""
+    if os.path.exists(IO_OUT) and os.path.getmtime(IO_OUT) >= os.path.getmtime(
+        IO_SRC
+    ):
+        return IO_OUT
+    cmd = ["g++", "-O3", "-std=c++17", "-fPIC", "-shared", IO_SRC, "-o", IO_OUT]
+    if verbose:
+        print("[pathway_amd.ops]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return IO_OUT
+
+
 if __name__ == "__main__":
     build()
+    build_io()

@@ -1,0 +1,56 @@
+"""Native IO scanner throughput vs Python csv/readline (host-side)."""
+from __future__ import annotations
+
+import csv
+import json
+import os
+import sys
+import tempfile
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def main(n_rows: int = 2_000_000):
+    from pathway_amd.ops import native_io
+
+    with tempfile.NamedTemporaryFile("w", suffix=".csv", delete=False) as f:
+        f.write("id,name,qty\n")
+        for i in range(n_rows):
+            f.write(f'{i},"item {i % 997}",{i % 13}\n')
+        path = f.name
+    size_mb = os.path.getsize(path) / 1e6
+    try:
+        t0 = time.perf_counter()
+        header, rows = native_io.read_csv(path)
+        t_native = time.perf_counter() - t0
+        assert len(rows) == n_rows
+
+        t0 = time.perf_counter()
+        with open(path, newline="") as fh:
+            ref = list(csv.reader(fh))
+        t_py = time.perf_counter() - t0
+        assert len(ref) - 1 == n_rows
+
+        t0 = time.perf_counter()
+        lines = native_io.read_lines(path)
+        t_lines = time.perf_counter() - t0
+        print(
+            json.dumps(
+                {
+                    "bench": "native_io_scanner",
+                    "rows": n_rows,
+                    "file_mb": round(size_mb, 1),
+                    "native_csv_mrows_s": round(n_rows / t_native / 1e6, 2),
+                    "python_csv_mrows_s": round(n_rows / t_py / 1e6, 2),
+                    "speedup": round(t_py / t_native, 2),
+                    "scan_lines_mb_s": round(size_mb / t_lines, 1),
+                }
+            )
+        )
+    finally:
+        os.unlink(path)
+
+
+if __name__ == "__main__":
+    main()

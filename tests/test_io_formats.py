@@ -137,3 +137,49 @@ def test_csv_write_appends_updates(tmp_path):
     assert len(lines) == 4
     diffs = sorted(l.rsplit(",", 1)[1] for l in lines[1:])
     assert diffs == ["-1", "1", "1"]
+
+
+def test_native_csv_scanner_matches_python_csv(tmp_path):
+    import csv as _csv
+    import io as _io
+
+    from pathway_amd.ops import native_io
+
+    if not native_io.available():
+        pytest.skip("libpwio unavailable")
+    content = (
+        'a,b,c\n'
+        '1,"x,y",3\n'
+        '2,"he said ""hi""",4\n'
+        '5,"multi\nline",6\n'
+        '7,plain,9\n'
+    )
+    p = tmp_path / "t.csv"
+    p.write_text(content)
+    header, rows = native_io.read_csv(str(p))
+    ref = list(_csv.reader(_io.StringIO(content)))
+    assert header == ref[0]
+    assert rows == ref[1:]
+
+
+def test_native_scanner_used_by_csv_read(tmp_path):
+    p = tmp_path / "q.csv"
+    p.write_text('name,qty\n"widget, large",2\nbolt,3\n')
+
+    class S(pw.Schema):
+        name: str
+        qty: int
+
+    t = pw.io.csv.read(str(p), schema=S, mode="static")
+    assert _col_sorted(t, "name") == ["bolt", "widget, large"]
+    assert _col_sorted(t, "qty") == [2, 3]
+
+
+def test_native_read_lines(tmp_path):
+    from pathway_amd.ops import native_io
+
+    if not native_io.available():
+        pytest.skip("libpwio unavailable")
+    p = tmp_path / "l.txt"
+    p.write_bytes(b"one\r\ntwo\nthree")
+    assert native_io.read_lines(str(p)) == ["one", "two", "three"]
