@@ -311,13 +311,13 @@ def test_inactivity_detection():
             self.next(v=1)
             self.commit()
             # then go silent; the utc_now ticks keep arriving
-            time.sleep(0.8)
+            time.sleep(1.5)
 
     t = pw.io.python.read(Activity(), schema=schema_from_types(v=int))
     inactive = t.inactivity_detection(
-        allowed_inactivity_period=datetime.timedelta(milliseconds=150),
+        allowed_inactivity_period=datetime.timedelta(milliseconds=100),
         refresh_rate=datetime.timedelta(milliseconds=50),
-        _max_ticks=14,
+        _max_ticks=26,
     )
     _, cols = pw.debug.table_to_dicts(inactive)
     stamps = list(cols["inactivity_timestamp_utc"].values())
