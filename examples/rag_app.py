@@ -3,6 +3,11 @@
 POST /v1/retrieve {"query": "...", "k": 3}
 POST /v1/pw_ai_answer {"prompt": "..."}
 """
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import pathway_amd as pw
 from pathway_amd.xpacks.llm.document_store import DocumentStore
 from pathway_amd.xpacks.llm.embedders import SentenceTransformerEmbedder
