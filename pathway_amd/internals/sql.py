@@ -213,11 +213,67 @@ def sql(query: str, **tables: Table) -> Table:
     if gb:
         gcols = [c.strip().split(".")[-1] for c in gb.split(",")]
         grouped = t.groupby(*[ex.ColumnReference(t, c) for c in gcols])
-        result = grouped.reduce(**named)
+        having_expr = None
+        tmp_names: list[str] = []
         if m.group("having"):
-            result = result.filter(_parse_scalar(m.group("having"), result))
+            # aggregates inside HAVING are computed in the reduce under
+            # temporary names, filtered on, then dropped (SQL semantics)
+            having_expr = _parse_scalar(m.group("having"), t)
+            counter = [0]
+
+            def hoist(e):
+                if isinstance(e, ex.ReducerExpression):
+                    nm = f"_pw_having_{counter[0]}"
+                    counter[0] += 1
+                    named[nm] = e
+                    tmp_names.append(nm)
+                    return ex.ColumnReference(None, nm)
+                if not isinstance(e, ex.ColumnExpression):
+                    return e
+                new = object.__new__(type(e))
+                new.__dict__.update(e.__dict__)
+                for attr, val in list(e.__dict__.items()):
+                    if isinstance(val, ex.ColumnExpression):
+                        new.__dict__[attr] = hoist(val)
+                    elif isinstance(val, tuple) and any(
+                        isinstance(v, ex.ColumnExpression) for v in val
+                    ):
+                        new.__dict__[attr] = tuple(
+                            hoist(v) if isinstance(v, ex.ColumnExpression) else v
+                            for v in val
+                        )
+                return new
+
+            having_expr = hoist(having_expr)
+        result = grouped.reduce(**named)
+        if having_expr is not None:
+            rebound = _rebind(having_expr, result)
+            result = result.filter(rebound)
+            if tmp_names:
+                result = result.without(*tmp_names)
         return result
     has_agg = any(isinstance(e, ex.ReducerExpression) for e in named.values())
     if has_agg:
         return t.groupby().reduce(**named)
     return t.select(**named)
+
+
+def _rebind(e, table):
+    """Re-point bare column refs at `table` (post-reduce scope)."""
+    if isinstance(e, ex.ColumnReference):
+        return ex.ColumnReference(table, e.name)
+    if not isinstance(e, ex.ColumnExpression):
+        return e
+    new = object.__new__(type(e))
+    new.__dict__.update(e.__dict__)
+    for attr, val in list(e.__dict__.items()):
+        if isinstance(val, ex.ColumnExpression):
+            new.__dict__[attr] = _rebind(val, table)
+        elif isinstance(val, tuple) and any(
+            isinstance(v, ex.ColumnExpression) for v in val
+        ):
+            new.__dict__[attr] = tuple(
+                _rebind(v, table) if isinstance(v, ex.ColumnExpression) else v
+                for v in val
+            )
+    return new
