@@ -128,7 +128,57 @@ def _align(docs_table, query_table):
 
 
 class AdaptiveRAGQuestionAnswerer(BaseRAGQuestionAnswerer):
-    pass
+    """Adaptive-k answerer (reference question_answering.py
+    AdaptiveRAGQuestionAnswerer): retrieve a generous doc set once, then
+    prompt the LLM with a geometrically growing prefix until it answers."""
+
+    def __init__(self, llm, indexer, *, n_starting_documents: int = 2,
+                 factor: int = 2, max_iterations: int = 4,
+                 strict_prompt: bool = False, **kwargs):
+        kwargs.setdefault(
+            "search_topk", n_starting_documents * factor ** max(max_iterations - 1, 0)
+        )
+        super().__init__(llm, indexer, **kwargs)
+        self.n_starting_documents = n_starting_documents
+        self.factor = factor
+        self.max_iterations = max_iterations
+        self.strict_prompt = strict_prompt
+
+    def answer_query(self, pw_ai_queries):
+        store = self._store()
+        queries = pw_ai_queries.select(
+            query=this.prompt,
+            k=self.search_topk,
+            metadata_filter=this.filters
+            if "filters" in pw_ai_queries._dtypes
+            else None,
+            filepath_globpattern=None,
+        )
+        docs = store.retrieve_query(queries)
+        from pathway_amd.xpacks.llm.prompts import prompt_qa
+
+        llm = self.llm
+        n0, factor, iters = self.n_starting_documents, self.factor, self.max_iterations
+
+        def adaptive_answer(prompt, result):
+            rv = result.value if hasattr(result, "value") else result
+            texts = [r.get("text") for r in (rv or [])]
+            n = n0
+            for _ in range(iters):
+                resp = str(llm.__wrapped__(prompt_qa(prompt, texts[:n])))
+                if resp and "No information" not in resp:
+                    return resp
+                n *= factor
+            return "No information found."
+
+        return pw_ai_queries.select(
+            result=common.apply_with_type(
+                adaptive_answer, dt.STR, this.prompt, _align(docs, pw_ai_queries)
+            )
+        )
+
+    answer = answer_query
+    pw_ai_answer = answer_query
 
 
 class DeckRetriever(BaseRAGQuestionAnswerer):

@@ -286,3 +286,48 @@ def test_simple_context_processor():
     ])
     assert "first doc (path: a.txt)" in ctx
     assert "second doc" in ctx
+
+
+def test_adaptive_rag_answerer():
+    from pathway_amd.xpacks.llm.document_store import DocumentStore
+    from pathway_amd.xpacks.llm.question_answering import AdaptiveRAGQuestionAnswerer
+    from pathway_amd.xpacks.llm.llms import BaseChat
+
+    class CountingChat(BaseChat):
+        """Refuses until it sees >= 2 docs in the prompt."""
+
+        def __init__(self):
+            super().__init__()
+            self.calls = []
+
+        def __wrapped__(self, prompt: str) -> str:
+            ndocs = sum(prompt.count(w) for w in ("alpha", "beta", "gamma"))
+            self.calls.append(ndocs)
+            if ndocs >= 2:
+                return f"answer with {ndocs} docs"
+            return "No information found."
+
+    schema = schema_from_types(data=bytes, _metadata=dict)
+    docs = table_from_rows(
+        schema,
+        [
+            (b"alpha streaming engines", {"path": "a.txt"}),
+            (b"beta streaming windows", {"path": "b.txt"}),
+            (b"gamma streaming joins", {"path": "c.txt"}),
+        ],
+    )
+    store = DocumentStore(docs)
+    chat = CountingChat()
+    qa = AdaptiveRAGQuestionAnswerer(
+        chat, store, n_starting_documents=1, factor=2, max_iterations=3
+    )
+    queries = table_from_rows(
+        schema_from_types(prompt=str, filters=str, model=str,
+                          return_context_docs=bool),
+        [("streaming", None, None, None)],
+    )
+    res = qa.answer_query(queries)
+    _, cols = pw.debug.table_to_dicts(res)
+    (ans,) = cols["result"].values()
+    assert "answer with" in ans
+    assert len(chat.calls) >= 2  # escalated at least once
