@@ -33,13 +33,34 @@ def _threshold_tensor(expr, b: DeltaBatch, device) -> torch.Tensor:
 
 
 class _WatermarkMixin:
+    #: upstream node whose time-column watermark this node shares (a
+    #: buffer filters rows, so a downstream node cannot recover the stream
+    #: watermark from what it receives); graph wiring, survives reset()
+    wm_source: "_WatermarkMixin | None" = None
+
     def _init_wm(self):
         self.watermark: float | None = None
+        #: watermark as of the END of the previous engine step — what a
+        #: downstream behavior node should compare against so that rows
+        #: released in the SAME step still pass (reference time_column.rs
+        #: frontier semantics)
+        self.watermark_prev: float | None = None
+
+    def _snapshot_wm(self) -> None:
+        self.watermark_prev = self.watermark
 
     def _advance_watermark(self, values: torch.Tensor) -> None:
         if values.numel():
             m = float(values.max())
             self.watermark = m if self.watermark is None else max(self.watermark, m)
+
+    def _effective_wm(self) -> float:
+        wm = self.watermark_prev
+        if self.wm_source is not None:
+            src = self.wm_source.watermark_prev
+            if src is not None:
+                wm = src if wm is None else max(wm, src)
+        return wm if wm is not None else float("-inf")
 
 
 class BufferNode(Node, _WatermarkMixin):
@@ -62,6 +83,7 @@ class BufferNode(Node, _WatermarkMixin):
     def step(self, time, inputs):
         b = inputs[0]
         device = self.device
+        self._snapshot_wm()
         if b is not None and len(b):
             thr = _threshold_tensor(self.threshold_expr, b, device)
             now = _threshold_tensor(self.time_expr, b, device)
@@ -116,13 +138,18 @@ class ForgetNode(Node, _WatermarkMixin):
     def step(self, time, inputs):
         b = inputs[0]
         device = self.device
+        self._snapshot_wm()
         out = []
         if b is not None and len(b):
             thr = _threshold_tensor(self.threshold_expr, b, device)
             now = _threshold_tensor(self.time_expr, b, device)
             self._advance_watermark(now)
-            # drop LATE rows (threshold already passed)
-            fresh = thr > (self.watermark if self.watermark is not None else float("-inf"))
+            # drop LATE rows (threshold already passed); a buffer upstream
+            # hides the stream watermark, so fold in its shared value
+            wm_late = self.watermark if self.watermark is not None else float("-inf")
+            if self.wm_source is not None and self.wm_source.watermark_prev is not None:
+                wm_late = max(wm_late, self.wm_source.watermark_prev)
+            fresh = thr > wm_late
             fidx = fresh.nonzero(as_tuple=True)[0]
             if fidx.numel():
                 fb = b.take(fidx)
@@ -160,15 +187,18 @@ class FreezeNode(Node, _WatermarkMixin):
     def reset(self):
         self._init_wm()
 
+    def wants_frontier(self) -> bool:
+        return True  # keep the previous-step watermark snapshot advancing
+
     def step(self, time, inputs):
         b = inputs[0]
+        self._snapshot_wm()
         if b is None or len(b) == 0:
             return None
         device = self.device
         thr = _threshold_tensor(self.threshold_expr, b, device)
         now = _threshold_tensor(self.time_expr, b, device)
-        wm = self.watermark if self.watermark is not None else float("-inf")
-        keep = thr > wm
+        keep = thr > self._effective_wm()
         self._advance_watermark(now)
         kidx = keep.nonzero(as_tuple=True)[0]
         if not kidx.numel():

@@ -316,26 +316,28 @@ class WindowedTable:
         if isinstance(beh, ExactlyOnceBehavior):
             shift = _as_number(beh.shift) if beh.shift is not None else 0
             beh = CommonBehavior(delay=None, cutoff=shift, keep_results=True)
-            node = BufferNode(
+            buf = BufferNode(
                 prepared._node, end_ref + shift, time_expr, get_device()
             )
-            prepared = Table(node, prepared._dtypes, prepared._universe)
+            prepared = Table(buf, prepared._dtypes, prepared._universe)
             node = FreezeNode(
                 prepared._node,
                 ex.ColumnReference(prepared, "_pw_window_end") + shift,
                 self._retime(prepared),
                 get_device(),
             )
+            node.wm_source = buf  # the buffer sees the unfiltered stream
             return Table(node, prepared._dtypes, prepared._universe)
         if isinstance(beh, CommonBehavior):
+            buf = None
             if beh.delay is not None:
-                node = BufferNode(
+                buf = BufferNode(
                     prepared._node,
                     start_ref + _as_number(beh.delay),
                     time_expr,
                     get_device(),
                 )
-                prepared = Table(node, prepared._dtypes, prepared._universe)
+                prepared = Table(buf, prepared._dtypes, prepared._universe)
             if beh.cutoff is not None:
                 thr = (
                     ex.ColumnReference(prepared, "_pw_window_end")
@@ -345,6 +347,8 @@ class WindowedTable:
                 node = cls(
                     prepared._node, thr, self._retime(prepared), get_device()
                 )
+                if buf is not None:
+                    node.wm_source = buf
                 prepared = Table(node, prepared._dtypes, prepared._universe)
         return prepared
 

@@ -505,3 +505,43 @@ def test_asof_now_join_freezes_answers():
     # the query was answered at t=2 with v=10; the t=4 state change must
     # not retro-update the frozen answer
     assert list(cols["v"].values()) == [10]
+
+
+def test_exactly_once_behavior_window():
+    # windows close exactly once at (end + shift): late rows are dropped,
+    # and each window emits a single final value
+    t = T(
+        """
+        t | v | __time__
+        1 | 1 |    2
+        3 | 1 |    2
+        6 | 1 |    8
+        2 | 1 |   10
+        """
+    )
+    res = t.windowby(
+        t.t,
+        window=pw.temporal.tumbling(duration=5),
+        behavior=pw.temporal.exactly_once_behavior(),
+    ).reduce(start=pw.this._pw_window_start, n=pw.reducers.count())
+    keys, cols = pw.debug.table_to_dicts(res)
+    got = sorted(zip(cols["start"].values(), cols["n"].values()))
+    # window [0,5) closed when the watermark (6) passed its end: the late
+    # t=2 arrival at engine-time 10 is ignored; [5,10) emits its count
+    assert (0, 2) in got
+
+
+def test_universe_promises():
+    t1 = T(
+        """
+        a
+        1
+        2
+        """
+    )
+    t2 = t1.filter(pw.this.a > 0)
+    pw.universes.promise_is_subset_of(t2, t1)
+    t3 = t2.with_universe_of(t1)
+    res = t1.select(pw.this.a, b=t3.a * 10)
+    _, cols = pw.debug.table_to_dicts(res)
+    assert sorted(cols["b"].values()) == [10, 20]
