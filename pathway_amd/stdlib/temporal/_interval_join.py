@@ -26,8 +26,37 @@ def interval(lower_bound: Any, upper_bound: Any) -> Interval:
     return Interval(lower_bound, upper_bound)
 
 
+def _behave(table, time_expr, behavior, side):
+    """Apply a CommonBehavior cutoff to one join input."""
+    cutoff = getattr(behavior, "cutoff", None)
+    if cutoff is None:
+        return table
+    from pathway_amd.engine.nodes_temporal import ForgetNode, FreezeNode
+    from pathway_amd.internals.config import get_device
+
+    te = thisclass.substitute_this(
+        ex.wrap_expr(time_expr),
+        {thisclass.this: table, thisclass.left: table, thisclass.right: table},
+    )
+    import pandas as pd
+
+    c = cutoff
+    if isinstance(c, pd.Timedelta):
+        c = int(c.value)
+    cls = FreezeNode if getattr(behavior, "keep_results", True) else ForgetNode
+    node = cls(table._node, te + c, te, get_device())
+    return Table(node, table._dtypes, table._universe)
+
+
 class IntervalJoinResult:
-    def __init__(self, left, right, left_time, right_time, itv, on, mode, how=None):
+    def __init__(self, left, right, left_time, right_time, itv, on, mode,
+                 how=None, behavior=None):
+        if behavior is not None:
+            # cutoff behavior: freeze (keep_results) or forget each input
+            # past watermark - cutoff before joining (reference
+            # interval_join behavior semantics)
+            left = _behave(left, left_time, behavior, "left")
+            right = _behave(right, right_time, behavior, "right")
         self._left = left
         self._right = right
         self._left_time = left_time
@@ -186,20 +215,23 @@ def interval_join(
     behavior=None,
     how: Any = None,
 ) -> IntervalJoinResult:
-    return IntervalJoinResult(self, other, self_time, other_time, interval, on, "inner")
+    return IntervalJoinResult(
+        self, other, self_time, other_time, interval, on, "inner",
+        behavior=behavior,
+    )
 
 
-def interval_join_inner(self, other, self_time, other_time, interval, *on, **kw):
-    return IntervalJoinResult(self, other, self_time, other_time, interval, on, "inner")
+def interval_join_inner(self, other, self_time, other_time, interval, *on, behavior=None, **kw):
+    return IntervalJoinResult(self, other, self_time, other_time, interval, on, "inner", behavior=behavior)
 
 
-def interval_join_left(self, other, self_time, other_time, interval, *on, **kw):
-    return IntervalJoinResult(self, other, self_time, other_time, interval, on, "left")
+def interval_join_left(self, other, self_time, other_time, interval, *on, behavior=None, **kw):
+    return IntervalJoinResult(self, other, self_time, other_time, interval, on, "left", behavior=behavior)
 
 
-def interval_join_right(self, other, self_time, other_time, interval, *on, **kw):
-    return IntervalJoinResult(self, other, self_time, other_time, interval, on, "right")
+def interval_join_right(self, other, self_time, other_time, interval, *on, behavior=None, **kw):
+    return IntervalJoinResult(self, other, self_time, other_time, interval, on, "right", behavior=behavior)
 
 
-def interval_join_outer(self, other, self_time, other_time, interval, *on, **kw):
-    return IntervalJoinResult(self, other, self_time, other_time, interval, on, "outer")
+def interval_join_outer(self, other, self_time, other_time, interval, *on, behavior=None, **kw):
+    return IntervalJoinResult(self, other, self_time, other_time, interval, on, "outer", behavior=behavior)

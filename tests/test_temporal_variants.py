@@ -201,3 +201,49 @@ def test_asof_now_join_left_pads():
     )
     res = t.asof_now_join_left(r, t.a == r.a).select(t.a, r.b)
     assert _rows(res, "a", "b") == [(1, "x"), (2, None)]
+
+
+def test_windowby_with_instance():
+    t = T(
+        """
+        t | u | v
+        1 | a | 1
+        2 | a | 1
+        1 | b | 5
+        """
+    )
+    res = t.windowby(
+        t.t, window=pw.temporal.tumbling(duration=5), instance=t.u
+    ).reduce(u=pw.this._pw_instance, s=pw.reducers.sum(pw.this.v))
+    assert _rows(res, "u", "s") == [("a", 2), ("b", 5)]
+
+
+def test_interval_join_with_behavior_cutoff():
+    # late left rows past the cutoff are dropped from the join output
+    l = T(
+        """
+        t | a | __time__
+        1 | p |    2
+        10 | q |    4
+        1 | r |    8
+        """
+    )
+    r = T(
+        """
+        s | b | __time__
+        1 | u |    2
+        10 | w |    4
+        """
+    )
+    res = l.interval_join(
+        r,
+        l.t,
+        r.s,
+        pw.temporal.interval(-1, 1),
+        behavior=pw.temporal.common_behavior(cutoff=2),
+    ).select(pw.left.a, pw.right.b)
+    got = _rows(res, "a", "b")
+    # p joined u (on time), q joined w; the r-row (t=1) arrived when the
+    # watermark (10) had passed 1+cutoff → dropped
+    assert ("p", "u") in got and ("q", "w") in got
+    assert ("r", "u") not in got
