@@ -247,3 +247,47 @@ def test_interval_join_with_behavior_cutoff():
     # watermark (10) had passed 1+cutoff → dropped
     assert ("p", "u") in got and ("q", "w") in got
     assert ("r", "u") not in got
+
+
+def test_window_factory_params():
+    # origin shifts the tumbling grid
+    t = T(
+        """
+        t | v
+        1 | 1
+        4 | 1
+        """
+    )
+    r = t.windowby(
+        t.t, window=pw.temporal.tumbling(duration=5, origin=1)
+    ).reduce(start=pw.this._pw_window_start, n=pw.reducers.count())
+    assert _rows(r, "start", "n") == [(1, 2)]
+
+
+def test_sliding_ratio():
+    t = T(
+        """
+        t | v
+        0 | 1
+        3 | 1
+        """
+    )
+    r = t.windowby(
+        t.t, window=pw.temporal.sliding(hop=2, ratio=2)
+    ).reduce(start=pw.this._pw_window_start, n=pw.reducers.count())
+    assert _rows(r, "start", "n") == [(-2, 1), (0, 2), (2, 1)]
+
+
+def test_session_predicate():
+    t = T(
+        """
+        t | v
+        1 | 1
+        2 | 1
+        10 | 1
+        """
+    )
+    r = t.windowby(
+        t.t, window=pw.temporal.session(predicate=lambda a, b: abs(a - b) <= 3)
+    ).reduce(n=pw.reducers.count())
+    assert sorted(x[0] for x in _rows(r, "n")) == [1, 2]
