@@ -221,8 +221,22 @@ def _tensor_binop(ta: torch.Tensor, tb: torch.Tensor, sym: str) -> torch.Tensor 
         if sym == "/":
             return ta.to(torch.float64) / tb.to(torch.float64)
         if sym == "//":
+            # integer zero-divisors: CUDA does not trap like CPU does —
+            # fall back to the host path so those rows become ERROR values
+            if (
+                ta.dtype == torch.int64
+                and tb.dtype == torch.int64
+                and bool((tb == 0).any())
+            ):
+                return None
             return torch.div(ta, tb, rounding_mode="floor")
         if sym == "%":
+            if (
+                ta.dtype == torch.int64
+                and tb.dtype == torch.int64
+                and bool((tb == 0).any())
+            ):
+                return None
             return ta - torch.div(ta, tb, rounding_mode="floor") * tb
         if sym == "**":
             if ta.dtype == torch.int64 and tb.dtype == torch.int64:

@@ -111,9 +111,10 @@ def xxh64_words(words: list[torch.Tensor], seed: int) -> torch.Tensor:
 
 
 def hash128_words(words: list[torch.Tensor]) -> tuple[torch.Tensor, torch.Tensor]:
-    if words[0].is_cuda:
+    if words[0].is_cuda and len(words) <= 20:
         # GPU path MUST go through the fused HIP kernel (ops raises loudly
-        # if libpwhip.so is missing on a GPU host)
+        # if libpwhip.so is missing on a GPU host); >20 words (very wide
+        # rows) use the torch formulation, which also runs on device
         from pathway_amd import ops
 
         return ops.hash128_words_gpu([w.contiguous() for w in words])
