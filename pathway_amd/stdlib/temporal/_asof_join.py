@@ -85,6 +85,44 @@ class AsofJoinResult:
         }
         out_columns = list(out_exprs.keys())
 
+        # --- tensor-native incremental path (engine/nodes_asof.py) ---
+        import os
+
+        int_like = (dt.INT, dt.DATE_TIME_NAIVE, dt.DATE_TIME_UTC, dt.DURATION)
+        lt_dt = dt.unoptionalize(left._dtypes.get(lt_name, dt.ANY))
+        rt_dt = dt.unoptionalize(right._dtypes.get(rt_name, dt.ANY))
+        if (
+            not os.environ.get("PW_ASOF_HOST")
+            and direction in (Direction.BACKWARD, Direction.FORWARD)
+            and mode in ("inner", "left")
+            and lt_dt in int_like
+            and rt_dt in int_like
+        ):
+            from pathway_amd.engine.nodes_asof import AsofJoinNode
+
+            lmap = {n: src for n, (side, src) in out_exprs.items() if side == "l"}
+            rmap = {n: src for n, (side, src) in out_exprs.items() if side == "r"}
+            node = AsofJoinNode(
+                left._node,
+                right._node,
+                [ex.ColumnReference(left, ln) for (ln, _) in on_pairs],
+                [ex.ColumnReference(right, rn) for (_, rn) in on_pairs],
+                ex.ColumnReference(left, lt_name),
+                ex.ColumnReference(right, rt_name),
+                lmap,
+                rmap,
+                mode,
+                direction.value,
+                get_device(),
+                defaults=defaults,
+            )
+            out_dtypes = {}
+            for name, (side, src) in out_exprs.items():
+                srcd = (left if side == "l" else right)._dtypes.get(src, dt.ANY)
+                opt = side == "r" and mode == "left"
+                out_dtypes[name] = dt.Optional(srcd) if opt else srcd
+            return Table(node, out_dtypes, Universe())
+
         def fn(in_rows, in_keys):
             lrows, rrows = in_rows
             lkeys, rkeys = in_keys
