@@ -255,7 +255,39 @@ def node_state_save(node):
         }
     if type(node).__name__ == "_BM25IndexNode":
         return {"kind": "bm25", "state": node.state, "answers": node.answers}
+    from pathway_amd.engine.nodes_asof import AsofJoinNode
+
+    if isinstance(node, AsofJoinNode):
+        return {
+            "kind": "asof",
+            "l": _asof_side_to_portable(node.L),
+            "r": _asof_side_to_portable(node.R),
+        }
     return None
+
+
+def _asof_side_to_portable(S):
+    return {
+        "words": [tensor_to_portable(w) for w in S.words],
+        "weights": tensor_to_portable(S.weights),
+        "cols": {n: column_to_portable(c) for n, c in S.cols.items()}
+        if S.cols is not None
+        else None,
+    }
+
+
+def _asof_side_from_portable(d, device):
+    from pathway_amd.engine.nodes_asof import _AsofSide
+
+    S = _AsofSide(device)
+    S.words = [tensor_from_portable(w, device) for w in d["words"]]
+    S.weights = tensor_from_portable(d["weights"], device)
+    S.cols = (
+        {n: column_from_portable(c, device) for n, c in d["cols"].items()}
+        if d["cols"] is not None
+        else None
+    )
+    return S
 
 
 def node_state_load(node, state, device) -> None:
@@ -315,3 +347,6 @@ def node_state_load(node, state, device) -> None:
     elif kind == "bm25":
         node.state = state["state"]
         node.answers = state["answers"]
+    elif kind == "asof":
+        node.L = _asof_side_from_portable(state["l"], device)
+        node.R = _asof_side_from_portable(state["r"], device)

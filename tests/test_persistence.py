@@ -146,3 +146,37 @@ def test_operator_persisting_recovery(tmp_path):
     combined = squash_updates(rows_before + list(cap2.rows))
     got = sorted(tuple(v) for v in combined.values())
     assert got == [("a", 3), ("c", 1)]
+
+
+def test_operator_snapshot_asof_state(tmp_path):
+    """Asof-join arrangements survive an operator-snapshot save/load."""
+    import torch
+
+    from pathway_amd.engine.nodes_asof import AsofJoinNode, _AsofSide
+    from pathway_amd.persistence.operator_snapshot import (
+        node_state_load,
+        node_state_save,
+    )
+    from pathway_amd.engine.column import TensorColumn, PointerColumn
+    from pathway_amd.internals import dtype as dt
+
+    node = AsofJoinNode.__new__(AsofJoinNode)
+    node.L = _AsofSide("cpu")
+    node.R = _AsofSide("cpu")
+    words = [torch.tensor([1, 2]), torch.tensor([0, 0]),
+             torch.tensor([10, 20]), torch.tensor([5, 6]), torch.tensor([7, 8])]
+    cols = {
+        "v": TensorColumn(torch.tensor([100, 200]), dt.INT),
+        "__rowkey__": PointerColumn(torch.tensor([[1, 2], [3, 4]])),
+    }
+    node.R.merge(words, torch.tensor([1, 1]), cols)
+    st = node_state_save(node)
+    assert st["kind"] == "asof"
+
+    node2 = AsofJoinNode.__new__(AsofJoinNode)
+    node2.L = _AsofSide("cpu")
+    node2.R = _AsofSide("cpu")
+    node_state_load(node2, st, torch.device("cpu"))
+    assert len(node2.R) == 2
+    assert node2.R.cols["v"].tensor.tolist() == [100, 200]
+    assert node2.R.words[2].tolist() == [10, 20]
