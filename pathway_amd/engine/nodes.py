@@ -24,13 +24,13 @@ pathway_amd.ops when running on gfx950.
 
 from __future__ import annotations
 
-from typing import Any, Callable, Sequence
+from typing import Any, Sequence
 
 import numpy as np
 import torch
 
 from pathway_amd.internals import dtype as dt
-from pathway_amd.internals.api import MASK64, SHARD_MASK
+from pathway_amd.internals.api import SHARD_MASK
 
 MASK_SHARD = SHARD_MASK
 from pathway_amd.engine import hashing

@@ -37,9 +37,7 @@ from pathway_amd.engine.expression_eval import EvalContext, evaluate
 from pathway_amd.engine.nodes import (
     Node,
     batch_vhash,
-    columns_equal_mask,
     consolidate_batch,
-    null_column,
     _mask_missing,
 )
 from pathway_amd.engine.state import (

@@ -17,7 +17,7 @@ from typing import Any
 import torch
 
 from pathway_amd.engine.batch import DeltaBatch
-from pathway_amd.engine.column import Column, TensorColumn, concat_columns
+from pathway_amd.engine.column import Column, TensorColumn
 from pathway_amd.engine.expression_eval import EvalContext, evaluate
 from pathway_amd.engine.nodes import Node, consolidate_batch
 

@@ -13,8 +13,8 @@ Two families, mirroring the reference:
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Any, Callable
+from dataclasses import dataclass
+from typing import Callable
 
 import torch
 

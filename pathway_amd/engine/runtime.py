@@ -12,7 +12,6 @@ nothing; cross-worker progress is a tiny min-allreduce per step
 
 from __future__ import annotations
 
-import itertools
 from typing import Any, Callable, Iterable
 
 import torch

@@ -8,7 +8,7 @@ from __future__ import annotations
 import queue
 import threading
 import time as _time
-from typing import Any, Callable, Iterable
+from typing import Any, Callable
 
 import torch
 

@@ -12,7 +12,6 @@ from __future__ import annotations
 
 from typing import Any
 
-import torch
 
 from pathway_amd.engine.batch import DeltaBatch
 from pathway_amd.engine.nodes import InputNode, Node, consolidate_batch

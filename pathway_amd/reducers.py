@@ -5,7 +5,7 @@ from __future__ import annotations
 
 from typing import Any
 
-from pathway_amd.internals.expression import ColumnExpression, ReducerExpression
+from pathway_amd.internals.expression import ReducerExpression
 
 
 def count(*args: Any) -> ReducerExpression:

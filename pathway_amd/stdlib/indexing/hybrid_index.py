@@ -2,7 +2,6 @@
 (reference stdlib/indexing/hybrid_index.py:14)."""
 from __future__ import annotations
 
-from typing import Any
 
 import pathway_amd.internals.common as common
 from pathway_amd.internals import dtype as dt

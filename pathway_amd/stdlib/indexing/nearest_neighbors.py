@@ -9,7 +9,7 @@ GPU index (exact ≥ approximate recall; perf via GEMM).  LshKnn likewise.
 from __future__ import annotations
 
 import enum
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any
 
 from pathway_amd.internals import expression as ex

@@ -3,8 +3,6 @@ create_hmm_reducer builds a stateful reducer that tracks the most likely
 current state of an HMM over a stream of observations (Viterbi update)."""
 from __future__ import annotations
 
-import math
-from typing import Any, Callable
 
 
 def create_hmm_reducer(

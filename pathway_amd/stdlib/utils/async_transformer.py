@@ -7,7 +7,6 @@ src/engine/dataflow/async_transformer.rs:297).
 from __future__ import annotations
 
 import asyncio
-from typing import Any
 
 from pathway_amd.internals import expression as ex
 from pathway_amd.internals.schema import SchemaMetaclass
