@@ -545,6 +545,28 @@ class Table(TableLike):
         iname = iref.name if isinstance(iref, ex.ColumnReference) else None
         src = self
 
+        import os
+
+        int_like = (dt.INT, dt.DATE_TIME_NAIVE, dt.DATE_TIME_UTC, dt.DURATION)
+        if (
+            not os.environ.get("PW_SORT_HOST")
+            and dt.unoptionalize(self._dtypes.get(kname, dt.ANY)) in int_like
+        ):
+            # tensor-native differential prev/next (engine/nodes_sort.py)
+            from pathway_amd.engine.nodes_sort import SortPrevNextNode
+
+            node = SortPrevNextNode(
+                src._node,
+                kref,
+                iref if isinstance(iref, ex.ColumnReference) else None,
+                get_device(),
+            )
+            out_dtypes = {
+                "prev": dt.Optional(dt.POINTER),
+                "next": dt.Optional(dt.POINTER),
+            }
+            return Table(node, out_dtypes, self._universe)
+
         def fn(in_rows, in_keys):
             rows, keys = in_rows[0], in_keys[0]
             groups: dict = {}

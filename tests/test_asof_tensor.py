@@ -126,3 +126,49 @@ def test_asof_tensor_retraction_promotes_predecessor():
             assert rows == [("x", "old")], (host, rows)
         finally:
             os.environ.pop("PW_ASOF_HOST", None)
+
+
+def test_sort_tensor_matches_host_random_streams():
+    """SortPrevNextNode vs the host recompute path under random
+    insert/retract streams, including instances and duplicate keys."""
+    for seed in range(6):
+        rng = random.Random(7000 + seed)
+        hdr = "g | t | v | __time__ | __diff__"
+        live = []
+        lines = [hdr]
+        serial = 0
+        for step in range(4):
+            etime = 2 * (step + 1)
+            for _ in range(rng.randint(1, 4)):
+                serial += 1
+                g = rng.choice(["a", "b"])
+                t = rng.randint(0, 6)  # duplicates likely
+                lines.append(f"{g} | {t} | v{serial} | {etime} | 1")
+                live.append((g, t, serial))
+            if live and rng.random() < 0.6:
+                victim = rng.choice(live)
+                live.remove(victim)
+                lines.append(
+                    f"{victim[0]} | {victim[1]} | v{victim[2]} | {etime} | -1"
+                )
+        md = "\n".join(lines)
+
+        outs = []
+        for host in (False, True):
+            pw.internals.rungraph.G.clear()
+            if host:
+                os.environ["PW_SORT_HOST"] = "1"
+            try:
+                tbl = T(md, id_from=["g", "t", "v"])
+                sorted_t = tbl.sort(key=tbl.t, instance=tbl.g)
+                _, cols = table_to_dicts(sorted_t)
+                ids = sorted(cols["prev"].keys(), key=repr)
+                outs.append(
+                    [
+                        (repr(i), repr(cols["prev"][i]), repr(cols["next"][i]))
+                        for i in ids
+                    ]
+                )
+            finally:
+                os.environ.pop("PW_SORT_HOST", None)
+        assert outs[0] == outs[1], f"seed {seed}\n{md}\n{outs[0]}\nvs\n{outs[1]}"
