@@ -194,6 +194,33 @@ class WindowedTable:
         src = t.select(**cols)
         max_gap = _as_number(w.max_gap) if w.max_gap is not None else None
         predicate = w.predicate
+
+        import os
+
+        tdt = dt.unoptionalize(src._dtypes.get("_pw_t", dt.ANY))
+        if (
+            not os.environ.get("PW_SESSION_HOST")
+            and predicate is None
+            and isinstance(max_gap, int)
+            and tdt in (dt.INT, dt.DATE_TIME_NAIVE, dt.DATE_TIME_UTC, dt.DURATION)
+        ):
+            # tensor path: affected-instance recompute on device
+            from pathway_amd.engine.nodes_session import SessionAssignNode
+
+            node = SessionAssignNode(
+                src._node,
+                "_pw_t",
+                "_pw_instance" if self._instance is not None else None,
+                max_gap,
+                get_device(),
+            )
+            out_dtypes = dict(src._dtypes)
+            out_dtypes["_pw_window_start"] = tdt
+            out_dtypes["_pw_window_end"] = tdt
+            gb = ["_pw_window_start", "_pw_window_end"]
+            if self._instance is not None:
+                gb.append("_pw_instance")
+            return Table(node, out_dtypes, Universe()), gb
         out_columns = list(src._dtypes.keys()) + [
             "_pw_window_start",
             "_pw_window_end",

@@ -172,3 +172,60 @@ def test_sort_tensor_matches_host_random_streams():
             finally:
                 os.environ.pop("PW_SORT_HOST", None)
         assert outs[0] == outs[1], f"seed {seed}\n{md}\n{outs[0]}\nvs\n{outs[1]}"
+
+
+def test_session_tensor_matches_host_random_streams():
+    """SessionAssignNode vs host recompute path under random streams:
+    session merges AND splits via retraction."""
+    for seed in range(6):
+        rng = random.Random(9000 + seed)
+        hdr = "g | t | v | __time__ | __diff__"
+        live = []
+        lines = [hdr]
+        serial = 0
+        for step in range(4):
+            etime = 2 * (step + 1)
+            for _ in range(rng.randint(1, 4)):
+                serial += 1
+                g = rng.choice(["a", "b"])
+                t = rng.randint(0, 30)
+                lines.append(f"{g} | {t} | {serial} | {etime} | 1")
+                live.append((g, t, serial))
+            if live and rng.random() < 0.6:
+                victim = rng.choice(live)
+                live.remove(victim)
+                lines.append(
+                    f"{victim[0]} | {victim[1]} | {victim[2]} | {etime} | -1"
+                )
+        md = "\n".join(lines)
+
+        outs = []
+        for host in (False, True):
+            pw.internals.rungraph.G.clear()
+            if host:
+                os.environ["PW_SESSION_HOST"] = "1"
+            try:
+                tbl = T(md, id_from=["g", "t", "v"])
+                res = tbl.windowby(
+                    tbl.t,
+                    window=pw.temporal.session(max_gap=4),
+                    instance=tbl.g,
+                ).reduce(
+                    g=pw.this._pw_instance,
+                    start=pw.this._pw_window_start,
+                    end=pw.this._pw_window_end,
+                    n=pw.reducers.count(),
+                    s=pw.reducers.sum(pw.this.v),
+                )
+                _, cols = table_to_dicts(res)
+                ids = list(cols["g"].keys())
+                outs.append(
+                    sorted(
+                        (cols["g"][i], cols["start"][i], cols["end"][i],
+                         cols["n"][i], cols["s"][i])
+                        for i in ids
+                    )
+                )
+            finally:
+                os.environ.pop("PW_SESSION_HOST", None)
+        assert outs[0] == outs[1], f"seed {seed}\n{md}\n{outs[0]}\nvs\n{outs[1]}"
