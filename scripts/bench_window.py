@@ -25,6 +25,8 @@ def main():
     p.add_argument("--batch", type=int, default=2_000_000, help="events per step")
     p.add_argument("--hop", type=int, default=10)
     p.add_argument("--duration", type=int, default=40)
+    p.add_argument("--window", choices=["sliding", "session"], default="sliding")
+    p.add_argument("--gap", type=int, default=3, help="session max_gap")
     args = p.parse_args()
 
     use_cuda = torch.cuda.is_available()
@@ -77,9 +79,11 @@ def main():
 
     src = Events()
     t_in = Table(InputNode(src, device), {"t": dt.INT, "v": dt.INT}, Universe())
-    win = t_in.windowby(
-        t_in.t, window=pw.temporal.sliding(hop=args.hop, duration=args.duration)
-    ).reduce(
+    if args.window == "session":
+        wspec = pw.temporal.session(max_gap=args.gap)
+    else:
+        wspec = pw.temporal.sliding(hop=args.hop, duration=args.duration)
+    win = t_in.windowby(t_in.t, window=wspec).reduce(
         start=pw.this._pw_window_start,
         s=pw.reducers.sum(pw.this.v),
         n=pw.reducers.count(),
@@ -109,11 +113,12 @@ def main():
     print(
         json.dumps(
             {
-                "bench": "sliding_window_sum",
+                "bench": f"{args.window}_window_sum",
                 "events_per_s": ev / el,
                 "ms_per_step": el / args.steps * 1000,
                 "events_per_step": B,
-                "windows_per_event": args.duration // args.hop,
+                "windows_per_event": (args.duration // args.hop)
+                if args.window == "sliding" else 1,
                 "window_updates_emitted": emitted[0],
                 "device": str(device),
             }
