@@ -263,6 +263,11 @@ def node_state_save(node):
             "l": _asof_side_to_portable(node.L),
             "r": _asof_side_to_portable(node.R),
         }
+    from pathway_amd.engine.nodes_session import SessionAssignNode
+    from pathway_amd.engine.nodes_sort import SortPrevNextNode
+
+    if isinstance(node, (SessionAssignNode, SortPrevNextNode)):
+        return {"kind": "sorted_side", "s": _asof_side_to_portable(node.S)}
     return None
 
 
@@ -350,3 +355,5 @@ def node_state_load(node, state, device) -> None:
     elif kind == "asof":
         node.L = _asof_side_from_portable(state["l"], device)
         node.R = _asof_side_from_portable(state["r"], device)
+    elif kind == "sorted_side":
+        node.S = _asof_side_from_portable(state["s"], device)
