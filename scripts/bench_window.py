@@ -27,6 +27,8 @@ def main():
     p.add_argument("--duration", type=int, default=40)
     p.add_argument("--window", choices=["sliding", "session"], default="sliding")
     p.add_argument("--gap", type=int, default=3, help="session max_gap")
+    p.add_argument("--instances", type=int, default=0,
+                   help="session instances (0 = one global instance)")
     args = p.parse_args()
 
     use_cuda = torch.cuda.is_available()
@@ -63,7 +65,8 @@ def main():
             ts = (
                 torch.rand(B, generator=g, device=dev) * args.duration + base
             ).to(torch.int64)
-            vals = torch.arange(B, dtype=torch.int64, device=dev) % 1000
+            nmod = args.instances if args.instances else 1000
+            vals = torch.arange(B, dtype=torch.int64, device=dev) % nmod
             seq = torch.arange(base * B, base * B + B, dtype=torch.int64, device=dev)
             klo, khi = hashing.value_hash_words(seq, TAG_INT)
             keys = torch.stack([klo, khi], dim=1)
@@ -83,7 +86,10 @@ def main():
         wspec = pw.temporal.session(max_gap=args.gap)
     else:
         wspec = pw.temporal.sliding(hop=args.hop, duration=args.duration)
-    win = t_in.windowby(t_in.t, window=wspec).reduce(
+    wkw = {}
+    if args.instances:
+        wkw["instance"] = t_in.v  # v is already `arange % 1000`-style
+    win = t_in.windowby(t_in.t, window=wspec, **wkw).reduce(
         start=pw.this._pw_window_start,
         s=pw.reducers.sum(pw.this.v),
         n=pw.reducers.count(),
