@@ -118,9 +118,13 @@ def table_from_markdown(
         rows_raw.append([_parse_value(c) for c in cells[: len(header)]])
 
     special = {"__time__", "__diff__", "__shard__"}
-    data_names = [h for h in header if h not in special]
+    # an explicit `id` column sets row keys (reference table_from_markdown)
+    has_id_col = "id" in header and not id_from
+    data_names = [h for h in header if h not in special and h != "id"]
     t_idx = header.index("__time__") if "__time__" in header else None
     d_idx = header.index("__diff__") if "__diff__" in header else None
+    id_idx = header.index("id") if "id" in header else None
+    id_vals = []
 
     columns: dict[str, list[Any]] = {n: [] for n in data_names}
     times, diffs = [], []
@@ -128,6 +132,8 @@ def table_from_markdown(
         for h, v in zip(header, r):
             if h in columns:
                 columns[h].append(v)
+        if id_idx is not None:
+            id_vals.append(r[id_idx])
         times.append(r[t_idx] if t_idx is not None else 0)
         diffs.append(r[d_idx] if d_idx is not None else 1)
 
@@ -160,6 +166,8 @@ def table_from_markdown(
         values = [columns[nm][i] for nm in data_names]
         if id_from:
             lo, hi = hash_values([columns[c][i] for c in id_from])
+        elif has_id_col:
+            lo, hi = hash_values([id_vals[i]])
         else:
             key_seq = i + 1
             lo, hi = hash_values([key_seq])

@@ -1025,14 +1025,10 @@ class StreamToTableNode(Node):
     def __init__(self, input_node, upsert_name: str, device):
         super().__init__([input_node], device)
         self.upsert_name = upsert_name
-        from pathway_amd.engine.nodes_join import _SideStore
-
-        self.store = _SideStore(device)
+        self.state: dict | None = {}
 
     def reset(self):
-        from pathway_amd.engine.nodes_join import _SideStore
-
-        self.store = _SideStore(self.device)
+        self.state = {}
 
     def step(self, time, inputs):
         b = inputs[0]
@@ -1042,20 +1038,51 @@ class StreamToTableNode(Node):
         src = b.columns.get("_pw_source_id")
         if src is None:
             raise ValueError("stream_to_table needs a _pw_source_id column")
-        keys = src.pairs
-        up = ups.tensor if isinstance(ups, TensorColumn) else torch.tensor(
-            [bool(v) for v in ups.to_pylist()], device=self.device
+        # true upsert semantics (arrange_from_upsert, upsert.rs:346): a new
+        # value for a key retracts the stored one; a deletion retracts the
+        # STORED row even when the event carries different column values
+        names = [
+            n for n in b.columns if n not in (self.upsert_name, "_pw_source_id")
+        ]
+        if not hasattr(self, "state") or self.state is None:
+            self.state = {}
+        pairs = src.pairs.cpu().tolist()
+        upl = [bool(v) for v in ups.to_pylist()]
+        vals = {n: b.columns[n].to_pylist() for n in names}
+        out_keys: list[tuple[int, int]] = []
+        out_rows: list[tuple] = []
+        out_diffs: list[int] = []
+        for i, key in enumerate(map(tuple, pairs)):
+            stored = self.state.get(key)
+            if upl[i]:
+                row = tuple(vals[n][i] for n in names)
+                if stored is not None:
+                    if stored == row:
+                        continue
+                    out_keys.append(key)
+                    out_rows.append(stored)
+                    out_diffs.append(-1)
+                out_keys.append(key)
+                out_rows.append(row)
+                out_diffs.append(1)
+                self.state[key] = row
+            elif stored is not None:
+                out_keys.append(key)
+                out_rows.append(stored)
+                out_diffs.append(-1)
+                del self.state[key]
+        if not out_keys:
+            return None
+        keys = torch.tensor(out_keys, dtype=torch.int64, device=self.device).reshape(
+            len(out_keys), 2
         )
-        diffs = torch.where(
-            up.to(torch.bool),
-            torch.ones(len(b), dtype=torch.int64, device=self.device),
-            -torch.ones(len(b), dtype=torch.int64, device=self.device),
-        )
-        cols = {
-            n: c
-            for n, c in b.columns.items()
-            if n not in (self.upsert_name, "_pw_source_id")
-        }
+        diffs = torch.tensor(out_diffs, dtype=torch.int64, device=self.device)
+        cols = {}
+        for j, n in enumerate(names):
+            proto = b.columns[n]
+            cols[n] = column_from_pylist(
+                [r[j] for r in out_rows], proto.dtype, self.device
+            )
         return DeltaBatch(keys, cols, diffs, time)
 
 

@@ -773,6 +773,32 @@ class Table(TableLike):
         }
         return Table(node, dtypes, Universe())
 
+    def from_streams(self, deletion_stream: "Table") -> "Table":
+        """Reconstruct current state from an upsert stream (self) and a
+        deletion stream (reference table.py from_streams /
+        merge_streams_to_table): per id the latest event wins, in event
+        order; deletion columns need not match."""
+        payload = {n: t for n, t in self._dtypes.items() if n != "id"}
+        cols = {n: ex.ColumnReference(self, n) for n in payload}
+        u = self.select(
+            **cols,
+            _pw_source_id=ex.ColumnReference(self, "id"),
+            is_upsert=True,
+        )
+        pad = {
+            n: ex.DeclareTypeExpression(
+                ex.ColumnConstExpression(None), dt.Optional(dt.unoptionalize(t))
+            )
+            for n, t in payload.items()
+        }
+        d = deletion_stream.select(
+            **pad,
+            _pw_source_id=ex.ColumnReference(deletion_stream, "id"),
+            is_upsert=False,
+        )
+        merged = Table.concat_reindex(u, d)
+        return merged.stream_to_table(ex.ColumnReference(merged, "is_upsert"))
+
     def unpack_snapshots(self, *args, **kwargs) -> "Table":
         raise NotImplementedError("unpack_snapshots lands in the next round")
 

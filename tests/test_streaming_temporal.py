@@ -545,3 +545,25 @@ def test_universe_promises():
     res = t1.select(pw.this.a, b=t3.a * 10)
     _, cols = pw.debug.table_to_dicts(res)
     assert sorted(cols["b"].values()) == [10, 20]
+
+
+def test_from_streams_reconstructs_state():
+    # updates: id 1 gets two versions; id 2 one; deletions remove id 2
+    t1 = T(
+        """
+        id | pet | age | __time__
+         1 | cat |  3  |     2
+         2 | dog | 11  |     2
+         1 | cat |  4  |     4
+        """
+    )
+    t2 = T(
+        """
+        id | pet | __time__
+         2 | dog |     6
+        """
+    )
+    res = t1.from_streams(t2)
+    _, cols = pw.debug.table_to_dicts(res)
+    rows = sorted(zip(cols["pet"].values(), cols["age"].values()))
+    assert rows == [("cat", 4)]
