@@ -1086,6 +1086,41 @@ class StreamToTableNode(Node):
         return DeltaBatch(keys, cols, diffs, time)
 
 
+class UnpackSnapshotsNode(Node):
+    """unpack_snapshots (reference table.py:3056): whenever the input
+    changes, append the FULL current table state as fresh rows keyed by
+    (minibatch time, original row key).  The state lives as a sorted GPU
+    arrangement; emission is one gather + one fused hash."""
+
+    def __init__(self, input_node: Node, device):
+        super().__init__([input_node], device)
+        self.state: Arrangement | None = None
+
+    def reset(self):
+        self.state = None
+
+    def step(self, time, inputs):
+        b = consolidate_batch(inputs[0])
+        if b is None or len(b) == 0:
+            return None
+        if self.state is None:
+            self.state = Arrangement(self.device, dict(b.columns))
+        v0, v1 = batch_vhash(b)
+        self.state.merge(b.keys, (v0, v1), b.diffs, dict(b.columns))
+        m = len(self.state)
+        if m == 0:
+            return None
+        k0 = self.state.key_words[0]
+        k1 = self.state.key_words[1]
+        tconst = torch.full((m,), int(time), dtype=torch.int64, device=self.device)
+        o0, o1 = hashing.hash128_words([k0, k1, tconst])
+        keys = torch.stack([o0, o1], dim=1)
+        cols = dict(self.state.columns)
+        out = DeltaBatch(keys, cols, self.state.weights.clamp(min=1), time)
+        out.consolidated = True
+        return out
+
+
 class FreezeAnswersNode(Node):
     """as-of-now answer freezing: the first emission per key wins; later
     positive updates for an answered key are dropped (retractions of the

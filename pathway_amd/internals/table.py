@@ -799,8 +799,13 @@ class Table(TableLike):
         merged = Table.concat_reindex(u, d)
         return merged.stream_to_table(ex.ColumnReference(merged, "is_upsert"))
 
-    def unpack_snapshots(self, *args, **kwargs) -> "Table":
-        raise NotImplementedError("unpack_snapshots lands in the next round")
+    def unpack_snapshots(self) -> "Table":
+        """Change stream → snapshot stream (reference table.py:3056): every
+        changed minibatch appends the full current state as fresh rows."""
+        from pathway_amd.engine.nodes import UnpackSnapshotsNode
+
+        node = UnpackSnapshotsNode(self._node, get_device())
+        return Table(node, dict(self._dtypes), Universe())
 
     def remove_errors(self) -> "Table":
         return self  # Error rows are filtered at output in this implementation

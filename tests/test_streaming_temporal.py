@@ -567,3 +567,30 @@ def test_from_streams_reconstructs_state():
     _, cols = pw.debug.table_to_dicts(res)
     rows = sorted(zip(cols["pet"].values(), cols["age"].values()))
     assert rows == [("cat", 4)]
+
+
+def test_unpack_snapshots():
+    t = T(
+        """
+        v | __time__ | __diff__
+        a |    2     |    1
+        b |    2     |    1
+        b |    4     |   -1
+        c |    4     |    1
+        """,
+        id_from=["v"],
+    )
+    snaps = t.unpack_snapshots()
+    from pathway_amd.engine.runtime import Runtime
+    from pathway_amd.debug import reset_all
+
+    cap = snaps._capture()
+    rt = Runtime([cap])
+    reset_all(rt.nodes)
+    rt.run()
+    by_time = {}
+    for r in cap.rows:
+        by_time.setdefault(r.time, []).append((r.values[0], r.diff))
+    assert sorted(v for v, d in by_time[2]) == ["a", "b"]
+    assert sorted(v for v, d in by_time[4]) == ["a", "c"]
+    assert all(d == 1 for vs in by_time.values() for _, vs_d in [(0, 0)] for v, d in vs)
