@@ -100,3 +100,13 @@ def write(
     node = OutputNode(table._node, writer, get_device())
     G.add_sink(node)
     return node
+
+
+def simple_read(server: str, topic: str, *, format: str = "raw", **kwargs):
+    """One-call raw Kafka reader (reference io/kafka simple_read)."""
+    return read(
+        {"bootstrap.servers": server, "group.id": "pathway-simple", "auto.offset.reset": "beginning"},
+        topic=topic,
+        format=format,
+        **kwargs,
+    )

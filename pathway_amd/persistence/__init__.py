@@ -51,3 +51,9 @@ class Config:
     @classmethod
     def simple_config(cls, backend: Backend, **kwargs) -> "Config":
         return cls(backend=backend, **kwargs)
+
+
+def get_persistence_engine_config(config):
+    """Engine-facing view of a persistence Config (reference
+    persistence/__init__.py helper)."""
+    return config
