@@ -157,6 +157,22 @@ class AsofJoinNode(Node):
         cols["__rowkey__"] = PointerColumn(b.keys)
         return [k0, k1, t, v0, v1], b.diffs, cols
 
+    def _ident(self, S: _AsofSide, idx, valid, proto_t):
+        """(valid, matched time, matched vhash words) — the answer's
+        identity triple; safe on an empty state."""
+        z = torch.zeros_like(proto_t)
+        if len(S) == 0:
+            f = torch.zeros(
+                proto_t.shape[0], dtype=torch.bool, device=self.device
+            )
+            return (f, z, z.clone(), z.clone())
+        return (
+            valid,
+            torch.where(valid, S.words[2].index_select(0, idx), z),
+            torch.where(valid, S.words[3].index_select(0, idx), z),
+            torch.where(valid, S.words[4].index_select(0, idx), z),
+        )
+
     def _probe(self, S: _AsofSide, qk0, qk1, qt):
         """(idx, valid) of the matched state row per query."""
         m = len(S)
@@ -246,7 +262,7 @@ class AsofJoinNode(Node):
         for out_name, src in self.lmap.items():
             out[out_name] = lcols[src].take(keep)
         for out_name, src in self.rmap.items():
-            if R.cols is not None and src in R.cols:
+            if R.cols is not None and src in R.cols and len(R) > 0:
                 c = R.cols[src].take(ridx)
                 out[out_name] = _mask_missing(c, qvalid, device)
                 if out_name in self.defaults:
@@ -310,12 +326,7 @@ class AsofJoinNode(Node):
             a_k1 = self.L.words[1].index_select(0, aff)
             a_t = self.L.words[2].index_select(0, aff)
             old_idx, old_valid = self._probe(R_old, a_k0, a_k1, a_t)
-            old_ident = (
-                old_valid,
-                torch.where(old_valid, R_old.words[2].index_select(0, old_idx), torch.zeros_like(a_t)),
-                torch.where(old_valid, R_old.words[3].index_select(0, old_idx), torch.zeros_like(a_t)),
-                torch.where(old_valid, R_old.words[4].index_select(0, old_idx), torch.zeros_like(a_t)),
-            )
+            old_ident = self._ident(R_old, old_idx, old_valid, a_t)
         # 2. merge ΔR into the right state (R becomes NEW)
         if dR is not None:
             # snapshot old arrays: merge builds new tensors, old refs stay
@@ -329,12 +340,7 @@ class AsofJoinNode(Node):
         # 3. re-answer affected rows against NEW right; emit changes
         if aff is not None:
             new_idx, new_valid = self._probe(self.R, a_k0, a_k1, a_t)
-            new_ident = (
-                new_valid,
-                torch.where(new_valid, self.R.words[2].index_select(0, new_idx), torch.zeros_like(a_t)),
-                torch.where(new_valid, self.R.words[3].index_select(0, new_idx), torch.zeros_like(a_t)),
-                torch.where(new_valid, self.R.words[4].index_select(0, new_idx), torch.zeros_like(a_t)),
-            )
+            new_ident = self._ident(self.R, new_idx, new_valid, a_t)
             unchanged = torch.ones_like(old_valid)
             for o, n in zip(old_ident, new_ident):
                 unchanged = unchanged & (o == n)

@@ -84,7 +84,7 @@ def _gen_stream(rng, side, n_rows, n_times, max_t):
 @pytest.mark.parametrize("direction", [Direction.BACKWARD, Direction.FORWARD])
 @pytest.mark.parametrize("how", ["inner", "left"])
 def test_asof_tensor_matches_host_random_streams(direction, how):
-    for seed in range(6):
+    for seed in range(10):
         rng = random.Random(100 * seed + (direction == Direction.FORWARD))
         lmd = _gen_stream(rng, "l", 3, 4, 20)
         rmd = _gen_stream(rng, "r", 3, 4, 20)
@@ -131,7 +131,7 @@ def test_asof_tensor_retraction_promotes_predecessor():
 def test_sort_tensor_matches_host_random_streams():
     """SortPrevNextNode vs the host recompute path under random
     insert/retract streams, including instances and duplicate keys."""
-    for seed in range(6):
+    for seed in range(10):
         rng = random.Random(7000 + seed)
         hdr = "g | t | v | __time__ | __diff__"
         live = []
@@ -177,7 +177,7 @@ def test_sort_tensor_matches_host_random_streams():
 def test_session_tensor_matches_host_random_streams():
     """SessionAssignNode vs host recompute path under random streams:
     session merges AND splits via retraction."""
-    for seed in range(6):
+    for seed in range(10):
         rng = random.Random(9000 + seed)
         hdr = "g | t | v | __time__ | __diff__"
         live = []
