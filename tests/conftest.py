@@ -12,3 +12,13 @@ def _clear_graph():
     G.clear()
     yield
     G.clear()
+
+
+def free_port() -> int:
+    """An OS-assigned free TCP port (avoids fixed-port collisions when
+    several pytest processes run concurrently)."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]

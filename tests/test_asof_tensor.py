@@ -229,3 +229,45 @@ def test_session_tensor_matches_host_random_streams():
             finally:
                 os.environ.pop("PW_SESSION_HOST", None)
         assert outs[0] == outs[1], f"seed {seed}\n{md}\n{outs[0]}\nvs\n{outs[1]}"
+
+
+def test_interval_join_matches_bruteforce_random():
+    """interval_join (bucketed engine path) vs a brute-force pair scan."""
+    for seed in range(8):
+        rng = random.Random(5000 + seed)
+        lrows, rrows = [], []
+        llines = ["k | t | a | __time__ | __diff__"]
+        rlines = ["k | s | b | __time__ | __diff__"]
+        serial = 0
+        for step in range(3):
+            etime = 2 * (step + 1)
+            for _ in range(rng.randint(1, 4)):
+                serial += 1
+                k, t = rng.randint(1, 2), rng.randint(0, 12)
+                llines.append(f"{k} | {t} | L{serial} | {etime} | 1")
+                lrows.append((k, t, f"L{serial}"))
+            for _ in range(rng.randint(1, 4)):
+                serial += 1
+                k, s_ = rng.randint(1, 2), rng.randint(0, 12)
+                rlines.append(f"{k} | {s_} | R{serial} | {etime} | 1")
+                rrows.append((k, s_, f"R{serial}"))
+            if lrows and rng.random() < 0.5:
+                v = rng.choice(lrows)
+                lrows.remove(v)
+                llines.append(f"{v[0]} | {v[1]} | {v[2]} | {etime} | -1")
+        pw.internals.rungraph.G.clear()
+        l = T("\n".join(llines), id_from=["a"])
+        r = T("\n".join(rlines), id_from=["b"])
+        lo, hi = -2, 3
+        res = l.interval_join(
+            r, l.t, r.s, pw.temporal.interval(lo, hi), l.k == r.k
+        ).select(pw.left.a, pw.right.b)
+        _, cols = table_to_dicts(res)
+        got = sorted(zip(cols["a"].values(), cols["b"].values()))
+        expected = sorted(
+            (a, b)
+            for (lk, lt, a) in lrows
+            for (rk, rs, b) in rrows
+            if lk == rk and lo <= rs - lt <= hi
+        )
+        assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"

@@ -173,7 +173,9 @@ def test_document_store_rest_server():
         ],
     )
     store = DocumentStore(docs)
-    port = 18231
+    from tests.conftest import free_port
+
+    port = free_port()
     srv = DocumentStoreServer("127.0.0.1", port, store)
     th = srv.run(threaded=True)
     try:
@@ -257,7 +259,9 @@ def test_document_store_client_and_rag_client():
     )
     store = DocumentStore(docs)
     qa = BaseRAGQuestionAnswerer(EchoChat(), store)
-    port = 18233
+    from tests.conftest import free_port
+
+    port = free_port()
     srv = QARestServer("127.0.0.1", port, qa)
     th = srv.run(threaded=True)
     try:

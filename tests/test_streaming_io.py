@@ -55,7 +55,10 @@ def test_rest_connector_serving():
 
     from pathway_amd.internals.rungraph import G
 
-    webserver = pw.io.http.PathwayWebserver("127.0.0.1", 18341)
+    from tests.conftest import free_port
+
+    port = free_port()
+    webserver = pw.io.http.PathwayWebserver("127.0.0.1", port)
     schema = schema_from_types(query=str)
     queries, response_writer = pw.io.http.rest_connector(
         webserver=webserver, schema=schema, route="/ask", delete_completed_queries=True
@@ -65,7 +68,7 @@ def test_rest_connector_serving():
     rt = pw.run(_serve_in_background=True)
     try:
         req = urllib.request.Request(
-            "http://127.0.0.1:18341/ask",
+            f"http://127.0.0.1:{port}/ask",
             data=json.dumps({"query": "hello"}).encode(),
             headers={"Content-Type": "application/json"},
         )
