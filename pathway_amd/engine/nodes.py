@@ -55,6 +55,10 @@ from pathway_amd.engine.state import (
 
 import os as _os
 
+#: fused segmented-reduce pre-agg (pw_seg_reduce) — default ON on device;
+#: PW_NO_SEGRED=1 falls back to the torch boundary chain (A/B)
+_PW_NO_SEGRED = bool(_os.environ.get("PW_NO_SEGRED"))
+
 #: opt-in HIP hash-aggregation pre-agg (PW_HASHAGG=1). Measured on MI355X:
 #: the sort path wins at both 50k-distinct (2.95 vs 3.63 ms/step — atomic
 #: contention on hot counters) and 50M-distinct (sort savings vanish when
@@ -511,6 +515,33 @@ class GroupReduceNode(Node):
                         0, first_idx
                     )
                 gcols_first = {n: c.take(rep_first) for n, c in gcols.items()}
+                hashagg = True
+            if hashagg is None and (
+                torch.device(device).type == "cuda"
+                and not _PW_NO_SEGRED
+                and all(c.dtype == torch.int64 for c in contribs.values())
+            ):
+                # sort + fused segmented reduce (pw_seg_reduce): the whole
+                # run-starts/compaction/per-acc-sum chain in two kernels
+                from pathway_amd import ops
+
+                words = [gkeys[:, 0].contiguous(), gkeys[:, 1].contiguous()]
+                perm = lex_sort_words(words)
+                sw0 = words[0].index_select(0, perm)
+                sw1 = words[1].index_select(0, perm)
+                names = list(contribs)
+                sc_list = [
+                    contribs[nm].index_select(0, perm) for nm in names
+                ]
+                uk0, uk1, first_sorted, accs = ops.seg_reduce_gpu(
+                    sw0, sw1, sc_list
+                )
+                ukeys_w = [uk0, uk1]
+                acc_deltas = dict(zip(names, accs))
+                gfirst_rows = perm.index_select(0, first_sorted)
+                gcols_first = {
+                    n: c.take(gfirst_rows) for n, c in gcols.items()
+                }
                 hashagg = True
             if hashagg is None:
                 # sort path: ONE lex sort of the batch's group keys, then

@@ -341,3 +341,59 @@ def hash_agg_gpu(
         [a.narrow(0, 0, m) for a in out_accs],
         out_rep.narrow(0, 0, m),
     )
+
+
+def seg_reduce_gpu(
+    k0: torch.Tensor, k1: torch.Tensor, contribs: Sequence[torch.Tensor]
+) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor, list]:
+    """Fused segmented reduce over rows SORTED by (k0, k1): returns
+    (unique_k0, unique_k1, first_row_idx, [per-seg sums...]) — the whole
+    run-starts/compaction/segment-sum chain in two kernels."""
+    lib = require_lib()
+    n = k0.shape[0]
+    dev = k0.device
+    nacc = len(contribs)
+    nblocks = (n + 255) // 256
+    block_counts = torch.empty(nblocks, dtype=torch.int32, device=dev)
+    rc = lib.pw_seg_reduce_count(
+        ctypes.c_void_p(k0.contiguous().data_ptr()),
+        ctypes.c_void_p(k1.contiguous().data_ptr()),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(block_counts.data_ptr()),
+        ctypes.c_int64(nblocks),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_seg_reduce_count failed: hip error {rc}")
+    csum = torch.cumsum(block_counts.to(torch.int64), 0)
+    nseg = int(csum[-1].item())
+    bases = torch.cat(
+        [torch.zeros(1, dtype=torch.int64, device=dev), csum[:-1]]
+    )
+    out_k0 = torch.empty(nseg, dtype=torch.int64, device=dev)
+    out_k1 = torch.empty(nseg, dtype=torch.int64, device=dev)
+    out_first = torch.empty(nseg, dtype=torch.int64, device=dev)
+    out_accs = [torch.zeros(nseg, dtype=torch.int64, device=dev) for _ in range(nacc)]
+    carr = (ctypes.c_void_p * max(nacc, 1))(
+        *[ctypes.c_void_p(c.contiguous().data_ptr()) for c in contribs]
+    )
+    oarr = (ctypes.c_void_p * max(nacc, 1))(
+        *[ctypes.c_void_p(c.data_ptr()) for c in out_accs]
+    )
+    rc = lib.pw_seg_reduce_emit(
+        ctypes.c_void_p(k0.contiguous().data_ptr()),
+        ctypes.c_void_p(k1.contiguous().data_ptr()),
+        carr,
+        ctypes.c_int(nacc),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(bases.data_ptr()),
+        ctypes.c_void_p(out_k0.data_ptr()),
+        ctypes.c_void_p(out_k1.data_ptr()),
+        ctypes.c_void_p(out_first.data_ptr()),
+        oarr,
+        ctypes.c_int64(nblocks),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_seg_reduce_emit failed: hip error {rc}")
+    return out_k0, out_k1, out_first, out_accs

@@ -569,3 +569,111 @@ extern "C" int pw_hash_agg(const void* k0, const void* k1,
                      (long long*)out_k1, op, (long long*)out_rep);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------------ fused seg-reduce --
+//
+// Replaces the segment-boundary chain of the sorted pre-aggregation
+// (run-starts → cumsum → nonzero → gathers → per-acc prefix-sum diffs,
+// ~15 small launches on a 4M-row batch) with two kernels + one tiny
+// cumsum.  Input: rows sorted by the 2-word key; output: compacted unique
+// keys, per-segment int64 sums for each accumulator, and the first-row
+// index of every segment (the carried-column representative).
+//
+// K2 does a wave-level segmented sum (shfl_up over 64 lanes): one
+// atomicAdd per run-per-wave instead of one per row — on sorted data with
+// d duplicates per key that is a ~min(64, d)× reduction in atomics.
+
+__global__ void k_segred_count(const long long* k0, const long long* k1,
+                               int64_t n, int* block_counts) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int flag = 0;
+  if (i < n)
+    flag = (i == 0 || k0[i] != k0[i - 1] || k1[i] != k1[i - 1]) ? 1 : 0;
+  // block reduction of flags (blockDim = 256)
+  __shared__ int cnt[PW_BLOCK / 64];
+  int w = __popcll(__ballot(flag));
+  if ((threadIdx.x & 63) == 0) cnt[threadIdx.x >> 6] = w;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int total = 0;
+    for (int j = 0; j < PW_BLOCK / 64; ++j) total += cnt[j];
+    block_counts[blockIdx.x] = total;
+  }
+}
+
+__global__ void k_segred_emit(const long long* k0, const long long* k1,
+                              AccPtrs contribs, int nacc, int64_t n,
+                              const long long* block_bases,
+                              long long* out_k0, long long* out_k1,
+                              long long* out_first, AccPtrsMut out_accs) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int lane = threadIdx.x & 63;
+  int flag = 0;
+  if (i < n)
+    flag = (i == 0 || k0[i] != k0[i - 1] || k1[i] != k1[i - 1]) ? 1 : 0;
+  // block-wide exclusive scan of flags via per-wave ballot + LDS
+  uint64_t ball = __ballot(flag);
+  int wave = threadIdx.x >> 6;
+  __shared__ int wave_tot[PW_BLOCK / 64];
+  if (lane == 0) wave_tot[wave] = __popcll(ball);
+  __syncthreads();
+  int wave_base = 0;
+  for (int j = 0; j < wave; ++j) wave_base += wave_tot[j];
+  uint64_t below = ball & ((lane == 0) ? 0ULL : ((1ULL << lane) - 1));
+  int local_excl = wave_base + __popcll(below);
+  long long base = block_bases[blockIdx.x];
+  // global segment id of row i (segments are 0-based)
+  long long sid = base + local_excl + flag - 1;
+  if (i < n && flag) {
+    out_k0[sid] = k0[i];
+    out_k1[sid] = k1[i];
+    out_first[sid] = i;
+  }
+  // wave-segmented sums per accumulator
+  for (int a = 0; a < nacc; ++a) {
+    long long v = (i < n) ? contribs.p[a][i] : 0;
+    long long s = sid;
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+      long long vv = __shfl_up(v, off, 64);
+      long long ss = __shfl_up(s, off, 64);
+      if (lane >= off && ss == s) v += vv;
+    }
+    long long nxt = __shfl_down(s, 1, 64);
+    bool tail = (lane == 63) || (nxt != s);
+    if (i < n && tail)
+      atomicAdd((unsigned long long*)&out_accs.p[a][s],
+                (unsigned long long)v);
+  }
+}
+
+extern "C" int pw_seg_reduce_count(const void* k0, const void* k1, int64_t n,
+                                   void* block_counts, int64_t nblocks,
+                                   void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_segred_count, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
+                     0, s, (const long long*)k0, (const long long*)k1, n,
+                     (int*)block_counts);
+  return (int)hipGetLastError();
+}
+
+extern "C" int pw_seg_reduce_emit(const void* k0, const void* k1,
+                                  const void** contrib_ptrs, int nacc,
+                                  int64_t n, const void* block_bases,
+                                  void* out_k0, void* out_k1, void* out_first,
+                                  void** out_acc_ptrs, int64_t nblocks,
+                                  void* stream) {
+  if (nacc > 8) return 1;
+  AccPtrs cp;
+  AccPtrsMut op;
+  for (int a = 0; a < nacc; ++a) {
+    cp.p[a] = (const long long*)contrib_ptrs[a];
+    op.p[a] = (long long*)out_acc_ptrs[a];
+  }
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_segred_emit, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
+                     0, s, (const long long*)k0, (const long long*)k1, cp,
+                     nacc, n, (const long long*)block_bases, (long long*)out_k0,
+                     (long long*)out_k1, (long long*)out_first, op);
+  return (int)hipGetLastError();
+}

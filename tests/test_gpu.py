@@ -373,3 +373,40 @@ print(json.dumps(out))
         assert res.returncode == 0, res.stderr[-2000:]
         outs.append(res.stdout.strip().splitlines()[-1])
     assert json.loads(outs[0]) == json.loads(outs[1])
+
+
+@pytest.mark.gpu
+def test_seg_reduce_matches_reference():
+    """pw_seg_reduce (fused run-starts + compaction + wave-segmented sums)
+    vs plain torch on sorted keys, incl. negative weights."""
+    from pathway_amd import ops
+
+    dev = torch.device("cuda:0")
+    g = torch.Generator(device="cpu").manual_seed(11)
+    n, vocab = 2_000_000, 37_000
+    ids = torch.randint(0, vocab, (n,), generator=g).to(dev)
+    k0 = (ids * 0x9E3779B185EBCA87 + 7).sort().values
+    k1 = k0 * 3 + 1
+    w = torch.where(
+        torch.rand(n, generator=g).to(dev) < 0.3,
+        torch.tensor(-1, device=dev),
+        torch.tensor(2, device=dev),
+    ).to(torch.int64)
+    v = torch.randint(-9, 9, (n,), generator=g).to(dev)
+
+    uk0, uk1, first, (aw, av) = ops.seg_reduce_gpu(k0, k1, [w, v])
+
+    # torch reference
+    starts = torch.ones(n, dtype=torch.bool, device=dev)
+    starts[1:] = (k0[1:] != k0[:-1]) | (k1[1:] != k1[:-1])
+    seg = torch.cumsum(starts.to(torch.int64), 0) - 1
+    fidx = starts.nonzero(as_tuple=True)[0]
+    nseg = int(fidx.numel())
+    assert uk0.shape[0] == nseg
+    assert torch.equal(uk0, k0.index_select(0, fidx))
+    assert torch.equal(uk1, k1.index_select(0, fidx))
+    assert torch.equal(first, fidx)
+    for got, src in ((aw, w), (av, v)):
+        ref = torch.zeros(nseg, dtype=torch.int64, device=dev)
+        ref.index_add_(0, seg, src)
+        assert torch.equal(got, ref)
