@@ -622,7 +622,9 @@ __global__ void k_segred_emit(const long long* k0, const long long* k1,
   uint64_t below = ball & ((lane == 0) ? 0ULL : ((1ULL << lane) - 1));
   int local_excl = wave_base + __popcll(below);
   long long base = block_bases[blockIdx.x];
-  // global segment id of row i (segments are 0-based)
+  // global segment id of row i (segments are 0-based); padding lanes get
+  // the sentinel -1 so a real row followed by padding still detects its
+  // run tail (dropping this cost the last segment whenever n % 64 != 0)
   long long sid = base + local_excl + flag - 1;
   if (i < n && flag) {
     out_k0[sid] = k0[i];
@@ -630,9 +632,9 @@ __global__ void k_segred_emit(const long long* k0, const long long* k1,
     out_first[sid] = i;
   }
   // wave-segmented sums per accumulator
+  long long s = (i < n) ? sid : -1;
   for (int a = 0; a < nacc; ++a) {
     long long v = (i < n) ? contribs.p[a][i] : 0;
-    long long s = sid;
 #pragma unroll
     for (int off = 1; off < 64; off <<= 1) {
       long long vv = __shfl_up(v, off, 64);
@@ -641,7 +643,7 @@ __global__ void k_segred_emit(const long long* k0, const long long* k1,
     }
     long long nxt = __shfl_down(s, 1, 64);
     bool tail = (lane == 63) || (nxt != s);
-    if (i < n && tail)
+    if (s >= 0 && tail)
       atomicAdd((unsigned long long*)&out_accs.p[a][s],
                 (unsigned long long)v);
   }
