@@ -774,6 +774,30 @@ class GroupReduceNode(Node):
 
         perm2 = merge_sorted_select(self.add_keys, ukeys_w)
         all_words = [w.index_select(0, perm2) for w in all_words]
+        if (
+            torch.device(device).type == "cuda"
+            and not _PW_NO_SEGRED
+            and all(a.dtype == torch.int64 for a in all_accs.values())
+        ):
+            from pathway_amd import ops
+
+            names2 = list(all_accs)
+            sorted_accs = [
+                all_accs[nm].index_select(0, perm2) for nm in names2
+            ]
+            out_words, first_idx, out_accs = ops.seg_reduce_words_gpu(
+                all_words, sorted_accs
+            )
+            merged = dict(zip(names2, out_accs))
+            keep = merged["__w__"] != 0
+            kidx = keep.nonzero(as_tuple=True)[0]
+            self.add_keys = [w.index_select(0, kidx) for w in out_words]
+            self.add_accs = {
+                name: acc.index_select(0, kidx) for name, acc in merged.items()
+            }
+            rep = perm2.index_select(0, first_idx).index_select(0, kidx)
+            self.add_carried = {n: c.take(rep) for n, c in all_carried.items()}
+            return
         starts2 = rows_ne(all_words)
         seg2 = torch.cumsum(starts2.to(torch.int64), 0) - 1
         first_idx = starts2.nonzero(as_tuple=True)[0]

@@ -85,6 +85,26 @@ def consolidate_sorted(
     n = weights.shape[0]
     if n == 0:
         return words, weights, columns
+    import os as _os
+
+    if (
+        words[0].is_cuda
+        and len(words) <= 8
+        and not _os.environ.get("PW_NO_SEGRED")
+    ):
+        # fused path: run-starts + compaction + weight segment-sum in two
+        # kernels (pw_seg_reduce)
+        from pathway_amd import ops
+
+        out_words, first_idx, (wsum,) = ops.seg_reduce_words_gpu(
+            words, [weights]
+        )
+        keep = (wsum != 0).nonzero(as_tuple=True)[0]
+        kept_first = first_idx.index_select(0, keep)
+        kept_w = wsum.index_select(0, keep)
+        kept_words = [w.index_select(0, keep) for w in out_words]
+        out_cols = {name: c.take(kept_first) for name, c in columns.items()}
+        return kept_words, kept_w, out_cols
     starts = rows_ne(words)
     seg = torch.cumsum(starts.to(torch.int64), 0) - 1
     nseg = int(seg[-1]) + 1

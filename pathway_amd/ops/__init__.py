@@ -343,21 +343,27 @@ def hash_agg_gpu(
     )
 
 
-def seg_reduce_gpu(
-    k0: torch.Tensor, k1: torch.Tensor, contribs: Sequence[torch.Tensor]
-) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor, list]:
-    """Fused segmented reduce over rows SORTED by (k0, k1): returns
-    (unique_k0, unique_k1, first_row_idx, [per-seg sums...]) — the whole
-    run-starts/compaction/segment-sum chain in two kernels."""
+def seg_reduce_words_gpu(
+    words: Sequence[torch.Tensor], contribs: Sequence[torch.Tensor]
+) -> tuple[list, torch.Tensor, list]:
+    """Fused segmented reduce over rows SORTED by `words` (≤8 int64 word
+    columns): returns (unique word columns, first_row_idx, [per-seg int64
+    sums...]) — the whole run-starts/compaction/segment-sum chain in two
+    kernels + one tiny cumsum."""
     lib = require_lib()
-    n = k0.shape[0]
-    dev = k0.device
+    n = words[0].shape[0]
+    dev = words[0].device
+    nw = len(words)
     nacc = len(contribs)
+    words = [w.contiguous() for w in words]
     nblocks = (n + 255) // 256
     block_counts = torch.empty(nblocks, dtype=torch.int32, device=dev)
+    warr = (ctypes.c_void_p * nw)(
+        *[ctypes.c_void_p(w.data_ptr()) for w in words]
+    )
     rc = lib.pw_seg_reduce_count(
-        ctypes.c_void_p(k0.contiguous().data_ptr()),
-        ctypes.c_void_p(k1.contiguous().data_ptr()),
+        warr,
+        ctypes.c_int(nw),
         ctypes.c_int64(n),
         ctypes.c_void_p(block_counts.data_ptr()),
         ctypes.c_int64(nblocks),
@@ -370,8 +376,7 @@ def seg_reduce_gpu(
     bases = torch.cat(
         [torch.zeros(1, dtype=torch.int64, device=dev), csum[:-1]]
     )
-    out_k0 = torch.empty(nseg, dtype=torch.int64, device=dev)
-    out_k1 = torch.empty(nseg, dtype=torch.int64, device=dev)
+    out_words = [torch.empty(nseg, dtype=torch.int64, device=dev) for _ in range(nw)]
     out_first = torch.empty(nseg, dtype=torch.int64, device=dev)
     out_accs = [torch.zeros(nseg, dtype=torch.int64, device=dev) for _ in range(nacc)]
     carr = (ctypes.c_void_p * max(nacc, 1))(
@@ -380,15 +385,17 @@ def seg_reduce_gpu(
     oarr = (ctypes.c_void_p * max(nacc, 1))(
         *[ctypes.c_void_p(c.data_ptr()) for c in out_accs]
     )
+    owarr = (ctypes.c_void_p * nw)(
+        *[ctypes.c_void_p(w.data_ptr()) for w in out_words]
+    )
     rc = lib.pw_seg_reduce_emit(
-        ctypes.c_void_p(k0.contiguous().data_ptr()),
-        ctypes.c_void_p(k1.contiguous().data_ptr()),
+        warr,
+        ctypes.c_int(nw),
         carr,
         ctypes.c_int(nacc),
         ctypes.c_int64(n),
         ctypes.c_void_p(bases.data_ptr()),
-        ctypes.c_void_p(out_k0.data_ptr()),
-        ctypes.c_void_p(out_k1.data_ptr()),
+        owarr,
         ctypes.c_void_p(out_first.data_ptr()),
         oarr,
         ctypes.c_int64(nblocks),
@@ -396,4 +403,12 @@ def seg_reduce_gpu(
     )
     if rc != 0:
         raise RuntimeError(f"pw_seg_reduce_emit failed: hip error {rc}")
-    return out_k0, out_k1, out_first, out_accs
+    return out_words, out_first, out_accs
+
+
+def seg_reduce_gpu(
+    k0: torch.Tensor, k1: torch.Tensor, contribs: Sequence[torch.Tensor]
+) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor, list]:
+    """2-word convenience wrapper over seg_reduce_words_gpu."""
+    out_words, out_first, out_accs = seg_reduce_words_gpu([k0, k1], contribs)
+    return out_words[0], out_words[1], out_first, out_accs

@@ -583,16 +583,25 @@ extern "C" int pw_hash_agg(const void* k0, const void* k1,
 // atomicAdd per run-per-wave instead of one per row — on sorted data with
 // d duplicates per key that is a ~min(64, d)× reduction in atomics.
 
-__global__ void k_segred_count(const long long* k0, const long long* k1,
-                               int64_t n, int* block_counts) {
+struct SegWords {
+  const int64_t* p[8];
+};
+
+__global__ void k_segred_count(SegWords w, int nw, int64_t n,
+                               int* block_counts) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int flag = 0;
-  if (i < n)
-    flag = (i == 0 || k0[i] != k0[i - 1] || k1[i] != k1[i - 1]) ? 1 : 0;
+  if (i < n) {
+    if (i == 0) flag = 1;
+    else {
+      for (int j = 0; j < nw; ++j)
+        if (w.p[j][i] != w.p[j][i - 1]) { flag = 1; break; }
+    }
+  }
   // block reduction of flags (blockDim = 256)
   __shared__ int cnt[PW_BLOCK / 64];
-  int w = __popcll(__ballot(flag));
-  if ((threadIdx.x & 63) == 0) cnt[threadIdx.x >> 6] = w;
+  int wavecnt = __popcll(__ballot(flag));
+  if ((threadIdx.x & 63) == 0) cnt[threadIdx.x >> 6] = wavecnt;
   __syncthreads();
   if (threadIdx.x == 0) {
     int total = 0;
@@ -601,16 +610,20 @@ __global__ void k_segred_count(const long long* k0, const long long* k1,
   }
 }
 
-__global__ void k_segred_emit(const long long* k0, const long long* k1,
-                              AccPtrs contribs, int nacc, int64_t n,
-                              const long long* block_bases,
-                              long long* out_k0, long long* out_k1,
-                              long long* out_first, AccPtrsMut out_accs) {
+__global__ void k_segred_emit(SegWords w, int nw, AccPtrs contribs, int nacc,
+                              int64_t n, const long long* block_bases,
+                              AccPtrsMut out_words, long long* out_first,
+                              AccPtrsMut out_accs) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int lane = threadIdx.x & 63;
   int flag = 0;
-  if (i < n)
-    flag = (i == 0 || k0[i] != k0[i - 1] || k1[i] != k1[i - 1]) ? 1 : 0;
+  if (i < n) {
+    if (i == 0) flag = 1;
+    else {
+      for (int j = 0; j < nw; ++j)
+        if (w.p[j][i] != w.p[j][i - 1]) { flag = 1; break; }
+    }
+  }
   // block-wide exclusive scan of flags via per-wave ballot + LDS
   uint64_t ball = __ballot(flag);
   int wave = threadIdx.x >> 6;
@@ -627,8 +640,7 @@ __global__ void k_segred_emit(const long long* k0, const long long* k1,
   // run tail (dropping this cost the last segment whenever n % 64 != 0)
   long long sid = base + local_excl + flag - 1;
   if (i < n && flag) {
-    out_k0[sid] = k0[i];
-    out_k1[sid] = k1[i];
+    for (int j = 0; j < nw; ++j) out_words.p[j][sid] = (long long)w.p[j][i];
     out_first[sid] = i;
   }
   // wave-segmented sums per accumulator
@@ -649,33 +661,37 @@ __global__ void k_segred_emit(const long long* k0, const long long* k1,
   }
 }
 
-extern "C" int pw_seg_reduce_count(const void* k0, const void* k1, int64_t n,
+extern "C" int pw_seg_reduce_count(const void** word_ptrs, int nw, int64_t n,
                                    void* block_counts, int64_t nblocks,
                                    void* stream) {
+  if (nw > 8) return 1;
+  SegWords w;
+  for (int j = 0; j < nw; ++j) w.p[j] = (const int64_t*)word_ptrs[j];
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(k_segred_count, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
-                     0, s, (const long long*)k0, (const long long*)k1, n,
-                     (int*)block_counts);
+                     0, s, w, nw, n, (int*)block_counts);
   return (int)hipGetLastError();
 }
 
-extern "C" int pw_seg_reduce_emit(const void* k0, const void* k1,
+extern "C" int pw_seg_reduce_emit(const void** word_ptrs, int nw,
                                   const void** contrib_ptrs, int nacc,
                                   int64_t n, const void* block_bases,
-                                  void* out_k0, void* out_k1, void* out_first,
+                                  void** out_word_ptrs, void* out_first,
                                   void** out_acc_ptrs, int64_t nblocks,
                                   void* stream) {
-  if (nacc > 8) return 1;
+  if (nacc > 8 || nw > 8) return 1;
+  SegWords w;
+  for (int j = 0; j < nw; ++j) w.p[j] = (const int64_t*)word_ptrs[j];
   AccPtrs cp;
-  AccPtrsMut op;
+  AccPtrsMut op, ow;
   for (int a = 0; a < nacc; ++a) {
     cp.p[a] = (const long long*)contrib_ptrs[a];
     op.p[a] = (long long*)out_acc_ptrs[a];
   }
+  for (int j = 0; j < nw; ++j) ow.p[j] = (long long*)out_word_ptrs[j];
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(k_segred_emit, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
-                     0, s, (const long long*)k0, (const long long*)k1, cp,
-                     nacc, n, (const long long*)block_bases, (long long*)out_k0,
-                     (long long*)out_k1, (long long*)out_first, op);
+                     0, s, w, nw, cp, nacc, n, (const long long*)block_bases,
+                     ow, (long long*)out_first, op);
   return (int)hipGetLastError();
 }
