@@ -44,6 +44,13 @@ class SessionAssignNode(Node):
 
     def reset(self):
         self.S = _AsofSide(self.device)
+        self._xmeta = {}
+
+    def wants_frontier(self) -> bool:
+        from pathway_amd.parallel import get_comm
+
+        c = get_comm()
+        return c is not None and c.world > 1
 
     def _prep(self, b: DeltaBatch):
         ctx = EvalContext(b.columns, b.keys, self.device)
@@ -120,6 +127,23 @@ class SessionAssignNode(Node):
 
     def step(self, time, inputs):
         b = consolidate_batch(inputs[0])
+        from pathway_amd.parallel import get_comm
+
+        comm = get_comm()
+        if comm is not None and comm.world > 1:
+            # co-locate rows of the same instance: exchange by instance
+            # hash (the reference centralizes time-column ops on one
+            # worker; sharding by instance keeps the same correctness
+            # with balanced load)
+            from pathway_amd.engine.nodes_join import _exchange_side
+
+            jk = None
+            if b is not None and len(b):
+                w5, _, _ = self._prep(b)
+                jk = torch.stack(w5[:2], dim=1)
+            if not hasattr(self, "_xmeta"):
+                self._xmeta = {}
+            b, _ = _exchange_side(comm, b, jk, time, self._xmeta)
         if b is None or len(b) == 0:
             return None
         device = self.device

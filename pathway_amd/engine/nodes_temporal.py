@@ -54,6 +54,21 @@ class _WatermarkMixin:
             m = float(values.max())
             self.watermark = m if self.watermark is None else max(self.watermark, m)
 
+    def _sync_watermark(self) -> None:
+        """Multi-worker: the time-column watermark is GLOBAL (the reference
+        centralizes time-column ops on worker 1, time_column.rs:44-51; here
+        every rank allreduce-maxes instead so behavior nodes act locally on
+        co-sharded rows with an identical watermark)."""
+        from pathway_amd.parallel import get_comm
+
+        comm = get_comm()
+        if comm is None or comm.world <= 1:
+            return
+        local = self.watermark if self.watermark is not None else float("-inf")
+        g = comm.allreduce_max_scalar(local)
+        if g != float("-inf"):
+            self.watermark = g
+
     def _effective_wm(self) -> float:
         wm = self.watermark_prev
         if self.wm_source is not None:
@@ -89,6 +104,7 @@ class BufferNode(Node, _WatermarkMixin):
             now = _threshold_tensor(self.time_expr, b, device)
             self._advance_watermark(now)
             self.held.append((b, thr))
+        self._sync_watermark()
         if self.watermark is None or not self.held:
             return None
         out = []
@@ -155,6 +171,7 @@ class ForgetNode(Node, _WatermarkMixin):
                 fb = b.take(fidx)
                 out.append(fb)
                 self.live.append((fb, thr.index_select(0, fidx)))
+        self._sync_watermark()
         # forget rows that just fell behind the watermark
         if self.watermark is not None and self.live:
             still = []
@@ -193,6 +210,7 @@ class FreezeNode(Node, _WatermarkMixin):
     def step(self, time, inputs):
         b = inputs[0]
         self._snapshot_wm()
+        self._sync_watermark()
         if b is None or len(b) == 0:
             return None
         device = self.device

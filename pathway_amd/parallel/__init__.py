@@ -109,6 +109,12 @@ class Comm:
         dist.all_reduce(t)
         return float(t.item())
 
+    def allreduce_max_scalar(self, v: float) -> float:
+        """Global max (watermark sync for time-column operators)."""
+        t = torch.tensor([v], dtype=torch.float64, device=self._comm_device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        return float(t.item())
+
 
 _COMM: Comm | None = None
 

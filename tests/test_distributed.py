@@ -151,3 +151,158 @@ def test_distributed_join_gloo():
         assert p.exitcode == 0
     all_rows = sorted(results[0] + results[1])
     assert all_rows == [(1, 10, "x"), (2, 20, "y"), (3, 30, "z")]
+
+
+def _worker_session_sort(rank: int, world: int, port: int, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["PW_DEVICE"] = "cpu"
+    import pathway_amd as pw
+    import pathway_amd.parallel as par
+    from pathway_amd.debug import table_from_markdown as T
+
+    par.init(backend="gloo")
+    # rows of instance "a" live on BOTH ranks: the exchange must co-locate
+    if rank == 0:
+        t = T(
+            """
+            g | t | v
+            a | 1 | 1
+            a | 9 | 1
+            """
+        )
+    else:
+        t = T(
+            """
+            g | t | v
+            a | 2 | 1
+            b | 5 | 1
+            """
+        )
+    res = t.windowby(
+        t.t, window=pw.temporal.session(max_gap=3), instance=t.g
+    ).reduce(
+        g=pw.this._pw_instance,
+        start=pw.this._pw_window_start,
+        n=pw.reducers.count(),
+    )
+    cap = res._capture()
+    from pathway_amd.engine.runtime import Runtime
+    from pathway_amd.internals.rungraph import reset_all
+
+    rt = Runtime([cap], comm=par.get_comm())
+    reset_all(rt.nodes)
+    rt.run()
+    from pathway_amd.internals.api import squash_updates
+
+    rows = sorted(tuple(v) for v in squash_updates(cap.rows).values())
+    q.put((rank, rows))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_distributed_session_windows_gloo():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [
+        ctx.Process(target=_worker_session_sort, args=(r, 2, 29581, q))
+        for r in range(2)
+    ]
+    for p in ps:
+        p.start()
+    results = {}
+    for _ in ps:
+        rank, rows = q.get()
+        results[rank] = rows
+    for p in ps:
+        p.join(60)
+    # union across ranks == the correct global sessions:
+    # instance a: times 1,2 merge (gap<=3), 9 alone; instance b: 5 alone
+    union = sorted(results[0] + results[1])
+    assert union == [("a", 1, 2), ("a", 9, 1), ("b", 5, 1)]
+
+
+def _worker_asof(rank: int, world: int, port: int, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["PW_DEVICE"] = "cpu"
+    import pathway_amd as pw
+    import pathway_amd.parallel as par
+    from pathway_amd.debug import table_from_markdown as T
+
+    par.init(backend="gloo")
+    # left rows on rank 0, right quotes on rank 1: key co-location is the
+    # asof exchange's job
+    if rank == 0:
+        l = T(
+            """
+            k | t | a
+            1 | 5 | x
+            1 | 9 | y
+            """,
+            id_from=["k", "t", "a"],
+        )
+        r = T(
+            """
+            k | s | b
+            2 | 1 | zz
+            """,
+            id_from=["k", "s", "b"],
+        )
+    else:
+        l = T(
+            """
+            k | t | a
+            2 | 2 | w
+            """,
+            id_from=["k", "t", "a"],
+        )
+        r = T(
+            """
+            k | s | b
+            1 | 4 | u
+            1 | 8 | v
+            """,
+            id_from=["k", "s", "b"],
+        )
+    res = l.asof_join(r, l.t, r.s, l.k == r.k, how="inner").select(
+        pw.left.a, pw.right.b
+    )
+    cap = res._capture()
+    from pathway_amd.engine.runtime import Runtime
+    from pathway_amd.internals.rungraph import reset_all
+
+    rt = Runtime([cap], comm=par.get_comm())
+    reset_all(rt.nodes)
+    rt.run()
+    from pathway_amd.internals.api import squash_updates
+
+    rows = sorted(tuple(v) for v in squash_updates(cap.rows).values())
+    q.put((rank, rows))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_distributed_asof_gloo():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_worker_asof, args=(r, 2, 29582, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    results = {}
+    for _ in ps:
+        rank, rows = q.get()
+        results[rank] = rows
+    for p in ps:
+        p.join(60)
+    union = sorted(results[0] + results[1])
+    # k=1: t=5→u(4), t=9→v(8); k=2: t=2→zz(1)
+    assert union == [("w", "zz"), ("x", "u"), ("y", "v")]
