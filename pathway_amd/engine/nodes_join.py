@@ -401,9 +401,30 @@ class SemiJoinNode(Node):
     def reset(self) -> None:
         self.lstore = _SideStore(self.device)
         self.rcount = _SideStore(self.device)
+        self._xmeta_l = {}
+        self._xmeta_r = {}
+
+    def wants_frontier(self) -> bool:
+        from pathway_amd.parallel import get_comm
+
+        c = get_comm()
+        return c is not None and c.world > 1
 
     def step(self, time, inputs):
         bl, br = inputs
+        from pathway_amd.parallel import get_comm
+
+        comm = get_comm()
+        if comm is not None and comm.world > 1:
+            if not hasattr(self, "_xmeta_l"):
+                self._xmeta_l = {}
+                self._xmeta_r = {}
+            bl, _ = _exchange_side(
+                comm, bl, bl.keys if bl is not None else None, time, self._xmeta_l
+            )
+            br, _ = _exchange_side(
+                comm, br, br.keys if br is not None else None, time, self._xmeta_r
+            )
         device = self.device
         affected = []
         if bl is not None and len(bl):
@@ -462,9 +483,32 @@ class KeyedMergeNode(Node):
     def reset(self) -> None:
         self.lstore = _SideStore(self.device)
         self.rstore = _SideStore(self.device)
+        self._xmeta_l = {}
+        self._xmeta_r = {}
+
+    def wants_frontier(self) -> bool:
+        from pathway_amd.parallel import get_comm
+
+        c = get_comm()
+        return c is not None and c.world > 1
 
     def step(self, time, inputs):
         bl, br = inputs
+        from pathway_amd.parallel import get_comm
+
+        comm = get_comm()
+        if comm is not None and comm.world > 1:
+            # the two tables of an update may shard the same row key to
+            # different ranks at the source: co-locate by row key
+            if not hasattr(self, "_xmeta_l"):
+                self._xmeta_l = {}
+                self._xmeta_r = {}
+            bl, _ = _exchange_side(
+                comm, bl, bl.keys if bl is not None else None, time, self._xmeta_l
+            )
+            br, _ = _exchange_side(
+                comm, br, br.keys if br is not None else None, time, self._xmeta_r
+            )
         device = self.device
         affected = []
         if bl is not None and len(bl):

@@ -306,3 +306,75 @@ def test_distributed_asof_gloo():
     union = sorted(results[0] + results[1])
     # k=1: t=5→u(4), t=9→v(8); k=2: t=2→zz(1)
     assert union == [("w", "zz"), ("x", "u"), ("y", "v")]
+
+
+def _worker_update_rows(rank: int, world: int, port: int, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["PW_DEVICE"] = "cpu"
+    import pathway_amd as pw
+    import pathway_amd.parallel as par
+    from pathway_amd.debug import table_from_markdown as T
+
+    par.init(backend="gloo")
+    # base rows on rank 0, overrides for the SAME keys on rank 1
+    if rank == 0:
+        base = T(
+            """
+            id | v
+             1 | 10
+             2 | 20
+            """
+        )
+        upd = pw.Table.empty(v=int)
+    else:
+        base = T(
+            """
+            id | v
+             3 | 30
+            """
+        )
+        upd = T(
+            """
+            id | v
+             2 | 99
+            """
+        )
+    res = base.update_rows(upd)
+    cap = res._capture()
+    from pathway_amd.engine.runtime import Runtime
+    from pathway_amd.internals.rungraph import reset_all
+
+    rt = Runtime([cap], comm=par.get_comm())
+    reset_all(rt.nodes)
+    rt.run()
+    from pathway_amd.internals.api import squash_updates
+
+    rows = sorted(tuple(v) for v in squash_updates(cap.rows).values())
+    q.put((rank, rows))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_distributed_update_rows_gloo():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [
+        ctx.Process(target=_worker_update_rows, args=(r, 2, 29583, q))
+        for r in range(2)
+    ]
+    for p in ps:
+        p.start()
+    results = {}
+    for _ in ps:
+        rank, rows = q.get()
+        results[rank] = rows
+    for p in ps:
+        p.join(60)
+    union = sorted(results[0] + results[1])
+    # key 2's base (rank 0) must meet its override (rank 1)
+    assert union == [(10,), (30,), (99,)]
