@@ -31,11 +31,19 @@ class Comm:
         self.backend = dist.get_backend()
         self.rank = dist.get_rank()
         self.world = dist.get_world_size()
-        self.device = device or (
-            torch.device(f"cuda:{self.rank % max(torch.cuda.device_count(), 1)}")
-            if torch.cuda.is_available()
-            else torch.device("cpu")
-        )
+        if device is None:
+            # honor the engine device (PW_DEVICE) — a CPU-engine worker on
+            # a GPU host must not stage exchanges through cuda
+            from pathway_amd.internals.config import get_device
+
+            env_dev = get_device()
+            if torch.device(env_dev).type == "cuda" and torch.cuda.is_available():
+                device = torch.device(
+                    f"cuda:{self.rank % max(torch.cuda.device_count(), 1)}"
+                )
+            else:
+                device = torch.device(env_dev)
+        self.device = torch.device(device)
         self._comm_device = (
             self.device if str(self.backend) == "nccl" else torch.device("cpu")
         )
