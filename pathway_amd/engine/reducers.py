@@ -175,7 +175,11 @@ _multiset(
 _multiset(
     "tuple",
     None,
-    lambda rows: tuple(v for r in rows for v in [r[0][0]] * r[1]),
+    lambda rows: tuple(
+        v
+        for r in sorted(rows, key=lambda r: (r[0][1] is None, r[0][1]))
+        for v in [r[0][0]] * r[1]
+    ),
     lambda args_dt: dt.List(dt.unoptionalize(args_dt[0])),
     n_args=2,  # (value, order_key)
 )
@@ -208,7 +212,7 @@ _multiset(
 _multiset(
     "ndarray",
     None,
-    lambda rows: _ndarray_host(rows),
+    lambda rows: _ndarray_host(sorted(rows, key=lambda r: (r[0][1] is None, r[0][1]))),
     lambda args_dt: dt.Array(),
     n_args=2,
 )

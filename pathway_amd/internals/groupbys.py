@@ -72,7 +72,16 @@ class GroupedTable:
                 if rname in ("argmin", "argmax") and len(rargs) == 1:
                     rargs = [rargs[0], ex.ColumnReference(table, "id")]
                 elif rname in ("earliest", "latest", "tuple", "ndarray") and len(rargs) == 1:
-                    rargs = [rargs[0], _raw_ref("__seq__")]
+                    if self._sort_by is not None:
+                        # sort_by overrides arrival order as the ordering
+                        # key of order-sensitive reducers (reference
+                        # groupby(sort_by=...) semantics)
+                        order = thisclass.substitute_this(
+                            ex.wrap_expr(self._sort_by), {thisclass.this: table}
+                        )
+                    else:
+                        order = _raw_ref("__seq__")
+                    rargs = [rargs[0], order]
                 out_name = f"_pw_r{counter[0]}"
                 counter[0] += 1
                 rkw = dict(e._kwargs)

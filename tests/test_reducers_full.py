@@ -265,3 +265,27 @@ def test_sorted_tuple_skip_nones():
     _, cols = table_to_dicts(r)
     (st,) = cols["st"].values()
     assert st == (1, 2)
+
+
+def test_groupby_sort_by_orders_reducers():
+    t = T(
+        """
+        g | v | o
+        a | 10 | 3
+        a | 20 | 1
+        a | 30 | 2
+        """
+    )
+    r = t.groupby(pw.this.g, sort_by=pw.this.o).reduce(
+        pw.this.g,
+        first=pw.reducers.earliest(pw.this.v),
+        last=pw.reducers.latest(pw.this.v),
+        tup=pw.reducers.tuple(pw.this.v),
+    )
+    _, cols = table_to_dicts(r)
+    (first,) = cols["first"].values()
+    (last,) = cols["last"].values()
+    (tup,) = cols["tup"].values()
+    # ordered by o: 20 (o=1), 30 (o=2), 10 (o=3)
+    assert first == 20 and last == 10
+    assert tup == (20, 30, 10)
