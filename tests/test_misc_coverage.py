@@ -152,3 +152,58 @@ def test_bin_namespace_exists():
     )
     e = pw.this.a.bin
     assert e is not None
+
+
+def test_window_with_datetime_durations():
+    import pandas as pd
+
+    t = table_from_rows(schema_from_types(s=str, v=int), [
+        ("2023-01-01 00:00:10", 1),
+        ("2023-01-01 00:00:50", 2),
+        ("2023-01-01 00:02:10", 3),
+    ])
+    d = t.select(ts=pw.this.s.dt.strptime("%Y-%m-%d %H:%M:%S"), v=pw.this.v)
+    res = d.windowby(
+        d.ts, window=pw.temporal.tumbling(duration=pd.Timedelta(minutes=1))
+    ).reduce(n=pw.reducers.count(), s=pw.reducers.sum(pw.this.v))
+    _, cols = table_to_dicts(res)
+    got = sorted(zip(cols["n"].values(), cols["s"].values()))
+    assert got == [(1, 3), (2, 3)]
+
+
+def test_join_with_instances():
+    l = T(
+        """
+        k | inst | a
+        1 |  x   | p
+        1 |  y   | q
+        """
+    )
+    r = T(
+        """
+        k | inst | b
+        1 |  x   | u
+        1 |  y   | v
+        """
+    )
+    res = l.join(
+        r, l.k == r.k, left_instance=l.inst, right_instance=r.inst
+    ).select(pw.left.a, pw.right.b)
+    _, cols = table_to_dicts(res)
+    got = sorted(zip(cols["a"].values(), cols["b"].values()))
+    # instance colocation: only same-instance pairs match
+    assert got == [("p", "u"), ("q", "v")]
+
+
+def test_flatten_with_extra_columns():
+    t = T(
+        """
+        g
+        a
+        """
+    )
+    lt = t.select(pw.this.g, items=pw.make_tuple(1, 2, 3))
+    flat = lt.flatten(pw.this.items)
+    _, cols = table_to_dicts(flat)
+    assert sorted(cols["items"].values()) == [1, 2, 3]
+    assert set(cols["g"].values()) == {"a"}
