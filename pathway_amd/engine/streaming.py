@@ -148,6 +148,8 @@ class FilePollReader:
         self.refresh_interval = refresh_interval
         self.max_polls = max_polls
         self.seen: dict[str, float] = {}
+        #: rows emitted per path (for retraction on delete/modify)
+        self.emitted: dict[str, list] = {}
 
     def _list_files(self) -> list[str]:
         import os
@@ -163,6 +165,12 @@ class FilePollReader:
         return sorted(glob.glob(self.path)) or (
             [self.path] if os.path.exists(self.path) else []
         )
+
+    def _retract_file(self, path: str) -> None:
+        """Object deletion/modification: retract the rows previously
+        emitted for this path (reference metadata/file_like.rs tracking)."""
+        for row in self.emitted.pop(path, []):
+            self.source.emit(list(row), diff=-1)
 
     def _emit_file(self, path: str):
         import csv as _csv
@@ -186,6 +194,12 @@ class FilePollReader:
         names = self.schema.column_names() if self.schema else None
         from pathway_amd.ops import native_io
 
+        rows_out = self.emitted.setdefault(path, [])
+
+        def emit_row(row):
+            rows_out.append(list(row))
+            self.source.emit(row)
+
         if self.format == "plaintext":
             if native_io.available():
                 lines = native_io.read_lines(path)
@@ -196,13 +210,13 @@ class FilePollReader:
                 row = [line]
                 if self.with_metadata:
                     row.append(meta)
-                self.source.emit(row)
+                emit_row(row)
         elif self.format == "binary":
             with open(path, "rb") as fh:
                 row = [fh.read()]
             if self.with_metadata:
                 row.append(meta)
-            self.source.emit(row)
+            emit_row(row)
         elif self.format == "csv":
             if native_io.available():
                 header, recs = native_io.read_csv(path)
@@ -214,7 +228,7 @@ class FilePollReader:
                     ]
                     if self.with_metadata:
                         row.append(meta)
-                    self.source.emit(row)
+                    emit_row(row)
             else:
                 with open(path, newline="") as fh:
                     reader = _csv.DictReader(fh)
@@ -222,7 +236,7 @@ class FilePollReader:
                         row = [_convert(rec.get(n), self.schema, n) for n in names]
                         if self.with_metadata:
                             row.append(meta)
-                        self.source.emit(row)
+                        emit_row(row)
         elif self.format in ("json", "jsonlines"):
             with open(path) as fh:
                 for line in fh:
@@ -233,21 +247,29 @@ class FilePollReader:
                     row = [rec.get(n) for n in names]
                     if self.with_metadata:
                         row.append(meta)
-                    self.source.emit(row)
+                    emit_row(row)
 
     def run(self):
         import os
 
         polls = 0
         while True:
+            current = set()
             for f in self._list_files():
                 try:
                     mtime = os.path.getmtime(f)
                 except OSError:
                     continue
+                current.add(f)
                 if self.seen.get(f) != mtime:
+                    if f in self.seen:
+                        # modified: retract the previous version first
+                        self._retract_file(f)
                     self.seen[f] = mtime
                     self._emit_file(f)
+            for gone in [p for p in self.seen if p not in current]:
+                del self.seen[gone]
+                self._retract_file(gone)
             polls += 1
             if self.mode == "static" or (
                 self.max_polls is not None and polls >= self.max_polls

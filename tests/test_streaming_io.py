@@ -334,3 +334,28 @@ def test_inactivity_detection():
     stamps = list(cols["inactivity_timestamp_utc"].values())
     assert len(stamps) >= 1  # the silence after the first row was flagged
     tu.utc_now.cache_clear()
+
+
+def test_fs_streaming_retracts_deleted_files(tmp_path):
+    d = tmp_path / "data"
+    d.mkdir()
+    (d / "a.txt").write_text("keepme\n")
+    (d / "b.txt").write_text("dropme\n")
+
+    t = pw.io.fs.read(
+        str(d), format="plaintext", mode="streaming", refresh_interval=0.05,
+        _max_polls=30,
+    )
+    res = t.groupby().reduce(n=pw.reducers.count())
+
+    import threading
+
+    def deleter():
+        time.sleep(0.4)
+        (d / "b.txt").unlink()
+
+    th = threading.Thread(target=deleter, daemon=True)
+    th.start()
+    _, cols = pw.debug.table_to_dicts(res)
+    # after the deletion retraction only a.txt's line remains
+    assert list(cols["n"].values()) == [1]
