@@ -207,3 +207,60 @@ def test_flatten_with_extra_columns():
     _, cols = table_to_dicts(flat)
     assert sorted(cols["items"].values()) == [1, 2, 3]
     assert set(cols["g"].values()) == {"a"}
+
+
+def test_gated_connectors_raise_helpfully():
+    with pytest.raises(Exception) as e:
+        pw.io.kafka.read({"bootstrap.servers": "localhost:9092"}, topic="t",
+                         format="raw", schema=schema_from_types(data=bytes))
+    assert "confluent" in str(e.value).lower() or "kafka" in str(e.value).lower()
+    with pytest.raises(Exception) as e2:
+        pw.io.elasticsearch.write(None, "http://localhost:9200")
+    assert "elasticsearch" in str(e2.value).lower()
+
+
+def test_sharepoint_surface_importable():
+    from pathway_amd.xpacks.connectors import sharepoint
+
+    assert hasattr(sharepoint, "read")
+
+
+def test_schema_from_csv(tmp_path):
+    from pathway_amd.internals.schema import schema_from_csv
+
+    p = tmp_path / "s.csv"
+    p.write_text("name,qty,price\nwidget,3,1.5\n")
+    S = schema_from_csv(str(p))
+    from pathway_amd.internals import dtype as dt
+
+    assert S.__columns__["name"].dtype == dt.STR
+    assert S.__columns__["qty"].dtype == dt.INT
+    assert S.__columns__["price"].dtype == dt.FLOAT
+
+
+def test_fs_static_with_metadata(tmp_path):
+    d = tmp_path / "docs"
+    d.mkdir()
+    (d / "x.txt").write_bytes(b"abc")
+    t = pw.io.fs.read(str(d), format="binary", mode="static", with_metadata=True)
+    _, cols = table_to_dicts(t)
+    (meta,) = cols["_metadata"].values()
+    mv = meta.value if hasattr(meta, "value") else meta
+    assert mv["path"].endswith("x.txt")
+    assert mv["size"] == 3
+
+
+def test_run_monitoring_in_out(capsys):
+    t = T(
+        """
+        a
+        1
+        """
+    )
+    import tempfile, os as _os
+
+    out = tempfile.mktemp(suffix=".csv")
+    pw.io.csv.write(t, out)
+    pw.run(monitoring_level=pw.MonitoringLevel.IN_OUT)
+    assert _os.path.exists(out)
+    _os.unlink(out)
