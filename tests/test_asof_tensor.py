@@ -271,3 +271,100 @@ def test_interval_join_matches_bruteforce_random():
             if lk == rk and lo <= rs - lt <= hi
         )
         assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
+
+
+def test_sliding_window_matches_bruteforce_random():
+    """windowby(sliding) under random insert/retract streams vs a brute
+    force over the surviving rows."""
+    for seed in range(6):
+        rng = random.Random(12000 + seed)
+        lines = ["t | v | __time__ | __diff__"]
+        live = []
+        serial = 0
+        for step in range(4):
+            etime = 2 * (step + 1)
+            for _ in range(rng.randint(1, 5)):
+                serial += 1
+                t_ = rng.randint(0, 25)
+                lines.append(f"{t_} | {serial} | {etime} | 1")
+                live.append((t_, serial))
+            if live and rng.random() < 0.6:
+                victim = rng.choice(live)
+                live.remove(victim)
+                lines.append(f"{victim[0]} | {victim[1]} | {etime} | -1")
+        md = "\n".join(lines)
+        pw.internals.rungraph.G.clear()
+        tbl = T(md, id_from=["t", "v"])
+        hop, dur = 5, 10
+        res = tbl.windowby(
+            tbl.t, window=pw.temporal.sliding(hop=hop, duration=dur)
+        ).reduce(
+            start=pw.this._pw_window_start,
+            n=pw.reducers.count(),
+            s=pw.reducers.sum(pw.this.v),
+        )
+        _, cols = table_to_dicts(res)
+        got = sorted(
+            zip(cols["start"].values(), cols["n"].values(), cols["s"].values())
+        )
+        # brute force over surviving rows
+        from collections import defaultdict
+
+        agg = defaultdict(lambda: [0, 0])
+        for (t_, v) in live:
+            first = ((t_ - dur) // hop + 1) * hop
+            w = first
+            while w <= t_:
+                if t_ < w + dur:
+                    agg[w][0] += 1
+                    agg[w][1] += v
+                w += hop
+        expected = sorted((w, n, s) for w, (n, s) in agg.items() if n > 0)
+        assert got == expected, f"seed {seed}\n{md}\n{got}\nvs\n{expected}"
+
+
+def test_groupreduce_multiset_fuzz_vs_pandas():
+    """min/max/count_distinct under random retraction streams vs pandas
+    on the surviving multiset."""
+    import pandas as pd
+
+    for seed in range(6):
+        # build the stream with explicit serials in the row id so
+        # retractions target the exact inserted row
+        md_lines = ["g | v | sid | __time__ | __diff__"]
+        live2 = []
+        rng2 = random.Random(15000 + seed)
+        serial = 0
+        for step in range(4):
+            etime = 2 * (step + 1)
+            for _ in range(rng2.randint(2, 5)):
+                serial += 1
+                g = rng2.choice(["a", "b"])
+                v = rng2.randint(-20, 20)
+                md_lines.append(f"{g} | {v} | {serial} | {etime} | 1")
+                live2.append((g, v, serial))
+            if live2 and rng2.random() < 0.7:
+                victim = rng2.choice(live2)
+                live2.remove(victim)
+                md_lines.append(
+                    f"{victim[0]} | {victim[1]} | {victim[2]} | {etime} | -1"
+                )
+        pw.internals.rungraph.G.clear()
+        tbl = T("\n".join(md_lines), id_from=["g", "v", "sid"])
+        r = tbl.groupby(pw.this.g).reduce(
+            pw.this.g,
+            mn=pw.reducers.min(pw.this.v),
+            mx=pw.reducers.max(pw.this.v),
+            nd=pw.reducers.count_distinct(pw.this.v),
+        )
+        _, cols = table_to_dicts(r)
+        got = sorted(
+            zip(cols["g"].values(), cols["mn"].values(), cols["mx"].values(),
+                cols["nd"].values())
+        )
+        df = pd.DataFrame(live2, columns=["g", "v", "sid"])
+        expected = sorted(
+            (g, int(sub["v"].min()), int(sub["v"].max()), int(sub["v"].nunique()))
+            for g, sub in df.groupby("g")
+        )
+        assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
