@@ -368,3 +368,41 @@ def test_groupreduce_multiset_fuzz_vs_pandas():
             for g, sub in df.groupby("g")
         )
         assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
+
+
+def test_set_ops_fuzz_vs_bruteforce():
+    """intersect/difference under random retraction streams."""
+    for seed in range(6):
+        rng = random.Random(19000 + seed)
+
+        def gen(tagged):
+            lines = ["id | v | __time__ | __diff__"]
+            live = set()
+            for step in range(4):
+                etime = 2 * (step + 1)
+                for _ in range(rng.randint(1, 4)):
+                    k = rng.randint(0, 9)
+                    if k in live:
+                        continue
+                    live.add(k)
+                    lines.append(f"{tagged}{k} | {k} | {etime} | 1")
+                if live and rng.random() < 0.5:
+                    k = rng.choice(sorted(live))
+                    live.remove(k)
+                    lines.append(f"{tagged}{k} | {k} | {etime} | -1")
+            return "\n".join(lines), live
+
+        # same id-space for both tables: tag must be identical
+        lmd, lleft = gen("r")
+        rmd, lright = gen("r")
+        pw.internals.rungraph.G.clear()
+        a = T(lmd)
+        b = T(rmd)
+        inter = a.intersect(b)
+        diffr = a.difference(b)
+        _, ci = table_to_dicts(inter)
+        _, cd = table_to_dicts(diffr)
+        got_i = sorted(ci["v"].values())
+        got_d = sorted(cd["v"].values())
+        assert got_i == sorted(lleft & lright), f"seed {seed} intersect"
+        assert got_d == sorted(lleft - lright), f"seed {seed} difference"
