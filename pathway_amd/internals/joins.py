@@ -119,6 +119,11 @@ class JoinResult(Joinable):
 
         out_exprs: dict[str, ex.ColumnExpression] = {}
         for a in args:
+            if isinstance(a, thisclass.ThisSplat):
+                src = left if a.cls in (thisclass.left, thisclass.this) else right
+                for n in src._dtypes:
+                    out_exprs[n] = ex.ColumnReference(src, n)
+                continue
             a = thisclass.substitute_this(
                 ex.wrap_expr(a),
                 {thisclass.left: left, thisclass.right: right},

@@ -133,6 +133,10 @@ class Table(TableLike):
     ) -> dict[str, ex.ColumnExpression]:
         out: dict[str, ex.ColumnExpression] = {}
         for a in args:
+            if isinstance(a, thisclass.ThisSplat):
+                for n in self._dtypes:
+                    out[n] = ex.ColumnReference(self, n)
+                continue
             if isinstance(a, thisclass.ThisMetaclass):
                 raise TypeError("pass pw.this.column, not pw.this")
             a = _substitute(ex.wrap_expr(a), {thisclass.this: self})

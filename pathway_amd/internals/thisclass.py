@@ -19,7 +19,9 @@ class ThisMetaclass(type):
         return ColumnReference(cls, name)
 
     def __iter__(cls):
-        raise TypeError(f"{cls._repr} is not iterable before resolution")
+        # `select(*pw.this)` / `select(*pw.left)`: defer the expansion to
+        # resolution time (reference table.py supports splatting this)
+        yield ThisSplat(cls)
 
     def __repr__(cls) -> str:
         return cls._repr
@@ -76,3 +78,11 @@ def substitute_this(expr: Any, mapping: dict[type, Any]) -> Any:
 
     _ = expr_mod  # keep import for clarity
     return sub(expr)
+
+
+class ThisSplat:
+    """Deferred `*pw.this` marker: expands to every column of the resolved
+    table inside select()/reduce() argument handling."""
+
+    def __init__(self, cls):
+        self.cls = cls

@@ -351,3 +351,34 @@ def test_update_cells_fuzz_vs_pandas():
             df.loc[k, "x"] = x
         expected = sorted(zip(df["x"].tolist(), df["y"].tolist()))
         assert got == expected, f"seed {seed}: {got} vs {expected}"
+
+
+def test_select_splat_this():
+    t = T(
+        """
+        a | b
+        1 | 2
+        """
+    )
+    r = t.select(*pw.this, c=pw.this.a + pw.this.b)
+    _, cols = table_to_dicts(r)
+    assert sorted(cols) == ["a", "b", "c"]
+    assert list(cols["c"].values()) == [3]
+
+
+def test_join_select_splat_left():
+    l = T(
+        """
+        k | a
+        1 | p
+        """
+    )
+    r = T(
+        """
+        k | b
+        1 | u
+        """
+    )
+    res = l.join(r, l.k == r.k).select(*pw.left, b=pw.right.b)
+    _, cols = table_to_dicts(res)
+    assert sorted(cols) == ["a", "b", "k"]
