@@ -122,7 +122,8 @@ class JoinResult(Joinable):
             if isinstance(a, thisclass.ThisSplat):
                 src = left if a.cls in (thisclass.left, thisclass.this) else right
                 for n in src._dtypes:
-                    out_exprs[n] = ex.ColumnReference(src, n)
+                    if n not in a.exclude:
+                        out_exprs[n] = ex.ColumnReference(src, n)
                 continue
             a = thisclass.substitute_this(
                 ex.wrap_expr(a),

@@ -135,7 +135,8 @@ class Table(TableLike):
         for a in args:
             if isinstance(a, thisclass.ThisSplat):
                 for n in self._dtypes:
-                    out[n] = ex.ColumnReference(self, n)
+                    if n not in a.exclude:
+                        out[n] = ex.ColumnReference(self, n)
                 continue
             if isinstance(a, thisclass.ThisMetaclass):
                 raise TypeError("pass pw.this.column, not pw.this")

@@ -11,6 +11,16 @@ class ThisMetaclass(type):
     def __getattr__(cls, name: str) -> ColumnReference:
         if name.startswith("__") and name.endswith("__"):
             raise AttributeError(name)
+        if name == "without":
+            # pw.this.without(pw.this.a, ...): deferred splat-with-exclusions
+            def _without(*cols):
+                excl = {
+                    c.name if isinstance(c, ColumnReference) else str(c)
+                    for c in cols
+                }
+                return _SplatIter(ThisSplat(cls, exclude=excl))
+
+            return _without
         return ColumnReference(cls, name)
 
     def __getitem__(cls, name: str) -> ColumnReference:
@@ -84,5 +94,16 @@ class ThisSplat:
     """Deferred `*pw.this` marker: expands to every column of the resolved
     table inside select()/reduce() argument handling."""
 
-    def __init__(self, cls):
+    def __init__(self, cls, exclude=None):
         self.cls = cls
+        self.exclude = set(exclude or ())
+
+
+class _SplatIter:
+    """Wrapper so `*pw.this.without(...)` splats to one ThisSplat marker."""
+
+    def __init__(self, splat):
+        self._splat = splat
+
+    def __iter__(self):
+        yield self._splat

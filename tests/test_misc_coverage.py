@@ -382,3 +382,15 @@ def test_join_select_splat_left():
     res = l.join(r, l.k == r.k).select(*pw.left, b=pw.right.b)
     _, cols = table_to_dicts(res)
     assert sorted(cols) == ["a", "b", "k"]
+
+
+def test_select_splat_without():
+    t = T(
+        """
+        a | b | c
+        1 | 2 | 3
+        """
+    )
+    r = t.select(*pw.this.without(pw.this.a))
+    _, cols = table_to_dicts(r)
+    assert sorted(cols) == ["b", "c"]
