@@ -406,3 +406,49 @@ def test_set_ops_fuzz_vs_bruteforce():
         got_d = sorted(cd["v"].values())
         assert got_i == sorted(lleft & lright), f"seed {seed} intersect"
         assert got_d == sorted(lleft - lright), f"seed {seed} difference"
+
+
+def test_left_join_fuzz_vs_bruteforce():
+    """join_left under random insert/retract streams on both sides:
+    pad→match and match→pad transitions must net out exactly."""
+    for seed in range(8):
+        rng = random.Random(21000 + seed)
+
+        def gen(side):
+            lines = [f"k | {side} | __time__ | __diff__"]
+            live = []
+            serial = 0
+            for step in range(4):
+                etime = 2 * (step + 1)
+                for _ in range(rng.randint(1, 3)):
+                    serial += 1
+                    k = rng.randint(1, 3)
+                    v = f"{side}{serial}"
+                    lines.append(f"{k} | {v} | {etime} | 1")
+                    live.append((k, v))
+                if live and rng.random() < 0.6:
+                    victim = rng.choice(live)
+                    live.remove(victim)
+                    lines.append(f"{victim[0]} | {victim[1]} | {etime} | -1")
+            return "\n".join(lines), live
+
+        lmd, llive = gen("a")
+        rmd, rlive = gen("b")
+        pw.internals.rungraph.G.clear()
+        l = T(lmd, id_from=["k", "a"])
+        r = T(rmd, id_from=["k", "b"])
+        res = l.join_left(r, l.k == r.k).select(pw.left.a, pw.right.b)
+        _, cols = table_to_dicts(res)
+        got = sorted(
+            zip(cols["a"].values(), cols["b"].values()),
+            key=lambda x: (x[0], x[1] is None, x[1]),
+        )
+        expected = []
+        for (lk, a) in llive:
+            matches = [b for (rk, b) in rlive if rk == lk]
+            if matches:
+                expected.extend((a, b) for b in matches)
+            else:
+                expected.append((a, None))
+        expected = sorted(expected, key=lambda x: (x[0], x[1] is None, x[1]))
+        assert got == expected, f"seed {seed}\nL:{lmd}\nR:{rmd}\n{got}\nvs\n{expected}"
