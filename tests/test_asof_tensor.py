@@ -452,3 +452,87 @@ def test_left_join_fuzz_vs_bruteforce():
                 expected.append((a, None))
         expected = sorted(expected, key=lambda x: (x[0], x[1] is None, x[1]))
         assert got == expected, f"seed {seed}\nL:{lmd}\nR:{rmd}\n{got}\nvs\n{expected}"
+
+
+def test_outer_join_fuzz_vs_bruteforce():
+    for seed in range(6):
+        rng = random.Random(23000 + seed)
+
+        def gen(side):
+            lines = [f"k | {side} | __time__ | __diff__"]
+            live = []
+            serial = 0
+            for step in range(3):
+                etime = 2 * (step + 1)
+                for _ in range(rng.randint(1, 3)):
+                    serial += 1
+                    k = rng.randint(1, 3)
+                    v = f"{side}{serial}"
+                    lines.append(f"{k} | {v} | {etime} | 1")
+                    live.append((k, v))
+                if live and rng.random() < 0.5:
+                    victim = rng.choice(live)
+                    live.remove(victim)
+                    lines.append(f"{victim[0]} | {victim[1]} | {etime} | -1")
+            return "\n".join(lines), live
+
+        lmd, llive = gen("a")
+        rmd, rlive = gen("b")
+        pw.internals.rungraph.G.clear()
+        l = T(lmd, id_from=["k", "a"])
+        r = T(rmd, id_from=["k", "b"])
+        res = l.join_outer(r, l.k == r.k).select(pw.left.a, pw.right.b)
+        _, cols = table_to_dicts(res)
+        key = lambda x: (x[0] is None, x[0], x[1] is None, x[1])
+        got = sorted(zip(cols["a"].values(), cols["b"].values()), key=key)
+        expected = []
+        for (lk, a) in llive:
+            ms = [b for (rk, b) in rlive if rk == lk]
+            expected.extend((a, b) for b in ms) if ms else expected.append((a, None))
+        for (rk, b) in rlive:
+            if not any(lk == rk for (lk, _) in llive):
+                expected.append((None, b))
+        expected = sorted(expected, key=key)
+        assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
+
+
+def test_window_join_fuzz_vs_bruteforce():
+    for seed in range(6):
+        rng = random.Random(25000 + seed)
+        dur = 5
+
+        def gen(side):
+            lines = [f"t | {side} | __time__ | __diff__"]
+            live = []
+            serial = 0
+            for step in range(3):
+                etime = 2 * (step + 1)
+                for _ in range(rng.randint(1, 3)):
+                    serial += 1
+                    t_ = rng.randint(0, 14)
+                    v = f"{side}{serial}"
+                    lines.append(f"{t_} | {v} | {etime} | 1")
+                    live.append((t_, v))
+                if live and rng.random() < 0.5:
+                    victim = rng.choice(live)
+                    live.remove(victim)
+                    lines.append(f"{victim[0]} | {victim[1]} | {etime} | -1")
+            return "\n".join(lines), live
+
+        lmd, llive = gen("a")
+        rmd, rlive = gen("b")
+        pw.internals.rungraph.G.clear()
+        l = T(lmd, id_from=["t", "a"])
+        r = T(rmd, id_from=["t", "b"])
+        res = l.window_join_inner(
+            r, l.t, r.t, pw.temporal.tumbling(duration=dur)
+        ).select(pw.left.a, pw.right.b)
+        _, cols = table_to_dicts(res)
+        got = sorted(zip(cols["a"].values(), cols["b"].values()))
+        expected = sorted(
+            (a, b)
+            for (lt, a) in llive
+            for (rt, b) in rlive
+            if lt // dur == rt // dur
+        )
+        assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
