@@ -1,6 +1,12 @@
-"""Tensor asof-join (engine/nodes_asof.py) vs the host RecomputeNode
-reference implementation: randomized update streams with inserts and
-retractions on both sides must produce identical final tables."""
+"""Randomized equivalence / brute-force fuzz harness (the analog of the
+reference's property-based operator tests).
+
+Every incremental operator family is driven with random insert+retract
+update streams and compared against an independent oracle: the host
+RecomputeNode implementations (asof, sort, session) or a plain
+brute-force/pandas evaluation of the surviving multiset (joins incl.
+outer padding, interval and window joins, sliding windows, multiset
+reducers, set ops)."""
 
 import os
 import random
