@@ -101,3 +101,29 @@ def test_louvain_two_cliques():
     _, qc = table_to_dicts(q)
     (qv,) = qc["modularity"].values()
     assert qv > 0.3  # two-clique split has high modularity
+
+
+def test_weighted_graph_contraction():
+    import pathway_amd as pw
+    from pathway_amd.debug import table_from_markdown as T, table_to_dicts
+    from pathway_amd.stdlib.graphs import louvain_level
+    from pathway_amd.stdlib.graphs.graph import Graph
+
+    raw = T(
+        """
+        a | b
+        1 | 2
+        2 | 3
+        4 | 5
+        3 | 4
+        """
+    )
+    edges = raw.select(u=raw.pointer_from(pw.this.a), v=raw.pointer_from(pw.this.b))
+    cl = louvain_level(edges)
+    g = Graph(None, edges)
+    g2 = g.contracted_to_simple_graph(cl)
+    _, cols = table_to_dicts(g2.E)
+    # contracted graph has at most as many edges as communities allow,
+    # and no self loops
+    for i in cols["u"]:
+        assert repr(cols["u"][i]) != repr(cols["v"][i])
