@@ -44,6 +44,17 @@ class RunGraph:
             from pathway_amd.persistence.engine import PersistenceManager
             from pathway_amd.internals.config import pathway_config
 
+            mode = getattr(persistence_config, "persistence_mode", None)
+            if mode is not None and "udf_caching" in str(mode).lower():
+                # UdfCaching: point the default DiskCache at the backend
+                # (reference PersistenceMode::UdfCaching — deterministic
+                # replay of non-deterministic UDFs)
+                import os as _os
+
+                backend = getattr(persistence_config, "backend", None)
+                path = getattr(backend, "path", None)
+                if path:
+                    _os.environ["PATHWAY_PERSISTENT_STORAGE"] = str(path)
             pm = PersistenceManager(persistence_config, worker=pathway_config.process_id)
         rt = Runtime(sinks, device=get_device(), comm=self.comm, persistence=pm)
         http_server = None

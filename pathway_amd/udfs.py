@@ -124,13 +124,19 @@ class InMemoryCache(CacheStrategy):
 
 class DiskCache(CacheStrategy):
     def __init__(self, directory: str | None = None, name: str | None = None):
-        self.directory = directory or os.path.join(
-            os.environ.get("PATHWAY_PERSISTENT_STORAGE", "/tmp/pw_udf_cache")
-        )
+        self._directory = directory
         self.name = name
 
+    @property
+    def directory(self) -> str:
+        # resolved lazily so PersistenceMode.UDF_CACHING (which points the
+        # env var at the persistence backend) takes effect even when the
+        # UDF was declared before pw.run()
+        return self._directory or os.environ.get(
+            "PATHWAY_PERSISTENT_STORAGE", "/tmp/pw_udf_cache"
+        )
+
     def wrap(self, fun: Callable) -> Callable:
-        os.makedirs(self.directory, exist_ok=True)
         prefix = self.name or getattr(fun, "__name__", "udf")
 
         @functools.wraps(fun)
@@ -144,6 +150,7 @@ class DiskCache(CacheStrategy):
                 with open(path, "rb") as f:
                     return pickle.load(f)
             result = fun(*args, **kwargs)
+            os.makedirs(self.directory, exist_ok=True)
             with open(path, "wb") as f:
                 pickle.dump(result, f)
             return result

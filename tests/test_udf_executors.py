@@ -184,3 +184,38 @@ def test_udf_class_style():
         """
     )
     assert _col(t.select(b=m(pw.this.a)), "b") == [15]
+
+
+def test_udf_caching_persistence_mode(tmp_path):
+    """PersistenceMode.UDF_CACHING routes DiskCache at the backend dir."""
+    calls = {"n": 0}
+
+    @pw.udf(cache_strategy=pw.udfs.DiskCache(name="pmodecache"))
+    def f(x: int) -> int:
+        calls["n"] += 1
+        return x * 7
+
+    t = T(
+        """
+        a
+        3
+        """
+    )
+    res = t.select(b=f(pw.this.a))
+    out = str(tmp_path / "out.csv")
+    pw.io.csv.write(res, out)
+    cfg = pw.persistence.Config(
+        backend=pw.persistence.Backend.filesystem(str(tmp_path / "snap")),
+        persistence_mode=pw.PersistenceMode.UDF_CACHING,
+    )
+    import os as _os
+
+    try:
+        pw.run(monitoring_level=pw.MonitoringLevel.NONE, persistence_config=cfg)
+        cached = [
+            p for p in (tmp_path / "snap").rglob("pmodecache_*.pkl")
+        ]
+        assert cached, "UDF result not cached under the persistence backend"
+    finally:
+        _os.environ.pop("PATHWAY_PERSISTENT_STORAGE", None)
+        pw.internals.rungraph.G.clear()
