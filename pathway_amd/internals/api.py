@@ -351,3 +351,11 @@ class _PendingValue:
 
 
 PENDING = _PendingValue()
+
+
+def unsafe_make_pointer(value) -> "BasePointer":
+    """Pointer directly from a trusted integer (reference api
+    unsafe_make_pointer) — no hashing; collision safety is the caller's
+    responsibility."""
+    v = int(value)
+    return Pointer(v & MASK64, 0)
