@@ -394,3 +394,35 @@ def test_select_splat_without():
     r = t.select(*pw.this.without(pw.this.a))
     _, cols = table_to_dicts(r)
     assert sorted(cols) == ["b", "c"]
+
+
+def test_yaml_loader_instantiates_objects():
+    import io as _io
+
+    yml = """
+$chat: !pw.xpacks.llm.llms.EchoChat
+  prefix: "A> "
+answerer:
+  llm: $chat
+  topk: 3
+"""
+    out = pw.load_yaml(_io.StringIO(yml))
+    assert type(out["answerer"]["llm"]).__name__ == "EchoChat"
+    assert out["answerer"]["topk"] == 3
+
+
+def test_compute_and_print_update_stream(capsys):
+    from pathway_amd.debug import compute_and_print_update_stream
+
+    t = T(
+        """
+        a | __time__ | __diff__
+        1 |    2     |    1
+        1 |    4     |   -1
+        2 |    4     |    1
+        """,
+        id_from=["a"],
+    )
+    compute_and_print_update_stream(t)
+    out = capsys.readouterr().out
+    assert "-1" in out and "2" in out
