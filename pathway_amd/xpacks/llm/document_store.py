@@ -79,9 +79,28 @@ class DocumentStore:
             ),
         )
         for post in self.doc_post_processors:
-            parsed = parsed.select(
-                text=common.apply_with_type(post, dt.STR, this.text),
-                _pw_meta=this._pw_meta,
+            # reference contract: post(text, metadata) -> (text, metadata)
+            def _apply_post(text, meta, _post=post):
+                mv = meta.value if hasattr(meta, "value") else meta
+                new_text, new_meta = _post(text, mv)
+                return Json({"text": new_text, "meta": dict(new_meta or {})})
+
+            packed = parsed.select(
+                _pw_pp=common.apply_with_type(
+                    _apply_post, dt.JSON, this.text, this._pw_meta
+                ),
+            )
+            parsed = packed.select(
+                text=common.apply_with_type(
+                    lambda p: p["text"].as_str(), dt.STR, this._pw_pp
+                ),
+                _pw_meta=common.apply_with_type(
+                    lambda p: Json(dict(p["meta"].as_dict())
+                                   if hasattr(p["meta"], "as_dict")
+                                   else p.value["meta"]),
+                    dt.JSON,
+                    this._pw_pp,
+                ),
             )
         chunks = parsed.select(
             _pw_chunks=self.splitter(this.text, this._pw_meta),

@@ -335,3 +335,24 @@ def test_adaptive_rag_answerer():
     (ans,) = cols["result"].values()
     assert "answer with" in ans
     assert len(chat.calls) >= 2  # escalated at least once
+
+
+def test_document_store_post_processors():
+    from pathway_amd.xpacks.llm.document_store import DocumentStore
+
+    schema = schema_from_types(data=bytes, _metadata=dict)
+    docs = table_from_rows(schema, [(b"hello world", {"path": "p.txt"})])
+
+    def shout(text, metadata):
+        return text.upper(), metadata
+
+    store = DocumentStore(docs, doc_post_processors=[shout])
+    qschema = DocumentStore.RetrieveQuerySchema
+    q = table_from_rows(
+        qschema, [("hello", 1, None, None)]
+    )
+    reply = store.retrieve_query(q)
+    _, cols = table_to_dicts(reply)
+    (res,) = cols["result"].values()
+    rv = res.value if hasattr(res, "value") else res
+    assert "HELLO WORLD" in rv[0]["text"]
