@@ -356,3 +356,33 @@ def test_document_store_post_processors():
     (res,) = cols["result"].values()
     rv = res.value if hasattr(res, "value") else res
     assert "HELLO WORLD" in rv[0]["text"]
+
+
+def test_default_document_index_factories():
+    import numpy as np
+
+    from pathway_amd.stdlib.indexing import (
+        default_brute_force_knn_document_index,
+        default_usearch_knn_document_index,
+    )
+
+    rng = np.random.default_rng(0)
+    vecs = [tuple(map(float, rng.normal(size=8))) for _ in range(20)]
+    docs = table_from_rows(
+        schema_from_types(vec=list, label=str),
+        [(list(v), f"d{i}") for i, v in enumerate(vecs)],
+    )
+    for factory in (
+        default_brute_force_knn_document_index,
+        default_usearch_knn_document_index,
+    ):
+        pwc = factory(docs.vec, docs, dimensions=8)
+        queries = table_from_rows(
+            schema_from_types(qv=list, k=int), [(list(vecs[3]), 3)]
+        )
+        res = pwc.query_as_of_now(
+            queries.qv, number_of_matches=queries.k, collapse_rows=True
+        )
+        _, cols = table_to_dicts(res)
+        name = [n for n in cols if "id" in n or "reply" in n]
+        assert cols, (factory, cols.keys())
