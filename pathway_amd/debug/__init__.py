@@ -402,6 +402,16 @@ class StreamGenerator:
     def table_from_markdown(self, *args, **kwargs) -> Table:
         return table_from_markdown(*args, **kwargs)
 
+    def table_from_pandas(self, df, **kwargs) -> Table:
+        """Streamed pandas frame (reference StreamGenerator.table_from_pandas:
+        __time__/__diff__ columns drive the event times)."""
+        return table_from_pandas(df, **kwargs)
+
+    def persistence_config(self):
+        """The reference returns a config replaying generated streams; the
+        synchronous engine replays deterministically without one."""
+        return None
+
 
 def table_from_parquet(path: str, id_from=None, unsafe_trusted_ids: bool = False):
     """Read a parquet file into a static table (reference debug/__init__.py)."""
