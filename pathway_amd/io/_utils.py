@@ -35,3 +35,20 @@ def check_deprecated_kwargs(kwargs, names, stacklevel=1):
 
 class RawDataSchema:
     pass
+
+
+def expand_paths(path: str) -> list[str]:
+    """Directory walk / glob pattern / single file → concrete file list
+    (the posix_like scanner's path semantics, data_storage sharding.rs)."""
+    import glob as _glob
+    import os as _os
+
+    if _os.path.isdir(path):
+        out = []
+        for root, _, fnames in _os.walk(path):
+            for f in sorted(fnames):
+                out.append(_os.path.join(root, f))
+        return out
+    if any(ch in path for ch in "*?["):
+        return sorted(_glob.glob(path))
+    return [path]

@@ -21,13 +21,9 @@ def read(
     from pathway_amd.debug import table_from_rows
     from pathway_amd.internals.schema import schema_from_csv
 
-    files = []
-    if os.path.isdir(path):
-        for root, _, fnames in os.walk(path):
-            for f in sorted(fnames):
-                files.append(os.path.join(root, f))
-    else:
-        files = [path]
+    from pathway_amd.io._utils import expand_paths
+
+    files = expand_paths(path)
     delimiter = ","
     quotechar = '"'
     if csv_settings is not None:

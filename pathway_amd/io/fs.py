@@ -94,13 +94,9 @@ def _read_binary(path: str, with_metadata: bool = False):
     from pathway_amd.debug import table_from_rows
     from pathway_amd.internals.schema import schema_from_types
 
-    files = []
-    if os.path.isdir(path):
-        for root, _, fnames in os.walk(path):
-            for f in sorted(fnames):
-                files.append(os.path.join(root, f))
-    else:
-        files = [path]
+    from pathway_amd.io._utils import expand_paths
+
+    files = expand_paths(path)
     rows = []
     for f in files:
         with open(f, "rb") as fh:

@@ -21,13 +21,9 @@ def read(
     from pathway_amd.debug import table_from_rows
     from pathway_amd.internals.json import Json
 
-    files = []
-    if os.path.isdir(path):
-        for root, _, fnames in os.walk(path):
-            for f in sorted(fnames):
-                files.append(os.path.join(root, f))
-    else:
-        files = [path]
+    from pathway_amd.io._utils import expand_paths
+
+    files = expand_paths(path)
     records = []
     for f in files:
         with open(f) as fh:

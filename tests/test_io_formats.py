@@ -183,3 +183,11 @@ def test_native_read_lines(tmp_path):
     p = tmp_path / "l.txt"
     p.write_bytes(b"one\r\ntwo\nthree")
     assert native_io.read_lines(str(p)) == ["one", "two", "three"]
+
+
+def test_static_reads_expand_globs(tmp_path):
+    (tmp_path / "a.txt").write_text("one\n")
+    (tmp_path / "b.txt").write_text("two\n")
+    (tmp_path / "c.log").write_text("skip\n")
+    t = pw.io.fs.read(str(tmp_path / "*.txt"), format="plaintext", mode="static")
+    assert _col_sorted(t, "data") == ["one", "two"]
