@@ -542,3 +542,43 @@ def test_window_join_fuzz_vs_bruteforce():
             if lt // dur == rt // dur
         )
         assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
+
+
+def test_groupby_by_id_fuzz_vs_pandas():
+    """groupby(id=pointer_expr) — the wordcount headline path — under
+    random retraction streams."""
+    import pandas as pd
+
+    for seed in range(6):
+        rng = random.Random(27000 + seed)
+        md_lines = ["w | n | sid | __time__ | __diff__"]
+        live = []
+        serial = 0
+        for step in range(4):
+            etime = 2 * (step + 1)
+            for _ in range(rng.randint(2, 6)):
+                serial += 1
+                w = rng.choice(["x", "y", "z"])
+                n = rng.randint(1, 5)
+                md_lines.append(f"{w} | {n} | {serial} | {etime} | 1")
+                live.append((w, n, serial))
+            if live and rng.random() < 0.6:
+                victim = rng.choice(live)
+                live.remove(victim)
+                md_lines.append(
+                    f"{victim[0]} | {victim[1]} | {victim[2]} | {etime} | -1"
+                )
+        pw.internals.rungraph.G.clear()
+        t = T("\n".join(md_lines), id_from=["w", "n", "sid"])
+        r = t.groupby(id=t.pointer_from(pw.this.w)).reduce(
+            w=pw.reducers.any(pw.this.w),
+            c=pw.reducers.count(),
+            s=pw.reducers.sum(pw.this.n),
+        )
+        _, cols = table_to_dicts(r)
+        got = sorted(zip(cols["w"].values(), cols["c"].values(), cols["s"].values()))
+        df = pd.DataFrame(live, columns=["w", "n", "sid"])
+        expected = sorted(
+            (w, int(len(sub)), int(sub["n"].sum())) for w, sub in df.groupby("w")
+        )
+        assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
