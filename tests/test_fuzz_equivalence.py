@@ -582,3 +582,31 @@ def test_groupby_by_id_fuzz_vs_pandas():
             (w, int(len(sub)), int(sub["n"].sum())) for w, sub in df.groupby("w")
         )
         assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
+
+
+def test_to_stream_roundtrip_fuzz():
+    """stream_to_table(to_stream(t)) reconstructs t under random updates."""
+    for seed in range(6):
+        rng = random.Random(29000 + seed)
+        md_lines = ["k | v | __time__ | __diff__"]
+        live = {}
+        for step in range(4):
+            etime = 2 * (step + 1)
+            for _ in range(rng.randint(1, 4)):
+                k = rng.randint(0, 5)
+                v = rng.randint(0, 99)
+                if k in live:
+                    # modification = retract old + insert new
+                    md_lines.append(f"{k} | {live[k]} | {etime} | -1")
+                md_lines.append(f"{k} | {v} | {etime} | 1")
+                live[k] = v
+            if live and rng.random() < 0.4:
+                k = rng.choice(sorted(live))
+                md_lines.append(f"{k} | {live.pop(k)} | {etime} | -1")
+        pw.internals.rungraph.G.clear()
+        t = T("\n".join(md_lines), id_from=["k"])
+        back = t.to_stream().stream_to_table()
+        _, cols = table_to_dicts(back)
+        got = sorted(zip(cols["k"].values(), cols["v"].values()))
+        expected = sorted(live.items())
+        assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
