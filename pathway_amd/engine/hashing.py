@@ -122,7 +122,13 @@ def hash128_words(words: list[torch.Tensor]) -> tuple[torch.Tensor, torch.Tensor
 
 
 def _float_bits(x: torch.Tensor) -> torch.Tensor:
-    return x.to(torch.float64).view(torch.int64)
+    # Normalize -0.0 -> 0.0 and NaN -> canonical 0x7FF8... so values that
+    # compare equal hash equal, matching api.serialize_value.
+    x = x.to(torch.float64)
+    x = torch.where(x == 0.0, torch.zeros_like(x), x)
+    bits = x.view(torch.int64)
+    canon_nan = torch.full_like(bits, 0x7FF8000000000000)
+    return torch.where(torch.isnan(x), canon_nan, bits)
 
 
 def value_hash_words(col_words: torch.Tensor, tag: int) -> tuple[torch.Tensor, torch.Tensor]:

@@ -146,7 +146,14 @@ def serialize_value(value: Any) -> bytes:
     if isinstance(value, (int, np.integer)):
         return w(TAG_INT) + struct.pack("<q", int(value))
     if isinstance(value, (float, np.floating)):
-        return w(TAG_FLOAT) + struct.pack("<d", float(value))
+        f = float(value)
+        # Normalize -0.0 -> 0.0 and NaN -> one canonical bit pattern so
+        # values that compare equal hash equal (reference value.rs:614-625).
+        if f == 0.0:
+            f = 0.0
+        elif f != f:
+            return w(TAG_FLOAT) + struct.pack("<Q", 0x7FF8000000000000)
+        return w(TAG_FLOAT) + struct.pack("<d", f)
     if isinstance(value, str):
         return w(TAG_STR) + value.encode("utf-8")
     if isinstance(value, bytes):
@@ -154,7 +161,12 @@ def serialize_value(value: Any) -> bytes:
     if isinstance(value, Json):
         return w(TAG_JSON) + value.dumps().encode("utf-8")
     if isinstance(value, np.ndarray):
-        return w(TAG_ARRAY) + value.tobytes()
+        # Shape + dtype kind are part of the key so that (2,2) vs (4,) or
+        # int64 vs float64 arrays with identical raw bytes do not collide
+        # (reference value.rs ArrayD hash_into hashes shape then elements).
+        header = struct.pack("<QQ", ord(value.dtype.kind), value.ndim)
+        header += b"".join(struct.pack("<Q", d) for d in value.shape)
+        return w(TAG_ARRAY) + header + value.tobytes()
     if isinstance(value, (tuple, list)):
         out = bytearray(w(TAG_TUPLE))
         for v in value:
@@ -164,12 +176,17 @@ def serialize_value(value: Any) -> bytes:
     import datetime
 
     if isinstance(value, datetime.timedelta):
-        ns = int(value.total_seconds() * 1e9)
-        return w(TAG_DURATION) + struct.pack("<q", ns)
+        # Exact integer-ns serialization (never float seconds): the device
+        # TensorColumn path hashes exact int64 ns, and the two must agree
+        # for pointer identity / joins on ns-precision values.
+        import pandas as pd
+
+        return w(TAG_DURATION) + struct.pack("<q", int(pd.Timedelta(value).value))
     if isinstance(value, datetime.datetime):
+        import pandas as pd
+
         tag = TAG_DT_UTC if value.tzinfo is not None else TAG_DT_NAIVE
-        ns = int(value.timestamp() * 1e9)
-        return w(tag) + struct.pack("<q", ns)
+        return w(tag) + struct.pack("<q", int(pd.Timestamp(value).value))
     # arbitrary python object: hash of repr as last resort
     return w(TAG_PYOBJ) + repr(value).encode("utf-8")
 

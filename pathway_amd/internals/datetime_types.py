@@ -54,7 +54,9 @@ def to_ns(value: Any) -> int:
     import datetime
 
     if isinstance(value, datetime.timedelta):
-        return int(value.total_seconds() * 1e9)
+        # Exact conversion through pandas — float seconds would round
+        # ns-precision values.
+        return int(pd.Timedelta(value).value)
     if isinstance(value, datetime.datetime):
         return int(pd.Timestamp(value).value)
     raise TypeError(f"not a datetime-like value: {value!r}")

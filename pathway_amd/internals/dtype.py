@@ -16,6 +16,9 @@ import datetime
 from typing import Any, Hashable
 
 import numpy as np
+import pandas as pd
+
+from pathway_amd.internals import datetime_types as _dt_types
 
 
 class DType:
@@ -214,6 +217,11 @@ def _init_py_type_map() -> None:
             typing.Any: ANY,
             datetime.datetime: DATE_TIME_NAIVE,
             datetime.timedelta: DURATION,
+            _dt_types.DateTimeNaive: DATE_TIME_NAIVE,
+            _dt_types.DateTimeUtc: DATE_TIME_UTC,
+            _dt_types.Duration: DURATION,
+            pd.Timestamp: DATE_TIME_NAIVE,
+            pd.Timedelta: DURATION,
             np.int64: INT,
             np.int32: INT,
             np.float64: FLOAT,
@@ -272,6 +280,12 @@ def wrap(input_type: Any) -> DType:
         pass
     if isinstance(input_type, Hashable) and input_type in _PY_TYPE_MAP:
         return _PY_TYPE_MAP[input_type]
+    # MRO fallback: subclasses of mapped types (e.g. user subclasses of
+    # pd.Timestamp) resolve to the mapped base instead of ANY.
+    if isinstance(input_type, type):
+        for base in input_type.__mro__[1:]:
+            if base in _PY_TYPE_MAP:
+                return _PY_TYPE_MAP[base]
     return ANY
 
 

@@ -65,3 +65,77 @@ def test_combined_row_hash_matches_hash_values():
         elo, ehi = api.hash_values([x, y])
         assert (int(lo[i]) & api.MASK64) == elo
         assert (int(hi[i]) & api.MASK64) == ehi
+
+
+def test_datetime_ns_precision_host_device_agree():
+    # ns-precision datetimes must hash identically via the
+    # host serialize path and the device int64-ns column path.
+    import pandas as pd
+
+    ns = 1694512345123456789
+    ts = pd.Timestamp(ns, unit="ns")
+    elo, ehi = api.hash128(api.serialize_value(ts))
+    t = torch.tensor([ns], dtype=torch.int64)
+    lo, hi = hashing.column_value_hash(t, "datetime_naive")
+    assert (int(lo[0]) & api.MASK64) == elo
+    assert (int(hi[0]) & api.MASK64) == ehi
+
+    # duration too
+    import datetime
+
+    td = datetime.timedelta(microseconds=123456789123)
+    expected_ns = int(pd.Timedelta(td).value)
+    elo2, _ = api.hash128(api.serialize_value(td))
+    lo2, _ = hashing.column_value_hash(
+        torch.tensor([expected_ns], dtype=torch.int64), "duration"
+    )
+    assert (int(lo2[0]) & api.MASK64) == elo2
+
+
+def test_float_zero_and_nan_normalization():
+    # -0.0 and 0.0 hash identically; all NaNs hash to one canonical pattern.
+    assert api.serialize_value(0.0) == api.serialize_value(-0.0)
+    nan1 = float("nan")
+    nan2 = struct.unpack("<d", struct.pack("<Q", 0x7FF8000000000001))[0]
+    assert api.serialize_value(nan1) == api.serialize_value(nan2)
+
+    t = torch.tensor([0.0, -0.0, nan1, nan2], dtype=torch.float64)
+    lo, _ = hashing.column_value_hash(t, "float")
+    assert int(lo[0]) == int(lo[1])
+    assert int(lo[2]) == int(lo[3])
+    elo, _ = api.hash128(api.serialize_value(0.0))
+    assert (int(lo[0]) & api.MASK64) == elo
+    enan, _ = api.hash128(api.serialize_value(nan1))
+    assert (int(lo[2]) & api.MASK64) == enan
+
+
+def test_ndarray_key_includes_shape_and_dtype():
+    import numpy as np
+
+    a = np.arange(4, dtype=np.int64).reshape(2, 2)
+    b = np.arange(4, dtype=np.int64)
+    assert api.serialize_value(a) != api.serialize_value(b)
+    c = b.view(np.float64)
+    assert api.serialize_value(b) != api.serialize_value(c)
+
+
+def test_wrap_datetime_public_types():
+    import pandas as pd
+
+    from pathway_amd.internals import dtype as dt
+    from pathway_amd.internals.datetime_types import (
+        DateTimeNaive,
+        DateTimeUtc,
+        Duration,
+    )
+
+    assert dt.wrap(DateTimeNaive) == dt.DATE_TIME_NAIVE
+    assert dt.wrap(DateTimeUtc) == dt.DATE_TIME_UTC
+    assert dt.wrap(Duration) == dt.DURATION
+    assert dt.wrap(pd.Timestamp) == dt.DATE_TIME_NAIVE
+    assert dt.wrap(pd.Timedelta) == dt.DURATION
+
+    class MyTs(pd.Timestamp):
+        pass
+
+    assert dt.wrap(MyTs) == dt.DATE_TIME_NAIVE
