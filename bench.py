@@ -4,11 +4,14 @@ BASELINE.md headline: pathway's published WordCount figure is 2,000,000
 messages/s with p95 end-to-end latency < 50 ms (4 CPU cores/system,
 docs/.../50.how-live-data-framework-connectors-work.md:276).  This bench
 runs the same shape — a stream of words → incremental groupby(word).count
-→ consolidated output deltas — through the pathway_amd GPU engine:
-per step one micro-batch of synthetic words is keyed (HIP xxh64-128
-kernel), shuffled by key shard (RCCL all-to-all over xGMI for N>1),
-segment-reduced and merged into GPU-resident state, and the (-old,+new)
-count deltas are emitted.
+→ consolidated output deltas — through the pathway_amd GPU engine, with
+the WHOLE-NODE ingest path billed inside the timed region: newline-
+separated wire bytes start in pinned HOST memory, are DMA'd H2D on a
+dedicated copy stream (double-buffered to overlap with compute), parsed
+(newline scan), token-hashed (HIP varlen xxh128 kernel), dictionary-
+decoded (sorted-pool binary-search kernel), keyed, shuffled by key shard
+(RCCL all-to-all over xGMI for N>1), segment-reduced and merged into
+GPU-resident state; the (-old,+new) count deltas are emitted.
 
 Driver contract: --gpus N --steps K --warmup W; launched via torchrun for
 N>1 (one rank per GPU, RCCL); rank 0 prints ONE json line.
@@ -29,7 +32,12 @@ def main() -> None:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch", type=int, default=4_000_000, help="events per step per GPU")
+    p.add_argument(
+        "--batch",
+        type=int,
+        default=None,
+        help="events per step per GPU (default 4M on GPU, 100k on CPU)",
+    )
     p.add_argument("--vocab", type=int, default=50_000)
     p.add_argument("--device", type=str, default=None)
     p.add_argument(
@@ -56,6 +64,8 @@ def main() -> None:
         torch.cuda.set_device(device)
 
     os.environ["PW_DEVICE"] = str(device)
+    if args.batch is None:
+        args.batch = 4_000_000 if use_cuda else 100_000
 
     import numpy as np
 
@@ -121,63 +131,132 @@ def main() -> None:
         def reset(self):
             self.counter = 0
 
-    class ByteStream:
-        """Raw ingest: per pull, a newline-separated byte buffer is
-        tokenized and xxh-128-keyed ON DEVICE (pw_varlen_hash) — the parse
-        path a GPU-native text connector runs (message parsing included in
-        the timed region)."""
+    class HostByteStream:
+        """Honest whole-node ingest.
 
-        def __init__(self, batch: int, vocab_n: int):
+        The word stream exists as newline-separated BYTES IN HOST MEMORY
+        (pinned staging buffers — the analog of a socket / page-cache
+        buffer a real connector reads from; like the reference harness's
+        pre-created 5M-row input, the wire bytes are synthesized before
+        the timed region).  Every timed step then performs the full
+        connector pipeline:
+
+          H2D DMA (hipMemcpyAsync on a dedicated copy stream,
+          double-buffered so copy overlaps the previous step's compute)
+          -> newline scan -> varlen xxh128 token hash (HIP kernel)
+          -> dictionary lookup (sorted-pool binary-search HIP kernel)
+          -> keyed DeltaBatch.
+
+        The output `word` column is DERIVED FROM THE PARSED BYTES via the
+        dictionary lookup; the codes used to synthesize the wire buffer
+        never enter the engine.  PCIe bytes are billed inside the timed
+        region (batch*(WW+1) bytes per step per GPU).
+        """
+
+        def __init__(self, batch: int, vocab_n: int, pool_buffers: int = 6):
             self.batch = batch
             self.vocab_n = vocab_n
             self.counter = 0
-            # device matrix of the vocabulary's word bytes (fixed width WW)
-            wb = np.frombuffer(
-                "".join(f"word{i:0{DIG}d}" for i in range(vocab_n)).encode(),
-                dtype=np.uint8,
+            self.nbytes = batch * (WW + 1)
+            # ---- host wire buffers (the "input stream"), pinned ----
+            vb = np.frombuffer(
+                "".join(vocab).encode(), dtype=np.uint8
             ).reshape(vocab_n, WW)
-            self.word_bytes = torch.from_numpy(wb.copy()).to(device)
+            rng = np.random.default_rng(1234 + rank)
+            self.host_bufs = []
+            for _ in range(pool_buffers):
+                codes = rng.integers(0, vocab_n, size=batch)
+                msg = np.empty((batch, WW + 1), dtype=np.uint8)
+                msg[:, :WW] = vb[codes]
+                msg[:, WW] = 10  # newline
+                tbuf = torch.from_numpy(msg.reshape(-1))
+                if use_cuda:
+                    tbuf = tbuf.pin_memory()
+                self.host_bufs.append(tbuf)
+            # ---- device dictionary: sorted pool-hash table ----
+            from pathway_amd.engine.state import lex_sort_words
+
+            plo, phi = GLOBAL_STRING_POOL.hash_tensors(device)
+            dperm = lex_sort_words([plo, phi])
+            self.dict_lo = plo.index_select(0, dperm).contiguous()
+            self.dict_hi = phi.index_select(0, dperm).contiguous()
+            self.dict_codes = dperm.contiguous()  # sorted pos -> pool code
+            self.miss = torch.zeros((), dtype=torch.int64, device=device)
+            # ---- double-buffered staging ----
+            if use_cuda:
+                self.copy_stream = torch.cuda.Stream(device)
+                self.staged = [
+                    torch.empty(self.nbytes, dtype=torch.uint8, device=device)
+                    for _ in range(2)
+                ]
+                self.ready = [torch.cuda.Event(), torch.cuda.Event()]
+                self.free = [torch.cuda.Event(), torch.cuda.Event()]
+                for ev in self.free:
+                    ev.record()  # both slots initially writable
+                self.next_buf = 0
+                self.cur = 0
+                self._prefetch(0)
+
+        def _prefetch(self, slot: int) -> None:
+            """Issue async H2D of the next host buffer into `slot`."""
+            with torch.cuda.stream(self.copy_stream):
+                self.copy_stream.wait_event(self.free[slot])
+                self.staged[slot].copy_(
+                    self.host_bufs[self.next_buf], non_blocking=True
+                )
+                self.ready[slot].record(self.copy_stream)
+            self.next_buf = (self.next_buf + 1) % len(self.host_bufs)
 
         def next_time(self):
             return None
 
         def pull(self, t, dev):
-            from pathway_amd import ops
             from pathway_amd.engine.column import PointerColumn
             from pathway_amd.internals.api import TAG_STR
 
             n = self.batch
-            codes = torch.randint(
-                0, self.vocab_n, (n,), dtype=torch.int64, generator=gen,
-                device=gen.device,
-            ).to(dev, non_blocking=True)
-            # build the wire buffer: "<word>\n" per message (device gather)
-            msg = torch.empty((n, WW + 1), dtype=torch.uint8, device=dev)
-            msg[:, :WW] = self.word_bytes.index_select(0, codes)
-            msg[:, WW] = 10  # newline
-            buf = msg.reshape(-1).contiguous()
-            # PARSE on device: newline scan -> token [start, end) spans
-            nl = (buf == 10).nonzero(as_tuple=True)[0]
-            starts = torch.cat(
-                [torch.zeros(1, dtype=torch.int64, device=dev), nl[:-1] + 1]
-            )
-            ends = nl
-            # per-token canonical string hash == the word's group key
             if dev.type == "cuda":
-                glo, ghi = ops.varlen_hash_se_gpu(buf, starts, ends, TAG_STR)
-            else:
-                from pathway_amd.internals.api import hash128, serialize_value
+                from pathway_amd import ops
 
-                bb = buf.cpu().numpy().tobytes()
-                st = starts.cpu().tolist()
-                en = ends.cpu().tolist()
-                glo_l, ghi_l = [], []
-                for a, b in zip(st, en):
-                    l_, h_ = hash128(serialize_value(bb[a:b].decode()))
-                    glo_l.append(l_ - (1 << 64) if l_ >= 1 << 63 else l_)
-                    ghi_l.append(h_ - (1 << 64) if h_ >= 1 << 63 else h_)
-                glo = torch.tensor(glo_l, dtype=torch.int64)
-                ghi = torch.tensor(ghi_l, dtype=torch.int64)
+                cur = self.cur
+                torch.cuda.current_stream().wait_event(self.ready[cur])
+                buf = self.staged[cur]
+                # PARSE on device: newline scan -> token [start, end) spans
+                nl = (buf == 10).nonzero(as_tuple=True)[0]
+                starts = torch.cat(
+                    [torch.zeros(1, dtype=torch.int64, device=dev), nl[:-1] + 1]
+                )
+                ends = nl
+                glo, ghi = ops.varlen_hash_se_gpu(buf, starts, ends, TAG_STR)
+                # word column derived from the parsed bytes: dictionary
+                # lookup of the token hash in the sorted pool table
+                pos, found = ops.lookup_gpu(
+                    [self.dict_lo, self.dict_hi], [glo, ghi]
+                )
+                codes = self.dict_codes.index_select(
+                    0, pos.clamp(0, self.dict_codes.shape[0] - 1)
+                )
+                codes = torch.where(found, codes, torch.full_like(codes, -1))
+                self.miss += (~found).sum()
+                # this step's reads of `buf` are enqueued; allow the copy
+                # stream to overwrite the slot for step t+2
+                self.free[cur].record()
+                self.cur = 1 - cur
+                self._prefetch(1 - cur)  # overlaps with this step's compute
+            else:
+                # CPU fallback (not the measured path): numpy parse of the
+                # same wire buffer -> dictionary-encode via the string pool
+                raw = self.host_bufs[self.counter // n % len(self.host_bufs)]
+                msg = np.ascontiguousarray(
+                    raw.numpy().reshape(n, WW + 1)[:, :WW]
+                )
+                toks = msg.view(f"S{WW}").ravel()
+                words_l = [t_.decode() for t_ in toks]
+                codes_np = GLOBAL_STRING_POOL.codes(words_l)
+                codes = torch.from_numpy(codes_np)
+                plo, phi = GLOBAL_STRING_POOL.hash_tensors(dev)
+                glo = plo.index_select(0, codes)
+                ghi = phi.index_select(0, codes)
             seq = torch.arange(
                 self.counter, self.counter + n, dtype=torch.int64, device=dev
             )
@@ -195,7 +274,7 @@ def main() -> None:
             self.counter = 0
 
     if args.ingest == "bytes":
-        source = ByteStream(args.batch, args.vocab)
+        source = HostByteStream(args.batch, args.vocab)
         in_node = InputNode(source, device)
         words = Table(
             in_node, {"word": dt.STR, "wkey": dt.POINTER}, Universe()
@@ -306,6 +385,14 @@ def main() -> None:
                         "ingest": args.ingest,
                         "p95_step_latency_ms": p95,
                         "emitted_delta_rows": emitted[0],
+                        "h2d_bytes_per_step": (
+                            source.nbytes if args.ingest == "bytes" else 0
+                        ),
+                        "dict_misses": (
+                            int(source.miss.item())
+                            if args.ingest == "bytes" and use_cuda
+                            else 0
+                        ),
                     },
                 }
             )
