@@ -30,11 +30,14 @@ class StreamingSource(Source):
         column_names: list[str],
         dtypes: list[dt.DType],
         name: str | None = None,
+        maxsize: int | None = None,
     ):
         self.column_names = column_names
         self.dtypes = dtypes
         self.name = name
-        self.q: queue.Queue = queue.Queue()
+        # bounded queue = reader backpressure (reference max_backlog_size,
+        # connectors/backlog.rs): emit() blocks when the engine is behind
+        self.q: queue.Queue = queue.Queue(maxsize=maxsize or 0)
         self._finished = threading.Event()
         self._seq = 0
         self._sync_group = None
@@ -50,6 +53,10 @@ class StreamingSource(Source):
             lo, hi = hash_values([self.name or "stream", self._seq])
             key = Pointer(lo, hi)
         self.q.put((key, values, diff))
+
+    def fail(self, exc: Exception) -> None:
+        """Reader-thread error: re-raised on the engine thread at next pull."""
+        self._error = exc
 
     def finish(self):
         self._finished.set()
@@ -80,6 +87,10 @@ class StreamingSource(Source):
         return None
 
     def pull(self, time: int, device) -> DeltaBatch | None:
+        err = getattr(self, "_error", None)
+        if err is not None:
+            self._error = None
+            raise RuntimeError(f"connector reader failed: {err}") from err
         rows = list(self._held)
         self._held = []
         try:
