@@ -48,11 +48,14 @@ class StreamingSource(Source):
     # -- producer side (reader thread) --
 
     def emit(self, values: list[Any], key: BasePointer | None = None, diff: int = 1):
+        """Enqueue one update; returns the row key so readers can later
+        retract the same row (deletions must reuse the insert's key)."""
         if key is None:
             self._seq += 1
             lo, hi = hash_values([self.name or "stream", self._seq])
             key = Pointer(lo, hi)
         self.q.put((key, values, diff))
+        return key
 
     def fail(self, exc: Exception) -> None:
         """Reader-thread error: re-raised on the engine thread at next pull."""
@@ -179,9 +182,10 @@ class FilePollReader:
 
     def _retract_file(self, path: str) -> None:
         """Object deletion/modification: retract the rows previously
-        emitted for this path (reference metadata/file_like.rs tracking)."""
-        for row in self.emitted.pop(path, []):
-            self.source.emit(list(row), diff=-1)
+        emitted for this path (reference metadata/file_like.rs tracking).
+        Each retraction reuses the row's original key."""
+        for key, row in self.emitted.pop(path, []):
+            self.source.emit(list(row), key=key, diff=-1)
 
     def _emit_file(self, path: str):
         import csv as _csv
@@ -208,8 +212,8 @@ class FilePollReader:
         rows_out = self.emitted.setdefault(path, [])
 
         def emit_row(row):
-            rows_out.append(list(row))
-            self.source.emit(row)
+            key = self.source.emit(row)
+            rows_out.append((key, list(row)))
 
         if self.format == "plaintext":
             if native_io.available():

@@ -125,19 +125,19 @@ class ObjectStoreReader:
         from pathway_amd.internals.json import Json
 
         listed = dict(self.store.list(self.prefix))
-        # deletions
+        # deletions (retractions reuse each row's original key)
         for key in list(self.seen.keys()):
             if key not in listed:
-                for row in self.emitted.pop(key, []):
-                    self.source.emit(list(row), diff=-1)
+                for rk, row in self.emitted.pop(key, []):
+                    self.source.emit(list(row), key=rk, diff=-1)
                 del self.seen[key]
         # new / modified
         for key, etag in sorted(listed.items()):
             if self.seen.get(key) == etag:
                 continue
             if key in self.seen:  # modified: retract old rows first
-                for row in self.emitted.pop(key, []):
-                    self.source.emit(list(row), diff=-1)
+                for rk, row in self.emitted.pop(key, []):
+                    self.source.emit(list(row), key=rk, diff=-1)
             data = self.store.get(key)
             if data is None:
                 continue
@@ -148,8 +148,8 @@ class ObjectStoreReader:
                 rows = [r + [meta] for r in rows]
             out = self.emitted.setdefault(key, [])
             for r in rows:
-                out.append(r)
-                self.source.emit(list(r))
+                rk = self.source.emit(list(r))
+                out.append((rk, r))
             self.seen[key] = etag
 
 
