@@ -60,6 +60,12 @@ class ObjectId:
     def __eq__(self, other: object) -> bool:
         return isinstance(other, ObjectId) and self.binary == other.binary
 
+    def __lt__(self, other: "ObjectId") -> bool:
+        return self.binary < other.binary
+
+    def __gt__(self, other: "ObjectId") -> bool:
+        return self.binary > other.binary
+
     def __hash__(self) -> int:
         return hash(self.binary)
 
