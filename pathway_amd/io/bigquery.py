@@ -1,19 +1,69 @@
-"""pw.io.bigquery (reference io/bigquery) — API-parity surface.
+"""pw.io.bigquery — BigQuery sink over the tabledata.insertAll REST API.
 
-Requires the google client library (offline image: raises at call time).
+Reference: src/connectors/data_storage/bigquery.rs (gcp-bigquery-client).
+Posts standard streaming-insert requests; credentials are a bearer token
+or a service-account object with a token attribute.
 """
+
 from __future__ import annotations
 
 from typing import Any
 
-from pathway_amd.io._utils import require_client
+from pathway_amd.io import _rest
+
+DEFAULT_BASE = "https://bigquery.googleapis.com/bigquery/v2"
 
 
-def read(*args: Any, schema=None, mode: str = "streaming", name: str | None = None, autocommit_duration_ms: int | None = 1500, **kwargs: Any):
-    require_client("google.cloud.bigquery", "bigquery")
-    raise NotImplementedError("pw.io.bigquery.read: client library loaded but offline transport is unavailable in this environment")
+def write(
+    table,
+    dataset_name: str,
+    table_name: str,
+    *,
+    project_id: str | None = None,
+    service_user_credentials_file: str | None = None,
+    credentials: Any = None,
+    base_url: str = DEFAULT_BASE,
+    name: str | None = None,
+    **kwargs: Any,
+):
+    from pathway_amd.engine.runtime import OutputNode
+    from pathway_amd.internals.config import get_device
+    from pathway_amd.internals.json import Json
+    from pathway_amd.internals.rungraph import G
 
+    if project_id is None and service_user_credentials_file:
+        import json as _json
 
-def write(table, *args: Any, name: str | None = None, **kwargs: Any):
-    require_client("google.cloud.bigquery", "bigquery")
-    raise NotImplementedError("pw.io.bigquery.write: client library loaded but offline transport is unavailable in this environment")
+        with open(service_user_credentials_file) as f:
+            project_id = _json.load(f).get("project_id")
+    headers = {}
+    token = getattr(credentials, "token", None) or (
+        credentials if isinstance(credentials, str) else None
+    )
+    if token:
+        headers["Authorization"] = f"Bearer {token}"
+    url = (f"{base_url}/projects/{project_id}/datasets/{dataset_name}"
+           f"/tables/{table_name}/insertAll")
+    names = table.column_names()
+
+    def writer(batch):
+        rows = []
+        for key, values, time, diff in batch.rows():
+            rec = {}
+            for n, v in zip(names, values):
+                rec[n] = v.value if isinstance(v, Json) else v
+            rec["time"] = time
+            rec["diff"] = diff
+            rows.append({"insertId": f"{key!r}-{time}-{diff}", "json": rec})
+        if rows:
+            out = _rest.request(
+                "POST", url, body={"kind": "bigquery#tableDataInsertAllRequest",
+                                   "rows": rows},
+                headers=headers,
+            )
+            if out and out.get("insertErrors"):
+                raise RuntimeError(f"bigquery insert errors: {out['insertErrors']}")
+
+    node = OutputNode(table._node, writer, get_device())
+    G.add_sink(node)
+    return node

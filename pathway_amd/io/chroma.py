@@ -1,14 +1,40 @@
-"""pw.io.chroma (reference io/chroma) — API-parity surface.
+"""pw.io.chroma — Chroma sink over the v1 collections REST API
+(reference src/connectors/data_storage/chroma.rs, 499 LoC)."""
 
-Requires the chromadb client library (offline image: raises at call time).
-"""
 from __future__ import annotations
 
 from typing import Any
 
-from pathway_amd.io._utils import require_client
+from pathway_amd.io import _rest
+from pathway_amd.io._vector_sink import make_vector_writer
 
 
-def write(table, *args: Any, name: str | None = None, **kwargs: Any):
-    require_client("chroma", "chromadb")
-    raise NotImplementedError("pw.io.chroma.write: client library loaded but offline transport is unavailable in this environment")
+def write(
+    table,
+    url: str,
+    collection_id: str,
+    *,
+    vector_column: str = "vector",
+    document_column: str | None = None,
+    name: str | None = None,
+    **kwargs: Any,
+):
+    base = url.rstrip("/")
+    coll = f"{base}/api/v1/collections/{collection_id}"
+
+    def upsert(points):
+        body = {
+            "ids": [p["id"] for p in points],
+            "embeddings": [p["vector"] for p in points],
+            "metadatas": [p["metadata"] for p in points],
+        }
+        if document_column:
+            body["documents"] = [
+                p["metadata"].get(document_column) for p in points
+            ]
+        _rest.request("POST", f"{coll}/upsert", body=body)
+
+    def delete(ids):
+        _rest.request("POST", f"{coll}/delete", body={"ids": ids})
+
+    return make_vector_writer(table, vector_column, upsert=upsert, delete=delete)
