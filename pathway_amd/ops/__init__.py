@@ -412,3 +412,27 @@ def seg_reduce_gpu(
     """2-word convenience wrapper over seg_reduce_words_gpu."""
     out_words, out_first, out_accs = seg_reduce_words_gpu([k0, k1], contribs)
     return out_words[0], out_words[1], out_first, out_accs
+
+
+def partition_gpu(dest: torch.Tensor, world: int) -> tuple[torch.Tensor, torch.Tensor]:
+    """Radix partition by destination rank: returns (perm, counts) with rows
+    grouped by destination (exchange shuffle pack; pact.rs:56 analog)."""
+    lib = require_lib()
+    n = dest.shape[0]
+    nblocks = max(1, min(2048, (n + 255) // 256))
+    perm = torch.empty(n, dtype=torch.int64, device=dest.device)
+    counts = torch.empty(world, dtype=torch.int64, device=dest.device)
+    scratch = torch.empty(nblocks * world, dtype=torch.int64, device=dest.device)
+    rc = lib.pw_partition(
+        ctypes.c_void_p(dest.contiguous().data_ptr()),
+        ctypes.c_int64(n),
+        ctypes.c_int(world),
+        ctypes.c_void_p(perm.data_ptr()),
+        ctypes.c_void_p(counts.data_ptr()),
+        ctypes.c_void_p(scratch.data_ptr()),
+        ctypes.c_int64(nblocks),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_partition failed: hip error {rc}")
+    return perm, counts

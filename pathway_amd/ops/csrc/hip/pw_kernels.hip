@@ -695,3 +695,89 @@ extern "C" int pw_seg_reduce_emit(const void** word_ptrs, int nw,
                      ow, (long long*)out_first, op);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------------------- partition --
+// Radix partition of rows by destination rank (the exchange shuffle pack,
+// reference pact.rs:56 / operators.rs:126).  dest[i] in [0, world);
+// world <= 64.  Output: perm such that rows grouped by destination
+// (sorted[j] = input[perm[j]]) and per-destination counts.  Order within a
+// destination is unspecified (the exchange does not require stability).
+//
+// Two passes over dest, each block owning a contiguous chunk:
+//   1. per-block histogram -> block_hist[b][d]
+//   2. single-block exclusive scan in destination-major order gives each
+//      (d, b) cell its global base
+//   3. scatter: per-block shared counters (atomicAdd in LDS) assign
+//      positions within the block's (d) quota.
+
+#define PW_PART_MAX_DEST 64
+
+__global__ void k_partition_count(const int64_t* dest, int64_t n,
+                                  int64_t chunk, int world,
+                                  long long* block_hist) {
+  __shared__ int hist[PW_PART_MAX_DEST];
+  for (int d = threadIdx.x; d < world; d += blockDim.x) hist[d] = 0;
+  __syncthreads();
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t end = min(start + chunk, n);
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+    atomicAdd(&hist[(int)dest[i]], 1);
+  __syncthreads();
+  for (int d = threadIdx.x; d < world; d += blockDim.x)
+    block_hist[(int64_t)blockIdx.x * world + d] = hist[d];
+}
+
+// exclusive scan over cells ordered (d major, b minor); also emits counts[d]
+__global__ void k_partition_scan(long long* block_hist, int64_t nblocks,
+                                 int world, long long* counts) {
+  // single block; simple sequential scan (cells <= 64 * 4096)
+  if (threadIdx.x == 0) {
+    long long run = 0;
+    for (int d = 0; d < world; ++d) {
+      long long dtotal = 0;
+      for (int64_t b = 0; b < nblocks; ++b) {
+        long long c = block_hist[b * world + d];
+        block_hist[b * world + d] = run;
+        run += c;
+        dtotal += c;
+      }
+      counts[d] = dtotal;
+    }
+  }
+}
+
+__global__ void k_partition_scatter(const int64_t* dest, int64_t n,
+                                    int64_t chunk, int world,
+                                    const long long* block_bases,
+                                    int64_t* perm) {
+  __shared__ unsigned int offs[PW_PART_MAX_DEST];
+  for (int d = threadIdx.x; d < world; d += blockDim.x) offs[d] = 0;
+  __syncthreads();
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t end = min(start + chunk, n);
+  const long long* base = block_bases + (int64_t)blockIdx.x * world;
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+    int d = (int)dest[i];
+    unsigned int p = atomicAdd(&offs[d], 1u);
+    perm[base[d] + p] = i;
+  }
+}
+
+extern "C" int pw_partition(const void* dest, int64_t n, int world,
+                            void* perm, void* counts, void* scratch,
+                            int64_t nblocks, void* stream) {
+  if (world > PW_PART_MAX_DEST) return 2;
+  hipStream_t s = (hipStream_t)stream;
+  int64_t chunk = (n + nblocks - 1) / nblocks;
+  if (chunk < 1) chunk = 1;
+  hipLaunchKernelGGL(k_partition_count, dim3((uint32_t)nblocks),
+                     dim3(PW_BLOCK), 0, s, (const int64_t*)dest, n, chunk,
+                     world, (long long*)scratch);
+  hipLaunchKernelGGL(k_partition_scan, dim3(1), dim3(1), 0, s,
+                     (long long*)scratch, nblocks, world,
+                     (long long*)counts);
+  hipLaunchKernelGGL(k_partition_scatter, dim3((uint32_t)nblocks),
+                     dim3(PW_BLOCK), 0, s, (const int64_t*)dest, n, chunk,
+                     world, (const long long*)scratch, (int64_t*)perm);
+  return (int)hipGetLastError();
+}

@@ -123,8 +123,15 @@ def exchange_bundle(
         tensors, columns = _empty_bundle_from_meta(merged, comm.device)
         dest = torch.zeros((0,), dtype=torch.int64, device=comm.device)
     columns = columns or {}
-    perm = torch.argsort(dest)
-    counts = torch.bincount(dest, minlength=world)
+    if dest.is_cuda:
+        # hand-written HIP radix partition (histogram + scan + scatter) —
+        # replaces torch.argsort + bincount
+        from pathway_amd import ops
+
+        perm, counts = ops.partition_gpu(dest, world)
+    else:
+        perm = torch.argsort(dest)
+        counts = torch.bincount(dest, minlength=world)
     out_tensors: dict[str, torch.Tensor] = {}
     for name, t in tensors.items():
         out_tensors[name] = comm.all_to_all_tensor(

@@ -412,3 +412,24 @@ def test_seg_reduce_matches_reference():
         ref = torch.zeros(nseg, dtype=torch.int64, device=dev)
         ref.index_add_(0, seg, src)
         assert torch.equal(got, ref)
+
+
+@gpu
+@requires_cuda
+def test_partition_kernel_vs_argsort():
+    from pathway_amd import ops
+
+    torch.manual_seed(11)
+    for n, world in [(0, 8), (1, 2), (1000, 3), (1 << 20, 8), (12345, 64)]:
+        dest = torch.randint(0, world, (n,), dtype=torch.int64, device="cuda")
+        perm, counts = ops.partition_gpu(dest, world)
+        ref_counts = torch.bincount(dest.cpu(), minlength=world)
+        assert torch.equal(counts.cpu(), ref_counts), (n, world)
+        # perm is a permutation and groups rows by destination
+        sorted_dest = dest.index_select(0, perm)
+        assert torch.equal(
+            sorted_dest.cpu(), torch.sort(dest.cpu()).values
+        ), (n, world)
+        assert torch.equal(
+            torch.sort(perm.cpu()).values, torch.arange(n, dtype=torch.int64)
+        ), (n, world)
