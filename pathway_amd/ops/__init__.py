@@ -436,3 +436,39 @@ def partition_gpu(dest: torch.Tensor, world: int) -> tuple[torch.Tensor, torch.T
     if rc != 0:
         raise RuntimeError(f"pw_partition failed: hip error {rc}")
     return perm, counts
+
+
+def gemm_bias_act_gpu(
+    a: torch.Tensor,
+    b: torch.Tensor,
+    bias: torch.Tensor | None = None,
+    act: str = "none",
+) -> torch.Tensor:
+    """Hand-written MFMA bf16 GEMM with fused bias(+GELU) epilogue.
+
+    a (M,K) bf16, b (K,N) bf16 row-major; bias (N,) float32; returns
+    (M,N) bf16.  mfma_f32_16x16x32_bf16 tiles, fp32 accumulate.
+    """
+    lib = require_lib()
+    assert a.dtype == torch.bfloat16 and b.dtype == torch.bfloat16
+    M, K = a.shape
+    K2, N = b.shape
+    assert K == K2
+    out = torch.empty((M, N), dtype=torch.bfloat16, device=a.device)
+    if bias is not None:
+        bias = bias.to(torch.float32).contiguous()
+    act_code = {"none": 0, "gelu": 1}[act]
+    rc = lib.pw_gemm_bf16(
+        ctypes.c_void_p(a.contiguous().data_ptr()),
+        ctypes.c_void_p(b.contiguous().data_ptr()),
+        ctypes.c_void_p(bias.data_ptr() if bias is not None else 0),
+        ctypes.c_void_p(out.data_ptr()),
+        ctypes.c_int64(M),
+        ctypes.c_int64(N),
+        ctypes.c_int64(K),
+        ctypes.c_int(act_code),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_gemm_bf16 failed: hip error {rc}")
+    return out

@@ -781,3 +781,158 @@ extern "C" int pw_partition(const void* dest, int64_t n, int world,
                      world, (const long long*)scratch, (int64_t*)perm);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------------------ MFMA GEMM --
+// Hand-written bf16 GEMM on gfx950 matrix cores with fused bias +
+// activation epilogue — the embedder's projection/FFN GEMMs (north-star:
+// "embedder forward on bf16 MFMA tiles").  C (M,N) = A (M,K) x B (K,N),
+// row-major bf16 inputs, fp32 accumulate via mfma_f32_16x16x32_bf16,
+// bf16 store.  Block = 128x128 tile, 4 waves (2x2), each wave owns a
+// 64x64 sub-tile = 4x4 fragments; K staged through LDS in BK=64 steps,
+// double-buffered.  act: 0 = none, 1 = exact GELU.
+//
+// Fragment layouts (cdna_hip_programming.md §3, verified by the gpu test
+// against torch fp32 matmul with asymmetric inputs):
+//   A: lane holds A[row0 + (l&15)][k0 + (l>>4)*8 + j], j=0..7
+//   B: lane holds B[k0 + (l>>4)*8 + j][col0 + (l&15)]
+//   C/D: lane l reg r -> row = row0 + (l>>4)*4 + r, col = col0 + (l&15)
+
+typedef __attribute__((ext_vector_type(8))) __bf16 pw_frag8;
+typedef __attribute__((ext_vector_type(4))) float pw_frag4f;
+
+#define PW_GEMM_BM 128
+#define PW_GEMM_BN 128
+#define PW_GEMM_BK 64
+
+__device__ __forceinline__ float pw_gelu(float x) {
+  return 0.5f * x * (1.0f + erff(x * 0.70710678118654752f));
+}
+
+__global__ __launch_bounds__(256) void k_gemm_bf16(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    const float* __restrict__ bias, __bf16* __restrict__ C, int64_t M,
+    int64_t N, int64_t K, int act, int nbm) {
+  __shared__ __bf16 lasA[2][PW_GEMM_BM][PW_GEMM_BK];
+  // B stored TRANSPOSED in LDS ([col][k]) so each lane's 8-element
+  // K-fragment is a contiguous ds_read_b128 (scattered writes happen
+  // once per element; scattered reads would repeat per MFMA)
+  __shared__ __bf16 lasBt[2][PW_GEMM_BN][PW_GEMM_BK];
+  // XCD-aware tile mapping: consecutive blockIdx tiles land on the same
+  // XCD's L2 by swizzling in chunks (guide: blockIdx->tile must be
+  // XCD-aware).  Simple bijective row-major with N-fastest is fine here
+  // because N <= 12 tiles for the encoder shapes.
+  int tile = blockIdx.x;
+  int bm = tile % nbm;
+  int bn = tile / nbm;
+  int64_t row0 = (int64_t)bm * PW_GEMM_BM;
+  int64_t col0 = (int64_t)bn * PW_GEMM_BN;
+  int tid = threadIdx.x;
+  int lane = tid & 63;
+  int wave = tid >> 6;  // 0..3
+  int wr = wave >> 1;   // wave row (0..1) -> 64-row strip
+  int wc = wave & 1;    // wave col (0..1) -> 64-col strip
+
+  pw_frag4f acc[4][4];
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+
+  int nk = (int)((K + PW_GEMM_BK - 1) / PW_GEMM_BK);
+
+  // cooperative tile load: 256 threads, A tile 128x64 (8192 elems = 32/thread),
+  // 8-wide vector loads -> 4 loads per thread
+  auto load_tiles = [&](int kt, int buf) {
+    int64_t k0 = (int64_t)kt * PW_GEMM_BK;
+    // A: each thread loads 4 rows' 8-element chunks
+    for (int it = tid; it < (PW_GEMM_BM * PW_GEMM_BK) / 8; it += 256) {
+      int r = it / (PW_GEMM_BK / 8);
+      int cchunk = it % (PW_GEMM_BK / 8);
+      int64_t grow = row0 + r;
+      int64_t gcol = k0 + cchunk * 8;
+      __bf16* dst = &lasA[buf][r][cchunk * 8];
+      if (grow < M && gcol + 8 <= K) {
+        *(pw_frag8*)dst = *(const pw_frag8*)&A[grow * K + gcol];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          dst[j] = (grow < M && gcol + j < K) ? A[grow * K + gcol + j]
+                                              : (__bf16)0.f;
+      }
+    }
+    // B: tile 64x128
+    for (int it = tid; it < (PW_GEMM_BK * PW_GEMM_BN) / 8; it += 256) {
+      int r = it / (PW_GEMM_BN / 8);
+      int cchunk = it % (PW_GEMM_BN / 8);
+      int64_t grow = k0 + r;
+      int64_t gcol = col0 + cchunk * 8;
+      if (grow < K && gcol + 8 <= N) {
+        pw_frag8 v = *(const pw_frag8*)&B[grow * N + gcol];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) lasBt[buf][cchunk * 8 + j][r] = v[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lasBt[buf][cchunk * 8 + j][r] =
+              (grow < K && gcol + j < N) ? B[grow * N + gcol + j]
+                                         : (__bf16)0.f;
+      }
+    }
+  };
+
+  load_tiles(0, 0);
+  __syncthreads();
+  for (int kt = 0; kt < nk; ++kt) {
+    int buf = kt & 1;
+    if (kt + 1 < nk) load_tiles(kt + 1, buf ^ 1);
+#pragma unroll
+    for (int kk = 0; kk < PW_GEMM_BK / 32; ++kk) {
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int arow = wr * 64 + m * 16 + (lane & 15);
+        int ak = kk * 32 + (lane >> 4) * 8;
+        pw_frag8 afrag = *(const pw_frag8*)&lasA[buf][arow][ak];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          int bcol = wc * 64 + n * 16 + (lane & 15);
+          int bk = kk * 32 + (lane >> 4) * 8;
+          pw_frag8 bfrag = *(const pw_frag8*)&lasBt[buf][bcol][bk];
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, bfrag, acc[m][n], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: bias + activation, bf16 store
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      int64_t col = col0 + wc * 64 + n * 16 + (lane & 15);
+      if (col >= N) continue;
+      float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int64_t row = row0 + wr * 64 + m * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r] + bv;
+        if (act == 1) v = pw_gelu(v);
+        C[row * N + col] = (__bf16)v;
+      }
+    }
+  }
+}
+
+extern "C" int pw_gemm_bf16(const void* A, const void* B, const void* bias,
+                            void* C, int64_t M, int64_t N, int64_t K,
+                            int act, void* stream) {
+  int nbm = (int)((M + PW_GEMM_BM - 1) / PW_GEMM_BM);
+  int nbn = (int)((N + PW_GEMM_BN - 1) / PW_GEMM_BN);
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_gemm_bf16, dim3((uint32_t)(nbm * nbn)), dim3(256), 0,
+                     s, (const __bf16*)A, (const __bf16*)B,
+                     (const float*)bias, (__bf16*)C, M, N, K, act, nbm);
+  return (int)hipGetLastError();
+}
