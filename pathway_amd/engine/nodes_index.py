@@ -35,77 +35,16 @@ def _as_matrix(col, device, dim: int | None = None) -> torch.Tensor:
     return torch.from_numpy(np.stack(arrs)).to(device)
 
 
-class VectorIndexState:
-    """GPU-resident brute-force KNN index with incremental add/remove."""
+from pathway_amd.engine.ann import FlatIndexState, IvfFlatState, LshState
 
-    def __init__(self, device, metric: str = "cos"):
-        self.device = device
-        self.metric = metric
-        self.keys = torch.zeros((0, 2), dtype=torch.int64, device=device)
-        self.vectors: torch.Tensor | None = None  # (m, d) f32
-        self.payload: dict[tuple[int, int], Any] = {}  # key -> filter data
+#: backwards-compatible name: the flat index with device tombstones
+VectorIndexState = FlatIndexState
 
-    def __len__(self) -> int:
-        return int(self.keys.shape[0])
-
-    def update(self, keys: torch.Tensor, vecs: torch.Tensor, diffs: torch.Tensor,
-               payloads: list[Any] | None = None):
-        adds = (diffs > 0).nonzero(as_tuple=True)[0]
-        dels = (diffs < 0).nonzero(as_tuple=True)[0]
-        if dels.numel():
-            del_keys = keys.index_select(0, dels)
-            # remove rows whose key matches any deleted key (host map — index
-            # mutation is control-plane rate)
-            dk = {tuple(k) for k in del_keys.cpu().tolist()}
-            if len(self):
-                keep_mask_host = [
-                    tuple(k) not in dk for k in self.keys.cpu().tolist()
-                ]
-                keep = torch.tensor(keep_mask_host, dtype=torch.bool, device=self.device)
-                kidx = keep.nonzero(as_tuple=True)[0]
-                self.keys = self.keys.index_select(0, kidx)
-                if self.vectors is not None:
-                    self.vectors = self.vectors.index_select(0, kidx)
-            for k in dk:
-                self.payload.pop(k, None)
-        if adds.numel():
-            add_keys = keys.index_select(0, adds)
-            add_vecs = vecs.index_select(0, adds)
-            if self.metric == "cos":
-                add_vecs = torch.nn.functional.normalize(add_vecs, dim=1, eps=1e-12)
-            self.keys = torch.cat([self.keys, add_keys])
-            self.vectors = (
-                add_vecs
-                if self.vectors is None or self.vectors.shape[0] == 0
-                else torch.cat([self.vectors, add_vecs])
-            )
-            if payloads is not None:
-                for i, k in zip(adds.cpu().tolist(), add_keys.cpu().tolist()):
-                    self.payload[tuple(k)] = payloads[i]
-
-    def search(self, q: torch.Tensor, k: int, filter_fns: list | None = None):
-        """q: (nq, d) f32 → (ids (nq,k,2) int64, scores (nq,k) f32, valid)."""
-        nq = q.shape[0]
-        m = len(self)
-        if m == 0 or self.vectors is None:
-            return (
-                torch.zeros((nq, 0, 2), dtype=torch.int64, device=self.device),
-                torch.zeros((nq, 0), dtype=torch.float32, device=self.device),
-                torch.zeros((nq, 0), dtype=torch.bool, device=self.device),
-            )
-        if self.metric == "cos":
-            qn = torch.nn.functional.normalize(q, dim=1, eps=1e-12)
-            scores = qn @ self.vectors.T  # higher = better
-        else:  # l2sq
-            # -||q - x||^2 so that higher = better uniformly
-            q2 = (q * q).sum(1, keepdim=True)
-            x2 = (self.vectors * self.vectors).sum(1)
-            scores = -(q2 + x2.unsqueeze(0) - 2.0 * (q @ self.vectors.T))
-        kk = min(k, m)
-        top_scores, top_idx = torch.topk(scores, kk, dim=1)
-        ids = self.keys.index_select(0, top_idx.reshape(-1)).reshape(nq, kk, 2)
-        valid = torch.ones((nq, kk), dtype=torch.bool, device=self.device)
-        return ids, top_scores, valid
+INDEX_KINDS = {
+    "flat": FlatIndexState,
+    "ivf": IvfFlatState,
+    "lsh": LshState,
+}
 
 
 class ExternalIndexNode(Node):
@@ -125,6 +64,8 @@ class ExternalIndexNode(Node):
         filter_data_col: str | None = None,
         query_filter_expr: Any | None = None,
         query_k_expr: Any | None = None,
+        index_kind: str = "flat",
+        index_params: dict | None = None,
     ):
         super().__init__([index_node, query_node], device)
         self.index_vec_col = index_vec_col
@@ -134,11 +75,15 @@ class ExternalIndexNode(Node):
         self.filter_data_col = filter_data_col
         self.query_filter_expr = query_filter_expr
         self.query_k_expr = query_k_expr
-        self.state = VectorIndexState(device, metric)
+        self.index_kind = index_kind
+        self.index_params = index_params or {}
+        self.state = INDEX_KINDS[index_kind](device, metric, **self.index_params)
         self.answers: dict[tuple[int, int], tuple] = {}  # query key -> values
 
     def reset(self):
-        self.state = VectorIndexState(self.device, self.metric)
+        self.state = INDEX_KINDS[self.index_kind](
+            self.device, self.metric, **self.index_params
+        )
         self.answers = {}
 
     def step(self, time, inputs):

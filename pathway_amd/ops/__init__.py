@@ -472,3 +472,29 @@ def gemm_bias_act_gpu(
     if rc != 0:
         raise RuntimeError(f"pw_gemm_bf16 failed: hip error {rc}")
     return out
+
+
+def topk_gpu(scores: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:
+    """Per-row top-k (higher=better) via the hand-written HIP kernel.
+
+    scores (nq, m) float32 -> (vals (nq, k) f32, idx (nq, k) int64).
+    """
+    lib = require_lib()
+    nq, m = scores.shape
+    assert scores.dtype == torch.float32
+    vals = torch.empty((nq, k), dtype=torch.float32, device=scores.device)
+    idx = torch.empty((nq, k), dtype=torch.int64, device=scores.device)
+    if nq == 0:
+        return vals, idx
+    rc = lib.pw_topk(
+        ctypes.c_void_p(scores.contiguous().data_ptr()),
+        ctypes.c_int64(nq),
+        ctypes.c_int64(m),
+        ctypes.c_int(k),
+        ctypes.c_void_p(vals.data_ptr()),
+        ctypes.c_void_p(idx.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_topk failed: hip error {rc}")
+    return vals, idx

@@ -936,3 +936,97 @@ extern "C" int pw_gemm_bf16(const void* A, const void* B, const void* bias,
                      (const float*)bias, (__bf16*)C, M, N, K, act, nbm);
   return (int)hipGetLastError();
 }
+
+// ----------------------------------------------------------------- top-k --
+// Per-query top-k over a (nq, m) score matrix (higher = better) — the
+// KNN answer selection (north-star: "cosine top-k" hand-written;
+// replaces torch.topk on the query path).  One block per query; each
+// thread keeps a k-deep insertion list over its strided slice, lists are
+// merged in LDS by k selection passes.  k <= 32.
+
+#define PW_TOPK_MAXK 32
+
+__global__ __launch_bounds__(256) void k_topk(
+    const float* __restrict__ scores, int64_t m, int k,
+    float* __restrict__ out_vals, int64_t* __restrict__ out_idx) {
+  const float NEG = -3.4e38f;
+  __shared__ float svals[256 * PW_TOPK_MAXK];
+  __shared__ int sidx[256 * PW_TOPK_MAXK];
+  int64_t q = blockIdx.x;
+  const float* row = scores + q * m;
+  int tid = threadIdx.x;
+  float lv[PW_TOPK_MAXK];
+  int li[PW_TOPK_MAXK];
+#pragma unroll
+  for (int j = 0; j < PW_TOPK_MAXK; ++j) {
+    lv[j] = NEG;
+    li[j] = -1;
+  }
+  for (int64_t i = tid; i < m; i += 256) {
+    float v = row[i];
+    if (v > lv[k - 1]) {
+      // insertion into the local sorted-descending list
+      int p = k - 1;
+      while (p > 0 && lv[p - 1] < v) {
+        lv[p] = lv[p - 1];
+        li[p] = li[p - 1];
+        --p;
+      }
+      lv[p] = v;
+      li[p] = (int)i;
+    }
+  }
+  for (int j = 0; j < k; ++j) {
+    svals[tid * k + j] = lv[j];
+    sidx[tid * k + j] = li[j];
+  }
+  __syncthreads();
+  // selection: k passes; wave 0 finds the global max of the 256*k
+  // candidates via wave reduction and invalidates it.  The barrier is
+  // block-uniform (every thread executes it each pass).
+  for (int sel = 0; sel < k; ++sel) {
+    if (tid < 64) {
+      float best = NEG;
+      int bestpos = -1;
+      for (int p = tid; p < 256 * k; p += 64) {
+        float v = svals[p];
+        if (v > best || (v == best && bestpos >= 0 && p < bestpos)) {
+          best = v;
+          bestpos = p;
+        }
+      }
+      // wave argmax reduction (ties -> lower position for determinism)
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) {
+        float ov = __shfl_down(best, off);
+        int op = __shfl_down(bestpos, off);
+        if (ov > best || (ov == best && op >= 0 &&
+                          (bestpos < 0 || op < bestpos))) {
+          best = ov;
+          bestpos = op;
+        }
+      }
+      if (tid == 0) {
+        if (bestpos >= 0 && best > NEG) {
+          out_vals[q * k + sel] = best;
+          out_idx[q * k + sel] = sidx[bestpos];
+          svals[bestpos] = NEG;
+        } else {
+          out_vals[q * k + sel] = NEG;
+          out_idx[q * k + sel] = -1;
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" int pw_topk(const void* scores, int64_t nq, int64_t m, int k,
+                       void* out_vals, void* out_idx, void* stream) {
+  if (k > PW_TOPK_MAXK || k < 1) return 2;
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_topk, dim3((uint32_t)nq), dim3(256), 0, s,
+                     (const float*)scores, m, k, (float*)out_vals,
+                     (int64_t*)out_idx);
+  return (int)hipGetLastError();
+}

@@ -1,9 +1,14 @@
 """KNN retrievers (reference stdlib/indexing/nearest_neighbors.py:65-574).
 
-BruteForceKnn: GPU brute-force cosine/L2 top-k — queries × index GEMM
-(hipBLASLt via torch on ROCm) + top-k, over the ExternalIndexNode.
-USearchKnn: HNSW-parity API; round 1 serves it from the same brute-force
-GPU index (exact ≥ approximate recall; perf via GEMM).  LshKnn likewise.
+BruteForceKnn: GPU brute-force cosine/L2 — queries × index GEMM + the
+hand-written pw_topk kernel, over the ExternalIndexNode.
+USearchKnn: the approximate index (reference: usearch HNSW,
+usearch_integration.rs:20-152).  The MI355X-native approximate engine is
+GPU IVF-Flat (engine/ann.py IvfFlatState): k-means coarse quantizer +
+nprobe candidate lists + exact rerank — same DataIndex API and as-of-now
+semantics; recall/perf measured in profiles/ann_r02.md.
+LshKnn: real random-hyperplane LSH with multi-table buckets
+(engine/ann.py LshState; reference _lsh.py semantics).
 """
 
 from __future__ import annotations
@@ -31,17 +36,22 @@ class USearchMetricKind(enum.Enum):
 class _BruteForceIndexBase:
     """InnerIndex implementation over ExternalIndexNode."""
 
+    #: engine/ann.py state class selector
+    index_kind = "flat"
+
     def __init__(
         self,
         data_column: ex.ColumnReference,
         metadata_column: ex.ColumnReference | None = None,
         metric: DistanceType = DistanceType.COS,
         embedder: Any = None,
+        index_params: dict | None = None,
     ):
         self.data_column = data_column
         self.metadata_column = metadata_column
         self.metric = metric
         self.embedder = embedder
+        self.index_params = index_params or {}
 
     def query_as_of_now(
         self,
@@ -89,6 +99,8 @@ class _BruteForceIndexBase:
             filter_data_col=filter_col,
             query_filter_expr=metadata_filter,
             query_k_expr=kexpr,
+            index_kind=self.index_kind,
+            index_params=self.index_params,
         )
         dtypes = {
             "_pw_index_reply_ids": dt.List(dt.POINTER),
@@ -107,12 +119,18 @@ class BruteForceKnn(_BruteForceIndexBase):
 
 
 class USearchKnn(_BruteForceIndexBase):
-    """HNSW-parity API (reference usearch_integration.rs:20-152); exact GPU
-    brute-force under the hood in round 1."""
+    """Approximate KNN with the usearch-factory API (reference
+    usearch_integration.rs:20-152).  Engine: GPU IVF-Flat
+    (engine/ann.py) — k-means lists + nprobe + exact rerank."""
+
+    index_kind = "ivf"
 
 
 class LshKnn(_BruteForceIndexBase):
-    pass
+    """Random-hyperplane LSH (reference _lsh.py): n_or tables x n_and
+    bits, bucket candidates, exact rerank."""
+
+    index_kind = "lsh"
 
 
 @dataclass
@@ -139,7 +157,11 @@ class UsearchKnnFactory:
 
     def build_index(self, data_column, metadata_column=None, **kwargs) -> USearchKnn:
         m = DistanceType.COS if self.metric in (USearchMetricKind.COS, USearchMetricKind.IP) else DistanceType.L2SQ
-        return USearchKnn(data_column, metadata_column, m, self.embedder)
+        params = {}
+        if self.expansion_search:
+            params["nprobe"] = max(1, int(self.expansion_search))
+        return USearchKnn(data_column, metadata_column, m, self.embedder,
+                          index_params=params)
 
 
 @dataclass
@@ -152,7 +174,10 @@ class LshKnnFactory:
     embedder: Any = None
 
     def build_index(self, data_column, metadata_column=None, **kwargs) -> LshKnn:
-        return LshKnn(data_column, metadata_column, self.distance_type, self.embedder)
+        return LshKnn(
+            data_column, metadata_column, self.distance_type, self.embedder,
+            index_params={"n_or": self.n_or, "n_and": self.n_and},
+        )
 
 
 class DefaultKnnFactory(BruteForceKnnFactory):

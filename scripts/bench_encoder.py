@@ -4,7 +4,10 @@ Run on MI355X: python scripts/bench_encoder.py
 """
 
 import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -501,3 +501,50 @@ def test_mfma_gemm_vs_torch():
         got_g = ops.gemm_bias_act_gpu(a, b, bias, act="gelu").float()
         errg = (got_g - ref_g).abs().max().item()
         assert errg / (ref_g.abs().max().item() + 1.0) < 0.02, (M, N, K, errg)
+
+
+@gpu
+@requires_cuda
+def test_topk_kernel_vs_torch():
+    from pathway_amd import ops
+
+    torch.manual_seed(13)
+    for nq, m, k in [(1, 100, 1), (7, 1000, 10), (64, 65536, 32), (3, 10, 8)]:
+        scores = torch.randn(nq, m, device="cuda")
+        vals, idx = ops.topk_gpu(scores, min(k, m))
+        rv, ri = torch.topk(scores, min(k, m), dim=1)
+        assert torch.allclose(vals, rv), (nq, m, k)
+        # indices may differ on exact ties; values fully determine correctness
+        gathered = scores.gather(1, idx)
+        assert torch.allclose(gathered, rv), (nq, m, k)
+
+
+@gpu
+@requires_cuda
+def test_ivf_gpu_recall_at_10():
+    from pathway_amd.engine.ann import IvfFlatState
+
+    torch.manual_seed(21)
+    n, d, k = 200_000, 384, 10
+    centers = (torch.randn(512, d, device="cuda") * 3.0)
+    assign = torch.randint(0, 512, (n,), device="cuda")
+    vecs = centers[assign] + 0.3 * torch.randn(n, d, device="cuda")
+    keys = torch.stack([
+        torch.arange(1, n + 1, dtype=torch.int64, device="cuda"),
+        torch.zeros(n, dtype=torch.int64, device="cuda"),
+    ], dim=1)
+    st = IvfFlatState("cuda", "cos", nprobe=16)
+    st.update(keys, vecs, torch.ones(n, dtype=torch.int64, device="cuda"))
+    assert st.centroids is not None
+    qa = torch.randint(0, 512, (100,), device="cuda")
+    q = centers[qa] + 0.3 * torch.randn(100, d, device="cuda")
+    vn = torch.nn.functional.normalize(vecs, dim=1)
+    ref = torch.topk(torch.nn.functional.normalize(q, dim=1) @ vn.T, k, 1).indices
+    ids, _, _ = st.search(q, k)
+    got = (ids[:, :, 0] - 1).cpu()
+    refc = ref.cpu()
+    recall = sum(
+        len(set(refc[i].tolist()) & set(got[i].tolist())) / k
+        for i in range(100)
+    ) / 100
+    assert recall >= 0.9, recall
