@@ -440,19 +440,20 @@ def partition_gpu(dest: torch.Tensor, world: int) -> tuple[torch.Tensor, torch.T
 
 def gemm_bias_act_gpu(
     a: torch.Tensor,
-    b: torch.Tensor,
+    b_t: torch.Tensor,
     bias: torch.Tensor | None = None,
     act: str = "none",
 ) -> torch.Tensor:
     """Hand-written MFMA bf16 GEMM with fused bias(+GELU) epilogue.
 
-    a (M,K) bf16, b (K,N) bf16 row-major; bias (N,) float32; returns
-    (M,N) bf16.  mfma_f32_16x16x32_bf16 tiles, fp32 accumulate.
+    a (M,K) bf16, b_t = B^T (N,K) bf16 row-major (weights are static, so
+    the transpose is one-time); bias (N,) float32; returns (M,N) bf16.
+    mfma_f32_16x16x32_bf16 tiles, fp32 accumulate.
     """
     lib = require_lib()
-    assert a.dtype == torch.bfloat16 and b.dtype == torch.bfloat16
+    assert a.dtype == torch.bfloat16 and b_t.dtype == torch.bfloat16
     M, K = a.shape
-    K2, N = b.shape
+    N, K2 = b_t.shape
     assert K == K2
     out = torch.empty((M, N), dtype=torch.bfloat16, device=a.device)
     if bias is not None:
@@ -460,7 +461,7 @@ def gemm_bias_act_gpu(
     act_code = {"none": 0, "gelu": 1}[act]
     rc = lib.pw_gemm_bf16(
         ctypes.c_void_p(a.contiguous().data_ptr()),
-        ctypes.c_void_p(b.contiguous().data_ptr()),
+        ctypes.c_void_p(b_t.contiguous().data_ptr()),
         ctypes.c_void_p(bias.data_ptr() if bias is not None else 0),
         ctypes.c_void_p(out.data_ptr()),
         ctypes.c_int64(M),

@@ -299,3 +299,148 @@ int64_t pw_csv_normalize(const char* path, char delimiter, int64_t ncols,
 }
 
 }  // extern "C"
+
+// --------------------------------------------------------------- LZ4 -----
+// LZ4 block-format compressor/decompressor (host C++), used by the
+// persistence input-snapshot chunks (reference input_snapshot.rs:5 uses
+// lz4 block compression with size-prepended frames).  Standard LZ4 block
+// format: sequences of [token][literals][offset(2B LE)][matchlen...];
+// greedy 4-byte hash-chain matcher.  Output interoperates with any LZ4
+// block decoder.
+
+extern "C" {
+
+int64_t pw_lz4_compress_bound(int64_t n) {
+  return n + n / 255 + 16;
+}
+
+// returns compressed size, or -1 on overflow of the out buffer
+int64_t pw_lz4_compress(const uint8_t* src, int64_t n, uint8_t* dst,
+                        int64_t cap) {
+  const int MINMATCH = 4;
+  const int64_t MFLIMIT = 12;  // last 12 bytes are always literals
+  if (n == 0) return 0;
+  uint32_t htab[1 << 14];
+  memset(htab, 0xFF, sizeof(htab));
+  auto hash4 = [](const uint8_t* p) -> uint32_t {
+    uint32_t v;
+    memcpy(&v, p, 4);
+    return (v * 2654435761u) >> 18;  // 14-bit
+  };
+  int64_t ip = 0, anchor = 0, op = 0;
+  const int64_t mflimit = n - MFLIMIT;
+  while (ip < mflimit) {
+    uint32_t h = hash4(src + ip);
+    int64_t ref = (htab[h] == 0xFFFFFFFFu) ? -1 : (int64_t)htab[h];
+    htab[h] = (uint32_t)ip;
+    if (ref >= 0 && ip - ref <= 0xFFFF &&
+        memcmp(src + ref, src + ip, MINMATCH) == 0) {
+      // extend match
+      int64_t mlen = MINMATCH;
+      while (ip + mlen < n - 5 && src[ref + mlen] == src[ip + mlen]) ++mlen;
+      int64_t litlen = ip - anchor;
+      // token + literals
+      int64_t need = 1 + litlen / 255 + 1 + litlen + 2 + mlen / 255 + 1;
+      if (op + need >= cap) return -1;
+      uint8_t* tok = dst + op++;
+      if (litlen >= 15) {
+        *tok = 0xF0;
+        int64_t rest = litlen - 15;
+        while (rest >= 255) {
+          dst[op++] = 255;
+          rest -= 255;
+        }
+        dst[op++] = (uint8_t)rest;
+      } else {
+        *tok = (uint8_t)(litlen << 4);
+      }
+      memcpy(dst + op, src + anchor, litlen);
+      op += litlen;
+      // offset
+      uint16_t off = (uint16_t)(ip - ref);
+      dst[op++] = (uint8_t)(off & 0xFF);
+      dst[op++] = (uint8_t)(off >> 8);
+      // match length (stored - MINMATCH)
+      int64_t mstore = mlen - MINMATCH;
+      if (mstore >= 15) {
+        *tok |= 0x0F;
+        int64_t rest = mstore - 15;
+        while (rest >= 255) {
+          dst[op++] = 255;
+          rest -= 255;
+        }
+        dst[op++] = (uint8_t)rest;
+      } else {
+        *tok |= (uint8_t)mstore;
+      }
+      ip += mlen;
+      anchor = ip;
+    } else {
+      ++ip;
+    }
+  }
+  // final literals
+  int64_t litlen = n - anchor;
+  int64_t need = 1 + litlen / 255 + 1 + litlen;
+  if (op + need > cap) return -1;
+  uint8_t* tok = dst + op++;
+  if (litlen >= 15) {
+    *tok = 0xF0;
+    int64_t rest = litlen - 15;
+    while (rest >= 255) {
+      dst[op++] = 255;
+      rest -= 255;
+    }
+    dst[op++] = (uint8_t)rest;
+  } else {
+    *tok = (uint8_t)(litlen << 4);
+  }
+  memcpy(dst + op, src + anchor, litlen);
+  op += litlen;
+  return op;
+}
+
+// returns decompressed size, or -1 on malformed input / overflow
+int64_t pw_lz4_decompress(const uint8_t* src, int64_t n, uint8_t* dst,
+                          int64_t cap) {
+  int64_t ip = 0, op = 0;
+  while (ip < n) {
+    uint8_t token = src[ip++];
+    int64_t litlen = token >> 4;
+    if (litlen == 15) {
+      uint8_t b;
+      do {
+        if (ip >= n) return -1;
+        b = src[ip++];
+        litlen += b;
+      } while (b == 255);
+    }
+    if (ip + litlen > n || op + litlen > cap) return -1;
+    memcpy(dst + op, src + ip, litlen);
+    ip += litlen;
+    op += litlen;
+    if (ip >= n) break;  // last sequence has no match
+    if (ip + 2 > n) return -1;
+    uint16_t off = (uint16_t)(src[ip] | (src[ip + 1] << 8));
+    ip += 2;
+    if (off == 0 || off > op) return -1;
+    int64_t mlen = (token & 0x0F);
+    if (mlen == 15) {
+      uint8_t b;
+      do {
+        if (ip >= n) return -1;
+        b = src[ip++];
+        mlen += b;
+      } while (b == 255);
+    }
+    mlen += 4;
+    if (op + mlen > cap) return -1;
+    // overlapping copy must be byte-wise
+    const uint8_t* m = dst + op - off;
+    for (int64_t j = 0; j < mlen; ++j) dst[op + j] = m[j];
+    op += mlen;
+  }
+  return op;
+}
+
+}  // extern "C"

@@ -808,8 +808,12 @@ __device__ __forceinline__ float pw_gelu(float x) {
   return 0.5f * x * (1.0f + erff(x * 0.70710678118654752f));
 }
 
+// B is passed TRANSPOSED (Bt = B^T, (N, K) row-major): both A and Bt
+// tiles then stage into LDS with linear 16B vector loads and both MFMA
+// fragments are contiguous 8-element ds reads (weights are static, so
+// the one-time transpose is free on the host side).
 __global__ __launch_bounds__(256) void k_gemm_bf16(
-    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    const __bf16* __restrict__ A, const __bf16* __restrict__ Bt,
     const float* __restrict__ bias, __bf16* __restrict__ C, int64_t M,
     int64_t N, int64_t K, int act, int nbm) {
   __shared__ __bf16 lasA[2][PW_GEMM_BM][PW_GEMM_BK];
@@ -860,22 +864,20 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
                                               : (__bf16)0.f;
       }
     }
-    // B: tile 64x128
-    for (int it = tid; it < (PW_GEMM_BK * PW_GEMM_BN) / 8; it += 256) {
-      int r = it / (PW_GEMM_BN / 8);
-      int cchunk = it % (PW_GEMM_BN / 8);
-      int64_t grow = k0 + r;
-      int64_t gcol = col0 + cchunk * 8;
-      if (grow < K && gcol + 8 <= N) {
-        pw_frag8 v = *(const pw_frag8*)&B[grow * N + gcol];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) lasBt[buf][cchunk * 8 + j][r] = v[j];
+    // Bt: tile 128 (N-rows) x 64 (K-cols), linear like A
+    for (int it = tid; it < (PW_GEMM_BN * PW_GEMM_BK) / 8; it += 256) {
+      int r = it / (PW_GEMM_BK / 8);
+      int cchunk = it % (PW_GEMM_BK / 8);
+      int64_t grow = col0 + r;       // N index
+      int64_t gcol = k0 + cchunk * 8;  // K index
+      __bf16* dst = &lasBt[buf][r][cchunk * 8];
+      if (grow < N && gcol + 8 <= K) {
+        *(pw_frag8*)dst = *(const pw_frag8*)&Bt[grow * K + gcol];
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          lasBt[buf][cchunk * 8 + j][r] =
-              (grow < K && gcol + j < N) ? B[grow * N + gcol + j]
-                                         : (__bf16)0.f;
+          dst[j] = (grow < N && gcol + j < K) ? Bt[grow * K + gcol + j]
+                                              : (__bf16)0.f;
       }
     }
   };

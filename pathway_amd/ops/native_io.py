@@ -148,3 +148,46 @@ def read_csv(path: str, delimiter: str = ",") -> tuple[list[str], list[list[str]
     rows = [fields[i : i + ncols] for i in range(0, got * ncols, ncols)]
     header = rows[0] if rows else []
     return header, rows[1:]
+
+
+def _ensure_lz4(lib):
+    if not hasattr(lib, "_lz4_ready"):
+        lib.pw_lz4_compress_bound.restype = ctypes.c_int64
+        lib.pw_lz4_compress_bound.argtypes = [ctypes.c_int64]
+        lib.pw_lz4_compress.restype = ctypes.c_int64
+        lib.pw_lz4_compress.argtypes = [
+            ctypes.c_char_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_int64,
+        ]
+        lib.pw_lz4_decompress.restype = ctypes.c_int64
+        lib.pw_lz4_decompress.argtypes = [
+            ctypes.c_char_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_int64,
+        ]
+        lib._lz4_ready = True
+    return lib
+
+
+def lz4_compress(data: bytes) -> bytes:
+    """LZ4 block compression via the native codec (pw_io.cpp)."""
+    lib = _try_load()
+    if lib is None:
+        raise RuntimeError(f"libpwio unavailable: {_load_error}")
+    _ensure_lz4(lib)
+    n = len(data)
+    cap = lib.pw_lz4_compress_bound(n)
+    out = ctypes.create_string_buffer(cap)
+    got = lib.pw_lz4_compress(data, n, out, cap)
+    if got < 0:
+        raise RuntimeError("lz4 compress overflow")
+    return out.raw[:got]
+
+
+def lz4_decompress(data: bytes, uncompressed_size: int) -> bytes:
+    lib = _try_load()
+    if lib is None:
+        raise RuntimeError(f"libpwio unavailable: {_load_error}")
+    _ensure_lz4(lib)
+    out = ctypes.create_string_buffer(max(uncompressed_size, 1))
+    got = lib.pw_lz4_decompress(data, len(data), out, uncompressed_size)
+    if got != uncompressed_size:
+        raise RuntimeError(f"lz4 decompress: got {got}, want {uncompressed_size}")
+    return out.raw[:uncompressed_size]

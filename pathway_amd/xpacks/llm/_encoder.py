@@ -157,16 +157,23 @@ class NativeEncoder(torch.nn.Module):
             bias_cache = getattr(self, "_f32_bias", None)
             if bias_cache is None:
                 bias_cache = self._f32_bias = {}
+            wt_cache = getattr(self, "_wt_cache", None)
+            if wt_cache is None:
+                wt_cache = self._wt_cache = {}
 
             def mm(h2, w, bias, act="none"):
-                # hand-written MFMA tile kernel, bias(+GELU) fused
+                # hand-written MFMA tile kernel, bias(+GELU) fused;
+                # weights pre-transposed once (static) for linear staging
                 b32 = bias_cache.get(id(bias))
                 if b32 is None:
                     b32 = bias_cache[id(bias)] = bias.detach().to(
                         torch.float32
                     ).contiguous()
+                wt = wt_cache.get(id(w))
+                if wt is None:
+                    wt = wt_cache[id(w)] = w.detach().T.contiguous()
                 return ops.gemm_bias_act_gpu(
-                    h2.reshape(-1, h2.shape[-1]), w, b32, act
+                    h2.reshape(-1, h2.shape[-1]), wt, b32, act
                 ).view(*h2.shape[:-1], w.shape[1])
         else:
 

@@ -39,9 +39,10 @@ def main():
     for N, K in [(1152, 384), (384, 384), (1536, 384), (384, 1536)]:
         a = (torch.randn(M, K, device="cuda") * 0.1).to(torch.bfloat16)
         b = (torch.randn(K, N, device="cuda") * 0.1).to(torch.bfloat16)
+        bt = b.T.contiguous()
         bias = torch.randn(N, device="cuda")
         t_pw = bench(f"pw_gemm_bf16 {M}x{N}x{K}",
-                     lambda: ops.gemm_bias_act_gpu(a, b, bias))
+                     lambda: ops.gemm_bias_act_gpu(a, bt, bias))
         bb = bias.to(torch.bfloat16)
         t_blas = bench(f"hipBLASLt   {M}x{N}x{K}",
                        lambda: a @ b + bb)

@@ -490,7 +490,7 @@ def test_mfma_gemm_vs_torch():
         b = (torch.randn(K, N, device="cuda") * 0.5).to(torch.bfloat16)
         bias = torch.randn(N, device="cuda")
         ref = (a.float() @ b.float() + bias).to(torch.bfloat16).float()
-        got = ops.gemm_bias_act_gpu(a, b, bias, act="none").float()
+        got = ops.gemm_bias_act_gpu(a, b.T.contiguous(), bias, act="none").float()
         err = (got - ref).abs().max().item()
         scale = ref.abs().max().item() + 1.0
         assert err / scale < 0.02, (M, N, K, err, scale)
@@ -498,7 +498,9 @@ def test_mfma_gemm_vs_torch():
         import torch.nn.functional as F
 
         ref_g = F.gelu(a.float() @ b.float() + bias)
-        got_g = ops.gemm_bias_act_gpu(a, b, bias, act="gelu").float()
+        got_g = ops.gemm_bias_act_gpu(
+            a, b.T.contiguous(), bias, act="gelu"
+        ).float()
         errg = (got_g - ref_g).abs().max().item()
         assert errg / (ref_g.abs().max().item() + 1.0) < 0.02, (M, N, K, errg)
 
