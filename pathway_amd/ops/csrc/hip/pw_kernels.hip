@@ -843,6 +843,32 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
     for (int n = 0; n < 4; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
 
   int nk = (int)((K + PW_GEMM_BK - 1) / PW_GEMM_BK);
+  bool interior_mn = (row0 + PW_GEMM_BM <= M) && (col0 + PW_GEMM_BN <= N);
+
+  // fast staging for interior tiles: async global->LDS DMA, 16B per lane
+  // (guide §5: global_load_lds dwordx4 — the compiler never auto-emits it)
+  auto load_tiles_gll = [&](int kt, int buf) {
+    int64_t k0 = (int64_t)kt * PW_GEMM_BK;
+    // A tile: 16 KiB = 16 wave-chunks of 1 KiB; 4 chunks per wave
+    typedef const __attribute__((address_space(1))) uint8_t* gptr_t;
+    typedef __attribute__((address_space(3))) uint8_t* lptr_t;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int wchunk = wave * 4 + i;
+      int eoff = wchunk * 512 + lane * 8;
+      int r = eoff >> 6, c = eoff & 63;
+      gptr_t srcA = (gptr_t)(const void*)&A[(row0 + r) * K + k0 + c];
+      lptr_t dstA = (lptr_t)(void*)(&lasA[buf][0][0] + wchunk * 512);
+      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)srcA,
+                                       (__attribute__((address_space(3))) void*)dstA,
+                                       16, 0, 0);
+      gptr_t srcB = (gptr_t)(const void*)&Bt[(col0 + r) * K + k0 + c];
+      lptr_t dstB = (lptr_t)(void*)(&lasBt[buf][0][0] + wchunk * 512);
+      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)srcB,
+                                       (__attribute__((address_space(3))) void*)dstB,
+                                       16, 0, 0);
+    }
+  };
 
   // cooperative tile load: 256 threads, A tile 128x64 (8192 elems = 32/thread),
   // 8-wide vector loads -> 4 loads per thread
@@ -882,11 +908,18 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
     }
   };
 
-  load_tiles(0, 0);
+  auto stage = [&](int kt, int buf) {
+    if (interior_mn && (int64_t)(kt + 1) * PW_GEMM_BK <= K)
+      load_tiles_gll(kt, buf);
+    else
+      load_tiles(kt, buf);
+  };
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     int buf = kt & 1;
-    if (kt + 1 < nk) load_tiles(kt + 1, buf ^ 1);
+    if (kt + 1 < nk) stage(kt + 1, buf ^ 1);
 #pragma unroll
     for (int kk = 0; kk < PW_GEMM_BK / 32; ++kk) {
 #pragma unroll
@@ -904,6 +937,7 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
         }
       }
     }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
   }
 
