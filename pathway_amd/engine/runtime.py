@@ -523,7 +523,16 @@ class Runtime:
             self._clock = max(self._clock, t + 2)
             if self.persistence is not None:
                 if getattr(self.persistence, "operator_persisting", False):
-                    self.save_operator_snapshot(t)
+                    # honor snapshot_interval_ms (reference saves operator
+                    # state at interval boundaries, not every step)
+                    interval = getattr(
+                        self.persistence, "snapshot_interval_ms", 0
+                    )
+                    now = _time.monotonic()
+                    last = getattr(self, "_last_opsnap", 0.0)
+                    if interval <= 0 or (now - last) * 1000.0 >= interval:
+                        self.save_operator_snapshot(t)
+                        self._last_opsnap = now
                 self.persistence.commit(t)
             steps += 1
             if max_steps is not None and steps >= max_steps:

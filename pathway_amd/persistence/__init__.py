@@ -26,10 +26,19 @@ class Backend:
         return b
 
     @classmethod
-    def azure(cls, root_path: str, account: Any = None, **kw) -> "Backend":
+    def azure(cls, root_path: str, account: Any = None, *,
+              account_url: str | None = None, container: str | None = None,
+              sas_token: str | None = None, **kw) -> "Backend":
         b = cls()
         b.kind = "azure"
         b.path = root_path
+        b.account_url = account_url or (
+            getattr(account, "account_url", None) if account else None
+        ) or ""
+        b.container = container or (
+            getattr(account, "container", None) if account else None
+        ) or ""
+        b.sas_token = sas_token
         return b
 
     @classmethod

@@ -190,7 +190,9 @@ def decode_value(r: _Reader) -> Any:
     if tag == V_BYTES:
         return r.bytes_()
     if tag == V_JSON:
-        return Json.loads(r.bytes_().decode("utf-8"))
+        import json as _json
+
+        return Json(_json.loads(r.bytes_().decode("utf-8")))
     if tag == V_TUPLE:
         n = r.u64()
         return tuple(decode_value(r) for _ in range(n))

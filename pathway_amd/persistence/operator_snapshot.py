@@ -80,29 +80,127 @@ def tensor_from_portable(a, device) -> torch.Tensor:
     return torch.from_numpy(np.ascontiguousarray(a)).to(device)
 
 
+class _RestrictedUnpickler(pickle.Unpickler):
+    """Allowlisting unpickler: operator snapshots are numpy/dtype/builtin
+    structures only — a writable persistence root must not allow code
+    execution on restart (ADVICE r1 finding 5)."""
+
+    _ALLOWED = {
+        ("builtins", None),  # module allowed, names checked below
+        ("numpy", None),
+        ("numpy.core.multiarray", None),
+        ("numpy._core.multiarray", None),
+        ("numpy.dtypes", None),
+        ("collections", None),
+        ("pathway_amd.internals.dtype", None),
+        ("pathway_amd.internals.api", None),
+        ("pathway_amd.internals.json", None),
+        ("pathway_amd.internals.datetime_types", None),
+        ("pandas._libs.tslibs.timestamps", None),
+        ("pandas._libs.tslibs.timedeltas", None),
+        ("datetime", None),
+        ("_codecs", None),
+    }
+    _DENY_NAMES = {"eval", "exec", "compile", "open", "__import__", "getattr",
+                   "setattr", "delattr", "input", "breakpoint"}
+
+    def find_class(self, module, name):
+        top = module.split(".")[0]
+        roots = {m.split(".")[0] for m, _ in self._ALLOWED}
+        if top not in roots or name in self._DENY_NAMES:
+            raise pickle.UnpicklingError(
+                f"operator snapshot references disallowed global "
+                f"{module}.{name}"
+            )
+        return super().find_class(module, name)
+
+
+def _safe_loads(data: bytes):
+    import io as _io
+
+    return _RestrictedUnpickler(_io.BytesIO(data)).load()
+
+
 class OperatorSnapshotStore:
+    """Operator-state snapshots over a persistence BlobStore.
+
+    Writes run on a background thread (the reference's background
+    snapshot writer/merger, operator_snapshot.rs:172-380 — amortization
+    here is by not blocking the step loop; state is full-serialize since
+    the engine's per-node states are already consolidated tensors).
+    Two rotations are kept; load takes the newest readable one.
+    """
+
     def __init__(self, root: str):
-        self.root = os.path.join(root, "operator_snapshots")
-        os.makedirs(self.root, exist_ok=True)
+        from pathway_amd.persistence.backends import FileStore
+
+        self.store = FileStore(root)
+        self.prefix = "operator_snapshots/0"
+        self._init_common()
+
+    @classmethod
+    def over_store(cls, store, worker: int) -> "OperatorSnapshotStore":
+        self = cls.__new__(cls)
+        self.store = store
+        self.prefix = f"operator_snapshots/{worker}"
+        self._init_common()
+        return self
+
+    def _init_common(self):
+        import threading
+
+        self._seq = 0
+        self._thread = None
+        self._lock = threading.Lock()
+
+    def _write(self, payload: bytes, seq: int) -> None:
+        self.store.put(f"{self.prefix}/snapshot-{seq}.bin", payload)
+        # keep the last two rotations
+        keys = sorted(
+            self.store.list(self.prefix + "/"),
+            key=lambda k: int(k.rsplit("-", 1)[-1].split(".")[0]),
+        )
+        for k in keys[:-2]:
+            self.store.delete(k)
 
     def save(self, snapshot_time: int, states: dict[str, Any]) -> None:
         payload = zlib.compress(
             pickle.dumps({"time": snapshot_time, "states": states}, protocol=4), 1
         )
-        tmp = os.path.join(self.root, "snapshot.bin.tmp")
-        with open(tmp, "wb") as f:
-            f.write(payload)
-            f.flush()
-            os.fsync(f.fileno())
-        os.replace(tmp, os.path.join(self.root, "snapshot.bin"))
+        with self._lock:
+            seq = self._seq
+            self._seq += 1
+            if self._thread is not None:
+                self._thread.join()
+            import threading
+
+            self._thread = threading.Thread(
+                target=self._write, args=(payload, seq), daemon=True
+            )
+            self._thread.start()
+
+    def wait(self) -> None:
+        with self._lock:
+            if self._thread is not None:
+                self._thread.join()
+                self._thread = None
 
     def load(self) -> tuple[int, dict[str, Any]] | None:
-        path = os.path.join(self.root, "snapshot.bin")
-        if not os.path.exists(path):
-            return None
-        with open(path, "rb") as f:
-            d = pickle.loads(zlib.decompress(f.read()))
-        return d["time"], d["states"]
+        self.wait()
+        keys = sorted(
+            self.store.list(self.prefix + "/"),
+            key=lambda k: int(k.rsplit("-", 1)[-1].split(".")[0]),
+        )
+        for key in reversed(keys):
+            data = self.store.get(key)
+            if not data:
+                continue
+            try:
+                d = _safe_loads(zlib.decompress(data))
+                return d["time"], d["states"]
+            except Exception:
+                continue  # fall back to the previous rotation
+        return None
 
 
 # ------------------------------------------------- node state dispatchers --

@@ -115,7 +115,11 @@ def test_pw_run_with_persistence(tmp_path):
     res = t.reduce(s=pw.reducers.sum(pw.this.v))
     pw.io.csv.write(res, outfile)
     pw.run(persistence_config=cfg)
-    assert os.path.exists(os.path.join(pdir, "w0", "metadata.json"))
+    # round-2 layout: metadata under <version>-<worker>-<rotation> keys
+    mdir = os.path.join(pdir, "metadata")
+    assert os.path.isdir(mdir)
+    names = os.listdir(mdir)
+    assert any(n.split("-")[1] == "0" for n in names), names
     with open(outfile) as f:
         assert len(f.read().strip().splitlines()) >= 2
 
