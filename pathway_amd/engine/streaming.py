@@ -44,6 +44,48 @@ class StreamingSource(Source):
         self._sync_member = None
         self._sync_idx = None
         self._held: list = []
+        # ingest/compute overlap: a stager thread
+        # pre-converts queued rows into CPU column bundles while the
+        # engine computes the previous timestamp; pull() then only stamps
+        # the time and moves tensors to the device.
+        self._staged: queue.Queue = queue.Queue(maxsize=2)
+        self._stager: threading.Thread | None = None
+
+    def _ensure_stager(self) -> None:
+        if self._stager is not None or self._sync_group is not None:
+            return
+        t = threading.Thread(target=self._stage_loop, daemon=True)
+        self._stager = t
+        t.start()
+
+    def _stage_loop(self) -> None:
+        import torch as _torch
+
+        while True:
+            rows = []
+            try:
+                rows.append(self.q.get(timeout=0.05))
+            except queue.Empty:
+                if self._finished.is_set() and self.q.empty():
+                    return
+                continue
+            try:
+                while len(rows) < MAX_EVENTS_PER_STEP:
+                    rows.append(self.q.get_nowait())
+            except queue.Empty:
+                pass
+            keys = _torch.tensor(
+                [list(k.as_signed_pair()) for k, _, _ in rows],
+                dtype=_torch.int64,
+            ).reshape(len(rows), 2)
+            diffs = _torch.tensor(
+                [d for _, _, d in rows], dtype=_torch.int64
+            )
+            cols = {}
+            for j, nme in enumerate(self.column_names):
+                vals = [v[j] if j < len(v) else None for _, v, _ in rows]
+                cols[nme] = column_from_pylist(vals, self.dtypes[j], "cpu")
+            self._staged.put((keys, cols, diffs))  # blocks at depth 2
 
     # -- producer side (reader thread) --
 
@@ -75,7 +117,11 @@ class StreamingSource(Source):
     # -- consumer side (engine loop) --
 
     def has_pending(self) -> bool:
-        return bool(self._held) or not self.q.empty()
+        return (
+            bool(self._held)
+            or not self.q.empty()
+            or not self._staged.empty()
+        )
 
     def is_live(self) -> bool:
         return not self._finished.is_set() or self.has_pending()
@@ -83,6 +129,13 @@ class StreamingSource(Source):
     def next_time(self):
         from pathway_amd.engine.runtime import STREAM_READY, STREAM_WAITING
 
+        if self._sync_group is None:
+            self._ensure_stager()
+            if self._held or not self._staged.empty():
+                return STREAM_READY
+            if self._finished.is_set() and self.q.empty():
+                return None
+            return STREAM_WAITING
         if self.has_pending():
             return STREAM_READY
         if self.is_live():
@@ -94,6 +147,18 @@ class StreamingSource(Source):
         if err is not None:
             self._error = None
             raise RuntimeError(f"connector reader failed: {err}") from err
+        if self._sync_group is None and self._stager is not None:
+            # staged path: the bundle was pre-built off-thread
+            try:
+                keys, cols, diffs = self._staged.get_nowait()
+            except queue.Empty:
+                return None
+            dev = torch.device(device)
+            if dev.type != "cpu":
+                keys = keys.to(dev, non_blocking=True)
+                diffs = diffs.to(dev, non_blocking=True)
+                cols = {n: c.to_device(dev) for n, c in cols.items()}
+            return DeltaBatch(keys, cols, diffs, time)
         rows = list(self._held)
         self._held = []
         try:
