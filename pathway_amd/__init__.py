@@ -127,6 +127,11 @@ from pathway_amd.stdlib import (  # noqa: E402
     utils,
     viz,
 )
+from pathway_amd.internals.exported import (  # noqa: E402
+    ExportedTable,
+    export_table,
+    import_table,
+)
 from pathway_amd.stdlib.utils.async_transformer import AsyncTransformer  # noqa: E402
 from pathway_amd.stdlib.utils.pandas_transformer import pandas_transformer  # noqa: E402
 
@@ -191,6 +196,9 @@ __all__ = [
     "GroupedTable",
     "iterate",
     "iterate_universe",
+    "ExportedTable",
+    "export_table",
+    "import_table",
     "JoinResult",
     "JoinMode",
     "GroupedJoinResult",

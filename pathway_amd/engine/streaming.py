@@ -50,6 +50,8 @@ class StreamingSource(Source):
         # the time and moves tensors to the device.
         self._staged: queue.Queue = queue.Queue(maxsize=2)
         self._stager: threading.Thread | None = None
+        self._drain_lock = threading.Lock()
+        self._inflight = False
 
     def _ensure_stager(self) -> None:
         if self._stager is not None or self._sync_group is not None:
@@ -62,18 +64,23 @@ class StreamingSource(Source):
         import torch as _torch
 
         while True:
-            rows = []
-            try:
-                rows.append(self.q.get(timeout=0.05))
-            except queue.Empty:
-                if self._finished.is_set() and self.q.empty():
+            if self.q.empty():
+                if self._finished.is_set():
                     return
+                _time.sleep(0.002)
                 continue
-            try:
-                while len(rows) < MAX_EVENTS_PER_STEP:
-                    rows.append(self.q.get_nowait())
-            except queue.Empty:
-                pass
+            # drain under the lock so next_time's done-check never sees
+            # rows vanish mid-transfer
+            with self._drain_lock:
+                rows = []
+                try:
+                    while len(rows) < MAX_EVENTS_PER_STEP:
+                        rows.append(self.q.get_nowait())
+                except queue.Empty:
+                    pass
+                if not rows:
+                    continue
+                self._inflight = True
             keys = _torch.tensor(
                 [list(k.as_signed_pair()) for k, _, _ in rows],
                 dtype=_torch.int64,
@@ -86,6 +93,8 @@ class StreamingSource(Source):
                 vals = [v[j] if j < len(v) else None for _, v, _ in rows]
                 cols[nme] = column_from_pylist(vals, self.dtypes[j], "cpu")
             self._staged.put((keys, cols, diffs))  # blocks at depth 2
+            with self._drain_lock:
+                self._inflight = False
 
     # -- producer side (reader thread) --
 
@@ -131,10 +140,15 @@ class StreamingSource(Source):
 
         if self._sync_group is None:
             self._ensure_stager()
-            if self._held or not self._staged.empty():
-                return STREAM_READY
-            if self._finished.is_set() and self.q.empty():
-                return None
+            with self._drain_lock:
+                if self._held or not self._staged.empty():
+                    return STREAM_READY
+                if (
+                    self._finished.is_set()
+                    and self.q.empty()
+                    and not self._inflight
+                ):
+                    return None
             return STREAM_WAITING
         if self.has_pending():
             return STREAM_READY

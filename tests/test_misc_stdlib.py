@@ -231,3 +231,105 @@ def test_mcp_server_surface():
     from pathway_amd.xpacks.llm.mcp_server import McpServable, McpServer
 
     assert hasattr(McpServer, "tool") or hasattr(McpServer, "serve") or McpServable
+
+
+def test_export_import_table_cross_graph():
+    """Cross-graph handoff (reference graph.rs:616-646 ExportedTable)."""
+    import pathway_amd as pw
+    from pathway_amd.internals.rungraph import G
+
+    G.clear()
+    t = pw.debug.table_from_markdown(
+        """
+        a | b
+        1 | x
+        2 | y
+        3 | x
+        """
+    )
+    agg = t.groupby(pw.this.b).reduce(pw.this.b, s=pw.reducers.sum(pw.this.a))
+    handle = pw.export_table(agg)
+    pw.run()
+    assert not handle.failed()
+    assert handle.frontier() >= 1
+    snap = handle.snapshot_at()
+    assert sorted((v[0], v[1]) for v in snap.values()) == [("x", 4), ("y", 2)]
+
+    # second graph: import and continue computing
+    G.clear()
+    back = pw.import_table(handle)
+    doubled = back.select(pw.this.b, d=pw.this.s * 2)
+    keys, cols = pw.debug.table_to_dicts(doubled)
+    assert sorted(zip(cols["b"].values(), cols["d"].values())) == [
+        ("x", 8), ("y", 4)
+    ]
+
+
+def test_export_import_preserves_retraction_stream():
+    import pathway_amd as pw
+    from pathway_amd.debug import table_from_rows
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.internals.schema import schema_from_types
+
+    G.clear()
+    t = table_from_rows(
+        schema_from_types(v=int), [(1, 0, 1), (2, 2, 1)], is_stream=True
+    )
+    total = t.reduce(s=pw.reducers.sum(pw.this.v))
+    handle = pw.export_table(total)
+    pw.run()
+    rows, _ = handle.data_from_offset(0)
+    # incremental stream: +1 at t=0, then -1/+3 at t=2
+    seq = sorted((r.time, r.diff, r.values[0]) for r in rows)
+    assert seq == [(0, 1, 1), (2, -1, 1), (2, 1, 3)]
+
+    G.clear()
+    back = pw.import_table(handle)
+    keys, cols = pw.debug.table_to_dicts(back)
+    assert list(cols["s"].values()) == [3]
+
+
+def test_async_transformer_views_and_options():
+    """Deepened AsyncTransformer (reference async_transformer.rs:297):
+    concurrent invoke via the async UDF executor; successful/failed
+    views split on per-row errors."""
+    import asyncio
+
+    import pathway_amd as pw
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.internals.schema import schema_from_types
+    from pathway_amd.stdlib.utils.async_transformer import AsyncTransformer
+
+    G.clear()
+    t = pw.debug.table_from_markdown(
+        """
+        a
+        1
+        2
+        3
+        4
+        """
+    )
+    calls = []
+
+    class MyT(AsyncTransformer):
+        output_schema = schema_from_types(doubled=int)
+
+        async def invoke(self, a):
+            calls.append(a)
+            await asyncio.sleep(0.01)
+            if a == 3:
+                raise ValueError("boom")
+            return {"doubled": a * 2}
+
+    tr = MyT(input_table=t).with_options(capacity=4)
+    ok = tr.successful
+    bad = tr.failed
+    keys, cols = pw.debug.table_to_dicts(ok)
+    assert sorted(cols["doubled"].values()) == [2, 4, 8]
+    G.clear()
+    tr2 = MyT(input_table=pw.debug.table_from_markdown("a\n3\n5\n"))
+    # failed rows carry Error values; count them without reading values
+    nfail = tr2.failed.reduce(c=pw.reducers.count())
+    _, cols2 = pw.debug.table_to_dicts(nfail)
+    assert list(cols2["c"].values()) == [1]  # only a==3 failed
