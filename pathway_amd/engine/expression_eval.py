@@ -578,7 +578,15 @@ def _eval_apply(e: expr.ApplyExpression, ctx: EvalContext, is_async: bool) -> Co
             for i in range(n):
                 args = [a[i] for a in arg_vals]
                 kwargs = {k: v[i] for k, v in kw_vals.items()}
-                if e._propagate_none and any(a is None for a in args):
+                if any(a is ERROR for a in args) or any(
+                    v is ERROR for v in kwargs.values()
+                ):
+                    # Value::Error propagates without invoking the UDF
+                    async def _err():
+                        return ERROR
+
+                    coros.append(_err())
+                elif e._propagate_none and any(a is None for a in args):
                     async def _none():
                         return None
 
@@ -603,6 +611,11 @@ def _eval_apply(e: expr.ApplyExpression, ctx: EvalContext, is_async: bool) -> Co
         for i in range(n):
             args = [a[i] for a in arg_vals]
             kwargs = {k: v[i] for k, v in kw_vals.items()}
+            if any(a is ERROR for a in args) or any(
+                v is ERROR for v in kwargs.values()
+            ):
+                out.append(ERROR)  # Value::Error propagates, UDF not called
+                continue
             if e._propagate_none and (
                 any(a is None for a in args) or any(v is None for v in kwargs.values())
             ):

@@ -813,7 +813,23 @@ class Table(TableLike):
         return Table(node, dict(self._dtypes), Universe())
 
     def remove_errors(self) -> "Table":
-        return self  # Error rows are filtered at output in this implementation
+        """Drop rows holding the Error sentinel in any column (reference
+        table.py remove_errors).  Probe trick: applying any function to
+        an Error value yields Error; fill_error maps that to False."""
+        from pathway_amd.internals.expression import (
+            ApplyExpression,
+            FillErrorExpression,
+        )
+
+        mask = None
+        for name in self._dtypes:
+            probe = FillErrorExpression(
+                ApplyExpression(lambda *a: True, None, self[name]), False
+            )
+            mask = probe if mask is None else mask & probe
+        if mask is None:
+            return self
+        return self.filter(mask)
 
     def await_futures(self) -> "Table":
         return self
