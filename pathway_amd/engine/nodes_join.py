@@ -216,12 +216,18 @@ class JoinNode(Node):
 
         if self.probe_only_left:
             # as-of-now: right state updates first (queries at time t see
-            # the state AS OF t), then left deltas probe; no dR emissions
+            # the state AS OF t), then left deltas probe; no dR emissions.
+            # Answers (incl. left-outer pads) are FROZEN at probe time —
+            # later right changes must not retract them, so the generic
+            # old/new pad mechanism below is skipped entirely.
             if jr is not None:
                 self.rstore.merge(jr, br)
             if jl is not None:
                 out_parts += self._emit_pairs(bl, jl, self.rstore, probe_is_right=False)
+                if pad_left:
+                    out_parts += self._asof_now_pads(bl, jl)
                 self.lstore.merge(jl, bl)
+            pad_left = pad_right = False
         else:
             # dR >< L_old
             if jr is not None:
@@ -313,6 +319,57 @@ class JoinNode(Node):
         if self.right_id_name:
             out_cols[self.right_id_name] = PointerColumn(rkeys)
         return DeltaBatch(okeys, out_cols, diffs, time)
+
+    def _asof_now_pads(self, bl: DeltaBatch, jl: torch.Tensor) -> list:
+        """Left-outer pads for as-of-now probes: the rows of THIS left
+        batch whose join key has no current right match, emitted with the
+        batch's own diffs and never revisited (frozen answers)."""
+        device = self.device
+        ukeys = unique_sorted_keys(jl)
+        ucnt = self.rstore.count_for(ukeys)
+        from pathway_amd.engine.state import searchsorted_words
+
+        pos = searchsorted_words(
+            [ukeys[:, 0].contiguous(), ukeys[:, 1].contiguous()],
+            [jl[:, 0].contiguous(), jl[:, 1].contiguous()],
+            side="left",
+        )
+        row_cnt = ucnt.index_select(0, pos.clamp(0, max(ukeys.shape[0] - 1, 0)))
+        un_idx = (row_cnt == 0).nonzero(as_tuple=True)[0]
+        if un_idx.shape[0] == 0:
+            return []
+        probe = bl.take(un_idx)
+        n = len(probe)
+        if self.key_mode == "left":
+            okeys = probe.keys
+        else:
+            lo, hi = hashing.derive_key_words(
+                self.lpad_salt,
+                [(probe.keys[:, 0].contiguous(), probe.keys[:, 1].contiguous())],
+            )
+            okeys = torch.stack([lo, hi], dim=1)
+        out_cols: dict[str, Column] = {}
+        for out_name, src in self.left_out.items():
+            out_cols[out_name] = probe.columns[src]
+        for out_name, src in self.right_out.items():
+            proto = (self._right_proto or {}).get(src)
+            if proto is None:
+                out_cols[out_name] = ObjectColumn(np.empty(n, dtype=object), dt.ANY)
+            else:
+                out_cols[out_name] = null_column(proto, n, device)
+        if self.left_id_name:
+            out_cols[self.left_id_name] = PointerColumn(probe.keys)
+        if self.right_id_name:
+            out_cols[self.right_id_name] = ObjectColumn(
+                np.empty(n, dtype=object), dt.Optional(dt.POINTER)
+            )
+        order = list(self.left_out.keys()) + list(self.right_out.keys())
+        if self.left_id_name:
+            order.append(self.left_id_name)
+        if self.right_id_name:
+            order.append(self.right_id_name)
+        out_cols = {k: out_cols[k] for k in order}
+        return [DeltaBatch(okeys, out_cols, probe.diffs, probe.time)]
 
     # -- padded rows for outer modes --
 

@@ -291,3 +291,31 @@ def test_session_predicate():
         t.t, window=pw.temporal.session(predicate=lambda a, b: abs(a - b) <= 3)
     ).reduce(n=pw.reducers.count())
     assert sorted(x[0] for x in _rows(r, "n")) == [1, 2]
+
+
+def test_asof_now_left_pads_frozen_before_first_right_row():
+    """as-of-now answers (incl. left-outer pads) freeze at probe time:
+    left rows arriving before ANY right row keep their null-padded
+    answer even after the right side later fills (r2 regression —
+    reference use-as-of-now semantics)."""
+    import datetime
+
+    import pathway_amd as pw
+    import pathway_amd.stdlib.temporal.time_utils as tu
+    from pathway_amd.internals.rungraph import G
+
+    tu.utc_now.cache_clear()
+    G.clear()
+    t = pw.debug.table_from_markdown("a\n1\n2\n")
+    right = tu.utc_now(
+        refresh_rate=datetime.timedelta(milliseconds=100),
+        initial_delay=datetime.timedelta(milliseconds=300),
+        max_ticks=2,
+    ).reduce(timestamp_utc=pw.reducers.latest(pw.this.timestamp_utc))
+    res = t.asof_now_join_left(right).select(
+        pw.left.a, ts=pw.right.timestamp_utc
+    )
+    _, cols = pw.debug.table_to_dicts(res)
+    assert sorted(cols["a"].values()) == [1, 2]
+    assert all(v is None for v in cols["ts"].values())
+    tu.utc_now.cache_clear()
