@@ -390,7 +390,10 @@ class Table(TableLike):
             else self._dtypes[n]
             for n in names
         }
-        return Table(node, dtypes, Universe())
+        # concat output universe = union of the inputs' universes
+        # (solver: every part is a subset; union ⊆ any common superset)
+        u = Universe.union_of(*[t._universe for t in tables])
+        return Table(node, dtypes, u)
 
     def concat_reindex(self, *others: "Table") -> "Table":
         from pathway_amd.engine.nodes import ConcatNode, DeriveKeyNode
@@ -423,18 +426,19 @@ class Table(TableLike):
     def intersect(self, *tables: "Table") -> "Table":
         from pathway_amd.engine.nodes_join import SemiJoinNode
 
-        node = self._node
         out = self
         for t in tables:
             n = SemiJoinNode(out._node, t._node, "intersect", get_device())
-            out = Table(n, out._dtypes, out._universe.subuniverse())
+            u = Universe.intersection_of(out._universe, t._universe)
+            out = Table(n, out._dtypes, u)
         return out
 
     def difference(self, other: "Table") -> "Table":
         from pathway_amd.engine.nodes_join import SemiJoinNode
 
         node = SemiJoinNode(self._node, other._node, "difference", get_device())
-        return Table(node, self._dtypes, self._universe.subuniverse())
+        u = Universe.difference_of(self._universe, other._universe)
+        return Table(node, self._dtypes, u)
 
     def restrict(self, other: TableLike) -> "Table":
         from pathway_amd.engine.nodes_join import SemiJoinNode

@@ -426,3 +426,64 @@ def test_compute_and_print_update_stream(capsys):
     compute_and_print_update_stream(t)
     out = capsys.readouterr().out
     assert "-1" in out and "2" in out
+
+
+def test_universe_solver_algebra():
+    """Relational universe reasoner (reference universe_solver.py SAT
+    encoding): consequences of union/intersection/difference promises
+    are derivable; unrelated universes are not conflated."""
+    from pathway_amd.internals.universe import (
+        Universe,
+        promise_are_pairwise_disjoint,
+        promise_is_subset_of,
+    )
+
+    a, b, w = Universe(), Universe(), Universe()
+    promise_is_subset_of(a, w)
+    promise_is_subset_of(b, w)
+    u = Universe.union_of(a, b)
+    # parts ⊆ union; union ⊆ any common superset
+    assert a.is_subset_of(u) and b.is_subset_of(u)
+    assert u.is_subset_of(w)
+    assert not w.is_subset_of(u)
+
+    i = Universe.intersection_of(a, b)
+    assert i.is_subset_of(a) and i.is_subset_of(b)
+    x = Universe()
+    promise_is_subset_of(x, a)
+    promise_is_subset_of(x, b)
+    # x ⊆ a and x ⊆ b  ->  x ⊆ a∩b
+    assert x.is_subset_of(i)
+
+    d = Universe.difference_of(a, b)
+    assert d.is_subset_of(a)
+    assert d.is_disjoint_from(b)
+    y = Universe()
+    promise_is_subset_of(y, b)
+    assert d.is_disjoint_from(y)
+
+    p, q = Universe(), Universe()
+    promise_are_pairwise_disjoint(p, q)
+    sp, sq = Universe(parent=p), Universe(parent=q)
+    assert sp.is_disjoint_from(sq)
+    # soundness: no invented facts
+    assert not p.is_subset_of(q)
+    assert not p.is_equal(q)
+
+
+def test_universe_solver_table_setops():
+    import pathway_amd as pw
+    from pathway_amd.internals.rungraph import G
+
+    G.clear()
+    t = pw.debug.table_from_markdown("a\n1\n2\n3\n")
+    evens = t.filter(pw.this.a % 2 == 0)
+    odds = t.filter(pw.this.a % 2 == 1)
+    u = evens.concat(odds)
+    # concat's universe is the union: contained in the source superset
+    assert u._universe.is_subset_of(t._universe)
+    inter = evens.intersect(t)
+    assert inter._universe.is_subset_of(evens._universe)
+    assert inter._universe.is_subset_of(t._universe)
+    diff = t.difference(evens)
+    assert diff._universe.is_subset_of(t._universe)
