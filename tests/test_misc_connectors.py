@@ -111,3 +111,65 @@ def test_pyfilesystem_read(tmp_path):
     t = pw.io.pyfilesystem.read(f"osfs://{d}", format="binary", mode="static")
     keys, cols = pw.debug.table_to_dicts(t)
     assert sorted(cols["data"].values()) == [b"\x01\x02", b"\x03"]
+
+
+def test_rabbitmq_roundtrip():
+    import json
+    import threading
+    import time
+
+    from pathway_amd.io._amqp_protocol import AmqpClient
+    from tests.fakes.fake_rabbitmq import FakeRabbit
+
+    srv = FakeRabbit().start()
+    try:
+        # raw client pub/consume
+        sub = AmqpClient(port=srv.port)
+        sub.queue_declare("q1")
+        sub.consume("q1")
+        pub = AmqpClient(port=srv.port)
+        pub.queue_declare("q1")
+        pub.publish("q1", b"hello")
+        rk, body = sub.next_delivery()
+        assert (rk, body) == ("q1", b"hello")
+        pub.close()
+        sub.close()
+
+        # table write -> broker
+        G.clear()
+        t = pw.debug.table_from_markdown(
+            """
+            a | b
+            1 | x
+            2 | y
+            """
+        )
+        pw.io.rabbitmq.write(t, f"amqp://127.0.0.1:{srv.port}", "rows")
+        pw.run()
+        deadline = time.time() + 5
+        while time.time() < deadline and sum(
+            1 for k, _ in srv.published if k == "rows"
+        ) < 2:
+            time.sleep(0.05)
+        recs = [json.loads(b) for k, b in srv.published if k == "rows"]
+        assert sorted((r["a"], r["b"]) for r in recs) == [(1, "x"), (2, "y")]
+
+        # streaming read
+        G.clear()
+        tbl = pw.io.rabbitmq.read(
+            f"amqp://127.0.0.1:{srv.port}", "live",
+            schema=schema_from_types(k=int), format="json", _max_messages=2,
+        )
+
+        def later():
+            time.sleep(0.3)
+            c = AmqpClient(port=srv.port)
+            c.publish("live", json.dumps({"k": 5}).encode())
+            c.publish("live", json.dumps({"k": 7}).encode())
+            c.close()
+
+        threading.Thread(target=later, daemon=True).start()
+        keys, cols = pw.debug.table_to_dicts(tbl)
+        assert sorted(cols["k"].values()) == [5, 7]
+    finally:
+        srv.stop()
