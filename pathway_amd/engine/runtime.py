@@ -353,9 +353,14 @@ class Runtime:
         pm = self.persistence
         import time as _t
 
+        from pathway_amd.internals import roctx
+
+        trace = roctx.enabled()
         ops_stats = self.stats.operators
         for node in self.nodes:
             _s0 = _t.perf_counter()
+            if trace:
+                roctx.range_push(f"pw::{type(node).__name__}@{time}")
             if isinstance(node, InputNode):
                 if injected is not None:
                     out = injected.get(id(node))
@@ -381,6 +386,8 @@ class Runtime:
                     st.rows_in += sum(len(b) for b in ins if b is not None)
                     if out is not None:
                         st.rows_out += len(out)
+            if trace:
+                roctx.range_pop()
             outputs[id(node)] = out
         for node in self.nodes:
             if isinstance(node, (SubscribeNode, OutputNode, CaptureNode)):

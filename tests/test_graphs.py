@@ -127,3 +127,43 @@ def test_weighted_graph_contraction():
     # and no self loops
     for i in cols["u"]:
         assert repr(cols["u"][i]) != repr(cols["v"][i])
+
+
+def test_iterate_with_streaming_outer_updates():
+    """Product-time semantics (reference dataflow.rs:5060-5190): each
+    outer timestamp re-runs the fixpoint on the updated input; outputs
+    at t=0 are retracted/updated by the t=2 delta."""
+    import pathway_amd as pw
+    from pathway_amd.debug import table_from_rows
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.internals.schema import schema_from_types
+
+    G.clear()
+    # x arrives at t=0 with value 40, at t=2 a second row 3 arrives
+    t = table_from_rows(
+        schema_from_types(v=int),
+        [(40, 0, 1), (3, 2, 1)],
+        is_stream=True,
+    )
+
+    def logic(t):
+        # halve values over 10 until all <= 10 (terminating fixpoint)
+        over = t.filter(pw.this.v > 10).select(v=pw.this.v // 2)
+        done = t.filter(pw.this.v <= 10)
+        return done.concat(over)
+
+    res = pw.iterate(logic, t=t)
+    keys, cols = pw.debug.table_to_dicts(res)
+    assert sorted(cols["v"].values()) == [3, 10]
+
+    # capture the update stream: the t=2 row must arrive at outer time 2
+    G.clear()
+    t2 = table_from_rows(
+        schema_from_types(v=int), [(40, 0, 1), (3, 2, 1)], is_stream=True
+    )
+    res2 = pw.iterate(logic, t=t2)
+    from pathway_amd.debug import _run_capture
+
+    rows = _run_capture(res2)
+    times = sorted({r.time for r in rows})
+    assert len(times) >= 2  # outputs at both outer timestamps
