@@ -50,10 +50,9 @@ def measure_rag_p50(device, *, n_docs: int = 100_000, n_queries: int = 128,
     # doc embeddings: embed a representative set, then tile with noise to
     # n_docs (index scale is what matters; embedding all n_docs would
     # measure the embedder, which the wordcount half already covers)
-    base = torch.stack([
-        torch.from_numpy(v) if isinstance(v, np.ndarray) else v
-        for v in [torch.tensor(e) for e in enc.encode(docs, batch_size=1024)]
-    ]).to(device)
+    base = torch.stack(
+        [torch.from_numpy(np.asarray(e)) for e in enc.encode(docs, batch_size=1024)]
+    ).to(device)
     reps = (n_docs + base.shape[0] - 1) // base.shape[0]
     vecs = base.repeat(reps, 1)[:n_docs]
     vecs = vecs + 0.01 * torch.randn(
