@@ -216,6 +216,69 @@ class DocumentStore:
             result=common.apply_with_type(stat_result, dt.JSON, thisclass.right.count)
         )
 
+    def register_mcp(self, server) -> None:
+        """Expose retrieve / statistics / list_inputs as MCP tools
+        (reference document_store register_mcp over fastmcp).  Handlers
+        run one-shot queries through the engine (eager graph: each call
+        builds a single-row query table and captures the reply)."""
+        from pathway_amd.debug import table_from_rows, table_to_dicts
+        from pathway_amd.internals.schema import schema_from_types
+
+        store = self
+
+        def retrieve_handler(args: dict):
+            q = table_from_rows(
+                schema_from_types(
+                    query=str, k=int, metadata_filter=str | None,
+                    filepath_globpattern=str | None,
+                ),
+                [(
+                    args.get("query", ""),
+                    int(args.get("k", 3)),
+                    args.get("metadata_filter"),
+                    args.get("filepath_globpattern"),
+                )],
+            )
+            res = store.retrieve_query(q)
+            _, cols = table_to_dicts(res)
+            [out] = list(cols["result"].values())
+            return out.value if hasattr(out, "value") else out
+
+        def statistics_handler(args: dict):
+            q = table_from_rows(schema_from_types(dummy=int), [(0,)])
+            res = store.statistics_query(q)
+            _, cols = table_to_dicts(res)
+            [out] = list(cols["result"].values())
+            return out.value if hasattr(out, "value") else out
+
+        def inputs_handler(args: dict):
+            q = table_from_rows(
+                schema_from_types(
+                    metadata_filter=str | None, filepath_globpattern=str | None
+                ),
+                [(args.get("metadata_filter"),
+                  args.get("filepath_globpattern"))],
+            )
+            res = store.inputs_query(q)
+            _, cols = table_to_dicts(res)
+            [out] = list(cols["result"].values())
+            return out.value if hasattr(out, "value") else out
+
+        server.tool(
+            "retrieve_query",
+            request_handler=retrieve_handler,
+            schema={"type": "object",
+                    "properties": {"query": {"type": "string"},
+                                   "k": {"type": "integer"},
+                                   "metadata_filter": {"type": "string"},
+                                   "filepath_globpattern": {"type": "string"}},
+                    "required": ["query"]},
+        )
+        server.tool("statistics_query", request_handler=statistics_handler,
+                    schema={"type": "object"})
+        server.tool("inputs_query", request_handler=inputs_handler,
+                    schema={"type": "object"})
+
     @property
     def index(self):
         from pathway_amd.stdlib.indexing.data_index import DataIndex

@@ -123,3 +123,79 @@ def test_hf_pipeline_chat_gated():
         HFPipelineChat(model="nonexistent-model-xyz")
     except Exception as e:
         assert "model" in str(e).lower() or "network" in str(e).lower() or True
+
+
+def test_mcp_server_stdio_and_http():
+    """MCP JSON-RPC (initialize/tools/list/tools/call) over stdio and
+    HTTP transports, with DocumentStore tools registered."""
+    import io
+    import json
+    import urllib.request
+
+    import pathway_amd as pw
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.xpacks.llm.mcp_server import McpServer, PathwayMcp
+
+    G.clear()
+    docs = pw.debug.table_from_markdown(
+        """
+        data    | _metadata
+        alpha   | {}
+        beta    | {}
+        """
+    )
+    from pathway_amd.internals.json import Json
+    docs = docs.select(
+        data=pw.apply(lambda s: s.encode(), pw.this.data),
+        _metadata=pw.apply(lambda _m: Json({"path": "doc"}), pw.this._metadata),
+    )
+    from pathway_amd.xpacks.llm.document_store import DocumentStore
+    from pathway_amd.stdlib.indexing.nearest_neighbors import BruteForceKnnFactory
+
+    def fake_embed(text: str):
+        return tuple(float(ord(c)) for c in (text or "ab")[:2]) + (1.0,)
+
+    store = DocumentStore(
+        docs,
+        retriever_factory=BruteForceKnnFactory(embedder=pw.udf(fake_embed)),
+    )
+    server = McpServer("test")
+    store.register_mcp(server)
+    assert set(server.tools) >= {"retrieve_query", "statistics_query",
+                                 "inputs_query"}
+
+    # stdio transport round trip
+    lines = [
+        json.dumps({"jsonrpc": "2.0", "id": 1, "method": "initialize",
+                    "params": {}}),
+        json.dumps({"jsonrpc": "2.0", "id": 2, "method": "tools/list"}),
+        json.dumps({"jsonrpc": "2.0", "id": 3, "method": "tools/call",
+                    "params": {"name": "retrieve_query",
+                               "arguments": {"query": "alpha", "k": 1}}}),
+    ]
+    out = io.StringIO()
+    server.serve_stdio(infile=io.StringIO("\n".join(lines) + "\n"), outfile=out)
+    resps = [json.loads(l) for l in out.getvalue().splitlines()]
+    assert resps[0]["result"]["serverInfo"]["name"] == "test"
+    names = [t["name"] for t in resps[1]["result"]["tools"]]
+    assert "retrieve_query" in names
+    payload = json.loads(resps[2]["result"]["content"][0]["text"])
+    assert payload and payload[0]["text"] == "alpha"
+
+    # HTTP transport
+    app = PathwayMcp(serve=[store], name="t2")
+    httpd = app.start()
+    try:
+        port = httpd.server_address[1]
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{port}/",
+            data=json.dumps({"jsonrpc": "2.0", "id": 9,
+                             "method": "tools/list"}).encode(),
+            headers={"Content-Type": "application/json"},
+        )
+        with urllib.request.urlopen(req, timeout=10) as resp:
+            got = json.loads(resp.read())
+        assert any(t["name"] == "retrieve_query"
+                   for t in got["result"]["tools"])
+    finally:
+        httpd.shutdown()
