@@ -333,3 +333,39 @@ def test_async_transformer_views_and_options():
     nfail = tr2.failed.reduce(c=pw.reducers.count())
     _, cols2 = pw.debug.table_to_dicts(nfail)
     assert list(cols2["c"].values()) == [1]  # only a==3 failed
+
+
+def test_viz_live_plot():
+    """Native live plot (reference stdlib/viz): subscriber state + SVG
+    renderer + HTTP endpoint."""
+    import json
+    import urllib.request
+
+    import pathway_amd as pw
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.stdlib.viz import plot, render_svg
+
+    svg = render_svg([1.0, 3.0, 2.0])
+    assert svg.startswith("<svg") and "polyline" in svg
+
+    G.clear()
+    t = pw.debug.table_from_markdown(
+        """
+        v | label
+        1 | a
+        5 | b
+        3 | c
+        """
+    )
+    lp = plot(t, value_column="v", serve=True)
+    pw.run()
+    port = lp.server.server_address[1]
+    with urllib.request.urlopen(f"http://127.0.0.1:{port}/data", timeout=10) as r:
+        data = json.loads(r.read())
+    assert data["columns"] == ["v", "label"]
+    assert len(data["rows"]) == 3
+    assert "polyline" in data["svg"]
+    with urllib.request.urlopen(f"http://127.0.0.1:{port}/", timeout=10) as r:
+        page = r.read().decode()
+    assert "pathway table" in page
+    lp.server.shutdown()
