@@ -173,3 +173,57 @@ def test_rabbitmq_roundtrip():
         assert sorted(cols["k"].values()) == [5, 7]
     finally:
         srv.stop()
+
+
+def test_pulsar_roundtrip():
+    import json
+    import threading
+    import time
+
+    pytest.importorskip("aiohttp")
+    from pathway_amd.io.pulsar import PulsarWsProducer
+    from tests.fakes.fake_pulsar import FakePulsar
+
+    srv = FakePulsar().start()
+    try:
+        # producer protocol
+        p = PulsarWsProducer(srv.url, "t1")
+        p.send(b"hello")
+        p.close()
+        tpath = "persistent/public/default/t1"
+        assert srv.messages.get(tpath) == [b"hello"]
+
+        # table write
+        G.clear()
+        t = pw.debug.table_from_markdown(
+            """
+            a | b
+            1 | x
+            2 | y
+            """
+        )
+        pw.io.pulsar.write(t, srv.url, "rows")
+        pw.run()
+        recs = [json.loads(m) for m in
+                srv.messages.get("persistent/public/default/rows", [])]
+        assert sorted((r["a"], r["b"]) for r in recs) == [(1, "x"), (2, "y")]
+
+        # streaming read (backlog + live)
+        G.clear()
+        tbl = pw.io.pulsar.read(
+            srv.url, "live", schema=schema_from_types(k=int), format="json",
+            subscription="sub1", _max_messages=2,
+        )
+
+        def later():
+            time.sleep(0.3)
+            p2 = PulsarWsProducer(srv.url, "live")
+            p2.send(json.dumps({"k": 5}).encode())
+            p2.send(json.dumps({"k": 7}).encode())
+            p2.close()
+
+        threading.Thread(target=later, daemon=True).start()
+        keys, cols = pw.debug.table_to_dicts(tbl)
+        assert sorted(cols["k"].values()) == [5, 7]
+    finally:
+        srv.stop()
