@@ -25,6 +25,8 @@ class _NetworkChat(BaseChat):
                  cache_strategy=None, temperature=None, max_tokens=None, **kwargs):
         super().__init__(cache_strategy=cache_strategy)
         self.model = model
+        self.temperature = temperature
+        self.max_tokens = max_tokens
         self.kwargs = kwargs
 
     def __wrapped__(self, messages: Any, **kwargs) -> str:
@@ -35,10 +37,53 @@ class _NetworkChat(BaseChat):
 
 
 class OpenAIChat(_NetworkChat):
+    """OpenAI chat-completions protocol (reference llms.py:43 OpenAIChat).
+
+    A real client for any OpenAI-compatible endpoint (OpenAI, vLLM,
+    llama.cpp server, TGI in openai mode): pass ``base_url`` to point it
+    at the server.  The request/response shapes are the standard
+    ``/chat/completions`` JSON — exercised against the fake HTTP service
+    in tests (no network in this environment, so the public api.openai.com
+    default is only reachable in deployments with egress)."""
+
     provider = "OpenAI"
 
+    def __init__(self, model: str | None = None, *, api_key: str | None = None,
+                 base_url: str | None = None, **kwargs):
+        super().__init__(model, **kwargs)
+        self.api_key = api_key
+        self.base_url = (base_url or "https://api.openai.com/v1").rstrip("/")
 
-class LiteLLMChat(_NetworkChat):
+    def __wrapped__(self, messages: Any, **kwargs) -> str:
+        from pathway_amd.io import _rest
+
+        if isinstance(messages, str):
+            messages = [{"role": "user", "content": messages}]
+        elif isinstance(messages, tuple):
+            messages = [
+                dict(m) if isinstance(m, dict) else {"role": "user", "content": str(m)}
+                for m in messages
+            ]
+        body: dict[str, Any] = {"model": self.model, "messages": messages}
+        if self.temperature is not None:
+            body["temperature"] = self.temperature
+        if self.max_tokens is not None:
+            body["max_tokens"] = self.max_tokens
+        body.update(kwargs)
+        headers = {}
+        if self.api_key:
+            headers["Authorization"] = f"Bearer {self.api_key}"
+        out = _rest.request(
+            "POST", f"{self.base_url}/chat/completions", body=body,
+            headers=headers,
+        )
+        return out["choices"][0]["message"]["content"]
+
+
+class LiteLLMChat(OpenAIChat):
+    """LiteLLM proxy speaks the OpenAI protocol; same client with the
+    proxy's base_url (reference llms.py LiteLLMChat)."""
+
     provider = "LiteLLM"
 
 

@@ -199,3 +199,41 @@ def test_mcp_server_stdio_and_http():
                    for t in got["result"]["tools"])
     finally:
         httpd.shutdown()
+
+
+def test_openai_chat_protocol():
+    """OpenAIChat speaks the real /chat/completions protocol — verified
+    against the capturing fake HTTP service (works with vLLM/llama.cpp
+    OpenAI-compatible servers)."""
+    import pathway_amd as pw
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.xpacks.llm.llms import LiteLLMChat, OpenAIChat
+    from tests.fakes.fake_http import FakeHTTPService
+
+    srv = FakeHTTPService().start()
+    srv.replies["/chat/completions"] = (200, {
+        "choices": [{"message": {"role": "assistant", "content": "42"}}]
+    })
+    try:
+        chat = OpenAIChat(model="m1", api_key="sk-x", base_url=srv.url,
+                          temperature=0.5)
+        G.clear()
+        t = pw.debug.table_from_markdown(
+            """
+            q
+            what_is_the_answer
+            """
+        )
+        res = t.select(a=chat(pw.this.q))
+        _, cols = pw.debug.table_to_dicts(res)
+        assert list(cols["a"].values()) == ["42"]
+        [req] = [r for r in srv.requests if r.path == "/chat/completions"]
+        body = req.json()
+        assert body["model"] == "m1"
+        assert body["temperature"] == 0.5
+        assert body["messages"][0] == {"role": "user",
+                                       "content": "what_is_the_answer"}
+        assert req.headers.get("Authorization") == "Bearer sk-x"
+        assert issubclass(LiteLLMChat, OpenAIChat)
+    finally:
+        srv.stop()
