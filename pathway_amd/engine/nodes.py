@@ -769,6 +769,36 @@ class GroupReduceNode(Node):
             n: concat_columns([self.add_carried[n], gcols_first[n]])
             for n in self.add_carried
         }
+        if (
+            torch.device(device).type == "cuda"
+            and not _PW_NO_SEGRED
+            and all(a.dtype == torch.int64 for a in all_accs.values())
+            and 1 <= len(self.add_accs) <= 8
+        ):
+            # fused LSM merge+consolidate (one count+emit kernel pair):
+            # unique-sorted state x unique-sorted delta, weight-0 rows
+            # dropped, representative row indices for carried columns
+            from pathway_amd import ops
+
+            names2 = list(self.add_accs)
+            # weight slot first (the kernel drops on acc[0] == 0)
+            names2.sort(key=lambda nm: nm != "__w__")
+            a_accs = [self.add_accs[nm].to(torch.int64) for nm in names2]
+            b_accs = [
+                acc_deltas.get(
+                    nm, torch.zeros(nseg, dtype=torch.int64, device=device)
+                )
+                for nm in names2
+            ]
+            out_words, out_accs, rep = ops.merge_consolidate_gpu(
+                self.add_keys, a_accs, ukeys_w, b_accs
+            )
+            self.add_keys = out_words
+            self.add_accs = dict(zip(names2, out_accs))
+            self.add_carried = {
+                n: c.take(rep) for n, c in all_carried.items()
+            }
+            return
         # state and delta are both sorted: O(m+n) merge, no re-sort
         from pathway_amd.engine.state import merge_sorted_select
 

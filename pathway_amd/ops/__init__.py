@@ -499,3 +499,80 @@ def topk_gpu(scores: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:
     if rc != 0:
         raise RuntimeError(f"pw_topk failed: hip error {rc}")
     return vals, idx
+
+
+def merge_consolidate_gpu(
+    a_words: Sequence[torch.Tensor],
+    a_accs: Sequence[torch.Tensor],
+    b_words: Sequence[torch.Tensor],
+    b_accs: Sequence[torch.Tensor],
+) -> tuple[list[torch.Tensor], list[torch.Tensor], torch.Tensor]:
+    """Fused LSM merge+consolidate of two unique lex-sorted key sets.
+
+    acc slot 0 is the weight; merged rows with zero weight are dropped.
+    Returns ([k0, k1], accs, rep) with rep indexing concat([A, B]) rows
+    (A preferred on matches) for carried-column gathers.
+    """
+    lib = require_lib()
+    m = a_words[0].shape[0]
+    n = b_words[0].shape[0]
+    nacc = len(a_accs)
+    assert len(b_accs) == nacc and 1 <= nacc <= 8
+    device = a_words[0].device
+    total_diag = m + n
+    nthreads = max(1, (total_diag + 7) // 8)
+    counts = torch.empty(nthreads, dtype=torch.int32, device=device)
+    a0 = a_words[0].contiguous()
+    a1 = a_words[1].contiguous()
+    b0 = b_words[0].contiguous()
+    b1 = b_words[1].contiguous()
+    aA = [t.contiguous() for t in a_accs]
+    aB = [t.contiguous() for t in b_accs]
+    rc = lib.pw_merge_consolidate_count(
+        ctypes.c_void_p(a0.data_ptr()),
+        ctypes.c_void_p(a1.data_ptr()),
+        _ptr_arr(aA),
+        ctypes.c_void_p(b0.data_ptr()),
+        ctypes.c_void_p(b1.data_ptr()),
+        _ptr_arr(aB),
+        ctypes.c_int(nacc),
+        ctypes.c_int64(m),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(counts.data_ptr()),
+        ctypes.c_int64(nthreads),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_merge_consolidate_count failed: {rc}")
+    csum = torch.cumsum(counts.to(torch.int64), 0)
+    total = int(csum[-1].item())
+    bases = torch.zeros(nthreads, dtype=torch.int64, device=device)
+    bases[1:] = csum[:-1]
+    o0 = torch.empty(total, dtype=torch.int64, device=device)
+    o1 = torch.empty(total, dtype=torch.int64, device=device)
+    out_accs = [
+        torch.empty(total, dtype=torch.int64, device=device)
+        for _ in range(nacc)
+    ]
+    rep = torch.empty(total, dtype=torch.int64, device=device)
+    rc = lib.pw_merge_consolidate_emit(
+        ctypes.c_void_p(a0.data_ptr()),
+        ctypes.c_void_p(a1.data_ptr()),
+        _ptr_arr(aA),
+        ctypes.c_void_p(b0.data_ptr()),
+        ctypes.c_void_p(b1.data_ptr()),
+        _ptr_arr(aB),
+        ctypes.c_int(nacc),
+        ctypes.c_int64(m),
+        ctypes.c_int64(n),
+        ctypes.c_void_p(bases.data_ptr()),
+        ctypes.c_int64(nthreads),
+        ctypes.c_void_p(o0.data_ptr()),
+        ctypes.c_void_p(o1.data_ptr()),
+        _ptr_arr(out_accs),
+        ctypes.c_void_p(rep.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_merge_consolidate_emit failed: {rc}")
+    return [o0, o1], out_accs, rep

@@ -1066,3 +1066,186 @@ extern "C" int pw_topk(const void* scores, int64_t nq, int64_t m, int k,
                      (int64_t*)out_idx);
   return (int)hipGetLastError();
 }
+
+// --------------------------------------------------- merge-consolidate --
+// Fused LSM merge of the additive-reduce state:
+// two UNIQUE lex-sorted 128-bit-key row sets (state A, delta B) with
+// int64 accumulators merge into unique sorted keys, accumulators summed
+// on key matches, rows whose weight slot (acc 0) sums to zero dropped.
+// Replaces the searchsorted-merge + gather + segmented-reduce + nonzero
+// + gather chain (~10 kernels) with count + emit over merge-path
+// partitions (per-thread diagonal chunks; matched pairs are owned by
+// the A side, a range starting on the B half of a match skips it).
+// rep[] returns each output row's source row (j in A, or m + i in B;
+// A preferred on matches) for carried-column gathers.
+
+struct McAcc {
+  const long long* p[8];
+};
+struct McAccMut {
+  long long* p[8];
+};
+
+__device__ __forceinline__ bool mc_le(const int64_t* a0, const int64_t* a1,
+                                      int64_t i, const int64_t* b0,
+                                      const int64_t* b1, int64_t j) {
+  if (a0[i] != b0[j]) return a0[i] < b0[j];
+  return a1[i] <= b1[j];
+}
+
+__device__ __forceinline__ bool mc_eq(const int64_t* a0, const int64_t* a1,
+                                      int64_t i, const int64_t* b0,
+                                      const int64_t* b1, int64_t j) {
+  return a0[i] == b0[j] && a1[i] == b1[j];
+}
+
+__device__ int64_t mc_merge_path(const int64_t* a0, const int64_t* a1,
+                                 int64_t m, const int64_t* b0,
+                                 const int64_t* b1, int64_t n, int64_t d) {
+  int64_t lo = d > n ? d - n : 0;
+  int64_t hi = d < m ? d : m;
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    if (mc_le(a0, a1, mid, b0, b1, d - mid - 1))
+      lo = mid + 1;
+    else
+      hi = mid;
+  }
+  return lo;  // ai; bi = d - ai
+}
+
+#define PW_MC_CHUNK 8
+
+// one walk function used by both phases: WRITE=false counts only
+template <bool WRITE>
+__device__ void mc_walk(const int64_t* a0, const int64_t* a1, McAcc accA,
+                        const int64_t* b0, const int64_t* b1, McAcc accB,
+                        int nacc, int64_t m, int64_t n, int64_t d0,
+                        int64_t d1, int* count_out, int64_t base,
+                        int64_t* o0, int64_t* o1, McAccMut accO,
+                        int64_t* rep) {
+  int64_t ai = mc_merge_path(a0, a1, m, b0, b1, n, d0);
+  int64_t bi = d0 - ai;
+  int64_t d = d0;
+  // skip rule: a range starting on the B half of a match
+  if (d < d1 && bi < n && ai > 0 && !(bi >= n) &&
+      !(ai < m && mc_le(a0, a1, ai, b0, b1, bi))) {
+    if (mc_eq(a0, a1, ai - 1, b0, b1, bi)) {
+      ++bi;
+      ++d;
+    }
+  }
+  int cnt = 0;
+  int64_t w = base;
+  while (d < d1) {
+    bool take_a = (bi >= n) || (ai < m && mc_le(a0, a1, ai, b0, b1, bi));
+    if (take_a) {
+      bool matched = (bi < n) && mc_eq(a0, a1, ai, b0, b1, bi);
+      long long wsum = accA.p[0][ai] + (matched ? accB.p[0][bi] : 0ll);
+      if (wsum != 0) {
+        if (WRITE) {
+          o0[w] = a0[ai];
+          o1[w] = a1[ai];
+          for (int c = 0; c < nacc; ++c)
+            accO.p[c][w] =
+                accA.p[c][ai] + (matched ? accB.p[c][bi] : 0ll);
+          rep[w] = ai;
+          ++w;
+        }
+        ++cnt;
+      }
+      ++ai;
+      ++d;
+      if (matched) {
+        ++bi;
+        ++d;
+      }
+    } else {
+      long long wsum = accB.p[0][bi];
+      if (wsum != 0) {
+        if (WRITE) {
+          o0[w] = b0[bi];
+          o1[w] = b1[bi];
+          for (int c = 0; c < nacc; ++c) accO.p[c][w] = accB.p[c][bi];
+          rep[w] = m + bi;
+          ++w;
+        }
+        ++cnt;
+      }
+      ++bi;
+      ++d;
+    }
+  }
+  if (!WRITE) *count_out = cnt;
+}
+
+__global__ void k_mc_count(const int64_t* a0, const int64_t* a1, McAcc accA,
+                           const int64_t* b0, const int64_t* b1, McAcc accB,
+                           int nacc, int64_t m, int64_t n,
+                           int* thread_counts, int64_t nthreads) {
+  int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= nthreads) return;
+  int64_t total = m + n;
+  int64_t d0 = t * PW_MC_CHUNK;
+  int64_t d1 = min(d0 + PW_MC_CHUNK, total);
+  McAccMut dummy{};
+  mc_walk<false>(a0, a1, accA, b0, b1, accB, nacc, m, n, d0, d1,
+                 &thread_counts[t], 0, nullptr, nullptr, dummy, nullptr);
+}
+
+__global__ void k_mc_emit(const int64_t* a0, const int64_t* a1, McAcc accA,
+                          const int64_t* b0, const int64_t* b1, McAcc accB,
+                          int nacc, int64_t m, int64_t n,
+                          const int64_t* bases, int64_t nthreads,
+                          int64_t* o0, int64_t* o1, McAccMut accO,
+                          int64_t* rep) {
+  int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= nthreads) return;
+  int64_t total = m + n;
+  int64_t d0 = t * PW_MC_CHUNK;
+  int64_t d1 = min(d0 + PW_MC_CHUNK, total);
+  mc_walk<true>(a0, a1, accA, b0, b1, accB, nacc, m, n, d0, d1, nullptr,
+                bases[t], o0, o1, accO, rep);
+}
+
+extern "C" int pw_merge_consolidate_count(
+    const void* a0, const void* a1, const void** accA, const void* b0,
+    const void* b1, const void** accB, int nacc, int64_t m, int64_t n,
+    void* thread_counts, int64_t nthreads, void* stream) {
+  if (nacc > 8 || nacc < 1) return 2;
+  McAcc A{}, B{};
+  for (int c = 0; c < nacc; ++c) {
+    A.p[c] = (const long long*)accA[c];
+    B.p[c] = (const long long*)accB[c];
+  }
+  hipStream_t s = (hipStream_t)stream;
+  int64_t blocks = (nthreads + PW_BLOCK - 1) / PW_BLOCK;
+  hipLaunchKernelGGL(k_mc_count, dim3((uint32_t)blocks), dim3(PW_BLOCK), 0,
+                     s, (const int64_t*)a0, (const int64_t*)a1, A,
+                     (const int64_t*)b0, (const int64_t*)b1, B, nacc, m, n,
+                     (int*)thread_counts, nthreads);
+  return (int)hipGetLastError();
+}
+
+extern "C" int pw_merge_consolidate_emit(
+    const void* a0, const void* a1, const void** accA, const void* b0,
+    const void* b1, const void** accB, int nacc, int64_t m, int64_t n,
+    const void* bases, int64_t nthreads, void* o0, void* o1, void** accO,
+    void* rep, void* stream) {
+  if (nacc > 8 || nacc < 1) return 2;
+  McAcc A{}, B{};
+  McAccMut O{};
+  for (int c = 0; c < nacc; ++c) {
+    A.p[c] = (const long long*)accA[c];
+    B.p[c] = (const long long*)accB[c];
+    O.p[c] = (long long*)accO[c];
+  }
+  hipStream_t s = (hipStream_t)stream;
+  int64_t blocks = (nthreads + PW_BLOCK - 1) / PW_BLOCK;
+  hipLaunchKernelGGL(k_mc_emit, dim3((uint32_t)blocks), dim3(PW_BLOCK), 0,
+                     s, (const int64_t*)a0, (const int64_t*)a1, A,
+                     (const int64_t*)b0, (const int64_t*)b1, B, nacc, m, n,
+                     (const int64_t*)bases, nthreads, (int64_t*)o0,
+                     (int64_t*)o1, O, (int64_t*)rep);
+  return (int)hipGetLastError();
+}
