@@ -166,3 +166,54 @@ class WorkloadTracker:
         if avg < self.low:
             return "down"
         return None
+
+
+class DetailedMetricsRecorder:
+    """Per-operator time series into SQLite (reference detailed-metrics
+    dir + web dashboard: set_monitoring_config(detailed_metrics_dir=...)
+    -> metrics.db read by web_dashboard/dashboard.py)."""
+
+    def __init__(self, directory: str, run_id: str | None = None):
+        import os
+        import sqlite3
+        import uuid
+
+        os.makedirs(directory, exist_ok=True)
+        self.path = os.path.join(directory, "metrics.db")
+        self.run_id = run_id or uuid.uuid4().hex[:12]
+        self.conn = sqlite3.connect(self.path)
+        self.conn.execute(
+            "CREATE TABLE IF NOT EXISTS operator_metrics ("
+            " run_id TEXT, ts REAL, engine_time INTEGER, operator TEXT,"
+            " steps INTEGER, rows_in INTEGER, rows_out INTEGER,"
+            " total_time_s REAL)"
+        )
+        self.conn.execute(
+            "CREATE TABLE IF NOT EXISTS run_metrics ("
+            " run_id TEXT, ts REAL, engine_time INTEGER, steps INTEGER,"
+            " rows_ingested INTEGER, rows_output INTEGER, p50_ms REAL,"
+            " p95_ms REAL)"
+        )
+        self.conn.commit()
+
+    def record(self, stats: "RunStats") -> None:
+        import time as _t
+
+        now = _t.time()
+        s = stats.snapshot()
+        self.conn.execute(
+            "INSERT INTO run_metrics VALUES (?,?,?,?,?,?,?,?)",
+            (self.run_id, now, s["current_time"] or 0, s["steps"],
+             s["rows_ingested"], s["rows_output"],
+             s["p50_step_ms"] or 0.0, s["p95_step_ms"] or 0.0),
+        )
+        for name, st in stats.operators.items():
+            self.conn.execute(
+                "INSERT INTO operator_metrics VALUES (?,?,?,?,?,?,?,?)",
+                (self.run_id, now, s["current_time"] or 0, name, st.steps,
+                 st.rows_in, st.rows_out, st.total_time_s),
+            )
+        self.conn.commit()
+
+    def close(self) -> None:
+        self.conn.close()

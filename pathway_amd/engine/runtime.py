@@ -318,6 +318,14 @@ class Runtime:
 
         self.stats = RunStats()
         self.monitor = None
+        self._metrics_recorder = None
+        from pathway_amd.internals.config import pathway_config
+
+        mdir = getattr(pathway_config, "detailed_metrics_dir", None)
+        if mdir:
+            from pathway_amd.engine.monitoring import DetailedMetricsRecorder
+
+            self._metrics_recorder = DetailedMetricsRecorder(mdir)
         for i, src in enumerate(self.sources):
             if not getattr(src, "persistent_id", None):
                 src.persistent_id = getattr(src.source, "name", None) or f"src{i}"
@@ -509,6 +517,11 @@ class Runtime:
             ingested, output = self.step_once(t)
             _now = _time.perf_counter()
             self.stats.record_step(t, _now - _s0, ingested, output)
+            if self._metrics_recorder is not None and (
+                _now - getattr(self, "_last_metrics_flush", 0.0) >= 1.0
+            ):
+                self._metrics_recorder.record(self.stats)
+                self._last_metrics_flush = _now
             if elastic is not None:
                 busy = (_now - _s0) / max(_now - _last_loop, 1e-9)
                 advice = elastic.add_point(min(busy, 1.0))
@@ -544,6 +557,8 @@ class Runtime:
             steps += 1
             if max_steps is not None and steps >= max_steps:
                 break
+        if self._metrics_recorder is not None:
+            self._metrics_recorder.record(self.stats)
         for node in self.nodes:
             if isinstance(node, SubscribeNode):
                 node.finish()

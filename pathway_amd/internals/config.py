@@ -21,6 +21,7 @@ class PathwayConfig:
     runtime_typechecking: bool = False
     license_key: str | None = os.environ.get("PATHWAY_LICENSE_KEY")
     monitoring_server: str | None = None
+    detailed_metrics_dir: str | None = None
     process_id: int = int(os.environ.get("PATHWAY_PROCESS_ID", "0"))
     processes: int = int(os.environ.get("PATHWAY_PROCESSES", "1"))
     threads: int = int(os.environ.get("PATHWAY_THREADS", "1"))
@@ -57,5 +58,8 @@ def set_license_key(key: str | None) -> None:
     pathway_config.license_key = key
 
 
-def set_monitoring_config(*, server_endpoint: str | None = None, **kwargs) -> None:
+def set_monitoring_config(*, server_endpoint: str | None = None,
+                          detailed_metrics_dir: str | None = None,
+                          **kwargs) -> None:
     pathway_config.monitoring_server = server_endpoint
+    pathway_config.detailed_metrics_dir = detailed_metrics_dir

@@ -369,3 +369,49 @@ def test_viz_live_plot():
         page = r.read().decode()
     assert "pathway table" in page
     lp.server.shutdown()
+
+
+def test_detailed_metrics_and_dashboard(tmp_path):
+    """set_monitoring_config(detailed_metrics_dir=...) -> metrics.db;
+    the web dashboard serves /api/history from it (reference
+    integration_tests/monitoring behavior)."""
+    import sqlite3
+
+    import pathway_amd as pw
+    from pathway_amd.internals.config import pathway_config
+    from pathway_amd.internals.rungraph import G
+
+    mdir = str(tmp_path / "metrics")
+    pw.set_monitoring_config(detailed_metrics_dir=mdir)
+    try:
+        G.clear()
+        t = T(
+            """
+            a
+            1
+            2
+            """
+        )
+        out = str(tmp_path / "o.csv")
+        pw.io.csv.write(t, out)
+        pw.run()
+        db = sqlite3.connect(f"{mdir}/metrics.db")
+        runs = db.execute("SELECT COUNT(*) FROM run_metrics").fetchone()[0]
+        ops = db.execute(
+            "SELECT COUNT(DISTINCT operator) FROM operator_metrics"
+        ).fetchone()[0]
+        db.close()
+        assert runs >= 1 and ops >= 1
+
+        from starlette.testclient import TestClient
+
+        from pathway_amd.web_dashboard.dashboard import create_app
+
+        app = create_app(detailed_metrics_dir=mdir)
+        client = TestClient(app)
+        hist = client.get("/api/history").json()
+        assert hist["points"] and hist["operators"]
+        page = client.get("/").text
+        assert "operators" in page and "svg" in page
+    finally:
+        pw.set_monitoring_config(detailed_metrics_dir=None)
