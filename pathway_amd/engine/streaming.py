@@ -65,10 +65,15 @@ class StreamingSource(Source):
 
         while True:
             if self.q.empty():
+                # order matters: observe finished FIRST — finish() is the
+                # reader's last call, so an empty queue seen after it is
+                # final; the reverse order can strand the last rows
                 if self._finished.is_set():
-                    return
-                _time.sleep(0.002)
-                continue
+                    if self.q.empty():
+                        return
+                else:
+                    _time.sleep(0.002)
+                    continue
             # drain under the lock so next_time's done-check never sees
             # rows vanish mid-transfer
             with self._drain_lock:
