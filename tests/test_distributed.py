@@ -6,6 +6,16 @@ import pytest
 import torch.multiprocessing as mp
 
 
+def _free_port() -> int:
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 def _worker_groupby(rank: int, world: int, port: int, q):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
@@ -378,3 +388,80 @@ def test_distributed_update_rows_gloo():
     union = sorted(results[0] + results[1])
     # key 2's base (rank 0) must meet its override (rank 1)
     assert union == [(10,), (30,), (99,)]
+
+
+def _worker_string_exchange(rank: int, world: int, port: int, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["PW_DEVICE"] = "cpu"
+    import pathway_amd as pw
+    import pathway_amd.parallel as par
+    from pathway_amd.debug import table_from_markdown as T
+    from pathway_amd.engine.column import GLOBAL_STRING_POOL
+
+    par.init(backend="gloo")
+    # force the byte-shipping path: pools are NOT synchronized and each
+    # rank interns different strings first (codes disagree across ranks)
+    GLOBAL_STRING_POOL.synchronized = False
+    if rank == 0:
+        GLOBAL_STRING_POOL.codes(["zzz", "yyy"])
+        t = T(
+            """
+            w      | v
+            alpha  | 1
+            beta   | 2
+            żółć   | 7
+            """
+        )
+    else:
+        GLOBAL_STRING_POOL.codes(["other"])
+        t = T(
+            """
+            w      | v
+            alpha  | 10
+            żółć   | 3
+            """
+        )
+    res = t.groupby(pw.this.w).reduce(
+        pw.this.w, s=pw.reducers.sum(pw.this.v), c=pw.reducers.count()
+    )
+    cap = res._capture()
+    from pathway_amd.engine.runtime import Runtime
+    from pathway_amd.internals.rungraph import reset_all
+
+    rt = Runtime([cap], comm=par.get_comm())
+    reset_all(rt.nodes)
+    rt.run()
+    from pathway_amd.internals.api import squash_updates
+
+    state = squash_updates(cap.rows)
+    rows = sorted(tuple(v) for v in state.values())
+    q.put((rank, rows))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_distributed_string_bytes_exchange():
+    """Unsynchronized string pools exchange utf-8 bytes as tensors (no
+    host pickling) and re-intern on the receiver."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_worker_string_exchange, args=(r, 2, port, q))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, rows = q.get()
+        results[rank] = rows
+    for p in procs:
+        p.join(60)
+    merged = sorted(set(results[0]) | set(results[1]))
+    assert merged == [("alpha", 11, 2), ("beta", 2, 1), ("żółć", 10, 2)]
