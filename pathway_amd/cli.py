@@ -59,6 +59,14 @@ def _launch(program: list[str], processes: int, threads: int, first_port: int, e
 
 def spawn(args) -> None:
     processes = args.processes
+    import os as _os
+
+    from pathway_amd.internals.license import check_worker_limit
+
+    check_worker_limit(
+        processes * max(args.threads, 1),
+        _os.environ.get("PATHWAY_LICENSE_KEY"),
+    )
     while True:
         rc = _launch(
             args.program, processes, args.threads, args.first_port, {}

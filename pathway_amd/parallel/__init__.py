@@ -136,8 +136,11 @@ def init(backend: str | None = None, device=None) -> Comm:
     global _COMM
     if _COMM is None:
         _COMM = Comm(backend, device)
+        from pathway_amd.internals.config import pathway_config
+        from pathway_amd.internals.license import check_worker_limit
         from pathway_amd.internals.rungraph import G
 
+        check_worker_limit(_COMM.world, pathway_config.license_key)
         G.comm = _COMM
     return _COMM
 

@@ -487,3 +487,41 @@ def test_universe_solver_table_setops():
     assert inter._universe.is_subset_of(t._universe)
     diff = t.difference(evens)
     assert diff._universe.is_subset_of(t._universe)
+
+
+def test_license_ed25519_and_entitlements():
+    """License keys (reference license.rs): pure-python RFC 8032
+    ed25519 sign/verify, entitlement parsing, MAX_WORKERS cap."""
+    from pathway_amd.internals import license as lic
+
+    # RFC 8032 test vector 1 (empty message)
+    seed = bytes.fromhex(
+        "9d61b19deffd5a60ba844af492ec2cc44449c5697b326919703bac031cae7f60"
+    )
+    pub = lic.ed25519_public_key(seed)
+    assert pub.hex() == (
+        "d75a980182b10ab7d54bfed3c964073a0ee172f3daa62325af021a68f707511a"
+    )
+    sig = lic.ed25519_sign(seed, b"")
+    assert sig.hex() == (
+        "e5564300c360ac729086e2cc806e828a84877f1eb8e5d974d873e06522490155"
+        "5fb8821590a33bacc61e39701cf9b46bd25bf5f0595bbe24655141438e7a100b"
+    )
+    assert lic.ed25519_verify(pub, b"", sig)
+    assert not lic.ed25519_verify(pub, b"x", sig)
+
+    key = lic.issue_key(["unlimited-workers", "monitoring"])
+    parsed = lic.parse_key(key)
+    assert parsed.valid and parsed.has("unlimited-workers")
+    assert parsed.max_workers() is None
+    # free tier: cap 8
+    assert lic.parse_key(None).max_workers() == 8
+    assert lic.parse_key("garbage").max_workers() == 8
+    # a tampered key falls back to free
+    bad = key[:-6] + "AAAAAA"
+    assert not lic.parse_key(bad).valid
+
+    lic.check_worker_limit(8, None)
+    with pytest.raises(RuntimeError):
+        lic.check_worker_limit(9, None)
+    lic.check_worker_limit(64, key)
