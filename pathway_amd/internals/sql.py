@@ -1,7 +1,10 @@
 """pw.sql — SQL to dataflow translation (reference internals/sql/, ~1500 LoC).
 
-Round-1: a compact translator covering SELECT / WHERE / GROUP BY / HAVING /
-simple JOIN ... ON / UNION ALL over registered tables.
+Covers SELECT / WHERE / GROUP BY / HAVING / JOIN (inner and LEFT) ON /
+UNION ALL / WITH ctes / FROM-subqueries, and in scalar expressions:
+arithmetic, comparisons, AND/OR/NOT, IN (...) / NOT IN, BETWEEN,
+LIKE (SQL wildcards), CASE WHEN ... THEN ... [ELSE ...] END, and the
+count/sum/avg/min/max aggregates.
 """
 
 from __future__ import annotations
@@ -57,6 +60,24 @@ def _parse_scalar(sql_expr: str, table: Table) -> ex.ColumnExpression:
         if tok.startswith("'"):
             eat()
             return ex.ColumnConstExpression(tok[1:-1])
+        if tok.upper() == "CASE":
+            eat()
+            whens = []
+            else_e = ex.ColumnConstExpression(None)
+            while peek() is not None and peek().upper() == "WHEN":
+                eat()
+                cond = or_expr()
+                eat("THEN")
+                val = or_expr()
+                whens.append((cond, val))
+            if peek() is not None and peek().upper() == "ELSE":
+                eat()
+                else_e = or_expr()
+            eat("END")
+            out = else_e
+            for cond, val in reversed(whens):
+                out = ex.IfElseExpression(cond, val, out)
+            return out
         name = eat()
         lname = name.lower()
         if lname in _AGGS and peek() == "(":
@@ -97,6 +118,52 @@ def _parse_scalar(sql_expr: str, table: Table) -> ex.ColumnExpression:
 
     def cmp_expr():
         e = add_expr()
+        negate = False
+        if peek() is not None and peek().upper() == "NOT":
+            nxt = toks[pos[0] + 1] if pos[0] + 1 < len(toks) else ""
+            if nxt.upper() in ("IN", "LIKE", "BETWEEN"):
+                eat()
+                negate = True
+        if peek() is not None and peek().upper() == "IN":
+            eat()
+            eat("(")
+            vals = [or_expr()]
+            while peek() == ",":
+                eat()
+                vals.append(or_expr())
+            eat(")")
+            out = None
+            for v in vals:
+                term = e == v
+                out = term if out is None else (out | term)
+            return ~out if negate else out
+        if peek() is not None and peek().upper() == "BETWEEN":
+            eat()
+            lo = add_expr()
+            eat("AND")
+            hi = add_expr()
+            out = (e >= lo) & (e <= hi)
+            return ~out if negate else out
+        if peek() is not None and peek().upper() == "LIKE":
+            eat()
+            pat = or_expr()
+            if not isinstance(pat, ex.ColumnConstExpression):
+                raise _SqlError("LIKE needs a literal pattern")
+            import fnmatch as _fn
+            import re as _re
+
+            rx = _re.compile(
+                "^" + _re.escape(str(pat._value)).replace("%", ".*").replace(
+                    "_", "."
+                ).replace("\\.\\*", "%") + "$",
+                _re.S,
+            )
+            out = ex.ApplyExpression(
+                lambda s, _rx=rx: bool(s is not None and _rx.match(str(s))),
+                bool,
+                e,
+            )
+            return ~out if negate else out
         if peek() in ("=", "!=", "<>", "<", "<=", ">", ">="):
             op = eat()
             r = add_expr()
@@ -134,8 +201,54 @@ def _parse_scalar(sql_expr: str, table: Table) -> ex.ColumnExpression:
     return or_expr()
 
 
+def _split_balanced(s: str, sep: str = ",") -> list[str]:
+    out, depth, cur = [], 0, []
+    for ch in s:
+        if ch == "(":
+            depth += 1
+        elif ch == ")":
+            depth -= 1
+        if ch == sep and depth == 0:
+            out.append("".join(cur).strip())
+            cur = []
+        else:
+            cur.append(ch)
+    if cur:
+        out.append("".join(cur).strip())
+    return out
+
+
 def sql(query: str, **tables: Table) -> Table:
     q = query.strip().rstrip(";")
+    mw = re.match(r"WITH\s+(.*?)\s+(SELECT\b.*)$", q, re.I | re.S)
+    if mw:
+        # WITH a AS (...), b AS (...) SELECT ... — split on balanced commas
+        ctes = _split_balanced(mw.group(1))
+        tables = dict(tables)
+        for cte in ctes:
+            mc = re.match(r"([A-Za-z_]\w*)\s+AS\s*\((.*)\)\s*$", cte, re.I | re.S)
+            if not mc:
+                raise _SqlError(f"unsupported WITH clause {cte!r}")
+            tables[mc.group(1)] = sql(mc.group(2), **tables)
+        return sql(mw.group(2), **tables)
+    mf = re.match(r"(SELECT\s+.*?\s+FROM)\s*\(", q, re.I | re.S)
+    if mf:
+        # FROM (subquery) [alias] — find the matching close paren by scan
+        start = mf.end()
+        depth, i = 1, start
+        while i < len(q) and depth:
+            if q[i] == "(":
+                depth += 1
+            elif q[i] == ")":
+                depth -= 1
+            i += 1
+        inner = q[start : i - 1]
+        rest = q[i:]
+        ma = re.match(r"\s*(?:AS\s+)?([A-Za-z_]\w*)?(.*)$", rest, re.I | re.S)
+        tables = dict(tables)
+        alias = ma.group(1) or "_pw_sub"
+        tables[alias] = sql(inner, **tables)
+        q = f"{mf.group(1)} {alias}{ma.group(2) or ''}"
     if re.search(r"\bUNION\s+ALL\b", q, re.I):
         parts = re.split(r"\bUNION\s+ALL\b", q, flags=re.I)
         result = sql(parts[0], **tables)
@@ -144,7 +257,7 @@ def sql(query: str, **tables: Table) -> Table:
         return result
     m = re.match(
         r"SELECT\s+(?P<sel>.*?)\s+FROM\s+(?P<from>[A-Za-z_][A-Za-z0-9_]*)"
-        r"(?:\s+(?:INNER\s+)?JOIN\s+(?P<jt>[A-Za-z_][A-Za-z0-9_]*)\s+ON\s+(?P<on>.*?))?"
+        r"(?:\s+(?P<jkind>INNER\s+|LEFT\s+(?:OUTER\s+)?)?JOIN\s+(?P<jt>[A-Za-z_][A-Za-z0-9_]*)\s+ON\s+(?P<on>.*?))?"
         r"(?:\s+WHERE\s+(?P<where>.*?))?"
         r"(?:\s+GROUP\s+BY\s+(?P<gb>.*?))?"
         r"(?:\s+HAVING\s+(?P<having>.*?))?$",
@@ -172,7 +285,11 @@ def sql(query: str, **tables: Table) -> Table:
         rcol = mo.group(2).split(".")[-1]
         if lcol not in t._dtypes:
             lcol, rcol = rcol, lcol
-        j = t.join(right, t[lcol] == right[rcol])
+        jkind = (m.group("jkind") or "").strip().upper()
+        if jkind.startswith("LEFT"):
+            j = t.join_left(right, t[lcol] == right[rcol])
+        else:
+            j = t.join(right, t[lcol] == right[rcol])
         # materialize all columns of both sides (suffix disambiguation)
         t = j._all_columns_table()
     if m.group("where"):
@@ -277,3 +394,15 @@ def _rebind(e, table):
                 for v in val
             )
     return new
+
+
+def _balanced(s: str) -> bool:
+    depth = 0
+    for ch in s:
+        if ch == "(":
+            depth += 1
+        elif ch == ")":
+            depth -= 1
+            if depth < 0:
+                return False
+    return depth == 0

@@ -91,3 +91,88 @@ def test_sql_join_on():
     )
     r = pw.sql("SELECT l.v AS v, r.w AS w FROM l JOIN r ON l.k = r.k", l=l, r=rt)
     assert _rows(r, "v", "w") == [("x", 10)]
+
+
+def test_sql_in_between_like_case():
+    t = T(
+        """
+        name  | qty
+        apple | 3
+        pear  | 8
+        plum  | 15
+        kiwi  | 1
+        """
+    )
+    r = pw.sql("SELECT name FROM t WHERE name IN ('apple', 'plum')", t=t)
+    _, cols = table_to_dicts(r)
+    assert sorted(cols["name"].values()) == ["apple", "plum"]
+
+    r = pw.sql("SELECT name FROM t WHERE qty BETWEEN 2 AND 10", t=t)
+    _, cols = table_to_dicts(r)
+    assert sorted(cols["name"].values()) == ["apple", "pear"]
+
+    r = pw.sql("SELECT name FROM t WHERE name LIKE 'p%'", t=t)
+    _, cols = table_to_dicts(r)
+    assert sorted(cols["name"].values()) == ["pear", "plum"]
+
+    r = pw.sql("SELECT name FROM t WHERE name NOT IN ('apple')", t=t)
+    _, cols = table_to_dicts(r)
+    assert len(cols["name"]) == 3
+
+    r = pw.sql(
+        "SELECT name, CASE WHEN qty > 10 THEN 'big' WHEN qty > 2 THEN 'mid' "
+        "ELSE 'small' END AS size FROM t",
+        t=t,
+    )
+    _, cols = table_to_dicts(r)
+    got = dict(zip(cols["name"].values(), cols["size"].values()))
+    assert got == {"apple": "mid", "pear": "mid", "plum": "big",
+                   "kiwi": "small"}
+
+
+def test_sql_with_cte_and_subquery():
+    t = T(
+        """
+        g | v
+        a | 1
+        a | 2
+        b | 5
+        """
+    )
+    r = pw.sql(
+        "WITH sums AS (SELECT g, SUM(v) AS s FROM t GROUP BY g) "
+        "SELECT g FROM sums WHERE s > 2",
+        t=t,
+    )
+    _, cols = table_to_dicts(r)
+    assert sorted(cols["g"].values()) == ["a", "b"]
+
+    r = pw.sql(
+        "SELECT s FROM (SELECT g, SUM(v) AS s FROM t GROUP BY g) q "
+        "WHERE s >= 3",
+        t=t,
+    )
+    _, cols = table_to_dicts(r)
+    assert sorted(cols["s"].values()) == [3, 5]
+
+
+def test_sql_left_join():
+    l = T(
+        """
+        k | a
+        1 | x
+        2 | y
+        """
+    )
+    r = T(
+        """
+        k | b
+        1 | p
+        """
+    )
+    res = pw.sql("SELECT a, b FROM l LEFT JOIN r ON l.k = r.k", l=l, r=r)
+    _, cols = table_to_dicts(res)
+    got = sorted(
+        zip(cols["a"].values(), cols["b"].values()), key=str
+    )
+    assert got == [("x", "p"), ("y", None)]
