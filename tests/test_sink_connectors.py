@@ -161,7 +161,12 @@ def test_nats_roundtrip():
         )
 
         def later():
-            time.sleep(0.3)
+            # NATS pub/sub is at-most-once: wait for the reader's SUB to
+            # land before publishing (real brokers drop subscriber-less
+            # messages the same way)
+            deadline = time.time() + 10
+            while time.time() < deadline and not srv.subs.get("live"):
+                time.sleep(0.01)
             c = NatsClient(srv.uri)
             c.publish("live", json.dumps({"k": 5}).encode())
             c.publish("live", json.dumps({"k": 7}).encode())
@@ -209,7 +214,9 @@ def test_mqtt_roundtrip():
         )
 
         def later():
-            time.sleep(0.3)
+            deadline = time.time() + 10
+            while time.time() < deadline and not srv.subs.get("live"):
+                time.sleep(0.01)
             c = MqttClient(srv.uri, client_id="late")
             c.publish("live", json.dumps({"k": 1}).encode())
             c.publish("live", json.dumps({"k": 2}).encode())
