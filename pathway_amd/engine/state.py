@@ -35,6 +35,12 @@ def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
     if n <= 1:
         return perm
     if len(words) == 2 and n > 2048:
+        # INVARIANT: this fast path is sound only
+        # because every caller sorts FULL row identities — equal (k0,k1)
+        # rows are interchangeable for grouping/merging.  A caller
+        # sorting by a strict prefix of a row's identity would break
+        # silently; tests/test_fuzz_equivalence.py pins the property and
+        # PW_DEBUG_SORT=1 verifies full lex order on every call.
         k0_sorted, perm0 = torch.sort(words[0])
         k1_sorted = words[1].index_select(0, perm0)
         # only a k0 collision between rows with DIFFERENT k1 violates lex
@@ -47,6 +53,19 @@ def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
             ).any()
         )
         if not violation:
+            import os as _os
+
+            if _os.environ.get("PW_DEBUG_SORT"):
+                ok = bool(
+                    (
+                        (k0_sorted[1:] > k0_sorted[:-1])
+                        | (
+                            (k0_sorted[1:] == k0_sorted[:-1])
+                            & (k1_sorted[1:] >= k1_sorted[:-1])
+                        )
+                    ).all()
+                )
+                assert ok, "lex_sort_words fast path produced non-lex order"
             return perm0
     for w in reversed(words):
         keys = w.index_select(0, perm)

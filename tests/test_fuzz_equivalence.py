@@ -610,3 +610,37 @@ def test_to_stream_roundtrip_fuzz():
         got = sorted(zip(cols["k"].values(), cols["v"].values()))
         expected = sorted(live.items())
         assert got == expected, f"seed {seed}\n{got}\nvs\n{expected}"
+
+
+def test_lex_sort_words_fast_path_invariant():
+    """lex_sort_words' k0-only fast path must yield true lexicographic
+    order whenever rows are full identities; the
+    collision-with-different-k1 case must fall back to the stable
+    multi-pass."""
+    import torch
+
+    from pathway_amd.engine.state import lex_sort_words
+
+    g = torch.Generator().manual_seed(77)
+    # force k0 collisions with differing k1
+    k0 = torch.randint(0, 50, (5000,), dtype=torch.int64, generator=g)
+    k1 = torch.randint(-(2**62), 2**62, (5000,), dtype=torch.int64, generator=g)
+    perm = lex_sort_words([k0, k1])
+    s0 = k0.index_select(0, perm)
+    s1 = k1.index_select(0, perm)
+    ok = ((s0[1:] > s0[:-1]) | ((s0[1:] == s0[:-1]) & (s1[1:] >= s1[:-1]))).all()
+    assert bool(ok)
+    # fast path (random 64-bit keys, no k0 collisions) under the debug
+    # verifier
+    import os
+
+    os.environ["PW_DEBUG_SORT"] = "1"
+    try:
+        r0 = torch.randint(-(2**62), 2**62, (100000,), dtype=torch.int64,
+                           generator=g)
+        r1 = torch.randint(-(2**62), 2**62, (100000,), dtype=torch.int64,
+                           generator=g)
+        perm = lex_sort_words([r0, r1])
+        assert perm.shape[0] == 100000
+    finally:
+        del os.environ["PW_DEBUG_SORT"]
