@@ -329,6 +329,22 @@ class Runtime:
         for i, src in enumerate(self.sources):
             if not getattr(src, "persistent_id", None):
                 src.persistent_id = getattr(src.source, "name", None) or f"src{i}"
+        if persistence is not None:
+            # connector offsets (e.g. Kafka (topic, partition) -> next
+            # offset) ride the AdvanceTime events so recovery can seek
+            # readers (reference OffsetAntichain, connectors/mod.rs:319)
+            def _collect():
+                out = {}
+                for s in self.sources:
+                    reader = getattr(s.source, "reader", None)
+                    offs = getattr(reader, "offsets", None)
+                    if offs:
+                        out[s.persistent_id] = [
+                            (str(k), str(v)) for k, v in offs.items()
+                        ]
+                return out
+
+            persistence._offsets_fn = _collect
 
     def _next_time(self):
         """Returns (time, waiting): time=None & waiting=False means done."""

@@ -193,3 +193,43 @@ def test_lz4_chunk_rotation(tmp_path):
     frames = list(SnapshotReader(store, "snapshots/0/src").frames())
     assert len(frames) == 10
     assert frames[3].startswith(b"payload-003")
+
+
+def test_kafka_offsets_ride_persistence(tmp_path):
+    """Connector offsets land in AdvanceTime events; source_offsets()
+    recovers them for seek (reference OffsetAntichain round trip)."""
+    import json as _json
+
+    import pathway_amd as pw
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.internals.schema import schema_from_types
+    from pathway_amd.persistence.engine import PersistenceManager
+    from tests.fakes.fake_kafka import FakeKafkaBroker
+
+    broker = FakeKafkaBroker(num_partitions=1).start()
+    try:
+        for i in range(3):
+            broker.seed("pt", 0, [(None, _json.dumps({"k": i}).encode())])
+        G.clear()
+        t = pw.io.kafka.read(
+            rdkafka_settings={"bootstrap.servers": broker.bootstrap},
+            topic="pt", schema=schema_from_types(k=int), format="json",
+            mode="static", name="ksrc",
+        )
+        out = str(tmp_path / "o.csv")
+        pw.io.csv.write(t, out)
+        backend = pw.persistence.Backend.filesystem(str(tmp_path / "p"))
+        cfg = pw.persistence.Config(backend=backend)
+        pw.run(persistence_config=cfg)
+
+        class Cfg:
+            pass
+
+        Cfg.backend = backend
+        Cfg.persistence_mode = None
+        Cfg.snapshot_interval_ms = 0
+        pm = PersistenceManager(Cfg(), worker=0)
+        offs = pm.source_offsets("ksrc")
+        assert offs.get("('pt', 0)") == "3"
+    finally:
+        broker.stop()
