@@ -18,10 +18,15 @@ from __future__ import annotations
 
 from typing import Sequence
 
+import os as _os
+
 import torch
 
 from pathway_amd.engine.batch import segmented_arange
 from pathway_amd.engine.column import Column, concat_columns
+
+#: kill switch for the hand-written radix sort (falls back to torch.sort)
+_PW_NO_PW_SORT = bool(_os.environ.get("PW_NO_PW_SORT"))
 
 
 def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
@@ -41,7 +46,12 @@ def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
         # sorting by a strict prefix of a row's identity would break
         # silently; tests/test_fuzz_equivalence.py pins the property and
         # PW_DEBUG_SORT=1 verifies full lex order on every call.
-        k0_sorted, perm0 = torch.sort(words[0])
+        if words[0].is_cuda and not _PW_NO_PW_SORT:
+            from pathway_amd import ops
+
+            k0_sorted, perm0 = ops.radix_sort64_gpu(words[0].contiguous())
+        else:
+            k0_sorted, perm0 = torch.sort(words[0])
         k1_sorted = words[1].index_select(0, perm0)
         # only a k0 collision between rows with DIFFERENT k1 violates lex
         # order (equal (k0,k1) rows may appear in any relative order —

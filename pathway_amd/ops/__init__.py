@@ -576,3 +576,75 @@ def merge_consolidate_gpu(
     if rc != 0:
         raise RuntimeError(f"pw_merge_consolidate_emit failed: {rc}")
     return [o0, o1], out_accs, rep
+
+
+def radix_sort64_gpu(keys: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """Hand-written LSD radix sort of int64 keys (signed order).
+
+    Returns (sorted_keys, perm) with perm stable: equal keys keep their
+    input order.  8x 8-bit digit passes; the (digit, block) scan runs as
+    one torch cumsum between kernel launches.
+    """
+    lib = require_lib()
+    n = keys.shape[0]
+    device = keys.device
+    perm = torch.arange(n, dtype=torch.int64, device=device)
+    if n <= 1:
+        return keys.clone(), perm
+    nblocks = max(1, min(1024, (n + (64 * 64) - 1) // (64 * 64)))
+    chunk = (n + nblocks * 64 - 1) // (nblocks * 64)
+    ka = torch.empty(n, dtype=torch.int64, device=device)  # uint64 bits
+    kb = torch.empty(n, dtype=torch.int64, device=device)
+    pa = perm
+    pb = torch.empty(n, dtype=torch.int64, device=device)
+    rc = lib.pw_radix_flip(
+        ctypes.c_void_p(keys.contiguous().data_ptr()),
+        ctypes.c_void_p(ka.data_ptr()),
+        ctypes.c_int64(n),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_radix_flip failed: {rc}")
+    counts = torch.empty(256 * nblocks, dtype=torch.int32, device=device)
+    for p in range(8):
+        shift = p * 8
+        rc = lib.pw_radix_count(
+            ctypes.c_void_p(ka.data_ptr()),
+            ctypes.c_int64(n),
+            ctypes.c_int(shift),
+            ctypes.c_int64(nblocks),
+            ctypes.c_int64(chunk),
+            ctypes.c_void_p(counts.data_ptr()),
+            _stream_ptr(),
+        )
+        if rc != 0:
+            raise RuntimeError(f"pw_radix_count failed: {rc}")
+        csum = torch.cumsum(counts.to(torch.int64), 0)
+        bases = torch.zeros_like(csum)
+        bases[1:] = csum[:-1]
+        rc = lib.pw_radix_scatter(
+            ctypes.c_void_p(ka.data_ptr()),
+            ctypes.c_void_p(pa.data_ptr()),
+            ctypes.c_int64(n),
+            ctypes.c_int(shift),
+            ctypes.c_int64(nblocks),
+            ctypes.c_int64(chunk),
+            ctypes.c_void_p(bases.data_ptr()),
+            ctypes.c_void_p(kb.data_ptr()),
+            ctypes.c_void_p(pb.data_ptr()),
+            _stream_ptr(),
+        )
+        if rc != 0:
+            raise RuntimeError(f"pw_radix_scatter failed: {rc}")
+        ka, kb = kb, ka
+        pa, pb = pb, pa
+    out = torch.empty(n, dtype=torch.int64, device=device)
+    rc = lib.pw_radix_unflip(
+        ctypes.c_void_p(ka.data_ptr()),
+        ctypes.c_void_p(out.data_ptr()),
+        ctypes.c_int64(n),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_radix_unflip failed: {rc}")
+    return out, pa

@@ -1249,3 +1249,138 @@ extern "C" int pw_merge_consolidate_emit(
                      (int64_t*)o1, O, (int64_t*)rep);
   return (int)hipGetLastError();
 }
+
+// -------------------------------------------------------------- radix sort --
+// Hand-written LSD radix sort of int64 keys with an int64 payload (the
+// permutation) — the mandated sort kernel for the 128-bit key path
+// (lex_sort_words' primary-word sort; the k1 tiebreak remains the
+// existing duplicate-check).  Signed order = unsigned order after the
+// sign bit is flipped once on input (flipped back on output).
+//
+// Per 8-bit digit pass:
+//   k_rs_count: one wave per block, each lane owns a contiguous chunk
+//     and counts serially (stable) into its LDS row; rows reduce to
+//     per-(block,digit) counts, laid out digit-major for the scan.
+//   <scan over (digit, block) runs in torch: one cumsum>
+//   k_rs_scatter: lanes re-walk their chunks; position = scanned base
+//     for (digit, block) + prefix of earlier lanes' counts (serial
+//     per-digit LDS prefix) + running in-lane count.
+
+#define PW_RS_THREADS 64
+#define PW_RS_DIGITS 256
+
+__global__ __launch_bounds__(PW_RS_THREADS) void k_rs_flip(
+    const int64_t* in, uint64_t* out, int64_t n, int dir) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (uint64_t)in[i] ^ 0x8000000000000000ull;
+}
+
+__global__ __launch_bounds__(PW_RS_THREADS) void k_rs_count(
+    const uint64_t* keys, int64_t n, int shift, int64_t chunk,
+    int* block_digit_counts /* [DIGITS][nblocks] digit-major */,
+    int64_t nblocks) {
+  __shared__ int cnt[PW_RS_THREADS][PW_RS_DIGITS];
+  int lane = threadIdx.x;
+  for (int d = 0; d < PW_RS_DIGITS; ++d) cnt[lane][d] = 0;
+  int64_t start = ((int64_t)blockIdx.x * PW_RS_THREADS + lane) * chunk;
+  int64_t end = min(start + chunk, n);
+  for (int64_t i = start; i < end; ++i) {
+    int d = (int)((keys[i] >> shift) & 0xFF);
+    ++cnt[lane][d];
+  }
+  __syncthreads();
+  // reduce lanes per digit: lane handles digits lane, lane+64, ...
+  for (int d = lane; d < PW_RS_DIGITS; d += PW_RS_THREADS) {
+    int s = 0;
+    for (int t = 0; t < PW_RS_THREADS; ++t) s += cnt[t][d];
+    block_digit_counts[(int64_t)d * nblocks + blockIdx.x] = s;
+  }
+}
+
+__global__ __launch_bounds__(PW_RS_THREADS) void k_rs_scatter(
+    const uint64_t* keys, const int64_t* payload, int64_t n, int shift,
+    int64_t chunk, const long long* bases /* [DIGITS][nblocks] excl */,
+    int64_t nblocks, uint64_t* out_keys, int64_t* out_payload) {
+  // counts accumulate into lane_base, then convert to exclusive prefix
+  // in place (one LDS array keeps us under the 160 KB/CU limit)
+  __shared__ long long lane_base[PW_RS_THREADS][PW_RS_DIGITS];
+  int lane = threadIdx.x;
+  for (int d = 0; d < PW_RS_DIGITS; ++d) lane_base[lane][d] = 0;
+  int64_t start = ((int64_t)blockIdx.x * PW_RS_THREADS + lane) * chunk;
+  int64_t end = min(start + chunk, n);
+  for (int64_t i = start; i < end; ++i) {
+    int d = (int)((keys[i] >> shift) & 0xFF);
+    ++lane_base[lane][d];
+  }
+  __syncthreads();
+  // serial exclusive prefix over lanes per digit (lane d' handles
+  // digits d', d'+64, ...)
+  for (int d = lane; d < PW_RS_DIGITS; d += PW_RS_THREADS) {
+    long long run = bases[(int64_t)d * nblocks + blockIdx.x];
+    for (int t = 0; t < PW_RS_THREADS; ++t) {
+      long long c = lane_base[t][d];
+      lane_base[t][d] = run;
+      run += c;
+    }
+  }
+  __syncthreads();
+  for (int64_t i = start; i < end; ++i) {
+    int d = (int)((keys[i] >> shift) & 0xFF);
+    long long p = lane_base[lane][d]++;
+    out_keys[p] = keys[i];
+    out_payload[p] = payload[i];
+  }
+}
+
+__global__ __launch_bounds__(PW_RS_THREADS) void k_rs_unflip(
+    const uint64_t* in, int64_t* out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (int64_t)(in[i] ^ 0x8000000000000000ull);
+}
+
+extern "C" int pw_radix_flip(const void* in, void* out, int64_t n,
+                             void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int64_t blocks = min((int64_t)2048, (n + PW_RS_THREADS - 1) / PW_RS_THREADS);
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(k_rs_flip, dim3((uint32_t)blocks), dim3(PW_RS_THREADS),
+                     0, s, (const int64_t*)in, (uint64_t*)out, n, 0);
+  return (int)hipGetLastError();
+}
+
+extern "C" int pw_radix_unflip(const void* in, void* out, int64_t n,
+                               void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int64_t blocks = min((int64_t)2048, (n + PW_RS_THREADS - 1) / PW_RS_THREADS);
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(k_rs_unflip, dim3((uint32_t)blocks),
+                     dim3(PW_RS_THREADS), 0, s, (const uint64_t*)in,
+                     (int64_t*)out, n);
+  return (int)hipGetLastError();
+}
+
+extern "C" int pw_radix_count(const void* keys, int64_t n, int shift,
+                              int64_t nblocks, int64_t chunk, void* counts,
+                              void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_rs_count, dim3((uint32_t)nblocks),
+                     dim3(PW_RS_THREADS), 0, s, (const uint64_t*)keys, n,
+                     shift, chunk, (int*)counts, nblocks);
+  return (int)hipGetLastError();
+}
+
+extern "C" int pw_radix_scatter(const void* keys, const void* payload,
+                                int64_t n, int shift, int64_t nblocks,
+                                int64_t chunk, const void* bases,
+                                void* out_keys, void* out_payload,
+                                void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_rs_scatter, dim3((uint32_t)nblocks),
+                     dim3(PW_RS_THREADS), 0, s, (const uint64_t*)keys,
+                     (const int64_t*)payload, n, shift, chunk,
+                     (const long long*)bases, nblocks, (uint64_t*)out_keys,
+                     (int64_t*)out_payload);
+  return (int)hipGetLastError();
+}

@@ -71,3 +71,34 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def bench_radix_sort():
+    """pw radix sort vs torch.sort (rocPRIM onesweep) on int64 keys."""
+    import time
+
+    import torch
+
+    from pathway_amd import ops
+
+    for n in (1 << 20, 1 << 22, 1 << 24):
+        keys = torch.randint(-(2**62), 2**62, (n,), dtype=torch.int64,
+                             device="cuda")
+        for label, fn in [
+            ("pw_radix_sort64", lambda: ops.radix_sort64_gpu(keys)),
+            ("torch.sort", lambda: torch.sort(keys)),
+        ]:
+            for _ in range(3):
+                fn()
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(10):
+                fn()
+            torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / 10
+            print(f"  {label:18s} n={n:>9}: {dt*1000:7.3f} ms "
+                  f"({n/dt/1e9:.2f} Gkeys/s)")
+
+
+if __name__ == "__main__" and __import__("os").environ.get("PW_BENCH_SORT"):
+    bench_radix_sort()

@@ -653,3 +653,26 @@ def test_groupreduce_fused_merge_matches_cpu():
     cpu_out = run("cpu")
     os.environ["PW_DEVICE"] = "cuda:0"
     assert gpu_out == cpu_out
+
+
+@gpu
+@requires_cuda
+def test_radix_sort_kernel_vs_torch():
+    from pathway_amd import ops
+
+    torch.manual_seed(41)
+    for n in (1, 2, 1000, 1 << 20, 12345):
+        keys = torch.randint(-(2**62), 2**62, (n,), dtype=torch.int64,
+                             device="cuda")
+        skeys, perm = ops.radix_sort64_gpu(keys)
+        ref_vals, ref_perm = torch.sort(keys, stable=True)
+        assert torch.equal(skeys, ref_vals), n
+        assert torch.equal(keys.index_select(0, perm), skeys), n
+        assert torch.equal(perm, ref_perm), n  # stability
+
+    # heavy duplicates exercise stability hard
+    keys = torch.randint(0, 7, (100000,), dtype=torch.int64, device="cuda")
+    skeys, perm = ops.radix_sort64_gpu(keys)
+    rv, rp = torch.sort(keys, stable=True)
+    assert torch.equal(skeys, rv)
+    assert torch.equal(perm, rp)
