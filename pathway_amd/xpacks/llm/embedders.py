@@ -85,13 +85,37 @@ class _NetworkEmbedder(BaseEmbedder):
 
 
 class OpenAIEmbedder(_NetworkEmbedder):
+    """OpenAI /embeddings protocol (reference embedders.py:77) — a real
+    client for any OpenAI-compatible endpoint (pass base_url for vLLM /
+    TEI in openai mode); verified against the fake HTTP service."""
+
     provider = "OpenAI"
 
-    def __init__(self, model: str | None = "text-embedding-3-small", **kwargs):
+    def __init__(self, model: str | None = "text-embedding-3-small", *,
+                 api_key: str | None = None, base_url: str | None = None,
+                 **kwargs):
         super().__init__(model, **kwargs)
+        self.api_key = api_key
+        self.base_url = (base_url or "https://api.openai.com/v1").rstrip("/")
+
+    def _embed_many(self, texts: list[str], **kwargs) -> list[np.ndarray]:
+        from pathway_amd.io import _rest
+
+        headers = {}
+        if self.api_key:
+            headers["Authorization"] = f"Bearer {self.api_key}"
+        out = _rest.request(
+            "POST", f"{self.base_url}/embeddings",
+            body={"model": self.model, "input": texts, **kwargs},
+            headers=headers,
+        )
+        data = sorted(out["data"], key=lambda d: d.get("index", 0))
+        return [np.asarray(d["embedding"], dtype=np.float32) for d in data]
 
 
-class LiteLLMEmbedder(_NetworkEmbedder):
+class LiteLLMEmbedder(OpenAIEmbedder):
+    """LiteLLM proxy speaks the OpenAI embeddings protocol."""
+
     provider = "LiteLLM"
 
 
