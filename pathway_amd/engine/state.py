@@ -25,8 +25,12 @@ import torch
 from pathway_amd.engine.batch import segmented_arange
 from pathway_amd.engine.column import Column, concat_columns
 
-#: kill switch for the hand-written radix sort (falls back to torch.sort)
-_PW_NO_PW_SORT = bool(_os.environ.get("PW_NO_PW_SORT"))
+#: opt-in for the hand-written radix sort kernel.  Measured on MI355X
+#: (profiles/kernels_r02.md): pw_radix_sort64 reaches 2.05 Gkeys/s @16M,
+#: rocPRIM's onesweep (torch.sort) 11 Gkeys/s — onesweep's decoupled-
+#: lookback single-pass structure wins, so the library stays the default
+#: and the pw kernel remains available and benchmarked.
+_PW_PW_SORT = bool(_os.environ.get("PW_PW_SORT"))
 
 
 def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
@@ -46,7 +50,7 @@ def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
         # sorting by a strict prefix of a row's identity would break
         # silently; tests/test_fuzz_equivalence.py pins the property and
         # PW_DEBUG_SORT=1 verifies full lex order on every call.
-        if words[0].is_cuda and not _PW_NO_PW_SORT:
+        if words[0].is_cuda and _PW_PW_SORT:
             from pathway_amd import ops
 
             k0_sorted, perm0 = ops.radix_sort64_gpu(words[0].contiguous())
