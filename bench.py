@@ -447,7 +447,9 @@ def main() -> None:
         print(
             json.dumps(
                 {
-                    "metric": "wordcount streaming throughput (events/sec, whole node)",
+                    # BASELINE.json metric; `value` is the events/s half,
+                    # config.rag_p50_ms carries the latency half
+                    "metric": "events/sec (whole node) on WordCount stream + p50 RAG query latency at fixed QPS",
                     "value": value,
                     "unit": "events/s",
                     "n_gpus": n_gpus,
