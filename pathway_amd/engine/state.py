@@ -300,7 +300,33 @@ class Arrangement:
         dwords = [w.index_select(0, dperm) for w in dwords]
         dweights = weights.index_select(0, dperm)
         dcols = {n: columns[n].take(dperm) for n in self.columns}
-        # 2. O(m+n) sorted merge with the state (no state re-sort)
+        # the delta may carry duplicate rows: consolidate it first so the
+        # fused merge's unique-rows precondition holds
+        dwords, dweights, dcols = consolidate_sorted(dwords, dweights, dcols)
+        if dwords[0].shape[0] == 0:
+            return
+        if dwords[0].is_cuda:
+            # 2a. fused merge-path merge+consolidate (pw HIP kernel):
+            # one count+emit pair over 4-word rows; zero-weight rows drop
+            # in-flight; carried columns gather once via rep
+            from pathway_amd import ops
+
+            out_words, out_accs, rep = ops.merge_consolidate_gpu(
+                self.words,
+                [self.weights],
+                dwords,
+                [dweights],
+                compare_words=2 if key_determined_vhash else 4,
+            )
+            self.key_words = out_words[:2]
+            self.vhash_words = out_words[2:]
+            self.weights = out_accs[0]
+            self.columns = {
+                n: concat_columns([self.columns[n], dcols[n]]).take(rep)
+                for n in self.columns
+            }
+            return
+        # 2b. host path: O(m+n) sorted merge + consolidation scan
         mwords = self.words[:2] if key_determined_vhash else self.words
         qwords = dwords[:2] if key_determined_vhash else dwords
         src_sel = merge_sorted_select(mwords, qwords)

@@ -506,35 +506,39 @@ def merge_consolidate_gpu(
     a_accs: Sequence[torch.Tensor],
     b_words: Sequence[torch.Tensor],
     b_accs: Sequence[torch.Tensor],
+    compare_words: int | None = None,
 ) -> tuple[list[torch.Tensor], list[torch.Tensor], torch.Tensor]:
-    """Fused LSM merge+consolidate of two unique lex-sorted key sets.
+    """Fused LSM merge+consolidate of two unique lex-sorted row sets.
 
-    acc slot 0 is the weight; merged rows with zero weight are dropped.
-    Returns ([k0, k1], accs, rep) with rep indexing concat([A, B]) rows
-    (A preferred on matches) for carried-column gathers.
+    Rows are nw (<=4) int64 words; comparison uses the first
+    `compare_words` (default all — pass 2 when the tail words are
+    key-determined).  acc slot 0 is the weight; merged rows with zero
+    weight are dropped.  Returns (out_words, accs, rep) with rep
+    indexing concat([A, B]) rows (A preferred on matches) for
+    carried-column gathers.
     """
     lib = require_lib()
     m = a_words[0].shape[0]
     n = b_words[0].shape[0]
+    nw = len(a_words)
+    nwc = compare_words or nw
     nacc = len(a_accs)
-    assert len(b_accs) == nacc and 1 <= nacc <= 8
+    assert len(b_accs) == nacc and 1 <= nacc <= 8 and 1 <= nwc <= nw <= 4
     device = a_words[0].device
     total_diag = m + n
     nthreads = max(1, (total_diag + 7) // 8)
     counts = torch.empty(nthreads, dtype=torch.int32, device=device)
-    a0 = a_words[0].contiguous()
-    a1 = a_words[1].contiguous()
-    b0 = b_words[0].contiguous()
-    b1 = b_words[1].contiguous()
+    wA = [t.contiguous() for t in a_words]
+    wB = [t.contiguous() for t in b_words]
     aA = [t.contiguous() for t in a_accs]
     aB = [t.contiguous() for t in b_accs]
     rc = lib.pw_merge_consolidate_count(
-        ctypes.c_void_p(a0.data_ptr()),
-        ctypes.c_void_p(a1.data_ptr()),
+        _ptr_arr(wA),
         _ptr_arr(aA),
-        ctypes.c_void_p(b0.data_ptr()),
-        ctypes.c_void_p(b1.data_ptr()),
+        _ptr_arr(wB),
         _ptr_arr(aB),
+        ctypes.c_int(nw),
+        ctypes.c_int(nwc),
         ctypes.c_int(nacc),
         ctypes.c_int64(m),
         ctypes.c_int64(n),
@@ -548,34 +552,35 @@ def merge_consolidate_gpu(
     total = int(csum[-1].item())
     bases = torch.zeros(nthreads, dtype=torch.int64, device=device)
     bases[1:] = csum[:-1]
-    o0 = torch.empty(total, dtype=torch.int64, device=device)
-    o1 = torch.empty(total, dtype=torch.int64, device=device)
+    out_words = [
+        torch.empty(total, dtype=torch.int64, device=device)
+        for _ in range(nw)
+    ]
     out_accs = [
         torch.empty(total, dtype=torch.int64, device=device)
         for _ in range(nacc)
     ]
     rep = torch.empty(total, dtype=torch.int64, device=device)
     rc = lib.pw_merge_consolidate_emit(
-        ctypes.c_void_p(a0.data_ptr()),
-        ctypes.c_void_p(a1.data_ptr()),
+        _ptr_arr(wA),
         _ptr_arr(aA),
-        ctypes.c_void_p(b0.data_ptr()),
-        ctypes.c_void_p(b1.data_ptr()),
+        _ptr_arr(wB),
         _ptr_arr(aB),
+        ctypes.c_int(nw),
+        ctypes.c_int(nwc),
         ctypes.c_int(nacc),
         ctypes.c_int64(m),
         ctypes.c_int64(n),
         ctypes.c_void_p(bases.data_ptr()),
         ctypes.c_int64(nthreads),
-        ctypes.c_void_p(o0.data_ptr()),
-        ctypes.c_void_p(o1.data_ptr()),
+        _ptr_arr(out_words),
         _ptr_arr(out_accs),
         ctypes.c_void_p(rep.data_ptr()),
         _stream_ptr(),
     )
     if rc != 0:
         raise RuntimeError(f"pw_merge_consolidate_emit failed: {rc}")
-    return [o0, o1], out_accs, rep
+    return out_words, out_accs, rep
 
 
 def radix_sort64_gpu(keys: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:

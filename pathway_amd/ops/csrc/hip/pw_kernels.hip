@@ -1085,28 +1085,36 @@ struct McAcc {
 struct McAccMut {
   long long* p[8];
 };
+struct McWords {
+  const int64_t* p[4];
+};
+struct McWordsMut {
+  int64_t* p[4];
+};
 
-__device__ __forceinline__ bool mc_le(const int64_t* a0, const int64_t* a1,
-                                      int64_t i, const int64_t* b0,
-                                      const int64_t* b1, int64_t j) {
-  if (a0[i] != b0[j]) return a0[i] < b0[j];
-  return a1[i] <= b1[j];
+// compare the first nwc words (lexicographic, signed)
+__device__ __forceinline__ bool mc_le(McWords A, int64_t i, McWords B,
+                                      int64_t j, int nwc) {
+  for (int w = 0; w < nwc; ++w) {
+    if (A.p[w][i] != B.p[w][j]) return A.p[w][i] < B.p[w][j];
+  }
+  return true;
 }
 
-__device__ __forceinline__ bool mc_eq(const int64_t* a0, const int64_t* a1,
-                                      int64_t i, const int64_t* b0,
-                                      const int64_t* b1, int64_t j) {
-  return a0[i] == b0[j] && a1[i] == b1[j];
+__device__ __forceinline__ bool mc_eq(McWords A, int64_t i, McWords B,
+                                      int64_t j, int nwc) {
+  for (int w = 0; w < nwc; ++w)
+    if (A.p[w][i] != B.p[w][j]) return false;
+  return true;
 }
 
-__device__ int64_t mc_merge_path(const int64_t* a0, const int64_t* a1,
-                                 int64_t m, const int64_t* b0,
-                                 const int64_t* b1, int64_t n, int64_t d) {
+__device__ int64_t mc_merge_path(McWords A, int64_t m, McWords B, int64_t n,
+                                 int64_t d, int nwc) {
   int64_t lo = d > n ? d - n : 0;
   int64_t hi = d < m ? d : m;
   while (lo < hi) {
     int64_t mid = (lo + hi) >> 1;
-    if (mc_le(a0, a1, mid, b0, b1, d - mid - 1))
+    if (mc_le(A, mid, B, d - mid - 1, nwc))
       lo = mid + 1;
     else
       hi = mid;
@@ -1116,21 +1124,22 @@ __device__ int64_t mc_merge_path(const int64_t* a0, const int64_t* a1,
 
 #define PW_MC_CHUNK 8
 
-// one walk function used by both phases: WRITE=false counts only
+// one walk function used by both phases: WRITE=false counts only.
+// nw = words carried to the output; nwc (<= nw) = comparison prefix
+// (callers guarantee the tail words are key-determined when nwc < nw).
 template <bool WRITE>
-__device__ void mc_walk(const int64_t* a0, const int64_t* a1, McAcc accA,
-                        const int64_t* b0, const int64_t* b1, McAcc accB,
-                        int nacc, int64_t m, int64_t n, int64_t d0,
-                        int64_t d1, int* count_out, int64_t base,
-                        int64_t* o0, int64_t* o1, McAccMut accO,
+__device__ void mc_walk(McWords A, McAcc accA, McWords B, McAcc accB,
+                        int nw, int nwc, int nacc, int64_t m, int64_t n,
+                        int64_t d0, int64_t d1, int* count_out,
+                        int64_t base, McWordsMut O, McAccMut accO,
                         int64_t* rep) {
-  int64_t ai = mc_merge_path(a0, a1, m, b0, b1, n, d0);
+  int64_t ai = mc_merge_path(A, m, B, n, d0, nwc);
   int64_t bi = d0 - ai;
   int64_t d = d0;
   // skip rule: a range starting on the B half of a match
-  if (d < d1 && bi < n && ai > 0 && !(bi >= n) &&
-      !(ai < m && mc_le(a0, a1, ai, b0, b1, bi))) {
-    if (mc_eq(a0, a1, ai - 1, b0, b1, bi)) {
+  if (d < d1 && bi < n && ai > 0 &&
+      !(ai < m && mc_le(A, ai, B, bi, nwc))) {
+    if (mc_eq(A, ai - 1, B, bi, nwc)) {
       ++bi;
       ++d;
     }
@@ -1138,14 +1147,13 @@ __device__ void mc_walk(const int64_t* a0, const int64_t* a1, McAcc accA,
   int cnt = 0;
   int64_t w = base;
   while (d < d1) {
-    bool take_a = (bi >= n) || (ai < m && mc_le(a0, a1, ai, b0, b1, bi));
+    bool take_a = (bi >= n) || (ai < m && mc_le(A, ai, B, bi, nwc));
     if (take_a) {
-      bool matched = (bi < n) && mc_eq(a0, a1, ai, b0, b1, bi);
+      bool matched = (bi < n) && mc_eq(A, ai, B, bi, nwc);
       long long wsum = accA.p[0][ai] + (matched ? accB.p[0][bi] : 0ll);
       if (wsum != 0) {
         if (WRITE) {
-          o0[w] = a0[ai];
-          o1[w] = a1[ai];
+          for (int k = 0; k < nw; ++k) O.p[k][w] = A.p[k][ai];
           for (int c = 0; c < nacc; ++c)
             accO.p[c][w] =
                 accA.p[c][ai] + (matched ? accB.p[c][bi] : 0ll);
@@ -1164,8 +1172,7 @@ __device__ void mc_walk(const int64_t* a0, const int64_t* a1, McAcc accA,
       long long wsum = accB.p[0][bi];
       if (wsum != 0) {
         if (WRITE) {
-          o0[w] = b0[bi];
-          o1[w] = b1[bi];
+          for (int k = 0; k < nw; ++k) O.p[k][w] = B.p[k][bi];
           for (int c = 0; c < nacc; ++c) accO.p[c][w] = accB.p[c][bi];
           rep[w] = m + bi;
           ++w;
@@ -1179,9 +1186,8 @@ __device__ void mc_walk(const int64_t* a0, const int64_t* a1, McAcc accA,
   if (!WRITE) *count_out = cnt;
 }
 
-__global__ void k_mc_count(const int64_t* a0, const int64_t* a1, McAcc accA,
-                           const int64_t* b0, const int64_t* b1, McAcc accB,
-                           int nacc, int64_t m, int64_t n,
+__global__ void k_mc_count(McWords A, McAcc accA, McWords B, McAcc accB,
+                           int nw, int nwc, int nacc, int64_t m, int64_t n,
                            int* thread_counts, int64_t nthreads) {
   int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= nthreads) return;
@@ -1189,64 +1195,72 @@ __global__ void k_mc_count(const int64_t* a0, const int64_t* a1, McAcc accA,
   int64_t d0 = t * PW_MC_CHUNK;
   int64_t d1 = min(d0 + PW_MC_CHUNK, total);
   McAccMut dummy{};
-  mc_walk<false>(a0, a1, accA, b0, b1, accB, nacc, m, n, d0, d1,
-                 &thread_counts[t], 0, nullptr, nullptr, dummy, nullptr);
+  McWordsMut wdummy{};
+  mc_walk<false>(A, accA, B, accB, nw, nwc, nacc, m, n, d0, d1,
+                 &thread_counts[t], 0, wdummy, dummy, nullptr);
 }
 
-__global__ void k_mc_emit(const int64_t* a0, const int64_t* a1, McAcc accA,
-                          const int64_t* b0, const int64_t* b1, McAcc accB,
-                          int nacc, int64_t m, int64_t n,
+__global__ void k_mc_emit(McWords A, McAcc accA, McWords B, McAcc accB,
+                          int nw, int nwc, int nacc, int64_t m, int64_t n,
                           const int64_t* bases, int64_t nthreads,
-                          int64_t* o0, int64_t* o1, McAccMut accO,
-                          int64_t* rep) {
+                          McWordsMut O, McAccMut accO, int64_t* rep) {
   int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= nthreads) return;
   int64_t total = m + n;
   int64_t d0 = t * PW_MC_CHUNK;
   int64_t d1 = min(d0 + PW_MC_CHUNK, total);
-  mc_walk<true>(a0, a1, accA, b0, b1, accB, nacc, m, n, d0, d1, nullptr,
-                bases[t], o0, o1, accO, rep);
+  mc_walk<true>(A, accA, B, accB, nw, nwc, nacc, m, n, d0, d1, nullptr,
+                bases[t], O, accO, rep);
 }
 
 extern "C" int pw_merge_consolidate_count(
-    const void* a0, const void* a1, const void** accA, const void* b0,
-    const void* b1, const void** accB, int nacc, int64_t m, int64_t n,
+    const void** wordsA, const void** accA, const void** wordsB,
+    const void** accB, int nw, int nwc, int nacc, int64_t m, int64_t n,
     void* thread_counts, int64_t nthreads, void* stream) {
-  if (nacc > 8 || nacc < 1) return 2;
+  if (nacc > 8 || nacc < 1 || nw > 4 || nwc > nw) return 2;
   McAcc A{}, B{};
+  McWords WA{}, WB{};
   for (int c = 0; c < nacc; ++c) {
     A.p[c] = (const long long*)accA[c];
     B.p[c] = (const long long*)accB[c];
   }
+  for (int k = 0; k < nw; ++k) {
+    WA.p[k] = (const int64_t*)wordsA[k];
+    WB.p[k] = (const int64_t*)wordsB[k];
+  }
   hipStream_t s = (hipStream_t)stream;
   int64_t blocks = (nthreads + PW_BLOCK - 1) / PW_BLOCK;
   hipLaunchKernelGGL(k_mc_count, dim3((uint32_t)blocks), dim3(PW_BLOCK), 0,
-                     s, (const int64_t*)a0, (const int64_t*)a1, A,
-                     (const int64_t*)b0, (const int64_t*)b1, B, nacc, m, n,
+                     s, WA, A, WB, B, nw, nwc, nacc, m, n,
                      (int*)thread_counts, nthreads);
   return (int)hipGetLastError();
 }
 
 extern "C" int pw_merge_consolidate_emit(
-    const void* a0, const void* a1, const void** accA, const void* b0,
-    const void* b1, const void** accB, int nacc, int64_t m, int64_t n,
-    const void* bases, int64_t nthreads, void* o0, void* o1, void** accO,
+    const void** wordsA, const void** accA, const void** wordsB,
+    const void** accB, int nw, int nwc, int nacc, int64_t m, int64_t n,
+    const void* bases, int64_t nthreads, void** out_words, void** accO,
     void* rep, void* stream) {
-  if (nacc > 8 || nacc < 1) return 2;
+  if (nacc > 8 || nacc < 1 || nw > 4 || nwc > nw) return 2;
   McAcc A{}, B{};
   McAccMut O{};
+  McWords WA{}, WB{};
+  McWordsMut WO{};
   for (int c = 0; c < nacc; ++c) {
     A.p[c] = (const long long*)accA[c];
     B.p[c] = (const long long*)accB[c];
     O.p[c] = (long long*)accO[c];
   }
+  for (int k = 0; k < nw; ++k) {
+    WA.p[k] = (const int64_t*)wordsA[k];
+    WB.p[k] = (const int64_t*)wordsB[k];
+    WO.p[k] = (int64_t*)out_words[k];
+  }
   hipStream_t s = (hipStream_t)stream;
   int64_t blocks = (nthreads + PW_BLOCK - 1) / PW_BLOCK;
   hipLaunchKernelGGL(k_mc_emit, dim3((uint32_t)blocks), dim3(PW_BLOCK), 0,
-                     s, (const int64_t*)a0, (const int64_t*)a1, A,
-                     (const int64_t*)b0, (const int64_t*)b1, B, nacc, m, n,
-                     (const int64_t*)bases, nthreads, (int64_t*)o0,
-                     (int64_t*)o1, O, (int64_t*)rep);
+                     s, WA, A, WB, B, nw, nwc, nacc, m, n,
+                     (const int64_t*)bases, nthreads, WO, O, (int64_t*)rep);
   return (int)hipGetLastError();
 }
 

@@ -676,3 +676,48 @@ def test_radix_sort_kernel_vs_torch():
     rv, rp = torch.sort(keys, stable=True)
     assert torch.equal(skeys, rv)
     assert torch.equal(perm, rp)
+
+
+@gpu
+@requires_cuda
+def test_arrangement_fused_merge_matches_cpu():
+    """Arrangement.merge via the fused 4-word kernel equals the host
+    merge+consolidate on randomized weighted multisets."""
+    from pathway_amd.engine.column import TensorColumn
+    from pathway_amd.engine.state import Arrangement
+    from pathway_amd.internals import dtype as dt_
+
+    torch.manual_seed(19)
+
+    def build(device):
+        proto = {"v": TensorColumn(torch.zeros(0, dtype=torch.int64,
+                                               device=device), dt_.INT)}
+        arr = Arrangement(torch.device(device), proto)
+        g = torch.Generator().manual_seed(99)
+        for batch in range(6):
+            n = 3000
+            keys = torch.randint(0, 500, (n, 2), dtype=torch.int64,
+                                 generator=g)
+            v0 = torch.randint(0, 50, (n,), dtype=torch.int64, generator=g)
+            v1 = v0 * 7 + 1
+            w = torch.randint(-2, 3, (n,), dtype=torch.int64, generator=g)
+            nz = w != 0
+            keys, v0, v1, w = keys[nz], v0[nz], v1[nz], w[nz]
+            cols = {"v": TensorColumn(v0.to(device), dt_.INT)}
+            arr.merge(keys.to(device), (v0.to(device), v1.to(device)),
+                      w.to(device), cols)
+        return arr
+
+    cpu = build("cpu")
+    gpu_ = build("cuda:0")
+    rows_cpu = sorted(zip(
+        cpu.key_words[0].tolist(), cpu.key_words[1].tolist(),
+        cpu.vhash_words[0].tolist(), cpu.weights.tolist(),
+        cpu.columns["v"].tensor.tolist(),
+    ))
+    rows_gpu = sorted(zip(
+        gpu_.key_words[0].cpu().tolist(), gpu_.key_words[1].cpu().tolist(),
+        gpu_.vhash_words[0].cpu().tolist(), gpu_.weights.cpu().tolist(),
+        gpu_.columns["v"].tensor.cpu().tolist(),
+    ))
+    assert rows_cpu == rows_gpu
