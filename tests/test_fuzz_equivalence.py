@@ -644,3 +644,120 @@ def test_lex_sort_words_fast_path_invariant():
         assert perm.shape[0] == 100000
     finally:
         del os.environ["PW_DEBUG_SORT"]
+
+
+def test_fuzz_persistence_codec_roundtrip():
+    """Random value streams survive encode->LZ4->decode bit-exactly."""
+    import random
+
+    import numpy as np
+
+    from pathway_amd.internals.api import Pointer
+    from pathway_amd.internals.json import Json
+    from pathway_amd.ops import native_io
+    from pathway_amd.persistence import codec
+
+    rng = random.Random(123)
+
+    def rand_value(depth=0):
+        kind = rng.randrange(0, 10 if depth > 1 else 12)
+        if kind == 0:
+            return None
+        if kind == 1:
+            return rng.choice([True, False])
+        if kind == 2:
+            return rng.randint(-(2**62), 2**62)
+        if kind == 3:
+            return rng.random() * 1e6 - 5e5
+        if kind == 4:
+            return "".join(chr(rng.randint(32, 0x2FF)) for _ in range(rng.randrange(20)))
+        if kind == 5:
+            return bytes(rng.randrange(256) for _ in range(rng.randrange(30)))
+        if kind == 6:
+            return Pointer(rng.getrandbits(64), rng.getrandbits(64))
+        if kind == 7:
+            return np.array([rng.randint(-9, 9) for _ in range(rng.randrange(1, 6))],
+                            dtype=np.int64)
+        if kind == 8:
+            return np.array([rng.random() for _ in range(rng.randrange(1, 5))])
+        if kind == 9:
+            return Json({"k": rng.randint(0, 9), "l": [1, None, "s"]})
+        if kind == 10:
+            return tuple(rand_value(depth + 1) for _ in range(rng.randrange(4)))
+        return rng.randint(0, 5)
+
+    for trial in range(30):
+        n = rng.randrange(1, 8)
+        events = []
+        blob = bytearray()
+        for _ in range(n):
+            key = Pointer(rng.getrandbits(64), rng.getrandbits(64))
+            values = [rand_value() for _ in range(rng.randrange(1, 5))]
+            kind = codec.E_INSERT if rng.random() < 0.7 else codec.E_DELETE
+            events.append((kind, key, values))
+            blob += codec.encode_event(kind, key=key, values=values)
+        comp = native_io.lz4_compress(bytes(blob))
+        raw = native_io.lz4_decompress(comp, len(blob))
+        assert raw == bytes(blob)
+        def same(a, b):
+            if isinstance(a, np.ndarray):
+                return isinstance(b, np.ndarray) and np.array_equal(a, b)
+            if isinstance(a, Json):
+                return a.value == b.value
+            if isinstance(a, float):
+                return a == b or (a != a and b != b)
+            if isinstance(a, tuple):
+                return (isinstance(b, tuple) and len(a) == len(b)
+                        and all(same(x, y) for x, y in zip(a, b)))
+            return a == b
+
+        i = 0
+        for kind, key, values in events:
+            k2, (key2, values2), i = codec.decode_event(raw, i)
+            assert k2 == kind and key2 == key
+            for a, b in zip(values, values2):
+                assert same(a, b), (a, b)
+        assert i == len(raw)
+
+
+def test_fuzz_ivf_with_deletions_vs_brute():
+    """IVF search with interleaved adds/deletes matches brute force over
+    the surviving rows (recall == 1 with nprobe == nlist)."""
+    import torch
+
+    from pathway_amd.engine.ann import FlatIndexState, IvfFlatState
+
+    g = torch.Generator().manual_seed(5)
+    d, k = 16, 5
+    ivf = IvfFlatState("cpu", "cos", min_train=200, nlist=8, nprobe=8,
+                      rebuild_every=300)
+    flat = FlatIndexState("cpu", "cos")
+    next_key = [1]
+
+    def batch(n_add, n_del):
+        keys_add = torch.stack([
+            torch.arange(next_key[0], next_key[0] + n_add, dtype=torch.int64),
+            torch.zeros(n_add, dtype=torch.int64),
+        ], dim=1)
+        next_key[0] += n_add
+        vecs = torch.randn(n_add, d, generator=g)
+        for st in (ivf, flat):
+            st.update(keys_add, vecs, torch.ones(n_add, dtype=torch.int64))
+        # delete a random previously-added prefix slice
+        if n_del and next_key[0] > n_del + 1:
+            start = int(torch.randint(1, next_key[0] - n_del, (1,), generator=g))
+            keys_del = torch.stack([
+                torch.arange(start, start + n_del, dtype=torch.int64),
+                torch.zeros(n_del, dtype=torch.int64),
+            ], dim=1)
+            dummy = torch.zeros(n_del, d)
+            for st in (ivf, flat):
+                st.update(keys_del, dummy, -torch.ones(n_del, dtype=torch.int64))
+
+    for step in (300, 150, 200, 100):
+        batch(step, step // 4)
+    q = torch.randn(20, d, generator=g)
+    ids_i, sc_i, _ = ivf.search(q, k)
+    ids_f, sc_f, _ = flat.search(q, k)
+    assert torch.allclose(sc_i, sc_f, atol=1e-5)
+    assert torch.equal(ids_i[:, :, 0], ids_f[:, :, 0])
