@@ -18,6 +18,13 @@ def read(
     name: str | None = None,
     **kwargs: Any,
 ):
+    if mode in ("streaming", "streaming_with_deletions"):
+        from pathway_amd.io import fs as io_fs
+
+        return io_fs.read(
+            path, format="csv", schema=schema, mode=mode, name=name,
+            **kwargs,
+        )
     from pathway_amd.debug import table_from_rows
     from pathway_amd.internals.schema import schema_from_csv
 

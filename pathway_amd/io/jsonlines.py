@@ -18,6 +18,15 @@ def read(
     name: str | None = None,
     **kwargs: Any,
 ):
+    if mode in ("streaming", "streaming_with_deletions"):
+        # live mode: the fs poller tails the directory (new/modified
+        # files append, deletions retract)
+        from pathway_amd.io import fs as io_fs
+
+        return io_fs.read(
+            path, format="json", schema=schema, mode=mode, name=name,
+            **kwargs,
+        )
     from pathway_amd.debug import table_from_rows
     from pathway_amd.internals.json import Json
 
