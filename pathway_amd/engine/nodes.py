@@ -121,7 +121,17 @@ def consolidate_batch(batch: DeltaBatch) -> DeltaBatch | None:
         return batch
     v0, v1 = batch_vhash(batch)
     words = [batch.keys[:, 0].contiguous(), batch.keys[:, 1].contiguous(), v0, v1]
-    perm = lex_sort_words(words)
+    if len(batch) > 2048 and batch.keys.is_cuda:
+        # consolidation only needs equal (key, vhash) rows adjacent — the
+        # order itself is irrelevant for outputs.  Sorting by a fused
+        # 128-bit row hash turns the 4-word multi-pass (4 stable
+        # argsorts) into the single-sort fast path; equal rows share the
+        # hash, so grouping is preserved (c0-collision guard inside
+        # lex_sort_words).
+        c0, c1 = hashing.hash128_words([w.contiguous() for w in words])
+        perm = lex_sort_words([c0, c1])
+    else:
+        perm = lex_sort_words(words)
     words = [w.index_select(0, perm) for w in words]
     weights = batch.diffs.index_select(0, perm)
     cols = {n: c.take(perm) for n, c in batch.columns.items()}
