@@ -415,3 +415,33 @@ def test_detailed_metrics_and_dashboard(tmp_path):
         assert "operators" in page and "svg" in page
     finally:
         pw.set_monitoring_config(detailed_metrics_dir=None)
+
+
+def test_telemetry_otlp_http_export():
+    """Telemetry exports OTLP/HTTP JSON (resourceSpans/resourceMetrics)
+    to the configured endpoint (reference telemetry.rs OTLP export)."""
+    from pathway_amd.internals.telemetry import Telemetry
+    from tests.fakes.fake_http import FakeHTTPService
+
+    srv = FakeHTTPService().start()
+    try:
+        t = Telemetry(endpoint=srv.url, service_name="svc1")
+        with t.span("graph_runner.run", worker=0):
+            pass
+        t.gauge("pathway.steps", 5.0, worker=0)
+        t.flush_otlp()
+        t.close()
+        traces = [r for r in srv.requests if r.path == "/v1/traces"]
+        metrics = [r for r in srv.requests if r.path == "/v1/metrics"]
+        assert traces and metrics
+        sp = traces[0].json()["resourceSpans"][0]
+        res_attrs = {a["key"]: a["value"] for a in sp["resource"]["attributes"]}
+        assert res_attrs["service.name"]["stringValue"] == "svc1"
+        span = sp["scopeSpans"][0]["spans"][0]
+        assert span["name"] == "graph_runner.run"
+        assert int(span["endTimeUnixNano"]) >= int(span["startTimeUnixNano"])
+        m = metrics[0].json()["resourceMetrics"][0]["scopeMetrics"][0]["metrics"][0]
+        assert m["name"] == "pathway.steps"
+        assert m["gauge"]["dataPoints"][0]["asDouble"] == 5.0
+    finally:
+        srv.stop()
