@@ -296,7 +296,8 @@ def main() -> None:
                 torch.cuda.current_stream().wait_event(self.ready[cur])
                 buf = self.staged[cur]
                 # PARSE on device: newline scan -> token [start, end) spans
-                nl = (buf == 10).nonzero(as_tuple=True)[0]
+                # (pw_scan_positions: ordered match positions, count+emit)
+                nl = ops.scan_positions_gpu(buf, 10)
                 starts = torch.cat(
                     [torch.zeros(1, dtype=torch.int64, device=dev), nl[:-1] + 1]
                 )

@@ -653,3 +653,40 @@ def radix_sort64_gpu(keys: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     if rc != 0:
         raise RuntimeError(f"pw_radix_unflip failed: {rc}")
     return out, pa
+
+
+def scan_positions_gpu(buf: torch.Tensor, target: int) -> torch.Tensor:
+    """Ordered positions where buf == target (uint8 buffer) — the
+    newline/separator scan of the ingest parse (replaces nonzero)."""
+    lib = require_lib()
+    n = buf.shape[0]
+    device = buf.device
+    nblocks = max(1, min(2048, (n + 4095) // 4096))
+    counts = torch.empty(nblocks, dtype=torch.int64, device=device)
+    rc = lib.pw_scan_positions(
+        ctypes.c_void_p(buf.contiguous().data_ptr()),
+        ctypes.c_int64(n),
+        ctypes.c_int(int(target)),
+        ctypes.c_void_p(counts.data_ptr()),
+        ctypes.c_int64(nblocks),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_scan_positions failed: {rc}")
+    csum = torch.cumsum(counts, 0)
+    total = int(csum[-1].item())
+    bases = torch.zeros(nblocks, dtype=torch.int64, device=device)
+    bases[1:] = csum[:-1]
+    out = torch.empty(total, dtype=torch.int64, device=device)
+    rc = lib.pw_scan_emit(
+        ctypes.c_void_p(buf.contiguous().data_ptr()),
+        ctypes.c_int64(n),
+        ctypes.c_int(int(target)),
+        ctypes.c_void_p(bases.data_ptr()),
+        ctypes.c_int64(nblocks),
+        ctypes.c_void_p(out.data_ptr()),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_scan_emit failed: {rc}")
+    return out

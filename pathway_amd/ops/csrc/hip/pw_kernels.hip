@@ -1398,3 +1398,80 @@ extern "C" int pw_radix_scatter(const void* keys, const void* payload,
                      (int64_t*)out_payload);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------------------ mask scan --
+// Ordered positions of matching bytes / true mask entries — the newline
+// scan of the ingest parse and the run-start extraction, replacing
+// rocprim partition (torch nonzero) with a count + emit pair whose scan
+// is one torch cumsum.
+
+__global__ void k_scan_count(const uint8_t* buf, int64_t n, int target,
+                             int64_t chunk, long long* block_counts) {
+  __shared__ int cnt;
+  if (threadIdx.x == 0) cnt = 0;
+  __syncthreads();
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t end = min(start + chunk, n);
+  int local = 0;
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+    if (buf[i] == (uint8_t)target) ++local;
+  atomicAdd(&cnt, local);
+  __syncthreads();
+  if (threadIdx.x == 0) block_counts[blockIdx.x] = cnt;
+}
+
+__global__ void k_scan_emit(const uint8_t* buf, int64_t n, int target,
+                            int64_t chunk, const long long* bases,
+                            int64_t* out) {
+  // ordered within the block: lanes cooperate wave-sequentially via a
+  // per-block running cursor advanced in lane order per 256-wide tile
+  __shared__ long long cursor;
+  if (threadIdx.x == 0) cursor = bases[blockIdx.x];
+  __syncthreads();
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t end = min(start + chunk, n);
+  for (int64_t base = start; base < end; base += blockDim.x) {
+    int64_t i = base + threadIdx.x;
+    int hit = (i < end && buf[i] == (uint8_t)target) ? 1 : 0;
+    // intra-tile exclusive prefix of hits (ballot across the block via LDS)
+    __shared__ int tile[PW_BLOCK];
+    tile[threadIdx.x] = hit;
+    __syncthreads();
+    // simple scan (block is 256 wide: log-step scan)
+    for (int off = 1; off < PW_BLOCK; off <<= 1) {
+      int v = (threadIdx.x >= off) ? tile[threadIdx.x - off] : 0;
+      __syncthreads();
+      tile[threadIdx.x] += v;
+      __syncthreads();
+    }
+    int incl = tile[threadIdx.x];
+    if (hit) out[cursor + incl - 1] = i;
+    __syncthreads();
+    if (threadIdx.x == PW_BLOCK - 1) cursor += incl;
+    __syncthreads();
+  }
+}
+
+extern "C" int pw_scan_positions(const void* buf, int64_t n, int target,
+                                 void* block_counts, int64_t nblocks,
+                                 void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int64_t chunk = (n + nblocks - 1) / nblocks;
+  if (chunk < 1) chunk = 1;
+  hipLaunchKernelGGL(k_scan_count, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
+                     0, s, (const uint8_t*)buf, n, target,
+                     chunk, (long long*)block_counts);
+  return (int)hipGetLastError();
+}
+
+extern "C" int pw_scan_emit(const void* buf, int64_t n, int target,
+                            const void* bases, int64_t nblocks, void* out,
+                            void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int64_t chunk = (n + nblocks - 1) / nblocks;
+  if (chunk < 1) chunk = 1;
+  hipLaunchKernelGGL(k_scan_emit, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
+                     0, s, (const uint8_t*)buf, n, target, chunk,
+                     (const long long*)bases, (int64_t*)out);
+  return (int)hipGetLastError();
+}

@@ -721,3 +721,16 @@ def test_arrangement_fused_merge_matches_cpu():
         gpu_.columns["v"].tensor.cpu().tolist(),
     ))
     assert rows_cpu == rows_gpu
+
+
+@gpu
+@requires_cuda
+def test_scan_positions_kernel():
+    from pathway_amd import ops
+
+    torch.manual_seed(3)
+    for n in (0, 1, 1000, 1 << 20):
+        buf = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+        got = ops.scan_positions_gpu(buf, 10)
+        ref = (buf == 10).nonzero(as_tuple=True)[0]
+        assert torch.equal(got, ref), n
