@@ -132,8 +132,9 @@ def consolidate_batch(batch: DeltaBatch) -> DeltaBatch | None:
         perm = lex_sort_words([c0, c1])
     else:
         perm = lex_sort_words(words)
-    words = [w.index_select(0, perm) for w in words]
-    weights = batch.diffs.index_select(0, perm)
+    from pathway_amd import ops
+
+    *words, weights = ops.gather_all(perm, words + [batch.diffs])
     cols = {n: c.take(perm) for n, c in batch.columns.items()}
     out_words, out_w, out_cols = consolidate_sorted(words, weights, cols)
     if out_w.shape[0] == 0:
@@ -537,12 +538,10 @@ class GroupReduceNode(Node):
 
                 words = [gkeys[:, 0].contiguous(), gkeys[:, 1].contiguous()]
                 perm = lex_sort_words(words)
-                sw0 = words[0].index_select(0, perm)
-                sw1 = words[1].index_select(0, perm)
                 names = list(contribs)
-                sc_list = [
-                    contribs[nm].index_select(0, perm) for nm in names
-                ]
+                sw0, sw1, *sc_list = ops.gather_all(
+                    perm, words + [contribs[nm] for nm in names]
+                )
                 uk0, uk1, first_sorted, accs = ops.seg_reduce_gpu(
                     sw0, sw1, sc_list
                 )

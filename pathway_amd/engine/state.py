@@ -133,9 +133,9 @@ def consolidate_sorted(
             words, [weights]
         )
         keep = (wsum != 0).nonzero(as_tuple=True)[0]
-        kept_first = first_idx.index_select(0, keep)
-        kept_w = wsum.index_select(0, keep)
-        kept_words = [w.index_select(0, keep) for w in out_words]
+        kept_first, kept_w, *kept_words = ops.gather_all(
+            keep, [first_idx, wsum] + out_words
+        )
         out_cols = {name: c.take(kept_first) for name, c in columns.items()}
         return kept_words, kept_w, out_cols
     starts = rows_ne(words)
@@ -297,8 +297,9 @@ class Arrangement:
         # 1. sort the (small) delta
         sort_words = dwords[:2] if key_determined_vhash else dwords
         dperm = lex_sort_words(sort_words)
-        dwords = [w.index_select(0, dperm) for w in dwords]
-        dweights = weights.index_select(0, dperm)
+        from pathway_amd import ops
+
+        *dwords, dweights = ops.gather_all(dperm, dwords + [weights])
         dcols = {n: columns[n].take(dperm) for n in self.columns}
         # the delta may carry duplicate rows: consolidate it first so the
         # fused merge's unique-rows precondition holds

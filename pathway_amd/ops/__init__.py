@@ -690,3 +690,51 @@ def scan_positions_gpu(buf: torch.Tensor, target: int) -> torch.Tensor:
     if rc != 0:
         raise RuntimeError(f"pw_scan_emit failed: {rc}")
     return out
+
+
+def gather_cols_gpu(idx: torch.Tensor, cols: list) -> list:
+    """Gather up to 8 8-byte columns through one shared int64 index in a
+    single launch (k_gather_cols) — replaces per-column index_select on
+    the arrange/merge/consolidate permutation paths.  Columns with a
+    non-8-byte itemsize fall back to torch indexing."""
+    lib = require_lib()
+    m = idx.shape[0]
+    outs: list = [None] * len(cols)
+    fused_pos: list[int] = []
+    for i, c in enumerate(cols):
+        if c.dim() == 1 and c.element_size() == 8 and c.is_contiguous():
+            fused_pos.append(i)
+        else:
+            outs[i] = c[idx] if c.dim() > 1 else c.index_select(0, idx)
+    idx = idx.contiguous()
+    for start in range(0, len(fused_pos), 8):
+        group = fused_pos[start : start + 8]
+        srcs = (ctypes.c_void_p * len(group))()
+        dsts = (ctypes.c_void_p * len(group))()
+        for k, i in enumerate(group):
+            c = cols[i]
+            out = torch.empty(m, dtype=c.dtype, device=c.device)
+            outs[i] = out
+            srcs[k] = c.data_ptr()
+            dsts[k] = out.data_ptr()
+        rc = lib.pw_gather_cols(
+            ctypes.c_void_p(idx.data_ptr()),
+            ctypes.c_int64(m),
+            ctypes.c_int(len(group)),
+            srcs,
+            dsts,
+            _stream_ptr(),
+        )
+        if rc != 0:
+            raise RuntimeError(f"pw_gather_cols failed: {rc}")
+    return outs
+
+
+def gather_all(idx: torch.Tensor, tensors: list) -> list:
+    """Gather a list of same-length tensors through one index — fused
+    k_gather_cols on device, per-tensor index_select elsewhere."""
+    if tensors and tensors[0].is_cuda and idx.shape[0] > 2048 and lib_available():
+        return gather_cols_gpu(idx, list(tensors))
+    return [
+        t.index_select(0, idx) if t.dim() == 1 else t[idx] for t in tensors
+    ]

@@ -1405,6 +1405,29 @@ extern "C" int pw_radix_scatter(const void* keys, const void* payload,
 // rocprim partition (torch nonzero) with a count + emit pair whose scan
 // is one torch cumsum.
 
+// 16-bit hit mask of `target` inside the 16-byte granule at byte offset g.
+// Vector path requires 16B alignment of buf+g (the host wrapper rounds the
+// per-block chunk to a multiple of 16 and torch allocations are 256B-aligned,
+// so interior granules qualify); the tail / unaligned path is scalar.
+__device__ __forceinline__ unsigned pw_granule_mask(const uint8_t* buf,
+                                                    int64_t g, int64_t end,
+                                                    uint8_t t, bool aligned) {
+  unsigned m = 0;
+  if (aligned && g + 16 <= end) {
+    uint4 v = *(const uint4*)(buf + g);
+    const uint32_t w[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+    for (int q = 0; q < 4; ++q)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        if (((w[q] >> (8 * j)) & 0xFFu) == t) m |= 1u << (4 * q + j);
+  } else {
+    for (int j = 0; j < 16 && g + j < end; ++j)
+      if (buf[g + j] == t) m |= 1u << j;
+  }
+  return m;
+}
+
 __global__ void k_scan_count(const uint8_t* buf, int64_t n, int target,
                              int64_t chunk, long long* block_counts) {
   __shared__ int cnt;
@@ -1412,9 +1435,13 @@ __global__ void k_scan_count(const uint8_t* buf, int64_t n, int target,
   __syncthreads();
   int64_t start = (int64_t)blockIdx.x * chunk;
   int64_t end = min(start + chunk, n);
+  const bool aligned = (((size_t)buf & 15) == 0);
   int local = 0;
-  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
-    if (buf[i] == (uint8_t)target) ++local;
+  for (int64_t tile = start; tile < end; tile += (int64_t)blockDim.x * 16) {
+    int64_t g = tile + (int64_t)threadIdx.x * 16;
+    if (g < end)
+      local += __popc(pw_granule_mask(buf, g, end, (uint8_t)target, aligned));
+  }
   atomicAdd(&cnt, local);
   __syncthreads();
   if (threadIdx.x == 0) block_counts[blockIdx.x] = cnt;
@@ -1423,31 +1450,42 @@ __global__ void k_scan_count(const uint8_t* buf, int64_t n, int target,
 __global__ void k_scan_emit(const uint8_t* buf, int64_t n, int target,
                             int64_t chunk, const long long* bases,
                             int64_t* out) {
-  // ordered within the block: lanes cooperate wave-sequentially via a
-  // per-block running cursor advanced in lane order per 256-wide tile
+  // Ordered emit: each 256-thread block walks its chunk in 4 KB tiles
+  // (one 16 B granule per lane).  Per tile: wave-level __shfl_up inclusive
+  // scan of per-granule hit counts + a 4-entry cross-wave prefix in LDS —
+  // 2 barriers per 4 KB instead of a 16-barrier log-scan per 256 B.
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
   __shared__ long long cursor;
+  __shared__ int wave_tot[PW_BLOCK / 64];
   if (threadIdx.x == 0) cursor = bases[blockIdx.x];
   __syncthreads();
   int64_t start = (int64_t)blockIdx.x * chunk;
   int64_t end = min(start + chunk, n);
-  for (int64_t base = start; base < end; base += blockDim.x) {
-    int64_t i = base + threadIdx.x;
-    int hit = (i < end && buf[i] == (uint8_t)target) ? 1 : 0;
-    // intra-tile exclusive prefix of hits (ballot across the block via LDS)
-    __shared__ int tile[PW_BLOCK];
-    tile[threadIdx.x] = hit;
-    __syncthreads();
-    // simple scan (block is 256 wide: log-step scan)
-    for (int off = 1; off < PW_BLOCK; off <<= 1) {
-      int v = (threadIdx.x >= off) ? tile[threadIdx.x - off] : 0;
-      __syncthreads();
-      tile[threadIdx.x] += v;
-      __syncthreads();
+  const bool aligned = (((size_t)buf & 15) == 0);
+  for (int64_t tile = start; tile < end; tile += (int64_t)blockDim.x * 16) {
+    int64_t g = tile + (int64_t)threadIdx.x * 16;
+    unsigned hmask =
+        (g < end) ? pw_granule_mask(buf, g, end, (uint8_t)target, aligned) : 0u;
+    int hits = __popc(hmask);
+    int incl = hits;
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+      int v = __shfl_up(incl, off, 64);
+      if (lane >= off) incl += v;
     }
-    int incl = tile[threadIdx.x];
-    if (hit) out[cursor + incl - 1] = i;
+    if (lane == 63) wave_tot[wave] = incl;
     __syncthreads();
-    if (threadIdx.x == PW_BLOCK - 1) cursor += incl;
+    int wbase = 0;
+    for (int w = 0; w < wave; ++w) wbase += wave_tot[w];
+    long long my = cursor + wbase + (incl - hits);
+    while (hmask) {
+      int j = __ffs(hmask) - 1;
+      hmask &= hmask - 1;
+      out[my++] = g + j;
+    }
+    __syncthreads();
+    if (threadIdx.x == (int)blockDim.x - 1) cursor += wbase + incl;
     __syncthreads();
   }
 }
@@ -1456,8 +1494,8 @@ extern "C" int pw_scan_positions(const void* buf, int64_t n, int target,
                                  void* block_counts, int64_t nblocks,
                                  void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  int64_t chunk = (n + nblocks - 1) / nblocks;
-  if (chunk < 1) chunk = 1;
+  int64_t chunk = ((n + nblocks - 1) / nblocks + 15) & ~(int64_t)15;
+  if (chunk < 16) chunk = 16;
   hipLaunchKernelGGL(k_scan_count, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
                      0, s, (const uint8_t*)buf, n, target,
                      chunk, (long long*)block_counts);
@@ -1468,10 +1506,54 @@ extern "C" int pw_scan_emit(const void* buf, int64_t n, int target,
                             const void* bases, int64_t nblocks, void* out,
                             void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  int64_t chunk = (n + nblocks - 1) / nblocks;
-  if (chunk < 1) chunk = 1;
+  int64_t chunk = ((n + nblocks - 1) / nblocks + 15) & ~(int64_t)15;
+  if (chunk < 16) chunk = 16;
   hipLaunchKernelGGL(k_scan_emit, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
                      0, s, (const uint8_t*)buf, n, target, chunk,
                      (const long long*)bases, (int64_t*)out);
+  return (int)hipGetLastError();
+}
+
+// ------------------------------------------------------- fused gather --
+// One launch gathers up to 8 8-byte columns through a shared index: the
+// arrange/merge/consolidate paths gather (key words, weights, value
+// columns) with the same permutation, and per-column index_select pays
+// the index read + launch overhead k times.  Pointer table passed by
+// value in the kernel args.
+
+struct PwGather8 {
+  const long long* src[8];
+  long long* dst[8];
+};
+
+__global__ void k_gather_cols(const long long* __restrict__ idx, int64_t m,
+                              int ncols, PwGather8 p) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < m; i += stride) {
+    long long j = idx[i];
+#pragma unroll 8
+    for (int c = 0; c < 8; ++c) {
+      if (c >= ncols) break;
+      p.dst[c][i] = p.src[c][j];
+    }
+  }
+}
+
+extern "C" int pw_gather_cols(const void* idx, int64_t m, int ncols,
+                              void* const* srcs, void* const* dsts,
+                              void* stream) {
+  if (ncols < 1 || ncols > 8) return -2;
+  hipStream_t s = (hipStream_t)stream;
+  PwGather8 p;
+  for (int c = 0; c < ncols; ++c) {
+    p.src[c] = (const long long*)srcs[c];
+    p.dst[c] = (long long*)dsts[c];
+  }
+  int64_t nblocks = (m + PW_BLOCK - 1) / PW_BLOCK;
+  if (nblocks > 4096) nblocks = 4096;
+  if (nblocks < 1) nblocks = 1;
+  hipLaunchKernelGGL(k_gather_cols, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
+                     0, s, (const long long*)idx, m, ncols, p);
   return (int)hipGetLastError();
 }

@@ -729,8 +729,35 @@ def test_scan_positions_kernel():
     from pathway_amd import ops
 
     torch.manual_seed(3)
-    for n in (0, 1, 1000, 1 << 20):
+    for n in (0, 1, 17, 1000, 4097, 1 << 20):
         buf = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
         got = ops.scan_positions_gpu(buf, 10)
         ref = (buf == 10).nonzero(as_tuple=True)[0]
         assert torch.equal(got, ref), n
+    # unaligned base (slice off one byte) exercises the scalar path
+    buf = torch.randint(0, 256, (100001,), dtype=torch.uint8, device="cuda")
+    sub = buf[1:]
+    got = ops.scan_positions_gpu(sub, 10)
+    ref = (sub == 10).nonzero(as_tuple=True)[0]
+    assert torch.equal(got, ref)
+
+
+@gpu
+@requires_cuda
+def test_gather_cols_kernel():
+    from pathway_amd import ops
+
+    torch.manual_seed(4)
+    n = 1 << 20
+    for ncols in (1, 3, 8, 11):
+        cols = [
+            torch.randint(-(1 << 60), 1 << 60, (n,), device="cuda")
+            if i % 2 == 0
+            else torch.randn(n, dtype=torch.float64, device="cuda")
+            for i in range(ncols)
+        ]
+        idx = torch.randint(0, n, (n // 3,), device="cuda")
+        got = ops.gather_cols_gpu(idx, cols)
+        for g, c in zip(got, cols):
+            assert g.dtype == c.dtype
+            assert torch.equal(g, c.index_select(0, idx))
