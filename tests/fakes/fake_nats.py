@@ -16,10 +16,17 @@ class FakeNats:
         server_self = self
 
         class Handler(socketserver.StreamRequestHandler):
+            # buffered writer: raw SocketIO.write is a single send() that
+            # may write PARTIALLY under load, desyncing the subscriber's
+            # frame stream (observed as a lost MSG); BufferedWriter.flush
+            # loops until the frame is fully on the wire
+            wbufsize = -1
+
             def handle(self):
                 self.wfile.write(
                     b"INFO " + json.dumps({"server_id": "fake", "version": "2"}).encode() + b"\r\n"
                 )
+                self.wfile.flush()
                 my_subs = []
                 try:
                     while True:
@@ -29,8 +36,10 @@ class FakeNats:
                         line = line.rstrip(b"\r\n")
                         if line.upper().startswith(b"CONNECT"):
                             self.wfile.write(b"+OK\r\n")
+                            self.wfile.flush()
                         elif line.upper() == b"PING":
                             self.wfile.write(b"PONG\r\n")
+                            self.wfile.flush()
                         elif line.upper().startswith(b"SUB "):
                             _, subject, sid = line.decode().split(" ")
                             with server_self.lock:
