@@ -465,3 +465,111 @@ def test_distributed_string_bytes_exchange():
         p.join(60)
     merged = sorted(set(results[0]) | set(results[1]))
     assert merged == [("alpha", 11, 2), ("beta", 2, 1), ("żółć", 10, 2)]
+
+
+def _ab_stream(seed: int):
+    """Seeded update stream of (uid, word, value, time, diff) rows with
+    ~20% retractions of previously inserted rows (same uid, diff=-1)."""
+    import random
+
+    rng = random.Random(seed)
+    words = [f"w{i:03d}" for i in range(40)]
+    live = []
+    events = []
+    t = 0
+    for step in range(400):
+        if step % 25 == 0:
+            t += 1
+        if live and rng.random() < 0.2:
+            uid, w, v = live.pop(rng.randrange(len(live)))
+            events.append((uid, w, v, t, -1))
+        else:
+            uid, w, v = step, rng.choice(words), rng.randrange(100)
+            live.append((uid, w, v))
+            events.append((uid, w, v, t, 1))
+    return events
+
+
+def _ab_oracle(events):
+    agg = {}
+    for _uid, w, v, _t, d in events:
+        c, s = agg.get(w, (0, 0))
+        agg[w] = (c + d, s + d * v)
+    return sorted((w, c, s) for w, (c, s) in agg.items() if c)
+
+
+def _ab_engine(events):
+    """groupby(word) -> (count, sum) over the update stream via the engine."""
+    import pathway_amd as pw
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.internals.schema import schema_builder, column_definition
+
+    G.clear()
+    schema = schema_builder(
+        {
+            "uid": column_definition(primary_key=True, dtype=int),
+            "w": column_definition(dtype=str),
+            "v": column_definition(dtype=int),
+        }
+    )
+    t = pw.debug.table_from_rows(schema, events, is_stream=True)
+    res = t.groupby(pw.this.w).reduce(
+        pw.this.w, c=pw.reducers.count(), s=pw.reducers.sum(pw.this.v)
+    )
+    keys, cols = pw.debug.table_to_dicts(res)
+    return sorted((cols["w"][k], cols["c"][k], cols["s"][k]) for k in keys)
+
+
+def _worker_ab(rank: int, world: int, port: int, seed: int, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["PW_DEVICE"] = "cpu"
+    import zlib
+
+    import pathway_amd as pw  # noqa: F401
+    import pathway_amd.parallel as par
+
+    par.init(backend="gloo")
+    events = _ab_stream(seed)
+    # deterministic partition of the stream by word (salt-free hash so
+    # every rank computes the same split; insert+retract share a word,
+    # hence a rank)
+    mine = [e for e in events if zlib.crc32(e[1].encode()) % world == rank]
+    rows = _ab_engine(mine)
+    q.put((rank, rows))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_worker_count_ab_determinism():
+    """results must be bit-identical across worker counts.
+    The same seeded insert+retract stream is run at world=1 (in-process)
+    and world=2 (gloo multiprocess); both must equal the host oracle."""
+    seed = 1234
+    events = _ab_stream(seed)
+    expected = _ab_oracle(events)
+    assert _ab_engine(events) == expected  # world = 1
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_worker_ab, args=(r, 2, port, seed, q))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, rows = q.get()
+        results[rank] = rows
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0
+    merged = sorted(results[0] + results[1])
+    assert merged == expected
+    assert not (set(results[0]) & set(results[1]))
