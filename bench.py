@@ -247,14 +247,14 @@ def main() -> None:
                 if use_cuda:
                     tbuf = tbuf.pin_memory()
                 self.host_bufs.append(tbuf)
-            # ---- device dictionary: sorted pool-hash table ----
-            from pathway_amd.engine.state import lex_sort_words
-
+            # ---- device dictionary: open-addressing hash table ----
+            # pool code == row index in the pool hash tensors, so the
+            # identity-valued table maps token hash -> code in O(1)
             plo, phi = GLOBAL_STRING_POOL.hash_tensors(device)
-            dperm = lex_sort_words([plo, phi])
-            self.dict_lo = plo.index_select(0, dperm).contiguous()
-            self.dict_hi = phi.index_select(0, dperm).contiguous()
-            self.dict_codes = dperm.contiguous()  # sorted pos -> pool code
+            if use_cuda:
+                from pathway_amd import ops
+
+                self.dict_ht = ops.DeviceHashTable(plo, phi)
             self.miss = torch.zeros((), dtype=torch.int64, device=device)
             # ---- double-buffered staging ----
             if use_cuda:
@@ -304,14 +304,9 @@ def main() -> None:
                 ends = nl
                 glo, ghi = ops.varlen_hash_se_gpu(buf, starts, ends, TAG_STR)
                 # word column derived from the parsed bytes: dictionary
-                # lookup of the token hash in the sorted pool table
-                pos, found = ops.lookup_gpu(
-                    [self.dict_lo, self.dict_hi], [glo, ghi]
-                )
-                codes = self.dict_codes.index_select(
-                    0, pos.clamp(0, self.dict_codes.shape[0] - 1)
-                )
-                codes = torch.where(found, codes, torch.full_like(codes, -1))
+                # probe of the token hash in the device hash table (codes
+                # are -1 on miss already)
+                codes, found = self.dict_ht.probe(glo, ghi)
                 self.miss += (~found).sum()
                 # this step's reads of `buf` are enqueued; allow the copy
                 # stream to overwrite the slot for step t+2

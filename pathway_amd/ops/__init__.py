@@ -730,6 +730,61 @@ def gather_cols_gpu(idx: torch.Tensor, cols: list) -> list:
     return outs
 
 
+class DeviceHashTable:
+    """Open-addressing device hash table over 128-bit keys -> int64 values
+    (k_ht_build / k_ht_probe).  Build once per dictionary version; probe
+    per delta batch.  Load factor <= 0.5 (nslots = next pow2 >= 2m)."""
+
+    def __init__(self, klo: torch.Tensor, khi: torch.Tensor,
+                 vals: torch.Tensor | None = None):
+        lib = require_lib()
+        m = klo.shape[0]
+        device = klo.device
+        nslots = 1 << max(4, (2 * m - 1).bit_length()) if m else 16
+        self.nslots = nslots
+        self.tab_lo = torch.empty(nslots, dtype=torch.int64, device=device)
+        self.tab_hi = torch.empty(nslots, dtype=torch.int64, device=device)
+        self.tab_val = torch.full((nslots,), -1, dtype=torch.int64, device=device)
+        if m:
+            rc = lib.pw_ht_build(
+                ctypes.c_void_p(klo.contiguous().data_ptr()),
+                ctypes.c_void_p(khi.contiguous().data_ptr()),
+                ctypes.c_void_p(vals.contiguous().data_ptr()) if vals is not None else None,
+                ctypes.c_int64(m),
+                ctypes.c_void_p(self.tab_lo.data_ptr()),
+                ctypes.c_void_p(self.tab_hi.data_ptr()),
+                ctypes.c_void_p(self.tab_val.data_ptr()),
+                ctypes.c_int64(nslots),
+                _stream_ptr(),
+            )
+            if rc != 0:
+                raise RuntimeError(f"pw_ht_build failed: hip error {rc}")
+
+    def probe(self, qlo: torch.Tensor, qhi: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+        """Returns (values, found); values are -1 where not found."""
+        lib = require_lib()
+        nq = qlo.shape[0]
+        out = torch.empty(nq, dtype=torch.int64, device=qlo.device)
+        found = torch.empty(nq, dtype=torch.bool, device=qlo.device)
+        if nq == 0:
+            return out, found
+        rc = lib.pw_ht_probe(
+            ctypes.c_void_p(qlo.contiguous().data_ptr()),
+            ctypes.c_void_p(qhi.contiguous().data_ptr()),
+            ctypes.c_int64(nq),
+            ctypes.c_void_p(self.tab_lo.data_ptr()),
+            ctypes.c_void_p(self.tab_hi.data_ptr()),
+            ctypes.c_void_p(self.tab_val.data_ptr()),
+            ctypes.c_int64(self.nslots),
+            ctypes.c_void_p(out.data_ptr()),
+            ctypes.c_void_p(found.data_ptr()),
+            _stream_ptr(),
+        )
+        if rc != 0:
+            raise RuntimeError(f"pw_ht_probe failed: hip error {rc}")
+        return out, found
+
+
 def gather_all(idx: torch.Tensor, tensors: list) -> list:
     """Gather a list of same-length tensors through one index — fused
     k_gather_cols on device, per-tensor index_select elsewhere."""

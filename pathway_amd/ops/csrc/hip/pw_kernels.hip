@@ -1540,6 +1540,90 @@ __global__ void k_gather_cols(const long long* __restrict__ idx, int64_t m,
   }
 }
 
+// ------------------------------------------------------ hash table --
+// Open-addressing (linear probe) device hash table over 128-bit keys —
+// O(1) replacement for the binary-search k_lookup on hot stable
+// dictionaries (the string-pool decode probes 4M tokens/step against a
+// 50k vocab: ~1.2 slot loads beat a 16-level dependent-load search).
+// Values are int64 >= 0; slot value -1 means empty.  Build never reads
+// keys (inputs are unique), so a single atomicCAS on the value slot
+// claims it.
+
+__global__ void k_ht_build(const int64_t* __restrict__ klo,
+                           const int64_t* __restrict__ khi,
+                           const int64_t* __restrict__ vals, int64_t m,
+                           int64_t* tab_lo, int64_t* tab_hi,
+                           long long* tab_val, int64_t mask) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < m; i += stride) {
+    uint64_t slot = (uint64_t)klo[i] & (uint64_t)mask;
+    while (atomicCAS((unsigned long long*)&tab_val[slot],
+                     (unsigned long long)(long long)-1,
+                     (unsigned long long)(long long)(vals ? vals[i] : i)) !=
+           (unsigned long long)(long long)-1)
+      slot = (slot + 1) & (uint64_t)mask;
+    tab_lo[slot] = klo[i];
+    tab_hi[slot] = khi[i];
+  }
+}
+
+__global__ void k_ht_probe(const int64_t* __restrict__ qlo,
+                           const int64_t* __restrict__ qhi, int64_t nq,
+                           const int64_t* __restrict__ tab_lo,
+                           const int64_t* __restrict__ tab_hi,
+                           const long long* __restrict__ tab_val,
+                           int64_t mask, int64_t* out, bool* found) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < nq; i += stride) {
+    int64_t lo = qlo[i], hi = qhi[i];
+    uint64_t slot = (uint64_t)lo & (uint64_t)mask;
+    int64_t v = -1;
+    for (;;) {
+      long long tv = tab_val[slot];
+      if (tv == -1) break;  // empty -> miss
+      if (tab_lo[slot] == lo && tab_hi[slot] == hi) {
+        v = (int64_t)tv;
+        break;
+      }
+      slot = (slot + 1) & (uint64_t)mask;
+    }
+    out[i] = v;
+    found[i] = v != -1;
+  }
+}
+
+extern "C" int pw_ht_build(const void* klo, const void* khi, const void* vals,
+                           int64_t m, void* tab_lo, void* tab_hi,
+                           void* tab_val, int64_t nslots, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int64_t nblocks = (m + PW_BLOCK - 1) / PW_BLOCK;
+  if (nblocks > 4096) nblocks = 4096;
+  if (nblocks < 1) nblocks = 1;
+  hipLaunchKernelGGL(k_ht_build, dim3((uint32_t)nblocks), dim3(PW_BLOCK), 0, s,
+                     (const int64_t*)klo, (const int64_t*)khi,
+                     (const int64_t*)vals, m, (int64_t*)tab_lo,
+                     (int64_t*)tab_hi, (long long*)tab_val, nslots - 1);
+  return (int)hipGetLastError();
+}
+
+extern "C" int pw_ht_probe(const void* qlo, const void* qhi, int64_t nq,
+                           const void* tab_lo, const void* tab_hi,
+                           const void* tab_val, int64_t nslots, void* out,
+                           void* found, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int64_t nblocks = (nq + PW_BLOCK - 1) / PW_BLOCK;
+  if (nblocks > 4096) nblocks = 4096;
+  if (nblocks < 1) nblocks = 1;
+  hipLaunchKernelGGL(k_ht_probe, dim3((uint32_t)nblocks), dim3(PW_BLOCK), 0, s,
+                     (const int64_t*)qlo, (const int64_t*)qhi, nq,
+                     (const int64_t*)tab_lo, (const int64_t*)tab_hi,
+                     (const long long*)tab_val, nslots - 1, (int64_t*)out,
+                     (bool*)found);
+  return (int)hipGetLastError();
+}
+
 extern "C" int pw_gather_cols(const void* idx, int64_t m, int ncols,
                               void* const* srcs, void* const* dsts,
                               void* stream) {

@@ -744,6 +744,34 @@ def test_scan_positions_kernel():
 
 @gpu
 @requires_cuda
+def test_device_hash_table():
+    from pathway_amd import ops
+
+    torch.manual_seed(5)
+    for m in (0, 1, 1000, 50_000):
+        klo = torch.randperm(1 << 20, device="cuda")[:m] * 7919 + 13
+        khi = torch.randint(-(1 << 62), 1 << 62, (m,), device="cuda")
+        vals = torch.randint(0, 1 << 40, (m,), device="cuda")
+        ht = ops.DeviceHashTable(klo, khi, vals)
+        # every inserted key must probe back to its value
+        got, found = ht.probe(klo, khi)
+        assert bool(found.all())
+        assert torch.equal(got, vals)
+        # identity-valued table (vals=None) returns the insert index
+        ht2 = ops.DeviceHashTable(klo, khi)
+        got2, found2 = ht2.probe(klo, khi)
+        assert bool(found2.all())
+        assert torch.equal(got2, torch.arange(m, device="cuda"))
+        # unknown keys miss with -1
+        miss_lo = klo + 1 if m else torch.tensor([42], device="cuda")
+        miss_hi = khi if m else torch.tensor([43], device="cuda")
+        gotm, foundm = ht.probe(miss_lo, miss_hi)
+        assert not bool(foundm.any())
+        assert bool((gotm == -1).all())
+
+
+@gpu
+@requires_cuda
 def test_gather_cols_kernel():
     from pathway_amd import ops
 
