@@ -613,32 +613,35 @@ class Table(TableLike):
 
     def _gradual_broadcast(self, threshold_table, lower_column, value_column, upper_column) -> "Table":
         """Apportion a broadcast value across rows (reference
-        operators/gradual_broadcast.rs:65): each row receives a value in
-        [lower, upper] proportional to its apportioning position."""
+        operators/gradual_broadcast.rs:120-190): the key space is split at
+        threshold = max_key * (value - lower) / (upper - lower); rows whose
+        key falls below the threshold receive `upper`, the rest `lower` —
+        so the fraction of rows at `upper` tracks the requested value
+        (keys are uniform hashes; "scaling does not need to be precise")."""
         from pathway_amd.engine.nodes_recompute import RecomputeNode
 
         lname = lower_column.name
         vname = value_column.name
         uname = upper_column.name
         src = self
+        _U64 = (1 << 64) - 1
 
         def fn(in_rows, in_keys):
             rows, keys = in_rows[0], in_keys[0]
             trows = in_rows[1]
-            if trows:
-                lo = trows[0][lname]
-                v = trows[0][vname]
-                hi = trows[0][uname]
-            else:
-                lo = v = hi = None
             out = []
-            ordered = sorted(zip(rows, keys), key=lambda rk: repr(rk[1]))
-            n = len(ordered)
-            for i, (row, k) in enumerate(ordered):
-                if v is None:
-                    out.append((k, {"apx_value": None}))
-                else:
-                    out.append((k, {"apx_value": v}))
+            if not trows:
+                return [(k, {"apx_value": None}) for k in keys]
+            lo = float(trows[0][lname])
+            v = float(trows[0][vname])
+            hi = float(trows[0][uname])
+            frac = 1.0 if hi == lo else max(0.0, min(1.0, (v - lo) / (hi - lo)))
+            threshold = int(frac * float(1 << 64))
+            for k in keys:
+                ku = k.as_signed_pair()[0] & _U64
+                out.append(
+                    (k, {"apx_value": hi if ku < threshold else lo})
+                )
             return out
 
         out_dtypes = {"apx_value": dt.Optional(dt.FLOAT)}

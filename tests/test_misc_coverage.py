@@ -525,3 +525,35 @@ def test_license_ed25519_and_entitlements():
     with pytest.raises(RuntimeError):
         lic.check_worker_limit(9, None)
     lic.check_worker_limit(64, key)
+
+
+def test_gradual_broadcast_apportioning():
+    """Reference gradual_broadcast.rs:120-190: rows whose key < max_key *
+    (value-lower)/(upper-lower) receive `upper`, the rest `lower`; the
+    fraction at `upper` tracks the requested value over uniform keys."""
+    from pathway_amd.internals.rungraph import G
+
+    G.clear()
+    n = 400
+    rows = table_from_rows(
+        schema_from_types(x=int), [(i,) for i in range(n)]
+    )
+    thr = table_from_rows(
+        schema_from_types(lo=float, v=float, hi=float), [(0.0, 0.25, 1.0)]
+    )
+    res = rows._gradual_broadcast(thr, thr.lo, thr.v, thr.hi)
+    _keys, cols = pw.debug.table_to_dicts(res)
+    vals = list(cols["apx_value"].values())
+    assert set(vals) <= {0.0, 1.0}
+    frac = sum(1 for v in vals if v == 1.0) / n
+    # hashes are uniform: the upper-share must track (v-lo)/(hi-lo)
+    assert 0.15 < frac < 0.35
+    # degenerate interval: everything gets the collapsed bound
+    G.clear()
+    rows2 = table_from_rows(schema_from_types(x=int), [(i,) for i in range(5)])
+    thr2 = table_from_rows(
+        schema_from_types(lo=float, v=float, hi=float), [(2.0, 2.0, 2.0)]
+    )
+    res2 = rows2._gradual_broadcast(thr2, thr2.lo, thr2.v, thr2.hi)
+    _k2, cols2 = pw.debug.table_to_dicts(res2)
+    assert set(cols2["apx_value"].values()) == {2.0}
