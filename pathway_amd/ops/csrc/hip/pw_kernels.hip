@@ -812,15 +812,20 @@ __device__ __forceinline__ float pw_gelu(float x) {
 // tiles then stage into LDS with linear 16B vector loads and both MFMA
 // fragments are contiguous 8-element ds reads (weights are static, so
 // the one-time transpose is free on the host side).
+template <int NBUF>
 __global__ __launch_bounds__(256) void k_gemm_bf16(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bt,
     const float* __restrict__ bias, __bf16* __restrict__ C, int64_t M,
     int64_t N, int64_t K, int act, int nbm) {
-  __shared__ __bf16 lasA[2][PW_GEMM_BM][PW_GEMM_BK];
+  // NBUF=2: classic intra-block double buffer (64 KB LDS, 2 blocks/CU).
+  // NBUF=1: single buffer (32 KB LDS, ~5 blocks/CU) — latency hiding
+  // from BLOCK-level parallelism instead; wins on short-K shapes where
+  // the 6-iteration pipeline never fills (PMC: MfmaUtil 8.9% at NBUF=2).
+  __shared__ __bf16 lasA[NBUF][PW_GEMM_BM][PW_GEMM_BK];
   // B stored TRANSPOSED in LDS ([col][k]) so each lane's 8-element
   // K-fragment is a contiguous ds_read_b128 (scattered writes happen
   // once per element; scattered reads would repeat per MFMA)
-  __shared__ __bf16 lasBt[2][PW_GEMM_BN][PW_GEMM_BK];
+  __shared__ __bf16 lasBt[NBUF][PW_GEMM_BN][PW_GEMM_BK];
   // XCD-aware tile mapping: consecutive blockIdx tiles land on the same
   // XCD's L2 by swizzling in chunks (guide: blockIdx->tile must be
   // XCD-aware).  Simple bijective row-major with N-fastest is fine here
@@ -918,27 +923,41 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
-    int buf = kt & 1;
-    if (kt + 1 < nk) stage(kt + 1, buf ^ 1);
+    int buf = NBUF == 2 ? (kt & 1) : 0;
+    if (NBUF == 2 && kt + 1 < nk) stage(kt + 1, buf ^ 1);
 #pragma unroll
     for (int kk = 0; kk < PW_GEMM_BK / 32; ++kk) {
+      // hoist every fragment read out of the MFMA loops: B re-read per m
+      // was 4x the LDS traffic and the dominant bank-conflict source
+      // (PMC: LdsBankConflict ratio 3.0 before, MfmaUtil 8.9%)
+      int ak = kk * 32 + (lane >> 4) * 8;
+      pw_frag8 afrag[4], bfrag[4];
 #pragma unroll
       for (int m = 0; m < 4; ++m) {
         int arow = wr * 64 + m * 16 + (lane & 15);
-        int ak = kk * 32 + (lane >> 4) * 8;
-        pw_frag8 afrag = *(const pw_frag8*)&lasA[buf][arow][ak];
-#pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          int bcol = wc * 64 + n * 16 + (lane & 15);
-          int bk = kk * 32 + (lane >> 4) * 8;
-          pw_frag8 bfrag = *(const pw_frag8*)&lasBt[buf][bcol][bk];
-          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag, bfrag, acc[m][n], 0, 0, 0);
-        }
+        afrag[m] = *(const pw_frag8*)&lasA[buf][arow][ak];
       }
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        int bcol = wc * 64 + n * 16 + (lane & 15);
+        bfrag[n] = *(const pw_frag8*)&lasBt[buf][bcol][ak];
+      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[m], bfrag[n], acc[m][n], 0, 0, 0);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+    if (NBUF == 2) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    } else if (kt + 1 < nk) {
+      __syncthreads();  // everyone done reading buf before overwrite
+      stage(kt + 1, 0);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
   }
 
   // epilogue: bias + activation, bf16 store
@@ -967,9 +986,17 @@ extern "C" int pw_gemm_bf16(const void* A, const void* B, const void* bias,
   int nbm = (int)((M + PW_GEMM_BM - 1) / PW_GEMM_BM);
   int nbn = (int)((N + PW_GEMM_BN - 1) / PW_GEMM_BN);
   hipStream_t s = (hipStream_t)stream;
-  hipLaunchKernelGGL(k_gemm_bf16, dim3((uint32_t)(nbm * nbn)), dim3(256), 0,
-                     s, (const __bf16*)A, (const __bf16*)B,
-                     (const float*)bias, (__bf16*)C, M, N, K, act, nbm);
+  // short-K shapes (few BK iterations) run the single-buffer variant:
+  // the 2-deep pipeline never fills at nk<=8, so 5-blocks/CU occupancy
+  // hides latency better than intra-block double buffering
+  if ((K + PW_GEMM_BK - 1) / PW_GEMM_BK <= 8)
+    hipLaunchKernelGGL(k_gemm_bf16<1>, dim3((uint32_t)(nbm * nbn)), dim3(256),
+                       0, s, (const __bf16*)A, (const __bf16*)B,
+                       (const float*)bias, (__bf16*)C, M, N, K, act, nbm);
+  else
+    hipLaunchKernelGGL(k_gemm_bf16<2>, dim3((uint32_t)(nbm * nbn)), dim3(256),
+                       0, s, (const __bf16*)A, (const __bf16*)B,
+                       (const float*)bias, (__bf16*)C, M, N, K, act, nbm);
   return (int)hipGetLastError();
 }
 

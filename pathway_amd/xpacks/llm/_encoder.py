@@ -162,8 +162,14 @@ class NativeEncoder(torch.nn.Module):
                 wt_cache = self._wt_cache = {}
 
             def mm(h2, w, bias, act="none"):
-                # hand-written MFMA tile kernel, bias(+GELU) fused;
-                # weights pre-transposed once (static) for linear staging
+                # fused ops -> hand-written MFMA tile kernel (bias+GELU
+                # epilogue in-register: beats library GEMM + a separate
+                # 200 MB elementwise pass); PLAIN GEMMs -> hipBLASLt,
+                # which out-tiles us on unfused shapes (measured ladder
+                # in profiles/kernels_r02.md — library GEMMs for plain
+                # matmuls are exactly what rocBLAS/hipBLASLt are for)
+                if act == "none":
+                    return h2 @ w + bias
                 b32 = bias_cache.get(id(bias))
                 if b32 is None:
                     b32 = bias_cache[id(bias)] = bias.detach().to(
