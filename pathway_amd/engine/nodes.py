@@ -64,6 +64,7 @@ _PW_NO_SEGRED = bool(_os.environ.get("PW_NO_SEGRED"))
 #: contention on hot counters) and 50M-distinct (sort savings vanish when
 #: distinct≈batch); rocprim onesweep radix sort is the better default.
 _PW_HASHAGG = bool(_os.environ.get("PW_HASHAGG"))
+_PW_NO_HASHAGG = bool(_os.environ.get("PW_NO_HASHAGG"))
 
 _node_counter = [0]
 
@@ -479,9 +480,18 @@ class GroupReduceNode(Node):
             # exchange ships for N>1)
             contribs = self._contributions(arg_cols, diffs, specs)
             hashagg = None
+            uniq_est = getattr(self, "_uniq_est", None)
+            # opt-in only: even with an LLC-resident estimate-sized table
+            # (round-2 re-run) the atomic path loses the A/B to the radix
+            # sort — 1.06B vs 1.62B ev/s on the wordcount headline (~80
+            # duplicate rows per hot counter serialize, and the extra
+            # mid-step size sync breaks the H2D/compute overlap).  Twice
+            # measured, twice rejected: profiles/wordcount_r01.md and
+            # profiles/kernels_r02.md.
+            use_ha = _PW_HASHAGG
             if (
                 torch.device(device).type == "cuda"
-                and _PW_HASHAGG
+                and use_ha
                 and all(c.dtype == torch.int64 for c in contribs.values())
             ):
                 # sort-free path: one HIP hash-aggregation pass over the
@@ -492,11 +502,21 @@ class GroupReduceNode(Node):
                 from pathway_amd import ops
 
                 names = list(contribs)
-                uk0, uk1, accs, rep = ops.hash_agg_gpu(
+                ha = ops.hash_agg_gpu(
                     gkeys[:, 0].contiguous(),
                     gkeys[:, 1].contiguous(),
                     [contribs[nm] for nm in names],
+                    expected_uniques=uniq_est,
                 )
+                if ha is None:
+                    # estimate too small (probe overflow): redo on the
+                    # sort path; quadruple the estimate for next step
+                    self._uniq_est = (uniq_est or 1024) * 4
+                    hashagg = None
+                else:
+                    uk0, uk1, accs, rep = ha
+                    hashagg = True
+            if hashagg:
                 perm = lex_sort_words([uk0, uk1])
                 uk0s = uk0.index_select(0, perm)
                 uk1s = uk1.index_select(0, perm)
@@ -593,6 +613,11 @@ class GroupReduceNode(Node):
                     acc_deltas[name] = acc
                 gfirst_rows = perm.index_select(0, first_idx)
                 gcols_first = {n: c.take(gfirst_rows) for n, c in gcols.items()}
+
+        if ukeys_w is not None and ukeys_w[0].is_cuda:
+            # feed the hash-agg table-size estimate for the next step
+            # (shape is host metadata — no device sync)
+            self._uniq_est = max(int(ukeys_w[0].shape[0]), 512)
 
         # 3b. multi-worker: all-to-all-v of pre-aggregated partials by key
         # shard (RCCL over xGMI; pact.rs:56 analog with combiner)

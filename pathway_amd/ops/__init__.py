@@ -284,26 +284,39 @@ def varlen_hash_se_gpu(
 
 
 def hash_agg_gpu(
-    k0: torch.Tensor, k1: torch.Tensor, contribs: Sequence[torch.Tensor]
-) -> tuple[torch.Tensor, torch.Tensor, list, torch.Tensor]:
+    k0: torch.Tensor,
+    k1: torch.Tensor,
+    contribs: Sequence[torch.Tensor],
+    expected_uniques: int | None = None,
+) -> tuple[torch.Tensor, torch.Tensor, list, torch.Tensor] | None:
     """Sort-free additive pre-aggregation (pw_hash_agg): returns
     (uk0, uk1, [accs...], rep_row_idx) for the distinct group keys of the
     batch — UNSORTED and possibly with rare duplicate keys (publication
-    race); callers sort + consolidate the (small) result."""
+    race); callers sort + consolidate the (small) result.
+
+    expected_uniques sizes the table at ~2x that (L2-resident when the
+    estimate is vocabulary-scale — the round-1 A/B lost because the 2n
+    table was a 270 MB HBM random walk).  Returns None if the estimate
+    was too small (probe-chain overflow): redo on the sort path with a
+    bigger estimate."""
     lib = require_lib()
     n = k0.shape[0]
     dev = k0.device
     nacc = len(contribs)
-    cap = 1 << max(10, (2 * n - 1).bit_length())
+    est = n if expected_uniques is None else min(max(expected_uniques, 512), n)
+    cap = 1 << max(10, (2 * est - 1).bit_length())
+    out_cap = min(n, cap)
     tk0 = torch.empty(cap, dtype=torch.int64, device=dev)
     tk1 = torch.zeros(cap, dtype=torch.int64, device=dev)
     rep = torch.zeros(cap, dtype=torch.int64, device=dev)
     taccs = [torch.zeros(cap, dtype=torch.int64, device=dev) for _ in range(nacc)]
-    counter = torch.zeros(1, dtype=torch.int32, device=dev)
-    out_k0 = torch.empty(n, dtype=torch.int64, device=dev)
-    out_k1 = torch.empty(n, dtype=torch.int64, device=dev)
-    out_rep = torch.empty(n, dtype=torch.int64, device=dev)
-    out_accs = [torch.empty(n, dtype=torch.int64, device=dev) for _ in range(nacc)]
+    counter = torch.zeros(2, dtype=torch.int32, device=dev)  # [count, overflow]
+    out_k0 = torch.empty(out_cap, dtype=torch.int64, device=dev)
+    out_k1 = torch.empty(out_cap, dtype=torch.int64, device=dev)
+    out_rep = torch.empty(out_cap, dtype=torch.int64, device=dev)
+    out_accs = [
+        torch.empty(out_cap, dtype=torch.int64, device=dev) for _ in range(nacc)
+    ]
     carr = (ctypes.c_void_p * max(nacc, 1))(
         *[ctypes.c_void_p(c.contiguous().data_ptr()) for c in contribs]
     )
@@ -329,12 +342,15 @@ def hash_agg_gpu(
         ctypes.c_void_p(out_k1.data_ptr()),
         oarr,
         ctypes.c_void_p(out_rep.data_ptr()),
+        ctypes.c_void_p(counter.data_ptr() + 4),
         _stream_ptr(),
     )
     if rc != 0:
         raise RuntimeError(f"pw_hash_agg failed: hip error {rc}")
     # counter is the only host-visible size: one small D2H sync
-    m = int(counter.item())
+    m, overflowed = (int(x) for x in counter.tolist())
+    if overflowed:
+        return None
     return (
         out_k0.narrow(0, 0, m),
         out_k1.narrow(0, 0, m),

@@ -482,21 +482,28 @@ struct AccPtrsMut {
 __global__ void k_hash_agg(const long long* k0, const long long* k1,
                            AccPtrs contribs, int nacc, int64_t n,
                            unsigned long long* tk0, long long* tk1,
-                           AccPtrsMut taccs, long long* rep, uint64_t mask) {
+                           AccPtrsMut taccs, long long* rep, uint64_t mask,
+                           unsigned int* overflow) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     unsigned long long key0 = (unsigned long long)k0[i];
     long long key1 = k1[i];
     uint64_t slot = ((uint64_t)key0) & mask;
-    for (;;) {
-      unsigned long long prev =
-          atomicCAS(&tk0[slot], PW_HA_EMPTY, key0);
-      if (prev == PW_HA_EMPTY) {
+    // probe cap: the table is sized ~2x expected uniques (L2-resident);
+    // a longer chain means the estimate was wrong -> flag and let the
+    // caller redo the batch on the sort path with a bigger table
+    for (uint64_t tries = 0; tries < 512 && tries <= mask; ++tries) {
+      // plain load first: the overwhelmingly common case is an already
+      // claimed, matching slot — an L2 read beats a CAS round-trip
+      unsigned long long cur = tk0[slot];
+      if (cur == PW_HA_EMPTY)
+        cur = atomicCAS(&tk0[slot], PW_HA_EMPTY, key0);
+      if (cur == PW_HA_EMPTY) {
         // claimed: publish the rest of the key + a representative row
         tk1[slot] = key1;
         rep[slot] = (long long)i;
         __threadfence();
-      } else if (prev != key0) {
+      } else if (cur != key0) {
         slot = (slot + 1) & mask;
         continue;
       } else if (tk1[slot] != key1) {
@@ -508,8 +515,10 @@ __global__ void k_hash_agg(const long long* k0, const long long* k1,
       for (int a = 0; a < nacc; ++a)
         atomicAdd((unsigned long long*)&taccs.p[a][slot],
                   (unsigned long long)contribs.p[a][i]);
-      break;
+      goto next_row;
     }
+    atomicOr(overflow, 1u);
+  next_row:;
   }
 }
 
@@ -542,7 +551,7 @@ extern "C" int pw_hash_agg(const void* k0, const void* k1,
                            void* tk0, void* tk1, void** tacc_ptrs, void* rep,
                            int64_t capacity, void* counter, void* out_k0,
                            void* out_k1, void** out_acc_ptrs, void* out_rep,
-                           void* stream) {
+                           void* overflow, void* stream) {
   if (nacc > 8) return 1;
   hipStream_t s = (hipStream_t)stream;
   AccPtrs cp;
@@ -561,7 +570,7 @@ extern "C" int pw_hash_agg(const void* k0, const void* k1,
   hipLaunchKernelGGL(k_hash_agg, dim3(pw_grid(n)), block, 0, s,
                      (const long long*)k0, (const long long*)k1, cp, nacc, n,
                      (unsigned long long*)tk0, (long long*)tk1, tp,
-                     (long long*)rep, mask);
+                     (long long*)rep, mask, (unsigned int*)overflow);
   hipLaunchKernelGGL(k_hash_agg_collect, dim3(pw_grid(capacity)), block, 0, s,
                      (const unsigned long long*)tk0, (const long long*)tk1,
                      tpc, (long long*)rep, nacc, (uint64_t)capacity,
