@@ -257,12 +257,32 @@ class FilePollReader:
             for root, _, files in os.walk(self.path):
                 for f in sorted(files):
                     out.append(os.path.join(root, f))
-            return out
-        import glob
+        else:
+            import glob
 
-        return sorted(glob.glob(self.path)) or (
-            [self.path] if os.path.exists(self.path) else []
-        )
+            out = sorted(glob.glob(self.path)) or (
+                [self.path] if os.path.exists(self.path) else []
+            )
+        return self._shard_paths(out)
+
+    @staticmethod
+    def _shard_paths(paths: list[str]) -> list[str]:
+        """Multi-worker file assignment by path hash (reference
+        connectors/data_storage/sharding.rs): each file is read by
+        exactly one worker; salt-free crc32 so every rank computes the
+        same split."""
+        from pathway_amd import parallel as par
+
+        comm = par.get_comm()
+        if comm is None or comm.world <= 1:
+            return paths
+        import zlib
+
+        return [
+            p
+            for p in paths
+            if zlib.crc32(p.encode()) % comm.world == comm.rank
+        ]
 
     def _retract_file(self, path: str) -> None:
         """Object deletion/modification: retract the rows previously

@@ -573,3 +573,56 @@ def test_worker_count_ab_determinism():
     merged = sorted(results[0] + results[1])
     assert merged == expected
     assert not (set(results[0]) & set(results[1]))
+
+
+def _worker_fs_shard(rank: int, world: int, port: int, tmpdir: str, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["PW_DEVICE"] = "cpu"
+    import pathway_amd as pw
+    import pathway_amd.parallel as par
+    from pathway_amd.internals.rungraph import G
+
+    par.init(backend="gloo")
+    G.clear()
+    t = pw.io.fs.read(
+        tmpdir, format="plaintext", mode="streaming", _max_polls=3,
+        refresh_interval=0.05,
+    )
+    keys, cols = pw.debug.table_to_dicts(t)
+    rows = sorted(cols["data"].values())
+    q.put((rank, rows))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_distributed_fs_file_sharding(tmp_path):
+    """Reference sharding.rs: in multi-worker mode each file is read by
+    exactly one worker (path-hash assignment) — no duplicated rows."""
+    # varied name lengths: crc32 is GF(2)-linear, so names differing in a
+    # single same-position digit can all share parity
+    for i in range(8):
+        (tmp_path / f"f{i}{'x' * i}.txt").write_text(f"line-{i}\n")
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_worker_fs_shard, args=(r, 2, port, str(tmp_path), q))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, rows = q.get()
+        results[rank] = rows
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0
+    assert sorted(results[0] + results[1]) == [f"line-{i}" for i in range(8)]
+    assert not (set(results[0]) & set(results[1]))
+    assert results[0] and results[1]  # both ranks got a share
