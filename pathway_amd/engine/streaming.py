@@ -214,7 +214,24 @@ class StreamingSource(Source):
         pass
 
 
-def spawn_reader(fn: Callable[[], None]) -> threading.Thread:
+def spawn_reader(
+    fn: Callable[[], None], source=None, sharded: bool = False
+) -> threading.Thread | None:
+    """Start a connector reader thread.
+
+    Multi-worker: a single-stream reader (CDC/WAL tail, change stream,
+    queue subscription) must run on exactly ONE worker or every rank
+    re-ingests the same rows (reference: connectors run per-worker only
+    when the source itself shards — kafka partitions, file path hashes).
+    Pass `source` to elect worker 0 and finish the source on the others;
+    `sharded=True` marks readers that do their own worker assignment."""
+    if source is not None and not sharded:
+        from pathway_amd import parallel as par
+
+        comm = par.get_comm()
+        if comm is not None and comm.world > 1 and comm.rank != 0:
+            source.finish()
+            return None
     th = threading.Thread(target=fn, daemon=True)
     th.start()
     return th

@@ -50,6 +50,6 @@ def read(rdkafka_settings: dict, topic_name: str, *, schema=None, autocommit_dur
         finally:
             src.finish()
 
-    spawn_reader(reader)
+    spawn_reader(reader, src)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())

@@ -85,7 +85,7 @@ def _read_streaming(
         schema, "streaming", with_metadata,
         refresh_interval=refresh_interval, max_polls=max_polls,
     )
-    spawn_reader(reader.run)
+    spawn_reader(reader.run, src, sharded=True)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 

@@ -197,7 +197,7 @@ def read(
         finally:
             src.finish()
 
-    spawn_reader(reader)
+    spawn_reader(reader, src)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 

@@ -81,8 +81,17 @@ class KafkaReader:
             self.settings.get("auto.offset.reset", "beginning")
         ).lower()
         parts: list[tuple[str, int]] = []
+        from pathway_amd import parallel as par
+
+        comm = par.get_comm()
+        world = comm.world if comm is not None else 1
+        rank = comm.rank if comm is not None else 0
         for t in self.topics:
             for p in client.partitions(t):
+                # multi-worker partition assignment (reference kafka.rs:
+                # each partition is consumed by exactly one worker)
+                if world > 1 and p % world != rank:
+                    continue
                 parts.append((t, p))
                 if (t, p) not in self.offsets:
                     if self.start_from_timestamp_ms is not None:
@@ -273,7 +282,7 @@ def read(
         max_polls=_max_polls,
     )
     src.reader = reader  # offsets exposed for persistence metadata
-    spawn_reader(reader.run)
+    spawn_reader(reader.run, src, sharded=True)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 

@@ -114,7 +114,7 @@ def read(
     reader = KinesisReader(src, _Api(endpoint, region), stream_name, parse,
                            max_polls=_max_polls)
     src.reader = reader
-    spawn_reader(reader.run)
+    spawn_reader(reader.run, src)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 

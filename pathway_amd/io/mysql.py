@@ -145,7 +145,7 @@ def read(
         max_polls=_max_polls, primary_key=primary_key,
     )
     src.reader = reader
-    spawn_reader(reader.run)
+    spawn_reader(reader.run, src)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 

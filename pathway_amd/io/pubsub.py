@@ -104,7 +104,7 @@ def read(
     reader = PubSubReader(src, base_url, project_id, subscription, parse,
                           _headers(credentials), max_polls=_max_polls)
     src.reader = reader
-    spawn_reader(reader.run)
+    spawn_reader(reader.run, src)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 

@@ -158,7 +158,7 @@ def read(
         max_messages=_max_messages,
     )
     src.reader = reader
-    spawn_reader(reader.run)
+    spawn_reader(reader.run, src)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 

@@ -114,6 +114,6 @@ def read(
         with_metadata=with_metadata,
     )
     src.reader = reader
-    spawn_reader(reader.run)
+    spawn_reader(reader.run, src)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())

@@ -108,7 +108,7 @@ def read(
         finally:
             src.finish()
 
-    spawn_reader(run_subject)
+    spawn_reader(run_subject, src)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 

@@ -97,7 +97,7 @@ def read(
     reader = RabbitReader(src, uri, queue_name, parse,
                           max_messages=_max_messages)
     src.reader = reader
-    spawn_reader(reader.run)
+    spawn_reader(reader.run, src)
     node = InputNode(src, get_device())
     return Table(node, {n: d for n, d in zip(names, dtypes)}, Universe())
 

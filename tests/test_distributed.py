@@ -626,3 +626,78 @@ def test_distributed_fs_file_sharding(tmp_path):
     assert sorted(results[0] + results[1]) == [f"line-{i}" for i in range(8)]
     assert not (set(results[0]) & set(results[1]))
     assert results[0] and results[1]  # both ranks got a share
+
+
+def _worker_single_stream(rank: int, world: int, port: int, nats_uri: str, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["PW_DEVICE"] = "cpu"
+    import pathway_amd as pw
+    import pathway_amd.parallel as par
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.internals.schema import schema_from_types
+
+    par.init(backend="gloo")
+    G.clear()
+    t = pw.io.nats.read(
+        nats_uri, "live", schema=schema_from_types(k=int), format="json",
+        _max_messages=2,
+    )
+    keys, cols = pw.debug.table_to_dicts(t)
+    q.put((rank, sorted(cols["k"].values())))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_distributed_single_stream_reader_election(tmp_path):
+    """A single-stream connector (NATS subscription) must be consumed by
+    exactly one worker — otherwise every rank re-ingests the stream
+    (at-most-once pub/sub would also double-deliver)."""
+    import json
+    import threading
+    import time
+
+    from tests.fakes.fake_nats import FakeNats
+    from pathway_amd.io.nats import NatsClient
+
+    srv = FakeNats().start()
+    try:
+        def later():
+            deadline = time.time() + 30
+            while time.time() < deadline and not srv.subs.get("live"):
+                time.sleep(0.01)
+            time.sleep(0.2)  # both ranks' readers would be up by now
+            c = NatsClient(srv.uri)
+            c.publish("live", json.dumps({"k": 5}).encode())
+            c.publish("live", json.dumps({"k": 7}).encode())
+            c.close()
+
+        threading.Thread(target=later, daemon=True).start()
+        ctx = mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        port = _free_port()
+        procs = [
+            ctx.Process(
+                target=_worker_single_stream, args=(r, 2, port, srv.uri, q)
+            )
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(2):
+            rank, rows = q.get()
+            results[rank] = rows
+        for p in procs:
+            p.join(60)
+            assert p.exitcode == 0
+        # exactly once: rank 0 has the stream, rank 1 has nothing
+        assert results[0] == [5, 7]
+        assert results[1] == []
+        assert len(srv.subs.get("live") or []) <= 1
+    finally:
+        srv.stop()
