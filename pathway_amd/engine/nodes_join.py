@@ -667,9 +667,13 @@ def _scatter_override(base: Column, qidx: torch.Tensor, repl: Column) -> Column:
 class FlattenNode(Node):
     """flatten_table: explode a sequence column into rows."""
 
-    def __init__(self, input_node: Node, flatten_name: str, device):
+    def __init__(
+        self, input_node: Node, flatten_name: str, device,
+        origin_id: str | None = None,
+    ):
         super().__init__([input_node], device)
         self.flatten_name = flatten_name
+        self.origin_id = origin_id
         self.salt = _salt("flatten", self.node_id)
 
     def step(self, time, inputs):
@@ -702,6 +706,10 @@ class FlattenNode(Node):
         fcol, _ = infer_and_build_column(flat_vals, device)
         cols = dict(base.columns)
         cols[self.flatten_name] = fcol
+        if self.origin_id:
+            from pathway_amd.engine.column import PointerColumn
+
+            cols[self.origin_id] = PointerColumn(base.keys.clone())
         ords = torch.tensor(ordinals, dtype=torch.int64, device=device)
         olo, ohi = hashing.value_hash_words(ords, 2)
         lo, hi = hashing.derive_key_words(

@@ -241,7 +241,12 @@ class Table(TableLike):
         return pos, neg
 
     def copy(self) -> "Table":
-        return Table(self._node, self._dtypes, self._universe)
+        # a DISTINCT node (same keys/values): self-joins resolve join
+        # sides by node identity, so t.join(t.copy(), ...) must see two
+        # different nodes (reference Table.copy gives a fresh table)
+        return self.select(
+            **{n: ex.ColumnReference(self, n) for n in self._dtypes}
+        )
 
     # -- renames / drops ---------------------------------------------------
 
@@ -509,8 +514,10 @@ class Table(TableLike):
         ref = self._resolve(to_flatten)
         if not isinstance(ref, ex.ColumnReference):
             raise ValueError("flatten expects a column reference")
-        node = FlattenNode(self._node, ref.name, get_device())
+        node = FlattenNode(self._node, ref.name, get_device(), origin_id=origin_id)
         dtypes = dict(self._dtypes)
+        if origin_id:
+            dtypes[origin_id] = dt.POINTER
         inner = self._dtypes.get(ref.name, dt.ANY)
         if isinstance(inner, dt.List):
             dtypes[ref.name] = inner.wrapped
