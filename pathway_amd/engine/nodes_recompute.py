@@ -22,6 +22,21 @@ from pathway_amd.engine.nodes import Node, consolidate_batch
 from pathway_amd.engine.nodes_join import _SideStore
 
 
+def _hashable(v):
+    """Stable hashable identity for a row value (ndarrays by bytes)."""
+    import numpy as np
+
+    if isinstance(v, np.ndarray):
+        return ("__nd__", v.shape, v.dtype.str, v.tobytes())
+    if isinstance(v, (list, dict, set)):
+        return repr(v)
+    try:
+        hash(v)
+        return v
+    except TypeError:
+        return repr(v)
+
+
 class RecomputeNode(Node):
     """fn(inputs: list[list[dict]], keys: list[list[Pointer]]) ->
     list[(Pointer, dict values)] — full output; node emits deltas."""
@@ -72,7 +87,7 @@ class RecomputeNode(Node):
                     w = lvl.weights.cpu().tolist()
                     names = list(cols.keys())
                     for i in range(len(ids)):
-                        row = tuple((n, cols[n][i]) for n in names)
+                        row = tuple((n, _hashable(cols[n][i])) for n in names)
                         k = (repr(ids[i]), row)
                         acc[k] += w[i]
                         payload[k] = (ids[i], {n: cols[n][i] for n in names})
@@ -89,22 +104,28 @@ class RecomputeNode(Node):
         new_norm = [
             (key, tuple(vals[n] for n in self.out_columns)) for key, vals in new_output
         ]
-        # diff vs previous
+        # diff vs previous (identities must be hashable: ndarray values
+        # are keyed by their bytes)
         from collections import Counter
 
-        old_c = Counter((repr(k), v) for k, v in self.prev_output)
-        new_c = Counter((repr(k), v) for k, v in new_norm)
+        def _ident(k, v):
+            return (repr(k), tuple(_hashable(x) for x in v))
+
+        old_c = Counter(_ident(k, v) for k, v in self.prev_output)
+        new_c = Counter(_ident(k, v) for k, v in new_norm)
         key_by_repr = {repr(k): k for k, _ in self.prev_output}
         key_by_repr.update({repr(k): k for k, _ in new_norm})
+        val_by_ident = {_ident(k, v): v for k, v in self.prev_output}
+        val_by_ident.update({_ident(k, v): v for k, v in new_norm})
         out_keys = []
         out_vals = []
         out_diffs = []
         for item in set(old_c) | set(new_c):
             d = new_c.get(item, 0) - old_c.get(item, 0)
             if d != 0:
-                krepr, vals = item
+                krepr, _vid = item
                 out_keys.append(key_by_repr[krepr])
-                out_vals.append(vals)
+                out_vals.append(val_by_ident[item])
                 out_diffs.append(d)
         self.prev_output = new_norm
         if not out_keys:

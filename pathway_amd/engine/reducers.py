@@ -216,6 +216,25 @@ _multiset(
     lambda args_dt: dt.Array(),
     n_args=2,
 )
+_multiset(
+    # sum over non-numeric-column values (ndarrays etc. — reference
+    # Reducer::ArraySum, reduce.rs); weight-scaled so retractions cancel
+    "array_sum",
+    None,
+    lambda rows: _array_sum_host(rows),
+    lambda args_dt: dt.unoptionalize(args_dt[0]),
+)
+
+
+def _array_sum_host(rows):
+    tot = None
+    for r in rows:
+        v, w = r[0][0], r[1]
+        if v is None:
+            continue
+        contrib = v * w
+        tot = contrib if tot is None else tot + contrib
+    return tot
 
 
 def _unique_host(rows):

@@ -69,6 +69,14 @@ class GroupedTable:
                     for a in e._args
                 ]
                 # engine arg augmentation
+                if rname == "sum" and len(rargs) == 1:
+                    # non-numeric sums (ndarray columns, reference
+                    # Reducer::ArraySum) take the host multiset family
+                    d = dt.unoptionalize(infer_dtype(rargs[0], table._dtypes))
+                    if d not in (
+                        dt.INT, dt.FLOAT, dt.BOOL, dt.DURATION,
+                    ):
+                        rname = "array_sum"
                 if rname in ("argmin", "argmax") and len(rargs) == 1:
                     rargs = [rargs[0], ex.ColumnReference(table, "id")]
                 elif rname in ("earliest", "latest", "tuple", "ndarray") and len(rargs) == 1:

@@ -240,6 +240,13 @@ class Table(TableLike):
         neg = self.filter(~ex.wrap_expr(self._resolve(split_expression)))
         return pos, neg
 
+    def __add__(self, other: "Table") -> "Table":
+        """Column-wise union of two same-universe tables (reference
+        Table.__add__); right-hand columns win on name clash."""
+        exprs = {n: ex.ColumnReference(self, n) for n in self._dtypes}
+        exprs.update({n: ex.ColumnReference(other, n) for n in other._dtypes})
+        return self.select(**exprs)
+
     def copy(self) -> "Table":
         # a DISTINCT node (same keys/values): self-joins resolve join
         # sides by node identity, so t.join(t.copy(), ...) must see two
