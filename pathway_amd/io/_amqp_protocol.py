@@ -189,6 +189,18 @@ class AmqpClient:
             return rk, body
 
     def close(self) -> None:
+        # half-close + drain: closing with unread inbound data RSTs the
+        # connection and can destroy still-buffered outbound frames
+        # server-side (see io/nats.py close)
+        import socket as _socket
+
+        try:
+            self.sock.shutdown(_socket.SHUT_WR)
+            self.sock.settimeout(0.25)
+            while self.sock.recv(65536):
+                pass
+        except OSError:
+            pass
         try:
             self.sock.close()
         except OSError:

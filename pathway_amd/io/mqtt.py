@@ -109,6 +109,18 @@ class MqttClient:
     def close(self) -> None:
         try:
             self._send_packet(DISCONNECT, 0, b"")
+        except OSError:
+            pass
+        import socket as _socket
+
+        try:
+            self.sock.shutdown(_socket.SHUT_WR)
+            self.sock.settimeout(0.25)
+            while self.sock.recv(65536):
+                pass
+        except OSError:
+            pass
+        try:
             self.sock.close()
         except OSError:
             pass

@@ -91,6 +91,17 @@ class NatsClient:
             # +OK / PONG / INFO: ignore
 
     def close(self) -> None:
+        # orderly shutdown: if the server sent anything we never read
+        # (+OK in verbose mode, INFO updates), closing with unread data
+        # RSTs the connection and can destroy our still-buffered
+        # outbound frames server-side — half-close and drain first
+        try:
+            self.sock.shutdown(socket.SHUT_WR)
+            self.sock.settimeout(0.25)
+            while self.sock.recv(65536):
+                pass
+        except OSError:
+            pass
         try:
             self.sock.close()
         except OSError:

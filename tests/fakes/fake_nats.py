@@ -35,8 +35,19 @@ class FakeNats:
                             break
                         line = line.rstrip(b"\r\n")
                         if line.upper().startswith(b"CONNECT"):
-                            self.wfile.write(b"+OK\r\n")
-                            self.wfile.flush()
+                            # +OK only in verbose mode (NATS protocol).
+                            # An unconditional +OK sits unread in a
+                            # publish-only client's receive buffer, so
+                            # its close() turns into a TCP RST that can
+                            # DESTROY still-buffered PUB frames server-
+                            # side (observed: second publish lost).
+                            try:
+                                opts = json.loads(line[7:].strip() or b"{}")
+                            except ValueError:
+                                opts = {}
+                            if opts.get("verbose"):
+                                self.wfile.write(b"+OK\r\n")
+                                self.wfile.flush()
                         elif line.upper() == b"PING":
                             self.wfile.write(b"PONG\r\n")
                             self.wfile.flush()
