@@ -240,6 +240,13 @@ class Table(TableLike):
         neg = self.filter(~ex.wrap_expr(self._resolve(split_expression)))
         return pos, neg
 
+    def eval_type(self, expression: Any) -> dt.DType:
+        """Static dtype of an expression over this table (reference
+        table.py eval_type)."""
+        from pathway_amd.internals.type_inference import infer_dtype
+
+        return infer_dtype(self._resolve(expression), self._dtypes)
+
     def __add__(self, other: "Table") -> "Table":
         """Column-wise union of two same-universe tables (reference
         Table.__add__); right-hand columns win on name clash."""

@@ -50,8 +50,22 @@ from pathway_amd.io import (
     slack,
     weaviate,
 )
-from pathway_amd.io._subscribe import subscribe
-from pathway_amd.io.synchronization import register_input_synchronization_group
+from pathway_amd.io._subscribe import (
+    OnChangeCallback,
+    OnChangeCallbackAsync,
+    OnFinishCallback,
+    subscribe,
+)
+from pathway_amd.io._utils import (
+    ENGINE_TIME,
+    CsvParserSettings,
+    DurationLike,
+    TLSSettings,
+)
+from pathway_amd.io.synchronization import (
+    SynchronizedColumn,
+    register_input_synchronization_group,
+)
 
 __all__ = [
     "airbyte", "azure", "bigquery", "clickhouse", "csv", "debezium",
@@ -60,4 +74,7 @@ __all__ = [
     "mongodb", "mqtt", "mssql", "mysql", "nats", "null", "plaintext",
     "postgres", "pubsub", "pulsar", "python", "questdb", "rabbitmq",
     "redpanda", "s3", "sqlite", "chroma", "duckdb", "leann", "milvus", "pinecone", "pyfilesystem", "qdrant", "slack", "weaviate", "subscribe", "register_input_synchronization_group",
+    "SynchronizedColumn", "ENGINE_TIME", "CsvParserSettings",
+    "DurationLike", "TLSSettings", "OnChangeCallback",
+    "OnChangeCallbackAsync", "OnFinishCallback",
 ]

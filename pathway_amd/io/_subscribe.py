@@ -7,6 +7,13 @@ from pathway_amd.engine.runtime import SubscribeNode
 from pathway_amd.internals.config import get_device
 from pathway_amd.internals.rungraph import G
 
+#: on_change(key, row, time, is_addition) — sync and async variants
+#: (reference internals/api OnChangeCallback / OnChangeCallbackAsync)
+OnChangeCallback = Callable[..., None]
+OnChangeCallbackAsync = Callable[..., Any]
+#: on_end() — called when the stream finishes
+OnFinishCallback = Callable[[], None]
+
 
 def subscribe(
     table,
