@@ -101,6 +101,39 @@ class DateTimeNamespace:
     def from_timestamp(self, unit: str):
         return self._m("from_timestamp", unit, return_type=dt.DATE_TIME_NAIVE)
 
+    def utc_from_timestamp(self, unit: str):
+        return self._m(
+            "utc_from_timestamp", unit, return_type=dt.DATE_TIME_UTC
+        )
+
+    def to_duration(self, unit: Any = "ns"):
+        return self._m("to_duration", unit, return_type=dt.DURATION)
+
+    def weeks(self):
+        return self._m("weeks", return_type=dt.INT)
+
+    def add_duration_in_timezone(self, duration: Any, timezone: Any):
+        """Wall-clock addition in a time zone (DST-aware; reference
+        date_time.py:855)."""
+        return self._m(
+            "add_duration_in_timezone", duration, timezone,
+            return_type=dt.DATE_TIME_NAIVE,
+        )
+
+    def subtract_duration_in_timezone(self, duration: Any, timezone: Any):
+        return self._m(
+            "subtract_duration_in_timezone", duration, timezone,
+            return_type=dt.DATE_TIME_NAIVE,
+        )
+
+    def subtract_date_time_in_timezone(self, date_time: Any, timezone: Any):
+        """a - b computed through the zone's wall clock (DST-aware;
+        reference date_time.py:943)."""
+        return self._m(
+            "subtract_date_time_in_timezone", date_time, timezone,
+            return_type=dt.DURATION,
+        )
+
     def utc_now(self):
         raise NotImplementedError("dt.utc_now is a table stream op; see pw.io")
 
@@ -239,4 +272,76 @@ register_method(
 register_method(
     "bin.base64_decode",
     _host_method(lambda s: __import__("base64").b64decode(s), dt.BYTES),
+)
+
+
+_UNIT_NS = {
+    "W": 7 * 24 * 3600 * 10**9,
+    "D": 24 * 3600 * 10**9, "day": 24 * 3600 * 10**9, "days": 24 * 3600 * 10**9,
+    "h": 3600 * 10**9, "hr": 3600 * 10**9, "hour": 3600 * 10**9, "hours": 3600 * 10**9,
+    "m": 60 * 10**9, "min": 60 * 10**9, "minute": 60 * 10**9, "minutes": 60 * 10**9,
+    "s": 10**9, "sec": 10**9, "second": 10**9, "seconds": 10**9,
+    "ms": 10**6, "millisecond": 10**6, "milliseconds": 10**6, "millis": 10**6, "milli": 10**6,
+    "us": 10**3,
+    "ns": 1, "nano": 1, "nanos": 1, "nanosecond": 1, "nanoseconds": 1,
+}
+
+
+def _to_duration(v, unit):
+    from pathway_amd.internals.datetime_types import Duration
+
+    return Duration(int(v) * _UNIT_NS[unit], unit="ns")
+
+
+def _utc_from_timestamp(v, unit):
+    import pandas as pd
+
+    from pathway_amd.internals.datetime_types import DateTimeUtc
+
+    mul = {"s": 10**9, "ms": 10**6, "us": 10**3, "ns": 1}[unit]
+    return DateTimeUtc(pd.Timestamp(int(v * mul), unit="ns", tz="UTC"))
+
+
+def _weeks(v):
+    import pandas as pd
+
+    return int(pd.Timedelta(v).value // (7 * 24 * 3600 * 10**9))
+
+
+def _add_dur_tz(v, dur, tz, sign=1):
+    import pandas as pd
+
+    from pathway_amd.internals.datetime_types import DateTimeNaive
+
+    aware = pd.Timestamp(v).tz_localize(tz, ambiguous=True)
+    res = aware + sign * pd.Timedelta(dur)
+    return DateTimeNaive(res.tz_convert(tz).tz_localize(None))
+
+
+def _sub_dt_tz(a, b, tz):
+    import pandas as pd
+
+    from pathway_amd.internals.datetime_types import Duration
+
+    aa = pd.Timestamp(a).tz_localize(tz, ambiguous=True)
+    bb = pd.Timestamp(b).tz_localize(tz, ambiguous=True)
+    return Duration((aa - bb).value, unit="ns")
+
+
+register_method("dt.to_duration", _host_method(_to_duration, dt.DURATION))
+register_method(
+    "dt.utc_from_timestamp", _host_method(_utc_from_timestamp, dt.DATE_TIME_UTC)
+)
+register_method("dt.weeks", _host_method(_weeks, dt.INT))
+register_method(
+    "dt.add_duration_in_timezone",
+    _host_method(_add_dur_tz, dt.DATE_TIME_NAIVE),
+)
+register_method(
+    "dt.subtract_duration_in_timezone",
+    _host_method(lambda v, d, tz: _add_dur_tz(v, d, tz, sign=-1), dt.DATE_TIME_NAIVE),
+)
+register_method(
+    "dt.subtract_date_time_in_timezone",
+    _host_method(_sub_dt_tz, dt.DURATION),
 )

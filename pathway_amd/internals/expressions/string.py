@@ -70,6 +70,9 @@ class StringNamespace:
     def swap_case(self):
         return self._m("swapcase", return_type=dt.STR)
 
+    # reference name (string.py swapcase); swap_case kept as an alias
+    swapcase = swap_case
+
     def ljust(self, width: Any, fillchar: Any = None):
         args = [width] + ([fillchar] if fillchar is not None else [])
         return self._m("ljust", *args, return_type=dt.STR)

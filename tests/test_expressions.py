@@ -298,3 +298,67 @@ def test_int_float_arithmetic_matrix():
     get = lambda n: next(iter(cols[n].values()))
     assert (get("q"), get("r"), get("p")) == (3, 1, 49)
     assert get("tdiv") == 3.5 and get("neg") == -7 and get("xor") == 5
+
+
+def test_dt_timezone_aware_arithmetic_and_conversions():
+    """Reference date_time.py:855-1600: wall-clock add/subtract in a time
+    zone (DST-aware), to_duration, weeks, utc_from_timestamp, swapcase."""
+    import pandas as pd
+
+    from pathway_amd.internals.rungraph import G
+
+    G.clear()
+    t = T(
+        """
+        s   | n
+        AbC | 3
+        """
+    )
+    r = t.select(
+        sw=t.s.str.swapcase(),
+        dur=t.n.dt.to_duration("h"),
+        wk=pw.this.n.dt.to_duration("W").dt.weeks(),
+        utc=t.n.dt.utc_from_timestamp("s"),
+    )
+    _k, c = table_to_dicts(r)
+    assert list(c["sw"].values())[0] == "aBc"
+    assert pd.Timedelta(list(c["dur"].values())[0]) == pd.Timedelta(hours=3)
+    assert list(c["wk"].values())[0] == 3
+    assert pd.Timestamp(list(c["utc"].values())[0]) == pd.Timestamp(
+        3, unit="s", tz="UTC"
+    )
+    G.clear()
+    t2 = T(
+        """
+        x
+        1
+        """
+    )
+    r2 = t2.select(
+        a=pw.apply_with_type(
+            lambda _: pd.Timestamp("2024-03-10 01:30"),
+            pw.DateTimeNaive,
+            pw.this.x,
+        )
+    ).select(
+        plus=pw.this.a.dt.add_duration_in_timezone(
+            pd.Timedelta(hours=1), "America/New_York"
+        ),
+        minus=pw.this.a.dt.subtract_duration_in_timezone(
+            pd.Timedelta(hours=1), "America/New_York"
+        ),
+        diff=pw.this.a.dt.subtract_date_time_in_timezone(
+            pd.Timestamp("2024-03-10 03:30"), "America/New_York"
+        ),
+    )
+    _k2, c2 = table_to_dicts(r2)
+    # 2024-03-10 02:00 does not exist in New York (DST spring forward):
+    # one wall-clock hour past 01:30 lands at 03:30
+    assert pd.Timestamp(list(c2["plus"].values())[0]) == pd.Timestamp(
+        "2024-03-10 03:30"
+    )
+    assert pd.Timestamp(list(c2["minus"].values())[0]) == pd.Timestamp(
+        "2024-03-10 00:30"
+    )
+    # 01:30 -> 03:30 spans ONE absolute hour across the gap
+    assert pd.Timedelta(list(c2["diff"].values())[0]) == pd.Timedelta(hours=-1)
