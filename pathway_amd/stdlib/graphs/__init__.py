@@ -1,6 +1,12 @@
 """Graph algorithms on pw.iterate (reference stdlib/graphs/)."""
 from pathway_amd.stdlib.graphs import graph
-from pathway_amd.stdlib.graphs.common import Edge, Vertex
+from pathway_amd.stdlib.graphs.common import (
+    Cluster,
+    Clustering,
+    Edge,
+    Vertex,
+    Weight,
+)
 from pathway_amd.stdlib.graphs.graph import Graph, WeightedGraph
 from pathway_amd.stdlib.graphs.pagerank import pagerank
 from pathway_amd.stdlib.graphs.bellman_ford import bellman_ford

@@ -14,6 +14,22 @@ class Edge(schema.Schema):
     v: Pointer
 
 
+class Weight(schema.Schema):
+    """Weight extension of Vertex / Edge (reference common.py:23)."""
+
+    weight: float
+
+
+class Cluster(Vertex):
+    pass
+
+
+class Clustering(schema.Schema):
+    """Vertex (id) -> cluster c membership (reference common.py:35)."""
+
+    c: Pointer
+
+
 class Graph:
     def __init__(self, V, E):
         self.V = V
