@@ -33,6 +33,10 @@ class RunStats:
     rows_output: int = 0
     step_latencies_ms: list = field(default_factory=list)
     operators: dict = field(default_factory=lambda: defaultdict(OperatorStats))
+    exchanges: int = 0
+    exchange_rows: int = 0
+    #: max-destination load over mean (1.0 = perfectly balanced); EWMA
+    exchange_skew: float = 1.0
 
     def record_step(self, t: int, latency_s: float, ingested: int, output: int):
         self.current_time = t
@@ -42,6 +46,13 @@ class RunStats:
         self.step_latencies_ms.append(latency_s * 1000)
         if len(self.step_latencies_ms) > 1000:
             self.step_latencies_ms = self.step_latencies_ms[-1000:]
+
+    def observe_exchange(self, rows: int, skew_ratio: float) -> None:
+        """Shard skew telemetry: max/mean destination load of
+        one all-to-all partition, EWMA-smoothed."""
+        self.exchanges += 1
+        self.exchange_rows += rows
+        self.exchange_skew = 0.9 * self.exchange_skew + 0.1 * skew_ratio
 
     def latency_quantile(self, q: float) -> float | None:
         if not self.step_latencies_ms:
@@ -58,6 +69,9 @@ class RunStats:
             "rows_output": self.rows_output,
             "p50_step_ms": self.latency_quantile(0.5),
             "p95_step_ms": self.latency_quantile(0.95),
+            "exchanges": self.exchanges,
+            "exchange_rows": self.exchange_rows,
+            "exchange_skew": round(self.exchange_skew, 4),
             "operators": {
                 name: vars(st) for name, st in self.operators.items()
             },
@@ -72,6 +86,8 @@ class RunStats:
             f"pathway_rows_ingested_total {s['rows_ingested']}",
             "# TYPE pathway_rows_output counter",
             f"pathway_rows_output_total {s['rows_output']}",
+            "# TYPE pathway_exchange_skew gauge",
+            f"pathway_exchange_skew {s['exchange_skew']}",
             "# TYPE pathway_step_latency_ms gauge",
             f"pathway_step_latency_ms{{quantile=\"0.5\"}} {s['p50_step_ms'] or 0}",
             f"pathway_step_latency_ms{{quantile=\"0.95\"}} {s['p95_step_ms'] or 0}",
