@@ -250,6 +250,11 @@ class IvfFlatState(FlatIndexState):
             outs.append(sc.argmax(1))
         return torch.cat(outs)
 
+    #: rerank working-set budget (bytes of gathered candidate vectors);
+    #: bounds the (chunk, width, d) transient so a 10M-row index at high
+    #: nprobe stays in a few GB instead of a ~100 GB burst allocation
+    RERANK_BUDGET_BYTES = 2 << 30
+
     def search(self, q: torch.Tensor, k: int, filter_fns=None):
         nq = q.shape[0]
         m = self.keys.shape[0]
@@ -261,13 +266,57 @@ class IvfFlatState(FlatIndexState):
         cs = self._scores_against(q, self.centroids)
         nprobe = min(self.nprobe, self.centroids.shape[0])
         _, probe = torch.topk(cs, nprobe, dim=1)  # (nq, nprobe)
-        # 2. candidate gather — ragged per query; flatten with offsets
-        starts = self.list_offsets.index_select(0, probe.reshape(-1))
-        ends = self.list_offsets.index_select(0, probe.reshape(-1) + 1)
-        lens = (ends - starts).reshape(nq, nprobe)
+        # 2. query chunking by rerank budget
+        starts_all = self.list_offsets.index_select(0, probe.reshape(-1))
+        ends_all = self.list_offsets.index_select(0, probe.reshape(-1) + 1)
+        lens_all = (ends_all - starts_all).reshape(nq, nprobe)
+        maxc_all = int(lens_all.sum(1).max().item()) if nq else 0
+        tail = m - self.clustered
+        d = self.vectors.shape[1]
+        row_bytes = max((maxc_all + tail) * d * self.vectors.element_size(), 1)
+        qchunk = max(1, int(self.RERANK_BUDGET_BYTES // row_bytes))
+        if qchunk < nq:
+            parts = [
+                self._search_clustered(
+                    q[i : i + qchunk], probe[i : i + qchunk],
+                    lens_all[i : i + qchunk],
+                    starts_all.reshape(nq, nprobe)[i : i + qchunk], k, tail,
+                )
+                for i in range(0, nq, qchunk)
+            ]
+            kkmax = max(p[1].shape[1] for p in parts)
+            padded = []
+            for ids_p, sc_p, va_p in parts:
+                pad = kkmax - sc_p.shape[1]
+                if pad:
+                    nqp = sc_p.shape[0]
+                    ids_p = torch.cat(
+                        [ids_p, torch.zeros(nqp, pad, 2, dtype=ids_p.dtype,
+                                            device=ids_p.device)], dim=1)
+                    sc_p = torch.cat(
+                        [sc_p, torch.full((nqp, pad), float("-inf"),
+                                          dtype=sc_p.dtype,
+                                          device=sc_p.device)], dim=1)
+                    va_p = torch.cat(
+                        [va_p, torch.zeros(nqp, pad, dtype=torch.bool,
+                                           device=va_p.device)], dim=1)
+                padded.append((ids_p, sc_p, va_p))
+            return (
+                torch.cat([p[0] for p in padded]),
+                torch.cat([p[1] for p in padded]),
+                torch.cat([p[2] for p in padded]),
+            )
+        return self._search_clustered(
+            q, probe, lens_all, starts_all.reshape(nq, nprobe), k, tail
+        )
+
+    def _search_clustered(self, q, probe, lens, starts2, k, tail):
+        nq = q.shape[0]
+        m = self.keys.shape[0]
+        nprobe = probe.shape[1]
+        starts = starts2.reshape(-1)
         total_per_q = lens.sum(1)  # (nq,)
         maxc = int(total_per_q.max().item()) if nq else 0
-        tail = m - self.clustered  # unclustered adds: always candidates
         width = maxc + tail
         if width == 0:
             return self._empty(nq)

@@ -164,3 +164,31 @@ def test_usearch_and_lsh_retrievers_end_to_end():
         # resolve back: the matched doc is 'beta' (exact self vector)
         dkeys, dcols = pw.debug.table_to_dicts(docs)
         assert dcols["doc"][ids[0]] == "beta"
+
+
+def test_ivf_chunked_rerank_equivalence():
+    """The budget-bounded rerank (RERANK_BUDGET_BYTES) must return
+    exactly the unchunked answers."""
+    import torch
+
+    from pathway_amd.engine.ann import IvfFlatState
+
+    torch.manual_seed(0)
+    n, d = 20000, 16
+    vecs = torch.randn(n, d)
+    keys = torch.stack(
+        [
+            torch.arange(1, n + 1, dtype=torch.int64),
+            torch.zeros(n, dtype=torch.int64),
+        ],
+        dim=1,
+    )
+    st = IvfFlatState("cpu", "cos", nprobe=8, min_train=1000)
+    st.update(keys, vecs, torch.ones(n, dtype=torch.int64))
+    q = torch.randn(64, d)
+    ids1, sc1, v1 = st.search(q, 10)
+    st.RERANK_BUDGET_BYTES = 1 << 14  # force many query chunks
+    ids2, sc2, v2 = st.search(q, 10)
+    assert torch.equal(ids1, ids2)
+    assert torch.allclose(sc1, sc2)
+    assert torch.equal(v1, v2)
