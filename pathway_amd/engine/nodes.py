@@ -187,8 +187,12 @@ class ExprMapNode(Node):
     """select / with_columns: evaluate expressions, keep keys.
 
     extra_inputs: same-universe tables referenced in the expressions
-    (reference: rowwise context over several universe-equal tables);
-    their batches are aligned to the main batch BY KEY before evaluation.
+    (reference: rowwise context over several universe-equal tables).
+    With extras the node is STATEFUL: per-key arrangements of the main
+    and extra tables, old-visible/merge/new-visible delta emission —
+    aligning against the other table's per-step delta is wrong under
+    retractions (the new main row would read the retracted old value)
+    and silent when only the extra side changes.
     """
 
     def __init__(
@@ -202,32 +206,78 @@ class ExprMapNode(Node):
         super().__init__([input_node] + [n for n, _ in extra_inputs], device)
         self.exprs = exprs
         self.extra_prefixes = [p for _, p in extra_inputs]
+        if self.extra_prefixes:
+            self.reset()
+
+    def reset(self) -> None:
+        if not self.extra_prefixes:
+            return
+        from pathway_amd.engine.nodes_join import _SideStore
+
+        self.main_store = _SideStore(self.device)
+        self.extra_stores = [_SideStore(self.device) for _ in self.extra_prefixes]
 
     def step(self, time, inputs):
         b = inputs[0]
-        if b is None or len(b) == 0:
+        if not self.extra_prefixes:
+            if b is None or len(b) == 0:
+                return None
+            ctx = EvalContext(b.columns, b.keys, self.device)
+            out_cols = {name: evaluate(e, ctx) for name, e in self.exprs.items()}
+            return DeltaBatch(b.keys, out_cols, b.diffs, time)
+        ebs = list(inputs[1:])
+        parts_in = [x for x in [b, *ebs] if x is not None and len(x)]
+        if not parts_in:
             return None
-        extra: dict[str, Column] = {}
-        if self.extra_prefixes:
-            from pathway_amd.engine.state import lex_sort_words, searchsorted_words
+        aff = unique_sorted_keys(
+            torch.cat([x.keys for x in parts_in], dim=0)
+        )
+        old = self._eval_visible(aff)
+        if b is not None and len(b):
+            self.main_store.merge(b.keys, b)
+        for st, eb in zip(self.extra_stores, ebs):
+            if eb is not None and len(eb):
+                st.merge(eb.keys, eb)
+        new = self._eval_visible(aff)
+        from pathway_amd.engine.nodes_join import _pad_delta
 
-            main_q = [b.keys[:, 0].contiguous(), b.keys[:, 1].contiguous()]
-            for prefix, eb in zip(self.extra_prefixes, inputs[1:]):
-                if eb is None or len(eb) == 0:
-                    raise RuntimeError(
-                        "same-universe referenced table produced no delta at a "
-                        "time its universe changed — universes are not equal"
+        parts = [p for p in _pad_delta(old, new, time, self.device) if p is not None and len(p)]
+        if not parts:
+            return None
+        return consolidate_batch(DeltaBatch.concat(parts))
+
+    def _eval_visible(self, aff: torch.Tensor):
+        """Evaluate the expressions over the CURRENT state rows of the
+        affected keys (None if no main rows)."""
+        from pathway_amd.engine.nodes_join import _scatter_override
+
+        cols, qidx, w = self.main_store.probe(aff)
+        if cols is None or qidx.shape[0] == 0:
+            return None
+        cols = dict(cols)
+        ids = cols.pop("__id__")
+        keys = ids.pairs
+        n = keys.shape[0]
+        extra: dict[str, Column] = {}
+        for st, prefix in zip(self.extra_stores, self.extra_prefixes):
+            ecols, eqidx, _ew = st.probe(keys)
+            if ecols is None:
+                continue
+            ecols = dict(ecols)
+            ecols.pop("__id__", None)
+            for cname, c in ecols.items():
+                if eqidx.shape[0] == n and bool(
+                    (eqidx == torch.arange(n, device=eqidx.device)).all()
+                ):
+                    extra[f"{prefix}{cname}"] = c
+                else:
+                    base = null_column(c, n, self.device)
+                    extra[f"{prefix}{cname}"] = _scatter_override(
+                        base, eqidx, c
                     )
-                ew = [eb.keys[:, 0].contiguous(), eb.keys[:, 1].contiguous()]
-                perm = lex_sort_words(ew)
-                sw = [w.index_select(0, perm) for w in ew]
-                pos = searchsorted_words(sw, main_q, side="left")
-                src = perm.index_select(0, pos.clamp(0, max(len(eb) - 1, 0)))
-                for n, c in eb.columns.items():
-                    extra[f"{prefix}{n}"] = c.take(src)
-        ctx = EvalContext(b.columns, b.keys, self.device, extra=extra)
+        ctx = EvalContext(cols, keys, self.device, extra=extra)
         out_cols = {name: evaluate(e, ctx) for name, e in self.exprs.items()}
-        return DeltaBatch(b.keys, out_cols, b.diffs, time)
+        return DeltaBatch(keys, out_cols, w, 0)
 
 
 class FilterNode(Node):

@@ -221,3 +221,38 @@ def test_by_hand_match_overrides():
     _k, cols = table_to_dicts(res)
     weights = sorted(cols["weight"].values())
     assert weights.count(99.0) == 2  # both identical pairs forced by hand
+
+
+def test_fuzzy_match_updates_incrementally():
+    """A retraction + new insert re-pairs the match (the node diffs its
+    output across engine times)."""
+    from pathway_amd.debug import table_from_rows
+    from pathway_amd.internals.schema import schema_builder, column_definition
+
+    G.clear()
+    schema = schema_builder(
+        {
+            "uid": column_definition(primary_key=True, dtype=int),
+            "name": column_definition(dtype=str),
+        }
+    )
+    # time 0: left matches "acme corp"; time 1: that right row is
+    # retracted and a better candidate arrives
+    rt_rows = [
+        (1, "acme corp", 0, 1),
+        (1, "acme corp", 1, -1),
+        (2, "acme corp holdings", 1, 1),
+    ]
+    lt = table_from_rows(
+        schema, [(10, "acme corp", 0, 1)], is_stream=True
+    )
+    rt = table_from_rows(schema, rt_rows, is_stream=True)
+    res = smart_fuzzy_match(lt.name, rt.name)
+    _k, cols = table_to_dicts(res)
+    # final state: exactly one pair, against the surviving right row
+    assert len(cols["left"]) == 1
+    _lk, lcols = table_to_dicts(lt)
+    _rk, rcols = table_to_dicts(rt)
+    rmap = {repr(k): v for k, v in rcols["name"].items()}
+    for k in cols["right"]:
+        assert rmap[repr(cols["right"][k])] == "acme corp holdings"
