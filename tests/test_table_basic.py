@@ -555,3 +555,38 @@ def test_sql_join():
             """
         ),
     )
+
+
+def test_select_with_ix_updates_under_retraction():
+    """Regression: a select combining own columns with an ix()-derived
+    column must re-evaluate from STATE when either side changes —
+    aligning against the other table's per-step delta read the retracted
+    old value (and missed extra-side-only changes)."""
+    from pathway_amd import reducers
+    from pathway_amd.debug import table_from_rows
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.internals.schema import column_definition, schema_builder
+    from pathway_amd.internals import thisclass
+
+    this = thisclass.this
+    G.clear()
+    schema = schema_builder(
+        {
+            "uid": column_definition(primary_key=True, dtype=int),
+            "g": column_definition(dtype=str),
+            "v": column_definition(dtype=int),
+        }
+    )
+    t = table_from_rows(
+        schema,
+        [(1, "a", 5, 0, 1), (2, "a", 9, 1, 1), (2, "a", 9, 2, -1)],
+        is_stream=True,
+    )
+    g = t.groupby(this.g).reduce(
+        g=this.g, best=reducers.argmax(this.v), w=reducers.max(this.v)
+    )
+    g2 = g.select(this.g, this.w, vv=t.ix(g.best).v)
+    keys, cols = pw.debug.table_to_dicts(g2)
+    # after insert(5) -> insert(9) -> retract(9): back to the uid=1 row
+    rows = [(cols["g"][k], cols["w"][k], cols["vv"][k]) for k in keys]
+    assert rows == [("a", 5, 5)]
