@@ -761,3 +761,53 @@ def test_fuzz_ivf_with_deletions_vs_brute():
     ids_f, sc_f, _ = flat.search(q, k)
     assert torch.allclose(sc_i, sc_f, atol=1e-5)
     assert torch.equal(ids_i[:, :, 0], ids_f[:, :, 0])
+
+
+def test_fuzz_stateful_select_with_foreign_columns():
+    """Randomized updates on two same-universe tables combined by one
+    select: final output must equal the oracle computed from final
+    states (stateful ExprMapNode path)."""
+    import random
+
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.internals.schema import column_definition, schema_builder
+    from pathway_amd.internals import thisclass
+    from pathway_amd.debug import table_from_rows, table_to_dicts
+
+    this = thisclass.this
+    rng = random.Random(99)
+    for trial in range(5):
+        G.clear()
+        schema = schema_builder(
+            {
+                "uid": column_definition(primary_key=True, dtype=int),
+                "v": column_definition(dtype=int),
+            }
+        )
+        # one shared universe: both tables keyed by the same uids; the
+        # second is derived (select) so universes match by construction
+        live: dict[int, int] = {}
+        events = []
+        t_now = 0
+        for step in range(60):
+            if step % 7 == 0:
+                t_now += 1
+            if live and rng.random() < 0.35:
+                uid = rng.choice(list(live))
+                events.append((uid, live.pop(uid), t_now, -1))
+            else:
+                uid = rng.randrange(8)
+                if uid in live:
+                    events.append((uid, live.pop(uid), t_now, -1))
+                v = rng.randrange(100)
+                live[uid] = v
+                events.append((uid, v, t_now, 1))
+        base = table_from_rows(schema, events, is_stream=True)
+        other = base.select(w=this.v * 10)
+        combined = base.select(this.uid, this.v, z=other.w + this.v)
+        _k, cols = table_to_dicts(combined)
+        got = sorted(
+            (cols["uid"][k], cols["v"][k], cols["z"][k]) for k in cols["uid"]
+        )
+        want = sorted((u, v, v * 10 + v) for u, v in live.items())
+        assert got == want, (trial, got, want)
