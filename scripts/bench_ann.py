@@ -21,6 +21,7 @@ def main():
     p.add_argument("--nq", type=int, default=1000)
     p.add_argument("--nprobe", type=int, default=16)
     p.add_argument("--centers", type=int, default=4096)
+    p.add_argument("--nlist", type=int, default=None)
     args = p.parse_args()
     assert torch.cuda.is_available()
     dev = "cuda"
@@ -43,7 +44,7 @@ def main():
         torch.zeros(n, dtype=torch.int64, device=dev),
     ], dim=1)
 
-    st = IvfFlatState(dev, "cos", nprobe=args.nprobe)
+    st = IvfFlatState(dev, "cos", nprobe=args.nprobe, nlist=args.nlist)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     st.update(keys, vecs, torch.ones(n, dtype=torch.int64, device=dev))
