@@ -203,7 +203,10 @@ class IvfFlatState(FlatIndexState):
             return self.nlist_cfg
         import math
 
-        return max(16, min(4096, int(math.sqrt(m) * 2)))
+        # 2*sqrt(m), capped at 16k: at 50M rows the 4096 cap left ~12k-row
+        # lists and rerank cost dominated (2.4k qps); 16384 lists measured
+        # 7.1k qps at the same recall (profiles/kernels_r02.md)
+        return max(16, min(16384, int(math.sqrt(m) * 2)))
 
     def _rebuild(self):
         m = self.keys.shape[0]
