@@ -701,3 +701,72 @@ def test_distributed_single_stream_reader_election(tmp_path):
         assert len(srv.subs.get("live") or []) <= 1
     finally:
         srv.stop()
+
+
+def _worker_kafka_parts(rank: int, world: int, port: int, bootstrap: str, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["PW_DEVICE"] = "cpu"
+    import pathway_amd as pw
+    import pathway_amd.parallel as par
+    from pathway_amd.internals.rungraph import G
+    from pathway_amd.internals.schema import schema_from_types
+
+    par.init(backend="gloo")
+    G.clear()
+    t = pw.io.kafka.read(
+        {"bootstrap.servers": bootstrap},
+        topic="events",
+        schema=schema_from_types(k=int),
+        format="json",
+        mode="static",
+    )
+    keys, cols = pw.debug.table_to_dicts(t)
+    q.put((rank, sorted(cols["k"].values())))
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_distributed_kafka_partition_assignment():
+    """Reference kafka.rs: each partition consumed by exactly one worker
+    (p % world == rank) — union complete, shares disjoint."""
+    import json
+
+    from tests.fakes.fake_kafka import FakeKafkaBroker
+    from pathway_amd.io._kafka_protocol import KafkaClient
+
+    b = FakeKafkaBroker(num_partitions=2).start()
+    try:
+        bootstrap = f"127.0.0.1:{b.port}"
+        c = KafkaClient(bootstrap)
+        for i in range(10):
+            c.produce("events", i % 2, [(None, json.dumps({"k": i}).encode())])
+        c.close()
+        ctx = mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        port = _free_port()
+        procs = [
+            ctx.Process(
+                target=_worker_kafka_parts, args=(r, 2, port, bootstrap, q)
+            )
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(2):
+            rank, rows = q.get()
+            results[rank] = rows
+        for p in procs:
+            p.join(60)
+            assert p.exitcode == 0
+        assert sorted(results[0] + results[1]) == list(range(10))
+        # partition p -> rank p % 2: even keys on rank 0, odd on rank 1
+        assert results[0] == [0, 2, 4, 6, 8]
+        assert results[1] == [1, 3, 5, 7, 9]
+    finally:
+        b.stop()
