@@ -50,10 +50,32 @@ def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
         # sorting by a strict prefix of a row's identity would break
         # silently; tests/test_fuzz_equivalence.py pins the property and
         # PW_DEBUG_SORT=1 verifies full lex order on every call.
-        if words[0].is_cuda and _PW_PW_SORT:
+        if words[0].is_cuda:
             from pathway_amd import ops
 
-            k0_sorted, perm0 = ops.radix_sort64_gpu(words[0].contiguous())
+            if _PW_PW_SORT:
+                k0_sorted, perm0 = ops.radix_sort64_gpu(words[0].contiguous())
+            else:
+                k0_sorted, perm0 = torch.sort(words[0])
+            if ops.lib_available():
+                # sync-free collision handling: odd-even repair of word1
+                # order inside equal-word0 runs (k_sort_repair).  The
+                # old host `.any()` check cost 0.27 ms/step in syncs —
+                # profiles/kernels_r02.md.
+                k1_sorted = words[1].index_select(0, perm0)
+                ops.sort_repair_gpu(k0_sorted, k1_sorted, perm0)
+                if _os.environ.get("PW_DEBUG_SORT"):
+                    ok = bool(
+                        (
+                            (k0_sorted[1:] > k0_sorted[:-1])
+                            | (
+                                (k0_sorted[1:] == k0_sorted[:-1])
+                                & (k1_sorted[1:] >= k1_sorted[:-1])
+                            )
+                        ).all()
+                    )
+                    assert ok, "sort repair produced non-lex order"
+                return perm0
         else:
             k0_sorted, perm0 = torch.sort(words[0])
         k1_sorted = words[1].index_select(0, perm0)
@@ -67,8 +89,6 @@ def lex_sort_words(words: Sequence[torch.Tensor]) -> torch.Tensor:
             ).any()
         )
         if not violation:
-            import os as _os
-
             if _os.environ.get("PW_DEBUG_SORT"):
                 ok = bool(
                     (

@@ -708,6 +708,31 @@ def scan_positions_gpu(buf: torch.Tensor, target: int) -> torch.Tensor:
     return out
 
 
+def sort_repair_gpu(
+    k0_sorted: torch.Tensor,
+    k1_sorted: torch.Tensor,
+    perm: torch.Tensor,
+    passes: int = 4,
+) -> None:
+    """In-place odd-even repair of word1 order inside equal-word0 runs
+    (k_sort_repair) — the sync-free replacement for the 2-word sort fast
+    path's collision check.  Mutates k1_sorted and perm."""
+    lib = require_lib()
+    n = k0_sorted.shape[0]
+    if n < 2:
+        return
+    rc = lib.pw_sort_repair(
+        ctypes.c_void_p(k0_sorted.data_ptr()),
+        ctypes.c_void_p(k1_sorted.data_ptr()),
+        ctypes.c_void_p(perm.data_ptr()),
+        ctypes.c_int64(n),
+        ctypes.c_int(int(passes)),
+        _stream_ptr(),
+    )
+    if rc != 0:
+        raise RuntimeError(f"pw_sort_repair failed: hip error {rc}")
+
+
 def gather_cols_gpu(idx: torch.Tensor, cols: list) -> list:
     """Gather up to 8 8-byte columns through one shared int64 index in a
     single launch (k_gather_cols) — replaces per-column index_select on

@@ -1550,6 +1550,44 @@ extern "C" int pw_scan_emit(const void* buf, int64_t n, int target,
   return (int)hipGetLastError();
 }
 
+// ------------------------------------------------------- sort repair --
+// After the 2-word fast-path sort by word0 only, rows in an equal-word0
+// run may be out of order on word1.  Host-sync-checking for collisions
+// cost 0.27 ms/step (10 syncs); instead: odd-even transposition passes
+// restricted to equal-k0 pairs, entirely on device.  `passes` bounds the
+// repairable run length; 64-bit hash keys make runs >8 over any real
+// row count astronomically improbable (n^9 / 2^512).  Signed compare
+// matches torch.sort / the multi-pass fallback.
+
+__global__ void k_sort_repair(const long long* __restrict__ k0,
+                              long long* k1, long long* perm, int64_t n,
+                              int parity) {
+  int64_t i = 2 * ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) + parity;
+  if (i + 1 >= n) return;
+  if (k0[i] != k0[i + 1]) return;
+  long long a = k1[i], b = k1[i + 1];
+  if (a > b) {
+    k1[i] = b;
+    k1[i + 1] = a;
+    long long p = perm[i];
+    perm[i] = perm[i + 1];
+    perm[i + 1] = p;
+  }
+}
+
+extern "C" int pw_sort_repair(const void* k0, void* k1, void* perm,
+                              int64_t n, int passes, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int64_t pairs = n / 2 + 1;
+  int64_t nblocks = (pairs + PW_BLOCK - 1) / PW_BLOCK;
+  if (nblocks < 1) nblocks = 1;
+  for (int p = 0; p < passes; ++p)
+    hipLaunchKernelGGL(k_sort_repair, dim3((uint32_t)nblocks), dim3(PW_BLOCK),
+                       0, s, (const long long*)k0, (long long*)k1,
+                       (long long*)perm, n, p & 1);
+  return (int)hipGetLastError();
+}
+
 // ------------------------------------------------------- fused gather --
 // One launch gathers up to 8 8-byte columns through a shared index: the
 // arrange/merge/consolidate paths gather (key words, weights, value

@@ -744,6 +744,35 @@ def test_scan_positions_kernel():
 
 @gpu
 @requires_cuda
+def test_sort_repair_collisions():
+    """lex_sort_words fast path with FORCED word0 collisions: the sync-free
+    odd-even repair (k_sort_repair) must restore full lex order, including
+    a 4-long equal-k0 run."""
+    import os
+
+    from pathway_amd.engine.state import lex_sort_words
+
+    torch.manual_seed(11)
+    n = 1 << 20
+    k0 = torch.randint(-(1 << 62), 1 << 62, (n,), device="cuda")
+    k1 = torch.randint(-(1 << 62), 1 << 62, (n,), device="cuda")
+    # collide: 200 duplicate pairs + one run of 4 equal k0
+    dup = torch.randint(0, n, (200,), device="cuda")
+    k0[dup] = k0[(dup + 1) % n]
+    k0[10:14] = k0[10]
+    os.environ["PW_DEBUG_SORT"] = "1"
+    try:
+        perm = lex_sort_words([k0, k1])  # debug flag asserts lex order
+    finally:
+        os.environ.pop("PW_DEBUG_SORT", None)
+    s0 = k0.index_select(0, perm)
+    s1 = k1.index_select(0, perm)
+    ok = (s0[1:] > s0[:-1]) | ((s0[1:] == s0[:-1]) & (s1[1:] >= s1[:-1]))
+    assert bool(ok.all())
+
+
+@gpu
+@requires_cuda
 def test_device_hash_table():
     from pathway_amd import ops
 
