@@ -240,6 +240,17 @@ class Table(TableLike):
         neg = self.filter(~ex.wrap_expr(self._resolve(split_expression)))
         return pos, neg
 
+    @property
+    def C(self) -> "_ColumnAccessor":  # noqa: N802
+        """Typed column accessor (reference Table.C): ``t.C.name`` is
+        ``t.name`` — useful when a column name collides with a Table
+        method."""
+        return _ColumnAccessor(self)
+
+    @property
+    def _C(self) -> "_ColumnAccessor":  # noqa: N802
+        return self.C
+
     def eval_type(self, expression: Any) -> dt.DType:
         """Static dtype of an expression over this table (reference
         table.py eval_type)."""
@@ -873,3 +884,21 @@ class Table(TableLike):
         from pathway_amd.engine.runtime import CaptureNode
 
         return CaptureNode(self._node, get_device(), list(self._dtypes.keys()))
+
+
+class _ColumnAccessor:
+    """`table.C.<name>` -> ColumnReference (reference Table.C)."""
+
+    def __init__(self, table: Table):
+        object.__setattr__(self, "_table", table)
+
+    def __getattr__(self, name: str):
+        t = object.__getattribute__(self, "_table")
+        if name == "id":
+            return ex.ColumnReference(t, "id")
+        if name not in t._dtypes:
+            raise AttributeError(f"no column {name!r}")
+        return ex.ColumnReference(t, name)
+
+    def __getitem__(self, name: str):
+        return self.__getattr__(name)
